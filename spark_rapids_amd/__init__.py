@@ -1,0 +1,16 @@
+"""spark_rapids_amd: MI355X-native columnar SQL engine with the capabilities
+of the RAPIDS Accelerator for Apache Spark (see SURVEY.md for the blueprint).
+
+Public surface: Session / DataFrame, the expression DSL (col, lit, when),
+aggregates (sum_, avg, count, count_star, min_, max_), DType, and the
+spark.rapids.* config registry.
+"""
+from .api import DataFrame, Session
+from .column import Column, ColumnBatch, Field, Schema
+from .config import RapidsConf, help_doc
+from .expr.aggregates import avg, count, count_star, max_, min_, sum_
+from .expr.expressions import CaseWhen, col, lit, when
+from .types import (BOOL, DATE32, FLOAT32, FLOAT64, INT8, INT16, INT32, INT64,
+                    STRING, TIMESTAMP, DType)
+
+__version__ = "0.1.0"

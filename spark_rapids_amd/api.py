@@ -1,0 +1,222 @@
+"""User-facing session + DataFrame API.
+
+The DataFrame surface mirrors what the reference accelerates underneath
+Spark SQL; here it is the engine's own frontend (no JVM in the loop). A
+Session owns a RapidsConf (the `spark.rapids.*` surface), a catalog, and the
+plan pipeline: logical plan -> GPU overrides tagging -> physical plan ->
+columnar execution on HIP kernels with CPU fallback per operator.
+"""
+from __future__ import annotations
+
+from typing import Dict, Iterable, List, Optional, Sequence, Union
+
+import numpy as np
+
+from .column import Column, ColumnBatch, Field, Schema
+from .config import RapidsConf, CONCURRENT_GPU_TASKS
+from .expr.aggregates import AggExpr
+from .expr.expressions import ColumnRef, Expression, col as _col
+from .memory.semaphore import GpuSemaphore
+from .plan import logical as L
+from .plan.overrides import plan_physical
+from .types import DType
+
+
+class MemTable:
+    """In-memory partitioned table source."""
+
+    def __init__(self, batches: List[ColumnBatch], schema: Schema):
+        self.batches = batches
+        self.schema = schema
+
+    def partitions(self) -> Iterable[ColumnBatch]:
+        return iter(self.batches)
+
+
+class DataFrame:
+    def __init__(self, session: "Session", plan: L.LogicalPlan):
+        self.session = session
+        self.plan = plan
+
+    # ---- transformations ----------------------------------------------
+    def filter(self, condition: Expression) -> "DataFrame":
+        return DataFrame(self.session, L.Filter(condition, self.plan))
+
+    where = filter
+
+    def select(self, *exprs: Union[str, Expression]) -> "DataFrame":
+        es = [(_col(e) if isinstance(e, str) else e) for e in exprs]
+        return DataFrame(self.session, L.Project(es, self.plan))
+
+    def with_column(self, name: str, expr: Expression) -> "DataFrame":
+        sch = self.plan.schema()
+        es: List[Expression] = []
+        replaced = False
+        for f in sch.fields:
+            if f.name == name:
+                es.append(expr.alias(name))
+                replaced = True
+            else:
+                es.append(_col(f.name))
+        if not replaced:
+            es.append(expr.alias(name))
+        return DataFrame(self.session, L.Project(es, self.plan))
+
+    def group_by(self, *keys: Union[str, Expression]) -> "GroupedData":
+        es = [(_col(k) if isinstance(k, str) else k) for k in keys]
+        return GroupedData(self, es)
+
+    def agg(self, *aggs: AggExpr) -> "DataFrame":
+        return DataFrame(self.session, L.Aggregate([], list(aggs), self.plan))
+
+    def join(self, other: "DataFrame", on: Union[str, Sequence[str]],
+             how: str = "inner", right_on: Optional[Sequence[str]] = None) -> "DataFrame":
+        if isinstance(on, str):
+            on = [on]
+        r_on = list(right_on) if right_on is not None else list(on)
+        return DataFrame(self.session,
+                         L.Join(self.plan, other.plan, list(on), r_on, how))
+
+    def sort(self, *keys: str, descending: Union[bool, List[bool]] = False) -> "DataFrame":
+        ks = list(keys)
+        if isinstance(descending, bool):
+            desc = [descending] * len(ks)
+        else:
+            desc = list(descending)
+        return DataFrame(self.session, L.Sort(self.plan, ks, desc))
+
+    def limit(self, n: int) -> "DataFrame":
+        return DataFrame(self.session, L.Limit(self.plan, n))
+
+    def union(self, other: "DataFrame") -> "DataFrame":
+        return DataFrame(self.session, L.Union([self.plan, other.plan]))
+
+    # ---- actions --------------------------------------------------------
+    @property
+    def schema(self) -> Schema:
+        return self.plan.schema()
+
+    def physical_plan(self):
+        return plan_physical(self.plan, self.session.conf)
+
+    def collect_batch(self) -> ColumnBatch:
+        exec_ = self.physical_plan()
+        sem = GpuSemaphore.get()
+        with sem.held():
+            batches = [b.cpu() for b in exec_.execute()]
+        if not batches:
+            return ColumnBatch(
+                [Column.from_pylist([], f.dtype) for f in self.schema.fields], 0)
+        if len(batches) == 1:
+            return batches[0]
+        from . import ops
+
+        return ops.concat_batches(batches)
+
+    def collect(self) -> List[tuple]:
+        batch = self.collect_batch()
+        cols = [c.to_pylist() for c in batch.columns]
+        return list(zip(*cols)) if cols else []
+
+    def to_pydict(self) -> Dict[str, list]:
+        batch = self.collect_batch()
+        return {f.name: c.to_pylist()
+                for f, c in zip(self.schema.fields, batch.columns)}
+
+    def count(self) -> int:
+        from .expr.aggregates import count_star
+
+        rows = self.agg(count_star()).collect()
+        return rows[0][0] if rows else 0
+
+    def explain(self) -> str:
+        exec_ = self.physical_plan()
+        out = [exec_.tree_string()]
+        notes = getattr(exec_, "tag_notes", [])
+        for n in notes:
+            for r in n.reasons:
+                out.append(f"  !{n.node}: {r}")
+        return "\n".join(out)
+
+
+class GroupedData:
+    def __init__(self, df: DataFrame, keys: List[Expression]):
+        self.df = df
+        self.keys = keys
+
+    def agg(self, *aggs: AggExpr) -> DataFrame:
+        return DataFrame(self.df.session,
+                         L.Aggregate(self.keys, list(aggs), self.df.plan))
+
+
+class Session:
+    def __init__(self, conf: Optional[Dict] = None):
+        self.conf = RapidsConf(conf)
+        self.catalog: Dict[str, MemTable] = {}
+        GpuSemaphore.initialize(self.conf.get(CONCURRENT_GPU_TASKS))
+
+    # ---- conf ----------------------------------------------------------
+    def set(self, key: str, value) -> "Session":
+        self.conf.set(key, value)
+        return self
+
+    # ---- data ingestion -------------------------------------------------
+    def create_dataframe(self, data: Dict[str, list],
+                         dtypes: Optional[Dict[str, DType]] = None,
+                         num_partitions: int = 1) -> DataFrame:
+        names = list(data)
+        cols = []
+        for n in names:
+            v = data[n]
+            dt = (dtypes or {}).get(n)
+            if isinstance(v, np.ndarray):
+                cols.append(Column.from_numpy(v, dt))
+            else:
+                if dt is None:
+                    dt = _infer_list_dtype(v)
+                cols.append(Column.from_pylist(list(v), dt))
+        batch = ColumnBatch(cols)
+        schema = Schema([Field(n, c.dtype) for n, c in zip(names, cols)])
+        batches = _split_partitions(batch, num_partitions)
+        table = MemTable(batches, schema)
+        return DataFrame(self, L.Scan(table, schema, "memory"))
+
+    def from_batches(self, batches: List[ColumnBatch], schema: Schema,
+                     label: str = "memory") -> DataFrame:
+        return DataFrame(self, L.Scan(MemTable(batches, schema), schema, label))
+
+    def register(self, name: str, df: DataFrame):
+        self.catalog[name] = df
+
+    def table(self, name: str) -> DataFrame:
+        return self.catalog[name]
+
+    def read_parquet(self, path: str, num_partitions: int = 1) -> DataFrame:
+        from .io.parquet import ParquetTable
+
+        src = ParquetTable(path, num_partitions)
+        return DataFrame(self, L.Scan(src, src.schema, f"parquet:{path}"))
+
+
+def _infer_list_dtype(v: list) -> DType:
+    from .expr.expressions import _infer_literal_dtype
+
+    for x in v:
+        if x is not None:
+            return _infer_literal_dtype(x)
+    return DType.int32()
+
+
+def _split_partitions(batch: ColumnBatch, n: int) -> List[ColumnBatch]:
+    if n <= 1 or batch.num_rows == 0:
+        return [batch]
+    from . import ops
+
+    rows = batch.num_rows
+    per = (rows + n - 1) // n
+    out = []
+    for s in range(0, rows, per):
+        idx = Column.from_numpy(
+            np.arange(s, min(s + per, rows), dtype=np.int32))
+        out.append(ops.gather(batch, idx))
+    return out

@@ -1,0 +1,75 @@
+"""Aggregate expression definitions.
+
+Reference analogue: org/apache/spark/sql/rapids/aggregate/aggregateFunctions.scala
+(GpuSum, GpuCount, GpuMin, GpuMax, GpuAverage...). An AggExpr names the
+aggregate op, the input expression, and the output type; the physical hash
+aggregate lowers these to the groupby kernel's (op, column, dtype) form with
+Spark result-type rules (sum(int) -> bigint, avg -> double, count -> bigint).
+"""
+from __future__ import annotations
+
+from typing import Optional
+
+from ..column import Schema
+from ..types import DType, FLOAT64, INT64
+from .expressions import Expression, _as_expr
+
+
+class AggExpr:
+    def __init__(self, op: str, child: Optional[Expression], name: Optional[str] = None):
+        self.op = op
+        self.child = _as_expr(child) if child is not None else None
+        self._name = name
+
+    def alias(self, name: str) -> "AggExpr":
+        return AggExpr(self.op, self.child, name)
+
+    def output_name(self) -> str:
+        if self._name:
+            return self._name
+        if self.child is None:
+            return f"{self.op}(*)"
+        return f"{self.op}({self.child})"
+
+    def out_dtype(self, schema: Schema) -> DType:
+        if self.op in ("count", "count_all"):
+            return INT64
+        ct = self.child.dtype(schema)
+        if self.op == "sum":
+            if ct.is_floating:
+                return FLOAT64
+            if ct.is_decimal:
+                return DType.decimal(min(ct.precision + 10, 38), ct.scale)
+            return INT64
+        if self.op in ("mean", "stddev", "variance"):
+            return FLOAT64
+        if self.op in ("min", "max", "first", "last"):
+            return ct
+        raise NotImplementedError(f"agg {self.op}")
+
+    def __str__(self):
+        return self.output_name()
+
+
+def sum_(e) -> AggExpr:
+    return AggExpr("sum", e)
+
+
+def avg(e) -> AggExpr:
+    return AggExpr("mean", e)
+
+
+def count(e) -> AggExpr:
+    return AggExpr("count", e)
+
+
+def count_star() -> AggExpr:
+    return AggExpr("count_all", None)
+
+
+def min_(e) -> AggExpr:
+    return AggExpr("min", e)
+
+
+def max_(e) -> AggExpr:
+    return AggExpr("max", e)

@@ -1,0 +1,404 @@
+"""Expression tree + evaluation.
+
+The engine analogue of the reference's GpuExpression.columnarEval path
+(reference: sql-plugin/src/main/scala/com/nvidia/spark/rapids/GpuExpressions.scala:139-334):
+expressions evaluate batch-at-a-time to whole Columns; the same tree evaluates
+on the CPU backend or on the GPU backend depending on where the batch lives.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+from ..column import Column, ColumnBatch, Schema
+from ..types import BOOL, DType, FLOAT64, INT32, INT64, STRING, TypeId, promote
+from .. import ops
+
+
+class Expression:
+    def dtype(self, schema: Schema) -> DType:
+        raise NotImplementedError
+
+    def nullable(self, schema: Schema) -> bool:
+        return True
+
+    @property
+    def children(self) -> Sequence["Expression"]:
+        return ()
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        raise NotImplementedError
+
+    def output_name(self) -> str:
+        return str(self)
+
+    # ---- operator DSL --------------------------------------------------
+    def _bin(self, op, other, swap=False):
+        other = _as_expr(other)
+        l, r = (other, self) if swap else (self, other)
+        return BinaryExpr(op, l, r)
+
+    def __add__(self, o):
+        return self._bin("add", o)
+
+    def __radd__(self, o):
+        return self._bin("add", o, swap=True)
+
+    def __sub__(self, o):
+        return self._bin("sub", o)
+
+    def __rsub__(self, o):
+        return self._bin("sub", o, swap=True)
+
+    def __mul__(self, o):
+        return self._bin("mul", o)
+
+    def __rmul__(self, o):
+        return self._bin("mul", o, swap=True)
+
+    def __truediv__(self, o):
+        return self._bin("div", o)
+
+    def __mod__(self, o):
+        return self._bin("mod", o)
+
+    def __lt__(self, o):
+        return self._bin("lt", o)
+
+    def __le__(self, o):
+        return self._bin("le", o)
+
+    def __gt__(self, o):
+        return self._bin("gt", o)
+
+    def __ge__(self, o):
+        return self._bin("ge", o)
+
+    def __eq__(self, o):  # noqa: PLE0302 - DSL
+        return self._bin("eq", o)
+
+    def __ne__(self, o):
+        return self._bin("ne", o)
+
+    def __and__(self, o):
+        return self._bin("and", o)
+
+    def __or__(self, o):
+        return self._bin("or", o)
+
+    def __invert__(self):
+        return UnaryExpr("not", self)
+
+    def __neg__(self):
+        return UnaryExpr("neg", self)
+
+    def __hash__(self):
+        return id(self)
+
+    def alias(self, name: str) -> "Alias":
+        return Alias(self, name)
+
+    def cast(self, to: DType) -> "CastExpr":
+        return CastExpr(self, to)
+
+    def is_null(self) -> "IsNull":
+        return IsNull(self)
+
+    def is_not_null(self) -> "Expression":
+        return UnaryExpr("not", IsNull(self))
+
+
+def _as_expr(v) -> Expression:
+    if isinstance(v, Expression):
+        return v
+    return Literal(v)
+
+
+class ColumnRef(Expression):
+    def __init__(self, name: str):
+        self.name = name
+
+    def dtype(self, schema: Schema) -> DType:
+        return schema.field(self.name).dtype
+
+    def nullable(self, schema: Schema) -> bool:
+        return schema.field(self.name).nullable
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return batch.columns[schema.index(self.name)]
+
+    def output_name(self) -> str:
+        return self.name
+
+    def __str__(self):
+        return self.name
+
+
+def _infer_literal_dtype(v) -> DType:
+    if isinstance(v, bool):
+        return BOOL
+    if isinstance(v, int):
+        return INT32 if -(2 ** 31) <= v < 2 ** 31 else INT64
+    if isinstance(v, float):
+        return FLOAT64
+    if isinstance(v, str):
+        return STRING
+    if v is None:
+        return DType(TypeId.NULL)
+    raise TypeError(f"unsupported literal {v!r}")
+
+
+class Literal(Expression):
+    def __init__(self, value, dtype: Optional[DType] = None):
+        self.value = value
+        self._dtype = dtype or _infer_literal_dtype(value)
+
+    def dtype(self, schema: Schema) -> DType:
+        return self._dtype
+
+    def nullable(self, schema: Schema) -> bool:
+        return self.value is None
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        if self.value is None:
+            return Column.nulls(self._dtype if self._dtype.id is not TypeId.NULL
+                                else INT32, batch.num_rows, batch.device)
+        return Column.from_pylist([self.value] * batch.num_rows, self._dtype,
+                                  batch.device)
+
+    def __str__(self):
+        return repr(self.value)
+
+
+# ops whose result is boolean
+_BOOL_OPS = {"eq", "ne", "lt", "le", "gt", "ge", "and", "or", "eq_null_safe"}
+# ops that force double output (Spark `/`)
+_DOUBLE_OPS = {"div", "pow"}
+
+
+class BinaryExpr(Expression):
+    def __init__(self, op: str, left: Expression, right: Expression):
+        self.op = op
+        self.left = left
+        self.right = right
+
+    @property
+    def children(self):
+        return (self.left, self.right)
+
+    def _in_dtype(self, schema) -> DType:
+        lt, rt = self.left.dtype(schema), self.right.dtype(schema)
+        if lt.id is TypeId.NULL:
+            return rt
+        if rt.id is TypeId.NULL:
+            return lt
+        if lt == rt:
+            return lt
+        if self.op in _DOUBLE_OPS:
+            return FLOAT64
+        return promote(lt, rt)
+
+    def dtype(self, schema: Schema) -> DType:
+        if self.op in _BOOL_OPS:
+            return BOOL
+        if self.op in _DOUBLE_OPS:
+            return FLOAT64
+        it = self._in_dtype(schema)
+        if it.is_decimal and self.op in ("add", "sub"):
+            return DType.decimal(min(it.precision + 1, 38), it.scale)
+        return it
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        common = self._in_dtype(schema)
+        if self.op in _DOUBLE_OPS and not common.is_decimal:
+            common = FLOAT64
+        out = self.dtype(schema)
+        # scalar fast path: literal on either side
+        if isinstance(self.right, Literal) and self.right.value is not None \
+                and not common.is_decimal:
+            lcol = ops.cast(self.left.eval(batch, schema), common)
+            return ops.binary_op_scalar(self.op, lcol, _coerce_py(self.right.value, common), out)
+        lcol = ops.cast(self.left.eval(batch, schema), common)
+        rcol = ops.cast(self.right.eval(batch, schema), common)
+        return ops.binary_op(self.op, lcol, rcol, out)
+
+    def __str__(self):
+        return f"({self.left} {self.op} {self.right})"
+
+
+def _coerce_py(v, dtype: DType):
+    if dtype.is_floating:
+        return float(v)
+    if dtype.is_integral or dtype.is_decimal:
+        return int(v)
+    return v
+
+
+_UNARY_OUT = {
+    "not": lambda t: BOOL,
+    "is_nan": lambda t: BOOL,
+    "neg": lambda t: t,
+    "abs": lambda t: t,
+    "sqrt": lambda t: FLOAT64,
+    "exp": lambda t: FLOAT64,
+    "log": lambda t: FLOAT64,
+    "sin": lambda t: FLOAT64,
+    "cos": lambda t: FLOAT64,
+    "tan": lambda t: FLOAT64,
+    "floor": lambda t: INT64 if not t.is_decimal else t,
+    "ceil": lambda t: INT64 if not t.is_decimal else t,
+    "length": lambda t: INT32,
+    "upper": lambda t: STRING,
+    "lower": lambda t: STRING,
+    "year": lambda t: INT32,
+    "month": lambda t: INT32,
+    "day": lambda t: INT32,
+}
+
+
+class UnaryExpr(Expression):
+    def __init__(self, op: str, child: Expression):
+        self.op = op
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return _UNARY_OUT[self.op](self.child.dtype(schema))
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        if self.op in ("sqrt", "exp", "log", "sin", "cos", "tan"):
+            c = ops.cast(c, FLOAT64)
+        return ops.unary_op(self.op, c, self.dtype(schema))
+
+    def __str__(self):
+        return f"{self.op}({self.child})"
+
+
+class IsNull(Expression):
+    def __init__(self, child: Expression):
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return BOOL
+
+    def nullable(self, schema: Schema) -> bool:
+        return False
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.is_null(self.child.eval(batch, schema))
+
+    def __str__(self):
+        return f"isnull({self.child})"
+
+
+class CastExpr(Expression):
+    def __init__(self, child: Expression, to: DType):
+        self.child = child
+        self.to = to
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return self.to
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.cast(self.child.eval(batch, schema), self.to)
+
+    def __str__(self):
+        return f"cast({self.child} as {self.to})"
+
+
+class Alias(Expression):
+    def __init__(self, child: Expression, name: str):
+        self.child = child
+        self.name = name
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return self.child.dtype(schema)
+
+    def nullable(self, schema: Schema) -> bool:
+        return self.child.nullable(schema)
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return self.child.eval(batch, schema)
+
+    def output_name(self) -> str:
+        return self.name
+
+    def __str__(self):
+        return f"{self.child} AS {self.name}"
+
+
+class CaseWhen(Expression):
+    """CASE WHEN cond THEN v ... ELSE e END  (pairs of (cond, value))."""
+
+    def __init__(self, branches, else_expr: Optional[Expression] = None):
+        self.branches = [(c, _as_expr(v)) for c, v in branches]
+        self.else_expr = _as_expr(else_expr) if else_expr is not None else None
+
+    @property
+    def children(self):
+        out = []
+        for c, v in self.branches:
+            out.extend([c, v])
+        if self.else_expr is not None:
+            out.append(self.else_expr)
+        return tuple(out)
+
+    def dtype(self, schema: Schema) -> DType:
+        t = self.branches[0][1].dtype(schema)
+        for _, v in self.branches[1:]:
+            t = promote(t, v.dtype(schema))
+        if self.else_expr is not None:
+            et = self.else_expr.dtype(schema)
+            if et.id is not TypeId.NULL:
+                t = promote(t, et)
+        return t
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        out_t = self.dtype(schema)
+        # evaluate as nested if_else from the last branch backwards
+        if self.else_expr is not None:
+            acc = ops.cast(self.else_expr.eval(batch, schema), out_t)
+        else:
+            acc = Column.nulls(out_t, batch.num_rows, batch.device)
+        backend = ops.backend_for(*batch.columns) if batch.columns else None
+        for cond, val in reversed(self.branches):
+            c = cond.eval(batch, schema)
+            v = ops.cast(val.eval(batch, schema), out_t)
+            acc = ops.backend_for(c, v, acc).if_else(c, v, acc)
+        return acc
+
+    def __str__(self):
+        parts = " ".join(f"WHEN {c} THEN {v}" for c, v in self.branches)
+        e = f" ELSE {self.else_expr}" if self.else_expr is not None else ""
+        return f"CASE {parts}{e} END"
+
+
+# ---------------------------------------------------------------------------
+# public DSL
+# ---------------------------------------------------------------------------
+
+def col(name: str) -> ColumnRef:
+    return ColumnRef(name)
+
+
+def lit(v, dtype: Optional[DType] = None) -> Literal:
+    return Literal(v, dtype)
+
+
+def when(cond: Expression, value) -> CaseWhen:
+    return CaseWhen([(cond, value)])

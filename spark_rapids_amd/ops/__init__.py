@@ -1,0 +1,114 @@
+"""Columnar op dispatch: device columns -> hand-written HIP kernels (hipdf),
+host columns -> CPU reference backend.
+
+The GPU path NEVER silently falls back to eager torch/CPU compute: if the
+native extension is missing on a machine with a GPU, ops raise. This is the
+boundary the reference crosses via JNI into libcudf
+(reference: SURVEY.md section 2.8 kernel surface).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+from ..column import Column, ColumnBatch
+from ..types import DType
+
+from . import cpu_backend
+
+_gpu = None
+_gpu_err: Optional[str] = None
+
+
+def _gpu_backend():
+    global _gpu, _gpu_err
+    if _gpu is None and _gpu_err is None:
+        try:
+            from . import gpu_backend as g
+            _gpu = g
+        except Exception as e:  # noqa: BLE001
+            _gpu_err = f"hipdf native extension unavailable: {e!r}"
+    if _gpu is None:
+        raise RuntimeError(_gpu_err)
+    return _gpu
+
+
+def backend_for(*cols: Column):
+    if any(c.is_cuda for c in cols):
+        return _gpu_backend()
+    return cpu_backend
+
+
+def binary_op(op: str, lhs: Column, rhs: Column, out_dtype: DType) -> Column:
+    return backend_for(lhs, rhs).binary_op(op, lhs, rhs, out_dtype)
+
+
+def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
+    return backend_for(lhs).binary_op_scalar(op, lhs, scalar, out_dtype)
+
+
+def unary_op(op: str, col: Column, out_dtype: Optional[DType] = None) -> Column:
+    return backend_for(col).unary_op(op, col, out_dtype or col.dtype)
+
+
+def cast(col: Column, to: DType) -> Column:
+    if col.dtype == to:
+        return col
+    return backend_for(col).cast(col, to)
+
+
+def is_null(col: Column) -> Column:
+    return backend_for(col).is_null(col)
+
+
+def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
+    """Filter: keep rows where mask is true (null mask = drop)."""
+    return backend_for(*batch.columns, mask).apply_boolean_mask(batch, mask)
+
+
+def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> ColumnBatch:
+    """Take rows by int32 index column; negative index -> null row
+    (OutOfBoundsPolicy.NULLIFY analogue for join gather maps)."""
+    return backend_for(*batch.columns, indices).gather(batch, indices, check_bounds)
+
+
+def murmur3_hash(cols: Sequence[Column], seed: int = 42) -> Column:
+    """Spark-compatible murmur3_x86_32 row hash over the columns."""
+    return backend_for(*cols).murmur3_hash(list(cols), seed)
+
+
+def hash_partition(batch: ColumnBatch, key_idx: Sequence[int], num_parts: int) -> Tuple[ColumnBatch, List[int]]:
+    """Reorder rows so rows of one partition are contiguous; returns the
+    reordered batch plus partition start offsets (len num_parts+1)."""
+    return backend_for(*batch.columns).hash_partition(batch, list(key_idx), num_parts)
+
+
+def reduce(op: str, col: Column):
+    """Whole-column reduction -> python scalar or None."""
+    return backend_for(col).reduce(op, col)
+
+
+def group_by_aggregate(batch: ColumnBatch, key_idx: Sequence[int],
+                       aggs: Sequence[Tuple[str, int, DType]]) -> ColumnBatch:
+    """Hash group-by. aggs = [(op, value_col_idx, out_dtype)].
+    Returns batch [keys..., agg results...]."""
+    return backend_for(*batch.columns).group_by_aggregate(batch, list(key_idx), list(aggs))
+
+
+def sort_order(batch: ColumnBatch, key_idx: Sequence[int],
+               descending: Sequence[bool], nulls_last: Sequence[bool]) -> Column:
+    """Return int32 permutation that sorts the batch by the keys."""
+    return backend_for(*batch.columns).sort_order(batch, list(key_idx), list(descending), list(nulls_last))
+
+
+def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
+                     left_keys: Sequence[int], right_keys: Sequence[int],
+                     how: str) -> Tuple[Column, Optional[Column]]:
+    """Equi-join gather maps (left_map, right_map). For semi/anti only
+    left_map is returned."""
+    return backend_for(*left.columns, *right.columns).join_gather_maps(
+        left, right, list(left_keys), list(right_keys), how)
+
+
+def concat_batches(batches: Sequence[ColumnBatch]) -> ColumnBatch:
+    assert batches
+    return backend_for(*batches[0].columns).concat_batches(list(batches))

@@ -1,0 +1,678 @@
+"""CPU reference backend with Spark SQL semantics.
+
+This is (a) the fallback execution path when an op is not GPU-enabled and
+(b) the golden reference that GPU kernel numerics tests compare against
+(reference analogue: CPU Spark itself in the integration-test harness,
+integration_tests/src/main/python/asserts.py).
+
+Spark-semantics notes implemented here:
+- AND/OR use Kleene three-valued logic
+- division / modulo by zero yields NULL (non-ANSI mode)
+- integer arithmetic wraps (non-ANSI)
+- float -> int cast truncates toward zero
+- sum/min/max/avg ignore NULLs; all-NULL group aggregates to NULL
+- murmur3 hash matches org.apache.spark.sql.catalyst.expressions.Murmur3Hash
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from ..column import Column, ColumnBatch, make_validity, torch_dtype
+from ..types import DType, TypeId
+
+
+# ---------------------------------------------------------------------------
+# helpers
+# ---------------------------------------------------------------------------
+
+def _vals(col: Column):
+    if col.dtype.id is TypeId.STRING:
+        return np.array(col.to_pylist(), dtype=object)
+    return col.data.cpu().numpy()[: col.size]
+
+
+def _valid(col: Column) -> np.ndarray:
+    return col.valid_array()
+
+
+def _make(vals: np.ndarray, valid: Optional[np.ndarray], dtype: DType) -> Column:
+    if dtype.id is TypeId.STRING:
+        out = []
+        for i, v in enumerate(vals):
+            if valid is not None and not valid[i]:
+                out.append(None)
+            else:
+                out.append(v)
+        return Column.from_pylist(out, dtype)
+    vals = np.ascontiguousarray(vals, dtype=dtype.numpy_dtype())
+    if valid is not None and not valid.all():
+        return Column(dtype, len(vals), torch.from_numpy(vals.copy()),
+                      make_validity(valid), null_count=int(len(vals) - valid.sum()))
+    return Column(dtype, len(vals), torch.from_numpy(vals.copy()), None, null_count=0)
+
+
+def _bool_col(vals: np.ndarray, valid: Optional[np.ndarray]) -> Column:
+    return _make(vals.astype(np.uint8), valid, DType.bool_())
+
+
+# ---------------------------------------------------------------------------
+# binary ops
+# ---------------------------------------------------------------------------
+
+_CMP_OPS = {"eq", "ne", "lt", "le", "gt", "ge"}
+
+
+def binary_op(op: str, lhs: Column, rhs: Column, out_dtype: DType) -> Column:
+    a, av = _vals(lhs), _valid(lhs)
+    b, bv = _vals(rhs), _valid(rhs)
+    return _binary_impl(op, a, av, b, bv, lhs.dtype, out_dtype)
+
+
+def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
+    a, av = _vals(lhs), _valid(lhs)
+    if scalar is None:
+        b = np.zeros(1, dtype=a.dtype if a.dtype != object else np.int64)
+        b = np.broadcast_to(b, a.shape)
+        bv = np.zeros(len(a), dtype=bool)
+    else:
+        if lhs.dtype.id is TypeId.STRING:
+            b = np.array([scalar] * len(a), dtype=object)
+        else:
+            b = np.broadcast_to(np.asarray(scalar), a.shape)
+        bv = np.ones(len(a), dtype=bool)
+    return _binary_impl(op, a, av, b, bv, lhs.dtype, out_dtype)
+
+
+def _binary_impl(op, a, av, b, bv, in_dtype: DType, out_dtype: DType) -> Column:
+    n = len(a)
+    if op == "and":
+        at = a.astype(bool) if a.dtype != object else a
+        bt = b.astype(bool)
+        res = at & bt
+        # Kleene: NULL unless (false present) or both valid
+        valid = (av & bv) | (av & ~at.astype(bool)) | (bv & ~bt.astype(bool))
+        return _bool_col(res.astype(np.uint8), valid)
+    if op == "or":
+        at = a.astype(bool)
+        bt = b.astype(bool)
+        res = at | bt
+        valid = (av & bv) | (av & at) | (bv & bt)
+        return _bool_col(res.astype(np.uint8), valid)
+    if op == "eq_null_safe":
+        eq = _safe_eq(a, b)
+        res = np.where(av & bv, eq, av == bv)
+        return _bool_col(res.astype(np.uint8), None)
+
+    valid = av & bv
+    if op in _CMP_OPS:
+        if in_dtype.id is TypeId.STRING:
+            # elementwise python compare on object arrays
+            res = np.array([_str_cmp(op, x, y) for x, y in zip(a, b)], dtype=bool)
+        else:
+            with np.errstate(invalid="ignore"):
+                res = {
+                    "eq": lambda: _safe_eq(a, b),
+                    "ne": lambda: ~_safe_eq(a, b),
+                    "lt": lambda: a < b,
+                    "le": lambda: a <= b,
+                    "gt": lambda: a > b,
+                    "ge": lambda: a >= b,
+                }[op]()
+        return _bool_col(res.astype(np.uint8), valid if not valid.all() else None)
+
+    np_out = out_dtype.numpy_dtype()
+    with np.errstate(divide="ignore", invalid="ignore", over="ignore"):
+        if op == "add":
+            res = (a.astype(np_out) + b.astype(np_out)).astype(np_out)
+        elif op == "sub":
+            res = (a.astype(np_out) - b.astype(np_out)).astype(np_out)
+        elif op == "mul":
+            res = (a.astype(np_out) * b.astype(np_out)).astype(np_out)
+        elif op == "div":  # Spark `/` -> double (planner casts); 0 divisor -> NULL
+            bf = b.astype(np.float64)
+            res = (a.astype(np.float64) / np.where(bf == 0, 1, bf)).astype(np_out)
+            valid = valid & (bf != 0)
+        elif op == "int_div":
+            bz = b == 0
+            bb = np.where(bz, 1, b)
+            q = np.trunc(a.astype(np.float64) / bb.astype(np.float64))
+            res = q.astype(np_out)
+            valid = valid & ~bz
+        elif op == "mod":
+            bz = b == 0
+            bb = np.where(bz, 1, b)
+            res = np.fmod(a, bb).astype(np_out)
+            valid = valid & ~bz
+        elif op == "pmod":
+            bz = b == 0
+            bb = np.where(bz, 1, b)
+            r = np.fmod(a, bb)
+            r = np.where((r != 0) & ((r < 0) != (bb < 0)), r + bb, r)
+            res = r.astype(np_out)
+            valid = valid & ~bz
+        elif op == "pow":
+            res = np.power(a.astype(np.float64), b.astype(np.float64)).astype(np_out)
+        elif op == "bitand":
+            res = (a & b).astype(np_out)
+        elif op == "bitor":
+            res = (a | b).astype(np_out)
+        elif op == "bitxor":
+            res = (a ^ b).astype(np_out)
+        elif op == "shiftleft":
+            res = (a.astype(np_out) << (b.astype(np.int32) & _shift_mask(np_out))).astype(np_out)
+        elif op == "shiftright":
+            res = (a.astype(np_out) >> (b.astype(np.int32) & _shift_mask(np_out))).astype(np_out)
+        elif op == "min":
+            res = np.minimum(a, b).astype(np_out)
+        elif op == "max":
+            res = np.maximum(a, b).astype(np_out)
+        else:
+            raise NotImplementedError(f"cpu binary op {op}")
+    return _make(res, valid if not valid.all() else None, out_dtype)
+
+
+def _shift_mask(np_out):
+    return 63 if np.dtype(np_out).itemsize == 8 else 31
+
+
+def _safe_eq(a, b):
+    if a.dtype == object:
+        return np.array([x == y for x, y in zip(a, b)], dtype=bool)
+    return a == b
+
+
+def _str_cmp(op, x, y):
+    x = x if x is not None else ""
+    y = y if y is not None else ""
+    return {"eq": x == y, "ne": x != y, "lt": x < y, "le": x <= y,
+            "gt": x > y, "ge": x >= y}[op]
+
+
+# ---------------------------------------------------------------------------
+# unary ops / cast
+# ---------------------------------------------------------------------------
+
+def unary_op(op: str, col: Column, out_dtype: DType) -> Column:
+    a, av = _vals(col), _valid(col)
+    np_out = out_dtype.numpy_dtype() if out_dtype.id is not TypeId.STRING else None
+    valid = av
+    with np.errstate(invalid="ignore", divide="ignore", over="ignore"):
+        if op == "neg":
+            res = (-a).astype(np_out)
+        elif op == "abs":
+            res = np.abs(a).astype(np_out)
+        elif op == "not":
+            res = (~a.astype(bool)).astype(np.uint8)
+        elif op == "sqrt":
+            res = np.sqrt(a.astype(np.float64)).astype(np_out)
+        elif op == "exp":
+            res = np.exp(a.astype(np.float64)).astype(np_out)
+        elif op == "log":
+            af = a.astype(np.float64)
+            res = np.log(np.where(af <= 0, 1, af)).astype(np_out)
+            valid = av & (af > 0)  # Spark: log(<=0) -> NULL
+        elif op == "floor":
+            res = np.floor(a).astype(np_out)
+        elif op == "ceil":
+            res = np.ceil(a).astype(np_out)
+        elif op == "sin":
+            res = np.sin(a.astype(np.float64)).astype(np_out)
+        elif op == "cos":
+            res = np.cos(a.astype(np.float64)).astype(np_out)
+        elif op == "tan":
+            res = np.tan(a.astype(np.float64)).astype(np_out)
+        elif op == "is_nan":
+            res = np.isnan(a.astype(np.float64)).astype(np.uint8)
+            res = np.where(av, res, 0)
+            return _bool_col(res, None)
+        elif op == "length":  # string length in chars
+            res = np.array([len(x) if x is not None else 0 for x in a], dtype=np.int32)
+        elif op == "upper":
+            return _make(np.array([x.upper() if x is not None else None for x in a],
+                                  dtype=object), av, out_dtype)
+        elif op == "lower":
+            return _make(np.array([x.lower() if x is not None else None for x in a],
+                                  dtype=object), av, out_dtype)
+        elif op == "year":
+            res = _dt_field(a, "year").astype(np_out)
+        elif op == "month":
+            res = _dt_field(a, "month").astype(np_out)
+        elif op == "day":
+            res = _dt_field(a, "day").astype(np_out)
+        else:
+            raise NotImplementedError(f"cpu unary op {op}")
+    return _make(res, valid if not valid.all() else None, out_dtype)
+
+
+def _dt_field(days: np.ndarray, field: str) -> np.ndarray:
+    dt = days.astype("datetime64[D]")
+    y = dt.astype("datetime64[Y]").astype(np.int64) + 1970
+    if field == "year":
+        return y
+    m_idx = dt.astype("datetime64[M]").astype(np.int64)
+    month = m_idx % 12 + 1
+    if field == "month":
+        return month
+    day = (dt - dt.astype("datetime64[M]")).astype(np.int64) + 1
+    return day
+
+
+def cast(col: Column, to: DType) -> Column:
+    a, av = _vals(col), _valid(col)
+    src = col.dtype
+    if to.id is TypeId.STRING:
+        out = []
+        for v, ok in zip(a, av):
+            if not ok:
+                out.append(None)
+            elif src.id is TypeId.BOOL:
+                out.append("true" if v else "false")
+            elif src.is_decimal:
+                out.append(_dec_str(int(v), src.scale))
+            else:
+                out.append(str(v))
+        return Column.from_pylist(out, to)
+    if src.id is TypeId.STRING:
+        res = np.zeros(len(a), dtype=to.numpy_dtype())
+        valid = av.copy()
+        for i, v in enumerate(a):
+            if not av[i]:
+                continue
+            try:
+                res[i] = float(v) if to.is_floating else int(float(v))
+            except (ValueError, TypeError):
+                valid[i] = False
+        return _make(res, valid if not valid.all() else None, to)
+    if src.is_decimal and to.is_decimal:
+        shift = to.scale - src.scale
+        res = a * (10 ** shift) if shift >= 0 else _round_half_up_div(a, 10 ** (-shift))
+        return _make(res, av if not av.all() else None, to)
+    if src.is_decimal:
+        f = a.astype(np.float64) / (10 ** src.scale)
+        res = f.astype(to.numpy_dtype())
+        return _make(res, av if not av.all() else None, to)
+    if to.is_decimal:
+        scaled = np.round(a.astype(np.float64) * (10 ** to.scale)) if src.is_floating \
+            else a.astype(np.int64) * (10 ** to.scale)
+        return _make(scaled.astype(np.int64), av if not av.all() else None, to)
+    with np.errstate(invalid="ignore", over="ignore"):
+        res = a.astype(to.numpy_dtype())
+    return _make(res, av if not av.all() else None, to)
+
+
+def _round_half_up_div(a: np.ndarray, d: int) -> np.ndarray:
+    q = a // d
+    r = a - q * d
+    adj = (np.abs(r) * 2 >= d).astype(np.int64) * np.sign(a)
+    return q + np.where(np.sign(a) < 0, np.minimum(adj, 0), np.maximum(adj, 0))
+
+
+def _dec_str(unscaled: int, scale: int) -> str:
+    if scale == 0:
+        return str(unscaled)
+    sign = "-" if unscaled < 0 else ""
+    s = str(abs(unscaled)).rjust(scale + 1, "0")
+    return f"{sign}{s[:-scale]}.{s[-scale:]}"
+
+
+def if_else(cond: Column, a: Column, b: Column) -> Column:
+    """Rowwise cond ? a : b. A NULL condition selects b (Spark CASE WHEN)."""
+    c = _vals(cond).astype(bool) & _valid(cond)
+    if a.dtype.id is TypeId.STRING:
+        av_, bv_ = _vals(a), _vals(b)
+        aok, bok = _valid(a), _valid(b)
+        out = [av_[i] if c[i] else bv_[i] for i in range(len(c))]
+        ok = np.where(c, aok, bok)
+        return _make(np.array(out, dtype=object), ok if not ok.all() else None, a.dtype)
+    res = np.where(c, _vals(a), _vals(b))
+    ok = np.where(c, _valid(a), _valid(b))
+    return _make(res, ok if not ok.all() else None, a.dtype)
+
+
+def is_null(col: Column) -> Column:
+    valid = _valid(col)
+    return _bool_col((~valid).astype(np.uint8), None)
+
+
+# ---------------------------------------------------------------------------
+# selection
+# ---------------------------------------------------------------------------
+
+def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
+    m = _vals(mask).astype(bool) & _valid(mask)
+    idx = np.nonzero(m)[0].astype(np.int32)
+    return _gather_idx(batch, idx, np.ones(len(idx), dtype=bool))
+
+
+def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> ColumnBatch:
+    idx = _vals(indices).astype(np.int64)
+    ok = idx >= 0
+    if check_bounds:
+        ok &= idx < batch.num_rows
+    return _gather_idx(batch, np.where(ok, idx, 0).astype(np.int64), ok)
+
+
+def _gather_idx(batch: ColumnBatch, idx: np.ndarray, row_ok: np.ndarray) -> ColumnBatch:
+    cols = []
+    for c in batch.columns:
+        a, av = _vals(c), _valid(c)
+        if len(a) == 0:
+            vals = np.zeros(len(idx), dtype=object if c.dtype.id is TypeId.STRING
+                            else c.dtype.numpy_dtype())
+            valid = np.zeros(len(idx), dtype=bool)
+        else:
+            vals = a[idx]
+            valid = av[idx] & row_ok
+        cols.append(_make(vals, valid if not valid.all() else None, c.dtype))
+    return ColumnBatch(cols, len(idx))
+
+
+def concat_batches(batches: List[ColumnBatch]) -> ColumnBatch:
+    ncols = batches[0].num_columns
+    cols = []
+    for i in range(ncols):
+        dtype = batches[0].columns[i].dtype
+        if dtype.id is TypeId.STRING:
+            vals = []
+            for b in batches:
+                vals.extend(b.columns[i].to_pylist())
+            cols.append(Column.from_pylist(vals, dtype))
+        else:
+            vals = np.concatenate([_vals(b.columns[i]) for b in batches])
+            valid = np.concatenate([_valid(b.columns[i]) for b in batches])
+            cols.append(_make(vals, valid if not valid.all() else None, dtype))
+    return ColumnBatch(cols)
+
+
+# ---------------------------------------------------------------------------
+# Spark-compatible murmur3_x86_32
+# ---------------------------------------------------------------------------
+
+_U32 = 0xFFFFFFFF
+
+
+def _rotl(x, r):
+    return ((x << r) | (x >> (32 - r))) & _U32
+
+
+def _mix_k1(k1):
+    k1 = (k1 * 0xCC9E2D51) & _U32
+    k1 = _rotl(k1, 15)
+    return (k1 * 0x1B873593) & _U32
+
+
+def _mix_h1(h1, k1):
+    h1 = h1 ^ k1
+    h1 = _rotl(h1, 13)
+    return (h1 * 5 + 0xE6546B64) & _U32
+
+
+def _fmix(h1, length):
+    h1 ^= length
+    h1 ^= h1 >> 16
+    h1 = (h1 * 0x85EBCA6B) & _U32
+    h1 ^= h1 >> 13
+    h1 = (h1 * 0xC2B2AE35) & _U32
+    h1 ^= h1 >> 16
+    return h1
+
+
+def _hash_int(v: np.ndarray, seed: np.ndarray) -> np.ndarray:
+    k1 = _mix_k1(v.astype(np.int64) & _U32)
+    h1 = _mix_h1(seed, k1)
+    return _fmix(h1, 4)
+
+
+def _hash_long(v: np.ndarray, seed: np.ndarray) -> np.ndarray:
+    v = v.astype(np.int64)
+    low = v & _U32
+    high = (v >> 32) & _U32
+    h1 = _mix_h1(seed, _mix_k1(low))
+    h1 = _mix_h1(h1, _mix_k1(high))
+    return _fmix(h1, 8)
+
+
+def _hash_bytes_one(data: bytes, seed: int) -> int:
+    h1 = seed
+    n = len(data)
+    i = 0
+    # Spark hashUnsafeBytes: 4-byte little-endian blocks then per-byte tail
+    while i + 4 <= n:
+        k1 = int.from_bytes(data[i:i + 4], "little")
+        h1 = _mix_h1(h1, _mix_k1(k1))
+        i += 4
+    while i < n:
+        b = data[i]
+        if b >= 128:
+            b -= 256
+        h1 = _mix_h1(h1, _mix_k1(b & _U32))
+        i += 1
+    return _fmix(h1, n)
+
+
+def murmur3_hash(cols: List[Column], seed: int = 42) -> Column:
+    n = cols[0].size
+    h = np.full(n, seed, dtype=np.int64)
+    for c in cols:
+        a, av = _vals(c), _valid(c)
+        if c.dtype.id is TypeId.STRING:
+            nh = np.array([_hash_bytes_one((x or "").encode("utf-8"), int(s))
+                           for x, s in zip(a, h)], dtype=np.int64)
+        elif c.dtype.id in (TypeId.INT64, TypeId.TIMESTAMP, TypeId.DECIMAL64):
+            nh = _hash_long(a, h)
+        elif c.dtype.id is TypeId.FLOAT64:
+            af = a.astype(np.float64).copy()
+            af[af == 0.0] = 0.0  # -0.0 -> 0.0
+            nh = _hash_long(af.view(np.int64), h)
+        elif c.dtype.id is TypeId.FLOAT32:
+            af = a.astype(np.float32).copy()
+            af[af == 0.0] = 0.0
+            nh = _hash_int(af.view(np.int32), h)
+        else:  # int8/16/32, bool, date: hashed as int
+            nh = _hash_int(a.astype(np.int32), h)
+        h = np.where(av, nh, h)
+    return _make(h.astype(np.int32), None, DType.int32())
+
+
+# ---------------------------------------------------------------------------
+# partition / groupby / join / sort
+# ---------------------------------------------------------------------------
+
+def hash_partition(batch: ColumnBatch, key_idx: List[int], num_parts: int):
+    keys = [batch.columns[i] for i in key_idx]
+    h = _vals(murmur3_hash(keys, 42)).astype(np.int64)
+    part = ((h % num_parts) + num_parts) % num_parts  # pmod
+    order = np.argsort(part, kind="stable").astype(np.int64)
+    counts = np.bincount(part, minlength=num_parts)
+    offsets = np.zeros(num_parts + 1, dtype=np.int64)
+    np.cumsum(counts, out=offsets[1:])
+    out = _gather_idx(batch, order, np.ones(len(order), dtype=bool))
+    return out, offsets.tolist()
+
+
+def reduce(op: str, col: Column):
+    a, av = _vals(col), _valid(col)
+    a = a[av]
+    if op == "count":
+        return int(av.sum())
+    if op == "count_all":
+        return col.size
+    if len(a) == 0:
+        return None
+    if op == "sum":
+        if col.dtype.is_floating:
+            return float(np.sum(a.astype(np.float64)))
+        return int(np.sum(a.astype(np.int64)))
+    if op == "min":
+        return a.min().item() if a.dtype != object else min(a)
+    if op == "max":
+        return a.max().item() if a.dtype != object else max(a)
+    if op == "mean":
+        return float(np.mean(a.astype(np.float64)))
+    raise NotImplementedError(f"cpu reduce {op}")
+
+
+def _group_codes(keys: List[Column]) -> Tuple[np.ndarray, np.ndarray]:
+    """Return (codes, first_row_index_per_group)."""
+    arrs = []
+    for k in keys:
+        a, av = _vals(k), _valid(k)
+        if a.dtype == object:
+            # encode strings (None distinct)
+            uniq, inv = np.unique(np.array([x if x is not None else "\0\0NULL" for x in a]),
+                                  return_inverse=True)
+            arrs.append(inv.astype(np.int64))
+        else:
+            arrs.append(a.view(np.int64) if a.dtype.itemsize == 8 else a.astype(np.int64))
+            arrs.append(av.astype(np.int64))
+    stacked = np.stack(arrs, axis=1) if arrs else np.zeros((len(keys[0]._vals), 0))
+    uniq, first_idx, codes = np.unique(stacked, axis=0, return_index=True,
+                                       return_inverse=True)
+    return codes.ravel(), first_idx
+
+
+def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
+                       aggs: List[Tuple[str, int, DType]]) -> ColumnBatch:
+    keys = [batch.columns[i] for i in key_idx]
+    n = batch.num_rows
+    if not key_idx:
+        codes = np.zeros(n, dtype=np.int64)
+        first_idx = np.array([0] if n else [], dtype=np.int64)
+        ngroups = 1 if n else 0
+    else:
+        codes, first_idx = _group_codes(keys)
+        ngroups = len(first_idx)
+    out_cols: List[Column] = []
+    for k in keys:
+        out_cols.append(_gather_idx(ColumnBatch([k]), first_idx,
+                                    np.ones(ngroups, dtype=bool)).columns[0])
+    for op, vidx, out_dtype in aggs:
+        if op == "count_all":
+            res = np.bincount(codes, minlength=ngroups).astype(np.int64)
+            out_cols.append(_make(res, None, out_dtype))
+            continue
+        vc = batch.columns[vidx]
+        a, av = _vals(vc), _valid(vc)
+        if op == "count":
+            res = np.bincount(codes[av], minlength=ngroups).astype(np.int64)
+            out_cols.append(_make(res, None, out_dtype))
+            continue
+        cnt = np.bincount(codes[av], minlength=ngroups)
+        gvalid = cnt > 0
+        gc = codes[av]
+        if op == "sum" and not out_dtype.is_floating:
+            # integral/decimal sum: accumulate in int64 (wraps like Spark non-ANSI)
+            s = np.zeros(ngroups, dtype=np.int64)
+            np.add.at(s, gc, a[av].astype(np.int64))
+            out_cols.append(_make(s, gvalid if not gvalid.all() else None, out_dtype))
+            continue
+        af = a[av].astype(np.float64)
+        if op in ("sum", "mean", "m2"):
+            s = np.zeros(ngroups)
+            np.add.at(s, gc, af)
+            if op == "sum":
+                res = s
+            elif op == "mean":
+                res = s / np.where(cnt == 0, 1, cnt)
+            else:  # m2: sum of squared deviations
+                mean = s / np.where(cnt == 0, 1, cnt)
+                d = af - mean[gc]
+                res = np.zeros(ngroups)
+                np.add.at(res, gc, d * d)
+        elif op in ("min", "max"):
+            init = np.inf if op == "min" else -np.inf
+            res = np.full(ngroups, init)
+            ufunc = np.minimum if op == "min" else np.maximum
+            ufunc.at(res, gc, af)
+        else:
+            raise NotImplementedError(f"cpu groupby agg {op}")
+        if out_dtype.is_floating:
+            res = res.astype(out_dtype.numpy_dtype())
+        else:
+            res = res.astype(np.int64)
+        out_cols.append(_make(res, gvalid if not gvalid.all() else None, out_dtype))
+    return ColumnBatch(out_cols, ngroups)
+
+
+def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
+                     left_keys: List[int], right_keys: List[int], how: str):
+    import pandas as pd
+
+    lk = {f"k{i}": _key_series(left.columns[c]) for i, c in enumerate(left_keys)}
+    rk = {f"k{i}": _key_series(right.columns[c]) for i, c in enumerate(right_keys)}
+    ldf = pd.DataFrame({**lk, "_l": np.arange(left.num_rows, dtype=np.int64)})
+    rdf = pd.DataFrame({**rk, "_r": np.arange(right.num_rows, dtype=np.int64)})
+    # Spark equi-join: NULL keys never match
+    ldf_nn = ldf.dropna(subset=[f"k{i}" for i in range(len(left_keys))])
+    rdf_nn = rdf.dropna(subset=[f"k{i}" for i in range(len(right_keys))])
+    on = [f"k{i}" for i in range(len(left_keys))]
+    if how == "inner":
+        m = ldf_nn.merge(rdf_nn, on=on, how="inner")
+        return (_make(m["_l"].to_numpy().astype(np.int32), None, DType.int32()),
+                _make(m["_r"].to_numpy().astype(np.int32), None, DType.int32()))
+    if how == "left":
+        m = ldf.merge(rdf_nn, on=on, how="left")
+        lmap = m["_l"].to_numpy().astype(np.int32)
+        rvals = m["_r"].to_numpy()
+        rmap = np.where(np.isnan(rvals), -1, np.nan_to_num(rvals)).astype(np.int32)
+        return (_make(lmap, None, DType.int32()),
+                _make(rmap, None, DType.int32()))
+    if how in ("semi", "anti"):
+        keys_set = rdf_nn[on].drop_duplicates()
+        m = ldf.merge(keys_set, on=on, how="left", indicator=True)
+        hit = (m["_merge"] == "both").to_numpy()
+        sel = hit if how == "semi" else ~hit
+        return (_make(m["_l"].to_numpy()[sel].astype(np.int32), None, DType.int32()),
+                None)
+    raise NotImplementedError(f"cpu join {how}")
+
+
+def _key_series(col: Column):
+    import pandas as pd
+
+    a, av = _vals(col), _valid(col)
+    if a.dtype == object:
+        return pd.Series(a, dtype=object).where(av, other=None)
+    s = pd.Series(a)
+    if not av.all():
+        s = s.astype("float64").where(av, other=np.nan) if a.dtype.kind in "iuf" \
+            else s.where(av, other=None)
+    return s
+
+
+def sort_order(batch: ColumnBatch, key_idx: List[int],
+               descending: List[bool], nulls_last: List[bool]) -> Column:
+    n = batch.num_rows
+    keys = []
+    # np.lexsort: last key is primary -> iterate reversed
+    for i, ci in enumerate(reversed(key_idx)):
+        ri = len(key_idx) - 1 - i
+        c = batch.columns[ci]
+        a, av = _vals(c), _valid(c)
+        desc = descending[ri]
+        nl = nulls_last[ri]
+        if a.dtype == object:
+            uniq, codes = np.unique(
+                np.array([x if x is not None else "" for x in a]), return_inverse=True)
+            a = codes.ravel().astype(np.int64)
+        else:
+            a = a.copy()
+        if a.dtype.kind == "f":
+            key = a.astype(np.float64)
+            if desc:
+                key = -key
+            null_key = np.inf if nl else -np.inf
+            key = np.where(av, key, null_key)
+            keys.append(key)
+        else:
+            key = a.astype(np.int64)
+            if desc:
+                key = -key
+            null_rank = np.where(av, 0, 1 if nl else -1)
+            keys.append(key)
+            keys.append(null_rank)
+    order = np.lexsort(tuple(keys)) if keys else np.arange(n)
+    return _make(order.astype(np.int32), None, DType.int32())

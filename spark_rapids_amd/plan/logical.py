@@ -1,0 +1,159 @@
+"""Logical plan nodes.
+
+The engine's analogue of the Catalyst logical plan that the reference's
+GpuOverrides pass consumes (reference: GpuOverrides.scala wrapAndTagPlan).
+Deliberately small: Scan, Filter, Project, Aggregate, Join, Sort, Limit, Union,
+Exchange (inserted by the planner for distributed runs).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence, Tuple
+
+from ..column import Field, Schema
+from ..expr.aggregates import AggExpr
+from ..expr.expressions import Expression
+from ..types import BOOL
+
+
+class LogicalPlan:
+    @property
+    def children(self) -> Sequence["LogicalPlan"]:
+        return ()
+
+    def schema(self) -> Schema:
+        raise NotImplementedError
+
+    def name(self) -> str:
+        return type(self).__name__
+
+
+class Scan(LogicalPlan):
+    def __init__(self, source, schema: Schema, label: str = "scan"):
+        self.source = source  # object with .partitions() -> iterable[ColumnBatch]
+        self._schema = schema
+        self.label = label
+
+    def schema(self) -> Schema:
+        return self._schema
+
+    def name(self) -> str:
+        return f"Scan({self.label})"
+
+
+class Filter(LogicalPlan):
+    def __init__(self, condition: Expression, child: LogicalPlan):
+        self.condition = condition
+        self.child = child
+        assert condition.dtype(child.schema()) == BOOL, "filter needs boolean"
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        return self.child.schema()
+
+
+class Project(LogicalPlan):
+    def __init__(self, exprs: List[Expression], child: LogicalPlan):
+        self.exprs = exprs
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        cs = self.child.schema()
+        return Schema([Field(e.output_name(), e.dtype(cs), e.nullable(cs))
+                       for e in self.exprs])
+
+
+class Aggregate(LogicalPlan):
+    def __init__(self, group_exprs: List[Expression], aggs: List[AggExpr],
+                 child: LogicalPlan):
+        self.group_exprs = group_exprs
+        self.aggs = aggs
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        cs = self.child.schema()
+        fields = [Field(e.output_name(), e.dtype(cs), e.nullable(cs))
+                  for e in self.group_exprs]
+        fields += [Field(a.output_name(), a.out_dtype(cs), True) for a in self.aggs]
+        return Schema(fields)
+
+
+class Join(LogicalPlan):
+    def __init__(self, left: LogicalPlan, right: LogicalPlan,
+                 left_on: List[str], right_on: List[str], how: str = "inner"):
+        self.left = left
+        self.right = right
+        self.left_on = left_on
+        self.right_on = right_on
+        self.how = how
+
+    @property
+    def children(self):
+        return (self.left, self.right)
+
+    def schema(self) -> Schema:
+        ls = self.left.schema()
+        if self.how in ("semi", "anti"):
+            return ls
+        rs = self.right.schema()
+        right_fields = []
+        for f in rs.fields:
+            nullable = f.nullable or self.how == "left"
+            right_fields.append(Field(f.name, f.dtype, nullable))
+        return Schema(list(ls.fields) + right_fields)
+
+    def name(self) -> str:
+        return f"Join({self.how})"
+
+
+class Sort(LogicalPlan):
+    def __init__(self, child: LogicalPlan, keys: List[str],
+                 descending: Optional[List[bool]] = None,
+                 nulls_last: Optional[List[bool]] = None):
+        self.child = child
+        self.keys = keys
+        self.descending = descending or [False] * len(keys)
+        # Spark default: NULLS FIRST for asc, NULLS LAST for desc
+        self.nulls_last = nulls_last or [d for d in self.descending]
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        return self.child.schema()
+
+
+class Limit(LogicalPlan):
+    def __init__(self, child: LogicalPlan, n: int):
+        self.child = child
+        self.n = n
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        return self.child.schema()
+
+
+class Union(LogicalPlan):
+    def __init__(self, plans: List[LogicalPlan]):
+        self.plans = plans
+
+    @property
+    def children(self):
+        return tuple(self.plans)
+
+    def schema(self) -> Schema:
+        return self.plans[0].schema()

@@ -1,0 +1,220 @@
+"""GPU overrides: tag the logical plan, place each operator on GPU or CPU,
+insert device transitions, and produce explain output.
+
+Reference analogue: GpuOverrides.scala (wrapAndTagPlan -> tag -> convert,
+registries at :4260/:4466) + RapidsMeta.scala (willNotWorkOnGpu reasons) +
+GpuTransitionOverrides.scala (transition insertion). The same contract holds:
+any unsupported node stays on CPU with transitions inserted around it, and
+`spark.rapids.sql.explain=NOT_ON_GPU|ALL` reports per-node reasons.
+"""
+from __future__ import annotations
+
+from typing import List, Optional
+
+from ..config import RapidsConf
+from ..expr.expressions import (Alias, BinaryExpr, CaseWhen, CastExpr, ColumnRef,
+                                Expression, IsNull, Literal, UnaryExpr)
+from ..types import DType, TypeId, TypeSig
+from . import logical as L
+from . import physical as P
+
+# ---------------------------------------------------------------------------
+# supported-type signatures per exec / expression family (grows every round)
+# ---------------------------------------------------------------------------
+
+_BASIC = TypeSig.all_basic()
+# GPU kernel support for group/join/sort keys: fixed-width only this round;
+# string keys fall back to CPU with a recorded reason.
+_FIXED_KEYS = TypeSig({
+    TypeId.BOOL, TypeId.INT8, TypeId.INT16, TypeId.INT32, TypeId.INT64,
+    TypeId.FLOAT32, TypeId.FLOAT64, TypeId.DECIMAL64, TypeId.DATE32,
+    TypeId.TIMESTAMP,
+})
+_NUMERIC = TypeSig.numeric()
+
+_GPU_BINARY_OPS = {
+    "add", "sub", "mul", "div", "int_div", "mod", "pmod", "pow",
+    "eq", "ne", "lt", "le", "gt", "ge", "eq_null_safe",
+    "and", "or", "bitand", "bitor", "bitxor", "shiftleft", "shiftright",
+    "min", "max",
+}
+_GPU_UNARY_OPS = {
+    "neg", "abs", "not", "sqrt", "exp", "log", "floor", "ceil",
+    "sin", "cos", "tan", "is_nan", "year", "month", "day",
+}
+# ops that only work on fixed-width inputs on GPU this round
+_GPU_STRING_OK = set()  # string compute exprs not yet on GPU (pass-through is)
+
+
+class TagReason:
+    def __init__(self, node: str, reasons: List[str]):
+        self.node = node
+        self.reasons = reasons
+
+
+class Tagger:
+    def __init__(self, conf: RapidsConf):
+        self.conf = conf
+        self.notes: List[TagReason] = []
+
+    # ---- expressions ---------------------------------------------------
+    def expr_reasons(self, e: Expression, schema) -> List[str]:
+        out: List[str] = []
+        self._tag_expr(e, schema, out)
+        return out
+
+    def _tag_expr(self, e: Expression, schema, out: List[str]):
+        if isinstance(e, (ColumnRef, Literal)):
+            pass
+        elif isinstance(e, Alias):
+            pass
+        elif isinstance(e, CastExpr):
+            src = e.child.dtype(schema)
+            if src.id is TypeId.STRING or e.to.id is TypeId.STRING:
+                out.append(f"cast {src} -> {e.to} not supported on GPU yet")
+            if not self.conf.expr_enabled("Cast"):
+                out.append("expression Cast disabled by conf")
+        elif isinstance(e, BinaryExpr):
+            in_t = e._in_dtype(schema)
+            if e.op not in _GPU_BINARY_OPS:
+                out.append(f"binary op {e.op} has no GPU kernel")
+            elif in_t.id is TypeId.STRING and e.op not in _GPU_STRING_OK:
+                out.append(f"binary op {e.op} on string not on GPU yet")
+            if not self.conf.expr_enabled(e.op):
+                out.append(f"expression {e.op} disabled by conf")
+        elif isinstance(e, UnaryExpr):
+            in_t = e.child.dtype(schema)
+            if e.op not in _GPU_UNARY_OPS:
+                out.append(f"unary op {e.op} has no GPU kernel")
+            elif in_t.id is TypeId.STRING:
+                out.append(f"unary op {e.op} on string not on GPU yet")
+        elif isinstance(e, (IsNull, CaseWhen)):
+            pass
+        else:
+            out.append(f"expression {type(e).__name__} not supported on GPU")
+        for c in e.children:
+            self._tag_expr(c, schema, out)
+
+    # ---- plan nodes ----------------------------------------------------
+    def exec_reasons(self, node: L.LogicalPlan) -> List[str]:
+        reasons: List[str] = []
+        name = type(node).__name__
+        if not self.conf.exec_enabled(name):
+            reasons.append(f"exec {name} disabled by conf")
+        cs = node.children[0].schema() if node.children else None
+        if isinstance(node, L.Scan):
+            for f in node.schema().fields:
+                r = _BASIC.supports(f.dtype)
+                if r:
+                    reasons.append(f"column {f.name}: {r}")
+        elif isinstance(node, L.Filter):
+            reasons += self.expr_reasons(node.condition, cs)
+        elif isinstance(node, L.Project):
+            for e in node.exprs:
+                reasons += self.expr_reasons(e, cs)
+        elif isinstance(node, L.Aggregate):
+            for e in node.group_exprs:
+                r = _FIXED_KEYS.supports(e.dtype(cs))
+                if r:
+                    reasons.append(f"group key {e}: {r}")
+                reasons += self.expr_reasons(e, cs)
+            for a in node.aggs:
+                if a.child is not None:
+                    t = a.child.dtype(cs)
+                    r = _NUMERIC.supports(t)
+                    if r and a.op not in ("count", "count_all", "min", "max"):
+                        reasons.append(f"agg {a.op}({a.child}): {r}")
+                    reasons += self.expr_reasons(a.child, cs)
+        elif isinstance(node, L.Join):
+            ls, rs = node.left.schema(), node.right.schema()
+            for k in node.left_on:
+                r = _FIXED_KEYS.supports(ls.field(k).dtype)
+                if r:
+                    reasons.append(f"join key {k}: {r}")
+            for k in node.right_on:
+                r = _FIXED_KEYS.supports(rs.field(k).dtype)
+                if r:
+                    reasons.append(f"join key {k}: {r}")
+            if node.how not in ("inner", "left", "semi", "anti"):
+                reasons.append(f"join type {node.how} not on GPU")
+        elif isinstance(node, L.Sort):
+            for k in node.keys:
+                r = _FIXED_KEYS.supports(node.schema().field(k).dtype)
+                if r:
+                    reasons.append(f"sort key {k}: {r}")
+        return reasons
+
+
+def plan_physical(node: L.LogicalPlan, conf: RapidsConf,
+                  tagger: Optional[Tagger] = None) -> P.PhysicalExec:
+    """Convert logical -> physical with GPU placement + transitions."""
+    top = tagger is None
+    if tagger is None:
+        tagger = Tagger(conf)
+    gpu_wanted = conf.sql_enabled and _gpu_available()
+    exec_ = _convert(node, conf, tagger, gpu_wanted)
+    if top:
+        exec_ = _ensure_device(exec_, "cpu")  # results surface on host
+        explain = conf.explain
+        if explain in ("NOT_ON_GPU", "ALL"):
+            for note in tagger.notes:
+                for r in note.reasons:
+                    print(f"!Exec {note.node} cannot run on GPU because {r}")
+        if conf.test_enabled and gpu_wanted:
+            bad = [n for n in tagger.notes if n.reasons]
+            if bad:
+                msgs = "; ".join(f"{n.node}: {n.reasons[0]}" for n in bad)
+                raise AssertionError(
+                    f"spark.rapids.sql.test.enabled: ops fell back to CPU: {msgs}")
+        _attach_tags(exec_, tagger)
+    return exec_
+
+
+def _attach_tags(exec_: P.PhysicalExec, tagger: Tagger):
+    exec_.tag_notes = tagger.notes  # type: ignore[attr-defined]
+
+
+def _gpu_available() -> bool:
+    import torch
+
+    return torch.cuda.is_available()
+
+
+def _ensure_device(exec_: P.PhysicalExec, device: str) -> P.PhysicalExec:
+    if exec_.device == device:
+        return exec_
+    return P.DeviceTransferExec(exec_, device)
+
+
+def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
+             gpu_wanted: bool) -> P.PhysicalExec:
+    reasons = tagger.exec_reasons(node) if gpu_wanted else []
+    on_gpu = gpu_wanted and not reasons
+    if gpu_wanted and reasons:
+        tagger.notes.append(TagReason(node.name(), reasons))
+    device = "cuda" if on_gpu else "cpu"
+
+    if isinstance(node, L.Scan):
+        return P.ScanExec(device, node.schema(), node.source, node.label)
+
+    kids = [_ensure_device(_convert(c, conf, tagger, gpu_wanted), device)
+            for c in node.children]
+
+    if isinstance(node, L.Filter):
+        return P.FilterExec(device, node.condition, kids[0])
+    if isinstance(node, L.Project):
+        return P.ProjectExec(device, node.exprs, kids[0], node.schema())
+    if isinstance(node, L.Aggregate):
+        return P.HashAggregateExec(device, node.group_exprs, node.aggs,
+                                   kids[0], node.schema())
+    if isinstance(node, L.Join):
+        return P.HashJoinExec(device, kids[0], kids[1], node.left_on,
+                              node.right_on, node.how, node.schema())
+    if isinstance(node, L.Sort):
+        return P.SortExec(device, node.keys, node.descending,
+                          node.nulls_last, kids[0])
+    if isinstance(node, L.Limit):
+        return P.LimitExec(device, node.n, kids[0])
+    if isinstance(node, L.Union):
+        return P.UnionExec(device, kids, node.schema())
+    raise NotImplementedError(f"plan node {type(node).__name__}")

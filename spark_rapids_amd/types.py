@@ -1,0 +1,278 @@
+"""Spark-semantics data type system for the MI355X columnar engine.
+
+Mirrors the role of the reference's TypeSig/TypeChecks lattice
+(reference: sql-plugin/src/main/scala/com/nvidia/spark/rapids/TypeChecks.scala:95-716)
+but is a fresh design: a small closed enum of physical types plus per-operator
+supported-type signatures used by the GPU-overrides tagging pass.
+"""
+from __future__ import annotations
+
+import enum
+from dataclasses import dataclass
+from typing import Optional
+
+import numpy as np
+
+
+class TypeId(enum.Enum):
+    BOOL = "boolean"
+    INT8 = "tinyint"
+    INT16 = "smallint"
+    INT32 = "int"
+    INT64 = "bigint"
+    FLOAT32 = "float"
+    FLOAT64 = "double"
+    DECIMAL64 = "decimal64"   # int64 backing + scale
+    DECIMAL128 = "decimal128"  # 2x int64 backing + scale
+    DATE32 = "date"           # days since epoch, int32 backing
+    TIMESTAMP = "timestamp"   # micros since epoch, int64 backing
+    STRING = "string"         # arrow offsets + bytes
+    NULL = "void"
+    LIST = "array"
+    STRUCT = "struct"
+
+
+_FIXED_WIDTH_BYTES = {
+    TypeId.BOOL: 1,
+    TypeId.INT8: 1,
+    TypeId.INT16: 2,
+    TypeId.INT32: 4,
+    TypeId.INT64: 8,
+    TypeId.FLOAT32: 4,
+    TypeId.FLOAT64: 8,
+    TypeId.DECIMAL64: 8,
+    TypeId.DECIMAL128: 16,
+    TypeId.DATE32: 4,
+    TypeId.TIMESTAMP: 8,
+}
+
+_NUMPY_DTYPES = {
+    TypeId.BOOL: np.uint8,
+    TypeId.INT8: np.int8,
+    TypeId.INT16: np.int16,
+    TypeId.INT32: np.int32,
+    TypeId.INT64: np.int64,
+    TypeId.FLOAT32: np.float32,
+    TypeId.FLOAT64: np.float64,
+    TypeId.DECIMAL64: np.int64,
+    TypeId.DATE32: np.int32,
+    TypeId.TIMESTAMP: np.int64,
+}
+
+_INTEGRALS = {TypeId.INT8, TypeId.INT16, TypeId.INT32, TypeId.INT64}
+_FLOATS = {TypeId.FLOAT32, TypeId.FLOAT64}
+
+
+@dataclass(frozen=True)
+class DType:
+    """A data type instance: a TypeId plus parameters (decimal precision/scale,
+    list element type, struct children)."""
+
+    id: TypeId
+    precision: int = 0
+    scale: int = 0
+    children: tuple = ()  # tuple[DType, ...] for LIST/STRUCT
+
+    # ---- constructors -------------------------------------------------
+    @staticmethod
+    def bool_() -> "DType":
+        return DType(TypeId.BOOL)
+
+    @staticmethod
+    def int8() -> "DType":
+        return DType(TypeId.INT8)
+
+    @staticmethod
+    def int16() -> "DType":
+        return DType(TypeId.INT16)
+
+    @staticmethod
+    def int32() -> "DType":
+        return DType(TypeId.INT32)
+
+    @staticmethod
+    def int64() -> "DType":
+        return DType(TypeId.INT64)
+
+    @staticmethod
+    def float32() -> "DType":
+        return DType(TypeId.FLOAT32)
+
+    @staticmethod
+    def float64() -> "DType":
+        return DType(TypeId.FLOAT64)
+
+    @staticmethod
+    def decimal(precision: int, scale: int) -> "DType":
+        if precision <= 18:
+            return DType(TypeId.DECIMAL64, precision, scale)
+        return DType(TypeId.DECIMAL128, precision, scale)
+
+    @staticmethod
+    def date32() -> "DType":
+        return DType(TypeId.DATE32)
+
+    @staticmethod
+    def timestamp() -> "DType":
+        return DType(TypeId.TIMESTAMP)
+
+    @staticmethod
+    def string() -> "DType":
+        return DType(TypeId.STRING)
+
+    @staticmethod
+    def list_(elem: "DType") -> "DType":
+        return DType(TypeId.LIST, children=(elem,))
+
+    @staticmethod
+    def struct(*fields: "DType") -> "DType":
+        return DType(TypeId.STRUCT, children=tuple(fields))
+
+    # ---- predicates ---------------------------------------------------
+    @property
+    def is_fixed_width(self) -> bool:
+        return self.id in _FIXED_WIDTH_BYTES
+
+    @property
+    def itemsize(self) -> int:
+        return _FIXED_WIDTH_BYTES[self.id]
+
+    @property
+    def is_integral(self) -> bool:
+        return self.id in _INTEGRALS
+
+    @property
+    def is_floating(self) -> bool:
+        return self.id in _FLOATS
+
+    @property
+    def is_numeric(self) -> bool:
+        return (
+            self.id in _INTEGRALS
+            or self.id in _FLOATS
+            or self.id in (TypeId.DECIMAL64, TypeId.DECIMAL128)
+        )
+
+    @property
+    def is_decimal(self) -> bool:
+        return self.id in (TypeId.DECIMAL64, TypeId.DECIMAL128)
+
+    @property
+    def is_timelike(self) -> bool:
+        return self.id in (TypeId.DATE32, TypeId.TIMESTAMP)
+
+    @property
+    def is_nested(self) -> bool:
+        return self.id in (TypeId.LIST, TypeId.STRUCT)
+
+    def numpy_dtype(self):
+        return np.dtype(_NUMPY_DTYPES[self.id])
+
+    def __str__(self) -> str:
+        if self.is_decimal:
+            return f"decimal({self.precision},{self.scale})"
+        if self.id is TypeId.LIST:
+            return f"array<{self.children[0]}>"
+        if self.id is TypeId.STRUCT:
+            return f"struct<{', '.join(str(c) for c in self.children)}>"
+        return self.id.value
+
+
+BOOL = DType.bool_()
+INT8 = DType.int8()
+INT16 = DType.int16()
+INT32 = DType.int32()
+INT64 = DType.int64()
+FLOAT32 = DType.float32()
+FLOAT64 = DType.float64()
+DATE32 = DType.date32()
+TIMESTAMP = DType.timestamp()
+STRING = DType.string()
+
+# Numeric widening order used by binary-op type promotion (Spark semantics:
+# result of int op float is the wider float, etc.)
+_PROMOTION_ORDER = [
+    TypeId.BOOL,
+    TypeId.INT8,
+    TypeId.INT16,
+    TypeId.INT32,
+    TypeId.INT64,
+    TypeId.FLOAT32,
+    TypeId.FLOAT64,
+]
+
+
+def promote(a: DType, b: DType) -> DType:
+    """Common wider type for arithmetic between a and b (non-decimal path)."""
+    if a == b:
+        return a
+    if a.is_decimal or b.is_decimal:
+        # decimal + integral -> decimal with enough precision; handled by caller
+        d = a if a.is_decimal else b
+        return d
+    ia = _PROMOTION_ORDER.index(a.id)
+    ib = _PROMOTION_ORDER.index(b.id)
+    return DType(_PROMOTION_ORDER[max(ia, ib)])
+
+
+class TypeSig:
+    """A set of supported TypeIds for one operator slot, with optional extra
+    predicates (e.g. decimal precision cap). Used by the overrides tagging pass
+    to explain why an op cannot run on GPU (reference analogue:
+    TypeChecks.scala TypeSig)."""
+
+    def __init__(self, ids, max_decimal_precision: int = 38, allow_nested: bool = False):
+        self.ids = frozenset(ids)
+        self.max_decimal_precision = max_decimal_precision
+        self.allow_nested = allow_nested
+
+    @staticmethod
+    def all_basic() -> "TypeSig":
+        return TypeSig(
+            {
+                TypeId.BOOL,
+                TypeId.INT8,
+                TypeId.INT16,
+                TypeId.INT32,
+                TypeId.INT64,
+                TypeId.FLOAT32,
+                TypeId.FLOAT64,
+                TypeId.DECIMAL64,
+                TypeId.DECIMAL128,
+                TypeId.DATE32,
+                TypeId.TIMESTAMP,
+                TypeId.STRING,
+            }
+        )
+
+    @staticmethod
+    def numeric() -> "TypeSig":
+        return TypeSig(
+            {
+                TypeId.INT8,
+                TypeId.INT16,
+                TypeId.INT32,
+                TypeId.INT64,
+                TypeId.FLOAT32,
+                TypeId.FLOAT64,
+                TypeId.DECIMAL64,
+                TypeId.DECIMAL128,
+            }
+        )
+
+    @staticmethod
+    def orderable() -> "TypeSig":
+        return TypeSig.all_basic()
+
+    def supports(self, dt: DType) -> Optional[str]:
+        """Return None if supported, else a human-readable reason."""
+        if dt.is_nested and not self.allow_nested:
+            return f"nested type {dt} is not supported"
+        if dt.id not in self.ids:
+            return f"type {dt} is not supported"
+        if dt.is_decimal and dt.precision > self.max_decimal_precision:
+            return (
+                f"decimal precision {dt.precision} exceeds max "
+                f"{self.max_decimal_precision}"
+            )
+        return None

@@ -1,0 +1,111 @@
+"""Operator-level tests: aggregate, join, sort, limit, union on the CPU path."""
+import pytest
+
+from spark_rapids_amd import (Session, avg, col, count, count_star, lit, max_,
+                              min_, sum_)
+
+
+def test_groupby_sum_count(session):
+    df = session.create_dataframe({
+        "k": [1, 2, 1, 2, 1, None],
+        "v": [10, 20, 30, None, 50, 60],
+    })
+    out = df.group_by("k").agg(
+        sum_(col("v")), count(col("v")), count_star()).sort("k").collect()
+    # NULLS FIRST on ascending sort
+    assert out == [(None, 60, 1, 1), (1, 90, 3, 3), (2, 20, 1, 2)]
+
+
+def test_groupby_avg_ignores_nulls(session):
+    df = session.create_dataframe({"k": [1, 1, 2], "v": [2.0, None, 6.0]})
+    out = df.group_by("k").agg(avg(col("v"))).sort("k").collect()
+    assert out == [(1, 2.0), (2, 6.0)]
+
+
+def test_groupby_all_null_group_sums_to_null(session):
+    df = session.create_dataframe({"k": [1, 1], "v": [None, None]},
+                                  dtypes={"v": __import__("spark_rapids_amd").INT32})
+    out = df.group_by("k").agg(sum_(col("v"))).collect()
+    assert out == [(1, None)]
+
+
+def test_global_agg(session):
+    df = session.create_dataframe({"v": [1, 2, 3, 4]})
+    assert df.agg(sum_(col("v")), min_(col("v")), max_(col("v"))).collect() \
+        == [(10, 1, 4)]
+
+
+def test_multi_partition_agg(session):
+    df = session.create_dataframe({"k": [1, 2] * 50, "v": list(range(100))},
+                                  num_partitions=7)
+    out = dict((k, v) for k, v in
+               df.group_by("k").agg(sum_(col("v"))).collect())
+    assert out == {1: sum(range(0, 100, 2)), 2: sum(range(1, 100, 2))}
+
+
+def test_inner_join(session):
+    left = session.create_dataframe({"k": [1, 2, 3, None], "a": [10, 20, 30, 40]})
+    right = session.create_dataframe({"k": [2, 3, 4, None], "b": [200, 300, 400, 500]})
+    out = sorted(left.join(right, on="k").select("a", "b").collect())
+    # NULL keys never match
+    assert out == [(20, 200), (30, 300)]
+
+
+def test_left_join(session):
+    left = session.create_dataframe({"k": [1, 2], "a": [10, 20]})
+    right = session.create_dataframe({"k": [2], "b": [200]})
+    out = sorted(left.join(right, on="k", how="left").select("a", "b").collect())
+    assert out == [(10, None), (20, 200)]
+
+
+def test_semi_anti_join(session):
+    left = session.create_dataframe({"k": [1, 2, 3], "a": [10, 20, 30]})
+    right = session.create_dataframe({"k": [2, 2, 3]})
+    semi = sorted(left.join(right, on="k", how="semi").select("a").to_pydict()["a"])
+    anti = sorted(left.join(right, on="k", how="anti").select("a").to_pydict()["a"])
+    assert semi == [20, 30]
+    assert anti == [10]
+
+
+def test_join_duplicate_keys_cross_product(session):
+    left = session.create_dataframe({"k": [1, 1], "a": [1, 2]})
+    right = session.create_dataframe({"k": [1, 1], "b": [3, 4]})
+    out = left.join(right, on="k").collect()
+    assert len(out) == 4
+
+
+def test_sort_orders(session):
+    df = session.create_dataframe({"x": [3, None, 1, 2]})
+    asc = df.sort("x").to_pydict()["x"]
+    assert asc == [None, 1, 2, 3]  # Spark asc = NULLS FIRST
+    desc = df.sort("x", descending=True).to_pydict()["x"]
+    assert desc == [3, 2, 1, None]  # Spark desc = NULLS LAST
+
+
+def test_sort_multi_key(session):
+    df = session.create_dataframe({"a": [1, 1, 2, 2], "b": [2, 1, 4, 3]})
+    out = df.sort("a", "b", descending=[False, True]).collect()
+    assert out == [(1, 2), (1, 1), (2, 4), (2, 3)]
+
+
+def test_limit(session):
+    df = session.create_dataframe({"x": list(range(10))}, num_partitions=3)
+    assert df.limit(5).count() == 5
+
+
+def test_union(session):
+    a = session.create_dataframe({"x": [1, 2]})
+    b = session.create_dataframe({"x": [3]})
+    assert sorted(a.union(b).to_pydict()["x"]) == [1, 2, 3]
+
+
+def test_count(session):
+    df = session.create_dataframe({"x": [1, None, 3]})
+    assert df.count() == 3
+
+
+def test_string_group_keys_cpu(session):
+    df = session.create_dataframe({"s": ["a", "b", "a", None],
+                                   "v": [1, 2, 3, 4]})
+    out = dict(df.group_by("s").agg(sum_(col("v"))).collect())
+    assert out == {"a": 4, "b": 2, None: 4}

@@ -1,0 +1,81 @@
+"""Tagging / fallback / explain / config-surface tests (no GPU needed:
+tagging logic is identical; placement just resolves to CPU here)."""
+import pytest
+
+from spark_rapids_amd import Session, col, lit, sum_
+from spark_rapids_amd.config import RapidsConf, help_doc, registry
+from spark_rapids_amd.plan.overrides import Tagger
+from spark_rapids_amd.plan import logical as L
+
+
+def _logical(df):
+    return df.plan
+
+
+def test_config_defaults_and_set():
+    c = RapidsConf()
+    assert c.sql_enabled is True
+    c.set("spark.rapids.sql.enabled", "false")
+    assert c.sql_enabled is False
+    c.set("spark.rapids.sql.batchSizeBytes", "512m")
+    assert c.batch_size_bytes == 512 << 20
+
+
+def test_config_doc_generation():
+    doc = help_doc()
+    assert "spark.rapids.sql.enabled" in doc
+    assert "spark.rapids.memory.pinnedPool.size" in doc
+    assert len(registry()) >= 20
+
+
+def test_tagger_accepts_numeric_plan(session):
+    df = session.create_dataframe({"a": [1], "b": [2.0]})
+    plan = L.Filter((col("a") > 0), _logical(df))
+    t = Tagger(session.conf)
+    assert t.exec_reasons(plan) == []
+
+
+def test_tagger_rejects_string_group_key(session):
+    df = session.create_dataframe({"s": ["x"], "v": [1]})
+    plan = L.Aggregate([col("s")], [sum_(col("v"))], _logical(df))
+    t = Tagger(session.conf)
+    reasons = t.exec_reasons(plan)
+    assert reasons and "string" in reasons[0]
+
+
+def test_per_exec_disable_conf(session):
+    session.set("spark.rapids.sql.exec.Filter", "false")
+    df = session.create_dataframe({"a": [1]})
+    plan = L.Filter(col("a") > 0, _logical(df))
+    t = Tagger(session.conf)
+    assert any("disabled by conf" in r for r in t.exec_reasons(plan))
+
+
+def test_per_expression_disable_conf(session):
+    session.set("spark.rapids.sql.expression.add", "false")
+    df = session.create_dataframe({"a": [1]})
+    t = Tagger(session.conf)
+    reasons = t.expr_reasons(col("a") + lit(1), df.schema)
+    assert any("disabled by conf" in r for r in reasons)
+
+
+def test_sql_enabled_false_runs_cpu(cpu_session):
+    df = cpu_session.create_dataframe({"a": [1, 2, 3]})
+    exec_ = df.filter(col("a") > 1).physical_plan()
+    assert "Cpu" in exec_.tree_string()
+    assert df.filter(col("a") > 1).count() == 2
+
+
+def test_explain_tree(session):
+    df = session.create_dataframe({"a": [1, 2]})
+    s = df.filter(col("a") > 0).select((col("a") * 2).alias("b")).explain()
+    assert "Project" in s and "Filter" in s and "Scan" in s
+
+
+def test_results_identical_cpu_vs_default(session, cpu_session):
+    data = {"k": [1, 2, 1, 2], "v": [1.0, 2.0, 3.0, 4.0]}
+    q = lambda s: (s.create_dataframe(data)
+                   .filter(col("v") > 1.0)
+                   .group_by("k").agg(sum_(col("v")))
+                   .sort("k").collect())
+    assert q(session) == q(cpu_session)
