@@ -1,0 +1,304 @@
+// pybind11 bindings for the hipdf CDNA4 kernel library.
+// Python (spark_rapids_amd/ops/gpu_backend.py) passes torch tensor
+// data_ptr()s, sizes and the current torch HIP stream; all device buffers are
+// allocated by the PyTorch-ROCm caching allocator (the RMM-style pool on
+// 288 GB HBM3E) so hipdf itself never calls hipMalloc.
+#include <pybind11/pybind11.h>
+#include <pybind11/stl.h>
+
+#include <hip/hip_runtime.h>
+
+#include <cstdint>
+#include <stdexcept>
+#include <string>
+
+namespace py = pybind11;
+
+extern "C" {
+void hipdf_binary_arith(int, int, const void*, const void*, double, int64_t,
+                        int, const void*, const void*, void*, void*, int64_t,
+                        hipStream_t);
+void hipdf_binary_cmp(int, int, const void*, const void*, double, int64_t, int,
+                      const void*, const void*, void*, void*, int64_t,
+                      hipStream_t);
+void hipdf_binary_bool(int, const void*, const void*, int, int, const void*,
+                       const void*, void*, void*, int64_t, hipStream_t);
+void hipdf_unary(int, int, const void*, const void*, void*, void*, int64_t,
+                 hipStream_t);
+void hipdf_cast(int, int, const void*, void*, int64_t, hipStream_t);
+void hipdf_decimal_rescale(const void*, void*, int64_t, int, int64_t,
+                           hipStream_t);
+void hipdf_if_else(int, const void*, const void*, const void*, const void*,
+                   const void*, const void*, void*, void*, int64_t,
+                   hipStream_t);
+void hipdf_mask_expand(const void*, void*, int, int64_t, hipStream_t);
+int64_t sel_num_blocks(int64_t);
+void hipdf_mask_count(const void*, const void*, void*, int64_t, hipStream_t);
+void hipdf_mask_scatter(const void*, const void*, const void*, void*, int64_t,
+                        hipStream_t);
+void hipdf_gather_fixed(int, const void*, const void*, void*, int64_t,
+                        hipStream_t);
+void hipdf_gather_validity(const void*, int, const void*, void*, int64_t,
+                           hipStream_t);
+void hipdf_gather_str_lens(const void*, const void*, void*, int64_t,
+                           hipStream_t);
+void hipdf_gather_str_bytes(const void*, const void*, const void*,
+                            const void*, void*, int64_t, hipStream_t);
+void hipdf_narrow_i64_i32(const void*, void*, int64_t, hipStream_t);
+void hipdf_copy_valid_range(const void*, int, int64_t, void*, int64_t,
+                            hipStream_t);
+int64_t scan_num_blocks(int64_t);
+void hipdf_scan_block(const void*, void*, void*, int64_t, hipStream_t);
+void hipdf_scan_add_offsets(void*, const void*, int64_t, hipStream_t);
+void hipdf_reduce(int, int, const void*, const void*, void*, void*, int64_t,
+                  hipStream_t);
+void hipdf_murmur3_col(int, int, const void*, const void*, void*, int64_t,
+                       hipStream_t);
+void hipdf_murmur3_str(const void*, const void*, const void*, void*, int64_t,
+                       hipStream_t);
+void hipdf_pmod_part(const void*, int, void*, int64_t, hipStream_t);
+void hipdf_gb_build(const void*, const void*, int, void*, void*, int64_t,
+                    int64_t, hipStream_t);
+void hipdf_gb_number(const void*, void*, void*, void*, int64_t, hipStream_t);
+void hipdf_gb_rowgid(const void*, const void*, void*, int64_t, hipStream_t);
+void hipdf_gb_agg(int, int, const void*, const void*, const void*, void*,
+                  void*, int, int32_t, int64_t, hipStream_t);
+void hipdf_mask_from_nonzero(const void*, void*, int64_t, hipStream_t);
+void hipdf_join_build(const void*, const void*, int, void*, void*, int64_t,
+                      int64_t, hipStream_t);
+void hipdf_join_count(int, const void*, const void*, const void*, int,
+                      const void*, const void*, int64_t, void*, int64_t,
+                      hipStream_t);
+void hipdf_join_fill(int, const void*, const void*, const void*, int,
+                     const void*, const void*, int64_t, const void*, void*,
+                     void*, int64_t, hipStream_t);
+int64_t part_num_blocks(int64_t);
+void hipdf_part_hist(const void*, int, void*, int64_t, hipStream_t);
+void hipdf_part_scatter(const void*, int, const void*, void*, int64_t,
+                        hipStream_t);
+}
+
+static void check_async() {
+  hipError_t e = hipGetLastError();
+  if (e != hipSuccess)
+    throw std::runtime_error(std::string("hipdf kernel launch: ") +
+                             hipGetErrorString(e));
+}
+
+#define P(x) reinterpret_cast<const void*>(x)
+#define PM(x) reinterpret_cast<void*>(x)
+#define S(x) reinterpret_cast<hipStream_t>(x)
+
+PYBIND11_MODULE(hipdf, m) {
+  m.doc() = "hand-written CDNA4 (gfx950) columnar kernels for MI355X";
+
+  m.def("build_arch", []() { return std::string("gfx950"); });
+  m.def("device_count", []() {
+    int n = 0;
+    hipGetDeviceCount(&n);
+    return n;
+  });
+  m.def("synchronize", []() {
+    hipError_t e = hipDeviceSynchronize();
+    if (e != hipSuccess)
+      throw std::runtime_error(hipGetErrorString(e));
+  });
+
+  m.def("binary_arith",
+        [](int op, int t, int64_t a, int64_t b, double sd, int64_t si,
+           bool scalar_rhs, int64_t av, int64_t bv, int64_t out, int64_t ov,
+           int64_t n, int64_t stream) {
+          hipdf_binary_arith(op, t, P(a), P(b), sd, si, scalar_rhs, P(av),
+                             P(bv), PM(out), PM(ov), n, S(stream));
+          check_async();
+        });
+  m.def("binary_cmp",
+        [](int op, int t, int64_t a, int64_t b, double sd, int64_t si,
+           bool scalar_rhs, int64_t av, int64_t bv, int64_t out, int64_t ov,
+           int64_t n, int64_t stream) {
+          hipdf_binary_cmp(op, t, P(a), P(b), sd, si, scalar_rhs, P(av), P(bv),
+                           PM(out), PM(ov), n, S(stream));
+          check_async();
+        });
+  m.def("binary_bool",
+        [](int op, int64_t a, int64_t b, int sb, bool scalar_rhs, int64_t av,
+           int64_t bv, int64_t out, int64_t ov, int64_t n, int64_t stream) {
+          hipdf_binary_bool(op, P(a), P(b), sb, scalar_rhs, P(av), P(bv),
+                            PM(out), PM(ov), n, S(stream));
+          check_async();
+        });
+  m.def("unary", [](int op, int t, int64_t a, int64_t av, int64_t out,
+                    int64_t ov, int64_t n, int64_t stream) {
+    hipdf_unary(op, t, P(a), P(av), PM(out), PM(ov), n, S(stream));
+    check_async();
+  });
+  m.def("cast", [](int ft, int tt, int64_t a, int64_t out, int64_t n,
+                   int64_t stream) {
+    hipdf_cast(ft, tt, P(a), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("decimal_rescale", [](int64_t a, int64_t out, int64_t pow10, bool up,
+                              int64_t n, int64_t stream) {
+    hipdf_decimal_rescale(P(a), PM(out), pow10, up, n, S(stream));
+    check_async();
+  });
+
+  m.def("if_else", [](int t, int64_t cond, int64_t cv, int64_t a, int64_t av,
+                      int64_t b, int64_t bv, int64_t out, int64_t ov,
+                      int64_t n, int64_t stream) {
+    hipdf_if_else(t, P(cond), P(cv), P(a), P(av), P(b), P(bv), PM(out),
+                  PM(ov), n, S(stream));
+    check_async();
+  });
+  m.def("mask_expand", [](int64_t mask, int64_t out, bool invert, int64_t n,
+                          int64_t stream) {
+    hipdf_mask_expand(P(mask), PM(out), invert, n, S(stream));
+    check_async();
+  });
+
+  m.def("sel_num_blocks", &sel_num_blocks);
+  m.def("mask_count", [](int64_t mask, int64_t mv, int64_t counts, int64_t n,
+                         int64_t stream) {
+    hipdf_mask_count(P(mask), P(mv), PM(counts), n, S(stream));
+    check_async();
+  });
+  m.def("mask_scatter", [](int64_t mask, int64_t mv, int64_t offsets,
+                           int64_t out_idx, int64_t n, int64_t stream) {
+    hipdf_mask_scatter(P(mask), P(mv), P(offsets), PM(out_idx), n, S(stream));
+    check_async();
+  });
+  m.def("gather_fixed", [](int esize, int64_t in, int64_t idx, int64_t out,
+                           int64_t n_out, int64_t stream) {
+    hipdf_gather_fixed(esize, P(in), P(idx), PM(out), n_out, S(stream));
+    check_async();
+  });
+  m.def("gather_validity", [](int64_t in_valid, bool in_has, int64_t idx,
+                              int64_t out_valid, int64_t n_out,
+                              int64_t stream) {
+    hipdf_gather_validity(P(in_valid), in_has, P(idx), PM(out_valid), n_out,
+                          S(stream));
+    check_async();
+  });
+  m.def("gather_str_lens", [](int64_t offs, int64_t idx, int64_t lens,
+                              int64_t n_out, int64_t stream) {
+    hipdf_gather_str_lens(P(offs), P(idx), PM(lens), n_out, S(stream));
+    check_async();
+  });
+  m.def("gather_str_bytes", [](int64_t in_bytes, int64_t in_offs, int64_t idx,
+                               int64_t out_offs, int64_t out_bytes,
+                               int64_t n_out, int64_t stream) {
+    hipdf_gather_str_bytes(P(in_bytes), P(in_offs), P(idx), P(out_offs),
+                           PM(out_bytes), n_out, S(stream));
+    check_async();
+  });
+  m.def("narrow_i64_i32", [](int64_t in, int64_t out, int64_t n,
+                             int64_t stream) {
+    hipdf_narrow_i64_i32(P(in), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("copy_valid_range", [](int64_t src, bool src_has, int64_t dst_off,
+                               int64_t dst, int64_t n, int64_t stream) {
+    hipdf_copy_valid_range(P(src), src_has, dst_off, PM(dst), n, S(stream));
+    check_async();
+  });
+
+  m.def("scan_num_blocks", &scan_num_blocks);
+  m.def("scan_block", [](int64_t in, int64_t out, int64_t sums, int64_t n,
+                         int64_t stream) {
+    hipdf_scan_block(P(in), PM(out), PM(sums), n, S(stream));
+    check_async();
+  });
+  m.def("scan_add_offsets", [](int64_t out, int64_t sums, int64_t n,
+                               int64_t stream) {
+    hipdf_scan_add_offsets(PM(out), P(sums), n, S(stream));
+    check_async();
+  });
+  m.def("reduce", [](int op, int t, int64_t a, int64_t av, int64_t acc,
+                     int64_t cnt, int64_t n, int64_t stream) {
+    hipdf_reduce(op, t, P(a), P(av), PM(acc), PM(cnt), n, S(stream));
+    check_async();
+  });
+
+  m.def("murmur3_col", [](int kind, int t, int64_t a, int64_t av,
+                          int64_t seeds, int64_t n, int64_t stream) {
+    hipdf_murmur3_col(kind, t, P(a), P(av), PM(seeds), n, S(stream));
+    check_async();
+  });
+  m.def("murmur3_str", [](int64_t offs, int64_t bytes, int64_t av,
+                          int64_t seeds, int64_t n, int64_t stream) {
+    hipdf_murmur3_str(P(offs), P(bytes), P(av), PM(seeds), n, S(stream));
+    check_async();
+  });
+  m.def("pmod_part", [](int64_t h, int nparts, int64_t part, int64_t n,
+                        int64_t stream) {
+    hipdf_pmod_part(P(h), nparts, PM(part), n, S(stream));
+    check_async();
+  });
+
+  m.def("gb_build", [](int64_t hashes, int64_t keys, int nkeys,
+                       int64_t slot_row, int64_t row_slot, int64_t cap,
+                       int64_t n, int64_t stream) {
+    hipdf_gb_build(P(hashes), P(keys), nkeys, PM(slot_row), PM(row_slot), cap,
+                   n, S(stream));
+    check_async();
+  });
+  m.def("gb_number", [](int64_t slot_row, int64_t slot_gid, int64_t ngroups,
+                        int64_t leaders, int64_t cap, int64_t stream) {
+    hipdf_gb_number(P(slot_row), PM(slot_gid), PM(ngroups), PM(leaders), cap,
+                    S(stream));
+    check_async();
+  });
+  m.def("gb_rowgid", [](int64_t row_slot, int64_t slot_gid, int64_t row_gid,
+                        int64_t n, int64_t stream) {
+    hipdf_gb_rowgid(P(row_slot), P(slot_gid), PM(row_gid), n, S(stream));
+    check_async();
+  });
+  m.def("gb_agg", [](int op, int t, int64_t vals, int64_t vvalid,
+                     int64_t row_gid, int64_t acc, int64_t cnt,
+                     bool acc_is_double, int ngroups, int64_t n,
+                     int64_t stream) {
+    hipdf_gb_agg(op, t, P(vals), P(vvalid), P(row_gid), PM(acc), PM(cnt),
+                 acc_is_double, ngroups, n, S(stream));
+    check_async();
+  });
+  m.def("mask_from_nonzero", [](int64_t cnt, int64_t mask, int64_t n,
+                                int64_t stream) {
+    hipdf_mask_from_nonzero(P(cnt), PM(mask), n, S(stream));
+    check_async();
+  });
+
+  m.def("join_build", [](int64_t hashes, int64_t keys, int nkeys,
+                         int64_t head, int64_t next, int64_t cap, int64_t n,
+                         int64_t stream) {
+    hipdf_join_build(P(hashes), P(keys), nkeys, PM(head), PM(next), cap, n,
+                     S(stream));
+    check_async();
+  });
+  m.def("join_count", [](int how, int64_t lh, int64_t lk, int64_t rk,
+                         int nkeys, int64_t head, int64_t next, int64_t cap,
+                         int64_t counts, int64_t n, int64_t stream) {
+    hipdf_join_count(how, P(lh), P(lk), P(rk), nkeys, P(head), P(next), cap,
+                     PM(counts), n, S(stream));
+    check_async();
+  });
+  m.def("join_fill", [](int how, int64_t lh, int64_t lk, int64_t rk,
+                        int nkeys, int64_t head, int64_t next, int64_t cap,
+                        int64_t offsets, int64_t lmap, int64_t rmap,
+                        int64_t n, int64_t stream) {
+    hipdf_join_fill(how, P(lh), P(lk), P(rk), nkeys, P(head), P(next), cap,
+                    P(offsets), PM(lmap), PM(rmap), n, S(stream));
+    check_async();
+  });
+
+  m.def("part_num_blocks", &part_num_blocks);
+  m.def("part_hist", [](int64_t part, int nparts, int64_t counts, int64_t n,
+                        int64_t stream) {
+    hipdf_part_hist(P(part), nparts, PM(counts), n, S(stream));
+    check_async();
+  });
+  m.def("part_scatter", [](int64_t part, int nparts, int64_t offsets,
+                           int64_t perm, int64_t n, int64_t stream) {
+    hipdf_part_scatter(P(part), nparts, P(offsets), PM(perm), n, S(stream));
+    check_async();
+  });
+}

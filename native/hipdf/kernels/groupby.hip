@@ -1,0 +1,234 @@
+// Hash group-by aggregation (reference analogue: cudf groupBy().aggregate
+// reached from GpuHashAggregateExec — SURVEY.md §3.5).
+//
+// Structure (race-free 3-pass build + per-agg accumulate):
+//   1. k_gb_build: open-addressing CAS on a row-index slot table; every row
+//      finds the slot whose leader row has equal keys (Spark semantics:
+//      nulls group together, NaN==NaN) and records its slot id.
+//   2. k_gb_number: slots with a leader get dense group ids (atomic counter).
+//   3. k_gb_rowgid: row -> group id via its slot.
+//   4. k_gb_agg: per (op, value column) accumulation. Low-cardinality groups
+//      (the common SQL case) take the LDS-tiled path: per-block accumulators
+//      in LDS, one device atomic per (block, group) at flush — this is what
+//      makes 4-group q1-style aggregations run at HBM bandwidth instead of
+//      serializing on 4 hot atomics.
+#include "hipdf_common.h"
+#include "keys.h"
+
+#define GB_EMPTY (-1)
+#define GB_LDS_GROUPS 2048
+
+__global__ void k_gb_build(const int32_t* __restrict__ hashes,
+                           const KeyCol* __restrict__ keys, int nkeys,
+                           int32_t* __restrict__ slot_row,
+                           int32_t* __restrict__ row_slot, uint32_t slot_mask,
+                           int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t slot = slot_of((uint32_t)hashes[i], slot_mask);
+    while (true) {
+      int32_t cur = atomicCAS(&slot_row[slot], GB_EMPTY, (int32_t)i);
+      if (cur == GB_EMPTY || cur == (int32_t)i ||
+          rows_equal(keys, keys, nkeys, i, cur)) {
+        row_slot[i] = (int32_t)slot;
+        break;
+      }
+      slot = (slot + 1) & slot_mask;
+    }
+  }
+}
+
+__global__ void k_gb_number(const int32_t* __restrict__ slot_row,
+                            int32_t* __restrict__ slot_gid,
+                            int32_t* __restrict__ ngroups,
+                            int32_t* __restrict__ leaders, int64_t cap) {
+  for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
+       s += (int64_t)gridDim.x * blockDim.x) {
+    int32_t r = slot_row[s];
+    if (r != GB_EMPTY) {
+      int32_t gid = atomicAdd(ngroups, 1);
+      slot_gid[s] = gid;
+      leaders[gid] = r;
+    }
+  }
+}
+
+__global__ void k_gb_rowgid(const int32_t* __restrict__ row_slot,
+                            const int32_t* __restrict__ slot_gid,
+                            int32_t* __restrict__ row_gid, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    row_gid[i] = slot_gid[row_slot[i]];
+}
+
+// ---- aggregation ---------------------------------------------------------
+enum GbOp : int { GB_SUM = 0, GB_MIN, GB_MAX, GB_COUNT, GB_COUNT_ALL };
+
+template <typename ACC>
+__device__ __forceinline__ void acc_atomic(int op, ACC* addr, ACC v);
+
+template <>
+__device__ __forceinline__ void acc_atomic<int64_t>(int op, int64_t* addr,
+                                                    int64_t v) {
+  if (op == GB_MIN) atomicMin((long long*)addr, (long long)v);
+  else if (op == GB_MAX) atomicMax((long long*)addr, (long long)v);
+  else atomicAdd((unsigned long long*)addr, (unsigned long long)v);
+}
+
+template <>
+__device__ __forceinline__ void acc_atomic<double>(int op, double* addr,
+                                                   double v) {
+  if (op == GB_SUM) {
+    atomicAdd(addr, v);
+    return;
+  }
+  unsigned long long* up = (unsigned long long*)addr;
+  unsigned long long old = *up, assumed;
+  do {
+    assumed = old;
+    double cur = __longlong_as_double((long long)assumed);
+    bool better = op == GB_MIN ? (v < cur || (isnan(cur) && !isnan(v)))
+                               : (v > cur && !isnan(cur)) || (isnan(v));
+    // Spark ordering: NaN is the greatest value
+    if (!better) break;
+    old = atomicCAS(up, assumed, (unsigned long long)__double_as_longlong(v));
+  } while (old != assumed);
+}
+
+template <typename ACC>
+__device__ __forceinline__ ACC acc_init(int op) {
+  if (op == GB_MIN) return std::numeric_limits<ACC>::max();
+  if (op == GB_MAX) return std::numeric_limits<ACC>::lowest();
+  return (ACC)0;
+}
+
+// one value column, one op; acc/cnt indexed by group id
+template <typename T, typename ACC, bool USE_LDS>
+__global__ void k_gb_agg(int op, const T* __restrict__ vals,
+                         const uint64_t* __restrict__ vvalid,
+                         const int32_t* __restrict__ row_gid,
+                         ACC* __restrict__ acc, int64_t* __restrict__ cnt,
+                         int32_t ngroups, int64_t n) {
+  extern __shared__ char lds_raw[];
+  ACC* lacc = (ACC*)lds_raw;
+  int64_t* lcnt = (int64_t*)(lds_raw + (USE_LDS ? sizeof(ACC) * ngroups : 0));
+  if (USE_LDS) {
+    for (int g = threadIdx.x; g < ngroups; g += blockDim.x) {
+      lacc[g] = acc_init<ACC>(op);
+      lcnt[g] = 0;
+    }
+    __syncthreads();
+  }
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t g = row_gid[i];
+    if (op == GB_COUNT_ALL) {
+      if (USE_LDS) atomicAdd((unsigned long long*)&lcnt[g], 1ull);
+      else atomicAdd((unsigned long long*)&cnt[g], 1ull);
+      continue;
+    }
+    if (!valid_bit(vvalid, i)) continue;
+    ACC v = (ACC)vals[i];
+    if (USE_LDS) {
+      if (op != GB_COUNT) acc_atomic<ACC>(op, &lacc[g], v);
+      atomicAdd((unsigned long long*)&lcnt[g], 1ull);
+    } else {
+      if (op != GB_COUNT) acc_atomic<ACC>(op, &acc[g], v);
+      atomicAdd((unsigned long long*)&cnt[g], 1ull);
+    }
+  }
+  if (USE_LDS) {
+    __syncthreads();
+    for (int g = threadIdx.x; g < ngroups; g += blockDim.x) {
+      if (lcnt[g]) {
+        if (op != GB_COUNT && op != GB_COUNT_ALL)
+          acc_atomic<ACC>(op, &acc[g], lacc[g]);
+        atomicAdd((unsigned long long*)&cnt[g], (unsigned long long)lcnt[g]);
+      }
+    }
+  }
+}
+
+// init global accumulators for min/max identities
+template <typename ACC>
+__global__ void k_gb_acc_init(int op, ACC* __restrict__ acc, int32_t ngroups) {
+  for (int g = blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += gridDim.x * blockDim.x)
+    acc[g] = acc_init<ACC>(op);
+}
+
+// bitmask from nonzero counts (output validity of aggregates)
+__global__ void k_mask_from_nonzero(const int64_t* __restrict__ cnt,
+                                    uint64_t* __restrict__ mask,
+                                    int64_t nstripe, int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t s = wave_global; s < nstripe; s += wave_count) {
+    int64_t i = s * WAVE + lane;
+    bool ok = i < n && cnt[i] != 0;
+    uint64_t ballot = __ballot(ok);
+    if (lane == 0) mask[s] = ballot;
+  }
+}
+
+extern "C" {
+
+void hipdf_gb_build(const void* hashes, const void* keys, int nkeys,
+                    void* slot_row, void* row_slot, int64_t cap, int64_t n,
+                    hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_build, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (const int32_t*)hashes, (const KeyCol*)keys, nkeys,
+                     (int32_t*)slot_row, (int32_t*)row_slot,
+                     (uint32_t)(cap - 1), n);
+}
+
+void hipdf_gb_number(const void* slot_row, void* slot_gid, void* ngroups,
+                     void* leaders, int64_t cap, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_number, flat_grid(cap), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)slot_row, (int32_t*)slot_gid,
+                     (int32_t*)ngroups, (int32_t*)leaders, cap);
+}
+
+void hipdf_gb_rowgid(const void* row_slot, const void* slot_gid, void* row_gid,
+                     int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_rowgid, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (const int32_t*)row_slot, (const int32_t*)slot_gid,
+                     (int32_t*)row_gid, n);
+}
+
+void hipdf_gb_agg(int op, int t, const void* vals, const void* vvalid,
+                  const void* row_gid, void* acc, void* cnt, int acc_is_double,
+                  int32_t ngroups, int64_t n, hipStream_t stream) {
+  bool use_lds = ngroups <= GB_LDS_GROUPS;
+  size_t lds = use_lds ? (size_t)ngroups * (8 + 8) : 0;
+  dim3 grid = flat_grid(n, 4);
+  auto launch = [&]<typename T, typename ACC>() {
+    // init global accumulators to the op identity first
+    hipLaunchKernelGGL((k_gb_acc_init<ACC>), flat_grid(ngroups),
+                       dim3(HIPDF_BLOCK), 0, stream, op, (ACC*)acc, ngroups);
+    if (use_lds)
+      hipLaunchKernelGGL((k_gb_agg<T, ACC, true>), grid, dim3(HIPDF_BLOCK),
+                         lds, stream, op, (const T*)vals,
+                         (const uint64_t*)vvalid, (const int32_t*)row_gid,
+                         (ACC*)acc, (int64_t*)cnt, ngroups, n);
+    else
+      hipLaunchKernelGGL((k_gb_agg<T, ACC, false>), grid, dim3(HIPDF_BLOCK),
+                         0, stream, op, (const T*)vals,
+                         (const uint64_t*)vvalid, (const int32_t*)row_gid,
+                         (ACC*)acc, (int64_t*)cnt, ngroups, n);
+  };
+  dispatch_type(t, [&]<typename T>() {
+    if (acc_is_double) launch.template operator()<T, double>();
+    else launch.template operator()<T, int64_t>();
+  });
+}
+
+void hipdf_mask_from_nonzero(const void* cnt, void* mask, int64_t n,
+                             hipStream_t stream) {
+  hipLaunchKernelGGL(k_mask_from_nonzero, stripe_grid(n), dim3(HIPDF_BLOCK),
+                     0, stream, (const int64_t*)cnt, (uint64_t*)mask,
+                     n_stripes(n), n);
+}
+
+}  // extern "C"

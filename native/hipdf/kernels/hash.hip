@@ -1,0 +1,132 @@
+// Spark-compatible murmur3_x86_32 row hashing + partition-id computation.
+// Must produce bit-identical results to the CPU reference
+// (spark_rapids_amd/ops/cpu_backend.py murmur3_hash) and to Spark's
+// Murmur3Hash so GPU shuffle partitioning lines up with CPU Spark
+// (reference analogue: spark-rapids-jni Hash.murmurHash32 — SURVEY.md §2.8B).
+#include "hipdf_common.h"
+
+__device__ __forceinline__ uint32_t rotl32(uint32_t x, int r) {
+  return (x << r) | (x >> (32 - r));
+}
+__device__ __forceinline__ uint32_t mix_k1(uint32_t k1) {
+  k1 *= 0xCC9E2D51u;
+  k1 = rotl32(k1, 15);
+  return k1 * 0x1B873593u;
+}
+__device__ __forceinline__ uint32_t mix_h1(uint32_t h1, uint32_t k1) {
+  h1 ^= k1;
+  h1 = rotl32(h1, 13);
+  return h1 * 5u + 0xE6546B64u;
+}
+__device__ __forceinline__ uint32_t fmix(uint32_t h1, uint32_t len) {
+  h1 ^= len;
+  h1 ^= h1 >> 16;
+  h1 *= 0x85EBCA6Bu;
+  h1 ^= h1 >> 13;
+  h1 *= 0xC2B2AE35u;
+  h1 ^= h1 >> 16;
+  return h1;
+}
+__device__ __forceinline__ uint32_t hash_int(uint32_t v, uint32_t seed) {
+  return fmix(mix_h1(seed, mix_k1(v)), 4);
+}
+__device__ __forceinline__ uint32_t hash_long(uint64_t v, uint32_t seed) {
+  uint32_t h1 = mix_h1(seed, mix_k1((uint32_t)v));
+  h1 = mix_h1(h1, mix_k1((uint32_t)(v >> 32)));
+  return fmix(h1, 8);
+}
+
+// hash kinds: how the column's value maps into murmur input
+enum HashKind : int {
+  HK_INT = 0,   // int8/16/32, bool, date: sign-extended to int32
+  HK_LONG = 1,  // int64 / timestamp / decimal64 unscaled
+  HK_FLOAT = 2,
+  HK_DOUBLE = 3,
+  HK_STRING = 4,
+};
+
+template <typename T>
+__global__ void k_murmur3_col(int kind, const T* __restrict__ a,
+                              const uint64_t* __restrict__ av,
+                              int32_t* __restrict__ seeds, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (!valid_bit(av, i)) continue;  // null keeps previous hash
+    uint32_t seed = (uint32_t)seeds[i];
+    uint32_t h;
+    T v = a[i];
+    if (kind == HK_LONG) {
+      h = hash_long((uint64_t)(int64_t)v, seed);
+    } else if (kind == HK_FLOAT) {
+      float f = (float)v;
+      if (f == 0.0f) f = 0.0f;  // -0.0 -> 0.0
+      h = hash_int(__float_as_uint(f), seed);
+    } else if (kind == HK_DOUBLE) {
+      double d = (double)v;
+      if (d == 0.0) d = 0.0;
+      h = hash_long((uint64_t)__double_as_longlong(d), seed);
+    } else {
+      h = hash_int((uint32_t)(int32_t)(int64_t)v, seed);
+    }
+    seeds[i] = (int32_t)h;
+  }
+}
+
+__global__ void k_murmur3_str(const int32_t* __restrict__ offsets,
+                              const uint8_t* __restrict__ bytes,
+                              const uint64_t* __restrict__ av,
+                              int32_t* __restrict__ seeds, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if (!valid_bit(av, i)) continue;
+    uint32_t h1 = (uint32_t)seeds[i];
+    int32_t s = offsets[i], e = offsets[i + 1];
+    int32_t len = e - s, p = s;
+    // Spark hashUnsafeBytes: 4-byte LE words then SIGNED per-byte tail
+    for (; p + 4 <= e; p += 4) {
+      uint32_t w = (uint32_t)bytes[p] | ((uint32_t)bytes[p + 1] << 8) |
+                   ((uint32_t)bytes[p + 2] << 16) |
+                   ((uint32_t)bytes[p + 3] << 24);
+      h1 = mix_h1(h1, mix_k1(w));
+    }
+    for (; p < e; ++p) h1 = mix_h1(h1, mix_k1((uint32_t)(int32_t)(int8_t)bytes[p]));
+    seeds[i] = (int32_t)fmix(h1, (uint32_t)len);
+  }
+}
+
+// partition id = pmod(hash, nparts) as int32
+__global__ void k_pmod_part(const int32_t* __restrict__ h, int32_t nparts,
+                            int32_t* __restrict__ part, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t r = h[i] % nparts;
+    part[i] = r < 0 ? r + nparts : r;
+  }
+}
+
+extern "C" {
+
+void hipdf_murmur3_col(int kind, int t, const void* a, const void* av,
+                       void* seeds, int64_t n, hipStream_t stream) {
+  dim3 grid = flat_grid(n);
+  dispatch_type(t, [&]<typename T>() {
+    hipLaunchKernelGGL((k_murmur3_col<T>), grid, dim3(HIPDF_BLOCK), 0, stream,
+                       kind, (const T*)a, (const uint64_t*)av,
+                       (int32_t*)seeds, n);
+  });
+}
+
+void hipdf_murmur3_str(const void* offsets, const void* bytes, const void* av,
+                       void* seeds, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_murmur3_str, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)offsets, (const uint8_t*)bytes,
+                     (const uint64_t*)av, (int32_t*)seeds, n);
+}
+
+void hipdf_pmod_part(const void* h, int nparts, void* part, int64_t n,
+                     hipStream_t stream) {
+  hipLaunchKernelGGL(k_pmod_part, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (const int32_t*)h, (int32_t)nparts, (int32_t*)part, n);
+}
+
+}  // extern "C"

@@ -298,6 +298,22 @@ def cast(col: Column, to: DType) -> Column:
         scaled = np.round(a.astype(np.float64) * (10 ** to.scale)) if src.is_floating \
             else a.astype(np.int64) * (10 ** to.scale)
         return _make(scaled.astype(np.int64), av if not av.all() else None, to)
+    if src.is_floating and (to.is_integral or to.id is TypeId.BOOL):
+        if to.id is TypeId.BOOL:
+            res = (a != 0).astype(np.uint8)
+            return _make(res, av if not av.all() else None, to)
+        # Spark non-ANSI: NaN -> 0, saturate at integral bounds, trunc to zero
+        info = np.iinfo(to.numpy_dtype())
+        t = np.trunc(a.astype(np.float64))
+        res = np.zeros(len(t), dtype=to.numpy_dtype())
+        nan = np.isnan(t)
+        big = ~nan & (t >= float(info.max))
+        small = ~nan & (t <= float(info.min))
+        mid = ~(nan | big | small)
+        res[big] = info.max
+        res[small] = info.min
+        res[mid] = t[mid].astype(to.numpy_dtype())
+        return _make(res, av if not av.all() else None, to)
     with np.errstate(invalid="ignore", over="ignore"):
         res = a.astype(to.numpy_dtype())
     return _make(res, av if not av.all() else None, to)

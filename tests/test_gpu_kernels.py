@@ -1,0 +1,314 @@
+"""GPU kernel numerics tests: every hipdf kernel vs the CPU reference
+backend on randomized data with nulls (reference analogue: the
+CPU-vs-GPU equality harness of integration_tests/)."""
+import numpy as np
+import pytest
+
+import spark_rapids_amd as sr
+from spark_rapids_amd import Column, ColumnBatch, DType
+from spark_rapids_amd.ops import cpu_backend
+from spark_rapids_amd.types import FLOAT32, FLOAT64, INT8, INT16, INT32, INT64, STRING
+
+pytestmark = pytest.mark.gpu
+
+RNG = np.random.default_rng(42)
+
+
+def _rand_col(dtype, n=10_000, nulls=0.2, lo=-100, hi=100):
+    valid = RNG.random(n) >= nulls
+    if dtype.id.value == "string":
+        words = ["", "a", "bb", "spark", "rapids", "mi355x", "wörld", "xyzzy"]
+        vals = [None if not v else words[i % len(words)]
+                for i, v in enumerate(valid)]
+        return Column.from_pylist(vals, dtype)
+    if dtype.is_floating:
+        vals = RNG.uniform(lo, hi, n).astype(dtype.numpy_dtype())
+    else:
+        vals = RNG.integers(lo, hi, n).astype(dtype.numpy_dtype())
+    return Column.from_numpy(vals, dtype, valid if nulls else None)
+
+
+def _cols_equal(a: Column, b: Column, approx=False):
+    la, lb = a.to_pylist(), b.to_pylist()
+    assert len(la) == len(lb)
+    for i, (x, y) in enumerate(zip(la, lb)):
+        if x is None or y is None:
+            assert x is None and y is None, f"row {i}: {x} != {y}"
+        elif approx and isinstance(x, float):
+            assert x == pytest.approx(y, rel=1e-12, abs=1e-9), f"row {i}"
+        else:
+            assert x == y, f"row {i}: {x} != {y}"
+
+
+@pytest.mark.parametrize("dtype", [INT32, INT64, FLOAT32, FLOAT64, INT16, INT8])
+@pytest.mark.parametrize("op", ["add", "sub", "mul", "min", "max"])
+def test_binary_arith(dtype, op):
+    a, b = _rand_col(dtype), _rand_col(dtype)
+    cpu = cpu_backend.binary_op(op, a, b, dtype)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.binary_op(op, a.cuda(), b.cuda(), dtype).cpu()
+    _cols_equal(cpu, gpu, approx=dtype.is_floating)
+
+
+@pytest.mark.parametrize("op", ["div", "mod", "pmod", "int_div"])
+def test_binary_null_producing(op):
+    dt = FLOAT64 if op == "div" else INT64
+    a = _rand_col(dt)
+    b = _rand_col(dt, lo=-3, hi=3)  # plenty of zero divisors
+    cpu = cpu_backend.binary_op(op, a, b, dt)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.binary_op(op, a.cuda(), b.cuda(), dt).cpu()
+    _cols_equal(cpu, gpu, approx=True)
+
+
+@pytest.mark.parametrize("op", ["eq", "ne", "lt", "le", "gt", "ge", "eq_null_safe"])
+def test_binary_cmp(op):
+    a = _rand_col(INT32, lo=-5, hi=5)
+    b = _rand_col(INT32, lo=-5, hi=5)
+    cpu = cpu_backend.binary_op(op, a, b, DType.bool_())
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.binary_op(op, a.cuda(), b.cuda(), DType.bool_()).cpu()
+    _cols_equal(cpu, gpu)
+
+
+def test_cmp_nan_semantics():
+    a = Column.from_numpy(np.array([np.nan, 1.0, np.nan, -0.0]), FLOAT64)
+    b = Column.from_numpy(np.array([np.nan, np.nan, 2.0, 0.0]), FLOAT64)
+    from spark_rapids_amd.ops import gpu_backend
+    for op, exp in [("eq", [True, False, False, True]),
+                    ("lt", [False, True, False, False]),
+                    ("gt", [False, False, True, False])]:
+        gpu = gpu_backend.binary_op(op, a.cuda(), b.cuda(), DType.bool_()).cpu()
+        assert gpu.to_pylist() == exp, op
+
+
+def test_kleene_bool():
+    a = Column.from_pylist([True, False, None] * 3, DType.bool_())
+    b = Column.from_pylist([True] * 3 + [False] * 3 + [None] * 3, DType.bool_())
+    from spark_rapids_amd.ops import gpu_backend
+    for op in ("and", "or"):
+        cpu = cpu_backend.binary_op(op, a, b, DType.bool_())
+        gpu = gpu_backend.binary_op(op, a.cuda(), b.cuda(), DType.bool_()).cpu()
+        _cols_equal(cpu, gpu)
+
+
+@pytest.mark.parametrize("op", ["add", "mul", "lt"])
+def test_binary_scalar(op):
+    a = _rand_col(INT64)
+    out_t = DType.bool_() if op == "lt" else INT64
+    cpu = cpu_backend.binary_op_scalar(op, a, 7, out_t)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.binary_op_scalar(op, a.cuda(), 7, out_t).cpu()
+    _cols_equal(cpu, gpu)
+
+
+@pytest.mark.parametrize("op", ["neg", "abs", "sqrt", "exp", "log", "floor", "ceil"])
+def test_unary(op):
+    dt = FLOAT64
+    a = _rand_col(dt)
+    cpu = cpu_backend.unary_op(op, a, dt)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.unary_op(op, a.cuda(), dt).cpu()
+    _cols_equal(cpu, gpu, approx=True)
+
+
+@pytest.mark.parametrize("src,dst", [
+    (FLOAT64, INT64), (FLOAT64, INT32), (INT64, FLOAT64), (INT32, INT64),
+    (INT64, INT32), (FLOAT32, FLOAT64), (INT32, DType.bool_()),
+])
+def test_cast(src, dst):
+    a = _rand_col(src, lo=-1000, hi=1000)
+    cpu = cpu_backend.cast(a, dst)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.cast(a.cuda(), dst).cpu()
+    _cols_equal(cpu, gpu, approx=dst.is_floating)
+
+
+def test_cast_nan_saturation():
+    a = Column.from_numpy(np.array([np.nan, np.inf, -np.inf, 1e20, -1e20, 2.9]),
+                          FLOAT64)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.cast(a.cuda(), INT32).cpu().to_pylist()
+    cpu = cpu_backend.cast(a, INT32).to_pylist()
+    assert gpu == cpu == [0, 2**31 - 1, -2**31, 2**31 - 1, -2**31, 2]
+
+
+def test_filter_and_gather_with_strings():
+    batch = ColumnBatch([
+        _rand_col(INT64), _rand_col(FLOAT64), _rand_col(STRING),
+    ])
+    mask = _rand_col(DType.bool_(), nulls=0.1, lo=0, hi=2)
+    cpu = cpu_backend.apply_boolean_mask(batch, mask)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.apply_boolean_mask(batch.cuda(), mask.cuda()).cpu()
+    assert cpu.num_rows == gpu.num_rows
+    for c, g in zip(cpu.columns, gpu.columns):
+        _cols_equal(c, g, approx=True)
+
+
+def test_gather_negative_indices_nullify():
+    batch = ColumnBatch([_rand_col(INT64, n=100, nulls=0.0)])
+    idx = Column.from_numpy(np.array([0, -1, 5, 99, -1], dtype=np.int32))
+    cpu = cpu_backend.gather(batch, idx)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.gather(batch.cuda(), idx.cuda()).cpu()
+    for c, g in zip(cpu.columns, gpu.columns):
+        _cols_equal(c, g)
+
+
+def test_concat_batches():
+    batches = [ColumnBatch([_rand_col(INT64, n=n), _rand_col(STRING, n=n)])
+               for n in (100, 37, 1, 200)]
+    cpu = cpu_backend.concat_batches(batches)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.concat_batches([b.cuda() for b in batches]).cpu()
+    for c, g in zip(cpu.columns, gpu.columns):
+        _cols_equal(c, g)
+
+
+@pytest.mark.parametrize("dtype", [INT32, INT64, FLOAT32, FLOAT64, STRING, INT8])
+def test_murmur3_matches_cpu(dtype):
+    c = _rand_col(dtype)
+    cpu = cpu_backend.murmur3_hash([c], 42)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.murmur3_hash([c.cuda()], 42).cpu()
+    _cols_equal(cpu, gpu)
+
+
+def test_murmur3_multi_column_chain():
+    cols = [_rand_col(INT64), _rand_col(INT32), _rand_col(FLOAT64)]
+    cpu = cpu_backend.murmur3_hash(cols, 42)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.murmur3_hash([c.cuda() for c in cols], 42).cpu()
+    _cols_equal(cpu, gpu)
+
+
+def test_hash_partition_agrees_with_cpu():
+    batch = ColumnBatch([_rand_col(INT64), _rand_col(FLOAT64)])
+    nparts = 16
+    cpu, cpu_offs = cpu_backend.hash_partition(batch, [0], nparts)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu, gpu_offs = gpu_backend.hash_partition(batch.cuda(), [0], nparts)
+    gpu = gpu.cpu()
+    assert cpu_offs == gpu_offs
+    # same rows in each partition (order within a partition may differ)
+    for p in range(nparts):
+        cs = sorted(zip(*[c.to_pylist()[cpu_offs[p]:cpu_offs[p + 1]]
+                          for c in cpu.columns]), key=repr)
+        gs = sorted(zip(*[c.to_pylist()[gpu_offs[p]:gpu_offs[p + 1]]
+                          for c in gpu.columns]), key=repr)
+        assert cs == gs
+
+
+@pytest.mark.parametrize("op", ["sum", "min", "max", "count", "mean"])
+def test_reduce(op):
+    for dt in (INT64, FLOAT64):
+        c = _rand_col(dt)
+        cpu = cpu_backend.reduce(op, c)
+        from spark_rapids_amd.ops import gpu_backend
+        gpu = gpu_backend.reduce(op, c.cuda())
+        if isinstance(cpu, float):
+            assert gpu == pytest.approx(cpu)
+        else:
+            assert gpu == cpu
+
+
+def _sorted_rows(batch: ColumnBatch):
+    return sorted(zip(*[c.to_pylist() for c in batch.columns]), key=repr)
+
+
+@pytest.mark.parametrize("nkeys,ngroups", [(1, 7), (1, 5000), (2, 100)])
+def test_group_by_aggregate(nkeys, ngroups):
+    n = 50_000
+    keys = [Column.from_numpy(
+        RNG.integers(0, ngroups, n).astype(np.int64), INT64,
+        RNG.random(n) >= 0.05) for _ in range(nkeys)]
+    vals = _rand_col(FLOAT64, n=n)
+    ivals = _rand_col(INT64, n=n)
+    batch = ColumnBatch(keys + [vals, ivals])
+    aggs = [("sum", nkeys, FLOAT64), ("count", nkeys, INT64),
+            ("min", nkeys + 1, INT64), ("max", nkeys + 1, INT64),
+            ("sum", nkeys + 1, INT64), ("count_all", -1, INT64)]
+    cpu = cpu_backend.group_by_aggregate(batch, list(range(nkeys)), aggs)
+    from spark_rapids_amd.ops import gpu_backend
+    gpu = gpu_backend.group_by_aggregate(batch.cuda(), list(range(nkeys)),
+                                         aggs).cpu()
+    assert cpu.num_rows == gpu.num_rows
+    crows = _sorted_rows(cpu)
+    grows = _sorted_rows(gpu)
+    for cr, gr in zip(crows, grows):
+        for x, y in zip(cr, gr):
+            if isinstance(x, float) and y is not None and x is not None:
+                assert x == pytest.approx(y, rel=1e-9), (cr, gr)
+            else:
+                assert x == y, (cr, gr)
+
+
+@pytest.mark.parametrize("how", ["inner", "left", "semi", "anti"])
+def test_join(how):
+    n = 20_000
+    left = ColumnBatch([
+        Column.from_numpy(RNG.integers(0, 1000, n).astype(np.int64), INT64,
+                          RNG.random(n) >= 0.05),
+        _rand_col(FLOAT64, n=n),
+    ])
+    right = ColumnBatch([
+        Column.from_numpy(RNG.integers(0, 1500, 5000).astype(np.int64), INT64,
+                          RNG.random(5000) >= 0.05),
+        _rand_col(INT32, n=5000),
+    ])
+    from spark_rapids_amd.ops import cpu_backend as cb, gpu_backend as gb
+    lcpu, rcpu = cb.join_gather_maps(left, right, [0], [0], how)
+    lgpu, rgpu = gb.join_gather_maps(left.cuda(), right.cuda(), [0], [0], how)
+    if how in ("inner", "left"):
+        cpu_out = ColumnBatch(cb.gather(left, lcpu).columns +
+                              cb.gather(right, rcpu).columns)
+        gpu_out = ColumnBatch(
+            gb.gather(left.cuda(), lgpu).columns +
+            gb.gather(right.cuda(), rgpu).columns).cpu()
+    else:
+        cpu_out = cb.gather(left, lcpu)
+        gpu_out = gb.gather(left.cuda(), lgpu).cpu()
+    assert cpu_out.num_rows == gpu_out.num_rows, how
+    assert _sorted_rows(cpu_out) == _sorted_rows(gpu_out)
+
+
+def test_end_to_end_query_gpu_vs_cpu():
+    n = 100_000
+    data = {
+        "k": RNG.integers(0, 50, n).astype(np.int64),
+        "v": RNG.uniform(0, 100, n),
+        "w": RNG.integers(-10, 10, n).astype(np.int32),
+    }
+
+    def run(enabled):
+        s = sr.Session({"spark.rapids.sql.enabled": enabled})
+        df = s.create_dataframe(dict(data), num_partitions=4)
+        return (df.filter((sr.col("v") > 10.0) & (sr.col("w") != 0))
+                  .with_column("vw", sr.col("v") * sr.col("w").cast(sr.FLOAT64))
+                  .group_by("k")
+                  .agg(sr.sum_(sr.col("vw")), sr.count_star(),
+                       sr.avg(sr.col("v")), sr.min_(sr.col("w")))
+                  .sort("k").collect())
+
+    gpu, cpu = run(True), run(False)
+    assert len(gpu) == len(cpu)
+    for g, c in zip(gpu, cpu):
+        assert g[0] == c[0] and g[2] == c[2] and g[4] == c[4]
+        assert g[1] == pytest.approx(c[1], rel=1e-9)
+        assert g[3] == pytest.approx(c[3], rel=1e-9)
+
+
+def test_gpu_plan_placement():
+    s = sr.Session()
+    df = s.create_dataframe({"a": [1, 2, 3]})
+    tree = df.filter(sr.col("a") > 1).physical_plan().tree_string()
+    assert "GpuFilter" in tree, tree
+
+
+def test_retry_split_on_gpu():
+    from spark_rapids_amd.memory.retry import oom_injector
+    s = sr.Session()
+    df = s.create_dataframe({"a": list(range(1000))})
+    oom_injector.arm(1, split=True)
+    assert df.filter(sr.col("a") >= 500).count() == 500
