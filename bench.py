@@ -1,0 +1,163 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: NDS-like power run on N MI355X GPUs (weak scaling).
+
+Driver contract: `python bench.py --gpus N --steps K --warmup W` (launched
+under torch.distributed.run for N>1, one rank per GPU over RCCL). Each rank
+owns a fixed-size synthetic partition of an NDS-style star schema
+(BASELINE.json metric: NDS power-run wall-clock + speedup vs CPU Spark; the
+CPU baseline here is this engine's own CPU backend on identical data — the
+reference's in-repo baseline number is its 3.0x default operator speedup,
+tools/generated_files/operatorsScore.csv).
+
+One step = the full power-run query suite executed on-GPU via hipdf kernels.
+Timing: barrier + torch.cuda.synchronize on both sides, MAX over ranks.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import sys
+import time
+
+sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
+
+import torch
+
+from spark_rapids_amd import Session
+from spark_rapids_amd.api import MemTable
+from spark_rapids_amd.bench import datagen
+from spark_rapids_amd.bench.queries import POWER_RUN, run_power
+from spark_rapids_amd.plan import logical as L
+
+
+def _dist_env():
+    rank = int(os.environ.get("RANK", "0"))
+    world = int(os.environ.get("WORLD_SIZE", "1"))
+    local_rank = int(os.environ.get("LOCAL_RANK", "0"))
+    return rank, world, local_rank
+
+
+def _make_tables(session, rows: int, seed: int, device: str,
+                 partitions: int):
+    per = rows // partitions
+    fact_batches = [datagen.gen_fact_partition(per, seed * 1000 + p)
+                    for p in range(partitions)]
+    if device == "cuda":
+        fact_batches = [b.cuda() for b in fact_batches]
+    items = datagen.gen_items()
+    stores = datagen.gen_stores()
+    if device == "cuda":
+        items, stores = items.cuda(), stores.cuda()
+    return {
+        "store_sales": session.from_batches(fact_batches,
+                                            datagen.fact_schema(), "store_sales"),
+        "item": session.from_batches([items], datagen.item_schema(), "item"),
+        "store": session.from_batches([stores], datagen.store_schema(), "store"),
+    }
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--gpus", type=int, default=1)
+    ap.add_argument("--steps", type=int, default=5)
+    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--rows-per-gpu", type=int, default=20_000_000)
+    ap.add_argument("--partitions", type=int, default=8)
+    ap.add_argument("--cpu-baseline-steps", type=int, default=1,
+                    help="0 disables the CPU-backend baseline measurement")
+    args = ap.parse_args()
+
+    rank, world, local_rank = _dist_env()
+    n_gpus = max(world, 1)
+    distributed = world > 1
+
+    use_gpu = torch.cuda.is_available()
+    if use_gpu:
+        torch.cuda.set_device(local_rank)
+    if distributed:
+        torch.distributed.init_process_group(
+            backend="nccl" if use_gpu else "gloo")
+
+    device = "cuda" if use_gpu else "cpu"
+    session = Session({"spark.rapids.sql.enabled": use_gpu})
+    tables = _make_tables(session, args.rows_per_gpu, seed=rank + 1,
+                          device=device, partitions=args.partitions)
+
+    def barrier_sync():
+        if distributed:
+            torch.distributed.barrier()
+        if use_gpu:
+            torch.cuda.synchronize()
+
+    # ---- warmup ----
+    for _ in range(args.warmup):
+        run_power(tables)
+    barrier_sync()
+
+    # ---- timed region ----
+    t0 = time.perf_counter()
+    for _ in range(args.steps):
+        run_power(tables)
+    barrier_sync()
+    elapsed = time.perf_counter() - t0
+
+    # MAX over ranks
+    if distributed:
+        t = torch.tensor([elapsed], dtype=torch.float64,
+                         device=device if use_gpu else "cpu")
+        torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
+        elapsed = float(t.item())
+
+    ms_per_step = elapsed / args.steps * 1000.0
+    # whole-job aggregate: fact rows scanned per second across all queries
+    rows_per_step = args.rows_per_gpu * len(POWER_RUN) * n_gpus
+    value = rows_per_step * args.steps / elapsed
+
+    # ---- CPU baseline (rank 0, once, same per-GPU data size) ----
+    speedup = None
+    if rank == 0 and args.cpu_baseline_steps > 0:
+        cpu_session = Session({"spark.rapids.sql.enabled": False})
+        cpu_tables = _make_tables(cpu_session, args.rows_per_gpu, seed=1,
+                                  device="cpu", partitions=args.partitions)
+        tc0 = time.perf_counter()
+        for _ in range(args.cpu_baseline_steps):
+            run_power(cpu_tables)
+        tcpu = (time.perf_counter() - tc0) / args.cpu_baseline_steps
+        speedup = tcpu / (elapsed / args.steps)
+
+    if rank == 0:
+        result = {
+            "metric": "NDS-like power-run fact rows/s (speedup vs CPU in config)",
+            "value": value,
+            "unit": "rows/s",
+            "n_gpus": n_gpus,
+            "steps": args.steps,
+            "warmup": args.warmup,
+            "ms_per_step": ms_per_step,
+            "higher_is_better": True,
+            "scaling": "weak",
+            "vs_baseline": (speedup / 3.0) if speedup is not None else None,
+            "dtype": "fp64",
+            "data": "synthetic",
+            "config": {
+                "model": "nds_like_power_run_5q",
+                "global_batch": args.rows_per_gpu * n_gpus,
+                "seq_len": None,
+                "parallelism": f"dp{n_gpus}",
+                "rows_per_gpu": args.rows_per_gpu,
+                "queries": [q for q, _ in POWER_RUN],
+                "speedup_vs_cpu_backend": speedup,
+                "baseline_definition": "vs_baseline = (cpu_backend_time / "
+                "gpu_time) / 3.0, where 3.0x is the reference's default "
+                "operator speedup score (BASELINE.md)",
+            },
+        }
+        print(json.dumps(result))
+
+    if distributed:
+        torch.distributed.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()

@@ -1,0 +1,81 @@
+"""NDS-like power-run query suite over the synthetic star schema.
+
+Query shapes mirror the reference's headline workloads (SURVEY.md §6 /
+BASELINE.json configs): scan->filter->project->hash aggregate (TPC-H q1 /
+TPC-DS q3 shape), join->aggregate (q5 shape), selective filter revenue (q6
+shape), and a high-cardinality customer rollup (exercises the big hash
+table + repartition paths).
+"""
+from __future__ import annotations
+
+from typing import Callable, Dict, List
+
+from ..api import DataFrame, Session
+from ..expr.aggregates import avg, count_star, max_, min_, sum_
+from ..expr.expressions import col, lit
+from ..types import FLOAT64, INT64
+
+
+def q1_pricing_summary(t: Dict[str, DataFrame]) -> DataFrame:
+    ss = t["store_sales"]
+    return (ss.filter(col("ss_sold_date") <= lit(10_900))
+            .group_by("ss_promo")
+            .agg(sum_(col("ss_quantity")),
+                 sum_(col("ss_sales_price")),
+                 sum_(col("ss_sales_price") * (lit(1.0) - col("ss_discount"))),
+                 avg(col("ss_list_price")),
+                 avg(col("ss_discount")),
+                 count_star()))
+
+
+def q2_join_agg(t: Dict[str, DataFrame]) -> DataFrame:
+    ss, item = t["store_sales"], t["item"]
+    return (ss.join(item, on="ss_item_id", right_on=["i_item_id"])
+            .filter(col("i_category") < 3)
+            .group_by("ss_store_id")
+            .agg(sum_(col("ss_sales_price")), count_star()))
+
+
+def q3_selective_revenue(t: Dict[str, DataFrame]) -> DataFrame:
+    ss = t["store_sales"]
+    return (ss.filter((col("ss_sold_date") >= 10_200)
+                      & (col("ss_sold_date") < 10_565)
+                      & (col("ss_discount") >= 0.05)
+                      & (col("ss_discount") <= 0.07)
+                      & (col("ss_quantity") < 24))
+            .agg(sum_(col("ss_list_price") * col("ss_discount"))))
+
+
+def q4_customer_rollup(t: Dict[str, DataFrame]) -> DataFrame:
+    ss = t["store_sales"]
+    return (ss.group_by("ss_customer_id")
+            .agg(sum_(col("ss_sales_price")), count_star())
+            .filter(col("sum(ss_sales_price)") > 4000.0)
+            .agg(count_star()))
+
+
+def q5_store_join(t: Dict[str, DataFrame]) -> DataFrame:
+    ss, store, item = t["store_sales"], t["store"], t["item"]
+    return (ss.join(store, on="ss_store_id", right_on=["s_store_id"])
+            .join(item, on="ss_item_id", right_on=["i_item_id"])
+            .filter(col("s_state") < 25)
+            .group_by("s_state", "i_category")
+            .agg(sum_(col("ss_sales_price") * col("ss_quantity").cast(FLOAT64)),
+                 count_star()))
+
+
+POWER_RUN: List = [
+    ("q1", q1_pricing_summary),
+    ("q2", q2_join_agg),
+    ("q3", q3_selective_revenue),
+    ("q4", q4_customer_rollup),
+    ("q5", q5_store_join),
+]
+
+
+def run_power(tables: Dict[str, DataFrame]) -> List[tuple]:
+    """Run the suite; returns the collected result rows (forces execution)."""
+    out = []
+    for name, fn in POWER_RUN:
+        out.append((name, fn(tables).collect()))
+    return out

@@ -298,10 +298,10 @@ def cast(col: Column, to: DType) -> Column:
         scaled = np.round(a.astype(np.float64) * (10 ** to.scale)) if src.is_floating \
             else a.astype(np.int64) * (10 ** to.scale)
         return _make(scaled.astype(np.int64), av if not av.all() else None, to)
-    if src.is_floating and (to.is_integral or to.id is TypeId.BOOL):
-        if to.id is TypeId.BOOL:
-            res = (a != 0).astype(np.uint8)
-            return _make(res, av if not av.all() else None, to)
+    if to.id is TypeId.BOOL:
+        res = (a != 0).astype(np.uint8)
+        return _make(res, av if not av.all() else None, to)
+    if src.is_floating and to.is_integral:
         # Spark non-ANSI: NaN -> 0, saturate at integral bounds, trunc to zero
         info = np.iinfo(to.numpy_dtype())
         t = np.trunc(a.astype(np.float64))
@@ -542,7 +542,16 @@ def _group_codes(keys: List[Column]) -> Tuple[np.ndarray, np.ndarray]:
                                   return_inverse=True)
             arrs.append(inv.astype(np.int64))
         else:
-            arrs.append(a.view(np.int64) if a.dtype.itemsize == 8 else a.astype(np.int64))
+            code = a.view(np.int64) if a.dtype.itemsize == 8 else a.astype(np.int64)
+            # normalize values under NULLs so all-null keys form one group,
+            # and fold -0.0 / NaN payloads like the GPU grouping semantics
+            if a.dtype.kind == "f":
+                af = a.astype(np.float64).copy()
+                af[af == 0.0] = 0.0
+                af[np.isnan(af)] = np.nan
+                code = af.view(np.int64)
+            code = np.where(av, code, 0)
+            arrs.append(code)
             arrs.append(av.astype(np.int64))
     stacked = np.stack(arrs, axis=1) if arrs else np.zeros((len(keys[0]._vals), 0))
     uniq, first_idx, codes = np.unique(stacked, axis=0, return_index=True,

@@ -210,6 +210,9 @@ def promote(a: DType, b: DType) -> DType:
         # decimal + integral -> decimal with enough precision; handled by caller
         d = a if a.is_decimal else b
         return d
+    if a.is_timelike or b.is_timelike:
+        # date/timestamp vs integral literal: compare in the timelike domain
+        return a if a.is_timelike else b
     ia = _PROMOTION_ORDER.index(a.id)
     ib = _PROMOTION_ORDER.index(b.id)
     return DType(_PROMOTION_ORDER[max(ia, ib)])

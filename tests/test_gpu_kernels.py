@@ -35,7 +35,10 @@ def _cols_equal(a: Column, b: Column, approx=False):
         if x is None or y is None:
             assert x is None and y is None, f"row {i}: {x} != {y}"
         elif approx and isinstance(x, float):
-            assert x == pytest.approx(y, rel=1e-12, abs=1e-9), f"row {i}"
+            if np.isnan(x) or np.isnan(y):
+                assert np.isnan(x) and np.isnan(y), f"row {i}: {x} != {y}"
+            else:
+                assert x == pytest.approx(y, rel=1e-12, abs=1e-9), f"row {i}"
         else:
             assert x == y, f"row {i}: {x} != {y}"
 
