@@ -57,9 +57,10 @@ void hipdf_murmur3_col(int, int, const void*, const void*, void*, int64_t,
 void hipdf_murmur3_str(const void*, const void*, const void*, void*, int64_t,
                        hipStream_t);
 void hipdf_pmod_part(const void*, int, void*, int64_t, hipStream_t);
-void hipdf_gb_build(const void*, const void*, int, void*, void*, int64_t,
-                    int64_t, hipStream_t);
-void hipdf_gb_number(const void*, void*, void*, void*, int64_t, hipStream_t);
+void hipdf_gb_build(const void*, const void*, int, void*, void*, void*,
+                    void*, int64_t, int64_t, hipStream_t);
+void hipdf_gb_number(const void*, const void*, void*, const void*, void*,
+                     int64_t, hipStream_t);
 void hipdf_gb_rowgid(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_gb_agg(int, int, const void*, const void*, const void*, void*,
                   void*, int, int32_t, int64_t, hipStream_t);
@@ -236,16 +237,18 @@ PYBIND11_MODULE(hipdf, m) {
   });
 
   m.def("gb_build", [](int64_t hashes, int64_t keys, int nkeys,
-                       int64_t slot_row, int64_t row_slot, int64_t cap,
+                       int64_t slot_row, int64_t row_slot,
+                       int64_t claimed_slots, int64_t ngroups, int64_t cap,
                        int64_t n, int64_t stream) {
-    hipdf_gb_build(P(hashes), P(keys), nkeys, PM(slot_row), PM(row_slot), cap,
-                   n, S(stream));
+    hipdf_gb_build(P(hashes), P(keys), nkeys, PM(slot_row), PM(row_slot),
+                   PM(claimed_slots), PM(ngroups), cap, n, S(stream));
     check_async();
   });
-  m.def("gb_number", [](int64_t slot_row, int64_t slot_gid, int64_t ngroups,
-                        int64_t leaders, int64_t cap, int64_t stream) {
-    hipdf_gb_number(P(slot_row), PM(slot_gid), PM(ngroups), PM(leaders), cap,
-                    S(stream));
+  m.def("gb_number", [](int64_t claimed_slots, int64_t slot_row,
+                        int64_t slot_gid, int64_t ngroups, int64_t leaders,
+                        int64_t max_groups, int64_t stream) {
+    hipdf_gb_number(P(claimed_slots), P(slot_row), PM(slot_gid), P(ngroups),
+                    PM(leaders), max_groups, S(stream));
     check_async();
   });
   m.def("gb_rowgid", [](int64_t row_slot, int64_t slot_gid, int64_t row_gid,

@@ -21,15 +21,23 @@
 __global__ void k_gb_build(const int32_t* __restrict__ hashes,
                            const KeyCol* __restrict__ keys, int nkeys,
                            int32_t* __restrict__ slot_row,
-                           int32_t* __restrict__ row_slot, uint32_t slot_mask,
+                           int32_t* __restrict__ row_slot,
+                           int32_t* __restrict__ claimed_slots,
+                           int32_t* __restrict__ ngroups, uint32_t slot_mask,
                            int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t slot = slot_of((uint32_t)hashes[i], slot_mask);
     while (true) {
       int32_t cur = atomicCAS(&slot_row[slot], GB_EMPTY, (int32_t)i);
-      if (cur == GB_EMPTY || cur == (int32_t)i ||
-          rows_equal(keys, keys, nkeys, i, cur)) {
+      if (cur == GB_EMPTY) {
+        // claimed a new group: record its slot so numbering touches only
+        // live slots instead of sweeping the whole table
+        claimed_slots[atomicAdd(ngroups, 1)] = (int32_t)slot;
+        row_slot[i] = (int32_t)slot;
+        break;
+      }
+      if (cur == (int32_t)i || rows_equal(keys, keys, nkeys, i, cur)) {
         row_slot[i] = (int32_t)slot;
         break;
       }
@@ -38,18 +46,17 @@ __global__ void k_gb_build(const int32_t* __restrict__ hashes,
   }
 }
 
-__global__ void k_gb_number(const int32_t* __restrict__ slot_row,
+__global__ void k_gb_number(const int32_t* __restrict__ claimed_slots,
+                            const int32_t* __restrict__ slot_row,
                             int32_t* __restrict__ slot_gid,
-                            int32_t* __restrict__ ngroups,
-                            int32_t* __restrict__ leaders, int64_t cap) {
-  for (int64_t s = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; s < cap;
-       s += (int64_t)gridDim.x * blockDim.x) {
-    int32_t r = slot_row[s];
-    if (r != GB_EMPTY) {
-      int32_t gid = atomicAdd(ngroups, 1);
-      slot_gid[s] = gid;
-      leaders[gid] = r;
-    }
+                            const int32_t* __restrict__ ngroups,
+                            int32_t* __restrict__ leaders) {
+  int32_t ng = *ngroups;
+  for (int64_t g = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; g < ng;
+       g += (int64_t)gridDim.x * blockDim.x) {
+    int32_t s = claimed_slots[g];
+    slot_gid[s] = (int32_t)g;
+    leaders[g] = slot_row[s];
   }
 }
 
@@ -175,19 +182,23 @@ __global__ void k_mask_from_nonzero(const int64_t* __restrict__ cnt,
 extern "C" {
 
 void hipdf_gb_build(const void* hashes, const void* keys, int nkeys,
-                    void* slot_row, void* row_slot, int64_t cap, int64_t n,
+                    void* slot_row, void* row_slot, void* claimed_slots,
+                    void* ngroups, int64_t cap, int64_t n,
                     hipStream_t stream) {
   hipLaunchKernelGGL(k_gb_build, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
                      (const int32_t*)hashes, (const KeyCol*)keys, nkeys,
                      (int32_t*)slot_row, (int32_t*)row_slot,
+                     (int32_t*)claimed_slots, (int32_t*)ngroups,
                      (uint32_t)(cap - 1), n);
 }
 
-void hipdf_gb_number(const void* slot_row, void* slot_gid, void* ngroups,
-                     void* leaders, int64_t cap, hipStream_t stream) {
-  hipLaunchKernelGGL(k_gb_number, flat_grid(cap), dim3(HIPDF_BLOCK), 0,
-                     stream, (const int32_t*)slot_row, (int32_t*)slot_gid,
-                     (int32_t*)ngroups, (int32_t*)leaders, cap);
+void hipdf_gb_number(const void* claimed_slots, const void* slot_row,
+                     void* slot_gid, const void* ngroups, void* leaders,
+                     int64_t max_groups, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_number, flat_grid(max_groups), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)claimed_slots,
+                     (const int32_t*)slot_row, (int32_t*)slot_gid,
+                     (const int32_t*)ngroups, (int32_t*)leaders);
 }
 
 void hipdf_gb_rowgid(const void* row_slot, const void* slot_gid, void* row_gid,

@@ -182,6 +182,26 @@ class Column:
         return Column.from_numpy(dense, dtype, None if valid.all() else valid, device)
 
     @staticmethod
+    def full(value, dtype: DType, size: int, device: str = "cpu") -> "Column":
+        """Broadcast a non-null scalar to a column without a python list."""
+        assert value is not None
+        if dtype.id is TypeId.STRING:
+            b = str(value).encode("utf-8")
+            data = torch.from_numpy(
+                np.frombuffer(b * size, dtype=np.uint8).copy()) if size else \
+                torch.zeros(0, dtype=torch.uint8)
+            offsets = torch.arange(0, (size + 1) * len(b) or 1, len(b) or 1,
+                                   dtype=torch.int32)[: size + 1] if b else \
+                torch.zeros(size + 1, dtype=torch.int32)
+            col = Column(dtype, size, data, None, offsets, 0)
+            return col.to(device) if device != "cpu" else col
+        if dtype.id is TypeId.BOOL:
+            value = int(bool(value))
+        data = torch.full((size,), value, dtype=torch_dtype(dtype),
+                          device=device)
+        return Column(dtype, size, data, None, null_count=0)
+
+    @staticmethod
     def nulls(dtype: DType, size: int, device: str = "cpu") -> "Column":
         data = torch.zeros(size, dtype=torch_dtype(dtype), device=device)
         validity = torch.zeros(mask_nbytes(size), dtype=torch.uint8, device=device)

@@ -162,8 +162,8 @@ class Literal(Expression):
         if self.value is None:
             return Column.nulls(self._dtype if self._dtype.id is not TypeId.NULL
                                 else INT32, batch.num_rows, batch.device)
-        return Column.from_pylist([self.value] * batch.num_rows, self._dtype,
-                                  batch.device)
+        return Column.full(self.value, self._dtype, batch.num_rows,
+                           batch.device)
 
     def __str__(self):
         return repr(self.value)
@@ -207,6 +207,11 @@ class BinaryExpr(Expression):
             return DType.decimal(min(it.precision + 1, 38), it.scale)
         return it
 
+    # ops where `scalar OP col` can run through the col-scalar kernel
+    _COMMUTATIVE = {"add", "mul", "eq", "ne", "eq_null_safe", "and", "or",
+                    "bitand", "bitor", "bitxor", "min", "max"}
+    _SWAP_CMP = {"lt": "gt", "gt": "lt", "le": "ge", "ge": "le"}
+
     def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
         common = self._in_dtype(schema)
         if self.op in _DOUBLE_OPS and not common.is_decimal:
@@ -217,6 +222,20 @@ class BinaryExpr(Expression):
                 and not common.is_decimal:
             lcol = ops.cast(self.left.eval(batch, schema), common)
             return ops.binary_op_scalar(self.op, lcol, _coerce_py(self.right.value, common), out)
+        if isinstance(self.left, Literal) and self.left.value is not None \
+                and not common.is_decimal:
+            swapped = self.op if self.op in self._COMMUTATIVE \
+                else self._SWAP_CMP.get(self.op)
+            if swapped is not None:
+                rcol = ops.cast(self.right.eval(batch, schema), common)
+                return ops.binary_op_scalar(
+                    swapped, rcol, _coerce_py(self.left.value, common), out)
+            if self.op == "sub":
+                # c - x == -(x - c): two scalar kernels, no literal column
+                rcol = ops.cast(self.right.eval(batch, schema), common)
+                d = ops.binary_op_scalar(
+                    "sub", rcol, _coerce_py(self.left.value, common), out)
+                return ops.unary_op("neg", d, out)
         lcol = ops.cast(self.left.eval(batch, schema), common)
         rcol = ops.cast(self.right.eval(batch, schema), common)
         return ops.binary_op(self.op, lcol, rcol, out)

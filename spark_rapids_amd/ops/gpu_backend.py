@@ -521,13 +521,16 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         desc = _key_desc(keys)
         slot_row = torch.full((cap,), -1, dtype=torch.int32, device="cuda")
         row_slot = torch.empty(n, dtype=torch.int32, device="cuda")
-        ext.gb_build(h.data_ptr(), desc.data_ptr(), len(keys),
-                     slot_row.data_ptr(), row_slot.data_ptr(), cap, n, s)
-        slot_gid = torch.full((cap,), -1, dtype=torch.int32, device="cuda")
+        claimed = torch.empty(n, dtype=torch.int32, device="cuda")
         ngroups_t = torch.zeros(1, dtype=torch.int32, device="cuda")
+        ext.gb_build(h.data_ptr(), desc.data_ptr(), len(keys),
+                     slot_row.data_ptr(), row_slot.data_ptr(),
+                     claimed.data_ptr(), ngroups_t.data_ptr(), cap, n, s)
+        slot_gid = torch.empty(cap, dtype=torch.int32, device="cuda")
         leaders = torch.empty(n, dtype=torch.int32, device="cuda")
-        ext.gb_number(slot_row.data_ptr(), slot_gid.data_ptr(),
-                      ngroups_t.data_ptr(), leaders.data_ptr(), cap, s)
+        ext.gb_number(claimed.data_ptr(), slot_row.data_ptr(),
+                      slot_gid.data_ptr(), ngroups_t.data_ptr(),
+                      leaders.data_ptr(), n, s)
         ngroups = int(ngroups_t.item())
         row_gid = torch.empty(n, dtype=torch.int32, device="cuda")
         ext.gb_rowgid(row_slot.data_ptr(), slot_gid.data_ptr(),
