@@ -61,8 +61,21 @@ _GB = {"sum": 0, "min": 1, "max": 2, "count": 3, "count_all": 4}
 _JOIN = {"inner": 0, "left": 1, "semi": 2, "anti": 3}
 
 
+_cached_stream: Optional[int] = None
+
+
 def _stream() -> int:
-    return torch.cuda.current_stream().cuda_stream
+    # one stream per process (one process per GPU); refreshed only via
+    # set_stream when the engine switches streams for copy/compute overlap
+    global _cached_stream
+    if _cached_stream is None:
+        _cached_stream = torch.cuda.current_stream().cuda_stream
+    return _cached_stream
+
+
+def set_stream(handle: Optional[int]) -> None:
+    global _cached_stream
+    _cached_stream = handle
 
 
 def _ptr(t: Optional[torch.Tensor]) -> int:

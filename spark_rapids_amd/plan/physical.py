@@ -187,6 +187,22 @@ class HashAggregateExec(PhysicalExec):
             return
         merged_in = ops.concat_batches(partial_results) if len(partial_results) > 1 \
             else partial_results[0]
+
+        # distributed merge: keyed -> RCCL all-to-all hash exchange so each
+        # rank owns a disjoint key range; keyless -> all-gather partials and
+        # merge identically on every rank
+        from ..shuffle import dist as _dist
+        if _dist.ctx().is_multi:
+            from ..shuffle.exchange import exchange_by_hash, gather_all
+            if nkeys:
+                received = exchange_by_hash(merged_in, list(range(nkeys)))
+            else:
+                received = gather_all(merged_in)
+            received = [b for b in received if b.num_rows]
+            if not received:
+                return
+            merged_in = ops.concat_batches(received) if len(received) > 1 \
+                else received[0]
         merge_specs = [(op, nkeys + j, partial[j][2])
                        for j, op in enumerate(merge_ops)]
         merged = ops.group_by_aggregate(merged_in, list(range(nkeys)), merge_specs)

@@ -1,0 +1,49 @@
+"""Device-to-device shuffle exchange over RCCL/xGMI.
+
+Reference analogue: GpuShuffleExchangeExecBase + GpuHashPartitioningBase
+(murmur3 -> pmod -> partition -> contiguousSplit -> transport, SURVEY.md
+§3.4), with the UCX client/server replaced by one RCCL all-to-all per
+exchange wave — xGMI is fully connected intra-node so pairwise traffic
+scales with per-link bandwidth.
+"""
+from __future__ import annotations
+
+from typing import List, Sequence
+
+import numpy as np
+import torch
+
+from .. import ops
+from ..column import Column, ColumnBatch, Field, Schema
+from . import dist
+from .serializer import deserialize_batch, serialize_batch
+
+
+def _slice_batch(batch: ColumnBatch, start: int, end: int) -> ColumnBatch:
+    idx = Column.from_numpy(np.arange(start, end, dtype=np.int32),
+                            device=batch.device)
+    return ops.gather(batch, idx)
+
+
+def batch_schema(batch: ColumnBatch) -> Schema:
+    return Schema([Field(f"c{i}", c.dtype) for i, c in enumerate(batch.columns)])
+
+
+def exchange_by_hash(batch: ColumnBatch, key_idx: Sequence[int]) -> List[ColumnBatch]:
+    """Hash-partition rows across the world by key and exchange; returns the
+    batches received from every rank (caller concatenates)."""
+    c = dist.ctx()
+    schema = batch_schema(batch)
+    parted, offsets = ops.hash_partition(batch, list(key_idx), c.world)
+    send = [serialize_batch(_slice_batch(parted, offsets[r], offsets[r + 1]))
+            for r in range(c.world)]
+    recv = dist.all_to_all_bytes(send)
+    return [deserialize_batch(b, schema) for b in recv]
+
+
+def gather_all(batch: ColumnBatch) -> List[ColumnBatch]:
+    """All-gather the batch from every rank (keyless aggregate merge /
+    broadcast build side)."""
+    schema = batch_schema(batch)
+    bufs = dist.all_gather_bytes(serialize_batch(batch))
+    return [deserialize_batch(b, schema) for b in bufs]

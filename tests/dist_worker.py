@@ -1,0 +1,71 @@
+"""Worker for multi-process distributed tests (gloo, CPU). Launched by
+test_distributed.py via torch.distributed.run; asserts that distributed
+aggregation over the exchange layer matches a single-process run."""
+import os
+import sys
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import numpy as np
+import torch
+import torch.distributed as td
+
+from spark_rapids_amd import Session, col, count_star, sum_, avg
+
+
+def main():
+    td.init_process_group(backend="gloo")
+    rank = td.get_rank()
+    world = td.get_world_size()
+
+    # each rank owns a deterministic partition; the union is rows 0..N-1
+    n_total = 10_000
+    per = n_total // world
+    lo, hi = rank * per, (rank + 1) * per
+    rows = np.arange(lo, hi, dtype=np.int64)
+
+    s = Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({
+        "k": (rows % 97),
+        "v": rows.astype(np.float64),
+        "c": (rows % 5),
+    }, num_partitions=3)
+
+    # keyed aggregate: global result sharded by key across ranks
+    out = df.group_by("k").agg(sum_(col("v")), count_star()).collect()
+    # validate: every key's global sum; keys disjoint across ranks
+    expect = {}
+    all_rows = np.arange(0, n_total, dtype=np.int64)
+    for k in range(97):
+        m = all_rows % 97 == k
+        expect[k] = (float(all_rows[m].sum()), int(m.sum()))
+    for k, sv, cnt in out:
+        ek, ec = expect[k]
+        assert sv == ek and cnt == ec, (k, sv, cnt, expect[k])
+    # keys disjoint + complete across ranks
+    my_keys = sorted(k for k, _, _ in out)
+    gathered = [None] * world
+    td.all_gather_object(gathered, my_keys)
+    if rank == 0:
+        flat = [k for ks in gathered for k in ks]
+        assert sorted(flat) == list(range(97)), "keys lost or duplicated"
+        assert len(set(flat)) == len(flat)
+
+    # keyless aggregate: every rank sees the global result
+    tot = df.agg(sum_(col("v")), count_star(), avg(col("c"))).collect()
+    assert tot[0][1] == n_total
+    assert tot[0][0] == float(all_rows.sum())
+
+    # filter + keyed agg + keyless rollup (the q4 shape)
+    out2 = (df.filter(col("c") != 0).group_by("k")
+            .agg(sum_(col("v"))).agg(count_star()).collect())
+    assert out2[0][0] == 97, out2
+
+    td.barrier()
+    if rank == 0:
+        print("DIST_OK")
+    td.destroy_process_group()
+
+
+if __name__ == "__main__":
+    main()
