@@ -74,6 +74,13 @@ void hipdf_join_fill(int, const void*, const void*, const void*, int,
                      const void*, const void*, int64_t, const void*, void*,
                      void*, int64_t, hipStream_t);
 int64_t part_num_blocks(int64_t);
+int64_t sort_num_blocks(int64_t);
+int hipdf_sort_key_width(int);
+void hipdf_make_sort_keys(int, const void*, const void*, const void*, int,
+                          int, int, void*, int64_t, hipStream_t);
+void hipdf_radix_count(const void*, int, void*, int64_t, hipStream_t);
+void hipdf_radix_scatter(const void*, const void*, int, const void*, void*,
+                         void*, int64_t, hipStream_t);
 void hipdf_part_hist(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_part_scatter(const void*, int, const void*, void*, int64_t,
                         hipStream_t);
@@ -294,6 +301,27 @@ PYBIND11_MODULE(hipdf, m) {
   });
 
   m.def("part_num_blocks", &part_num_blocks);
+  m.def("sort_num_blocks", &sort_num_blocks);
+  m.def("sort_key_width", &hipdf_sort_key_width);
+  m.def("make_sort_keys", [](int t, int64_t data, int64_t valid, int64_t perm,
+                             bool desc, bool nulls_last, bool has_valid,
+                             int64_t keys, int64_t n, int64_t stream) {
+    hipdf_make_sort_keys(t, P(data), P(valid), P(perm), desc, nulls_last,
+                         has_valid, PM(keys), n, S(stream));
+    check_async();
+  });
+  m.def("radix_count", [](int64_t keys, int shift, int64_t counts, int64_t n,
+                          int64_t stream) {
+    hipdf_radix_count(P(keys), shift, PM(counts), n, S(stream));
+    check_async();
+  });
+  m.def("radix_scatter", [](int64_t keys_in, int64_t perm_in, int shift,
+                            int64_t offsets, int64_t keys_out,
+                            int64_t perm_out, int64_t n, int64_t stream) {
+    hipdf_radix_scatter(P(keys_in), P(perm_in), shift, P(offsets),
+                        PM(keys_out), PM(perm_out), n, S(stream));
+    check_async();
+  });
   m.def("part_hist", [](int64_t part, int nparts, int64_t counts, int64_t n,
                         int64_t stream) {
     hipdf_part_hist(P(part), nparts, PM(counts), n, S(stream));

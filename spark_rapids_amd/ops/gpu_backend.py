@@ -623,13 +623,45 @@ def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
 
 
 # ---------------------------------------------------------------------------
-# sort (CPU fallback until the radix sort lands; overrides tags SortExec off
-# GPU so this path is only reachable programmatically)
+# sort: stable LSD radix over order-preserving u64 keys (native/hipdf sort.hip)
 # ---------------------------------------------------------------------------
 
 def sort_order(batch: ColumnBatch, key_idx: List[int], descending: List[bool],
                nulls_last: List[bool]) -> Column:
-    from . import cpu_backend
-
-    host = batch.cpu()
-    return cpu_backend.sort_order(host, key_idx, descending, nulls_last)
+    n = batch.num_rows
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.int32())
+    perm: Optional[torch.Tensor] = None
+    keys_a = torch.empty(n, dtype=torch.int64, device="cuda")
+    keys_b = torch.empty(n, dtype=torch.int64, device="cuda")
+    perm_a = torch.empty(n, dtype=torch.int32, device="cuda")
+    perm_b = torch.empty(n, dtype=torch.int32, device="cuda")
+    nb = ext.sort_num_blocks(n)
+    counts = torch.empty(256 * nb, dtype=torch.int64, device="cuda")
+    # least-significant key first: stability carries earlier orders forward
+    for ci, desc, nl in reversed(list(zip(key_idx, descending, nulls_last))):
+        col = batch.columns[ci]
+        t = _ht(col.dtype)
+        has_valid = col.validity is not None
+        ext.make_sort_keys(t, col.data.data_ptr(), _ptr(col.validity),
+                           0 if perm is None else perm.data_ptr(), desc, nl,
+                           has_valid, keys_a.data_ptr(), n, s)
+        width = ext.sort_key_width(t) + (1 if has_valid else 0)
+        cur_keys, alt_keys = keys_a, keys_b
+        cur_perm, alt_perm = perm, perm_a if perm is not perm_a else perm_b
+        for p in range(width):
+            ext.radix_count(cur_keys.data_ptr(), 8 * p, counts.data_ptr(), n, s)
+            offsets, _ = _exclusive_scan_i64(counts)
+            ext.radix_scatter(cur_keys.data_ptr(),
+                              0 if cur_perm is None else cur_perm.data_ptr(),
+                              8 * p, offsets.data_ptr(), alt_keys.data_ptr(),
+                              alt_perm.data_ptr(), n, s)
+            cur_keys, alt_keys = alt_keys, cur_keys
+            nxt = perm_b if alt_perm is perm_a else perm_a
+            cur_perm, alt_perm = alt_perm, nxt
+        perm = cur_perm
+        keys_a, keys_b = cur_keys, alt_keys
+    if perm is None:
+        perm = torch.arange(n, dtype=torch.int32, device="cuda")
+    return Column(DType.int32(), n, perm.clone(), None, null_count=0)

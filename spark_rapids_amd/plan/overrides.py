@@ -138,7 +138,6 @@ class Tagger:
             if node.how not in ("inner", "left", "semi", "anti"):
                 reasons.append(f"join type {node.how} not on GPU")
         elif isinstance(node, L.Sort):
-            reasons.append("sort has no GPU kernel yet (radix sort pending)")
             for k in node.keys:
                 r = _FIXED_KEYS.supports(node.schema().field(k).dtype)
                 if r:
