@@ -27,6 +27,7 @@ KERNELS = [
     "kernels/groupby.hip",
     "kernels/join.hip",
     "kernels/partition.hip",
+    "kernels/sort.hip",
 ]
 
 CXXFLAGS = ["-O3", "-std=c++20", "-fPIC", f"--offload-arch={ARCH}",
@@ -82,6 +83,10 @@ def build(verbose: bool = True) -> str:
             os.path.getmtime(o) > os.path.getmtime(out) for o in link_inputs):
         subprocess.run([HIPCC, "-shared", "-fPIC", "-o", out] + link_inputs +
                        [f"--offload-arch={ARCH}"], check=True)
+    # import check in a fresh interpreter (catches missing symbols at link)
+    subprocess.run([sys.executable, "-c",
+                    f"import sys; sys.path.insert(0, {REPO!r}); import hipdf"],
+                   check=True)
     if verbose:
         print(f"[hipdf] built {out}")
     return out
