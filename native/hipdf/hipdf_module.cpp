@@ -304,10 +304,10 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("sort_num_blocks", &sort_num_blocks);
   m.def("sort_key_width", &hipdf_sort_key_width);
   m.def("make_sort_keys", [](int t, int64_t data, int64_t valid, int64_t perm,
-                             bool desc, bool nulls_last, bool has_valid,
+                             bool desc, bool nulls_last, bool null_only,
                              int64_t keys, int64_t n, int64_t stream) {
     hipdf_make_sort_keys(t, P(data), P(valid), P(perm), desc, nulls_last,
-                         has_valid, PM(keys), n, S(stream));
+                         null_only, PM(keys), n, S(stream));
     check_async();
   });
   m.def("radix_count", [](int64_t keys, int shift, int64_t counts, int64_t n,

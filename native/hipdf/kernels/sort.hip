@@ -66,20 +66,27 @@ __global__ void k_make_sort_keys(const T* __restrict__ data,
                                  const uint64_t* __restrict__ valid,
                                  const int32_t* __restrict__ perm, int desc,
                                  int width_bytes, int null_byte_null,
-                                 uint64_t* __restrict__ keys, int64_t n) {
+                                 int null_only, uint64_t* __restrict__ keys,
+                                 int64_t n) {
+  // width 8 leaves no room for the null byte in 64 bits: value-only keys
+  // here, and the caller runs one extra null_only pass afterwards.
   uint64_t vmask = width_bytes >= 8 ? ~0ull : ((1ull << (8 * width_bytes)) - 1);
+  bool embed_null = width_bytes < 8;
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     int32_t r = perm ? perm[i] : (int32_t)i;
     bool ok = valid_bit(valid, r);
+    uint64_t nb = ok ? (uint64_t)(1 - null_byte_null) : (uint64_t)null_byte_null;
+    if (null_only) {
+      keys[i] = nb;
+      continue;
+    }
     uint64_t k = 0;
     if (ok) {
       k = sort_key_of<T>(data[r]) & vmask;
       if (desc) k = (~k) & vmask;
     }
-    // null byte just above the value bytes: 0/1 chosen by null ordering
-    uint64_t nb = ok ? (uint64_t)(1 - null_byte_null) : (uint64_t)null_byte_null;
-    keys[i] = k | (nb << (8 * width_bytes));
+    keys[i] = embed_null ? (k | (nb << (8 * width_bytes))) : k;
   }
 }
 
@@ -161,18 +168,17 @@ int hipdf_sort_key_width(int t) { return sort_key_width(t); }
 
 void hipdf_make_sort_keys(int t, const void* data, const void* valid,
                           const void* perm, int desc, int nulls_last,
-                          int has_valid, void* keys, int64_t n,
+                          int null_only, void* keys, int64_t n,
                           hipStream_t stream) {
   int w = sort_key_width(t);
   // null byte: NULLS LAST -> null rows get 1 (sort after valid rows);
   // NULLS FIRST -> null rows get 0 and valid rows 1
   int null_byte_null = nulls_last ? 1 : 0;
-  (void)has_valid;
   dispatch_type(t, [&]<typename T>() {
     hipLaunchKernelGGL((k_make_sort_keys<T>), flat_grid(n), dim3(HIPDF_BLOCK),
                        0, stream, (const T*)data, (const uint64_t*)valid,
                        (const int32_t*)perm, desc, w, null_byte_null,
-                       (uint64_t*)keys, n);
+                       null_only, (uint64_t*)keys, n);
   });
 }
 

@@ -686,18 +686,23 @@ def sort_order(batch: ColumnBatch, key_idx: List[int],
         else:
             a = a.copy()
         if a.dtype.kind == "f":
-            key = a.astype(np.float64)
+            key = a.astype(np.float64).copy()
+            key[np.isnan(key)] = np.inf  # NaN greatest (then refined below)
+            nan_rank = np.isnan(a.astype(np.float64)).astype(np.int64)
             if desc:
                 key = -key
-            null_key = np.inf if nl else -np.inf
-            key = np.where(av, key, null_key)
-            keys.append(key)
+                nan_rank = -nan_rank
+            key = np.where(av, key, 0.0)
+            nan_rank = np.where(av, nan_rank, 0)
         else:
             key = a.astype(np.int64)
+            nan_rank = np.zeros(len(a), dtype=np.int64)
             if desc:
                 key = -key
-            null_rank = np.where(av, 0, 1 if nl else -1)
-            keys.append(key)
-            keys.append(null_rank)
+            key = np.where(av, key, 0)
+        null_rank = np.where(av, 0, 1 if nl else -1)
+        keys.append(key)
+        keys.append(nan_rank)
+        keys.append(null_rank)
     order = np.lexsort(tuple(keys)) if keys else np.arange(n)
     return _make(order.astype(np.int32), None, DType.int32())

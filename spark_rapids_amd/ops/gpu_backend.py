@@ -644,22 +644,36 @@ def sort_order(batch: ColumnBatch, key_idx: List[int], descending: List[bool],
         col = batch.columns[ci]
         t = _ht(col.dtype)
         has_valid = col.validity is not None
-        ext.make_sort_keys(t, col.data.data_ptr(), _ptr(col.validity),
-                           0 if perm is None else perm.data_ptr(), desc, nl,
-                           has_valid, keys_a.data_ptr(), n, s)
-        width = ext.sort_key_width(t) + (1 if has_valid else 0)
+        vwidth = ext.sort_key_width(t)
+        embedded_null = vwidth < 8  # null byte fits above the value bytes
+
         cur_keys, alt_keys = keys_a, keys_b
         cur_perm, alt_perm = perm, perm_a if perm is not perm_a else perm_b
-        for p in range(width):
-            ext.radix_count(cur_keys.data_ptr(), 8 * p, counts.data_ptr(), n, s)
-            offsets, _ = _exclusive_scan_i64(counts)
-            ext.radix_scatter(cur_keys.data_ptr(),
-                              0 if cur_perm is None else cur_perm.data_ptr(),
-                              8 * p, offsets.data_ptr(), alt_keys.data_ptr(),
-                              alt_perm.data_ptr(), n, s)
-            cur_keys, alt_keys = alt_keys, cur_keys
-            nxt = perm_b if alt_perm is perm_a else perm_a
-            cur_perm, alt_perm = alt_perm, nxt
+
+        def _passes(npasses, shift0=0):
+            nonlocal cur_keys, alt_keys, cur_perm, alt_perm
+            for p in range(npasses):
+                sh = 8 * (shift0 + p)
+                ext.radix_count(cur_keys.data_ptr(), sh, counts.data_ptr(), n, s)
+                offsets, _ = _exclusive_scan_i64(counts)
+                ext.radix_scatter(cur_keys.data_ptr(),
+                                  0 if cur_perm is None else cur_perm.data_ptr(),
+                                  sh, offsets.data_ptr(), alt_keys.data_ptr(),
+                                  alt_perm.data_ptr(), n, s)
+                cur_keys, alt_keys = alt_keys, cur_keys
+                nxt = perm_b if alt_perm is perm_a else perm_a
+                cur_perm, alt_perm = alt_perm, nxt
+
+        ext.make_sort_keys(t, col.data.data_ptr(), _ptr(col.validity),
+                           0 if cur_perm is None else cur_perm.data_ptr(),
+                           desc, nl, False, cur_keys.data_ptr(), n, s)
+        _passes(vwidth + (1 if (has_valid and embedded_null) else 0))
+        if has_valid and not embedded_null:
+            # 8-byte keys: run one extra null-ordering pass on null-only keys
+            ext.make_sort_keys(t, col.data.data_ptr(), _ptr(col.validity),
+                               0 if cur_perm is None else cur_perm.data_ptr(),
+                               desc, nl, True, cur_keys.data_ptr(), n, s)
+            _passes(1)
         perm = cur_perm
         keys_a, keys_b = cur_keys, alt_keys
     if perm is None:
