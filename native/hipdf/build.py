@@ -28,6 +28,7 @@ KERNELS = [
     "kernels/join.hip",
     "kernels/partition.hip",
     "kernels/sort.hip",
+    "kernels/decode.hip",
 ]
 
 CXXFLAGS = ["-O3", "-std=c++20", "-fPIC", f"--offload-arch={ARCH}",

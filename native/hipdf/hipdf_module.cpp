@@ -75,6 +75,11 @@ void hipdf_join_fill(int, const void*, const void*, const void*, int,
                      void*, int64_t, hipStream_t);
 int64_t part_num_blocks(int64_t);
 int64_t sort_num_blocks(int64_t);
+void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
+                             hipStream_t);
+void hipdf_scatter_fixed(int, const void*, const void*, void*, int64_t,
+                         hipStream_t);
+void hipdf_levels_to_mask(const void*, int, void*, int64_t, hipStream_t);
 int hipdf_sort_key_width(int);
 void hipdf_make_sort_keys(int, const void*, const void*, const void*, int,
                           int, int, void*, int64_t, hipStream_t);
@@ -297,6 +302,22 @@ PYBIND11_MODULE(hipdf, m) {
                         int64_t n, int64_t stream) {
     hipdf_join_fill(how, P(lh), P(lk), P(rk), nkeys, P(head), P(next), cap,
                     P(offsets), PM(lmap), PM(rmap), n, S(stream));
+    check_async();
+  });
+
+  m.def("rle_hybrid_decode", [](int64_t data, int64_t nbytes, int bw,
+                                int64_t out, int64_t n, int64_t stream) {
+    hipdf_rle_hybrid_decode(P(data), nbytes, bw, PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("scatter_fixed", [](int esize, int64_t vals, int64_t idx, int64_t out,
+                            int64_t n, int64_t stream) {
+    hipdf_scatter_fixed(esize, P(vals), P(idx), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("levels_to_mask", [](int64_t levels, int max_level, int64_t mask,
+                             int64_t n, int64_t stream) {
+    hipdf_levels_to_mask(P(levels), max_level, PM(mask), n, S(stream));
     check_async();
   });
 

@@ -191,11 +191,26 @@ class Session:
     def table(self, name: str) -> DataFrame:
         return self.catalog[name]
 
-    def read_parquet(self, path: str, num_partitions: int = 1) -> DataFrame:
+    def read_parquet(self, path: str, columns=None) -> DataFrame:
+        import torch
+
+        from .config import PARQUET_MT_THREADS, PARQUET_READER_TYPE
         from .io.parquet import ParquetTable
 
-        src = ParquetTable(path, num_partitions)
+        reader = str(self.conf.get(PARQUET_READER_TYPE)).upper()
+        if reader == "AUTO":
+            reader = "GPU_DECODE" if (torch.cuda.is_available()
+                                      and self.conf.sql_enabled) else "CPU"
+        src = ParquetTable(path, columns=columns, reader=reader,
+                           prefetch_threads=self.conf.get(PARQUET_MT_THREADS))
         return DataFrame(self, L.Scan(src, src.schema, f"parquet:{path}"))
+
+    def write_parquet(self, df: DataFrame, path: str,
+                      compression: str = "snappy"):
+        from .io.parquet import write_parquet
+
+        batch = df.collect_batch()
+        write_parquet(batch, df.schema, path, compression)
 
 
 def _infer_list_dtype(v: list) -> DType:

@@ -132,6 +132,18 @@ HOST_SPILL_STORAGE_SIZE = bytes_conf(
 SPILL_PATH = str_conf(
     "spark.rapids.memory.spillPath", "/tmp/rapids_spill",
     "Local directory for disk spill files.")
+PARQUET_READER_TYPE = str_conf(
+    "spark.rapids.sql.format.parquet.reader.type", "AUTO",
+    "Parquet reader: AUTO (GPU decode when a GPU is present, CPU otherwise), "
+    "GPU_DECODE (hipdf kernels decode PLAIN/dictionary pages on device, "
+    "per-file CPU fallback for unsupported features), CPU (pyarrow host "
+    "decode feeding device batches — the hybrid-scan analogue).")
+PARQUET_ENABLED = bool_conf(
+    "spark.rapids.sql.format.parquet.enabled", True,
+    "Enable parquet scans on GPU.")
+PARQUET_MT_THREADS = int_conf(
+    "spark.rapids.sql.format.parquet.multiThreadedRead.numThreads", 4,
+    "Threads in the multithreaded parquet prefetch pool.")
 SHUFFLE_MODE = str_conf(
     "spark.rapids.shuffle.mode", "MULTITHREADED",
     "Shuffle transport: MULTITHREADED (host staging) or RCCL (device-to-device "
