@@ -52,8 +52,11 @@ def _make_tables(session, rows: int, seed: int, device: str,
     return {
         "store_sales": session.from_batches(fact_batches,
                                             datagen.fact_schema(), "store_sales"),
-        "item": session.from_batches([items], datagen.item_schema(), "item"),
-        "store": session.from_batches([stores], datagen.store_schema(), "store"),
+        # dimension tables are generated identically on every rank
+        "item": session.from_batches([items], datagen.item_schema(), "item",
+                                     replicated=True),
+        "store": session.from_batches([stores], datagen.store_schema(),
+                                      "store", replicated=True),
     }
 
 

@@ -25,9 +25,11 @@ from .types import DType
 class MemTable:
     """In-memory partitioned table source."""
 
-    def __init__(self, batches: List[ColumnBatch], schema: Schema):
+    def __init__(self, batches: List[ColumnBatch], schema: Schema,
+                 replicated: bool = False):
         self.batches = batches
         self.schema = schema
+        self.replicated = replicated
 
     def partitions(self) -> Iterable[ColumnBatch]:
         return iter(self.batches)
@@ -163,7 +165,8 @@ class Session:
     # ---- data ingestion -------------------------------------------------
     def create_dataframe(self, data: Dict[str, list],
                          dtypes: Optional[Dict[str, DType]] = None,
-                         num_partitions: int = 1) -> DataFrame:
+                         num_partitions: int = 1,
+                         replicated: bool = False) -> DataFrame:
         names = list(data)
         cols = []
         for n in names:
@@ -178,12 +181,13 @@ class Session:
         batch = ColumnBatch(cols)
         schema = Schema([Field(n, c.dtype) for n, c in zip(names, cols)])
         batches = _split_partitions(batch, num_partitions)
-        table = MemTable(batches, schema)
+        table = MemTable(batches, schema, replicated)
         return DataFrame(self, L.Scan(table, schema, "memory"))
 
     def from_batches(self, batches: List[ColumnBatch], schema: Schema,
-                     label: str = "memory") -> DataFrame:
-        return DataFrame(self, L.Scan(MemTable(batches, schema), schema, label))
+                     label: str = "memory", replicated: bool = False) -> DataFrame:
+        return DataFrame(self, L.Scan(MemTable(batches, schema, replicated),
+                                      schema, label))
 
     def register(self, name: str, df: DataFrame):
         self.catalog[name] = df

@@ -33,6 +33,10 @@ class Scan(LogicalPlan):
         self._schema = schema
         self.label = label
 
+    @property
+    def replicated(self) -> bool:
+        return bool(getattr(self.source, "replicated", False))
+
     def schema(self) -> Schema:
         return self._schema
 
@@ -157,3 +161,26 @@ class Union(LogicalPlan):
 
     def schema(self) -> Schema:
         return self.plans[0].schema()
+
+
+def is_replicated(plan: LogicalPlan) -> bool:
+    """True if every rank holds identical full data for this subtree (so
+    distributed exchanges/broadcasts must be skipped)."""
+    if isinstance(plan, Scan):
+        return plan.replicated
+    if isinstance(plan, Join):
+        # the build (right) side is either replicated or broadcast
+        # (all-gathered) by the exec, so the output's distribution follows
+        # the stream (left) side
+        return is_replicated(plan.left)
+    if isinstance(plan, Union):
+        return all(is_replicated(p) for p in plan.plans)
+    if isinstance(plan, Aggregate):
+        # replicated input -> local agg is global; keyless sharded input ->
+        # all-gather merge makes the output replicated too
+        if is_replicated(plan.child):
+            return True
+        return len(plan.group_exprs) == 0
+    if plan.children:
+        return all(is_replicated(c) for c in plan.children)
+    return False

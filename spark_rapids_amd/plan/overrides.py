@@ -206,10 +206,12 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
         return P.ProjectExec(device, node.exprs, kids[0], node.schema())
     if isinstance(node, L.Aggregate):
         return P.HashAggregateExec(device, node.group_exprs, node.aggs,
-                                   kids[0], node.schema())
+                                   kids[0], node.schema(),
+                                   input_replicated=L.is_replicated(node.child))
     if isinstance(node, L.Join):
         return P.HashJoinExec(device, kids[0], kids[1], node.left_on,
-                              node.right_on, node.how, node.schema())
+                              node.right_on, node.how, node.schema(),
+                              right_replicated=L.is_replicated(node.right))
     if isinstance(node, L.Sort):
         return P.SortExec(device, node.keys, node.descending,
                           node.nulls_last, kids[0])

@@ -162,10 +162,12 @@ class HashAggregateExec(PhysicalExec):
     (GpuAggregateExec.scala:1942,896)."""
 
     def __init__(self, device: str, group_exprs: List[Expression],
-                 aggs: List[AggExpr], child: PhysicalExec, schema: Schema):
+                 aggs: List[AggExpr], child: PhysicalExec, schema: Schema,
+                 input_replicated: bool = False):
         super().__init__(device, schema, [child])
         self.group_exprs = group_exprs
         self.aggs = aggs
+        self.input_replicated = input_replicated
 
     def execute(self) -> Iterator[ColumnBatch]:
         in_schema = self.children[0].schema
@@ -192,7 +194,7 @@ class HashAggregateExec(PhysicalExec):
         # rank owns a disjoint key range; keyless -> all-gather partials and
         # merge identically on every rank
         from ..shuffle import dist as _dist
-        if _dist.ctx().is_multi:
+        if _dist.ctx().is_multi and not self.input_replicated:
             from ..shuffle.exchange import exchange_by_hash, gather_all
             if nkeys:
                 received = exchange_by_hash(merged_in, list(range(nkeys)))
@@ -231,15 +233,33 @@ class HashJoinExec(PhysicalExec):
     probes (reference: GpuShuffledHashJoinExec / GpuHashJoin.scala)."""
 
     def __init__(self, device: str, left: PhysicalExec, right: PhysicalExec,
-                 left_on: List[str], right_on: List[str], how: str, schema: Schema):
+                 left_on: List[str], right_on: List[str], how: str,
+                 schema: Schema, right_replicated: bool = True):
         super().__init__(device, schema, [left, right])
         self.left_on = left_on
         self.right_on = right_on
         self.how = how
+        self.right_replicated = right_replicated
 
     def execute(self) -> Iterator[ColumnBatch]:
         left, right = self.children
         rbatches = list(right.execute())
+        # distributed: a sharded build side must be broadcast (all-gathered)
+        # so every rank probes against the full table (broadcast-join; the
+        # shuffled-join strategy for large build sides is a later round)
+        from ..shuffle import dist as _dist
+        if _dist.ctx().is_multi and not self.right_replicated:
+            from ..shuffle.exchange import gather_all
+            if rbatches:
+                local = ops.concat_batches(rbatches) if len(rbatches) > 1 \
+                    else rbatches[0]
+            else:
+                local = ColumnBatch(
+                    [Column.from_pylist([], f.dtype) for f in
+                     right.schema.fields], 0)
+                if self.gpu:
+                    local = local.cuda()
+            rbatches = [b for b in gather_all(local) if b.num_rows]
         if not rbatches:
             if self.how in ("inner", "semi"):
                 return

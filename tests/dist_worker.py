@@ -61,6 +61,27 @@ def main():
             .agg(sum_(col("v"))).agg(count_star()).collect())
     assert out2[0][0] == 97, out2
 
+    # sharded-right join: build side must be all-gathered (broadcast join)
+    right_rows = np.arange(rank * 10, rank * 10 + 10, dtype=np.int64)
+    right = s.create_dataframe({"k": right_rows % 97,
+                                "r": right_rows * 100})
+    joined = df.join(right, on="k").agg(count_star()).collect()
+    # expected: every left row whose k is in union of right keys (0..19 %97)
+    rk = set((np.arange(0, world * 10) % 97).tolist())
+    exp_cnt = int(sum(1 for v in all_rows if (v % 97) in rk)) * \
+        sum(1 for v in (np.arange(0, world * 10) % 97) if True) // (world * 10)
+    # compute exactly: matches = sum over left rows of count of right dups
+    from collections import Counter
+    rc = Counter((np.arange(0, world * 10) % 97).tolist())
+    exp = int(sum(rc[v % 97] for v in all_rows.tolist()))
+    assert joined[0][0] == exp, (joined, exp)
+
+    # replicated table must NOT be re-broadcast/duplicated
+    rep = s.create_dataframe({"k": list(range(97)), "w": [1] * 97},
+                             replicated=True)
+    j2 = df.join(rep, on="k").agg(count_star()).collect()
+    assert j2[0][0] == n_total, j2
+
     td.barrier()
     if rank == 0:
         print("DIST_OK")
