@@ -50,7 +50,12 @@ class DataFrame:
         es = [(_col(e) if isinstance(e, str) else e) for e in exprs]
         return DataFrame(self.session, L.Project(es, self.plan))
 
-    def with_column(self, name: str, expr: Expression) -> "DataFrame":
+    def with_column(self, name: str, expr) -> "DataFrame":
+        from .expr.windows import WindowExpr
+
+        if isinstance(expr, WindowExpr):
+            return DataFrame(self.session,
+                             L.Window([expr.alias(name)], self.plan))
         sch = self.plan.schema()
         es: List[Expression] = []
         replaced = False

@@ -1,0 +1,123 @@
+"""Window expressions.
+
+Reference analogue: the window exec family (GpuWindowExec / running /
+bounded / double-pass variants, sql-plugin .../rapids/window/, SURVEY.md
+§2.4). Supported this round: ranking (row_number, rank, dense_rank),
+running and whole-partition aggregates (sum/count/min/max/avg), lag/lead.
+Frames: ROWS UNBOUNDED PRECEDING..CURRENT ROW ("running", the default for
+ordered aggregates like Spark) and UNBOUNDED..UNBOUNDED (whole partition,
+the default when no order is given).
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Sequence
+
+from ..types import DType, FLOAT64, INT32, INT64
+from .expressions import Expression, _as_expr
+
+RANKING = {"row_number", "rank", "dense_rank"}
+AGGS = {"sum", "count", "min", "max", "mean"}
+OFFSETS = {"lag", "lead"}
+
+
+class WindowSpec:
+    def __init__(self, partition_by: Sequence[str] = (),
+                 order_by: Sequence[str] = (),
+                 descending: Optional[Sequence[bool]] = None):
+        self.partition_by = list(partition_by)
+        self.order_by = list(order_by)
+        self.descending = list(descending) if descending is not None \
+            else [False] * len(self.order_by)
+
+
+class WindowFunc:
+    """An unbound window function; bind with .over(...)."""
+
+    def __init__(self, op: str, child: Optional[Expression] = None,
+                 offset: int = 1, default=None):
+        self.op = op
+        self.child = _as_expr(child) if child is not None else None
+        self.offset = offset
+        self.default = default
+
+    def over(self, partition_by: Sequence[str] = (),
+             order_by: Sequence[str] = (),
+             descending: Optional[Sequence[bool]] = None) -> "WindowExpr":
+        return WindowExpr(self, WindowSpec(partition_by, order_by, descending))
+
+
+class WindowExpr:
+    def __init__(self, func: WindowFunc, spec: WindowSpec,
+                 name: Optional[str] = None):
+        self.func = func
+        self.spec = spec
+        self._name = name
+        if func.op in RANKING and not spec.order_by:
+            raise ValueError(f"{func.op} requires order_by")
+        if func.op in OFFSETS and not spec.order_by:
+            raise ValueError(f"{func.op} requires order_by")
+
+    def alias(self, name: str) -> "WindowExpr":
+        return WindowExpr(self.func, self.spec, name)
+
+    def output_name(self) -> str:
+        if self._name:
+            return self._name
+        c = f"({self.func.child})" if self.func.child is not None else "()"
+        return f"{self.func.op}{c}"
+
+    def out_dtype(self, schema) -> DType:
+        op = self.func.op
+        if op in RANKING:
+            return INT32
+        if op == "count":
+            return INT64
+        ct = self.func.child.dtype(schema)
+        if op == "sum":
+            return FLOAT64 if ct.is_floating else INT64
+        if op == "mean":
+            return FLOAT64
+        return ct  # min/max/lag/lead keep input type
+
+    def nullable(self, schema) -> bool:
+        return self.func.op not in RANKING
+
+
+def row_number() -> WindowFunc:
+    return WindowFunc("row_number")
+
+
+def rank() -> WindowFunc:
+    return WindowFunc("rank")
+
+
+def dense_rank() -> WindowFunc:
+    return WindowFunc("dense_rank")
+
+
+def win_sum(e) -> WindowFunc:
+    return WindowFunc("sum", e)
+
+
+def win_count(e) -> WindowFunc:
+    return WindowFunc("count", e)
+
+
+def win_min(e) -> WindowFunc:
+    return WindowFunc("min", e)
+
+
+def win_max(e) -> WindowFunc:
+    return WindowFunc("max", e)
+
+
+def win_avg(e) -> WindowFunc:
+    return WindowFunc("mean", e)
+
+
+def lag(e, offset: int = 1, default=None) -> WindowFunc:
+    return WindowFunc("lag", e, offset, default)
+
+
+def lead(e, offset: int = 1, default=None) -> WindowFunc:
+    return WindowFunc("lead", e, offset, default)

@@ -120,6 +120,24 @@ class Join(LogicalPlan):
         return f"Join({self.how})"
 
 
+class Window(LogicalPlan):
+    def __init__(self, window_exprs, child: LogicalPlan):
+        self.window_exprs = list(window_exprs)
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        cs = self.child.schema()
+        fields = list(cs.fields)
+        for w in self.window_exprs:
+            fields.append(Field(w.output_name(), w.out_dtype(cs),
+                                w.nullable(cs)))
+        return Schema(fields)
+
+
 class Sort(LogicalPlan):
     def __init__(self, child: LogicalPlan, keys: List[str],
                  descending: Optional[List[bool]] = None,

@@ -137,6 +137,9 @@ class Tagger:
                     reasons.append(f"join key {k}: {r}")
             if node.how not in ("inner", "left", "semi", "anti"):
                 reasons.append(f"join type {node.how} not on GPU")
+        elif isinstance(node, L.Window):
+            reasons.append("window has no GPU kernels yet "
+                           "(segmented scans pending)")
         elif isinstance(node, L.Sort):
             for k in node.keys:
                 r = _FIXED_KEYS.supports(node.schema().field(k).dtype)
@@ -212,6 +215,10 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
         return P.HashJoinExec(device, kids[0], kids[1], node.left_on,
                               node.right_on, node.how, node.schema(),
                               right_replicated=L.is_replicated(node.right))
+    if isinstance(node, L.Window):
+        from .window_exec import WindowExec
+
+        return WindowExec(device, node.window_exprs, kids[0], node.schema())
     if isinstance(node, L.Sort):
         return P.SortExec(device, node.keys, node.descending,
                           node.nulls_last, kids[0])

@@ -1,0 +1,104 @@
+"""Window function tests vs hand-computed expectations (Spark semantics)."""
+import numpy as np
+import pytest
+
+from spark_rapids_amd import (Session, col, dense_rank, lag, lead, rank,
+                              row_number, win_avg, win_count, win_max,
+                              win_min, win_sum)
+
+
+@pytest.fixture
+def df(session):
+    return session.create_dataframe({
+        "p": [1, 1, 1, 2, 2, 1],
+        "o": [10, 30, 20, 5, 5, 30],
+        "v": [1.0, 2.0, None, 4.0, 5.0, 6.0],
+    })
+
+
+def test_row_number(df):
+    out = df.with_column("rn", row_number().over(["p"], ["o"])).collect()
+    # sorted by (p, o): p=1 o=10,20,30,30 ; p=2 o=5,5
+    rns = [(r[0], r[1], r[3]) for r in out]
+    assert rns == [(1, 10, 1), (1, 20, 2), (1, 30, 3), (1, 30, 4),
+                   (2, 5, 1), (2, 5, 2)]
+
+
+def test_rank_and_dense_rank(df):
+    out = df.with_column("rk", rank().over(["p"], ["o"])).collect()
+    assert [r[3] for r in out] == [1, 2, 3, 3, 1, 1]
+    out = df.with_column("dr", dense_rank().over(["p"], ["o"])).collect()
+    assert [r[3] for r in out] == [1, 2, 3, 3, 1, 1]
+
+
+def test_running_sum_ignores_nulls(df):
+    out = df.with_column("s", win_sum(col("v")).over(["p"], ["o"])).collect()
+    # p=1 sorted by o: v = 1.0, None(o=20), then o=30 twice (2.0, 6.0 in
+    # original row order for ties)
+    vals = [r[3] for r in out]
+    assert vals[0] == 1.0
+    assert vals[1] == 1.0  # null ignored, frame sum so far
+    assert vals[2] + vals[3] >= vals[2]  # monotone over ties
+    assert vals[4] == 4.0 and vals[5] == 9.0
+
+
+def test_partition_agg_no_order(df):
+    out = df.with_column("t", win_sum(col("v")).over(["p"])).collect()
+    for p, o, v, t in out:
+        assert t == (9.0 if p == 1 else 9.0)  # p1: 1+2+6, p2: 4+5
+
+
+def test_win_count_avg_min_max(df):
+    out = df.with_column("c", win_count(col("v")).over(["p"])).collect()
+    assert [r[3] for r in out] == [3, 3, 3, 3, 2, 2]
+    out = df.with_column("m", win_min(col("o").cast(
+        __import__("spark_rapids_amd").INT64)).over(["p"])).collect()
+    assert [r[3] for r in out] == [10, 10, 10, 10, 5, 5]
+    out = df.with_column("a", win_avg(col("v")).over(["p"])).collect()
+    assert [r[3] for r in out] == pytest.approx([3.0, 3.0, 3.0, 3.0, 4.5, 4.5])
+
+
+def test_lag_lead(df):
+    out = df.with_column("lg", lag(col("o")).over(["p"], ["o"])).collect()
+    assert [r[3] for r in out] == [None, 10, 20, 30, None, 5]
+    out = df.with_column("ld", lead(col("o")).over(["p"], ["o"])).collect()
+    assert [r[3] for r in out] == [20, 30, 30, None, 5, None]
+
+
+def test_lag_default(df):
+    out = df.with_column("lg", lag(col("o"), 1, -1).over(["p"], ["o"])).collect()
+    assert [r[3] for r in out] == [-1, 10, 20, 30, -1, 5]
+
+
+def test_window_running_min(session):
+    df = session.create_dataframe({"p": [1] * 5, "o": [1, 2, 3, 4, 5],
+                                   "v": [3.0, 1.0, None, 2.0, 0.5]})
+    out = df.with_column("m", win_min(col("v")).over(["p"], ["o"])).collect()
+    assert [r[3] for r in out] == [3.0, 1.0, 1.0, 1.0, 0.5]
+
+
+def test_window_falls_back_to_cpu(session):
+    df = session.create_dataframe({"p": [1], "o": [1], "v": [1.0]})
+    tree = df.with_column("rn", row_number().over(["p"], ["o"])).explain()
+    assert "Window" in tree
+
+
+def test_window_large_random(session):
+    rng = np.random.default_rng(5)
+    n = 20_000
+    df = session.create_dataframe({
+        "p": rng.integers(0, 50, n),
+        "o": rng.integers(0, 1000, n),
+        "v": rng.uniform(0, 10, n),
+    })
+    out = df.with_column("rn", row_number().over(["p"], ["o"])).to_pydict()
+    # validate per-partition: rn is 1..len(partition) in order
+    import collections
+
+    per = collections.defaultdict(list)
+    for p, o, v, rn in zip(out["p"], out["o"], out["v"], out["rn"]):
+        per[p].append((rn, o))
+    for p, rows in per.items():
+        assert [r[0] for r in rows] == list(range(1, len(rows) + 1))
+        os_ = [r[1] for r in rows]
+        assert os_ == sorted(os_)
