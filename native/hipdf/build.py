@@ -29,6 +29,7 @@ KERNELS = [
     "kernels/partition.hip",
     "kernels/sort.hip",
     "kernels/decode.hip",
+    "kernels/strings.hip",
 ]
 
 CXXFLAGS = ["-O3", "-std=c++20", "-fPIC", f"--offload-arch={ARCH}",

@@ -80,6 +80,20 @@ void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
 void hipdf_scatter_fixed(int, const void*, const void*, void*, int64_t,
                          hipStream_t);
 void hipdf_levels_to_mask(const void*, int, void*, int64_t, hipStream_t);
+void hipdf_str_cmp(int, const void*, const void*, const void*, const void*,
+                   void*, int64_t, hipStream_t);
+void hipdf_str_cmp_scalar(int, const void*, const void*, const void*, int,
+                          void*, int64_t, hipStream_t);
+void hipdf_str_find(int, const void*, const void*, const void*, int, void*,
+                    int64_t, hipStream_t);
+void hipdf_str_like(const void*, const void*, const void*, int, void*,
+                    int64_t, hipStream_t);
+void hipdf_str_length(const void*, const void*, void*, int64_t, hipStream_t);
+void hipdf_str_case(int, const void*, void*, int64_t, hipStream_t);
+void hipdf_substr_ranges(const void*, const void*, int, int, void*, void*,
+                         int64_t, hipStream_t);
+void hipdf_substr_copy(const void*, const void*, const void*, const void*,
+                       void*, int64_t, hipStream_t);
 int hipdf_sort_key_width(int);
 void hipdf_make_sort_keys(int, const void*, const void*, const void*, int,
                           int, int, void*, int64_t, hipStream_t);
@@ -318,6 +332,51 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("levels_to_mask", [](int64_t levels, int max_level, int64_t mask,
                              int64_t n, int64_t stream) {
     hipdf_levels_to_mask(P(levels), max_level, PM(mask), n, S(stream));
+    check_async();
+  });
+
+  m.def("str_cmp", [](int op, int64_t ao, int64_t ab, int64_t bo, int64_t bb,
+                      int64_t out, int64_t n, int64_t stream) {
+    hipdf_str_cmp(op, P(ao), P(ab), P(bo), P(bb), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("str_cmp_scalar", [](int op, int64_t ao, int64_t ab, int64_t pat,
+                             int plen, int64_t out, int64_t n, int64_t stream) {
+    hipdf_str_cmp_scalar(op, P(ao), P(ab), P(pat), plen, PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("str_find", [](int mode, int64_t ao, int64_t ab, int64_t pat, int plen,
+                       int64_t out, int64_t n, int64_t stream) {
+    hipdf_str_find(mode, P(ao), P(ab), P(pat), plen, PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("str_like", [](int64_t ao, int64_t ab, int64_t pat, int plen,
+                       int64_t out, int64_t n, int64_t stream) {
+    hipdf_str_like(P(ao), P(ab), P(pat), plen, PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("str_length", [](int64_t ao, int64_t ab, int64_t out, int64_t n,
+                         int64_t stream) {
+    hipdf_str_length(P(ao), P(ab), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("str_case", [](bool upper, int64_t in, int64_t out, int64_t nbytes,
+                       int64_t stream) {
+    hipdf_str_case(upper, P(in), PM(out), nbytes, S(stream));
+    check_async();
+  });
+  m.def("substr_ranges", [](int64_t ao, int64_t ab, int start, int slen,
+                            int64_t bstart, int64_t blen, int64_t n,
+                            int64_t stream) {
+    hipdf_substr_ranges(P(ao), P(ab), start, slen, PM(bstart), PM(blen), n,
+                        S(stream));
+    check_async();
+  });
+  m.def("substr_copy", [](int64_t ab, int64_t bstart, int64_t blen,
+                          int64_t out_off, int64_t out_bytes, int64_t n,
+                          int64_t stream) {
+    hipdf_substr_copy(P(ab), P(bstart), P(blen), P(out_off), PM(out_bytes), n,
+                      S(stream));
     check_async();
   });
 

@@ -1,0 +1,268 @@
+// String kernels over Arrow offsets+bytes columns (reference analogue: the
+// cudf strings family reached from stringFunctions.scala — SURVEY.md §2.4).
+// Byte-wise UTF-8: comparisons are memcmp-order (== codepoint order),
+// length/substring count codepoints, upper/lower transform ASCII bytes
+// (non-ASCII passes through; gated by incompatibleOps like the reference's
+// incompat ops).
+#include "hipdf_common.h"
+
+enum StrCmpOp : int { SC_EQ = 0, SC_NE, SC_LT, SC_LE, SC_GT, SC_GE };
+enum StrFindOp : int { SF_CONTAINS = 0, SF_STARTS, SF_ENDS };
+
+__device__ __forceinline__ int str_cmp(const uint8_t* a, int32_t la,
+                                       const uint8_t* b, int32_t lb) {
+  int32_t n = la < lb ? la : lb;
+  for (int32_t i = 0; i < n; ++i) {
+    if (a[i] != b[i]) return a[i] < b[i] ? -1 : 1;
+  }
+  return la == lb ? 0 : (la < lb ? -1 : 1);
+}
+
+__device__ __forceinline__ bool cmp_result(int op, int c) {
+  switch (op) {
+    case SC_EQ: return c == 0;
+    case SC_NE: return c != 0;
+    case SC_LT: return c < 0;
+    case SC_LE: return c <= 0;
+    case SC_GT: return c > 0;
+    default: return c >= 0;
+  }
+}
+
+// column vs column compare; null handling done by caller's validity AND
+__global__ void k_str_cmp(int op, const int32_t* __restrict__ ao,
+                          const uint8_t* __restrict__ ab,
+                          const int32_t* __restrict__ bo,
+                          const uint8_t* __restrict__ bb,
+                          uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = str_cmp(ab + ao[i], ao[i + 1] - ao[i], bb + bo[i],
+                    bo[i + 1] - bo[i]);
+    out[i] = (uint8_t)cmp_result(op, c);
+  }
+}
+
+// column vs scalar pattern
+__global__ void k_str_cmp_scalar(int op, const int32_t* __restrict__ ao,
+                                 const uint8_t* __restrict__ ab,
+                                 const uint8_t* __restrict__ pat, int32_t plen,
+                                 uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int c = str_cmp(ab + ao[i], ao[i + 1] - ao[i], pat, plen);
+    out[i] = (uint8_t)cmp_result(op, c);
+  }
+}
+
+__global__ void k_str_find(int mode, const int32_t* __restrict__ ao,
+                           const uint8_t* __restrict__ ab,
+                           const uint8_t* __restrict__ pat, int32_t plen,
+                           uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* s = ab + ao[i];
+    int32_t len = ao[i + 1] - ao[i];
+    bool r = false;
+    if (plen == 0) {
+      r = true;
+    } else if (mode == SF_STARTS) {
+      r = len >= plen && str_cmp(s, plen, pat, plen) == 0;
+    } else if (mode == SF_ENDS) {
+      r = len >= plen && str_cmp(s + len - plen, plen, pat, plen) == 0;
+    } else {
+      for (int32_t p = 0; p + plen <= len; ++p) {
+        if (str_cmp(s + p, plen, pat, plen) == 0) {
+          r = true;
+          break;
+        }
+      }
+    }
+    out[i] = (uint8_t)r;
+  }
+}
+
+// SQL LIKE: % any-run, _ one char (byte-approx: one codepoint via lead byte)
+__device__ bool like_match(const uint8_t* s, int32_t sl, const uint8_t* p,
+                           int32_t pl) {
+  int32_t si = 0, pi = 0, star_p = -1, star_s = 0;
+  while (si < sl) {
+    if (pi < pl && p[pi] == '%') {
+      star_p = ++pi;
+      star_s = si;
+    } else if (pi < pl && (p[pi] == '_' || p[pi] == s[si])) {
+      if (p[pi] == '_') {
+        // skip one UTF-8 codepoint
+        ++si;
+        while (si < sl && (s[si] & 0xC0) == 0x80) ++si;
+      } else {
+        ++si;
+      }
+      ++pi;
+    } else if (star_p >= 0) {
+      pi = star_p;
+      si = ++star_s;
+    } else {
+      return false;
+    }
+  }
+  while (pi < pl && p[pi] == '%') ++pi;
+  return pi == pl;
+}
+
+__global__ void k_str_like(const int32_t* __restrict__ ao,
+                           const uint8_t* __restrict__ ab,
+                           const uint8_t* __restrict__ pat, int32_t plen,
+                           uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (uint8_t)like_match(ab + ao[i], ao[i + 1] - ao[i], pat, plen);
+}
+
+// codepoint length
+__global__ void k_str_length(const int32_t* __restrict__ ao,
+                             const uint8_t* __restrict__ ab,
+                             int32_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t cnt = 0;
+    for (int32_t p = ao[i]; p < ao[i + 1]; ++p)
+      cnt += (ab[p] & 0xC0) != 0x80;
+    out[i] = cnt;
+  }
+}
+
+// ASCII upper/lower (offsets unchanged)
+__global__ void k_str_case(int upper, const uint8_t* __restrict__ in,
+                           uint8_t* __restrict__ out, int64_t nbytes) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < nbytes;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint8_t c = in[i];
+    if (upper && c >= 'a' && c <= 'z') c -= 32;
+    if (!upper && c >= 'A' && c <= 'Z') c += 32;
+    out[i] = c;
+  }
+}
+
+// substring (1-based start in codepoints, length in codepoints; Spark
+// semantics: start 0 behaves like 1, negative counts from the end).
+// pass 1: byte [start,len) per row
+__global__ void k_substr_ranges(const int32_t* __restrict__ ao,
+                                const uint8_t* __restrict__ ab, int32_t start,
+                                int32_t slen, int32_t* __restrict__ bstart,
+                                int64_t* __restrict__ blen, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t s0 = ao[i], s1 = ao[i + 1];
+    // count codepoints
+    int32_t cps = 0;
+    for (int32_t p = s0; p < s1; ++p) cps += (ab[p] & 0xC0) != 0x80;
+    int32_t begin = start > 0 ? start - 1 : (start < 0 ? cps + start : 0);
+    if (begin < 0) begin = 0;
+    int32_t end = slen < 0 ? cps : begin + slen;
+    if (end > cps) end = cps;
+    if (begin >= cps || end <= begin) {
+      bstart[i] = s0;
+      blen[i] = 0;
+      continue;
+    }
+    // walk to byte positions
+    int32_t cp = 0, bs = s1, be = s1;
+    for (int32_t p = s0; p < s1; ++p) {
+      if ((ab[p] & 0xC0) == 0x80) continue;
+      if (cp == begin) bs = p;
+      if (cp == end) {
+        be = p;
+        break;
+      }
+      ++cp;
+    }
+    if (cp < end) be = s1;
+    bstart[i] = bs;
+    blen[i] = be - bs;
+  }
+}
+
+// pass 2: copy using scanned output offsets
+__global__ void k_substr_copy(const uint8_t* __restrict__ ab,
+                              const int32_t* __restrict__ bstart,
+                              const int64_t* __restrict__ blen,
+                              const int64_t* __restrict__ out_off,
+                              uint8_t* __restrict__ out_bytes, int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t i = wave_global; i < n; i += wave_count) {
+    int64_t len = blen[i];
+    int64_t dst = out_off[i];
+    int32_t src = bstart[i];
+    for (int64_t b = lane; b < len; b += WAVE)
+      out_bytes[dst + b] = ab[src + b];
+  }
+}
+
+extern "C" {
+
+void hipdf_str_cmp(int op, const void* ao, const void* ab, const void* bo,
+                   const void* bb, void* out, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_cmp, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     op, (const int32_t*)ao, (const uint8_t*)ab,
+                     (const int32_t*)bo, (const uint8_t*)bb, (uint8_t*)out, n);
+}
+
+void hipdf_str_cmp_scalar(int op, const void* ao, const void* ab,
+                          const void* pat, int plen, void* out, int64_t n,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_cmp_scalar, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, op, (const int32_t*)ao, (const uint8_t*)ab,
+                     (const uint8_t*)pat, (int32_t)plen, (uint8_t*)out, n);
+}
+
+void hipdf_str_find(int mode, const void* ao, const void* ab, const void* pat,
+                    int plen, void* out, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_find, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     mode, (const int32_t*)ao, (const uint8_t*)ab,
+                     (const uint8_t*)pat, (int32_t)plen, (uint8_t*)out, n);
+}
+
+void hipdf_str_like(const void* ao, const void* ab, const void* pat, int plen,
+                    void* out, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_like, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (const int32_t*)ao, (const uint8_t*)ab,
+                     (const uint8_t*)pat, (int32_t)plen, (uint8_t*)out, n);
+}
+
+void hipdf_str_length(const void* ao, const void* ab, void* out, int64_t n,
+                      hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_length, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (const int32_t*)ao, (const uint8_t*)ab, (int32_t*)out, n);
+}
+
+void hipdf_str_case(int upper, const void* in, void* out, int64_t nbytes,
+                    hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_case, flat_grid(nbytes), dim3(HIPDF_BLOCK), 0,
+                     stream, upper, (const uint8_t*)in, (uint8_t*)out, nbytes);
+}
+
+void hipdf_substr_ranges(const void* ao, const void* ab, int start, int slen,
+                         void* bstart, void* blen, int64_t n,
+                         hipStream_t stream) {
+  hipLaunchKernelGGL(k_substr_ranges, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)ao, (const uint8_t*)ab,
+                     (int32_t)start, (int32_t)slen, (int32_t*)bstart,
+                     (int64_t*)blen, n);
+}
+
+void hipdf_substr_copy(const void* ab, const void* bstart, const void* blen,
+                       const void* out_off, void* out_bytes, int64_t n,
+                       hipStream_t stream) {
+  int64_t blocks = (n * WAVE + HIPDF_BLOCK - 1) / HIPDF_BLOCK;
+  if (blocks > 4 * HIPDF_MAX_BLOCKS) blocks = 4 * HIPDF_MAX_BLOCKS;
+  if (blocks < 1) blocks = 1;
+  hipLaunchKernelGGL(k_substr_copy, dim3((uint32_t)blocks), dim3(HIPDF_BLOCK),
+                     0, stream, (const uint8_t*)ab, (const int32_t*)bstart,
+                     (const int64_t*)blen, (const int64_t*)out_off,
+                     (uint8_t*)out_bytes, n);
+}
+
+}  // extern "C"

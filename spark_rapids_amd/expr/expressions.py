@@ -106,6 +106,31 @@ class Expression:
     def is_not_null(self) -> "Expression":
         return UnaryExpr("not", IsNull(self))
 
+    # string DSL
+    def contains(self, pattern: str) -> "StringPredicate":
+        return StringPredicate("contains", self, pattern)
+
+    def startswith(self, pattern: str) -> "StringPredicate":
+        return StringPredicate("starts_with", self, pattern)
+
+    def endswith(self, pattern: str) -> "StringPredicate":
+        return StringPredicate("ends_with", self, pattern)
+
+    def like(self, pattern: str) -> "StringPredicate":
+        return StringPredicate("like", self, pattern)
+
+    def substr(self, pos: int, length: int = -1) -> "Substring":
+        return Substring(self, pos, length)
+
+    def length(self) -> "UnaryExpr":
+        return UnaryExpr("length", self)
+
+    def upper(self) -> "UnaryExpr":
+        return UnaryExpr("upper", self)
+
+    def lower(self) -> "UnaryExpr":
+        return UnaryExpr("lower", self)
+
 
 def _as_expr(v) -> Expression:
     if isinstance(v, Expression):
@@ -405,6 +430,54 @@ class CaseWhen(Expression):
         parts = " ".join(f"WHEN {c} THEN {v}" for c, v in self.branches)
         e = f" ELSE {self.else_expr}" if self.else_expr is not None else ""
         return f"CASE {parts}{e} END"
+
+
+class StringPredicate(Expression):
+    """contains / starts_with / ends_with / LIKE against a literal pattern
+    (reference analogue: GpuContains/GpuStartsWith/GpuLike)."""
+
+    def __init__(self, op: str, child: Expression, pattern: str):
+        self.op = op
+        self.child = child
+        self.pattern = pattern
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return BOOL
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        return ops.backend_for(c).str_predicate(self.op, c, self.pattern)
+
+    def __str__(self):
+        return f"{self.op}({self.child}, {self.pattern!r})"
+
+
+class Substring(Expression):
+    """Spark substring(str, pos, len): 1-based pos in codepoints, negative
+    pos counts from the end; len < 0 means to-the-end."""
+
+    def __init__(self, child: Expression, pos: int, length: int = -1):
+        self.child = child
+        self.pos = pos
+        self.length = length
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return STRING
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        return ops.backend_for(c).substring(c, self.pos, self.length)
+
+    def __str__(self):
+        return f"substring({self.child}, {self.pos}, {self.length})"
 
 
 # ---------------------------------------------------------------------------

@@ -334,6 +334,40 @@ def _dec_str(unscaled: int, scale: int) -> str:
     return f"{sign}{s[:-scale]}.{s[-scale:]}"
 
 
+def str_predicate(op: str, col: Column, pattern: str) -> Column:
+    a, av = _vals(col), _valid(col)
+    if op == "like":
+        import re
+
+        rx = re.escape(pattern).replace(r"\%", ".*").replace("%", ".*") \
+            .replace(r"\_", ".").replace("_", ".")
+        prog = re.compile(f"^{rx}$", re.DOTALL)
+        res = np.array([bool(prog.match(x)) if x is not None else False
+                        for x in a], dtype=np.uint8)
+    else:
+        fn = {"contains": lambda x: pattern in x,
+              "starts_with": lambda x: x.startswith(pattern),
+              "ends_with": lambda x: x.endswith(pattern)}[op]
+        res = np.array([fn(x) if x is not None else False for x in a],
+                       dtype=np.uint8)
+    return _make(res, av if not av.all() else None, DType.bool_())
+
+
+def substring(col: Column, pos: int, length: int = -1) -> Column:
+    a, av = _vals(col), _valid(col)
+    out = []
+    for x, ok in zip(a, av):
+        if not ok or x is None:
+            out.append(None)
+            continue
+        n = len(x)
+        begin = pos - 1 if pos > 0 else (n + pos if pos < 0 else 0)
+        begin = max(begin, 0)
+        end = n if length < 0 else min(begin + length, n)
+        out.append(x[begin:end] if begin < n else "")
+    return Column.from_pylist(out, DType.string())
+
+
 def if_else(cond: Column, a: Column, b: Column) -> Column:
     """Rowwise cond ? a : b. A NULL condition selects b (Spark CASE WHEN)."""
     c = _vals(cond).astype(bool) & _valid(cond)

@@ -115,10 +115,29 @@ def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
     return _binary(op, lhs, None, scalar, out_dtype)
 
 
+_STR_CMP = {"eq": 0, "ne": 1, "lt": 2, "le": 3, "gt": 4, "ge": 5}
+
+
 def _binary(op, lhs: Column, rhs: Optional[Column], scalar, out_dtype) -> Column:
     n = lhs.size
     s = _stream()
     scalar_rhs = rhs is None
+    if lhs.dtype.id is TypeId.STRING:
+        if op not in _STR_CMP:
+            raise NotImplementedError(f"string op {op} not on GPU")
+        out = _alloc(n, out_dtype)
+        if scalar_rhs:
+            pat = _pattern_tensor(scalar)
+            ext.str_cmp_scalar(_STR_CMP[op], lhs.offsets.data_ptr(),
+                               lhs.data.data_ptr(), pat.data_ptr(),
+                               pat.numel(), out.data_ptr(), n, s)
+            v = lhs.validity.clone() if lhs.validity is not None else None
+            return Column(out_dtype, n, out, v, null_count=lhs._null_count)
+        ext.str_cmp(_STR_CMP[op], lhs.offsets.data_ptr(), lhs.data.data_ptr(),
+                    rhs.offsets.data_ptr(), rhs.data.data_ptr(),
+                    out.data_ptr(), n, s)
+        # combine validities via the bool AND kernel on expanded masks
+        return _with_and_validity(out, lhs, rhs, out_dtype, n, s)
     t = _ht(lhs.dtype)
     sd, si = 0.0, 0
     if scalar_rhs and scalar is not None:
@@ -158,9 +177,88 @@ def _binary(op, lhs: Column, rhs: Optional[Column], scalar, out_dtype) -> Column
     return Column(out_dtype, n, out, ov, null_count=None if need_mask else 0)
 
 
+def _pattern_tensor(pattern) -> torch.Tensor:
+    b = str(pattern).encode("utf-8")
+    if not b:
+        return torch.zeros(1, dtype=torch.uint8, device="cuda")[:0]
+    return torch.frombuffer(bytearray(b), dtype=torch.uint8).cuda()
+
+
+def _and_masks(a: Optional[torch.Tensor], b: Optional[torch.Tensor]):
+    if a is None:
+        return b.clone() if b is not None else None
+    if b is None:
+        return a.clone()
+    n = min(a.numel(), b.numel())
+    return (a[:n].view(torch.int32) & b[:n].view(torch.int32)).view(torch.uint8)
+
+
+def _with_and_validity(out, lhs, rhs, out_dtype, n, s):
+    v = _and_masks(lhs.validity, rhs.validity)
+    return Column(out_dtype, n, out, v,
+                  null_count=None if v is not None else 0)
+
+
+def str_predicate(op: str, col: Column, pattern: str) -> Column:
+    n = col.size
+    s = _stream()
+    out = _alloc(n, DType.bool_())
+    pat = _pattern_tensor(pattern)
+    if op == "like":
+        ext.str_like(col.offsets.data_ptr(), col.data.data_ptr(),
+                     pat.data_ptr(), pat.numel(), out.data_ptr(), n, s)
+    else:
+        mode = {"contains": 0, "starts_with": 1, "ends_with": 2}[op]
+        ext.str_find(mode, col.offsets.data_ptr(), col.data.data_ptr(),
+                     pat.data_ptr(), pat.numel(), out.data_ptr(), n, s)
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(DType.bool_(), n, out, v, null_count=col._null_count)
+
+
+def substring(col: Column, pos: int, length: int = -1) -> Column:
+    n = col.size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.string())
+    bstart = torch.empty(n, dtype=torch.int32, device="cuda")
+    blen = torch.empty(n, dtype=torch.int64, device="cuda")
+    ext.substr_ranges(col.offsets.data_ptr(), col.data.data_ptr(), pos,
+                      length, bstart.data_ptr(), blen.data_ptr(), n, s)
+    scanned, total = _exclusive_scan_i64(blen)
+    out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
+                            device="cuda")[:total]
+    if total:
+        ext.substr_copy(col.data.data_ptr(), bstart.data_ptr(),
+                        blen.data_ptr(), scanned.data_ptr(),
+                        out_bytes.data_ptr(), n, s)
+    offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+    offs[n] = total
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(DType.string(), n, out_bytes, v, offs,
+                  null_count=col._null_count)
+
+
 def unary_op(op: str, col: Column, out_dtype: DType) -> Column:
     n = col.size
     s = _stream()
+    if col.dtype.id is TypeId.STRING:
+        v = col.validity.clone() if col.validity is not None else None
+        if op == "length":
+            out = _alloc(n, out_dtype)
+            ext.str_length(col.offsets.data_ptr(), col.data.data_ptr(),
+                           out.data_ptr(), n, s)
+            return Column(out_dtype, n, out, v, null_count=col._null_count)
+        if op in ("upper", "lower"):
+            nb = int(col.data.numel())
+            ob = torch.empty(max(nb, 1), dtype=torch.uint8,
+                             device="cuda")[:nb]
+            if nb:
+                ext.str_case(op == "upper", col.data.data_ptr(),
+                             ob.data_ptr(), nb, s)
+            return Column(out_dtype, n, ob, v, col.offsets.clone(),
+                          null_count=col._null_count)
+        raise NotImplementedError(f"string unary {op} not on GPU")
     out = _alloc(n, out_dtype)
     if op in ("not", "is_nan", "year", "month", "day"):
         ext.unary(_UN_OPS[op], _ht(col.dtype), col.data.data_ptr(),

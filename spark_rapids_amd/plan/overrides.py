@@ -12,8 +12,10 @@ from __future__ import annotations
 from typing import List, Optional
 
 from ..config import RapidsConf
-from ..expr.expressions import (Alias, BinaryExpr, CaseWhen, CastExpr, ColumnRef,
-                                Expression, IsNull, Literal, UnaryExpr)
+from ..config import ALLOW_INCOMPAT as _ALLOW_INCOMPAT
+from ..expr.expressions import (Alias, BinaryExpr, CaseWhen, CastExpr,
+                                ColumnRef, Expression, IsNull, Literal,
+                                StringPredicate, Substring, UnaryExpr)
 from ..types import DType, TypeId, TypeSig
 from . import logical as L
 from . import physical as P
@@ -42,8 +44,9 @@ _GPU_UNARY_OPS = {
     "neg", "abs", "not", "sqrt", "exp", "log", "floor", "ceil",
     "sin", "cos", "tan", "is_nan", "year", "month", "day",
 }
-# ops that only work on fixed-width inputs on GPU this round
-_GPU_STRING_OK = set()  # string compute exprs not yet on GPU (pass-through is)
+# string ops with GPU kernels (strings.hip); eq_null_safe still CPU-only
+_GPU_STRING_OK = {"eq", "ne", "lt", "le", "gt", "ge"}
+_GPU_STRING_UNARY = {"length", "upper", "lower"}
 
 
 class TagReason:
@@ -84,11 +87,20 @@ class Tagger:
                 out.append(f"expression {e.op} disabled by conf")
         elif isinstance(e, UnaryExpr):
             in_t = e.child.dtype(schema)
-            if e.op not in _GPU_UNARY_OPS:
+            if in_t.id is TypeId.STRING:
+                if e.op not in _GPU_STRING_UNARY:
+                    out.append(f"unary op {e.op} on string not on GPU yet")
+                elif e.op in ("upper", "lower") and \
+                        not self.conf.get(_ALLOW_INCOMPAT):
+                    out.append(f"{e.op} on GPU is ASCII-only "
+                               "(spark.rapids.sql.incompatibleOps.enabled)")
+            elif e.op not in _GPU_UNARY_OPS:
                 out.append(f"unary op {e.op} has no GPU kernel")
-            elif in_t.id is TypeId.STRING:
-                out.append(f"unary op {e.op} on string not on GPU yet")
         elif isinstance(e, (IsNull, CaseWhen)):
+            pass
+        elif isinstance(e, StringPredicate):
+            pass  # contains/starts/ends/like have GPU kernels
+        elif isinstance(e, Substring):
             pass
         else:
             out.append(f"expression {type(e).__name__} not supported on GPU")

@@ -1,0 +1,123 @@
+"""String function tests: CPU semantics + GPU kernel equality."""
+import numpy as np
+import pytest
+
+import spark_rapids_amd as sr
+from spark_rapids_amd import Column, ColumnBatch, STRING, col
+from spark_rapids_amd.ops import cpu_backend
+
+WORDS = ["hello", "Hello World", "", "spark", None, "SPARKLE", "park", "spa",
+         "end%", "a_b", "wörld", "sp"]
+
+
+def _scol(n=1200):
+    return Column.from_pylist([WORDS[i % len(WORDS)] for i in range(n)], STRING)
+
+
+def test_cpu_contains_like(session):
+    df = session.create_dataframe({"s": WORDS})
+    out = df.select(col("s").contains("park").alias("x")).to_pydict()["x"]
+    assert out == [False, False, False, True, None, False, True, False,
+                   False, False, False, False]
+    out = df.select(col("s").like("sp%").alias("x")).to_pydict()["x"]
+    assert out == [False, False, False, True, None, False, False, True,
+                   False, False, False, True]
+    out = df.select(col("s").like("a_b").alias("x")).to_pydict()["x"]
+    assert out[9] is True
+
+
+def test_cpu_substring(session):
+    df = session.create_dataframe({"s": ["hello", None, "ab", ""]})
+    out = df.select(col("s").substr(2, 3).alias("x")).to_pydict()["x"]
+    assert out == ["ell", None, "b", ""]
+    out = df.select(col("s").substr(-3, 2).alias("x")).to_pydict()["x"]
+    assert out == ["ll", None, "ab", ""]
+
+
+def test_cpu_length_case(session):
+    df = session.create_dataframe({"s": ["abc", "wörld", None, ""]})
+    out = df.select(col("s").length().alias("x")).to_pydict()["x"]
+    assert out == [3, 5, None, 0]
+    out = df.select(col("s").upper().alias("x")).to_pydict()["x"]
+    assert out == ["ABC", "WÖRLD", None, ""]
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("op", ["contains", "starts_with", "ends_with"])
+def test_gpu_str_find(op):
+    from spark_rapids_amd.ops import gpu_backend
+    c = _scol()
+    for pat in ("spark", "sp", "", "park"):
+        cpu = cpu_backend.str_predicate(op, c, pat)
+        gpu = gpu_backend.str_predicate(op, c.cuda(), pat).cpu()
+        assert cpu.to_pylist() == gpu.to_pylist(), (op, pat)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("pat", ["sp%", "%ark%", "a_b", "%", "_", "he__o",
+                                 "w_rld"])
+def test_gpu_like(pat):
+    from spark_rapids_amd.ops import gpu_backend
+    c = _scol()
+    cpu = cpu_backend.str_predicate("like", c, pat)
+    gpu = gpu_backend.str_predicate("like", c.cuda(), pat).cpu()
+    assert cpu.to_pylist() == gpu.to_pylist(), pat
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("op", ["eq", "ne", "lt", "ge"])
+def test_gpu_str_cmp(op):
+    from spark_rapids_amd.ops import gpu_backend
+    from spark_rapids_amd import DType
+    a, b = _scol(), Column.from_pylist(
+        [WORDS[(i * 7 + 3) % len(WORDS)] for i in range(1200)], STRING)
+    cpu = cpu_backend.binary_op(op, a, b, DType.bool_())
+    gpu = gpu_backend.binary_op(op, a.cuda(), b.cuda(), DType.bool_()).cpu()
+    assert cpu.to_pylist() == gpu.to_pylist()
+    cpu = cpu_backend.binary_op_scalar(op, a, "spark", DType.bool_())
+    gpu = gpu_backend.binary_op_scalar(op, a.cuda(), "spark",
+                                       DType.bool_()).cpu()
+    assert cpu.to_pylist() == gpu.to_pylist()
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("pos,ln", [(2, 3), (1, -1), (-3, 2), (0, 2), (7, 5)])
+def test_gpu_substring(pos, ln):
+    from spark_rapids_amd.ops import gpu_backend
+    c = _scol()
+    cpu = cpu_backend.substring(c, pos, ln)
+    gpu = gpu_backend.substring(c.cuda(), pos, ln).cpu()
+    assert cpu.to_pylist() == gpu.to_pylist()
+
+
+@pytest.mark.gpu
+def test_gpu_length_and_case_ascii():
+    from spark_rapids_amd.ops import gpu_backend
+    from spark_rapids_amd.types import INT32, STRING as STR
+    ascii_col = Column.from_pylist(["abc", "XYZ", None, "", "MiXeD1"], STRING)
+    cpu = cpu_backend.unary_op("length", ascii_col, INT32)
+    gpu = gpu_backend.unary_op("length", ascii_col.cuda(), INT32).cpu()
+    assert cpu.to_pylist() == gpu.to_pylist()
+    for op in ("upper", "lower"):
+        cpu = cpu_backend.unary_op(op, ascii_col, STR)
+        gpu = gpu_backend.unary_op(op, ascii_col.cuda(), STR).cpu()
+        assert cpu.to_pylist() == gpu.to_pylist()
+
+
+@pytest.mark.gpu
+def test_gpu_string_filter_e2e():
+    s = sr.Session()
+    n = 30_000
+    df = s.create_dataframe({
+        "s": [WORDS[i % len(WORDS)] for i in range(n)],
+        "v": list(range(n)),
+    })
+    tree = df.filter(col("s").like("sp%")).physical_plan().tree_string()
+    assert "GpuFilter" in tree, tree
+    gpu = df.filter(col("s").like("sp%")).count()
+    s2 = sr.Session({"spark.rapids.sql.enabled": False})
+    df2 = s2.create_dataframe({
+        "s": [WORDS[i % len(WORDS)] for i in range(n)],
+        "v": list(range(n)),
+    })
+    assert gpu == df2.filter(col("s").like("sp%")).count()
