@@ -221,6 +221,35 @@ class Session:
         batch = df.collect_batch()
         write_parquet(batch, df.schema, path, compression)
 
+    def read_csv(self, path: str, header: bool = True,
+                 delimiter: str = ",") -> DataFrame:
+        from .io.formats import CsvTable
+
+        src = CsvTable(path, header, delimiter)
+        return DataFrame(self, L.Scan(src, src.schema, f"csv:{path}"))
+
+    def read_orc(self, path: str) -> DataFrame:
+        from .io.formats import OrcTable
+
+        src = OrcTable(path)
+        return DataFrame(self, L.Scan(src, src.schema, f"orc:{path}"))
+
+    def read_json(self, path: str) -> DataFrame:
+        from .io.formats import JsonTable
+
+        src = JsonTable(path)
+        return DataFrame(self, L.Scan(src, src.schema, f"json:{path}"))
+
+    def write_csv(self, df: DataFrame, path: str):
+        from .io.formats import write_csv
+
+        write_csv(df.collect_batch(), df.schema, path)
+
+    def write_orc(self, df: DataFrame, path: str):
+        from .io.formats import write_orc
+
+        write_orc(df.collect_batch(), df.schema, path)
+
 
 def _infer_list_dtype(v: list) -> DType:
     from .expr.expressions import _infer_literal_dtype
