@@ -40,6 +40,7 @@ void hipdf_gather_fixed(int, const void*, const void*, void*, int64_t,
                         hipStream_t);
 void hipdf_gather_validity(const void*, int, const void*, void*, int64_t,
                            hipStream_t);
+void hipdf_gather_table(const void*, int, const void*, int64_t, hipStream_t);
 void hipdf_gather_str_lens(const void*, const void*, void*, int64_t,
                            hipStream_t);
 void hipdf_gather_str_bytes(const void*, const void*, const void*,
@@ -64,6 +65,9 @@ void hipdf_gb_number(const void*, const void*, void*, const void*, void*,
 void hipdf_gb_rowgid(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_gb_agg(int, int, const void*, const void*, const void*, void*,
                   void*, int, int32_t, int64_t, hipStream_t);
+void hipdf_gb_agg_multi(const void*, int, const void*, int32_t, int64_t,
+                        hipStream_t);
+void hipdf_gb_acc_init(int, void*, int, int32_t, hipStream_t);
 void hipdf_mask_from_nonzero(const void*, void*, int64_t, hipStream_t);
 void hipdf_join_build(const void*, const void*, int, void*, void*, int64_t,
                       int64_t, hipStream_t);
@@ -199,6 +203,11 @@ PYBIND11_MODULE(hipdf, m) {
     hipdf_gather_fixed(esize, P(in), P(idx), PM(out), n_out, S(stream));
     check_async();
   });
+  m.def("gather_table", [](int64_t cols, int ncols, int64_t idx,
+                           int64_t n_out, int64_t stream) {
+    hipdf_gather_table(P(cols), ncols, P(idx), n_out, S(stream));
+    check_async();
+  });
   m.def("gather_validity", [](int64_t in_valid, bool in_has, int64_t idx,
                               int64_t out_valid, int64_t n_out,
                               int64_t stream) {
@@ -288,6 +297,16 @@ PYBIND11_MODULE(hipdf, m) {
                      int64_t stream) {
     hipdf_gb_agg(op, t, P(vals), P(vvalid), P(row_gid), PM(acc), PM(cnt),
                  acc_is_double, ngroups, n, S(stream));
+    check_async();
+  });
+  m.def("gb_acc_init", [](int op, int64_t acc, bool is_double, int ngroups,
+                          int64_t stream) {
+    hipdf_gb_acc_init(op, PM(acc), is_double, ngroups, S(stream));
+    check_async();
+  });
+  m.def("gb_agg_multi", [](int64_t aggs, int naggs, int64_t row_gid,
+                           int ngroups, int64_t n, int64_t stream) {
+    hipdf_gb_agg_multi(P(aggs), naggs, P(row_gid), ngroups, n, S(stream));
     check_async();
   });
   m.def("mask_from_nonzero", [](int64_t cnt, int64_t mask, int64_t n,

@@ -156,6 +156,113 @@ __global__ void k_gb_agg(int op, const T* __restrict__ vals,
   }
 }
 
+// ---- fused multi-aggregate: one pass over row_gid and every value column
+// (saves one full re-read of row_gid + one kernel launch per aggregate) ----
+struct AggDesc {
+  int op;             // GbOp
+  int type;           // HType of the value column
+  int acc_is_double;  // accumulator: double (1) or int64 (0)
+  const void* vals;
+  const uint64_t* valid;
+  void* acc;          // [ngroups] double or int64
+  int64_t* cnt;       // [ngroups]
+};
+
+__device__ __forceinline__ double load_as_double(const void* p, int t,
+                                                 int64_t i) {
+  switch (t) {
+    case HT_U8: return ((const uint8_t*)p)[i];
+    case HT_I8: return ((const int8_t*)p)[i];
+    case HT_I16: return ((const int16_t*)p)[i];
+    case HT_I32: return ((const int32_t*)p)[i];
+    case HT_I64: return (double)((const int64_t*)p)[i];
+    case HT_F32: return ((const float*)p)[i];
+    default: return ((const double*)p)[i];
+  }
+}
+
+__device__ __forceinline__ int64_t load_as_i64(const void* p, int t,
+                                               int64_t i) {
+  switch (t) {
+    case HT_U8: return ((const uint8_t*)p)[i];
+    case HT_I8: return ((const int8_t*)p)[i];
+    case HT_I16: return ((const int16_t*)p)[i];
+    case HT_I32: return ((const int32_t*)p)[i];
+    case HT_I64: return ((const int64_t*)p)[i];
+    case HT_F32: return (int64_t)((const float*)p)[i];
+    default: return (int64_t)((const double*)p)[i];
+  }
+}
+
+template <bool USE_LDS>
+__global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
+                               const int32_t* __restrict__ row_gid,
+                               int32_t ngroups, int64_t n) {
+  extern __shared__ char lds_raw[];
+  // LDS layout per agg a: acc[a][ngroups] (8B each) then cnt[a][ngroups]
+  if (USE_LDS) {
+    for (int a = 0; a < naggs; ++a) {
+      char* base = lds_raw + (size_t)a * ngroups * 16;
+      int op = aggs[a].op;
+      if (aggs[a].acc_is_double) {
+        double* lacc = (double*)base;
+        for (int g = threadIdx.x; g < ngroups; g += blockDim.x)
+          lacc[g] = acc_init<double>(op);
+      } else {
+        int64_t* lacc = (int64_t*)base;
+        for (int g = threadIdx.x; g < ngroups; g += blockDim.x)
+          lacc[g] = acc_init<int64_t>(op);
+      }
+      int64_t* lcnt = (int64_t*)(base + (size_t)ngroups * 8);
+      for (int g = threadIdx.x; g < ngroups; g += blockDim.x) lcnt[g] = 0;
+    }
+    __syncthreads();
+  }
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t g = row_gid[i];
+    for (int a = 0; a < naggs; ++a) {
+      const AggDesc& d = aggs[a];
+      char* base = lds_raw + (size_t)a * ngroups * 16;
+      int64_t* cnt_p = USE_LDS ? (int64_t*)(base + (size_t)ngroups * 8)
+                               : d.cnt;
+      if (d.op == GB_COUNT_ALL) {
+        atomicAdd((unsigned long long*)&cnt_p[g], 1ull);
+        continue;
+      }
+      if (!valid_bit(d.valid, i)) continue;
+      atomicAdd((unsigned long long*)&cnt_p[g], 1ull);
+      if (d.op == GB_COUNT) continue;
+      if (d.acc_is_double) {
+        double* acc_p = USE_LDS ? (double*)base : (double*)d.acc;
+        acc_atomic<double>(d.op, &acc_p[g], load_as_double(d.vals, d.type, i));
+      } else {
+        int64_t* acc_p = USE_LDS ? (int64_t*)base : (int64_t*)d.acc;
+        acc_atomic<int64_t>(d.op, &acc_p[g], load_as_i64(d.vals, d.type, i));
+      }
+    }
+  }
+  if (USE_LDS) {
+    __syncthreads();
+    for (int a = 0; a < naggs; ++a) {
+      const AggDesc& d = aggs[a];
+      char* base = lds_raw + (size_t)a * ngroups * 16;
+      int64_t* lcnt = (int64_t*)(base + (size_t)ngroups * 8);
+      for (int g = threadIdx.x; g < ngroups; g += blockDim.x) {
+        if (!lcnt[g]) continue;
+        atomicAdd((unsigned long long*)&d.cnt[g],
+                  (unsigned long long)lcnt[g]);
+        if (d.op == GB_COUNT || d.op == GB_COUNT_ALL) continue;
+        if (d.acc_is_double)
+          acc_atomic<double>(d.op, &((double*)d.acc)[g], ((double*)base)[g]);
+        else
+          acc_atomic<int64_t>(d.op, &((int64_t*)d.acc)[g],
+                              ((int64_t*)base)[g]);
+      }
+    }
+  }
+}
+
 // init global accumulators for min/max identities
 template <typename ACC>
 __global__ void k_gb_acc_init(int op, ACC* __restrict__ acc, int32_t ngroups) {
@@ -233,6 +340,33 @@ void hipdf_gb_agg(int op, int t, const void* vals, const void* vvalid,
     if (acc_is_double) launch.template operator()<T, double>();
     else launch.template operator()<T, int64_t>();
   });
+}
+
+void hipdf_gb_acc_init(int op, void* acc, int acc_is_double, int32_t ngroups,
+                       hipStream_t stream) {
+  if (acc_is_double)
+    hipLaunchKernelGGL((k_gb_acc_init<double>), flat_grid(ngroups),
+                       dim3(HIPDF_BLOCK), 0, stream, op, (double*)acc,
+                       ngroups);
+  else
+    hipLaunchKernelGGL((k_gb_acc_init<int64_t>), flat_grid(ngroups),
+                       dim3(HIPDF_BLOCK), 0, stream, op, (int64_t*)acc,
+                       ngroups);
+}
+
+void hipdf_gb_agg_multi(const void* aggs, int naggs, const void* row_gid,
+                        int32_t ngroups, int64_t n, hipStream_t stream) {
+  size_t lds = (size_t)naggs * ngroups * 16;
+  bool use_lds = lds > 0 && lds <= 64 * 1024;
+  dim3 grid = flat_grid(n, 4);
+  if (use_lds)
+    hipLaunchKernelGGL((k_gb_agg_multi<true>), grid, dim3(HIPDF_BLOCK), lds,
+                       stream, (const AggDesc*)aggs, naggs,
+                       (const int32_t*)row_gid, ngroups, n);
+  else
+    hipLaunchKernelGGL((k_gb_agg_multi<false>), grid, dim3(HIPDF_BLOCK), 0,
+                       stream, (const AggDesc*)aggs, naggs,
+                       (const int32_t*)row_gid, ngroups, n);
 }
 
 void hipdf_mask_from_nonzero(const void* cnt, void* mask, int64_t n,

@@ -108,6 +108,47 @@ __global__ void k_gather_validity(const uint64_t* __restrict__ in_valid,
   }
 }
 
+// fused whole-table gather for fixed-width columns: one launch gathers
+// every column (the index load and ballot amortize across columns)
+struct GatherCol {
+  int esize;
+  int in_has_valid;
+  const void* in;
+  const uint64_t* in_valid;
+  void* out;
+  uint64_t* out_valid;  // may be null
+};
+
+__global__ void k_gather_table(const GatherCol* __restrict__ cols, int ncols,
+                               const int32_t* __restrict__ idx,
+                               int64_t nstripe, int64_t n_out) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t s = wave_global; s < nstripe; s += wave_count) {
+    int64_t j = s * WAVE + lane;
+    int32_t i = j < n_out ? idx[j] : -1;
+    for (int c = 0; c < ncols; ++c) {
+      const GatherCol& g = cols[c];
+      if (j < n_out) {
+        int32_t src = i >= 0 ? i : 0;
+        switch (g.esize) {
+          case 1: ((uint8_t*)g.out)[j] = i >= 0 ? ((const uint8_t*)g.in)[src] : 0; break;
+          case 2: ((uint16_t*)g.out)[j] = i >= 0 ? ((const uint16_t*)g.in)[src] : 0; break;
+          case 4: ((uint32_t*)g.out)[j] = i >= 0 ? ((const uint32_t*)g.in)[src] : 0; break;
+          default: ((uint64_t*)g.out)[j] = i >= 0 ? ((const uint64_t*)g.in)[src] : 0; break;
+        }
+      }
+      if (g.out_valid) {
+        bool ok = j < n_out && i >= 0 &&
+                  (!g.in_has_valid || valid_bit(g.in_valid, i));
+        uint64_t ballot = __ballot(ok);
+        if (lane == 0) g.out_valid[s] = ballot;
+      }
+    }
+  }
+}
+
 // string gather phase 1: per-output-row byte length
 __global__ void k_gather_str_lens(const int32_t* __restrict__ offsets,
                                   const int32_t* __restrict__ idx,
@@ -226,6 +267,13 @@ void hipdf_gather_fixed(int esize, const void* in, const void* idx, void* out,
     default:
       throw std::runtime_error("gather: bad element size");
   }
+}
+
+void hipdf_gather_table(const void* cols, int ncols, const void* idx,
+                        int64_t n_out, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gather_table, stripe_grid(n_out), dim3(HIPDF_BLOCK), 0,
+                     stream, (const GatherCol*)cols, ncols,
+                     (const int32_t*)idx, n_stripes(n_out), n_out);
 }
 
 void hipdf_gather_validity(const void* in_valid, int in_has_valid,

@@ -434,7 +434,32 @@ def _gather_col(c: Column, idx: torch.Tensor, n_out: int,
 
 def _gather_by_idx(batch: ColumnBatch, idx: torch.Tensor, n_out: int,
                    maybe_negative: bool) -> ColumnBatch:
-    cols = [_gather_col(c, idx, n_out, maybe_negative) for c in batch.columns]
+    s = _stream()
+    cols: List[Optional[Column]] = [None] * len(batch.columns)
+    fixed: List[tuple] = []  # (position, column)
+    for ci, c in enumerate(batch.columns):
+        if c.dtype.id is TypeId.STRING or n_out == 0:
+            cols[ci] = _gather_col(c, idx, n_out, maybe_negative)
+        else:
+            fixed.append((ci, c))
+    if fixed:
+        blobs = []
+        outs = []
+        for ci, c in fixed:
+            out = _alloc(n_out, c.dtype)
+            need_valid = c.validity is not None or maybe_negative
+            ov = _alloc_mask(n_out) if need_valid else None
+            outs.append((ci, c, out, ov))
+            blobs.append(struct.pack(
+                "<iiqqqq", c.dtype.itemsize,
+                1 if c.validity is not None else 0, c.data.data_ptr(),
+                _ptr(c.validity), out.data_ptr(), _ptr(ov)))
+        desc = torch.frombuffer(bytearray(b"".join(blobs)),
+                                dtype=torch.uint8).cuda()
+        ext.gather_table(desc.data_ptr(), len(fixed), idx.data_ptr(), n_out, s)
+        for ci, c, out, ov in outs:
+            cols[ci] = Column(c.dtype, n_out, out, ov,
+                              null_count=None if ov is not None else 0)
     return ColumnBatch(cols, n_out)
 
 
@@ -647,8 +672,14 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         ext.gb_rowgid(row_slot.data_ptr(), slot_gid.data_ptr(),
                       row_gid.data_ptr(), n, s)
     leaders = leaders[:ngroups]
-    out_cols = [_gather_col(k, leaders, ngroups, maybe_negative=False)
-                for k in keys]
+    key_batch = _gather_by_idx(ColumnBatch(keys, n) if keys else
+                               ColumnBatch([], 0), leaders, ngroups,
+                               maybe_negative=False) if keys else None
+    out_cols = list(key_batch.columns) if key_batch is not None else []
+
+    # fused multi-aggregate: one kernel pass accumulates every agg
+    allocs = []
+    blobs = []
     for op, vidx, out_dtype in aggs:
         vc = batch.columns[vidx] if vidx >= 0 else None
         acc_is_double = out_dtype.is_floating or (
@@ -658,15 +689,24 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                           device="cuda")
         cnt = torch.zeros(max(ngroups, 1), dtype=torch.int64, device="cuda")
         t = _ht(vc.dtype) if vc is not None else 4
-        ext.gb_agg(_GB[op], t, _ptr(vc.data if vc is not None else None),
-                   _ptr(vc.validity if vc is not None else None),
-                   row_gid.data_ptr(), acc.data_ptr(), cnt.data_ptr(),
-                   acc_is_double, ngroups, n, s)
+        if op not in ("count", "count_all"):
+            ext.gb_acc_init(_GB[op], acc.data_ptr(), acc_is_double, ngroups, s)
+        allocs.append((op, out_dtype, acc_is_double, acc, cnt))
+        blobs.append(struct.pack(
+            "<iiiiqqqq", _GB[op], t, acc_is_double, 0,
+            _ptr(vc.data if vc is not None else None),
+            _ptr(vc.validity if vc is not None else None),
+            acc.data_ptr(), cnt.data_ptr()))
+    if aggs:
+        desc = torch.frombuffer(bytearray(b"".join(blobs)),
+                                dtype=torch.uint8).cuda()
+        ext.gb_agg_multi(desc.data_ptr(), len(aggs), row_gid.data_ptr(),
+                         ngroups, n, s)
+    for op, out_dtype, acc_is_double, acc, cnt in allocs:
         if op in ("count", "count_all"):
             out_cols.append(Column(out_dtype, ngroups, cnt[:ngroups].clone(),
                                    None, null_count=0))
             continue
-        # value: cast accumulator to out dtype; validity from count>0
         acc_dt = DType.float64() if acc_is_double else DType.int64()
         data = acc[:ngroups]
         if torch_dtype(out_dtype) != data.dtype:
