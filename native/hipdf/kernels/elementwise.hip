@@ -10,7 +10,7 @@
 enum BinOp : int {
   OP_ADD = 0, OP_SUB, OP_MUL, OP_DIV, OP_INT_DIV, OP_MOD, OP_PMOD, OP_POW,
   OP_EQ, OP_NE, OP_LT, OP_LE, OP_GT, OP_GE, OP_EQ_NS, OP_AND, OP_OR,
-  OP_BITAND, OP_BITOR, OP_BITXOR, OP_SHL, OP_SHR, OP_MIN, OP_MAX,
+  OP_BITAND, OP_BITOR, OP_BITXOR, OP_SHL, OP_SHR, OP_MIN, OP_MAX, OP_ROUND,
 };
 
 enum UnOp : int {
@@ -97,6 +97,13 @@ __device__ __forceinline__ T arith_one(int op, T a, T b, bool& ok) {
     case OP_SHR: return (T)(to_i64(a) >> (to_i64(b) & (sizeof(T) == 8 ? 63 : 31)));
     case OP_MIN: return spark_lt(a, b) ? a : b;
     case OP_MAX: return spark_lt(a, b) ? b : a;
+    case OP_ROUND: {
+      // HALF_UP at scale encoded as b = 10^scale (Spark round())
+      double p = (double)b;
+      double x = (double)a * p;
+      double r = x >= 0 ? floor(x + 0.5) : ceil(x - 0.5);
+      return (T)(r / p);
+    }
     default: return (T)0;
   }
 }

@@ -9,7 +9,9 @@ from .api import DataFrame, Session
 from .column import Column, ColumnBatch, Field, Schema
 from .config import RapidsConf, help_doc
 from .expr.aggregates import avg, count, count_star, max_, min_, sum_
-from .expr.expressions import CaseWhen, col, lit, when
+from .expr.expressions import (CaseWhen, coalesce, col, date_add, date_sub,
+                               datediff, greatest, isin, least, lit, round_,
+                               when)
 from .expr.windows import (dense_rank, lag, lead, rank, row_number, win_avg,
                            win_count, win_max, win_min, win_sum)
 from .types import (BOOL, DATE32, FLOAT32, FLOAT64, INT8, INT16, INT32, INT64,

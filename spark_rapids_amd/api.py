@@ -108,6 +108,10 @@ class DataFrame:
 
     def collect_batch(self) -> ColumnBatch:
         exec_ = self.physical_plan()
+        from .metrics import instrument
+
+        instrument(exec_)
+        self._last_exec = exec_
         sem = GpuSemaphore.get()
         with sem.held():
             batches = [b.cpu() for b in exec_.execute()]
@@ -136,6 +140,14 @@ class DataFrame:
         rows = self.agg(count_star()).collect()
         return rows[0][0] if rows else 0
 
+    def metrics(self):
+        """Per-operator metrics from the last action (opTimeMs inclusive of
+        children, output rows/batches)."""
+        from .metrics import collect_metrics
+
+        exec_ = getattr(self, "_last_exec", None)
+        return collect_metrics(exec_) if exec_ is not None else []
+
     def explain(self) -> str:
         exec_ = self.physical_plan()
         out = [exec_.tree_string()]
@@ -161,6 +173,10 @@ class Session:
         self.conf = RapidsConf(conf)
         self.catalog: Dict[str, MemTable] = {}
         GpuSemaphore.initialize(self.conf.get(CONCURRENT_GPU_TASKS))
+        from .config import ROCTX_ENABLED
+        from .metrics import enable_roctx
+
+        enable_roctx(self.conf.get(ROCTX_ENABLED))
 
     # ---- conf ----------------------------------------------------------
     def set(self, key: str, value) -> "Session":

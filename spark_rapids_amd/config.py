@@ -164,6 +164,10 @@ GPU_OOM_INJECTION = int_conf(
 STABLE_SORT = bool_conf(
     "spark.rapids.sql.stableSort.enabled", False,
     "Use a stable sort on GPU (matches CPU tie ordering; slightly slower).")
+ROCTX_ENABLED = bool_conf(
+    "spark.rapids.sql.rocTx.enabled", False,
+    "Emit rocTX ranges around each operator (view with rocprofv3 "
+    "--marker-trace; reference analogue: NVTX ranges + nsys).")
 ALLOW_INCOMPAT = bool_conf(
     "spark.rapids.sql.incompatibleOps.enabled", True,
     "Allow operators whose GPU results can differ from the CPU in corner "

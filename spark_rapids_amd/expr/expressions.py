@@ -227,6 +227,9 @@ class BinaryExpr(Expression):
             return BOOL
         if self.op in _DOUBLE_OPS:
             return FLOAT64
+        lt, rt = self.left.dtype(schema), self.right.dtype(schema)
+        if self.op == "sub" and lt.is_timelike and rt.is_timelike:
+            return INT32  # datediff domain
         it = self._in_dtype(schema)
         if it.is_decimal and self.op in ("add", "sub"):
             return DType.decimal(min(it.precision + 1, 38), it.scale)
@@ -484,6 +487,73 @@ class Substring(Expression):
 # public DSL
 # ---------------------------------------------------------------------------
 
+class Coalesce(Expression):
+    def __init__(self, *exprs):
+        self.exprs = [_as_expr(e) for e in exprs]
+
+    @property
+    def children(self):
+        return tuple(self.exprs)
+
+    def dtype(self, schema: Schema) -> DType:
+        t = self.exprs[0].dtype(schema)
+        for e in self.exprs[1:]:
+            et = e.dtype(schema)
+            if t.id is TypeId.NULL:
+                t = et
+            elif et.id is not TypeId.NULL and et != t:
+                t = promote(t, et)
+        return t
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        out_t = self.dtype(schema)
+        acc = ops.cast(self.exprs[-1].eval(batch, schema), out_t)
+        for e in reversed(self.exprs[:-1]):
+            c = ops.cast(e.eval(batch, schema), out_t)
+            acc = ops.backend_for(c, acc).if_else(_not_null(c), c, acc)
+        return acc
+
+    def __str__(self):
+        return f"coalesce({', '.join(str(e) for e in self.exprs)})"
+
+
+def _not_null(c: Column) -> Column:
+    return ops.unary_op("not", ops.is_null(c), BOOL)
+
+
+class Round(Expression):
+    """Spark round(): HALF_UP at `scale` decimal places."""
+
+    def __init__(self, child, scale: int = 0):
+        self.child = _as_expr(child)
+        self.scale = scale
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        t = self.child.dtype(schema)
+        return t if t.is_decimal or t.is_integral else FLOAT64
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        t = self.child.dtype(schema)
+        c = self.child.eval(batch, schema)
+        if t.is_integral and self.scale >= 0:
+            return c
+        if t.is_decimal:
+            shift = t.scale - self.scale
+            if shift <= 0:
+                return c
+            rescaled = ops.cast(c, DType.decimal(t.precision, self.scale))
+            return ops.cast(rescaled, t)
+        c = ops.cast(c, FLOAT64)
+        return ops.backend_for(c).round_half_up(c, self.scale)
+
+    def __str__(self):
+        return f"round({self.child}, {self.scale})"
+
+
 def col(name: str) -> ColumnRef:
     return ColumnRef(name)
 
@@ -494,3 +564,52 @@ def lit(v, dtype: Optional[DType] = None) -> Literal:
 
 def when(cond: Expression, value) -> CaseWhen:
     return CaseWhen([(cond, value)])
+
+
+def coalesce(*exprs) -> Coalesce:
+    return Coalesce(*exprs)
+
+
+def round_(e, scale: int = 0) -> Round:
+    return Round(e, scale)
+
+
+def _null_aware_fold(op: str, exprs):
+    """greatest/least: Spark skips NULLs (result null only if all null)."""
+    es = [_as_expr(e) for e in exprs]
+    acc = es[0]
+    for e in es[1:]:
+        pick = BinaryExpr(op, acc, e)
+        acc = CaseWhen([(IsNull(acc), e), (IsNull(e), acc)], pick)
+    return acc
+
+
+def greatest(*exprs) -> Expression:
+    return _null_aware_fold("max", exprs)
+
+
+def least(*exprs) -> Expression:
+    return _null_aware_fold("min", exprs)
+
+
+def isin(e, *values) -> Expression:
+    e = _as_expr(e)
+    acc = BinaryExpr("eq", e, Literal(values[0]))
+    for v in values[1:]:
+        acc = BinaryExpr("or", acc, BinaryExpr("eq", e, Literal(v)))
+    return acc
+
+
+def date_add(d, days) -> Expression:
+    return BinaryExpr("add", _as_expr(d), _as_expr(days))
+
+
+def date_sub(d, days) -> Expression:
+    return BinaryExpr("sub", _as_expr(d), _as_expr(days))
+
+
+def datediff(end, start) -> Expression:
+    from ..types import DATE32
+
+    return BinaryExpr("sub", CastExpr(_as_expr(end), DATE32),
+                      CastExpr(_as_expr(start), DATE32))

@@ -334,6 +334,14 @@ def _dec_str(unscaled: int, scale: int) -> str:
     return f"{sign}{s[:-scale]}.{s[-scale:]}"
 
 
+def round_half_up(col: Column, scale: int) -> Column:
+    a, av = _vals(col), _valid(col)
+    p = 10.0 ** scale
+    x = a.astype(np.float64) * p
+    r = np.where(x >= 0, np.floor(x + 0.5), np.ceil(x - 0.5)) / p
+    return _make(r, av if not av.all() else None, col.dtype)
+
+
 def str_predicate(op: str, col: Column, pattern: str) -> Column:
     a, av = _vals(col), _valid(col)
     if op == "like":

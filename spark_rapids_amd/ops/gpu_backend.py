@@ -45,7 +45,7 @@ _BIN_OPS = {
     "pmod": 6, "pow": 7, "eq": 8, "ne": 9, "lt": 10, "le": 11, "gt": 12,
     "ge": 13, "eq_null_safe": 14, "and": 15, "or": 16, "bitand": 17,
     "bitor": 18, "bitxor": 19, "shiftleft": 20, "shiftright": 21,
-    "min": 22, "max": 23,
+    "min": 22, "max": 23, "round": 24,
 }
 _CMP_OPS = {"eq", "ne", "lt", "le", "gt", "ge", "eq_null_safe"}
 _BOOL_OPS = {"and", "or"}
@@ -237,6 +237,10 @@ def substring(col: Column, pos: int, length: int = -1) -> Column:
     v = col.validity.clone() if col.validity is not None else None
     return Column(DType.string(), n, out_bytes, v, offs,
                   null_count=col._null_count)
+
+
+def round_half_up(col: Column, scale: int) -> Column:
+    return binary_op_scalar("round", col, float(10 ** scale), col.dtype)
 
 
 def unary_op(op: str, col: Column, out_dtype: DType) -> Column:

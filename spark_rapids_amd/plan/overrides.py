@@ -14,8 +14,9 @@ from typing import List, Optional
 from ..config import RapidsConf
 from ..config import ALLOW_INCOMPAT as _ALLOW_INCOMPAT
 from ..expr.expressions import (Alias, BinaryExpr, CaseWhen, CastExpr,
-                                ColumnRef, Expression, IsNull, Literal,
-                                StringPredicate, Substring, UnaryExpr)
+                                Coalesce, ColumnRef, Expression, IsNull,
+                                Literal, Round, StringPredicate, Substring,
+                                UnaryExpr)
 from ..types import DType, TypeId, TypeSig
 from . import logical as L
 from . import physical as P
@@ -96,7 +97,7 @@ class Tagger:
                                "(spark.rapids.sql.incompatibleOps.enabled)")
             elif e.op not in _GPU_UNARY_OPS:
                 out.append(f"unary op {e.op} has no GPU kernel")
-        elif isinstance(e, (IsNull, CaseWhen)):
+        elif isinstance(e, (IsNull, CaseWhen, Coalesce, Round)):
             pass
         elif isinstance(e, StringPredicate):
             pass  # contains/starts/ends/like have GPU kernels

@@ -1,0 +1,54 @@
+"""coalesce / round / greatest / least / isin / date arithmetic tests."""
+import pytest
+
+from spark_rapids_amd import (DATE32, Session, coalesce, col, date_add,
+                              datediff, greatest, isin, least, lit, round_)
+
+
+def test_coalesce(session):
+    df = session.create_dataframe({"a": [None, 2, None], "b": [10, None, None]})
+    out = df.select(coalesce(col("a"), col("b"), lit(-1)).alias("c")).to_pydict()["c"]
+    assert out == [10, 2, -1]
+
+
+def test_round_half_up(session):
+    df = session.create_dataframe({"x": [2.5, 3.5, -2.5, 1.234, 1.235]})
+    out = df.select(round_(col("x")).alias("r")).to_pydict()["r"]
+    assert out == [3.0, 4.0, -3.0, 1.0, 1.0]
+    out = df.select(round_(col("x"), 2).alias("r")).to_pydict()["r"]
+    assert out == [2.5, 3.5, -2.5, 1.23, pytest.approx(1.24)]
+
+
+def test_greatest_least_skip_nulls(session):
+    df = session.create_dataframe({"a": [1, None, None], "b": [5, 3, None]})
+    out = df.select(greatest(col("a"), col("b")).alias("g")).to_pydict()["g"]
+    assert out == [5, 3, None]
+    out = df.select(least(col("a"), col("b")).alias("l")).to_pydict()["l"]
+    assert out == [1, 3, None]
+
+
+def test_isin(session):
+    df = session.create_dataframe({"a": [1, 2, 3, None]})
+    out = df.filter(isin(col("a"), 1, 3)).to_pydict()["a"]
+    assert out == [1, 3]
+
+
+def test_date_arithmetic(session):
+    df = session.create_dataframe({"d": [10957, 10958]},
+                                  dtypes={"d": DATE32})  # 2000-01-01, -02
+    out = df.select(date_add(col("d"), lit(10)).alias("x")).to_pydict()["x"]
+    assert out == [10967, 10968]
+    out = df.select(datediff(col("d"), lit(10950)).alias("x")).to_pydict()["x"]
+    assert out == [7, 8]
+    sch = df.select(datediff(col("d"), lit(10950)).alias("x")).schema
+    assert str(sch.fields[0].dtype) == "int"
+
+
+def test_year_month_day(session):
+    df = session.create_dataframe({"d": [10957, 11323]}, dtypes={"d": DATE32})
+    from spark_rapids_amd.expr.expressions import UnaryExpr
+
+    y = df.select(UnaryExpr("year", col("d")).alias("y")).to_pydict()["y"]
+    m = df.select(UnaryExpr("month", col("d")).alias("m")).to_pydict()["m"]
+    d = df.select(UnaryExpr("day", col("d")).alias("dd")).to_pydict()["dd"]
+    assert y == [2000, 2001] and m == [1, 1] and d == [1, 1]

@@ -79,3 +79,12 @@ def test_results_identical_cpu_vs_default(session, cpu_session):
                    .group_by("k").agg(sum_(col("v")))
                    .sort("k").collect())
     assert q(session) == q(cpu_session)
+
+
+def test_metrics_collection(session):
+    df = session.create_dataframe({"a": list(range(100))})
+    q = df.filter(__import__("spark_rapids_amd").col("a") > 10)
+    q.collect()
+    m = q.metrics()
+    assert m and any(x["numOutputRows"] == 89 for x in m)
+    assert all("opTimeMs" in x for x in m)
