@@ -177,6 +177,9 @@ class Session:
         from .metrics import enable_roctx
 
         enable_roctx(self.conf.get(ROCTX_ENABLED))
+        from .tools import lore
+
+        lore.configure(self.conf.get_raw("spark.rapids.sql.lore.dumpPath"))
 
     # ---- conf ----------------------------------------------------------
     def set(self, key: str, value) -> "Session":

@@ -164,6 +164,10 @@ GPU_OOM_INJECTION = int_conf(
 STABLE_SORT = bool_conf(
     "spark.rapids.sql.stableSort.enabled", False,
     "Use a stable sort on GPU (matches CPU tie ordering; slightly slower).")
+LORE_DUMP_PATH = str_conf(
+    "spark.rapids.sql.lore.dumpPath", "",
+    "When set, dump every operator's output batches to this directory as "
+    "parquet for offline replay (LORE analogue; tools/lore.py replay()).")
 ROCTX_ENABLED = bool_conf(
     "spark.rapids.sql.rocTx.enabled", False,
     "Emit rocTX ranges around each operator (view with rocprofv3 "

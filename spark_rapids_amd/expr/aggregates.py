@@ -73,3 +73,13 @@ def min_(e) -> AggExpr:
 
 def max_(e) -> AggExpr:
     return AggExpr("max", e)
+
+
+def stddev(e) -> AggExpr:
+    """stddev_samp"""
+    return AggExpr("stddev", e)
+
+
+def variance(e) -> AggExpr:
+    """var_samp"""
+    return AggExpr("variance", e)

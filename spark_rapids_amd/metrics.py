@@ -33,8 +33,11 @@ def instrument(exec_) -> None:
         orig = e.execute
         e.metrics = {"opTimeMs": 0.0, "numOutputRows": 0,
                      "numOutputBatches": 0}
+        from .tools import lore
 
-        def wrapped(_orig=orig, _e=e):
+        lore_dir = lore.next_exec_dir(e.name())
+
+        def wrapped(_orig=orig, _e=e, _lore=lore_dir):
             it = _orig()
             while True:
                 t0 = time.perf_counter()
@@ -54,6 +57,11 @@ def instrument(exec_) -> None:
                 _e.metrics["opTimeMs"] += (time.perf_counter() - t0) * 1e3
                 _e.metrics["numOutputRows"] += batch.num_rows
                 _e.metrics["numOutputBatches"] += 1
+                if _lore is not None:
+                    from .tools import lore as _l
+
+                    _l.dump_batch(_lore, batch, _e.schema,
+                                  _e.metrics["numOutputBatches"] - 1)
                 yield batch
 
         e.execute = wrapped

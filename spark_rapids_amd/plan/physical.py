@@ -151,6 +151,20 @@ def _lower_aggs(aggs: List[AggExpr], in_schema: Schema):
             partial.append(("count", v, INT64))
             merge.extend(["sum", "sum"])
             final.append(("div", js, js + 1, ct))
+        elif a.op in ("stddev", "variance"):
+            from ..expr.expressions import BinaryExpr, CastExpr
+            from ..types import FLOAT64 as F64
+
+            ce = CastExpr(a.child, F64)
+            value_exprs.append(ce)
+            value_exprs.append(BinaryExpr("mul", ce, ce))
+            v = len(value_exprs) - 2
+            js = len(partial)
+            partial.append(("sum", v, FLOAT64))
+            partial.append(("sum", v + 1, FLOAT64))
+            partial.append(("count", v, INT64))
+            merge.extend(["sum", "sum", "sum"])
+            final.append(("var", js, js + 1, js + 2, a.op == "stddev"))
         else:
             raise NotImplementedError(f"agg {a.op}")
     return value_exprs, partial, merge, final
@@ -220,6 +234,20 @@ class HashAggregateExec(PhysicalExec):
                 s = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
                 c = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
                 out_cols.append(ops.binary_op("div", s, c, FLOAT64))
+            elif spec[0] == "var":
+                # sample variance from (sum, sumsq, count):
+                # (sumsq - sum^2/n) / (n-1); NULL when n < 2
+                sm = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
+                sq = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
+                cn = ops.cast(merged.columns[nkeys + spec[3]], FLOAT64)
+                mean_sq = ops.binary_op(
+                    "div", ops.binary_op("mul", sm, sm, FLOAT64), cn, FLOAT64)
+                num = ops.binary_op("sub", sq, mean_sq, FLOAT64)
+                den = ops.binary_op_scalar("sub", cn, 1.0, FLOAT64)
+                var = ops.binary_op("div", num, den, FLOAT64)
+                if spec[4]:
+                    var = ops.unary_op("sqrt", var, FLOAT64)
+                out_cols.append(var)
         yield ColumnBatch(out_cols, merged.num_rows)
 
     def describe(self):

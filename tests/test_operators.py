@@ -109,3 +109,26 @@ def test_string_group_keys_cpu(session):
                                    "v": [1, 2, 3, 4]})
     out = dict(df.group_by("s").agg(sum_(col("v"))).collect())
     assert out == {"a": 4, "b": 2, None: 4}
+
+
+def test_stddev_variance(session):
+    import statistics
+
+    from spark_rapids_amd import stddev, variance
+
+    data = [2.0, 4.0, 4.0, 4.0, 5.0, 5.0, 7.0, 9.0]
+    df = session.create_dataframe({"k": [1] * len(data) , "v": data})
+    out = df.group_by("k").agg(stddev(col("v")), variance(col("v"))).collect()
+    assert out[0][1] == pytest.approx(statistics.stdev(data))
+    assert out[0][2] == pytest.approx(statistics.variance(data))
+
+
+def test_stddev_multi_partition(session):
+    import statistics
+
+    from spark_rapids_amd import stddev
+
+    data = [float(i % 17) for i in range(1000)]
+    df = session.create_dataframe({"v": data}, num_partitions=4)
+    out = df.agg(stddev(col("v"))).collect()
+    assert out[0][0] == pytest.approx(statistics.stdev(data), rel=1e-9)

@@ -88,3 +88,20 @@ def test_metrics_collection(session):
     m = q.metrics()
     assert m and any(x["numOutputRows"] == 89 for x in m)
     assert all("opTimeMs" in x for x in m)
+
+
+def test_lore_dump_and_replay(tmp_path):
+    from spark_rapids_amd import Session, col, sum_
+    import os
+
+    s = Session({"spark.rapids.sql.lore.dumpPath": str(tmp_path)})
+    df = s.create_dataframe({"k": [1, 2, 1], "v": [1.0, 2.0, 3.0]})
+    df.filter(col("v") > 0.5).group_by("k").agg(sum_(col("v"))).collect()
+    dirs = sorted(os.listdir(tmp_path))
+    assert any("Filter" in d for d in dirs), dirs
+    from spark_rapids_amd.tools.lore import replay
+
+    s2 = Session()
+    fdir = next(os.path.join(tmp_path, d) for d in dirs if "Filter" in d)
+    re_df = replay(s2, fdir)
+    assert re_df.count() == 3
