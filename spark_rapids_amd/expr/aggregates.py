@@ -24,12 +24,15 @@ class AggExpr:
     def alias(self, name: str) -> "AggExpr":
         return AggExpr(self.op, self.child, name)
 
+    _DISPLAY = {"count_all": "count", "mean": "avg"}
+
     def output_name(self) -> str:
         if self._name:
             return self._name
+        disp = self._DISPLAY.get(self.op, self.op)
         if self.child is None:
-            return f"{self.op}(*)"
-        return f"{self.op}({self.child})"
+            return f"{disp}(*)"
+        return f"{disp}({self.child})"
 
     def out_dtype(self, schema: Schema) -> DType:
         if self.op in ("count", "count_all"):
