@@ -30,6 +30,7 @@ KERNELS = [
     "kernels/sort.hip",
     "kernels/decode.hip",
     "kernels/strings.hip",
+    "kernels/window.hip",
 ]
 
 CXXFLAGS = ["-O3", "-std=c++20", "-fPIC", f"--offload-arch={ARCH}",

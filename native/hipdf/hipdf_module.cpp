@@ -79,6 +79,10 @@ void hipdf_join_fill(int, const void*, const void*, const void*, int,
                      void*, int64_t, hipStream_t);
 int64_t part_num_blocks(int64_t);
 int64_t sort_num_blocks(int64_t);
+void hipdf_change_flags(const void*, int, void*, int64_t, hipStream_t);
+void hipdf_iota_i32(void*, int64_t, hipStream_t);
+void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
+void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
                              hipStream_t);
 void hipdf_scatter_fixed(int, const void*, const void*, void*, int64_t,
@@ -338,6 +342,25 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
 
+  m.def("change_flags", [](int64_t keys, int nkeys, int64_t out, int64_t n,
+                           int64_t stream) {
+    hipdf_change_flags(P(keys), nkeys, PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("iota_i32", [](int64_t out, int64_t n, int64_t stream) {
+    hipdf_iota_i32(PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("scan_block_f64", [](int64_t in, int64_t out, int64_t sums, int64_t n,
+                             int64_t stream) {
+    hipdf_scan_block_f64(P(in), PM(out), PM(sums), n, S(stream));
+    check_async();
+  });
+  m.def("scan_add_offsets_f64", [](int64_t out, int64_t sums, int64_t n,
+                                   int64_t stream) {
+    hipdf_scan_add_offsets_f64(PM(out), P(sums), n, S(stream));
+    check_async();
+  });
   m.def("rle_hybrid_decode", [](int64_t data, int64_t nbytes, int bw,
                                 int64_t out, int64_t n, int64_t stream) {
     hipdf_rle_hybrid_decode(P(data), nbytes, bw, PM(out), n, S(stream));

@@ -1,0 +1,114 @@
+// Window building blocks (reference analogue: the GpuWindowExec family's
+// device pieces — SURVEY.md §2.4 window variants). The heavy lifting
+// (sorting, scans, per-segment aggregation, gathers) reuses the sort /
+// scan / groupby kernels; this file adds the segment machinery:
+//  - change flags: row differs from the previous row on the given keys
+//  - iota: row index column
+//  - double-typed device-wide scan blocks (running sums over f64)
+#include "hipdf_common.h"
+#include "keys.h"
+
+__global__ void k_change_flags(const KeyCol* __restrict__ keys, int nkeys,
+                               uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    out[i] = (i == 0 || !rows_equal(keys, keys, nkeys, i, i - 1)) ? 1 : 0;
+  }
+}
+
+__global__ void k_iota_i32(int32_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x)
+    out[i] = (int32_t)i;
+}
+
+#define SCAN_ITEMS_W 8
+
+__global__ void k_scan_block_f64(const double* __restrict__ in,
+                                 double* __restrict__ out,
+                                 double* __restrict__ sums, int64_t n) {
+  __shared__ double wave_sums[HIPDF_BLOCK / WAVE];
+  int64_t base = (int64_t)blockIdx.x * HIPDF_BLOCK * SCAN_ITEMS_W;
+  int tid = threadIdx.x;
+  int lane = lane_id();
+  int wid = tid / WAVE;
+  double vals[SCAN_ITEMS_W];
+  double tsum = 0;
+  int64_t tbase = base + (int64_t)tid * SCAN_ITEMS_W;
+#pragma unroll
+  for (int k = 0; k < SCAN_ITEMS_W; ++k) {
+    int64_t i = tbase + k;
+    vals[k] = i < n ? in[i] : 0.0;
+    tsum += vals[k];
+  }
+  double incl = tsum;
+  for (int off = 1; off < WAVE; off <<= 1) {
+    double up = __shfl_up(incl, off);
+    if (lane >= off) incl += up;
+  }
+  if (lane == WAVE - 1) wave_sums[wid] = incl;
+  __syncthreads();
+  if (tid == 0) {
+    double acc = 0;
+    for (int w = 0; w < HIPDF_BLOCK / WAVE; ++w) {
+      double c = wave_sums[w];
+      wave_sums[w] = acc;
+      acc += c;
+    }
+    sums[blockIdx.x] = acc;
+  }
+  __syncthreads();
+  double excl = wave_sums[wid] + incl - tsum;
+#pragma unroll
+  for (int k = 0; k < SCAN_ITEMS_W; ++k) {
+    int64_t i = tbase + k;
+    if (i < n) out[i] = excl;
+    excl += vals[k];
+  }
+}
+
+__global__ void k_scan_add_offsets_f64(double* __restrict__ out,
+                                       const double* __restrict__ sums,
+                                       int64_t n) {
+  double off = sums[blockIdx.x];
+  int64_t base = (int64_t)blockIdx.x * HIPDF_BLOCK * SCAN_ITEMS_W;
+  for (int k = 0; k < SCAN_ITEMS_W; ++k) {
+    int64_t i = base + (int64_t)threadIdx.x + (int64_t)k * HIPDF_BLOCK;
+    if (i < n) out[i] += off;
+  }
+}
+
+extern "C" {
+
+void hipdf_change_flags(const void* keys, int nkeys, void* out, int64_t n,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(k_change_flags, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const KeyCol*)keys, nkeys, (uint8_t*)out, n);
+}
+
+void hipdf_iota_i32(void* out, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_iota_i32, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (int32_t*)out, n);
+}
+
+void hipdf_scan_block_f64(const void* in, void* out, void* sums, int64_t n,
+                          hipStream_t stream) {
+  int64_t per = (int64_t)HIPDF_BLOCK * SCAN_ITEMS_W;
+  int64_t nb = (n + per - 1) / per;
+  if (nb < 1) nb = 1;
+  hipLaunchKernelGGL(k_scan_block_f64, dim3((uint32_t)nb), dim3(HIPDF_BLOCK),
+                     0, stream, (const double*)in, (double*)out, (double*)sums,
+                     n);
+}
+
+void hipdf_scan_add_offsets_f64(void* out, const void* sums, int64_t n,
+                                hipStream_t stream) {
+  int64_t per = (int64_t)HIPDF_BLOCK * SCAN_ITEMS_W;
+  int64_t nb = (n + per - 1) / per;
+  if (nb < 1) nb = 1;
+  hipLaunchKernelGGL(k_scan_add_offsets_f64, dim3((uint32_t)nb),
+                     dim3(HIPDF_BLOCK), 0, stream, (double*)out,
+                     (const double*)sums, n);
+}
+
+}  // extern "C"

@@ -151,8 +151,35 @@ class Tagger:
             if node.how not in ("inner", "left", "semi", "anti"):
                 reasons.append(f"join type {node.how} not on GPU")
         elif isinstance(node, L.Window):
-            reasons.append("window has no GPU kernels yet "
-                           "(segmented scans pending)")
+            spec = node.window_exprs[0].spec
+            for k in spec.partition_by + spec.order_by:
+                r = _FIXED_KEYS.supports(cs.field(k).dtype)
+                if r:
+                    reasons.append(f"window key {k}: {r}")
+            for w in node.window_exprs:
+                op = w.func.op
+                if op in ("row_number", "rank", "dense_rank"):
+                    continue
+                vt = w.func.child.dtype(cs) if w.func.child is not None else None
+                if op in ("lag", "lead"):
+                    if vt is not None and vt.is_nested:
+                        reasons.append(f"lag/lead over {vt} not on GPU")
+                    if w.func.default is not None and vt is not None \
+                            and vt.id is TypeId.STRING:
+                        reasons.append("lag/lead string default not on GPU")
+                    continue
+                if op in ("sum", "count", "mean"):
+                    if vt is not None and not vt.is_numeric:
+                        reasons.append(f"window {op}({vt}) not on GPU")
+                    continue
+                if op in ("min", "max"):
+                    if spec.order_by:
+                        reasons.append(
+                            f"running {op} window has no GPU kernel yet")
+                    elif vt is not None and not vt.is_numeric:
+                        reasons.append(f"window {op}({vt}) not on GPU")
+                    continue
+                reasons.append(f"window function {op} not on GPU")
         elif isinstance(node, L.Sort):
             for k in node.keys:
                 r = _FIXED_KEYS.supports(node.schema().field(k).dtype)

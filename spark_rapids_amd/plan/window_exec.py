@@ -30,6 +30,14 @@ class WindowExec(PhysicalExec):
 
     def execute(self) -> Iterator[ColumnBatch]:
         child = self.children[0]
+        if self.gpu:
+            batches = [b.cuda() for b in child.execute()]
+            if not batches:
+                return
+            table = ops.concat_batches(batches) if len(batches) > 1 \
+                else batches[0]
+            yield self._execute_gpu(table, child.schema)
+            return
         batches = [b.cpu() for b in child.execute()]
         if not batches:
             return
@@ -65,6 +73,290 @@ class WindowExec(PhysicalExec):
     def describe(self):
         names = ", ".join(w.output_name() for w in self.window_exprs)
         return f"{self.name()}[{names}]"
+
+    # ------------------------------------------------------------------
+    # GPU path: segmented computation over the sorted table built from the
+    # radix sort + scan + groupby primitives (window.hip adds change-flags,
+    # iota and f64 scans). See GPU_WINDOW_FUNCS in overrides for the set
+    # that places here.
+    # ------------------------------------------------------------------
+    def _execute_gpu(self, table: ColumnBatch, cs) -> ColumnBatch:
+        import torch
+
+        from ..ops import gpu_backend as gb
+        from ..ops.gpu_backend import ext
+        from ..types import BOOL, FLOAT64, INT32, INT64, DType
+
+        spec = self.spec
+        n = table.num_rows
+        s = gb._stream()
+        key_idx = [cs.index(k) for k in spec.partition_by]
+        order_idx = [cs.index(k) for k in spec.order_by]
+        sort_idx = key_idx + order_idx
+        desc = [False] * len(key_idx) + list(spec.descending)
+        nl = [d for d in desc]
+        if sort_idx and n:
+            order = gb.sort_order(table, sort_idx, desc, nl)
+            table = gb.gather(table, order)
+        if n == 0:
+            cols = list(table.columns)
+            for w in self.window_exprs:
+                cols.append(gb._empty_col(w.out_dtype(cs)))
+            return ColumnBatch(cols, 0)
+
+        def _u8(t):
+            return Column(DType.bool_(), n, t, None, null_count=0)
+
+        def _i32col(t):
+            return Column(INT32, n, t, None, null_count=0)
+
+        # segment heads / order-change flags
+        heads = torch.empty(n, dtype=torch.uint8, device="cuda")
+        if key_idx:
+            kd = gb._key_desc([table.columns[i] for i in key_idx])
+            ext.change_flags(kd.data_ptr(), len(key_idx), heads.data_ptr(),
+                             n, s)
+        else:
+            heads.zero_()
+            heads[0] = 1
+        heads_i64 = torch.empty(n, dtype=torch.int64, device="cuda")
+        ext.cast(0, 4, heads.data_ptr(), heads_i64.data_ptr(), n, s)
+        excl_heads, nseg = gb._exclusive_scan_i64(heads_i64)
+        # seg_id = inclusive_scan(heads) - 1 = excl + head - 1
+        segid_col = gb.binary_op(
+            "add", Column(INT64, n, excl_heads, None, null_count=0),
+            Column(INT64, n, heads_i64, None, null_count=0), INT64)
+        segid_col = gb.binary_op_scalar("sub", segid_col, 1, INT64)
+        seg_id = gb.cast(segid_col, INT32)
+        # head positions + per-row segment start
+        head_pos = self._indices_of(heads, n)
+        seg_start = self._gather_i32(head_pos, seg_id.data, n)
+        iota = torch.empty(n, dtype=torch.int32, device="cuda")
+        ext.iota_i32(iota.data_ptr(), n, s)
+        iota_col = _i32col(iota)
+        seg_start_col = _i32col(seg_start)
+
+        out_cols = list(table.columns)
+        for w in self.window_exprs:
+            out_cols.append(self._compute_gpu_one(
+                w, table, cs, n, heads, seg_id, seg_start_col, iota_col,
+                head_pos, int(nseg)))
+        return ColumnBatch(out_cols, n)
+
+    def _indices_of(self, flags_u8: torch.Tensor, n: int):
+        """positions (i32) of set flags, via the compaction kernels."""
+        import torch
+
+        from ..ops import gpu_backend as gb
+        from ..ops.gpu_backend import ext
+
+        s = gb._stream()
+        nb = ext.sel_num_blocks(n)
+        counts = torch.empty(nb, dtype=torch.int64, device="cuda")
+        ext.mask_count(flags_u8.data_ptr(), 0, counts.data_ptr(), n, s)
+        offsets, total = gb._exclusive_scan_i64(counts)
+        idx = torch.empty(max(total, 1), dtype=torch.int32,
+                          device="cuda")[:total]
+        if total:
+            ext.mask_scatter(flags_u8.data_ptr(), 0, offsets.data_ptr(),
+                             idx.data_ptr(), n, s)
+        return idx
+
+    def _gather_i32(self, src: torch.Tensor, idx: torch.Tensor, n_out: int):
+        import torch
+
+        from ..ops import gpu_backend as gb
+        from ..ops.gpu_backend import ext
+
+        out = torch.empty(n_out, dtype=src.dtype, device="cuda")
+        ext.gather_fixed(src.element_size(), src.data_ptr(), idx.data_ptr(),
+                         out.data_ptr(), n_out, gb._stream())
+        return out
+
+    def _compute_gpu_one(self, w, table, cs, n, heads, seg_id, seg_start_col,
+                         iota_col, head_pos, nseg):
+        import torch
+
+        from ..column import mask_nbytes
+        from ..ops import gpu_backend as gb
+        from ..ops.gpu_backend import ext
+        from ..types import FLOAT64, INT32, INT64, DType
+
+        s = gb._stream()
+        op = w.func.op
+        out_dt = w.out_dtype(cs)
+
+        def _i32col(t):
+            return Column(INT32, n, t, None, null_count=0)
+
+        if op == "row_number":
+            rn = gb.binary_op("sub", iota_col, seg_start_col, INT32)
+            return gb.binary_op_scalar("add", rn, 1, INT32)
+        if op in ("rank", "dense_rank"):
+            all_keys = [table.columns[cs.index(k)] for k in
+                        self.spec.partition_by + self.spec.order_by]
+            och = torch.empty(n, dtype=torch.uint8, device="cuda")
+            kd = gb._key_desc(all_keys)
+            ext.change_flags(kd.data_ptr(), len(all_keys), och.data_ptr(), n, s)
+            och64 = torch.empty(n, dtype=torch.int64, device="cuda")
+            ext.cast(0, 4, och.data_ptr(), och64.data_ptr(), n, s)
+            excl, _ = gb._exclusive_scan_i64(och64)
+            if op == "rank":
+                run_pos = self._indices_of(och, n)
+                runid = gb.binary_op(
+                    "add", Column(INT64, n, excl, None, null_count=0),
+                    Column(INT64, n, och64, None, null_count=0), INT64)
+                runid = gb.cast(gb.binary_op_scalar("sub", runid, 1, INT64),
+                                INT32)
+                run_start = _i32col(self._gather_i32(run_pos, runid.data, n))
+                r = gb.binary_op("sub", run_start, seg_start_col, INT32)
+                return gb.binary_op_scalar("add", r, 1, INT32)
+            # dense_rank: inclusive run count - run count at segment start + 1
+            runid_incl = gb.binary_op(
+                "add", Column(INT64, n, excl, None, null_count=0),
+                Column(INT64, n, och64, None, null_count=0), INT64)
+            ri32 = gb.cast(runid_incl, INT32)
+            at_start = _i32col(self._gather_i32(ri32.data, seg_start_col.data, n))
+            dr = gb.binary_op("sub", ri32, at_start, INT32)
+            return gb.binary_op_scalar("add", dr, 1, INT32)
+
+        # value-based functions
+        vc = w.func.child.eval(table, cs)
+        if op in ("lag", "lead"):
+            k = w.func.offset if op == "lag" else -w.func.offset
+            src = gb.binary_op_scalar("sub", iota_col, k, INT32)
+            # segment end = next head position - 1 (last segment -> n-1)
+            hp_ext = torch.cat([head_pos, torch.tensor(
+                [n], dtype=torch.int32, device="cuda")])
+            segid_next = gb.binary_op_scalar("add", gb.cast(
+                Column(INT32, n, seg_id.data, None, null_count=0), INT32),
+                1, INT32)
+            seg_end = _i32col(self._gather_i32(hp_ext, segid_next.data, n))
+            seg_end = gb.binary_op_scalar("sub", seg_end, 1, INT32)
+            ok = gb.binary_op("and",
+                              gb.binary_op("ge", src, seg_start_col,
+                                           DType.bool_()),
+                              gb.binary_op("le", src, seg_end,
+                                           DType.bool_()), DType.bool_())
+            neg1 = Column.full(-1, INT32, n, "cuda")
+            srcm = gb.if_else(ok, src, neg1)
+            gathered = gb.gather(ColumnBatch([vc], n), srcm).columns[0]
+            if w.func.default is not None:
+                dcol = Column.full(w.func.default, gathered.dtype, n, "cuda")
+                return gb.if_else(ok, gathered, dcol)
+            return gathered
+
+        running = len(self.spec.order_by) > 0
+        valid_u8 = torch.ones(n, dtype=torch.uint8, device="cuda")
+        if vc.validity is not None:
+            ext.mask_expand(vc.validity.data_ptr(), valid_u8.data_ptr(),
+                            False, n, s)
+        nn_col = Column(DType.bool_(), n, valid_u8, None, null_count=0)
+        if running and op in ("sum", "count", "mean"):
+            use_f64 = out_dt.is_floating
+            work_t = FLOAT64 if use_f64 else INT64
+            v64 = gb.cast(Column(vc.dtype, n, vc.data, None, null_count=0),
+                          work_t)
+            vz = gb.binary_op("mul", v64, gb.cast(nn_col, work_t), work_t)
+            run_cnt = self._running_sum_i64(
+                gb.cast(nn_col, INT64).data, heads, seg_start_col, n)
+            if op == "count":
+                return Column(INT64, n, run_cnt, None, null_count=0)
+            if use_f64:
+                run_sum = self._running_sum_f64(vz.data, seg_start_col, n)
+            else:
+                run_sum = self._running_sum_i64(vz.data, heads,
+                                                seg_start_col, n)
+            sum_col = Column(work_t, n, run_sum, None, null_count=None)
+            ov = torch.empty(mask_nbytes(n), dtype=torch.uint8, device="cuda")
+            ext.mask_from_nonzero(run_cnt.data_ptr(), ov.data_ptr(), n, s)
+            sum_col = Column(work_t, n, run_sum, ov, null_count=None)
+            if op == "sum":
+                return gb.cast(sum_col, out_dt) if work_t != out_dt else sum_col
+            cnt_f = gb.cast(Column(INT64, n, run_cnt, None, null_count=0),
+                            FLOAT64)
+            return gb.binary_op("div", gb.cast(sum_col, FLOAT64), cnt_f,
+                                FLOAT64)
+        if not running and op in ("sum", "count", "mean", "min", "max"):
+            # per-segment aggregate via the groupby kernels (seg_id as the
+            # group id), then broadcast back with a gather
+            acc_is_double = out_dt.is_floating or vc.dtype.is_floating
+            acc = torch.empty(max(nseg, 1),
+                              dtype=torch.float64 if acc_is_double
+                              else torch.int64, device="cuda")
+            cnt = torch.zeros(max(nseg, 1), dtype=torch.int64, device="cuda")
+            gop = {"sum": 0, "min": 1, "max": 2, "count": 3, "mean": 0}[op]
+            if op != "count":
+                ext.gb_acc_init(gop, acc.data_ptr(), acc_is_double, nseg, s)
+            ext.gb_agg(gop if op != "mean" else 0, gb._ht(vc.dtype),
+                       vc.data.data_ptr(), gb._ptr(vc.validity),
+                       seg_id.data.data_ptr(), acc.data_ptr(), cnt.data_ptr(),
+                       acc_is_double, nseg, n, s)
+            cnt_rows = self._gather_i32(cnt, seg_id.data, n)
+            if op == "count":
+                return Column(INT64, n, cnt_rows, None, null_count=0)
+            acc_rows = self._gather_i32(acc, seg_id.data, n)
+            work_t = FLOAT64 if acc_is_double else INT64
+            ov = torch.empty(mask_nbytes(n), dtype=torch.uint8, device="cuda")
+            ext.mask_from_nonzero(cnt_rows.data_ptr(), ov.data_ptr(), n, s)
+            val = Column(work_t, n, acc_rows, ov, null_count=None)
+            if op == "mean":
+                cf = gb.cast(Column(INT64, n, cnt_rows, None, null_count=0),
+                             FLOAT64)
+                return gb.binary_op("div", gb.cast(val, FLOAT64), cf, FLOAT64)
+            return gb.cast(val, out_dt) if work_t != out_dt else val
+        raise NotImplementedError(f"gpu window {op}")
+
+    def _running_sum_i64(self, vz: "torch.Tensor", heads, seg_start_col, n):
+        import torch
+
+        from ..ops import gpu_backend as gb
+        from ..types import INT64
+
+        excl, _ = gb._exclusive_scan_i64(vz)
+        ec = Column(INT64, n, excl, None, null_count=0)
+        vcol = Column(INT64, n, vz, None, null_count=0)
+        at_start = Column(INT64, n, self._gather_i32(excl, seg_start_col.data,
+                                                     n), None, null_count=0)
+        incl = gb.binary_op("add", ec, vcol, INT64)
+        return gb.binary_op("sub", incl, at_start, INT64).data
+
+    def _running_sum_f64(self, vz: "torch.Tensor", seg_start_col, n):
+        import torch
+
+        from ..ops import gpu_backend as gb
+        from ..ops.gpu_backend import ext
+        from ..types import FLOAT64
+
+        s = gb._stream()
+        out = torch.empty(n, dtype=torch.float64, device="cuda")
+        per = 256 * 8
+        nb = max((n + per - 1) // per, 1)
+        sums = torch.empty(nb, dtype=torch.float64, device="cuda")
+        ext.scan_block_f64(vz.data_ptr(), out.data_ptr(), sums.data_ptr(),
+                           n, s)
+        if nb > 1:
+            # recursive scan of block sums (small; do on device via same path)
+            sums2 = torch.empty(nb, dtype=torch.float64, device="cuda")
+            partial = torch.empty(max((nb + per - 1) // per, 1),
+                                  dtype=torch.float64, device="cuda")
+            ext.scan_block_f64(sums.data_ptr(), sums2.data_ptr(),
+                               partial.data_ptr(), nb, s)
+            if partial.numel() > 1:
+                p2 = torch.empty_like(partial)
+                p3 = torch.empty(1, dtype=torch.float64, device="cuda")
+                ext.scan_block_f64(partial.data_ptr(), p2.data_ptr(),
+                                   p3.data_ptr(), partial.numel(), s)
+                ext.scan_add_offsets_f64(sums2.data_ptr(), p2.data_ptr(),
+                                         nb, s)
+            ext.scan_add_offsets_f64(out.data_ptr(), sums2.data_ptr(), n, s)
+        ec = Column(FLOAT64, n, out, None, null_count=0)
+        vcol = Column(FLOAT64, n, vz, None, null_count=0)
+        at_start = Column(FLOAT64, n,
+                          self._gather_i32(out, seg_start_col.data, n), None,
+                          null_count=0)
+        incl = gb.binary_op("add", ec, vcol, FLOAT64)
+        return gb.binary_op("sub", incl, at_start, FLOAT64).data
 
 
 def _change_flags(table: ColumnBatch, key_idx: List[int], n: int) -> np.ndarray:

@@ -102,3 +102,76 @@ def test_window_large_random(session):
         assert [r[0] for r in rows] == list(range(1, len(rows) + 1))
         os_ = [r[1] for r in rows]
         assert os_ == sorted(os_)
+
+
+import numpy as np
+import pytest as _pt
+
+import spark_rapids_amd as sr
+
+
+def _win_data(n=30_000, nulls=0.1):
+    rng = np.random.default_rng(11)
+    return {
+        "p": rng.integers(0, 40, n),
+        "o": rng.integers(0, 500, n),
+        "v": np.where(rng.random(n) < nulls, np.nan, rng.uniform(0, 10, n)),
+    }
+
+
+def _df(s, data, nulls=True):
+    import numpy as _np
+
+    valid = ~_np.isnan(data["v"])
+    from spark_rapids_amd import Column, ColumnBatch, Field, Schema
+    from spark_rapids_amd.types import FLOAT64, INT64
+
+    cols = [Column.from_numpy(data["p"].astype(_np.int64)),
+            Column.from_numpy(data["o"].astype(_np.int64)),
+            Column.from_numpy(_np.nan_to_num(data["v"]), FLOAT64,
+                              valid if not valid.all() else None)]
+    schema = Schema([Field("p", INT64), Field("o", INT64),
+                     Field("v", FLOAT64)])
+    return s.from_batches([ColumnBatch(cols)], schema)
+
+
+@_pt.mark.gpu
+@_pt.mark.parametrize("fn", ["row_number", "rank", "dense_rank",
+                             "run_sum", "run_count", "run_avg",
+                             "part_sum", "part_min", "part_max", "part_avg",
+                             "lag", "lead"])
+def test_gpu_window_matches_cpu(fn):
+    data = _win_data()
+    exprs = {
+        "row_number": lambda: row_number().over(["p"], ["o"]),
+        "rank": lambda: rank().over(["p"], ["o"]),
+        "dense_rank": lambda: dense_rank().over(["p"], ["o"]),
+        "run_sum": lambda: win_sum(col("v")).over(["p"], ["o"]),
+        "run_count": lambda: win_count(col("v")).over(["p"], ["o"]),
+        "run_avg": lambda: win_avg(col("v")).over(["p"], ["o"]),
+        "part_sum": lambda: win_sum(col("v")).over(["p"]),
+        "part_min": lambda: win_min(col("v")).over(["p"]),
+        "part_max": lambda: win_max(col("v")).over(["p"]),
+        "part_avg": lambda: win_avg(col("v")).over(["p"]),
+        "lag": lambda: lag(col("v"), 2).over(["p"], ["o"]),
+        "lead": lambda: lead(col("v"), 1, -5.0).over(["p"], ["o"]),
+    }
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    qg = _df(sg, data).with_column("w", exprs[fn]())
+    if fn not in ():
+        tree = qg.physical_plan().tree_string()
+        assert "GpuWindow" in tree, tree
+    gout = qg.to_pydict()["w"]
+    cout = _df(sc, data).with_column("w", exprs[fn]()).to_pydict()["w"]
+    assert len(gout) == len(cout)
+    for i, (g, c) in enumerate(zip(gout, cout)):
+        if c is None or g is None:
+            assert g is None and c is None, (fn, i, g, c)
+        elif isinstance(c, float):
+            if np.isnan(c) or np.isnan(g):
+                assert np.isnan(c) and np.isnan(g), (fn, i, g, c)
+            else:
+                assert g == _pt.approx(c, rel=1e-9), (fn, i)
+        else:
+            assert g == c, (fn, i, g, c)
