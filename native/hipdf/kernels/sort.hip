@@ -15,6 +15,7 @@
 
 #define SORT_BLOCK 256
 #define RADIX_BINS 256
+#define SORT_ITEMS 16  // rows per thread per pass (tile = 4096 rows/block)
 
 // ---- key transform -------------------------------------------------------
 
@@ -98,16 +99,23 @@ __global__ void k_radix_count(const uint64_t* __restrict__ keys, int shift,
   __shared__ int lcnt[RADIX_BINS];
   for (int b = threadIdx.x; b < RADIX_BINS; b += blockDim.x) lcnt[b] = 0;
   __syncthreads();
-  int64_t i = (int64_t)blockIdx.x * SORT_BLOCK + threadIdx.x;
-  if (i < n) {
-    int digit = (int)((keys[i] >> shift) & 255);
-    atomicAdd(&lcnt[digit], 1);
+  int64_t base = (int64_t)blockIdx.x * SORT_BLOCK * SORT_ITEMS;
+#pragma unroll 4
+  for (int it = 0; it < SORT_ITEMS; ++it) {
+    int64_t i = base + (int64_t)it * SORT_BLOCK + threadIdx.x;
+    if (i < n) {
+      int digit = (int)((keys[i] >> shift) & 255);
+      atomicAdd(&lcnt[digit], 1);
+    }
   }
   __syncthreads();
   for (int b = threadIdx.x; b < RADIX_BINS; b += blockDim.x)
     counts[(int64_t)b * nblocks + blockIdx.x] = lcnt[b];
 }
 
+// Stable multi-round scatter: each round handles 256 rows in row order,
+// maintaining a running per-digit offset in LDS across rounds so relative
+// order is preserved within the block (LSD stability requirement).
 __global__ void k_radix_scatter(const uint64_t* __restrict__ keys_in,
                                 const int32_t* __restrict__ perm_in,
                                 int shift,
@@ -116,51 +124,59 @@ __global__ void k_radix_scatter(const uint64_t* __restrict__ keys_in,
                                 uint64_t* __restrict__ keys_out,
                                 int32_t* __restrict__ perm_out, int64_t n) {
   __shared__ int wave_bin[SORT_BLOCK / WAVE][RADIX_BINS];
+  __shared__ int running[RADIX_BINS];
   int tid = threadIdx.x;
   int wid = tid / WAVE;
   int lane = tid & (WAVE - 1);
-  int64_t i = (int64_t)blockIdx.x * SORT_BLOCK + tid;
-  bool active = i < n;
-  uint64_t key = active ? keys_in[i] : 0;
-  int digit = (int)((key >> shift) & 255);
+  for (int b = tid; b < RADIX_BINS; b += blockDim.x) running[b] = 0;
+  int64_t base0 = (int64_t)blockIdx.x * SORT_BLOCK * SORT_ITEMS;
 
-  // wave multi-split: peers = active lanes in this wave with equal digit
-  uint64_t active_mask = __ballot(active);
-  uint64_t peers = active_mask;
-  for (int b = 0; b < 8; ++b) {
-    uint64_t m = __ballot((digit >> b) & 1);
-    peers &= ((digit >> b) & 1) ? m : ~m;
-  }
-  uint64_t lt = lane == 0 ? 0ull : (~0ull >> (64 - lane));
-  int rank_in_wave = __popcll(peers & lt);
+  for (int it = 0; it < SORT_ITEMS; ++it) {
+    int64_t i = base0 + (int64_t)it * SORT_BLOCK + tid;
+    bool active = i < n;
+    uint64_t key = active ? keys_in[i] : 0;
+    int digit = (int)((key >> shift) & 255);
 
-  for (int w = 0; w < SORT_BLOCK / WAVE; ++w)
-    for (int b = tid; b < RADIX_BINS; b += blockDim.x) wave_bin[w][b] = 0;
-  __syncthreads();
-  if (active && rank_in_wave == 0) wave_bin[wid][digit] = __popcll(peers);
-  __syncthreads();
-  // per-bin exclusive scan across the 4 waves
-  for (int b = tid; b < RADIX_BINS; b += blockDim.x) {
-    int acc = 0;
-    for (int w = 0; w < SORT_BLOCK / WAVE; ++w) {
-      int c = wave_bin[w][b];
-      wave_bin[w][b] = acc;
-      acc += c;
+    uint64_t active_mask = __ballot(active);
+    uint64_t peers = active_mask;
+    for (int b = 0; b < 8; ++b) {
+      uint64_t m = __ballot((digit >> b) & 1);
+      peers &= ((digit >> b) & 1) ? m : ~m;
     }
-  }
-  __syncthreads();
-  if (active) {
-    int64_t base = offsets[(int64_t)digit * nblocks + blockIdx.x];
-    int64_t pos = base + wave_bin[wid][digit] + rank_in_wave;
-    keys_out[pos] = key;
-    perm_out[pos] = perm_in ? perm_in[i] : (int32_t)i;
+    uint64_t lt = lane == 0 ? 0ull : (~0ull >> (64 - lane));
+    int rank_in_wave = __popcll(peers & lt);
+
+    for (int w = 0; w < SORT_BLOCK / WAVE; ++w)
+      for (int b = tid; b < RADIX_BINS; b += blockDim.x) wave_bin[w][b] = 0;
+    __syncthreads();
+    if (active && rank_in_wave == 0) wave_bin[wid][digit] = __popcll(peers);
+    __syncthreads();
+    // per-bin exclusive scan across the 4 waves, on top of `running`
+    for (int b = tid; b < RADIX_BINS; b += blockDim.x) {
+      int acc = running[b];
+      for (int w = 0; w < SORT_BLOCK / WAVE; ++w) {
+        int c = wave_bin[w][b];
+        wave_bin[w][b] = acc;
+        acc += c;
+      }
+      running[b] = acc;
+    }
+    __syncthreads();
+    if (active) {
+      int64_t base = offsets[(int64_t)digit * nblocks + blockIdx.x];
+      int64_t pos = base + wave_bin[wid][digit] + rank_in_wave;
+      keys_out[pos] = key;
+      perm_out[pos] = perm_in ? perm_in[i] : (int32_t)i;
+    }
+    __syncthreads();
   }
 }
 
 extern "C" {
 
 int64_t sort_num_blocks(int64_t n) {
-  int64_t nb = (n + SORT_BLOCK - 1) / SORT_BLOCK;
+  int64_t per = (int64_t)SORT_BLOCK * SORT_ITEMS;
+  int64_t nb = (n + per - 1) / per;
   return nb < 1 ? 1 : nb;
 }
 
