@@ -51,6 +51,39 @@ class PhysicalExec:
         return "\n".join(lines)
 
 
+def prefetched(it: Iterator, depth: int = 2) -> Iterator:
+    """Run an iterator on a background thread with a bounded queue.
+
+    The device syncs inside a batch's processing (allocation-sizing .item()
+    reads) release the GIL, so the producer thread enqueues the next
+    partition's kernels while the consumer waits — the pipeline-overlap
+    analogue of the reference's semaphore-bounded concurrent tasks
+    (GpuSemaphore + async coalesce iterators)."""
+    import queue
+    import threading
+
+    q: "queue.Queue" = queue.Queue(maxsize=depth)
+    _END = object()
+
+    def producer():
+        try:
+            for item in it:
+                q.put(item)
+            q.put(_END)
+        except BaseException as e:  # noqa: BLE001 - forwarded to consumer
+            q.put(e)
+
+    t = threading.Thread(target=producer, daemon=True)
+    t.start()
+    while True:
+        item = q.get()
+        if item is _END:
+            return
+        if isinstance(item, BaseException):
+            raise item
+        yield item
+
+
 class ScanExec(PhysicalExec):
     def __init__(self, device: str, schema: Schema, source, label: str):
         super().__init__(device, schema)
@@ -189,7 +222,9 @@ class HashAggregateExec(PhysicalExec):
         value_exprs, partial, merge_ops, final = _lower_aggs(self.aggs, in_schema)
 
         partial_results: List[ColumnBatch] = []
-        for batch in self.children[0].execute():
+        source = prefetched(self.children[0].execute()) if self.gpu \
+            else self.children[0].execute()
+        for batch in source:
             def task(b):
                 key_cols = [e.eval(b, in_schema) for e in self.group_exprs]
                 val_cols = [e.eval(b, in_schema) for e in value_exprs]
@@ -296,7 +331,8 @@ class HashJoinExec(PhysicalExec):
             rtable = ops.concat_batches(rbatches) if len(rbatches) > 1 else rbatches[0]
         lkidx = [left.schema.index(k) for k in self.left_on]
         rkidx = [right.schema.index(k) for k in self.right_on]
-        for lbatch in left.execute():
+        lsource = prefetched(left.execute()) if self.gpu else left.execute()
+        for lbatch in lsource:
             if lbatch.num_rows == 0:
                 continue
             if rtable is None or rtable.num_rows == 0:
