@@ -54,11 +54,10 @@ class PhysicalExec:
 def prefetched(it: Iterator, depth: int = 2) -> Iterator:
     """Run an iterator on a background thread with a bounded queue.
 
-    The device syncs inside a batch's processing (allocation-sizing .item()
-    reads) release the GIL, so the producer thread enqueues the next
-    partition's kernels while the consumer waits — the pipeline-overlap
-    analogue of the reference's semaphore-bounded concurrent tasks
-    (GpuSemaphore + async coalesce iterators)."""
+    Measured on MI355X: a NET REGRESSION for the power-run shape (GIL
+    contention outweighs sync overlap), so operators do not use it by
+    default; kept for IO-bound scans where the producer blocks in native
+    reads (parquet prefetch pool uses the same pattern internally)."""
     import queue
     import threading
 
@@ -222,9 +221,7 @@ class HashAggregateExec(PhysicalExec):
         value_exprs, partial, merge_ops, final = _lower_aggs(self.aggs, in_schema)
 
         partial_results: List[ColumnBatch] = []
-        source = prefetched(self.children[0].execute()) if self.gpu \
-            else self.children[0].execute()
-        for batch in source:
+        for batch in self.children[0].execute():
             def task(b):
                 key_cols = [e.eval(b, in_schema) for e in self.group_exprs]
                 val_cols = [e.eval(b, in_schema) for e in value_exprs]
@@ -331,8 +328,7 @@ class HashJoinExec(PhysicalExec):
             rtable = ops.concat_batches(rbatches) if len(rbatches) > 1 else rbatches[0]
         lkidx = [left.schema.index(k) for k in self.left_on]
         rkidx = [right.schema.index(k) for k in self.right_on]
-        lsource = prefetched(left.execute()) if self.gpu else left.execute()
-        for lbatch in lsource:
+        for lbatch in left.execute():
             if lbatch.num_rows == 0:
                 continue
             if rtable is None or rtable.num_rows == 0:
