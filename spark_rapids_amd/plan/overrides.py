@@ -238,7 +238,12 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
     device = "cuda" if on_gpu else "cpu"
 
     if isinstance(node, L.Scan):
-        return P.ScanExec(device, node.schema(), node.source, node.label)
+        scan = P.ScanExec(device, node.schema(), node.source, node.label)
+        if on_gpu:
+            from ..config import BATCH_SIZE_BYTES
+
+            return P.CoalesceBatchesExec(scan, conf.get(BATCH_SIZE_BYTES))
+        return scan
 
     kids = [_ensure_device(_convert(c, conf, tagger, gpu_wanted), device)
             for c in node.children]

@@ -112,6 +112,40 @@ class DeviceTransferExec(PhysicalExec):
         return f"DeviceTransfer(to={self.device})"
 
 
+class CoalesceBatchesExec(PhysicalExec):
+    """Concatenate small batches up to the spark.rapids.sql.batchSizeBytes
+    goal (reference: GpuCoalesceBatches / TargetSize goal,
+    GpuCoalesceBatches.scala). 288 GB of HBM3E favors big batches: fewer,
+    larger kernels and fewer allocation-sizing syncs."""
+
+    def __init__(self, child: PhysicalExec, target_bytes: int):
+        super().__init__(child.device, child.schema, [child])
+        self.target_bytes = target_bytes
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        pending: List[ColumnBatch] = []
+        pending_bytes = 0
+        for batch in self.children[0].execute():
+            if batch.num_rows == 0:
+                continue
+            nb = batch.nbytes
+            if pending and pending_bytes + nb > self.target_bytes:
+                yield self._flush(pending)
+                pending, pending_bytes = [], 0
+            pending.append(batch)
+            pending_bytes += nb
+        if pending:
+            yield self._flush(pending)
+
+    def _flush(self, pending: List[ColumnBatch]) -> ColumnBatch:
+        if len(pending) == 1:
+            return pending[0]
+        return ops.concat_batches(pending)
+
+    def describe(self):
+        return f"{self.name()}(target={self.target_bytes})"
+
+
 class FilterExec(PhysicalExec):
     def __init__(self, device: str, condition: Expression, child: PhysicalExec):
         super().__init__(device, child.schema, [child])

@@ -132,3 +132,27 @@ def test_stddev_multi_partition(session):
     df = session.create_dataframe({"v": data}, num_partitions=4)
     out = df.agg(stddev(col("v"))).collect()
     assert out[0][0] == pytest.approx(statistics.stdev(data), rel=1e-9)
+
+
+def test_coalesce_batches_exec():
+    from spark_rapids_amd import Column, ColumnBatch, INT64
+    from spark_rapids_amd.column import Field, Schema
+    from spark_rapids_amd.plan.physical import CoalesceBatchesExec, ScanExec
+
+    batches = [ColumnBatch([Column.from_pylist(list(range(i * 10, i * 10 + 10)),
+                                               INT64)]) for i in range(5)]
+
+    class Src:
+        def partitions(self):
+            return iter(batches)
+
+    schema = Schema([Field("x", INT64)])
+    scan = ScanExec("cpu", schema, Src(), "t")
+    # tiny target: batches pass through one-by-one
+    small = list(CoalesceBatchesExec(scan, 8).execute())
+    assert len(small) == 5
+    # big target: everything coalesces to one batch
+    scan2 = ScanExec("cpu", schema, Src(), "t")
+    big = list(CoalesceBatchesExec(scan2, 1 << 30).execute())
+    assert len(big) == 1 and big[0].num_rows == 50
+    assert big[0].columns[0].to_pylist() == list(range(50))
