@@ -29,13 +29,20 @@ __global__ void k_gb_build(const int32_t* __restrict__ hashes,
        i += (int64_t)gridDim.x * blockDim.x) {
     uint32_t slot = slot_of((uint32_t)hashes[i], slot_mask);
     while (true) {
-      int32_t cur = atomicCAS(&slot_row[slot], GB_EMPTY, (int32_t)i);
+      // plain read first: slots are write-once (EMPTY -> row), so a stale
+      // EMPTY just falls through to the CAS; this keeps the hot
+      // low-cardinality case (every row hitting a few claimed slots) on
+      // cached loads instead of serializing atomics across 256 CUs
+      int32_t cur = slot_row[slot];
       if (cur == GB_EMPTY) {
-        // claimed a new group: record its slot so numbering touches only
-        // live slots instead of sweeping the whole table
-        claimed_slots[atomicAdd(ngroups, 1)] = (int32_t)slot;
-        row_slot[i] = (int32_t)slot;
-        break;
+        cur = atomicCAS(&slot_row[slot], GB_EMPTY, (int32_t)i);
+        if (cur == GB_EMPTY) {
+          // claimed a new group: record its slot so numbering touches only
+          // live slots instead of sweeping the whole table
+          claimed_slots[atomicAdd(ngroups, 1)] = (int32_t)slot;
+          row_slot[i] = (int32_t)slot;
+          break;
+        }
       }
       if (cur == (int32_t)i || rows_equal(keys, keys, nkeys, i, cur)) {
         row_slot[i] = (int32_t)slot;
