@@ -164,6 +164,10 @@ GPU_OOM_INJECTION = int_conf(
 STABLE_SORT = bool_conf(
     "spark.rapids.sql.stableSort.enabled", False,
     "Use a stable sort on GPU (matches CPU tie ordering; slightly slower).")
+PRUNE_COLUMNS = bool_conf(
+    "spark.rapids.sql.optimizer.pruneColumns.enabled", True,
+    "Push projections below joins/aggregates so unused columns are never "
+    "gathered or transferred (Catalyst-optimizer analogue).")
 LORE_DUMP_PATH = str_conf(
     "spark.rapids.sql.lore.dumpPath", "",
     "When set, dump every operator's output batches to this directory as "

@@ -194,6 +194,11 @@ def plan_physical(node: L.LogicalPlan, conf: RapidsConf,
     top = tagger is None
     if tagger is None:
         tagger = Tagger(conf)
+        from ..config import PRUNE_COLUMNS
+        if conf.get(PRUNE_COLUMNS):
+            from .optimizer import prune_columns
+
+            node = prune_columns(node)
     gpu_wanted = conf.sql_enabled and _gpu_available()
     exec_ = _convert(node, conf, tagger, gpu_wanted)
     if top:

@@ -105,3 +105,27 @@ def test_lore_dump_and_replay(tmp_path):
     fdir = next(os.path.join(tmp_path, d) for d in dirs if "Filter" in d)
     re_df = replay(s2, fdir)
     assert re_df.count() == 3
+
+
+def test_column_pruning_under_join(session):
+    from spark_rapids_amd.plan.optimizer import prune_columns
+
+    fact = session.create_dataframe(
+        {"k": [1, 2], "a": [1, 2], "b": [3, 4], "c": [5, 6]})
+    dim = session.create_dataframe({"k": [1, 2], "x": [7, 8], "y": [9, 10]})
+    q = (fact.join(dim, on="k").group_by("x")
+         .agg(sum_(col("a"))))
+    pruned = prune_columns(q.plan)
+    # the join children should be narrowed to {k,a} and {k,x}
+    join = pruned.child if hasattr(pruned, "child") else pruned
+    from spark_rapids_amd.plan import logical as L
+
+    node = pruned
+    while not isinstance(node, L.Join):
+        node = node.children[0]
+    assert set(node.left.schema().names) == {"k", "a"}
+    assert set(node.right.schema().names) == {"k", "x"}
+    # results identical with and without pruning
+    res = q.collect()
+    session.conf.set("spark.rapids.sql.optimizer.pruneColumns.enabled", False)
+    assert sorted(res) == sorted(q.collect())
