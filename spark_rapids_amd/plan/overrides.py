@@ -258,9 +258,12 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
     if isinstance(node, L.Project):
         return P.ProjectExec(device, node.exprs, kids[0], node.schema())
     if isinstance(node, L.Aggregate):
+        from ..config import BATCH_SIZE_BYTES
+
         return P.HashAggregateExec(device, node.group_exprs, node.aggs,
                                    kids[0], node.schema(),
-                                   input_replicated=L.is_replicated(node.child))
+                                   input_replicated=L.is_replicated(node.child),
+                                   merge_target_bytes=conf.get(BATCH_SIZE_BYTES))
     if isinstance(node, L.Join):
         return P.HashJoinExec(device, kids[0], kids[1], node.left_on,
                               node.right_on, node.how, node.schema(),

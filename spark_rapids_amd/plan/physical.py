@@ -243,11 +243,13 @@ class HashAggregateExec(PhysicalExec):
 
     def __init__(self, device: str, group_exprs: List[Expression],
                  aggs: List[AggExpr], child: PhysicalExec, schema: Schema,
-                 input_replicated: bool = False):
+                 input_replicated: bool = False,
+                 merge_target_bytes: int = 2 << 30):
         super().__init__(device, schema, [child])
         self.group_exprs = group_exprs
         self.aggs = aggs
         self.input_replicated = input_replicated
+        self.merge_target_bytes = merge_target_bytes
 
     def execute(self) -> Iterator[ColumnBatch]:
         in_schema = self.children[0].schema
@@ -287,34 +289,56 @@ class HashAggregateExec(PhysicalExec):
                 else received[0]
         merge_specs = [(op, nkeys + j, partial[j][2])
                        for j, op in enumerate(merge_ops)]
-        merged = ops.group_by_aggregate(merged_in, list(range(nkeys)), merge_specs)
 
-        # final projection
-        out_cols: List[Column] = [merged.columns[i] for i in range(nkeys)]
+        # repartition-based fallback (reference: GpuMergeAggregateIterator's
+        # recursive bucket split, GpuAggregateExec.scala:205-306): when the
+        # merged partials exceed the batch-size goal, hash-split them into
+        # buckets and merge bucket-by-bucket so no single merge pass needs
+        # the whole key space in memory.
+        buckets = [merged_in]
+        if nkeys and merged_in.nbytes > self._merge_target_bytes():
+            nb = max(2, (merged_in.nbytes +
+                         self._merge_target_bytes() - 1) //
+                     self._merge_target_bytes())
+            parted, offs = ops.hash_partition(merged_in, list(range(nkeys)),
+                                              int(nb))
+            buckets = [_slice_rows(parted, offs[i], offs[i + 1])
+                       for i in range(int(nb))]
+
         cs = in_schema
-        for spec, agg in zip(final, self.aggs):
-            if spec[0] == "col":
-                c = merged.columns[nkeys + spec[1]]
-                out_cols.append(ops.cast(c, agg.out_dtype(cs)))
-            elif spec[0] == "div":
-                s = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
-                c = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
-                out_cols.append(ops.binary_op("div", s, c, FLOAT64))
-            elif spec[0] == "var":
-                # sample variance from (sum, sumsq, count):
-                # (sumsq - sum^2/n) / (n-1); NULL when n < 2
-                sm = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
-                sq = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
-                cn = ops.cast(merged.columns[nkeys + spec[3]], FLOAT64)
-                mean_sq = ops.binary_op(
-                    "div", ops.binary_op("mul", sm, sm, FLOAT64), cn, FLOAT64)
-                num = ops.binary_op("sub", sq, mean_sq, FLOAT64)
-                den = ops.binary_op_scalar("sub", cn, 1.0, FLOAT64)
-                var = ops.binary_op("div", num, den, FLOAT64)
-                if spec[4]:
-                    var = ops.unary_op("sqrt", var, FLOAT64)
-                out_cols.append(var)
-        yield ColumnBatch(out_cols, merged.num_rows)
+        for bucket in buckets:
+            if bucket.num_rows == 0:
+                continue
+            merged = ops.group_by_aggregate(bucket, list(range(nkeys)),
+                                            merge_specs)
+            out_cols: List[Column] = [merged.columns[i] for i in range(nkeys)]
+            for spec, agg in zip(final, self.aggs):
+                if spec[0] == "col":
+                    c = merged.columns[nkeys + spec[1]]
+                    out_cols.append(ops.cast(c, agg.out_dtype(cs)))
+                elif spec[0] == "div":
+                    s = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
+                    c = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
+                    out_cols.append(ops.binary_op("div", s, c, FLOAT64))
+                elif spec[0] == "var":
+                    # sample variance from (sum, sumsq, count):
+                    # (sumsq - sum^2/n) / (n-1); NULL when n < 2
+                    sm = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
+                    sq = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
+                    cn = ops.cast(merged.columns[nkeys + spec[3]], FLOAT64)
+                    mean_sq = ops.binary_op(
+                        "div", ops.binary_op("mul", sm, sm, FLOAT64), cn,
+                        FLOAT64)
+                    num = ops.binary_op("sub", sq, mean_sq, FLOAT64)
+                    den = ops.binary_op_scalar("sub", cn, 1.0, FLOAT64)
+                    var = ops.binary_op("div", num, den, FLOAT64)
+                    if spec[4]:
+                        var = ops.unary_op("sqrt", var, FLOAT64)
+                    out_cols.append(var)
+            yield ColumnBatch(out_cols, merged.num_rows)
+
+    def _merge_target_bytes(self) -> int:
+        return self.merge_target_bytes
 
     def describe(self):
         keys = ", ".join(str(e) for e in self.group_exprs)
@@ -453,6 +477,14 @@ class UnionExec(PhysicalExec):
     def execute(self) -> Iterator[ColumnBatch]:
         for c in self.children:
             yield from c.execute()
+
+
+def _slice_rows(batch: ColumnBatch, start: int, end: int) -> ColumnBatch:
+    import numpy as np
+
+    idx = Column.from_numpy(np.arange(start, end, dtype=np.int32),
+                            device=batch.device)
+    return ops.gather(batch, idx)
 
 
 def with_retry_split_single(task, batch):

@@ -156,3 +156,19 @@ def test_coalesce_batches_exec():
     big = list(CoalesceBatchesExec(scan2, 1 << 30).execute())
     assert len(big) == 1 and big[0].num_rows == 50
     assert big[0].columns[0].to_pylist() == list(range(50))
+
+
+def test_repartition_merge_fallback(session):
+    # force tiny merge target so the bucket-split path runs
+    session.conf.set("spark.rapids.sql.batchSizeBytes", 4096)
+    import numpy as np
+
+    n = 20_000
+    rng = np.random.default_rng(3)
+    df = session.create_dataframe({
+        "k": rng.integers(0, 5000, n),
+        "v": rng.uniform(0, 1, n),
+    }, num_partitions=6)
+    out = df.group_by("k").agg(sum_(col("v")), count_star()).collect()
+    assert sum(r[2] for r in out) == n
+    assert len(out) == len(set(r[0] for r in out))  # keys unique across buckets
