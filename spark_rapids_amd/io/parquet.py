@@ -162,13 +162,23 @@ class ParquetTable:
         tbl = pq.read_table(path, columns=self.columns)
         return arrow_table_to_batch(tbl)
 
+    def _my_files(self) -> List[str]:
+        # distributed scans shard files round-robin across ranks
+        from ..shuffle import dist
+
+        c = dist.ctx()
+        if c.is_multi and len(self.files) > 1:
+            return self.files[c.rank::c.world]
+        return self.files
+
     def partitions(self) -> Iterable[ColumnBatch]:
-        if len(self.files) == 1 or self.prefetch_threads <= 1:
-            for f in self.files:
+        files = self._my_files()
+        if len(files) <= 1 or self.prefetch_threads <= 1:
+            for f in files:
                 yield self._read_one(f)
             return
         with cf.ThreadPoolExecutor(self.prefetch_threads) as pool:
-            futures = [pool.submit(self._read_one, f) for f in self.files]
+            futures = [pool.submit(self._read_one, f) for f in files]
             for fut in futures:
                 yield fut.result()
 
