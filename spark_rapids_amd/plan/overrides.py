@@ -26,12 +26,18 @@ from . import physical as P
 # ---------------------------------------------------------------------------
 
 _BASIC = TypeSig.all_basic()
-# GPU kernel support for group/join/sort keys: fixed-width only this round;
-# string keys fall back to CPU with a recorded reason.
+# sort keys need the radix key transform: fixed-width only this round
 _FIXED_KEYS = TypeSig({
     TypeId.BOOL, TypeId.INT8, TypeId.INT16, TypeId.INT32, TypeId.INT64,
     TypeId.FLOAT32, TypeId.FLOAT64, TypeId.DECIMAL64, TypeId.DATE32,
     TypeId.TIMESTAMP,
+})
+# group-by / join keys go through murmur3 row hash + KeyCol row equality,
+# which handle strings too (hash.hip murmur3_str + keys.h byte compare)
+_HASH_KEYS = TypeSig({
+    TypeId.BOOL, TypeId.INT8, TypeId.INT16, TypeId.INT32, TypeId.INT64,
+    TypeId.FLOAT32, TypeId.FLOAT64, TypeId.DECIMAL64, TypeId.DATE32,
+    TypeId.TIMESTAMP, TypeId.STRING,
 })
 _NUMERIC = TypeSig.numeric()
 
@@ -127,7 +133,7 @@ class Tagger:
                 reasons += self.expr_reasons(e, cs)
         elif isinstance(node, L.Aggregate):
             for e in node.group_exprs:
-                r = _FIXED_KEYS.supports(e.dtype(cs))
+                r = _HASH_KEYS.supports(e.dtype(cs))
                 if r:
                     reasons.append(f"group key {e}: {r}")
                 reasons += self.expr_reasons(e, cs)
@@ -141,11 +147,11 @@ class Tagger:
         elif isinstance(node, L.Join):
             ls, rs = node.left.schema(), node.right.schema()
             for k in node.left_on:
-                r = _FIXED_KEYS.supports(ls.field(k).dtype)
+                r = _HASH_KEYS.supports(ls.field(k).dtype)
                 if r:
                     reasons.append(f"join key {k}: {r}")
             for k in node.right_on:
-                r = _FIXED_KEYS.supports(rs.field(k).dtype)
+                r = _HASH_KEYS.supports(rs.field(k).dtype)
                 if r:
                     reasons.append(f"join key {k}: {r}")
             if node.how not in ("inner", "left", "semi", "anti"):

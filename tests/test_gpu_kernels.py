@@ -387,3 +387,70 @@ def test_sort_exec_gpu_e2e():
     # data differs (rng advanced); just validate GPU output is sorted
     av = [r[0] for r in gpu]
     assert av == sorted(av)
+
+
+def test_group_by_string_keys_gpu():
+    words = ["alpha", "beta", "gamma", None, "", "alpha2"]
+    n = 30_000
+    keys = Column.from_pylist([words[i % 6] for i in range(n)], STRING)
+    vals = _rand_col(FLOAT64, n=n)
+    batch = ColumnBatch([keys, vals])
+    aggs = [("sum", 1, FLOAT64), ("count_all", -1, INT64)]
+    from spark_rapids_amd.ops import gpu_backend
+    cpu = cpu_backend.group_by_aggregate(batch, [0], aggs)
+    gpu = gpu_backend.group_by_aggregate(batch.cuda(), [0], aggs).cpu()
+    assert cpu.num_rows == gpu.num_rows == 6
+    crows = _sorted_rows(cpu)
+    grows = _sorted_rows(gpu)
+    for cr, gr in zip(crows, grows):
+        assert cr[0] == gr[0] and cr[2] == gr[2]
+        if cr[1] is not None:
+            assert gr[1] == pytest.approx(cr[1], rel=1e-9)
+
+
+def test_join_string_keys_gpu():
+    lwords = ["a", "bb", None, "ccc", "bb", ""]
+    rwords = ["bb", "ccc", "", "zz", None]
+    left = ColumnBatch([Column.from_pylist(lwords * 500, STRING),
+                        _rand_col(INT64, n=3000, nulls=0.0)])
+    right = ColumnBatch([Column.from_pylist(rwords * 100, STRING),
+                         _rand_col(INT32, n=500, nulls=0.0)])
+    from spark_rapids_amd.ops import cpu_backend as cb, gpu_backend as gb
+    lc, rc = cb.join_gather_maps(left, right, [0], [0], "inner")
+    lg, rg = gb.join_gather_maps(left.cuda(), right.cuda(), [0], [0], "inner")
+    cpu_out = ColumnBatch(cb.gather(left, lc).columns +
+                          cb.gather(right, rc).columns)
+    gpu_out = ColumnBatch(gb.gather(left.cuda(), lg).columns +
+                          gb.gather(right.cuda(), rg).columns).cpu()
+    assert _sorted_rows(cpu_out) == _sorted_rows(gpu_out)
+
+
+def test_spill_device_to_disk_roundtrip_gpu(tmp_path, monkeypatch):
+    monkeypatch.setenv("RAPIDS_SPILL_PATH", str(tmp_path))
+    from spark_rapids_amd.memory.spill import SpillableBatch
+
+    b = ColumnBatch([_rand_col(INT64, n=1000), _rand_col(STRING, n=1000)]).cuda()
+    ref = b.cpu()
+    h = SpillableBatch(b)
+    assert h.state == "device"
+    assert h.spill_to_host() > 0
+    assert h.state == "host"
+    assert h.spill_to_disk() > 0
+    got = h.get()  # resurrect back onto the device
+    assert got.is_cuda
+    for c, g in zip(ref.columns, got.cpu().columns):
+        assert c.to_pylist() == g.to_pylist()
+    h.close()
+
+
+def test_enabled_assert_mode_gpu():
+    s = sr.Session({"spark.rapids.sql.test.enabled": True})
+    df = s.create_dataframe({"a": [1.0, 2.0]})
+    # fully-GPU plan passes
+    assert df.filter(sr.col("a") > 1.0).count() == 1
+    # a CPU-fallback op (disabled exec) must raise
+    s2 = sr.Session({"spark.rapids.sql.test.enabled": True,
+                     "spark.rapids.sql.exec.Filter": False})
+    df2 = s2.create_dataframe({"a": [1.0, 2.0]})
+    with pytest.raises(AssertionError):
+        df2.filter(sr.col("a") > 1.0).count()

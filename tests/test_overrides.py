@@ -35,12 +35,14 @@ def test_tagger_accepts_numeric_plan(session):
     assert t.exec_reasons(plan) == []
 
 
-def test_tagger_rejects_string_group_key(session):
+def test_tagger_accepts_string_group_key_rejects_string_sort(session):
     df = session.create_dataframe({"s": ["x"], "v": [1]})
     plan = L.Aggregate([col("s")], [sum_(col("v"))], _logical(df))
     t = Tagger(session.conf)
-    reasons = t.exec_reasons(plan)
-    assert reasons and "string" in reasons[0]
+    assert t.exec_reasons(plan) == []  # hash keys handle strings
+    sort_plan = L.Sort(_logical(df), ["s"])
+    reasons = Tagger(session.conf).exec_reasons(sort_plan)
+    assert reasons and "string" in reasons[0]  # radix keys do not (yet)
 
 
 def test_per_exec_disable_conf(session):
