@@ -53,20 +53,20 @@ void hipdf_scan_block(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets(void*, const void*, int64_t, hipStream_t);
 void hipdf_reduce(int, int, const void*, const void*, void*, void*, int64_t,
                   hipStream_t);
-void hipdf_murmur3_col(int, int, const void*, const void*, void*, int64_t,
-                       hipStream_t);
-void hipdf_murmur3_str(const void*, const void*, const void*, void*, int64_t,
-                       hipStream_t);
+void hipdf_murmur3_col(int, int, const void*, const void*, const void*,
+                       void*, int64_t, hipStream_t);
+void hipdf_murmur3_str(const void*, const void*, const void*, const void*,
+                       void*, int64_t, hipStream_t);
 void hipdf_pmod_part(const void*, int, void*, int64_t, hipStream_t);
-void hipdf_gb_build(const void*, const void*, int, void*, void*, void*,
-                    void*, int64_t, int64_t, hipStream_t);
+void hipdf_gb_build(const void*, const void*, int, const void*, void*,
+                    void*, void*, void*, int64_t, int64_t, hipStream_t);
 void hipdf_gb_number(const void*, const void*, void*, const void*, void*,
                      int64_t, hipStream_t);
 void hipdf_gb_rowgid(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_gb_agg(int, int, const void*, const void*, const void*, void*,
                   void*, int, int32_t, int64_t, hipStream_t);
-void hipdf_gb_agg_multi(const void*, int, const void*, int32_t, int64_t,
-                        hipStream_t);
+void hipdf_gb_agg_multi(const void*, int, const void*, const void*,
+                        int32_t, int64_t, hipStream_t);
 void hipdf_gb_acc_init(int, void*, int, int32_t, hipStream_t);
 void hipdf_mask_from_nonzero(const void*, void*, int64_t, hipStream_t);
 void hipdf_join_build(const void*, const void*, int, void*, void*, int64_t,
@@ -260,13 +260,16 @@ PYBIND11_MODULE(hipdf, m) {
   });
 
   m.def("murmur3_col", [](int kind, int t, int64_t a, int64_t av,
-                          int64_t seeds, int64_t n, int64_t stream) {
-    hipdf_murmur3_col(kind, t, P(a), P(av), PM(seeds), n, S(stream));
+                          int64_t sel, int64_t seeds, int64_t n,
+                          int64_t stream) {
+    hipdf_murmur3_col(kind, t, P(a), P(av), P(sel), PM(seeds), n, S(stream));
     check_async();
   });
   m.def("murmur3_str", [](int64_t offs, int64_t bytes, int64_t av,
-                          int64_t seeds, int64_t n, int64_t stream) {
-    hipdf_murmur3_str(P(offs), P(bytes), P(av), PM(seeds), n, S(stream));
+                          int64_t sel, int64_t seeds, int64_t n,
+                          int64_t stream) {
+    hipdf_murmur3_str(P(offs), P(bytes), P(av), P(sel), PM(seeds), n,
+                      S(stream));
     check_async();
   });
   m.def("pmod_part", [](int64_t h, int nparts, int64_t part, int64_t n,
@@ -275,12 +278,13 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
 
-  m.def("gb_build", [](int64_t hashes, int64_t keys, int nkeys,
+  m.def("gb_build", [](int64_t hashes, int64_t keys, int nkeys, int64_t sel,
                        int64_t slot_row, int64_t row_slot,
                        int64_t claimed_slots, int64_t ngroups, int64_t cap,
                        int64_t n, int64_t stream) {
-    hipdf_gb_build(P(hashes), P(keys), nkeys, PM(slot_row), PM(row_slot),
-                   PM(claimed_slots), PM(ngroups), cap, n, S(stream));
+    hipdf_gb_build(P(hashes), P(keys), nkeys, P(sel), PM(slot_row),
+                   PM(row_slot), PM(claimed_slots), PM(ngroups), cap, n,
+                   S(stream));
     check_async();
   });
   m.def("gb_number", [](int64_t claimed_slots, int64_t slot_row,
@@ -309,8 +313,10 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
   m.def("gb_agg_multi", [](int64_t aggs, int naggs, int64_t row_gid,
-                           int ngroups, int64_t n, int64_t stream) {
-    hipdf_gb_agg_multi(P(aggs), naggs, P(row_gid), ngroups, n, S(stream));
+                           int64_t sel, int ngroups, int64_t n,
+                           int64_t stream) {
+    hipdf_gb_agg_multi(P(aggs), naggs, P(row_gid), P(sel), ngroups, n,
+                       S(stream));
     check_async();
   });
   m.def("mask_from_nonzero", [](int64_t cnt, int64_t mask, int64_t n,

@@ -20,14 +20,16 @@
 
 __global__ void k_gb_build(const int32_t* __restrict__ hashes,
                            const KeyCol* __restrict__ keys, int nkeys,
+                           const int32_t* __restrict__ sel,
                            int32_t* __restrict__ slot_row,
                            int32_t* __restrict__ row_slot,
                            int32_t* __restrict__ claimed_slots,
                            int32_t* __restrict__ ngroups, uint32_t slot_mask,
                            int64_t n) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    uint32_t slot = slot_of((uint32_t)hashes[i], slot_mask);
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int32_t i = sel ? sel[j] : (int32_t)j;
+    uint32_t slot = slot_of((uint32_t)hashes[j], slot_mask);
     while (true) {
       // plain read first: slots are write-once (EMPTY -> row), so a stale
       // EMPTY just falls through to the CAS; this keeps the hot
@@ -40,12 +42,12 @@ __global__ void k_gb_build(const int32_t* __restrict__ hashes,
           // claimed a new group: record its slot so numbering touches only
           // live slots instead of sweeping the whole table
           claimed_slots[atomicAdd(ngroups, 1)] = (int32_t)slot;
-          row_slot[i] = (int32_t)slot;
+          row_slot[j] = (int32_t)slot;
           break;
         }
       }
-      if (cur == (int32_t)i || rows_equal(keys, keys, nkeys, i, cur)) {
-        row_slot[i] = (int32_t)slot;
+      if (cur == i || rows_equal(keys, keys, nkeys, i, cur)) {
+        row_slot[j] = (int32_t)slot;
         break;
       }
       slot = (slot + 1) & slot_mask;
@@ -204,6 +206,7 @@ __device__ __forceinline__ int64_t load_as_i64(const void* p, int t,
 template <bool USE_LDS>
 __global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
                                const int32_t* __restrict__ row_gid,
+                               const int32_t* __restrict__ sel,
                                int32_t ngroups, int64_t n) {
   extern __shared__ char lds_raw[];
   // LDS layout per agg a: acc[a][ngroups] (8B each) then cnt[a][ngroups]
@@ -225,9 +228,10 @@ __global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
     }
     __syncthreads();
   }
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    int32_t g = row_gid[i];
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int32_t g = row_gid[j];
+    int64_t i = sel ? (int64_t)sel[j] : j;
     for (int a = 0; a < naggs; ++a) {
       const AggDesc& d = aggs[a];
       char* base = lds_raw + (size_t)a * ngroups * 16;
@@ -296,14 +300,14 @@ __global__ void k_mask_from_nonzero(const int64_t* __restrict__ cnt,
 extern "C" {
 
 void hipdf_gb_build(const void* hashes, const void* keys, int nkeys,
-                    void* slot_row, void* row_slot, void* claimed_slots,
-                    void* ngroups, int64_t cap, int64_t n,
-                    hipStream_t stream) {
+                    const void* sel, void* slot_row, void* row_slot,
+                    void* claimed_slots, void* ngroups, int64_t cap,
+                    int64_t n, hipStream_t stream) {
   hipLaunchKernelGGL(k_gb_build, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
                      (const int32_t*)hashes, (const KeyCol*)keys, nkeys,
-                     (int32_t*)slot_row, (int32_t*)row_slot,
-                     (int32_t*)claimed_slots, (int32_t*)ngroups,
-                     (uint32_t)(cap - 1), n);
+                     (const int32_t*)sel, (int32_t*)slot_row,
+                     (int32_t*)row_slot, (int32_t*)claimed_slots,
+                     (int32_t*)ngroups, (uint32_t)(cap - 1), n);
 }
 
 void hipdf_gb_number(const void* claimed_slots, const void* slot_row,
@@ -362,18 +366,21 @@ void hipdf_gb_acc_init(int op, void* acc, int acc_is_double, int32_t ngroups,
 }
 
 void hipdf_gb_agg_multi(const void* aggs, int naggs, const void* row_gid,
-                        int32_t ngroups, int64_t n, hipStream_t stream) {
+                        const void* sel, int32_t ngroups, int64_t n,
+                        hipStream_t stream) {
   size_t lds = (size_t)naggs * ngroups * 16;
   bool use_lds = lds > 0 && lds <= 64 * 1024;
   dim3 grid = flat_grid(n, 4);
   if (use_lds)
     hipLaunchKernelGGL((k_gb_agg_multi<true>), grid, dim3(HIPDF_BLOCK), lds,
                        stream, (const AggDesc*)aggs, naggs,
-                       (const int32_t*)row_gid, ngroups, n);
+                       (const int32_t*)row_gid, (const int32_t*)sel, ngroups,
+                       n);
   else
     hipLaunchKernelGGL((k_gb_agg_multi<false>), grid, dim3(HIPDF_BLOCK), 0,
                        stream, (const AggDesc*)aggs, naggs,
-                       (const int32_t*)row_gid, ngroups, n);
+                       (const int32_t*)row_gid, (const int32_t*)sel, ngroups,
+                       n);
 }
 
 void hipdf_mask_from_nonzero(const void* cnt, void* mask, int64_t n,

@@ -48,11 +48,13 @@ enum HashKind : int {
 template <typename T>
 __global__ void k_murmur3_col(int kind, const T* __restrict__ a,
                               const uint64_t* __restrict__ av,
+                              const int32_t* __restrict__ sel,
                               int32_t* __restrict__ seeds, int64_t n) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
     if (!valid_bit(av, i)) continue;  // null keeps previous hash
-    uint32_t seed = (uint32_t)seeds[i];
+    uint32_t seed = (uint32_t)seeds[j];
     uint32_t h;
     T v = a[i];
     if (kind == HK_LONG) {
@@ -68,18 +70,20 @@ __global__ void k_murmur3_col(int kind, const T* __restrict__ a,
     } else {
       h = hash_int((uint32_t)(int32_t)(int64_t)v, seed);
     }
-    seeds[i] = (int32_t)h;
+    seeds[j] = (int32_t)h;
   }
 }
 
 __global__ void k_murmur3_str(const int32_t* __restrict__ offsets,
                               const uint8_t* __restrict__ bytes,
                               const uint64_t* __restrict__ av,
+                              const int32_t* __restrict__ sel,
                               int32_t* __restrict__ seeds, int64_t n) {
-  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
-       i += (int64_t)gridDim.x * blockDim.x) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
     if (!valid_bit(av, i)) continue;
-    uint32_t h1 = (uint32_t)seeds[i];
+    uint32_t h1 = (uint32_t)seeds[j];
     int32_t s = offsets[i], e = offsets[i + 1];
     int32_t len = e - s, p = s;
     // Spark hashUnsafeBytes: 4-byte LE words then SIGNED per-byte tail
@@ -90,7 +94,7 @@ __global__ void k_murmur3_str(const int32_t* __restrict__ offsets,
       h1 = mix_h1(h1, mix_k1(w));
     }
     for (; p < e; ++p) h1 = mix_h1(h1, mix_k1((uint32_t)(int32_t)(int8_t)bytes[p]));
-    seeds[i] = (int32_t)fmix(h1, (uint32_t)len);
+    seeds[j] = (int32_t)fmix(h1, (uint32_t)len);
   }
 }
 
@@ -107,20 +111,23 @@ __global__ void k_pmod_part(const int32_t* __restrict__ h, int32_t nparts,
 extern "C" {
 
 void hipdf_murmur3_col(int kind, int t, const void* a, const void* av,
-                       void* seeds, int64_t n, hipStream_t stream) {
+                       const void* sel, void* seeds, int64_t n,
+                       hipStream_t stream) {
   dim3 grid = flat_grid(n);
   dispatch_type(t, [&]<typename T>() {
     hipLaunchKernelGGL((k_murmur3_col<T>), grid, dim3(HIPDF_BLOCK), 0, stream,
                        kind, (const T*)a, (const uint64_t*)av,
-                       (int32_t*)seeds, n);
+                       (const int32_t*)sel, (int32_t*)seeds, n);
   });
 }
 
 void hipdf_murmur3_str(const void* offsets, const void* bytes, const void* av,
-                       void* seeds, int64_t n, hipStream_t stream) {
+                       const void* sel, void* seeds, int64_t n,
+                       hipStream_t stream) {
   hipLaunchKernelGGL(k_murmur3_str, flat_grid(n), dim3(HIPDF_BLOCK), 0,
                      stream, (const int32_t*)offsets, (const uint8_t*)bytes,
-                     (const uint64_t*)av, (int32_t*)seeds, n);
+                     (const uint64_t*)av, (const int32_t*)sel,
+                     (int32_t*)seeds, n);
 }
 
 void hipdf_pmod_part(const void* h, int nparts, void* part, int64_t n,

@@ -378,8 +378,8 @@ def _exclusive_scan_i64(vals: torch.Tensor) -> Tuple[torch.Tensor, int]:
 # selection
 # ---------------------------------------------------------------------------
 
-def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
-    n = batch.num_rows
+def mask_to_sel(mask: Column, n: int) -> torch.Tensor:
+    """Selection vector (int32 row indices) of rows where mask is true."""
     s = _stream()
     nb = ext.sel_num_blocks(n)
     counts = torch.empty(nb, dtype=torch.int64, device="cuda")
@@ -390,7 +390,12 @@ def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
     if total:
         ext.mask_scatter(mask.data.data_ptr(), _ptr(mask.validity),
                          offsets.data_ptr(), idx.data_ptr(), n, s)
-    return _gather_by_idx(batch, idx, total, maybe_negative=False)
+    return idx
+
+
+def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
+    idx = mask_to_sel(mask, batch.num_rows)
+    return _gather_by_idx(batch, idx, idx.numel(), maybe_negative=False)
 
 
 def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> ColumnBatch:
@@ -525,19 +530,23 @@ _HASH_KIND = {
 }
 
 
-def _murmur3_tensor(cols: List[Column], seed: int) -> torch.Tensor:
-    n = cols[0].size
+def _murmur3_tensor(cols: List[Column], seed: int,
+                    sel: Optional[torch.Tensor] = None,
+                    n_out: Optional[int] = None) -> torch.Tensor:
+    n = cols[0].size if sel is None else (n_out if n_out is not None
+                                          else sel.numel())
     s = _stream()
     seeds = torch.full((max(n, 1),), seed, dtype=torch.int32, device="cuda")[:n]
+    selp = 0 if sel is None else sel.data_ptr()
     for c in cols:
         if n == 0:
             break
         if c.dtype.id is TypeId.STRING:
             ext.murmur3_str(c.offsets.data_ptr(), c.data.data_ptr(),
-                            _ptr(c.validity), seeds.data_ptr(), n, s)
+                            _ptr(c.validity), selp, seeds.data_ptr(), n, s)
         else:
             ext.murmur3_col(_HASH_KIND[c.dtype.id], _ht(c.dtype),
-                            c.data.data_ptr(), _ptr(c.validity),
+                            c.data.data_ptr(), _ptr(c.validity), selp,
                             seeds.data_ptr(), n, s)
     return seeds
 
@@ -642,28 +651,33 @@ def _next_pow2(x: int) -> int:
 
 
 def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
-                       aggs: List[Tuple[str, int, DType]]) -> ColumnBatch:
-    n = batch.num_rows
+                       aggs: List[Tuple[str, int, DType]],
+                       sel: Optional[torch.Tensor] = None) -> ColumnBatch:
+    """Hash group-by; with `sel` (int32 row indices) only the selected rows
+    participate — the filter-into-aggregate fusion path (no materialized
+    gather of the filtered batch)."""
+    n = batch.num_rows if sel is None else sel.numel()
     s = _stream()
     keys = [batch.columns[i] for i in key_idx]
     if n == 0:
         out = [_empty_col(k.dtype) for k in keys]
         out += [_empty_col(dt) for _, _, dt in aggs]
         return ColumnBatch(out, 0)
+    selp = 0 if sel is None else sel.data_ptr()
     if not key_idx:
         # global aggregate: single group
         row_gid = torch.zeros(n, dtype=torch.int32, device="cuda")
         ngroups = 1
         leaders = torch.zeros(1, dtype=torch.int32, device="cuda")
     else:
-        h = _murmur3_tensor(keys, 42)
+        h = _murmur3_tensor(keys, 42, sel, n)
         cap = max(1024, _next_pow2(2 * n))
         desc = _key_desc(keys)
         slot_row = torch.full((cap,), -1, dtype=torch.int32, device="cuda")
         row_slot = torch.empty(n, dtype=torch.int32, device="cuda")
         claimed = torch.empty(n, dtype=torch.int32, device="cuda")
         ngroups_t = torch.zeros(1, dtype=torch.int32, device="cuda")
-        ext.gb_build(h.data_ptr(), desc.data_ptr(), len(keys),
+        ext.gb_build(h.data_ptr(), desc.data_ptr(), len(keys), selp,
                      slot_row.data_ptr(), row_slot.data_ptr(),
                      claimed.data_ptr(), ngroups_t.data_ptr(), cap, n, s)
         slot_gid = torch.empty(cap, dtype=torch.int32, device="cuda")
@@ -705,7 +719,7 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         desc = torch.frombuffer(bytearray(b"".join(blobs)),
                                 dtype=torch.uint8).cuda()
         ext.gb_agg_multi(desc.data_ptr(), len(aggs), row_gid.data_ptr(),
-                         ngroups, n, s)
+                         selp, ngroups, n, s)
     for op, out_dtype, acc_is_double, acc, cnt in allocs:
         if op in ("count", "count_all"):
             out_cols.append(Column(out_dtype, ngroups, cnt[:ngroups].clone(),

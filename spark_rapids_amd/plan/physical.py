@@ -252,18 +252,39 @@ class HashAggregateExec(PhysicalExec):
         self.merge_target_bytes = merge_target_bytes
 
     def execute(self) -> Iterator[ColumnBatch]:
-        in_schema = self.children[0].schema
+        # filter-into-aggregate fusion: when the child is a GPU filter,
+        # evaluate the predicate to a selection vector and aggregate the
+        # selected rows directly — no materialized gather of the filtered
+        # batch (reference analogue: the pre-projection + filter fusion the
+        # tiered project/AST path gives cudf).
+        source = self.children[0]
+        fused_condition = None
+        if self.gpu and isinstance(source, FilterExec) and source.gpu:
+            fused_condition = source.condition
+            source = source.children[0]
+        in_schema = source.schema
         nkeys = len(self.group_exprs)
         value_exprs, partial, merge_ops, final = _lower_aggs(self.aggs, in_schema)
 
         partial_results: List[ColumnBatch] = []
-        for batch in self.children[0].execute():
+        for batch in source.execute():
             def task(b):
+                sel = None
+                if fused_condition is not None:
+                    from ..ops import gpu_backend as _gb
+
+                    mask = fused_condition.eval(b, in_schema)
+                    sel = _gb.mask_to_sel(mask, b.num_rows)
                 key_cols = [e.eval(b, in_schema) for e in self.group_exprs]
                 val_cols = [e.eval(b, in_schema) for e in value_exprs]
                 pre = ColumnBatch(key_cols + val_cols, b.num_rows)
                 specs = [(op, (nkeys + v) if v >= 0 else -1, dt)
                          for op, v, dt in partial]
+                if sel is not None:
+                    from ..ops import gpu_backend as _gb
+
+                    return _gb.group_by_aggregate(pre, list(range(nkeys)),
+                                                  specs, sel=sel)
                 return ops.group_by_aggregate(pre, list(range(nkeys)), specs)
             partial_results.append(with_retry_split_single(task, batch))
 
