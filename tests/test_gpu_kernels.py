@@ -454,3 +454,30 @@ def test_enabled_assert_mode_gpu():
     df2 = s2.create_dataframe({"a": [1.0, 2.0]})
     with pytest.raises(AssertionError):
         df2.filter(sr.col("a") > 1.0).count()
+
+
+def test_concurrent_queries_semaphore_gpu():
+    """Two threads running GPU queries concurrently under the semaphore
+    (reference analogue: concurrentGpuTasks)."""
+    import threading
+
+    s = sr.Session({"spark.rapids.sql.concurrentGpuTasks": 2})
+    n = 200_000
+    df = s.create_dataframe({
+        "k": RNG.integers(0, 100, n), "v": RNG.uniform(0, 1, n)})
+    results = [None, None]
+    errors = []
+
+    def work(slot):
+        try:
+            out = (df.filter(sr.col("v") > 0.5).group_by("k")
+                   .agg(sr.count_star()).collect())
+            results[slot] = sum(r[1] for r in out)
+        except Exception as e:  # noqa: BLE001
+            errors.append(e)
+
+    ts = [threading.Thread(target=work, args=(i,)) for i in range(2)]
+    [t.start() for t in ts]
+    [t.join() for t in ts]
+    assert not errors, errors
+    assert results[0] == results[1] and results[0] is not None
