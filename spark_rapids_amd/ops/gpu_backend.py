@@ -690,9 +690,10 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         ext.gb_rowgid(row_slot.data_ptr(), slot_gid.data_ptr(),
                       row_gid.data_ptr(), n, s)
     leaders = leaders[:ngroups]
-    key_batch = _gather_by_idx(ColumnBatch(keys, n) if keys else
-                               ColumnBatch([], 0), leaders, ngroups,
-                               maybe_negative=False) if keys else None
+    # leaders are ORIGINAL row ids (pre-selection), so gather keys from the
+    # full-length key columns
+    key_batch = _gather_by_idx(ColumnBatch(keys, keys[0].size), leaders,
+                               ngroups, maybe_negative=False) if keys else None
     out_cols = list(key_batch.columns) if key_batch is not None else []
 
     # fused multi-aggregate: one kernel pass accumulates every agg
