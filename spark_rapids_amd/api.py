@@ -92,6 +92,18 @@ class DataFrame:
             desc = list(descending)
         return DataFrame(self.session, L.Sort(self.plan, ks, desc))
 
+    def distinct(self) -> "DataFrame":
+        """Drop duplicate rows (group-by all columns with no aggregates)."""
+        keys = [_col(f.name) for f in self.plan.schema().fields]
+        return DataFrame(self.session, L.Aggregate(keys, [], self.plan))
+
+    def map_batches(self, fn, schema: Optional[Schema] = None) -> "DataFrame":
+        """Run a python function over host batches (the CPU-bridge / UDF
+        escape hatch, reference analogue: GpuCpuBridgeExpression). The
+        overrides pass keeps this on CPU with transitions around it."""
+        return DataFrame(self.session,
+                         L.MapBatches(fn, self.plan, schema))
+
     def limit(self, n: int) -> "DataFrame":
         return DataFrame(self.session, L.Limit(self.plan, n))
 
@@ -239,6 +251,17 @@ class Session:
 
         batch = df.collect_batch()
         write_parquet(batch, df.schema, path, compression)
+
+    def range(self, start: int, end: Optional[int] = None,
+              step: int = 1, num_partitions: int = 1) -> DataFrame:
+        """Integer range source (reference analogue: GpuRangeExec)."""
+        if end is None:
+            start, end = 0, start
+        import numpy as np
+
+        vals = np.arange(start, end, step, dtype=np.int64)
+        return self.create_dataframe({"id": vals},
+                                     num_partitions=num_partitions)
 
     def read_csv(self, path: str, header: bool = True,
                  delimiter: str = ",") -> DataFrame:

@@ -120,6 +120,22 @@ class Join(LogicalPlan):
         return f"Join({self.how})"
 
 
+class MapBatches(LogicalPlan):
+    """CPU python-function operator (UDF bridge); never places on GPU."""
+
+    def __init__(self, fn, child: LogicalPlan, schema=None):
+        self.fn = fn
+        self.child = child
+        self._schema = schema
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        return self._schema if self._schema is not None else self.child.schema()
+
+
 class Window(LogicalPlan):
     def __init__(self, window_exprs, child: LogicalPlan):
         self.window_exprs = list(window_exprs)

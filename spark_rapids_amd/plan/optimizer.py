@@ -100,6 +100,9 @@ def prune_columns(plan: L.LogicalPlan,
             child_needed &= set(plan.child.schema().names)
         return L.Window(plan.window_exprs,
                         prune_columns(plan.child, child_needed))
+    if isinstance(plan, L.MapBatches):
+        return L.MapBatches(plan.fn, prune_columns(plan.child, None),
+                            plan._schema)
     if isinstance(plan, L.Limit):
         return L.Limit(prune_columns(plan.child, needed), plan.n)
     if isinstance(plan, L.Union):

@@ -156,6 +156,8 @@ class Tagger:
                     reasons.append(f"join key {k}: {r}")
             if node.how not in ("inner", "left", "semi", "anti"):
                 reasons.append(f"join type {node.how} not on GPU")
+        elif isinstance(node, L.MapBatches):
+            reasons.append("python map_batches runs on CPU (UDF bridge)")
         elif isinstance(node, L.Window):
             spec = node.window_exprs[0].spec
             for k in spec.partition_by + spec.order_by:
@@ -274,6 +276,9 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
         return P.HashJoinExec(device, kids[0], kids[1], node.left_on,
                               node.right_on, node.how, node.schema(),
                               right_replicated=L.is_replicated(node.right))
+    if isinstance(node, L.MapBatches):
+        return P.MapBatchesExec(node.fn, _ensure_device(kids[0], "cpu"),
+                                node.schema())
     if isinstance(node, L.Window):
         from .window_exec import WindowExec
 

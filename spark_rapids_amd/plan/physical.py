@@ -467,6 +467,20 @@ class SortExec(PhysicalExec):
         return f"{self.name()}[{ks}]"
 
 
+class MapBatchesExec(PhysicalExec):
+    """Host python-function operator (CPU bridge / UDF escape hatch)."""
+
+    def __init__(self, fn, child: PhysicalExec, schema: Schema):
+        super().__init__("cpu", schema, [child])
+        self.fn = fn
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        for batch in self.children[0].execute():
+            out = self.fn(batch.cpu())
+            if out is not None and out.num_rows >= 0:
+                yield out
+
+
 class LimitExec(PhysicalExec):
     def __init__(self, device: str, n: int, child: PhysicalExec):
         super().__init__(device, child.schema, [child])

@@ -172,3 +172,30 @@ def test_repartition_merge_fallback(session):
     out = df.group_by("k").agg(sum_(col("v")), count_star()).collect()
     assert sum(r[2] for r in out) == n
     assert len(out) == len(set(r[0] for r in out))  # keys unique across buckets
+
+
+def test_distinct(session):
+    df = session.create_dataframe({"a": [1, 1, 2, 2, 3], "b": [1, 1, 2, 9, 3]})
+    out = sorted(df.distinct().collect())
+    assert out == [(1, 1), (2, 2), (2, 9), (3, 3)]
+
+
+def test_map_batches_udf_bridge(session):
+    df = session.create_dataframe({"a": [1, 2, 3, 4]})
+
+    def double(batch):
+        from spark_rapids_amd import Column, ColumnBatch, INT64
+
+        vals = [v * 2 for v in batch.columns[0].to_pylist()]
+        return ColumnBatch([Column.from_pylist(vals, INT64)])
+
+    out = df.map_batches(double).to_pydict()["a"]
+    assert out == [2, 4, 6, 8]
+    # composable with engine ops afterwards
+    assert df.map_batches(double).filter(col("a") > 4).count() == 2
+
+
+def test_session_range(session):
+    df = session.range(10)
+    assert df.count() == 10
+    assert session.range(2, 8, 2).to_pydict()["id"] == [2, 4, 6]
