@@ -210,8 +210,7 @@ class BinaryExpr(Expression):
     def children(self):
         return (self.left, self.right)
 
-    def _in_dtype(self, schema) -> DType:
-        lt, rt = self.left.dtype(schema), self.right.dtype(schema)
+    def _common(self, lt: DType, rt: DType) -> DType:
         if lt.id is TypeId.NULL:
             return rt
         if rt.id is TypeId.NULL:
@@ -222,7 +221,12 @@ class BinaryExpr(Expression):
             return FLOAT64
         return promote(lt, rt)
 
+    def _in_dtype(self, schema) -> DType:
+        return self._common(self.left.dtype(schema), self.right.dtype(schema))
+
     def dtype(self, schema: Schema) -> DType:
+        # NOTE: children's dtype() is called exactly once here — recursing
+        # more than once makes deep expression chains exponential
         if self.op in _BOOL_OPS:
             return BOOL
         if self.op in _DOUBLE_OPS:
@@ -230,7 +234,7 @@ class BinaryExpr(Expression):
         lt, rt = self.left.dtype(schema), self.right.dtype(schema)
         if self.op == "sub" and lt.is_timelike and rt.is_timelike:
             return INT32  # datediff domain
-        it = self._in_dtype(schema)
+        it = self._common(lt, rt)
         if it.is_decimal and self.op in ("add", "sub"):
             return DType.decimal(min(it.precision + 1, 38), it.scale)
         return it
