@@ -231,6 +231,13 @@ class Session:
     def table(self, name: str) -> DataFrame:
         return self.catalog[name]
 
+    def sql(self, query: str) -> DataFrame:
+        """Parse a SQL query over registered tables into the same logical
+        plans the DataFrame API builds (and through the same GPU overrides)."""
+        from .sql.parser import parse_sql
+
+        return parse_sql(self, query)
+
     def read_parquet(self, path: str, columns=None) -> DataFrame:
         import torch
 

@@ -1,0 +1,437 @@
+"""SQL frontend: a recursive-descent parser for the accelerated subset.
+
+The reference accelerates Spark SQL; this engine owns the frontend, so a
+compact SQL layer maps queries onto the same logical plans the DataFrame
+API builds (and therefore through the same GPU overrides pass).
+
+Supported: SELECT <exprs|*> FROM <table> [JOIN <table> ON a = b]...
+[WHERE <cond>] [GROUP BY <cols>] [HAVING <cond>] [ORDER BY <cols> [ASC|DESC]]
+[LIMIT n]. Expressions: + - * / %, comparisons, AND/OR/NOT, IS [NOT] NULL,
+IN (...), BETWEEN, LIKE, CASE WHEN, CAST(x AS t), literals, and the
+functions sum/avg/count/min/max/stddev/variance/coalesce/round/abs/sqrt/
+upper/lower/length/substring/year/month/day.
+"""
+from __future__ import annotations
+
+import re
+from typing import List, Optional
+
+from ..expr import aggregates as A
+from ..expr.expressions import (BinaryExpr, CaseWhen, CastExpr, Coalesce,
+                                ColumnRef, Expression, IsNull, Literal, Round,
+                                StringPredicate, Substring, UnaryExpr)
+from ..types import (BOOL, DATE32, DType, FLOAT32, FLOAT64, INT16, INT32,
+                     INT64, INT8, STRING, TIMESTAMP)
+
+_TOKEN_RE = re.compile(r"""
+    \s*(?:
+      (?P<num>\d+\.\d*(?:[eE][+-]?\d+)?|\.\d+|\d+(?:[eE][+-]?\d+)?)
+    | (?P<str>'(?:[^']|'')*')
+    | (?P<name>[A-Za-z_][A-Za-z_0-9]*)
+    | (?P<op><=|>=|<>|!=|=|<|>|\+|-|\*|/|%|\(|\)|,|\.)
+    )""", re.VERBOSE)
+
+_TYPES = {
+    "boolean": BOOL, "tinyint": INT8, "smallint": INT16, "int": INT32,
+    "integer": INT32, "bigint": INT64, "long": INT64, "float": FLOAT32,
+    "double": FLOAT64, "date": DATE32, "timestamp": TIMESTAMP,
+    "string": STRING,
+}
+
+_AGG_FUNCS = {"sum", "avg", "count", "min", "max", "stddev", "variance"}
+
+
+class SqlError(ValueError):
+    pass
+
+
+def tokenize(text: str) -> List[tuple]:
+    out = []
+    pos = 0
+    while pos < len(text):
+        m = _TOKEN_RE.match(text, pos)
+        if not m or m.end() == pos:
+            if text[pos:].strip() == "":
+                break
+            raise SqlError(f"cannot tokenize at: {text[pos:pos+20]!r}")
+        pos = m.end()
+        if m.group("num") is not None:
+            out.append(("num", m.group("num")))
+        elif m.group("str") is not None:
+            out.append(("str", m.group("str")[1:-1].replace("''", "'")))
+        elif m.group("name") is not None:
+            out.append(("name", m.group("name")))
+        else:
+            out.append(("op", m.group("op")))
+    out.append(("end", ""))
+    return out
+
+
+class Parser:
+    def __init__(self, text: str, session):
+        self.toks = tokenize(text)
+        self.i = 0
+        self.session = session
+
+    # -- token helpers ---------------------------------------------------
+    def peek(self, k=0):
+        return self.toks[min(self.i + k, len(self.toks) - 1)]
+
+    def next(self):
+        t = self.toks[self.i]
+        self.i += 1
+        return t
+
+    def kw(self, word) -> bool:
+        t = self.peek()
+        if t[0] == "name" and t[1].upper() == word.upper():
+            self.i += 1
+            return True
+        return False
+
+    def expect_kw(self, word):
+        if not self.kw(word):
+            raise SqlError(f"expected {word} at {self.peek()}")
+
+    def op(self, sym) -> bool:
+        t = self.peek()
+        if t[0] == "op" and t[1] == sym:
+            self.i += 1
+            return True
+        return False
+
+    def expect_op(self, sym):
+        if not self.op(sym):
+            raise SqlError(f"expected {sym!r} at {self.peek()}")
+
+    # -- grammar ---------------------------------------------------------
+    def parse_query(self):
+        self.expect_kw("SELECT")
+        star = False
+        items = []  # (expr_or_aggexpr, alias)
+        if self.op("*"):
+            star = True
+        else:
+            while True:
+                e = self.parse_select_item()
+                alias = None
+                if self.kw("AS"):
+                    alias = self.next()[1]
+                elif self.peek()[0] == "name" and self.peek()[1].upper() not in (
+                        "FROM", "WHERE", "GROUP", "ORDER", "LIMIT", "HAVING",
+                        "JOIN", "LEFT", "INNER", "ON", "AND", "OR", "ASC",
+                        "DESC", "BY"):
+                    alias = self.next()[1]
+                items.append((e, alias))
+                if not self.op(","):
+                    break
+        self.expect_kw("FROM")
+        df = self.parse_table()
+        while True:
+            how = None
+            up = self.peek()[1].upper() if self.peek()[0] == "name" else ""
+            if up == "JOIN":
+                self.next()
+                how = "inner"
+            elif up == "INNER":
+                self.next()
+                self.expect_kw("JOIN")
+                how = "inner"
+            elif up == "LEFT":
+                self.next()
+                self.kw("OUTER")
+                self.expect_kw("JOIN")
+                how = "left"
+            else:
+                break
+            right = self.parse_table()
+            self.expect_kw("ON")
+            lk, rk = self.parse_join_keys()
+            df = df.join(right, on=lk, right_on=rk, how=how)
+        where = None
+        if self.kw("WHERE"):
+            where = self.parse_expr()
+        group_cols = []
+        if self.kw("GROUP"):
+            self.expect_kw("BY")
+            while True:
+                group_cols.append(self.next()[1])
+                if not self.op(","):
+                    break
+        having = None
+        if self.kw("HAVING"):
+            having = self.parse_expr()
+        order = []
+        if self.kw("ORDER"):
+            self.expect_kw("BY")
+            while True:
+                name = self.next()[1]
+                desc = False
+                if self.kw("DESC"):
+                    desc = True
+                else:
+                    self.kw("ASC")
+                order.append((name, desc))
+                if not self.op(","):
+                    break
+        limit = None
+        if self.kw("LIMIT"):
+            limit = int(self.next()[1])
+        if self.peek()[0] != "end":
+            raise SqlError(f"unexpected trailing tokens at {self.peek()}")
+
+        # assemble plan
+        if where is not None:
+            df = df.filter(where)
+        has_aggs = any(isinstance(e, A.AggExpr) for e, _ in items)
+        if group_cols or has_aggs:
+            keys = group_cols
+            aggs = []
+            for e, alias in items:
+                if isinstance(e, A.AggExpr):
+                    aggs.append(e.alias(alias) if alias else e)
+                elif isinstance(e, ColumnRef) and e.name in group_cols:
+                    pass  # group key, included automatically
+                elif star:
+                    pass
+                else:
+                    raise SqlError(
+                        f"non-aggregate select item {e} not in GROUP BY")
+            df = df.group_by(*keys).agg(*aggs) if keys else df.agg(*aggs)
+        elif not star:
+            exprs = [(e.alias(alias) if alias else e) for e, alias in items]
+            # ORDER BY may reference pre-projection columns (Spark allows
+            # both); sort first when a key is not in the select list
+            if order and not all(n in [x.output_name() for x in exprs]
+                                 for n, _ in order):
+                df = df.sort(*[n for n, _ in order],
+                             descending=[d for _, d in order])
+                order = []
+            df = df.select(*exprs)
+        if having is not None:
+            df = df.filter(having)
+        if order:
+            df = df.sort(*[n for n, _ in order],
+                         descending=[d for _, d in order])
+        if limit is not None:
+            df = df.limit(limit)
+        return df
+
+    def parse_table(self):
+        name = self.next()[1]
+        df = self.session.table(name)
+        # optional alias (ignored name binding; columns stay unqualified)
+        t = self.peek()
+        if t[0] == "name" and t[1].upper() not in (
+                "JOIN", "LEFT", "INNER", "ON", "WHERE", "GROUP", "ORDER",
+                "LIMIT", "HAVING"):
+            self.next()
+        return df
+
+    def parse_join_keys(self):
+        lk, rk = [], []
+        while True:
+            a = self.parse_qualified_name()
+            self.expect_op("=")
+            b = self.parse_qualified_name()
+            lk.append(a)
+            rk.append(b)
+            if not self.kw("AND"):
+                break
+        return lk, rk
+
+    def parse_qualified_name(self) -> str:
+        n = self.next()[1]
+        if self.op("."):
+            n = self.next()[1]  # drop qualifier; names are engine-global
+        return n
+
+    def parse_select_item(self):
+        return self.parse_expr()
+
+    # expression precedence: OR < AND < NOT < cmp < add < mul < unary
+    def parse_expr(self):
+        e = self.parse_and()
+        while self.kw("OR"):
+            e = BinaryExpr("or", e, self.parse_and())
+        return e
+
+    def parse_and(self):
+        e = self.parse_not()
+        while self.kw("AND"):
+            e = BinaryExpr("and", e, self.parse_not())
+        return e
+
+    def parse_not(self):
+        if self.kw("NOT"):
+            return UnaryExpr("not", self.parse_not())
+        return self.parse_cmp()
+
+    def parse_cmp(self):
+        e = self.parse_add()
+        t = self.peek()
+        if t[0] == "op" and t[1] in ("=", "<>", "!=", "<", "<=", ">", ">="):
+            self.next()
+            opmap = {"=": "eq", "<>": "ne", "!=": "ne", "<": "lt",
+                     "<=": "le", ">": "gt", ">=": "ge"}
+            return BinaryExpr(opmap[t[1]], e, self.parse_add())
+        if self.kw("IS"):
+            neg = self.kw("NOT")
+            self.expect_kw("NULL")
+            x = IsNull(e)
+            return UnaryExpr("not", x) if neg else x
+        if self.kw("BETWEEN"):
+            lo = self.parse_add()
+            self.expect_kw("AND")
+            hi = self.parse_add()
+            return BinaryExpr("and", BinaryExpr("ge", e, lo),
+                              BinaryExpr("le", e, hi))
+        if self.kw("IN"):
+            self.expect_op("(")
+            vals = []
+            while True:
+                vals.append(self.parse_literal_value())
+                if not self.op(","):
+                    break
+            self.expect_op(")")
+            from ..expr.expressions import isin
+
+            return isin(e, *vals)
+        if self.kw("LIKE"):
+            pat = self.next()
+            if pat[0] != "str":
+                raise SqlError("LIKE needs a string literal")
+            return StringPredicate("like", e, pat[1])
+        return e
+
+    def parse_add(self):
+        e = self.parse_mul()
+        while True:
+            if self.op("+"):
+                e = BinaryExpr("add", e, self.parse_mul())
+            elif self.op("-"):
+                e = BinaryExpr("sub", e, self.parse_mul())
+            else:
+                return e
+
+    def parse_mul(self):
+        e = self.parse_unary()
+        while True:
+            if self.op("*"):
+                e = BinaryExpr("mul", e, self.parse_unary())
+            elif self.op("/"):
+                e = BinaryExpr("div", e, self.parse_unary())
+            elif self.op("%"):
+                e = BinaryExpr("mod", e, self.parse_unary())
+            else:
+                return e
+
+    def parse_unary(self):
+        if self.op("-"):
+            return UnaryExpr("neg", self.parse_unary())
+        return self.parse_primary()
+
+    def parse_literal_value(self):
+        t = self.next()
+        if t[0] == "num":
+            return float(t[1]) if any(c in t[1] for c in ".eE") else int(t[1])
+        if t[0] == "str":
+            return t[1]
+        if t[0] == "name" and t[1].upper() in ("TRUE", "FALSE"):
+            return t[1].upper() == "TRUE"
+        raise SqlError(f"expected literal at {t}")
+
+    def parse_primary(self):
+        t = self.peek()
+        if t[0] == "op" and t[1] == "(":
+            self.next()
+            e = self.parse_expr()
+            self.expect_op(")")
+            return e
+        if t[0] == "num":
+            self.next()
+            v = float(t[1]) if any(c in t[1] for c in ".eE") else int(t[1])
+            return Literal(v)
+        if t[0] == "str":
+            self.next()
+            return Literal(t[1])
+        if t[0] == "name":
+            up = t[1].upper()
+            if up == "NULL":
+                self.next()
+                return Literal(None)
+            if up in ("TRUE", "FALSE"):
+                self.next()
+                return Literal(up == "TRUE")
+            if up == "CASE":
+                return self.parse_case()
+            if up == "CAST":
+                self.next()
+                self.expect_op("(")
+                e = self.parse_expr()
+                self.expect_kw("AS")
+                tname = self.next()[1].lower()
+                if tname not in _TYPES:
+                    raise SqlError(f"unknown type {tname}")
+                self.expect_op(")")
+                return CastExpr(e, _TYPES[tname])
+            if self.peek(1) == ("op", "("):
+                return self.parse_func()
+            self.next()
+            return ColumnRef(self.parse_qualified_suffix(t[1]))
+        raise SqlError(f"unexpected token {t}")
+
+    def parse_qualified_suffix(self, first: str) -> str:
+        if self.op("."):
+            return self.next()[1]
+        return first
+
+    def parse_case(self):
+        self.expect_kw("CASE")
+        branches = []
+        else_e = None
+        while self.kw("WHEN"):
+            c = self.parse_expr()
+            self.expect_kw("THEN")
+            v = self.parse_expr()
+            branches.append((c, v))
+        if self.kw("ELSE"):
+            else_e = self.parse_expr()
+        self.expect_kw("END")
+        return CaseWhen(branches, else_e)
+
+    def parse_func(self):
+        name = self.next()[1].lower()
+        self.expect_op("(")
+        if name == "count" and self.op("*"):
+            self.expect_op(")")
+            return A.count_star()
+        args = []
+        if not self.op(")"):
+            while True:
+                args.append(self.parse_expr())
+                if not self.op(","):
+                    break
+            self.expect_op(")")
+        if name in _AGG_FUNCS:
+            ctor = {"sum": A.sum_, "avg": A.avg, "count": A.count,
+                    "min": A.min_, "max": A.max_, "stddev": A.stddev,
+                    "variance": A.variance}[name]
+            return ctor(args[0])
+        if name == "coalesce":
+            return Coalesce(*args)
+        if name == "round":
+            scale = args[1].value if len(args) > 1 else 0
+            return Round(args[0], int(scale))
+        if name in ("abs", "sqrt", "exp", "log", "floor", "ceil", "upper",
+                    "lower", "length", "year", "month", "day"):
+            return UnaryExpr(name, args[0])
+        if name == "substring" or name == "substr":
+            pos = int(args[1].value)
+            ln = int(args[2].value) if len(args) > 2 else -1
+            return Substring(args[0], pos, ln)
+        raise SqlError(f"unknown function {name}")
+
+
+def parse_sql(session, text: str):
+    return Parser(text, session).parse_query()

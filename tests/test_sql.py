@@ -1,0 +1,88 @@
+"""SQL frontend tests (parser -> logical plans -> engine)."""
+import pytest
+
+from spark_rapids_amd import Session
+
+
+@pytest.fixture
+def s():
+    sess = Session()
+    sess.register("t", sess.create_dataframe({
+        "k": [1, 2, 1, 2, 3],
+        "v": [10.0, 20.0, 30.0, None, 50.0],
+        "s": ["aa", "ab", "bb", "aa", None],
+        "d": [1, 2, 3, 4, 5],
+    }))
+    sess.register("dim", sess.create_dataframe({
+        "k": [1, 2], "name": ["one", "two"]}))
+    return sess
+
+
+def test_select_star_where(s):
+    out = s.sql("SELECT * FROM t WHERE v > 15").collect()
+    assert len(out) == 3
+
+
+def test_projection_arith_alias(s):
+    out = s.sql("SELECT v * 2 AS dbl, k FROM t WHERE k = 1").collect()
+    assert sorted(out) == [(20.0, 1), (60.0, 1)]
+
+
+def test_group_by_aggs(s):
+    out = s.sql("SELECT k, sum(v) AS sv, count(*) AS c FROM t "
+                "GROUP BY k ORDER BY k").collect()
+    assert out == [(1, 40.0, 2), (2, 20.0, 2), (3, 50.0, 1)]
+
+
+def test_global_agg(s):
+    out = s.sql("SELECT sum(v) AS total, avg(v) a, count(v) c FROM t").collect()
+    assert out[0][0] == 110.0 and out[0][2] == 4
+    assert out[0][1] == pytest.approx(110.0 / 4)
+
+
+def test_join_on(s):
+    out = s.sql("SELECT name, v FROM t JOIN dim ON t.k = dim.k "
+                "WHERE v IS NOT NULL ORDER BY v").collect()
+    assert out[0] == ("one", 10.0)
+    assert len(out) == 3
+
+
+def test_case_when_cast(s):
+    out = s.sql("SELECT CASE WHEN v > 15 THEN 1 ELSE 0 END AS big FROM t "
+                "WHERE v IS NOT NULL").collect()
+    assert [r[0] for r in out] == [0, 1, 1, 1]
+    out = s.sql("SELECT CAST(v AS bigint) AS i FROM t WHERE k = 1").collect()
+    assert sorted(r[0] for r in out) == [10, 30]
+
+
+def test_between_in_like(s):
+    assert s.sql("SELECT * FROM t WHERE d BETWEEN 2 AND 4").collect() != []
+    assert len(s.sql("SELECT * FROM t WHERE k IN (1, 3)").collect()) == 3
+    out = s.sql("SELECT s FROM t WHERE s LIKE 'a%'").collect()
+    assert sorted(r[0] for r in out) == ["aa", "aa", "ab"]
+
+
+def test_order_desc_limit(s):
+    out = s.sql("SELECT d FROM t ORDER BY d DESC LIMIT 2").collect()
+    assert out == [(5,), (4,)]
+
+
+def test_having_via_alias(s):
+    out = s.sql("SELECT k, sum(v) AS sv FROM t GROUP BY k HAVING sv > 25 "
+                "ORDER BY k").collect()
+    assert out == [(1, 40.0), (3, 50.0)]
+
+
+def test_functions(s):
+    out = s.sql("SELECT round(v / 3, 1) r, upper(s) u, length(s) l FROM t "
+                "WHERE k = 1 ORDER BY d").collect()
+    assert out[0] == (3.3, "AA", 2)
+
+
+def test_parse_errors(s):
+    from spark_rapids_amd.sql.parser import SqlError
+
+    with pytest.raises(SqlError):
+        s.sql("SELECT FROM t")
+    with pytest.raises(SqlError):
+        s.sql("SELECT * FROM t WHERE ???")
