@@ -31,6 +31,7 @@ KERNELS = [
     "kernels/decode.hip",
     "kernels/strings.hip",
     "kernels/window.hip",
+    "kernels/decimal128.hip",
 ]
 
 CXXFLAGS = ["-O3", "-std=c++20", "-fPIC", f"--offload-arch={ARCH}",

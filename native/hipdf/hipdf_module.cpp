@@ -79,6 +79,16 @@ void hipdf_join_fill(int, const void*, const void*, const void*, int,
                      void*, int64_t, hipStream_t);
 int64_t part_num_blocks(int64_t);
 int64_t sort_num_blocks(int64_t);
+void hipdf_i128_arith(int, const void*, const void*, const void*, const void*,
+                      void*, void*, int64_t, hipStream_t);
+void hipdf_i128_cmp(int, const void*, const void*, const void*, const void*,
+                    void*, void*, int64_t, hipStream_t);
+void hipdf_i64_to_i128(const void*, void*, int64_t, hipStream_t);
+void hipdf_i128_to_f64(const void*, void*, int64_t, hipStream_t);
+void hipdf_gb_sum_i64_to_i128(const void*, const void*, const void*,
+                              const void*, void*, void*, int64_t, hipStream_t);
+void hipdf_gb_sum_i128(const void*, const void*, const void*, const void*,
+                       void*, void*, int64_t, hipStream_t);
 void hipdf_change_flags(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_iota_i32(void*, int64_t, hipStream_t);
 void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
@@ -425,6 +435,41 @@ PYBIND11_MODULE(hipdf, m) {
                           int64_t stream) {
     hipdf_substr_copy(P(ab), P(bstart), P(blen), P(out_off), PM(out_bytes), n,
                       S(stream));
+    check_async();
+  });
+
+  m.def("i128_arith", [](int op, int64_t a, int64_t b, int64_t av, int64_t bv,
+                         int64_t out, int64_t ov, int64_t n, int64_t stream) {
+    hipdf_i128_arith(op, P(a), P(b), P(av), P(bv), PM(out), PM(ov), n,
+                     S(stream));
+    check_async();
+  });
+  m.def("i128_cmp", [](int op, int64_t a, int64_t b, int64_t av, int64_t bv,
+                       int64_t out, int64_t ov, int64_t n, int64_t stream) {
+    hipdf_i128_cmp(op, P(a), P(b), P(av), P(bv), PM(out), PM(ov), n,
+                   S(stream));
+    check_async();
+  });
+  m.def("i64_to_i128", [](int64_t in, int64_t out, int64_t n, int64_t stream) {
+    hipdf_i64_to_i128(P(in), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("i128_to_f64", [](int64_t in, int64_t out, int64_t n, int64_t stream) {
+    hipdf_i128_to_f64(P(in), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("gb_sum_i64_to_i128", [](int64_t vals, int64_t vvalid, int64_t row_gid,
+                                 int64_t sel, int64_t acc, int64_t cnt,
+                                 int64_t n, int64_t stream) {
+    hipdf_gb_sum_i64_to_i128(P(vals), P(vvalid), P(row_gid), P(sel), PM(acc),
+                             PM(cnt), n, S(stream));
+    check_async();
+  });
+  m.def("gb_sum_i128", [](int64_t vals, int64_t vvalid, int64_t row_gid,
+                          int64_t sel, int64_t acc, int64_t cnt, int64_t n,
+                          int64_t stream) {
+    hipdf_gb_sum_i128(P(vals), P(vvalid), P(row_gid), P(sel), PM(acc),
+                      PM(cnt), n, S(stream));
     check_async();
   });
 

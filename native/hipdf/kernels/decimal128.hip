@@ -1,0 +1,229 @@
+// 128-bit decimal kernels (reference analogue: spark-rapids-jni Arithmetic
+// — decimal128 add/sub/compare — SURVEY.md §2.8B). Layout: interleaved
+// little-endian (lo, hi) int64 word pairs per row (arrow decimal128 LE).
+#include "hipdf_common.h"
+
+struct i128 {
+  uint64_t lo;
+  int64_t hi;
+};
+
+__device__ __forceinline__ i128 load128(const int64_t* p, int64_t i) {
+  return {(uint64_t)p[2 * i], p[2 * i + 1]};
+}
+
+__device__ __forceinline__ void store128(int64_t* p, int64_t i, i128 v) {
+  p[2 * i] = (int64_t)v.lo;
+  p[2 * i + 1] = v.hi;
+}
+
+__device__ __forceinline__ i128 add128(i128 a, i128 b) {
+  uint64_t lo = a.lo + b.lo;
+  int64_t carry = lo < a.lo ? 1 : 0;
+  return {lo, a.hi + b.hi + carry};
+}
+
+__device__ __forceinline__ i128 neg128(i128 a) {
+  uint64_t lo = ~a.lo + 1;
+  int64_t hi = ~a.hi + (lo == 0 ? 1 : 0);
+  return {lo, hi};
+}
+
+__device__ __forceinline__ int cmp128(i128 a, i128 b) {
+  if (a.hi != b.hi) return a.hi < b.hi ? -1 : 1;
+  if (a.lo != b.lo) return a.lo < b.lo ? -1 : 1;
+  return 0;
+}
+
+// ---- elementwise ---------------------------------------------------------
+
+// op: 0 add, 1 sub, 22 min, 23 max (matches BinOp ids used by python)
+__global__ void k_i128_arith(int op, const int64_t* __restrict__ a,
+                             const int64_t* __restrict__ b,
+                             const uint64_t* __restrict__ av,
+                             const uint64_t* __restrict__ bv,
+                             int64_t* __restrict__ out,
+                             uint64_t* __restrict__ ov, int64_t nstripe,
+                             int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t s = wave_global; s < nstripe; s += wave_count) {
+    int64_t row = s * WAVE + lane;
+    bool ok = false;
+    if (row < n) {
+      ok = valid_bit(av, row) && valid_bit(bv, row);
+      i128 x = load128(a, row), y = load128(b, row);
+      i128 r{0, 0};
+      if (ok) {
+        switch (op) {
+          case 0: r = add128(x, y); break;
+          case 1: r = add128(x, neg128(y)); break;
+          case 22: r = cmp128(x, y) <= 0 ? x : y; break;
+          case 23: r = cmp128(x, y) >= 0 ? x : y; break;
+        }
+      }
+      store128(out, row, r);
+    }
+    uint64_t ballot = __ballot(ok);
+    write_valid_word(ov, s, ballot, lane);
+  }
+}
+
+// cmp op ids match StrCmpOp: 0 eq 1 ne 2 lt 3 le 4 gt 5 ge
+__global__ void k_i128_cmp(int op, const int64_t* __restrict__ a,
+                           const int64_t* __restrict__ b,
+                           const uint64_t* __restrict__ av,
+                           const uint64_t* __restrict__ bv,
+                           uint8_t* __restrict__ out,
+                           uint64_t* __restrict__ ov, int64_t nstripe,
+                           int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t s = wave_global; s < nstripe; s += wave_count) {
+    int64_t row = s * WAVE + lane;
+    bool ok = false;
+    if (row < n) {
+      ok = valid_bit(av, row) && valid_bit(bv, row);
+      int c = cmp128(load128(a, row), load128(b, row));
+      bool r;
+      switch (op) {
+        case 0: r = c == 0; break;
+        case 1: r = c != 0; break;
+        case 2: r = c < 0; break;
+        case 3: r = c <= 0; break;
+        case 4: r = c > 0; break;
+        default: r = c >= 0; break;
+      }
+      out[row] = ok ? (uint8_t)r : 0;
+    }
+    uint64_t ballot = __ballot(ok);
+    write_valid_word(ov, s, ballot, lane);
+  }
+}
+
+// sign-extend int64 -> int128 pairs (decimal64 -> decimal128 widening)
+__global__ void k_i64_to_i128(const int64_t* __restrict__ in,
+                              int64_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t v = in[i];
+    out[2 * i] = v;
+    out[2 * i + 1] = v < 0 ? -1 : 0;
+  }
+}
+
+// int128 pairs -> double (for casts / mean)
+__global__ void k_i128_to_f64(const int64_t* __restrict__ in,
+                              double* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    i128 v = load128(in, i);
+    out[i] = (double)v.hi * 18446744073709551616.0 + (double)v.lo;
+  }
+}
+
+// grouped sum of int64 values into int128 accumulators (carry-correct
+// split atomics: lo add returns the old value, carry derived from wrap)
+__global__ void k_gb_sum_i64_to_i128(const int64_t* __restrict__ vals,
+                                     const uint64_t* __restrict__ vvalid,
+                                     const int32_t* __restrict__ row_gid,
+                                     const int32_t* __restrict__ sel,
+                                     int64_t* __restrict__ acc,  // 2*ngroups
+                                     int64_t* __restrict__ cnt,
+                                     int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
+    if (!valid_bit(vvalid, i)) continue;
+    int32_t g = row_gid[j];
+    int64_t v = vals[i];
+    uint64_t lo = (uint64_t)v;
+    int64_t hi = v < 0 ? -1 : 0;
+    unsigned long long old = atomicAdd((unsigned long long*)&acc[2 * g],
+                                       (unsigned long long)lo);
+    if (old + lo < old)  // wrapped: carry into the high word
+      atomicAdd((unsigned long long*)&acc[2 * g + 1], 1ull);
+    if (hi)
+      atomicAdd((unsigned long long*)&acc[2 * g + 1],
+                (unsigned long long)hi);
+    atomicAdd((unsigned long long*)&cnt[g], 1ull);
+  }
+}
+
+// grouped sum of int128 values into int128 accumulators
+__global__ void k_gb_sum_i128(const int64_t* __restrict__ vals,
+                              const uint64_t* __restrict__ vvalid,
+                              const int32_t* __restrict__ row_gid,
+                              const int32_t* __restrict__ sel,
+                              int64_t* __restrict__ acc,
+                              int64_t* __restrict__ cnt, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
+    if (!valid_bit(vvalid, i)) continue;
+    int32_t g = row_gid[j];
+    i128 v = load128(vals, i);
+    unsigned long long old = atomicAdd((unsigned long long*)&acc[2 * g],
+                                       (unsigned long long)v.lo);
+    if (old + v.lo < old)
+      atomicAdd((unsigned long long*)&acc[2 * g + 1], 1ull);
+    if (v.hi)
+      atomicAdd((unsigned long long*)&acc[2 * g + 1],
+                (unsigned long long)v.hi);
+    atomicAdd((unsigned long long*)&cnt[g], 1ull);
+  }
+}
+
+extern "C" {
+
+void hipdf_i128_arith(int op, const void* a, const void* b, const void* av,
+                      const void* bv, void* out, void* ov, int64_t n,
+                      hipStream_t stream) {
+  hipLaunchKernelGGL(k_i128_arith, stripe_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, op, (const int64_t*)a, (const int64_t*)b,
+                     (const uint64_t*)av, (const uint64_t*)bv, (int64_t*)out,
+                     (uint64_t*)ov, n_stripes(n), n);
+}
+
+void hipdf_i128_cmp(int op, const void* a, const void* b, const void* av,
+                    const void* bv, void* out, void* ov, int64_t n,
+                    hipStream_t stream) {
+  hipLaunchKernelGGL(k_i128_cmp, stripe_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     op, (const int64_t*)a, (const int64_t*)b,
+                     (const uint64_t*)av, (const uint64_t*)bv, (uint8_t*)out,
+                     (uint64_t*)ov, n_stripes(n), n);
+}
+
+void hipdf_i64_to_i128(const void* in, void* out, int64_t n,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(k_i64_to_i128, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int64_t*)in, (int64_t*)out, n);
+}
+
+void hipdf_i128_to_f64(const void* in, void* out, int64_t n,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(k_i128_to_f64, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int64_t*)in, (double*)out, n);
+}
+
+void hipdf_gb_sum_i64_to_i128(const void* vals, const void* vvalid,
+                              const void* row_gid, const void* sel, void* acc,
+                              void* cnt, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_sum_i64_to_i128, flat_grid(n, 4),
+                     dim3(HIPDF_BLOCK), 0, stream, (const int64_t*)vals,
+                     (const uint64_t*)vvalid, (const int32_t*)row_gid,
+                     (const int32_t*)sel, (int64_t*)acc, (int64_t*)cnt, n);
+}
+
+void hipdf_gb_sum_i128(const void* vals, const void* vvalid,
+                       const void* row_gid, const void* sel, void* acc,
+                       void* cnt, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_sum_i128, flat_grid(n, 4), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int64_t*)vals, (const uint64_t*)vvalid,
+                     (const int32_t*)row_gid, (const int32_t*)sel,
+                     (int64_t*)acc, (int64_t*)cnt, n);
+}
+
+}  // extern "C"

@@ -150,6 +150,11 @@ void hipdf_scatter_fixed(int esize, const void* vals, const void* idx,
                          0, stream, (const uint64_t*)vals,
                          (const int32_t*)idx, (uint64_t*)out, n);
       break;
+    case 16:
+      hipLaunchKernelGGL((k_scatter_fixed<ulonglong2>), grid,
+                         dim3(HIPDF_BLOCK), 0, stream, (const ulonglong2*)vals,
+                         (const int32_t*)idx, (ulonglong2*)out, n);
+      break;
     default:
       throw std::runtime_error("scatter: bad element size");
   }

@@ -43,6 +43,7 @@ enum HashKind : int {
   HK_FLOAT = 2,
   HK_DOUBLE = 3,
   HK_STRING = 4,
+  HK_I128 = 5,  // decimal128: chained hash_long of (lo, hi)
 };
 
 template <typename T>
@@ -56,6 +57,14 @@ __global__ void k_murmur3_col(int kind, const T* __restrict__ a,
     if (!valid_bit(av, i)) continue;  // null keeps previous hash
     uint32_t seed = (uint32_t)seeds[j];
     uint32_t h;
+    if (kind == HK_I128) {
+      // interleaved pairs; only instantiated with T = int64_t
+      uint64_t lo = (uint64_t)(int64_t)a[2 * i];
+      uint64_t hi = (uint64_t)(int64_t)a[2 * i + 1];
+      h = hash_long(hi, hash_long(lo, seed));
+      seeds[j] = (int32_t)h;
+      continue;
+    }
     T v = a[i];
     if (kind == HK_LONG) {
       h = hash_long((uint64_t)(int64_t)v, seed);

@@ -50,6 +50,11 @@ __device__ __forceinline__ bool keyval_equal(const KeyCol& c, int64_t ra,
       if (isnan(a) && isnan(b)) return true;
       return a == b;
     }
+    case 7: {  // int128 pairs (decimal128)
+      const int64_t* pa = (const int64_t*)c.data;
+      const int64_t* pb = (const int64_t*)cb.data;
+      return pa[2 * ra] == pb[2 * rb] && pa[2 * ra + 1] == pb[2 * rb + 1];
+    }
   }
   return false;
 }

@@ -84,7 +84,8 @@ __global__ void k_gather_fixed(const T* __restrict__ in,
   for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n_out;
        j += (int64_t)gridDim.x * blockDim.x) {
     int32_t i = idx[j];
-    out[j] = i >= 0 ? in[i] : (T)0;
+    T zero{};
+    out[j] = i >= 0 ? in[i] : zero;
   }
 }
 
@@ -136,6 +137,11 @@ __global__ void k_gather_table(const GatherCol* __restrict__ cols, int ncols,
           case 1: ((uint8_t*)g.out)[j] = i >= 0 ? ((const uint8_t*)g.in)[src] : 0; break;
           case 2: ((uint16_t*)g.out)[j] = i >= 0 ? ((const uint16_t*)g.in)[src] : 0; break;
           case 4: ((uint32_t*)g.out)[j] = i >= 0 ? ((const uint32_t*)g.in)[src] : 0; break;
+          case 16: {
+            ulonglong2 z{0, 0};
+            ((ulonglong2*)g.out)[j] = i >= 0 ? ((const ulonglong2*)g.in)[src] : z;
+            break;
+          }
           default: ((uint64_t*)g.out)[j] = i >= 0 ? ((const uint64_t*)g.in)[src] : 0; break;
         }
       }
@@ -263,6 +269,11 @@ void hipdf_gather_fixed(int esize, const void* in, const void* idx, void* out,
       hipLaunchKernelGGL((k_gather_fixed<uint64_t>), grid, dim3(HIPDF_BLOCK),
                          0, stream, (const uint64_t*)in, (const int32_t*)idx,
                          (uint64_t*)out, n_out);
+      break;
+    case 16:
+      hipLaunchKernelGGL((k_gather_fixed<ulonglong2>), grid,
+                         dim3(HIPDF_BLOCK), 0, stream, (const ulonglong2*)in,
+                         (const int32_t*)idx, (ulonglong2*)out, n_out);
       break;
     default:
       throw std::runtime_error("gather: bad element size");

@@ -39,7 +39,41 @@ _TORCH_DTYPES = {
 
 
 def torch_dtype(dt: DType) -> torch.dtype:
+    if dt.id is TypeId.DECIMAL128:
+        return torch.int64  # 2 words per row, interleaved (lo, hi) LE
     return _TORCH_DTYPES[dt.id]
+
+
+def dec128_pack(values, scale: int) -> np.ndarray:
+    """Python ints/floats -> interleaved (lo, hi) int64 pairs."""
+    import decimal
+
+    out = np.zeros(2 * len(values), dtype=np.uint64)
+    for i, v in enumerate(values):
+        if v is None:
+            continue
+        if isinstance(v, float):
+            u = int(round(v * (10 ** scale)))
+        elif isinstance(v, decimal.Decimal):
+            u = int(v.scaleb(scale))
+        else:
+            u = int(v)
+        u &= (1 << 128) - 1
+        out[2 * i] = u & 0xFFFFFFFFFFFFFFFF
+        out[2 * i + 1] = (u >> 64) & 0xFFFFFFFFFFFFFFFF
+    return out.view(np.int64)
+
+
+def dec128_unpack(arr: np.ndarray):
+    """Interleaved pairs -> python ints (signed 128-bit)."""
+    u = arr.view(np.uint64)
+    out = []
+    for i in range(len(u) // 2):
+        v = int(u[2 * i]) | (int(u[2 * i + 1]) << 64)
+        if v >= 1 << 127:
+            v -= 1 << 128
+        out.append(v)
+    return out
 
 
 def mask_nbytes(size: int) -> int:
@@ -174,6 +208,13 @@ class Column:
             )
             return col.to(device) if device != "cpu" else col
         valid = np.array([v is not None for v in values], dtype=bool)
+        if dtype.id is TypeId.DECIMAL128:
+            packed = dec128_pack(values, dtype.scale)
+            data = torch.from_numpy(packed.copy())
+            validity = make_validity(valid) if not valid.all() else None
+            col = Column(dtype, n, data, validity,
+                         null_count=int(n - valid.sum()) if validity is not None else 0)
+            return col.to(device) if device != "cpu" else col
         np_dt = dtype.numpy_dtype()
         fill = 0
         dense = np.array([v if v is not None else fill for v in values], dtype=np_dt)
@@ -197,13 +238,21 @@ class Column:
             return col.to(device) if device != "cpu" else col
         if dtype.id is TypeId.BOOL:
             value = int(bool(value))
+        if dtype.id is TypeId.DECIMAL128:
+            one = torch.from_numpy(dec128_pack([value], dtype.scale).copy())
+            data = one.repeat(size).to(device) if device != "cpu" \
+                else one.repeat(size)
+            return Column(dtype, size, data, None, null_count=0)
+        if dtype.id is TypeId.DECIMAL64 and isinstance(value, float):
+            value = int(round(value * (10 ** dtype.scale)))
         data = torch.full((size,), value, dtype=torch_dtype(dtype),
                           device=device)
         return Column(dtype, size, data, None, null_count=0)
 
     @staticmethod
     def nulls(dtype: DType, size: int, device: str = "cpu") -> "Column":
-        data = torch.zeros(size, dtype=torch_dtype(dtype), device=device)
+        nwords = 2 * size if dtype.id is TypeId.DECIMAL128 else size
+        data = torch.zeros(nwords, dtype=torch_dtype(dtype), device=device)
         validity = torch.zeros(mask_nbytes(size), dtype=torch.uint8, device=device)
         offsets = None
         if dtype.id is TypeId.STRING:
@@ -233,6 +282,14 @@ class Column:
                 else:
                     out.append(raw[offs[i]: offs[i + 1]].decode("utf-8"))
             return out
+        if self.dtype.id is TypeId.DECIMAL128:
+            import decimal
+
+            ints = dec128_unpack(self.data.cpu().numpy()[: 2 * self.size])
+            scale = self.dtype.scale
+            ctx = decimal.Context(prec=50)  # decimal128 needs > default 28
+            return [(decimal.Decimal(v).scaleb(-scale, ctx) if scale else v)
+                    if ok else None for v, ok in zip(ints, valid)]
         arr = self.to_numpy()
         if self.dtype.id is TypeId.BOOL:
             arr = arr.astype(bool)

@@ -104,6 +104,11 @@ def _arrow_array_to_column(arr, dtype: DType) -> Column:
         dense = raw[0::2].copy()
         return Column(dtype, n, torch.from_numpy(dense), validity,
                       null_count=arr.null_count)
+    if dtype.id is TypeId.DECIMAL128:
+        # arrow 16-byte LE values == our interleaved (lo, hi) pairs
+        raw = np.frombuffer(bufs[1], dtype=np.int64, count=2 * n).copy()
+        return Column(dtype, n, torch.from_numpy(raw), validity,
+                      null_count=arr.null_count)
     np_dt = dtype.numpy_dtype()
     vals = np.frombuffer(bufs[1], dtype=np_dt, count=n).copy()
     return Column(dtype, n, torch.from_numpy(vals), validity,
@@ -211,9 +216,14 @@ def _column_to_arrow(c: Column, dtype: DType):
         import decimal
 
         scale = dtype.scale
-        vals = [None if not ok else
-                decimal.Decimal(int(v)).scaleb(-scale)
-                for v, ok in zip(c.to_numpy(), valid)]
+        if dtype.id is TypeId.DECIMAL128:
+            from ..column import dec128_unpack
+
+            ints = dec128_unpack(c.data.cpu().numpy()[: 2 * c.size])
+        else:
+            ints = c.to_numpy()
+        vals = [None if not ok else decimal.Decimal(int(v)).scaleb(-scale)
+                for v, ok in zip(ints, valid)]
         return pa.array(vals, type=pa.decimal128(dtype.precision, scale))
     if dtype.id is TypeId.DATE32:
         return pa.array(c.to_numpy(), type=pa.date32(), mask=mask)

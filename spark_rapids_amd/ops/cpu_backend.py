@@ -31,6 +31,11 @@ from ..types import DType, TypeId
 def _vals(col: Column):
     if col.dtype.id is TypeId.STRING:
         return np.array(col.to_pylist(), dtype=object)
+    if col.dtype.id is TypeId.DECIMAL128:
+        from ..column import dec128_unpack
+
+        return np.array(dec128_unpack(
+            col.data.cpu().numpy()[: 2 * col.size]), dtype=object)
     return col.data.cpu().numpy()[: col.size]
 
 
@@ -39,6 +44,16 @@ def _valid(col: Column) -> np.ndarray:
 
 
 def _make(vals: np.ndarray, valid: Optional[np.ndarray], dtype: DType) -> Column:
+    if dtype.id is TypeId.DECIMAL128:
+        from ..column import dec128_pack
+
+        packed = dec128_pack([int(v) for v in vals], 0)
+        if valid is not None and not valid.all():
+            return Column(dtype, len(vals), torch.from_numpy(packed.copy()),
+                          make_validity(valid),
+                          null_count=int(len(vals) - valid.sum()))
+        return Column(dtype, len(vals), torch.from_numpy(packed.copy()),
+                      None, null_count=0)
     if dtype.id is TypeId.STRING:
         out = []
         for i, v in enumerate(vals):
@@ -123,6 +138,20 @@ def _binary_impl(op, a, av, b, bv, in_dtype: DType, out_dtype: DType) -> Column:
                 }[op]()
         return _bool_col(res.astype(np.uint8), valid if not valid.all() else None)
 
+    if out_dtype.id is TypeId.DECIMAL128:
+        if op == "add":
+            res = a + b
+        elif op == "sub":
+            res = a - b
+        elif op == "min":
+            res = np.array([x if x <= y else y for x, y in zip(a, b)],
+                           dtype=object)
+        elif op == "max":
+            res = np.array([x if x >= y else y for x, y in zip(a, b)],
+                           dtype=object)
+        else:
+            raise NotImplementedError(f"decimal128 op {op}")
+        return _make(res, valid if not valid.all() else None, out_dtype)
     np_out = out_dtype.numpy_dtype()
     with np.errstate(divide="ignore", invalid="ignore", over="ignore"):
         if op == "add":
@@ -286,6 +315,29 @@ def cast(col: Column, to: DType) -> Column:
             except (ValueError, TypeError):
                 valid[i] = False
         return _make(res, valid if not valid.all() else None, to)
+    if src.id is TypeId.DECIMAL128 or to.id is TypeId.DECIMAL128:
+        if src.is_decimal and to.id is TypeId.DECIMAL128:
+            shift = to.scale - src.scale
+            vals = np.array([int(v) * (10 ** shift) if shift >= 0 else
+                             int(v) // (10 ** -shift) for v in a],
+                            dtype=object)
+            return _make(vals, av if not av.all() else None, to)
+        if src.id is TypeId.DECIMAL128 and to.is_floating:
+            f = np.array([float(int(v)) / (10 ** src.scale) for v in a])
+            return _make(f.astype(to.numpy_dtype()),
+                         av if not av.all() else None, to)
+        if src.id is TypeId.DECIMAL128 and to.id is TypeId.DECIMAL64:
+            shift = to.scale - src.scale
+            vals = np.array([int(v) * (10 ** shift) if shift >= 0 else
+                             int(v) // (10 ** -shift) for v in a],
+                            dtype=object)
+            return _make(np.array([int(v) for v in vals], dtype=np.int64),
+                         av if not av.all() else None, to)
+        if to.id is TypeId.DECIMAL128:
+            scaled = np.array([int(round(float(v) * (10 ** to.scale)))
+                               for v in a], dtype=object)
+            return _make(scaled, av if not av.all() else None, to)
+        raise NotImplementedError(f"cast {src} -> {to}")
     if src.is_decimal and to.is_decimal:
         shift = to.scale - src.scale
         res = a * (10 ** shift) if shift >= 0 else _round_half_up_div(a, 10 ** (-shift))
@@ -519,6 +571,12 @@ def murmur3_hash(cols: List[Column], seed: int = 42) -> Column:
         if c.dtype.id is TypeId.STRING:
             nh = np.array([_hash_bytes_one((x or "").encode("utf-8"), int(s))
                            for x, s in zip(a, h)], dtype=np.int64)
+        elif c.dtype.id is TypeId.DECIMAL128:
+            lo = np.array([int(v) & 0xFFFFFFFFFFFFFFFF for v in a],
+                          dtype=np.uint64).astype(np.int64)
+            hi = np.array([(int(v) >> 64) & 0xFFFFFFFFFFFFFFFF for v in a],
+                          dtype=np.uint64).astype(np.int64)
+            nh = _hash_long(hi, _hash_long(lo, h))
         elif c.dtype.id in (TypeId.INT64, TypeId.TIMESTAMP, TypeId.DECIMAL64):
             nh = _hash_long(a, h)
         elif c.dtype.id is TypeId.FLOAT64:
@@ -579,10 +637,12 @@ def _group_codes(keys: List[Column]) -> Tuple[np.ndarray, np.ndarray]:
     for k in keys:
         a, av = _vals(k), _valid(k)
         if a.dtype == object:
-            # encode strings (None distinct)
-            uniq, inv = np.unique(np.array([x if x is not None else "\0\0NULL" for x in a]),
-                                  return_inverse=True)
+            # encode strings / int128 objects (None distinct)
+            uniq, inv = np.unique(
+                np.array([repr(x) if x is not None else "\0\0NULL"
+                          for x in a]), return_inverse=True)
             arrs.append(inv.astype(np.int64))
+            arrs.append(av.astype(np.int64))
         else:
             code = a.view(np.int64) if a.dtype.itemsize == 8 else a.astype(np.int64)
             # normalize values under NULLs so all-null keys form one group,
@@ -627,9 +687,24 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
             res = np.bincount(codes[av], minlength=ngroups).astype(np.int64)
             out_cols.append(_make(res, None, out_dtype))
             continue
+        if batch.columns[vidx].dtype.id is TypeId.DECIMAL128 and \
+                op not in ("sum", "count", "count_all"):
+            raise NotImplementedError(
+                f"{op} over decimal128 (float path would lose precision)")
         cnt = np.bincount(codes[av], minlength=ngroups)
         gvalid = cnt > 0
         gc = codes[av]
+        if op == "sum" and out_dtype.id is TypeId.DECIMAL128:
+            sums = [0] * ngroups
+            has = [False] * ngroups
+            for g, v, ok in zip(codes, a, av):
+                if ok:
+                    sums[g] += int(v)
+                    has[g] = True
+            gv = np.array(has)
+            out_cols.append(_make(np.array(sums, dtype=object),
+                                  gv if not gv.all() else None, out_dtype))
+            continue
         if op == "sum" and not out_dtype.is_floating:
             # integral/decimal sum: accumulate in int64 (wraps like Spark non-ANSI)
             s = np.zeros(ngroups, dtype=np.int64)
@@ -721,6 +796,20 @@ def sort_order(batch: ColumnBatch, key_idx: List[int],
         a, av = _vals(c), _valid(c)
         desc = descending[ri]
         nl = nulls_last[ri]
+        if a.dtype == object and c.dtype.id is TypeId.DECIMAL128:
+            ints = [int(x) if x is not None else 0 for x in a]
+            if desc:
+                ints = [-v - 1 for v in ints]
+            hi = np.array([v >> 64 for v in ints], dtype=np.int64)
+            lo = np.array([v & ((1 << 64) - 1) for v in ints],
+                          dtype=np.uint64).view(np.int64)
+            # unsigned-order lo as secondary: bias to signed
+            lo = (lo ^ np.int64(-0x8000000000000000))
+            null_rank = np.where(av, 0, 1 if nl else -1)
+            keys.append(lo)
+            keys.append(hi)
+            keys.append(null_rank)
+            continue
         if a.dtype == object:
             uniq, codes = np.unique(
                 np.array([x if x is not None else "" for x in a]), return_inverse=True)

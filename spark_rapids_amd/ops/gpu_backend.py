@@ -87,8 +87,10 @@ def _ht(dt: DType) -> int:
 
 
 def _alloc(n: int, dt: DType) -> torch.Tensor:
-    return torch.empty(max(n, 1), dtype=torch_dtype(dt), device="cuda")[:n] \
-        if n == 0 else torch.empty(n, dtype=torch_dtype(dt), device="cuda")
+    words = 2 * n if dt.id is TypeId.DECIMAL128 else n
+    if words == 0:
+        return torch.empty(1, dtype=torch_dtype(dt), device="cuda")[:0]
+    return torch.empty(words, dtype=torch_dtype(dt), device="cuda")
 
 
 def _alloc_mask(n: int) -> torch.Tensor:
@@ -101,6 +103,9 @@ def _empty_col(dtype: DType) -> Column:
                       None, torch.zeros(1, dtype=torch.int32, device="cuda"), 0)
     return Column(dtype, 0, torch.zeros(0, dtype=torch_dtype(dtype),
                                         device="cuda"), None, None, 0)
+
+
+_I128_ARITH = {"add": 0, "sub": 1, "min": 22, "max": 23}
 
 
 # ---------------------------------------------------------------------------
@@ -138,6 +143,27 @@ def _binary(op, lhs: Column, rhs: Optional[Column], scalar, out_dtype) -> Column
                     out.data_ptr(), n, s)
         # combine validities via the bool AND kernel on expanded masks
         return _with_and_validity(out, lhs, rhs, out_dtype, n, s)
+    if lhs.dtype.id is TypeId.DECIMAL128:
+        if scalar_rhs:
+            raise NotImplementedError("decimal128 scalar ops not on GPU")
+        ov = _alloc_mask(n)
+        if op in _I128_ARITH and op in ("add", "sub", "min", "max"):
+            out = _alloc(n, out_dtype)
+            ext.i128_arith(_I128_ARITH[op], lhs.data.data_ptr(),
+                           rhs.data.data_ptr(), _ptr(lhs.validity),
+                           _ptr(rhs.validity), out.data_ptr(), ov.data_ptr(),
+                           n, s)
+            return Column(out_dtype, n, out, ov, null_count=None)
+        if op in _STR_CMP:
+            out = _alloc(n, out_dtype)
+            ext.i128_cmp(_STR_CMP[op], lhs.data.data_ptr(),
+                         rhs.data.data_ptr(), _ptr(lhs.validity),
+                         _ptr(rhs.validity), out.data_ptr(), ov.data_ptr(),
+                         n, s)
+            has_valid = lhs.validity is not None or rhs.validity is not None
+            return Column(out_dtype, n, out, ov if has_valid else None,
+                          null_count=None if has_valid else 0)
+        raise NotImplementedError(f"decimal128 op {op} not on GPU")
     t = _ht(lhs.dtype)
     sd, si = 0.0, 0
     if scalar_rhs and scalar is not None:
@@ -294,6 +320,30 @@ def cast(col: Column, to: DType) -> Column:
 def _cast_decimal(col: Column, to: DType, v) -> Column:
     n = col.size
     s = _stream()
+    if col.dtype.id is TypeId.DECIMAL128 or to.id is TypeId.DECIMAL128:
+        if col.dtype.id is TypeId.DECIMAL64 and to.id is TypeId.DECIMAL128:
+            src = col
+            if to.scale != col.dtype.scale:
+                src = _cast_decimal(col, DType.decimal(18, to.scale), v)
+            out = _alloc(n, to)
+            ext.i64_to_i128(src.data.data_ptr(), out.data_ptr(), n, s)
+            return Column(to, n, out, v, null_count=col._null_count)
+        if col.dtype.id is TypeId.DECIMAL128 and to.is_floating:
+            dbl = torch.empty(max(n, 1), dtype=torch.float64,
+                              device="cuda")[:n]
+            if n:
+                ext.i128_to_f64(col.data.data_ptr(), dbl.data_ptr(), n, s)
+            c = Column(DType.float64(), n, dbl, v, null_count=col._null_count)
+            scaled = binary_op_scalar("div", c, float(10 ** col.dtype.scale),
+                                      DType.float64())
+            scaled = Column(DType.float64(), n, scaled.data, v,
+                            null_count=col._null_count)
+            return cast(scaled, to) if to.id is not TypeId.FLOAT64 else scaled
+        if col.dtype.id is TypeId.DECIMAL128 and to.id is TypeId.DECIMAL128 \
+                and to.scale == col.dtype.scale:
+            return Column(to, n, col.data.clone(), v,
+                          null_count=col._null_count)
+        raise NotImplementedError(f"gpu cast {col.dtype} -> {to}")
     if col.dtype.is_decimal and to.is_decimal:
         shift = to.scale - col.dtype.scale
         out = _alloc(n, to)
@@ -544,6 +594,9 @@ def _murmur3_tensor(cols: List[Column], seed: int,
         if c.dtype.id is TypeId.STRING:
             ext.murmur3_str(c.offsets.data_ptr(), c.data.data_ptr(),
                             _ptr(c.validity), selp, seeds.data_ptr(), n, s)
+        elif c.dtype.id is TypeId.DECIMAL128:
+            ext.murmur3_col(5, 4, c.data.data_ptr(), _ptr(c.validity), selp,
+                            seeds.data_ptr(), n, s)
         else:
             ext.murmur3_col(_HASH_KIND[c.dtype.id], _ht(c.dtype),
                             c.data.data_ptr(), _ptr(c.validity), selp,
@@ -636,6 +689,9 @@ def _key_desc(cols: List[Column]) -> torch.Tensor:
         if c.dtype.id is TypeId.STRING:
             blobs.append(struct.pack("<iiqqq", 0, 1, c.offsets.data_ptr(),
                                      _ptr(c.validity), c.data.data_ptr()))
+        elif c.dtype.id is TypeId.DECIMAL128:
+            blobs.append(struct.pack("<iiqqq", 7, 0, c.data.data_ptr(),
+                                     _ptr(c.validity), 0))
         else:
             blobs.append(struct.pack("<iiqqq", _ht(c.dtype), 0,
                                      c.data.data_ptr(), _ptr(c.validity), 0))
@@ -696,11 +752,29 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                                ngroups, maybe_negative=False) if keys else None
     out_cols = list(key_batch.columns) if key_batch is not None else []
 
-    # fused multi-aggregate: one kernel pass accumulates every agg
+    # fused multi-aggregate: one kernel pass accumulates every agg;
+    # decimal128 sums go through dedicated carry-correct i128 kernels
     allocs = []
     blobs = []
     for op, vidx, out_dtype in aggs:
         vc = batch.columns[vidx] if vidx >= 0 else None
+        if op == "sum" and out_dtype.id is TypeId.DECIMAL128:
+            acc = torch.zeros(2 * max(ngroups, 1), dtype=torch.int64,
+                              device="cuda")
+            cnt = torch.zeros(max(ngroups, 1), dtype=torch.int64,
+                              device="cuda")
+            if n:
+                if vc.dtype.id is TypeId.DECIMAL128:
+                    ext.gb_sum_i128(vc.data.data_ptr(), _ptr(vc.validity),
+                                    row_gid.data_ptr(), selp, acc.data_ptr(),
+                                    cnt.data_ptr(), n, s)
+                else:
+                    ext.gb_sum_i64_to_i128(
+                        vc.data.data_ptr(), _ptr(vc.validity),
+                        row_gid.data_ptr(), selp, acc.data_ptr(),
+                        cnt.data_ptr(), n, s)
+            allocs.append(("sum_d128", out_dtype, False, acc, cnt))
+            continue
         acc_is_double = out_dtype.is_floating or (
             vc is not None and vc.dtype.is_floating)
         acc = torch.empty(max(ngroups, 1),
@@ -722,6 +796,12 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         ext.gb_agg_multi(desc.data_ptr(), len(aggs), row_gid.data_ptr(),
                          selp, ngroups, n, s)
     for op, out_dtype, acc_is_double, acc, cnt in allocs:
+        if op == "sum_d128":
+            ov = _alloc_mask(ngroups)
+            ext.mask_from_nonzero(cnt.data_ptr(), ov.data_ptr(), ngroups, s)
+            out_cols.append(Column(out_dtype, ngroups, acc[:2 * ngroups],
+                                   ov, null_count=None))
+            continue
         if op in ("count", "count_all"):
             out_cols.append(Column(out_dtype, ngroups, cnt[:ngroups].clone(),
                                    None, null_count=0))

@@ -36,9 +36,11 @@ _FIXED_KEYS = TypeSig({
 # which handle strings too (hash.hip murmur3_str + keys.h byte compare)
 _HASH_KEYS = TypeSig({
     TypeId.BOOL, TypeId.INT8, TypeId.INT16, TypeId.INT32, TypeId.INT64,
-    TypeId.FLOAT32, TypeId.FLOAT64, TypeId.DECIMAL64, TypeId.DATE32,
-    TypeId.TIMESTAMP, TypeId.STRING,
+    TypeId.FLOAT32, TypeId.FLOAT64, TypeId.DECIMAL64, TypeId.DECIMAL128,
+    TypeId.DATE32, TypeId.TIMESTAMP, TypeId.STRING,
 })
+_D128_BINARY_OK = {"add", "sub", "min", "max", "eq", "ne", "lt", "le",
+                   "gt", "ge"}
 _NUMERIC = TypeSig.numeric()
 
 _GPU_BINARY_OPS = {
@@ -90,6 +92,8 @@ class Tagger:
                 out.append(f"binary op {e.op} has no GPU kernel")
             elif in_t.id is TypeId.STRING and e.op not in _GPU_STRING_OK:
                 out.append(f"binary op {e.op} on string not on GPU yet")
+            elif in_t.id is TypeId.DECIMAL128 and e.op not in _D128_BINARY_OK:
+                out.append(f"binary op {e.op} on decimal128 not on GPU yet")
             if not self.conf.expr_enabled(e.op):
                 out.append(f"expression {e.op} disabled by conf")
         elif isinstance(e, UnaryExpr):
@@ -143,6 +147,10 @@ class Tagger:
                     r = _NUMERIC.supports(t)
                     if r and a.op not in ("count", "count_all", "min", "max"):
                         reasons.append(f"agg {a.op}({a.child}): {r}")
+                    if t.id is TypeId.DECIMAL128 and a.op not in (
+                            "sum", "count", "count_all"):
+                        reasons.append(
+                            f"agg {a.op} over decimal128 not supported yet")
                     reasons += self.expr_reasons(a.child, cs)
         elif isinstance(node, L.Join):
             ls, rs = node.left.schema(), node.right.schema()
@@ -177,7 +185,8 @@ class Tagger:
                         reasons.append("lag/lead string default not on GPU")
                     continue
                 if op in ("sum", "count", "mean"):
-                    if vt is not None and not vt.is_numeric:
+                    if vt is not None and (not vt.is_numeric
+                                           or vt.id is TypeId.DECIMAL128):
                         reasons.append(f"window {op}({vt}) not on GPU")
                     continue
                 if op in ("min", "max"):

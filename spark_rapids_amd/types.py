@@ -206,6 +206,9 @@ def promote(a: DType, b: DType) -> DType:
     """Common wider type for arithmetic between a and b (non-decimal path)."""
     if a == b:
         return a
+    if a.is_decimal and b.is_decimal:
+        return DType.decimal(max(a.precision, b.precision),
+                             max(a.scale, b.scale))
     if a.is_decimal or b.is_decimal:
         # decimal + integral -> decimal with enough precision; handled by caller
         d = a if a.is_decimal else b
