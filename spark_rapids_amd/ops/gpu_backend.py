@@ -790,10 +790,10 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
             _ptr(vc.data if vc is not None else None),
             _ptr(vc.validity if vc is not None else None),
             acc.data_ptr(), cnt.data_ptr()))
-    if aggs:
+    if blobs:
         desc = torch.frombuffer(bytearray(b"".join(blobs)),
                                 dtype=torch.uint8).cuda()
-        ext.gb_agg_multi(desc.data_ptr(), len(aggs), row_gid.data_ptr(),
+        ext.gb_agg_multi(desc.data_ptr(), len(blobs), row_gid.data_ptr(),
                          selp, ngroups, n, s)
     for op, out_dtype, acc_is_double, acc, cnt in allocs:
         if op == "sum_d128":
