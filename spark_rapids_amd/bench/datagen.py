@@ -12,6 +12,8 @@ import numpy as np
 from ..column import Column, ColumnBatch, Field, Schema
 from ..types import DATE32, FLOAT64, INT8, INT16, INT32, INT64, DType
 
+DEC72 = DType.decimal(7, 2)
+
 N_ITEMS = 102_000
 N_STORES = 1_002
 N_CUSTOMERS = 1_000_000
@@ -26,9 +28,9 @@ def fact_schema() -> Schema:
         Field("ss_customer_id", INT32),
         Field("ss_promo", INT8),
         Field("ss_quantity", INT32),
-        Field("ss_wholesale_cost", FLOAT64),
-        Field("ss_list_price", FLOAT64),
-        Field("ss_sales_price", FLOAT64),
+        Field("ss_wholesale_cost", DEC72),
+        Field("ss_list_price", DEC72),
+        Field("ss_sales_price", DEC72),
         Field("ss_discount", FLOAT64),
     ])
 
@@ -49,9 +51,10 @@ def gen_fact_partition(rows: int, seed: int) -> ColumnBatch:
         Column.from_numpy(rng.integers(0, N_CUSTOMERS, rows).astype(np.int32), INT32),
         Column.from_numpy(rng.integers(0, 4, rows).astype(np.int8), INT8),
         Column.from_numpy(qty, INT32),
-        Column.from_numpy(wholesale, FLOAT64),
-        Column.from_numpy(list_price, FLOAT64),
-        Column.from_numpy(sales_price, FLOAT64, price_valid),
+        Column.from_numpy(np.round(wholesale * 100).astype(np.int64), DEC72),
+        Column.from_numpy(np.round(list_price * 100).astype(np.int64), DEC72),
+        Column.from_numpy(np.round(sales_price * 100).astype(np.int64), DEC72,
+                          price_valid),
         Column.from_numpy(discount, FLOAT64),
     ]
     return ColumnBatch(cols, rows)

@@ -18,12 +18,13 @@ from ..types import FLOAT64, INT64
 
 def q1_pricing_summary(t: Dict[str, DataFrame]) -> DataFrame:
     ss = t["store_sales"]
+    price_f = col("ss_sales_price").cast(FLOAT64)
     return (ss.filter(col("ss_sold_date") <= lit(10_900))
             .group_by("ss_promo")
             .agg(sum_(col("ss_quantity")),
-                 sum_(col("ss_sales_price")),
-                 sum_(col("ss_sales_price") * (lit(1.0) - col("ss_discount"))),
-                 avg(col("ss_list_price")),
+                 sum_(col("ss_sales_price")),      # decimal sum -> decimal128
+                 sum_(price_f * (lit(1.0) - col("ss_discount"))),
+                 avg(col("ss_list_price")),        # decimal mean via double
                  avg(col("ss_discount")),
                  count_star()))
 
@@ -33,7 +34,7 @@ def q2_join_agg(t: Dict[str, DataFrame]) -> DataFrame:
     return (ss.join(item, on="ss_item_id", right_on=["i_item_id"])
             .filter(col("i_category") < 3)
             .group_by("ss_store_id")
-            .agg(sum_(col("ss_sales_price")), count_star()))
+            .agg(sum_(col("ss_sales_price")), count_star()))  # decimal sum
 
 
 def q3_selective_revenue(t: Dict[str, DataFrame]) -> DataFrame:
@@ -43,14 +44,16 @@ def q3_selective_revenue(t: Dict[str, DataFrame]) -> DataFrame:
                       & (col("ss_discount") >= 0.05)
                       & (col("ss_discount") <= 0.07)
                       & (col("ss_quantity") < 24))
-            .agg(sum_(col("ss_list_price") * col("ss_discount"))))
+            .agg(sum_(col("ss_list_price").cast(FLOAT64)
+                      * col("ss_discount"))))
 
 
 def q4_customer_rollup(t: Dict[str, DataFrame]) -> DataFrame:
     ss = t["store_sales"]
-    return (ss.group_by("ss_customer_id")
-            .agg(sum_(col("ss_sales_price")), count_star())
-            .filter(col("sum(ss_sales_price)") > 4000.0)
+    return (ss.with_column("price_f", col("ss_sales_price").cast(FLOAT64))
+            .group_by("ss_customer_id")
+            .agg(sum_(col("price_f")), count_star())
+            .filter(col("sum(price_f)") > 4000.0)
             .agg(count_star()))
 
 
@@ -60,7 +63,8 @@ def q5_store_join(t: Dict[str, DataFrame]) -> DataFrame:
             .join(item, on="ss_item_id", right_on=["i_item_id"])
             .filter(col("s_state") < 25)
             .group_by("s_state", "i_category")
-            .agg(sum_(col("ss_sales_price") * col("ss_quantity").cast(FLOAT64)),
+            .agg(sum_(col("ss_sales_price").cast(FLOAT64)
+                      * col("ss_quantity").cast(FLOAT64)),
                  count_star()))
 
 

@@ -211,6 +211,11 @@ class BinaryExpr(Expression):
         return (self.left, self.right)
 
     def _common(self, lt: DType, rt: DType) -> DType:
+        if (lt.is_decimal or rt.is_decimal) and self.op in (
+                "mul", "div", "int_div", "mod", "pmod", "pow"):
+            raise NotImplementedError(
+                f"decimal {self.op} needs scale arithmetic (not implemented "
+                "yet): cast to double first, e.g. col.cast(FLOAT64)")
         if lt.id is TypeId.NULL:
             return rt
         if rt.id is TypeId.NULL:

@@ -208,10 +208,16 @@ def _lower_aggs(aggs: List[AggExpr], in_schema: Schema):
             merge.append("sum" if a.op == "sum" else a.op)
             final.append(("col", j))
         elif a.op == "mean":
-            value_exprs.append(a.child)
+            ct = a.child.dtype(in_schema)
+            if ct.is_decimal:
+                # decimal mean: compute over the scaled double value
+                from ..expr.expressions import CastExpr
+
+                value_exprs.append(CastExpr(a.child, FLOAT64))
+            else:
+                value_exprs.append(a.child)
             v = len(value_exprs) - 1
             js = len(partial)
-            ct = a.child.dtype(in_schema)
             sum_t = FLOAT64 if (ct.is_floating or ct.is_decimal) else INT64
             partial.append(("sum", v, sum_t))
             partial.append(("count", v, INT64))
