@@ -167,7 +167,7 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": (speedup / 3.0) if speedup is not None else None,
-            "dtype": "fp64",
+            "dtype": "decimal(7,2)+fp64",
             "data": "synthetic",
             "config": {
                 "model": "nds_like_power_run_5q",

@@ -144,6 +144,11 @@ PARQUET_ENABLED = bool_conf(
 PARQUET_MT_THREADS = int_conf(
     "spark.rapids.sql.format.parquet.multiThreadedRead.numThreads", 4,
     "Threads in the multithreaded parquet prefetch pool.")
+BROADCAST_THRESHOLD = bytes_conf(
+    "spark.rapids.sql.join.broadcastThreshold", 512 << 20,
+    "Distributed joins all-gather (broadcast) the build side when its "
+    "global size is below this; larger builds hash-exchange BOTH sides "
+    "across ranks instead (shuffled hash join).")
 SHUFFLE_MODE = str_conf(
     "spark.rapids.shuffle.mode", "MULTITHREADED",
     "Shuffle transport: MULTITHREADED (host staging) or RCCL (device-to-device "

@@ -282,9 +282,12 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
                                    input_replicated=L.is_replicated(node.child),
                                    merge_target_bytes=conf.get(BATCH_SIZE_BYTES))
     if isinstance(node, L.Join):
+        from ..config import BROADCAST_THRESHOLD
+
         return P.HashJoinExec(device, kids[0], kids[1], node.left_on,
                               node.right_on, node.how, node.schema(),
-                              right_replicated=L.is_replicated(node.right))
+                              right_replicated=L.is_replicated(node.right),
+                              broadcast_threshold=conf.get(BROADCAST_THRESHOLD))
     if isinstance(node, L.MapBatches):
         return P.MapBatchesExec(node.fn, _ensure_device(kids[0], "cpu"),
                                 node.schema())

@@ -379,32 +379,61 @@ class HashJoinExec(PhysicalExec):
 
     def __init__(self, device: str, left: PhysicalExec, right: PhysicalExec,
                  left_on: List[str], right_on: List[str], how: str,
-                 schema: Schema, right_replicated: bool = True):
+                 schema: Schema, right_replicated: bool = True,
+                 broadcast_threshold: int = 512 << 20):
         super().__init__(device, schema, [left, right])
         self.left_on = left_on
         self.right_on = right_on
         self.how = how
         self.right_replicated = right_replicated
+        self.broadcast_threshold = broadcast_threshold
+        self._strategy = "local"
+
+    def _local_or_empty(self, batches: List[ColumnBatch], schema: Schema):
+        if batches:
+            return ops.concat_batches(batches) if len(batches) > 1 \
+                else batches[0]
+        empty = ColumnBatch(
+            [Column.from_pylist([], f.dtype) for f in schema.fields], 0)
+        return empty.cuda() if self.gpu else empty
 
     def execute(self) -> Iterator[ColumnBatch]:
         left, right = self.children
         rbatches = list(right.execute())
-        # distributed: a sharded build side must be broadcast (all-gathered)
-        # so every rank probes against the full table (broadcast-join; the
-        # shuffled-join strategy for large build sides is a later round)
+        lbatches_override = None
+        # distributed strategies for a sharded build side (every rank must
+        # take the same branch: the decision uses the all-gathered global
+        # size). Small build -> broadcast (all-gather); large build ->
+        # hash-exchange BOTH sides so each rank joins one key range
+        # (reference analogues: GpuBroadcastHashJoinExec vs
+        # GpuShuffledHashJoinExec).
         from ..shuffle import dist as _dist
         if _dist.ctx().is_multi and not self.right_replicated:
-            from ..shuffle.exchange import gather_all
-            if rbatches:
-                local = ops.concat_batches(rbatches) if len(rbatches) > 1 \
-                    else rbatches[0]
+            import torch as _torch
+            import torch.distributed as _td
+
+            from ..shuffle.exchange import exchange_by_hash, gather_all
+
+            local = self._local_or_empty(rbatches, right.schema)
+            dev = "cuda" if self.gpu else "cpu"
+            sz = _torch.tensor([local.nbytes], dtype=_torch.int64, device=dev)
+            _td.all_reduce(sz)
+            total = int(sz.item())
+            lkidx = [left.schema.index(k) for k in self.left_on]
+            rkidx = [right.schema.index(k) for k in self.right_on]
+            same_key_types = all(
+                left.schema.fields[a].dtype.id == right.schema.fields[b].dtype.id
+                for a, b in zip(lkidx, rkidx))
+            if total <= self.broadcast_threshold or not same_key_types:
+                self._strategy = "broadcast"
+                rbatches = [b for b in gather_all(local) if b.num_rows]
             else:
-                local = ColumnBatch(
-                    [Column.from_pylist([], f.dtype) for f in
-                     right.schema.fields], 0)
-                if self.gpu:
-                    local = local.cuda()
-            rbatches = [b for b in gather_all(local) if b.num_rows]
+                self._strategy = "shuffled"
+                rbatches = [b for b in exchange_by_hash(local, rkidx)
+                            if b.num_rows]
+                lall = self._local_or_empty(list(left.execute()), left.schema)
+                lbatches_override = [b for b in exchange_by_hash(lall, lkidx)
+                                     if b.num_rows]
         if not rbatches:
             if self.how in ("inner", "semi"):
                 return
@@ -413,7 +442,9 @@ class HashJoinExec(PhysicalExec):
             rtable = ops.concat_batches(rbatches) if len(rbatches) > 1 else rbatches[0]
         lkidx = [left.schema.index(k) for k in self.left_on]
         rkidx = [right.schema.index(k) for k in self.right_on]
-        for lbatch in left.execute():
+        lsource = lbatches_override if lbatches_override is not None \
+            else left.execute()
+        for lbatch in lsource:
             if lbatch.num_rows == 0:
                 continue
             if rtable is None or rtable.num_rows == 0:

@@ -82,6 +82,21 @@ def main():
     j2 = df.join(rep, on="k").agg(count_star()).collect()
     assert j2[0][0] == n_total, j2
 
+    # shuffled-hash-join path: force it with a tiny broadcast threshold
+    s2 = Session({"spark.rapids.sql.enabled": False,
+                  "spark.rapids.sql.join.broadcastThreshold": 1})
+    df2 = s2.create_dataframe({
+        "k": (rows % 97), "v": rows.astype(np.float64)}, num_partitions=2)
+    right2 = s2.create_dataframe({"k": right_rows % 97, "r": right_rows * 100})
+    j3 = df2.join(right2, on="k").agg(count_star()).collect()
+    assert j3[0][0] == exp, (j3, exp, "shuffled join path")
+    left_join = df2.join(right2, on="k", how="left").agg(count_star()).collect()
+    # left join rows = matched pairs + unmatched left rows
+    unmatched = sum(1 for v in rows.tolist() if rc[v % 97] == 0)
+    # per-rank left rows only; aggregate over ranks is global
+    gexp = int(sum(rc[v % 97] if rc[v % 97] else 1 for v in all_rows.tolist()))
+    assert left_join[0][0] == gexp, (left_join, gexp)
+
     td.barrier()
     if rank == 0:
         print("DIST_OK")
