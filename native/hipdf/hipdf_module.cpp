@@ -76,7 +76,7 @@ void hipdf_join_count(int, const void*, const void*, const void*, int,
                       hipStream_t);
 void hipdf_join_fill(int, const void*, const void*, const void*, int,
                      const void*, const void*, int64_t, const void*, void*,
-                     void*, int64_t, hipStream_t);
+                     void*, void*, int64_t, hipStream_t);
 int64_t part_num_blocks(int64_t);
 int64_t sort_num_blocks(int64_t);
 void hipdf_i128_arith(int, const void*, const void*, const void*, const void*,
@@ -352,9 +352,10 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("join_fill", [](int how, int64_t lh, int64_t lk, int64_t rk,
                         int nkeys, int64_t head, int64_t next, int64_t cap,
                         int64_t offsets, int64_t lmap, int64_t rmap,
-                        int64_t n, int64_t stream) {
+                        int64_t right_matched, int64_t n, int64_t stream) {
     hipdf_join_fill(how, P(lh), P(lk), P(rk), nkeys, P(head), P(next), cap,
-                    P(offsets), PM(lmap), PM(rmap), n, S(stream));
+                    P(offsets), PM(lmap), PM(rmap), PM(right_matched), n,
+                    S(stream));
     check_async();
   });
 

@@ -10,7 +10,7 @@
 #include "hipdf_common.h"
 #include "keys.h"
 
-enum JoinHow : int { J_INNER = 0, J_LEFT, J_SEMI, J_ANTI };
+enum JoinHow : int { J_INNER = 0, J_LEFT, J_SEMI, J_ANTI, J_FULL };
 
 __global__ void k_join_build(const int32_t* __restrict__ hashes,
                              const KeyCol* __restrict__ keys, int nkeys,
@@ -48,7 +48,7 @@ __global__ void k_join_count(int how, const int32_t* __restrict__ lhashes,
     int64_t c;
     switch (how) {
       case J_INNER: c = matches; break;
-      case J_LEFT: c = matches ? matches : 1; break;
+      case J_LEFT: case J_FULL: c = matches ? matches : 1; break;
       case J_SEMI: c = matches ? 1 : 0; break;
       default: c = matches ? 0 : 1; break;  // anti
     }
@@ -65,7 +65,8 @@ __global__ void k_join_fill(int how, const int32_t* __restrict__ lhashes,
                             uint32_t slot_mask,
                             const int64_t* __restrict__ offsets,
                             int32_t* __restrict__ lmap,
-                            int32_t* __restrict__ rmap, int64_t n) {
+                            int32_t* __restrict__ rmap,
+                            uint8_t* __restrict__ right_matched, int64_t n) {
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     int64_t off = offsets[i];
@@ -75,9 +76,10 @@ __global__ void k_join_fill(int how, const int32_t* __restrict__ lhashes,
       for (int32_t r = head[slot]; r != -1; r = next[r]) {
         if (rows_equal(lkeys, rkeys, nkeys, i, r)) {
           ++matches;
-          if (how == J_INNER || how == J_LEFT) {
+          if (how == J_INNER || how == J_LEFT || how == J_FULL) {
             lmap[off] = (int32_t)i;
             rmap[off] = r;
+            if (right_matched) right_matched[r] = 1;
             ++off;
           } else {
             break;
@@ -85,7 +87,7 @@ __global__ void k_join_fill(int how, const int32_t* __restrict__ lhashes,
         }
       }
     }
-    if (how == J_LEFT && matches == 0) {
+    if ((how == J_LEFT || how == J_FULL) && matches == 0) {
       lmap[off] = (int32_t)i;
       rmap[off] = -1;  // null right row
     } else if (how == J_SEMI && matches) {
@@ -120,13 +122,14 @@ void hipdf_join_count(int how, const void* lhashes, const void* lkeys,
 void hipdf_join_fill(int how, const void* lhashes, const void* lkeys,
                      const void* rkeys, int nkeys, const void* head,
                      const void* next, int64_t cap, const void* offsets,
-                     void* lmap, void* rmap, int64_t n, hipStream_t stream) {
+                     void* lmap, void* rmap, void* right_matched, int64_t n,
+                     hipStream_t stream) {
   hipLaunchKernelGGL(k_join_fill, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
                      how, (const int32_t*)lhashes, (const KeyCol*)lkeys,
                      (const KeyCol*)rkeys, nkeys, (const int32_t*)head,
                      (const int32_t*)next, (uint32_t)(cap - 1),
                      (const int64_t*)offsets, (int32_t*)lmap, (int32_t*)rmap,
-                     n);
+                     (uint8_t*)right_matched, n);
 }
 
 }  // extern "C"

@@ -102,11 +102,12 @@ def sort_order(batch: ColumnBatch, key_idx: Sequence[int],
 
 def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
                      left_keys: Sequence[int], right_keys: Sequence[int],
-                     how: str) -> Tuple[Column, Optional[Column]]:
+                     how: str, right_matched=None) -> Tuple[Column, Optional[Column]]:
     """Equi-join gather maps (left_map, right_map). For semi/anti only
-    left_map is returned."""
+    left_map is returned; full outer also records matched build rows into
+    right_matched."""
     return backend_for(*left.columns, *right.columns).join_gather_maps(
-        left, right, list(left_keys), list(right_keys), how)
+        left, right, list(left_keys), list(right_keys), how, right_matched)
 
 
 def concat_batches(batches: Sequence[ColumnBatch]) -> ColumnBatch:

@@ -740,7 +740,8 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
 
 
 def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
-                     left_keys: List[int], right_keys: List[int], how: str):
+                     left_keys: List[int], right_keys: List[int], how: str,
+                     right_matched=None):
     import pandas as pd
 
     lk = {f"k{i}": _key_series(left.columns[c]) for i, c in enumerate(left_keys)}
@@ -755,11 +756,14 @@ def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
         m = ldf_nn.merge(rdf_nn, on=on, how="inner")
         return (_make(m["_l"].to_numpy().astype(np.int32), None, DType.int32()),
                 _make(m["_r"].to_numpy().astype(np.int32), None, DType.int32()))
-    if how == "left":
+    if how in ("left", "full"):
         m = ldf.merge(rdf_nn, on=on, how="left")
         lmap = m["_l"].to_numpy().astype(np.int32)
         rvals = m["_r"].to_numpy()
         rmap = np.where(np.isnan(rvals), -1, np.nan_to_num(rvals)).astype(np.int32)
+        if how == "full" and right_matched is not None:
+            hit = rmap[rmap >= 0]
+            right_matched[hit] = True
         return (_make(lmap, None, DType.int32()),
                 _make(rmap, None, DType.int32()))
     if how in ("semi", "anti"):

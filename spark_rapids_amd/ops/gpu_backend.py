@@ -58,7 +58,7 @@ _UN_OPS = {
 _HK_INT, _HK_LONG, _HK_FLOAT, _HK_DOUBLE = 0, 1, 2, 3
 _RED = {"sum": 0, "min": 1, "max": 2, "count": 3}
 _GB = {"sum": 0, "min": 1, "max": 2, "count": 3, "count_all": 4}
-_JOIN = {"inner": 0, "left": 1, "semi": 2, "anti": 3}
+_JOIN = {"inner": 0, "left": 1, "semi": 2, "anti": 3, "full": 4}
 
 
 _cached_stream: Optional[int] = None
@@ -826,7 +826,8 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
 # ---------------------------------------------------------------------------
 
 def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
-                     left_keys: List[int], right_keys: List[int], how: str):
+                     left_keys: List[int], right_keys: List[int], how: str,
+                     right_matched: Optional[torch.Tensor] = None):
     s = _stream()
     nl, nr = left.num_rows, right.num_rows
     lk = [left.columns[i] for i in left_keys]
@@ -847,12 +848,12 @@ def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
     offsets, total = _exclusive_scan_i64(counts)
     lmap = torch.empty(max(total, 1), dtype=torch.int32, device="cuda")[:total]
     rmap = torch.empty(max(total, 1), dtype=torch.int32, device="cuda")[:total] \
-        if how in ("inner", "left") else None
+        if how in ("inner", "left", "full") else None
     if total:
         ext.join_fill(_JOIN[how], lh.data_ptr(), ldesc.data_ptr(),
                       rdesc.data_ptr(), len(lk), head.data_ptr(),
                       nxt.data_ptr(), cap, offsets.data_ptr(),
-                      lmap.data_ptr(), _ptr(rmap), nl, s)
+                      lmap.data_ptr(), _ptr(rmap), _ptr(right_matched), nl, s)
     lcol = Column(DType.int32(), total, lmap, None, null_count=0)
     rcol = Column(DType.int32(), total, rmap, None, null_count=0) \
         if rmap is not None else None

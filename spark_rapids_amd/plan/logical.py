@@ -110,11 +110,14 @@ class Join(LogicalPlan):
         if self.how in ("semi", "anti"):
             return ls
         rs = self.right.schema()
+        left_fields = [Field(f.name, f.dtype,
+                             f.nullable or self.how == "full")
+                       for f in ls.fields]
         right_fields = []
         for f in rs.fields:
-            nullable = f.nullable or self.how == "left"
+            nullable = f.nullable or self.how in ("left", "full")
             right_fields.append(Field(f.name, f.dtype, nullable))
-        return Schema(list(ls.fields) + right_fields)
+        return Schema(left_fields + right_fields)
 
     def name(self) -> str:
         return f"Join({self.how})"

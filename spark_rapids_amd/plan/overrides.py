@@ -162,7 +162,7 @@ class Tagger:
                 r = _HASH_KEYS.supports(rs.field(k).dtype)
                 if r:
                     reasons.append(f"join key {k}: {r}")
-            if node.how not in ("inner", "left", "semi", "anti"):
+            if node.how not in ("inner", "left", "semi", "anti", "full"):
                 reasons.append(f"join type {node.how} not on GPU")
         elif isinstance(node, L.MapBatches):
             reasons.append("python map_batches runs on CPU (UDF bridge)")

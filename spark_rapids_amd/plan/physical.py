@@ -442,19 +442,29 @@ class HashJoinExec(PhysicalExec):
             rtable = ops.concat_batches(rbatches) if len(rbatches) > 1 else rbatches[0]
         lkidx = [left.schema.index(k) for k in self.left_on]
         rkidx = [right.schema.index(k) for k in self.right_on]
+        right_matched = None
+        if self.how == "full" and rtable is not None and rtable.num_rows:
+            import numpy as _np
+            import torch as _torch
+
+            right_matched = (_torch.zeros(rtable.num_rows,
+                                          dtype=_torch.uint8, device="cuda")
+                             if self.gpu else
+                             _np.zeros(rtable.num_rows, dtype=bool))
         lsource = lbatches_override if lbatches_override is not None \
             else left.execute()
         for lbatch in lsource:
             if lbatch.num_rows == 0:
                 continue
             if rtable is None or rtable.num_rows == 0:
-                if self.how in ("left", "anti"):
+                if self.how in ("left", "anti", "full"):
                     if self.how == "anti":
                         yield lbatch
                     else:
                         yield self._left_with_null_right(lbatch)
                 continue
-            lmap, rmap = ops.join_gather_maps(lbatch, rtable, lkidx, rkidx, self.how)
+            lmap, rmap = ops.join_gather_maps(lbatch, rtable, lkidx, rkidx,
+                                              self.how, right_matched)
             if self.how in ("semi", "anti"):
                 out = ops.gather(lbatch, lmap)
                 if out.num_rows:
@@ -464,6 +474,46 @@ class HashJoinExec(PhysicalExec):
             rout = ops.gather(rtable, rmap)
             if lout.num_rows:
                 yield ColumnBatch(lout.columns + rout.columns, lout.num_rows)
+        if self.how == "full" and rtable is not None and rtable.num_rows:
+            extra = self._unmatched_right(rtable, right_matched, left.schema)
+            if extra is not None and extra.num_rows:
+                yield extra
+
+    def _unmatched_right(self, rtable: ColumnBatch, right_matched,
+                         left_schema: Schema):
+        """Full outer: rows of the build side no probe row matched, with
+        null left columns. In distributed broadcast/replicated mode every
+        rank holds the same build table, so only rank 0 emits them;
+        shuffled mode is co-partitioned and emits locally."""
+        from ..shuffle import dist as _dist
+
+        c = _dist.ctx()
+        if c.is_multi and self._strategy != "shuffled" and c.rank != 0:
+            return None
+        if self.gpu:
+            import torch as _torch
+
+            from ..ops import gpu_backend as _gb
+
+            # unmatched = rows where the matched flag is 0
+            notm = _torch.empty(rtable.num_rows, dtype=_torch.uint8,
+                                device="cuda")
+            _gb.ext.unary(_gb._UN_OPS["not"], 0, right_matched.data_ptr(), 0,
+                          notm.data_ptr(), 0, rtable.num_rows, _gb._stream())
+            mask_col = Column(DType.bool_(), rtable.num_rows, notm, None,
+                              null_count=0)
+            runm = ops.apply_boolean_mask(rtable, mask_col)
+        else:
+            import numpy as _np
+
+            idx = _np.nonzero(~right_matched)[0].astype(_np.int32)
+            runm = ops.gather(rtable, Column.from_numpy(idx))
+        if runm.num_rows == 0:
+            return None
+        cols = [Column.nulls(f.dtype, runm.num_rows,
+                             "cuda" if self.gpu else "cpu")
+                for f in left_schema.fields]
+        return ColumnBatch(cols + runm.columns, runm.num_rows)
 
     def _left_with_null_right(self, lbatch: ColumnBatch) -> ColumnBatch:
         nsch = self.schema

@@ -481,3 +481,33 @@ def test_concurrent_queries_semaphore_gpu():
     [t.join() for t in ts]
     assert not errors, errors
     assert results[0] == results[1] and results[0] is not None
+
+
+def test_full_outer_join_gpu():
+    s = sr.Session()
+    rng2 = np.random.default_rng(31)
+    left = s.create_dataframe({
+        "k": rng2.integers(0, 800, 5000).astype(np.int64),
+        "a": rng2.uniform(0, 1, 5000)})
+    right = s.create_dataframe({
+        "k": rng2.integers(400, 1200, 900).astype(np.int64),
+        "b": rng2.integers(0, 100, 900).astype(np.int32)})
+    tree = left.join(right, on="k", how="full").physical_plan().tree_string()
+    assert "GpuHashJoin" in tree, tree
+    gpu = sorted(left.join(right, on="k", how="full").collect(), key=repr)
+    s2 = sr.Session({"spark.rapids.sql.enabled": False})
+    left2 = s2.create_dataframe({
+        "k": rng2.integers(0, 800, 5000).astype(np.int64),
+        "a": rng2.uniform(0, 1, 5000)})
+    # regenerate identical data with the same seed stream is tricky; instead
+    # compare GPU vs CPU on THE SAME session data via conf flip
+    s3 = sr.Session({"spark.rapids.sql.enabled": False})
+    lcpu = s3.create_dataframe(
+        {"k": left.collect_batch().columns[0].to_numpy(),
+         "a": left.collect_batch().columns[1].to_numpy()})
+    rcpu = s3.create_dataframe(
+        {"k": right.collect_batch().columns[0].to_numpy(),
+         "b": right.collect_batch().columns[1].to_numpy()})
+    cpu = sorted(lcpu.join(rcpu, on="k", how="full").collect(), key=repr)
+    assert len(gpu) == len(cpu)
+    assert gpu == cpu

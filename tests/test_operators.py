@@ -199,3 +199,16 @@ def test_session_range(session):
     df = session.range(10)
     assert df.count() == 10
     assert session.range(2, 8, 2).to_pydict()["id"] == [2, 4, 6]
+
+
+def test_full_outer_join(session):
+    left = session.create_dataframe({"k": [1, 2, 3, None], "a": [10, 20, 30, 40]})
+    right = session.create_dataframe({"k": [2, 3, 4, None], "b": [200, 300, 400, 500]})
+    out = sorted(left.join(right, on="k", how="full").collect(), key=repr)
+    # matches: 2,3; left-unmatched: 1, None; right-unmatched: 4, None
+    assert len(out) == 6
+    d = {r[:2]: r[2:] for r in out}
+    assert d[(2, 20)] == (2, 200)
+    assert d[(3, 30)] == (3, 300)
+    assert d[(1, 10)] == (None, None)
+    assert (None, None, 4, 400) in out
