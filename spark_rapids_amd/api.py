@@ -84,6 +84,11 @@ class DataFrame:
         return DataFrame(self.session,
                          L.Join(self.plan, other.plan, list(on), r_on, how))
 
+    def cross_join(self, other: "DataFrame") -> "DataFrame":
+        """Cartesian product; combine with filter() for non-equi joins
+        (broadcast-nested-loop-join analogue)."""
+        return DataFrame(self.session, L.CrossJoin(self.plan, other.plan))
+
     def sort(self, *keys: str, descending: Union[bool, List[bool]] = False) -> "DataFrame":
         ks = list(keys)
         if isinstance(descending, bool):

@@ -23,11 +23,15 @@ OFFSETS = {"lag", "lead"}
 class WindowSpec:
     def __init__(self, partition_by: Sequence[str] = (),
                  order_by: Sequence[str] = (),
-                 descending: Optional[Sequence[bool]] = None):
+                 descending: Optional[Sequence[bool]] = None,
+                 rows_between=None):
         self.partition_by = list(partition_by)
         self.order_by = list(order_by)
         self.descending = list(descending) if descending is not None \
             else [False] * len(self.order_by)
+        # (preceding, following) row offsets, e.g. (-3, 0) = 3 PRECEDING..
+        # CURRENT ROW; None = default frame (running / whole partition)
+        self.rows_between = rows_between
 
 
 class WindowFunc:
@@ -42,8 +46,10 @@ class WindowFunc:
 
     def over(self, partition_by: Sequence[str] = (),
              order_by: Sequence[str] = (),
-             descending: Optional[Sequence[bool]] = None) -> "WindowExpr":
-        return WindowExpr(self, WindowSpec(partition_by, order_by, descending))
+             descending: Optional[Sequence[bool]] = None,
+             rows_between=None) -> "WindowExpr":
+        return WindowExpr(self, WindowSpec(partition_by, order_by, descending,
+                                           rows_between))
 
 
 class WindowExpr:
@@ -56,6 +62,14 @@ class WindowExpr:
             raise ValueError(f"{func.op} requires order_by")
         if func.op in OFFSETS and not spec.order_by:
             raise ValueError(f"{func.op} requires order_by")
+        if spec.rows_between is not None:
+            if func.op not in AGGS:
+                raise ValueError("rows_between needs an aggregate function")
+            if not spec.order_by:
+                raise ValueError("rows_between requires order_by")
+            lo, hi = spec.rows_between
+            if lo > 0 or hi < lo:
+                raise ValueError("rows_between must be (lo<=0, hi>=lo)")
 
     def alias(self, name: str) -> "WindowExpr":
         return WindowExpr(self.func, self.spec, name)

@@ -139,6 +139,23 @@ class MapBatches(LogicalPlan):
         return self._schema if self._schema is not None else self.child.schema()
 
 
+class CrossJoin(LogicalPlan):
+    """Cartesian product (reference analogue: GpuCartesianProductExec /
+    GpuBroadcastNestedLoopJoinExec — non-equi joins are cross + filter)."""
+
+    def __init__(self, left: LogicalPlan, right: LogicalPlan):
+        self.left = left
+        self.right = right
+
+    @property
+    def children(self):
+        return (self.left, self.right)
+
+    def schema(self) -> Schema:
+        return Schema(list(self.left.schema().fields) +
+                      list(self.right.schema().fields))
+
+
 class Window(LogicalPlan):
     def __init__(self, window_exprs, child: LogicalPlan):
         self.window_exprs = list(window_exprs)
@@ -205,7 +222,7 @@ def is_replicated(plan: LogicalPlan) -> bool:
     distributed exchanges/broadcasts must be skipped)."""
     if isinstance(plan, Scan):
         return plan.replicated
-    if isinstance(plan, Join):
+    if isinstance(plan, (Join, CrossJoin)):
         # the build (right) side is either replicated or broadcast
         # (all-gathered) by the exec, so the output's distribution follows
         # the stream (left) side

@@ -100,6 +100,16 @@ def prune_columns(plan: L.LogicalPlan,
             child_needed &= set(plan.child.schema().names)
         return L.Window(plan.window_exprs,
                         prune_columns(plan.child, child_needed))
+    if isinstance(plan, L.CrossJoin):
+        lnames = set(plan.left.schema().names)
+        if needed is None:
+            lneed = rneed = None
+        else:
+            lneed = needed & lnames
+            rneed = needed - lnames
+        left = _project_to(prune_columns(plan.left, lneed), lneed)
+        right = _project_to(prune_columns(plan.right, rneed), rneed)
+        return L.CrossJoin(left, right)
     if isinstance(plan, L.MapBatches):
         return L.MapBatches(plan.fn, prune_columns(plan.child, None),
                             plan._schema)

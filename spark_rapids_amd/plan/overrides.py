@@ -190,7 +190,10 @@ class Tagger:
                         reasons.append(f"window {op}({vt}) not on GPU")
                     continue
                 if op in ("min", "max"):
-                    if spec.order_by:
+                    if spec.rows_between is not None:
+                        reasons.append(
+                            f"bounded {op} window has no GPU kernel yet")
+                    elif spec.order_by:
                         reasons.append(
                             f"running {op} window has no GPU kernel yet")
                     elif vt is not None and not vt.is_numeric:
@@ -291,6 +294,9 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
     if isinstance(node, L.MapBatches):
         return P.MapBatchesExec(node.fn, _ensure_device(kids[0], "cpu"),
                                 node.schema())
+    if isinstance(node, L.CrossJoin):
+        return P.CrossJoinExec(device, kids[0], kids[1], node.schema(),
+                               right_replicated=L.is_replicated(node.right))
     if isinstance(node, L.Window):
         from .window_exec import WindowExec
 

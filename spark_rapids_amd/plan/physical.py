@@ -528,6 +528,69 @@ class HashJoinExec(PhysicalExec):
         return f"{self.name()}({self.how}, {pairs})"
 
 
+class CrossJoinExec(PhysicalExec):
+    """Cartesian product; output bounded by maxOutputRows to keep an
+    accidental unfiltered cross join from exploding memory. Gather maps are
+    computed with the iota + int-div/mod kernels on GPU."""
+
+    MAX_OUTPUT_ROWS = 1 << 28
+
+    def __init__(self, device: str, left: PhysicalExec, right: PhysicalExec,
+                 schema: Schema, right_replicated: bool = True):
+        super().__init__(device, schema, [left, right])
+        self.right_replicated = right_replicated
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        import numpy as np
+
+        left, right = self.children
+        rbatches = list(right.execute())
+        from ..shuffle import dist as _dist
+        if _dist.ctx().is_multi and not self.right_replicated:
+            from ..shuffle.exchange import gather_all
+
+            local = rbatches[0] if len(rbatches) == 1 else (
+                ops.concat_batches(rbatches) if rbatches else None)
+            if local is None:
+                local = ColumnBatch([Column.from_pylist([], f.dtype)
+                                     for f in right.schema.fields], 0)
+                if self.gpu:
+                    local = local.cuda()
+            rbatches = [b for b in gather_all(local) if b.num_rows]
+        if not rbatches:
+            return
+        rtable = ops.concat_batches(rbatches) if len(rbatches) > 1             else rbatches[0]
+        nr = rtable.num_rows
+        for lbatch in left.execute():
+            nl = lbatch.num_rows
+            if nl == 0 or nr == 0:
+                continue
+            total = nl * nr
+            if total > self.MAX_OUTPUT_ROWS:
+                raise MemoryError(
+                    f"cross join would produce {total} rows "
+                    f"(> {self.MAX_OUTPUT_ROWS}); filter the inputs first")
+            if self.gpu:
+                import torch as _torch
+
+                from ..ops import gpu_backend as _gb
+
+                iota = _torch.empty(total, dtype=_torch.int32, device="cuda")
+                _gb.ext.iota_i32(iota.data_ptr(), total, _gb._stream())
+                icol = Column(DType.int32(), total, iota, None, null_count=0)
+                lmap = _gb.binary_op_scalar("int_div", icol, nr,
+                                            DType.int32())
+                rmap = _gb.binary_op_scalar("mod", icol, nr, DType.int32())
+            else:
+                li = np.repeat(np.arange(nl, dtype=np.int32), nr)
+                ri = np.tile(np.arange(nr, dtype=np.int32), nl)
+                lmap = Column.from_numpy(li)
+                rmap = Column.from_numpy(ri)
+            lout = ops.gather(lbatch, lmap)
+            rout = ops.gather(rtable, rmap)
+            yield ColumnBatch(lout.columns + rout.columns, total)
+
+
 class SortExec(PhysicalExec):
     def __init__(self, device: str, keys: List[str], descending: List[bool],
                  nulls_last: List[bool], child: PhysicalExec):

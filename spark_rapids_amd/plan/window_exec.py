@@ -252,6 +252,73 @@ class WindowExec(PhysicalExec):
             ext.mask_expand(vc.validity.data_ptr(), valid_u8.data_ptr(),
                             False, n, s)
         nn_col = Column(DType.bool_(), n, valid_u8, None, null_count=0)
+        rb = self.spec.rows_between
+        if rb is not None and op in ("sum", "count", "mean"):
+            lo_off, hi_off = rb
+            hp_ext = torch.cat([head_pos, torch.tensor(
+                [n], dtype=torch.int32, device="cuda")])
+            segid_next = gb.binary_op_scalar(
+                "add", Column(INT32, n, seg_id.data, None, null_count=0), 1,
+                INT32)
+            seg_end = Column(INT32, n, self._gather_i32(hp_ext,
+                                                        segid_next.data, n),
+                             None, null_count=0)
+            seg_end = gb.binary_op_scalar("sub", seg_end, 1, INT32)
+            b_idx = gb.binary_op("min",
+                                 gb.binary_op_scalar("add", iota_col, hi_off,
+                                                     INT32), seg_end, INT32)
+            a_idx = gb.binary_op("max",
+                                 gb.binary_op_scalar("add", iota_col, lo_off,
+                                                     INT32), seg_start_col,
+                                 INT32)
+            am1 = gb.binary_op_scalar("sub", a_idx, 1, INT32)
+            use_f64 = out_dt.is_floating
+            work_t = FLOAT64 if use_f64 else INT64
+            v64 = gb.cast(Column(vc.dtype, n, vc.data, None, null_count=0),
+                          work_t)
+            vz = gb.binary_op("mul", v64, gb.cast(nn_col, work_t), work_t)
+            nn64 = gb.cast(nn_col, INT64)
+            # inclusive global prefix sums; gather_fixed zero-fills index -1
+            if use_f64:
+                excl = self._scan_f64(vz.data, n)
+                incl = gb.binary_op("add",
+                                    Column(work_t, n, excl, None,
+                                           null_count=0), vz, work_t)
+            else:
+                excl_t, _ = gb._exclusive_scan_i64(vz.data)
+                incl = gb.binary_op("add",
+                                    Column(INT64, n, excl_t, None,
+                                           null_count=0), vz, INT64)
+            excl_n, _ = gb._exclusive_scan_i64(nn64.data)
+            incl_n = gb.binary_op("add",
+                                  Column(INT64, n, excl_n, None,
+                                         null_count=0), nn64, INT64)
+            incl_b = Column(work_t, n,
+                            self._gather_i32(incl.data, b_idx.data, n), None,
+                            null_count=0)
+            incl_a = Column(work_t, n,
+                            self._gather_i32(incl.data, am1.data, n), None,
+                            null_count=0)
+            cnt_b = Column(INT64, n,
+                           self._gather_i32(incl_n.data, b_idx.data, n),
+                           None, null_count=0)
+            cnt_a = Column(INT64, n,
+                           self._gather_i32(incl_n.data, am1.data, n),
+                           None, null_count=0)
+            cnt_r = gb.binary_op("sub", cnt_b, cnt_a, INT64)
+            # empty frames (b < a) would go negative; clamp via max(cnt, 0)
+            cnt_r = gb.binary_op_scalar("max", cnt_r, 0, INT64)
+            if op == "count":
+                return cnt_r
+            sum_r = gb.binary_op("sub", incl_b, incl_a, work_t)
+            ov = torch.empty(mask_nbytes(n), dtype=torch.uint8,
+                             device="cuda")
+            ext.mask_from_nonzero(cnt_r.data.data_ptr(), ov.data_ptr(), n, s)
+            sum_c = Column(work_t, n, sum_r.data, ov, null_count=None)
+            if op == "sum":
+                return gb.cast(sum_c, out_dt) if work_t != out_dt else sum_c
+            cf = gb.cast(cnt_r, FLOAT64)
+            return gb.binary_op("div", gb.cast(sum_c, FLOAT64), cf, FLOAT64)
         if running and op in ("sum", "count", "mean"):
             use_f64 = out_dt.is_floating
             work_t = FLOAT64 if use_f64 else INT64
@@ -321,12 +388,12 @@ class WindowExec(PhysicalExec):
         incl = gb.binary_op("add", ec, vcol, INT64)
         return gb.binary_op("sub", incl, at_start, INT64).data
 
-    def _running_sum_f64(self, vz: "torch.Tensor", seg_start_col, n):
+    def _scan_f64(self, vz: "torch.Tensor", n: int):
+        """Global exclusive prefix sum of an f64 tensor (device kernels)."""
         import torch
 
         from ..ops import gpu_backend as gb
         from ..ops.gpu_backend import ext
-        from ..types import FLOAT64
 
         s = gb._stream()
         out = torch.empty(n, dtype=torch.float64, device="cuda")
@@ -336,7 +403,6 @@ class WindowExec(PhysicalExec):
         ext.scan_block_f64(vz.data_ptr(), out.data_ptr(), sums.data_ptr(),
                            n, s)
         if nb > 1:
-            # recursive scan of block sums (small; do on device via same path)
             sums2 = torch.empty(nb, dtype=torch.float64, device="cuda")
             partial = torch.empty(max((nb + per - 1) // per, 1),
                                   dtype=torch.float64, device="cuda")
@@ -350,6 +416,15 @@ class WindowExec(PhysicalExec):
                 ext.scan_add_offsets_f64(sums2.data_ptr(), p2.data_ptr(),
                                          nb, s)
             ext.scan_add_offsets_f64(out.data_ptr(), sums2.data_ptr(), n, s)
+        return out
+
+    def _running_sum_f64(self, vz: "torch.Tensor", seg_start_col, n):
+        import torch
+
+        from ..ops import gpu_backend as gb
+        from ..types import FLOAT64
+
+        out = self._scan_f64(vz, n)
         ec = Column(FLOAT64, n, out, None, null_count=0)
         vcol = Column(FLOAT64, n, vz, None, null_count=0)
         at_start = Column(FLOAT64, n,
@@ -408,8 +483,44 @@ def _compute(w: WindowExpr, table: ColumnBatch, cs, n, heads, ochange, idx,
                      else outv, outvalid if not outvalid.all() else None, out_dt)
 
     vv = np.where(valid, v, 0.0)
-    running = len(w.spec.order_by) > 0
     cnt_f = valid.astype(np.int64)
+    rb = w.spec.rows_between
+    if rb is not None:
+        lo_off, hi_off = rb
+        seg_end = _segment_ends(heads, idx, n)
+        a = np.maximum(idx + lo_off, seg_start)
+        b = np.minimum(idx + hi_off, seg_end)
+        empty_frame = b < a
+        if op in ("min", "max"):
+            fn = min if op == "min" else max
+            res = np.zeros(n)
+            ok = np.zeros(n, dtype=bool)
+            for i in range(n):
+                if empty_frame[i]:
+                    continue
+                vals_in = [v[j] for j in range(a[i], b[i] + 1) if valid[j]]
+                if vals_in:
+                    res[i] = fn(vals_in)
+                    ok[i] = True
+            res = res.astype(out_dt.numpy_dtype())
+            return _make(res, ok if not ok.all() else None, out_dt)
+        csv = np.cumsum(vv)
+        csn = np.cumsum(cnt_f)
+        am1 = np.maximum(a - 1, 0)
+        base_v = np.where(a > 0, csv[am1], 0.0)
+        base_n = np.where(a > 0, csn[am1], 0)
+        sum_r = np.where(empty_frame, 0.0, csv[b] - base_v)
+        cnt_r = np.where(empty_frame, 0, csn[b] - base_n)
+        if op == "count":
+            return _make(cnt_r.astype(np.int64), None, out_dt)
+        ok = cnt_r > 0
+        if op == "mean":
+            res = sum_r / np.maximum(cnt_r, 1)
+        else:
+            res = sum_r
+        return _make(res.astype(out_dt.numpy_dtype()),
+                     ok if not ok.all() else None, out_dt)
+    running = len(w.spec.order_by) > 0
     if running:
         cum = np.cumsum(vv)
         run_sum = cum - cum[seg_start] + vv[seg_start]

@@ -212,3 +212,13 @@ def test_full_outer_join(session):
     assert d[(3, 30)] == (3, 300)
     assert d[(1, 10)] == (None, None)
     assert (None, None, 4, 400) in out
+
+
+def test_cross_join_and_non_equi(session):
+    a = session.create_dataframe({"x": [1, 2, 3]})
+    b = session.create_dataframe({"y": [10, 20]})
+    out = sorted(a.cross_join(b).collect())
+    assert len(out) == 6 and out[0] == (1, 10)
+    # non-equi: x * 10 < y
+    ne = sorted(a.cross_join(b).filter(col("x") * 10 < col("y")).collect())
+    assert ne == [(1, 20)]

@@ -175,3 +175,44 @@ def test_gpu_window_matches_cpu(fn):
                 assert g == _pt.approx(c, rel=1e-9), (fn, i)
         else:
             assert g == c, (fn, i, g, c)
+
+
+def test_bounded_frame_cpu(session):
+    df = session.create_dataframe({
+        "p": [1] * 6 + [2] * 3,
+        "o": [1, 2, 3, 4, 5, 6, 1, 2, 3],
+        "v": [1.0, 2.0, None, 4.0, 5.0, 6.0, 10.0, 20.0, 30.0],
+    })
+    e = win_sum(col("v")).over(["p"], ["o"], rows_between=(-1, 1))
+    out = df.with_column("s", e).collect()
+    vals = [r[3] for r in out]
+    assert vals[:6] == [3.0, 3.0, 6.0, 9.0, 15.0, 11.0]
+    assert vals[6:] == [30.0, 60.0, 50.0]
+    e2 = win_count(col("v")).over(["p"], ["o"], rows_between=(-2, 0))
+    cnts = [r[3] for r in df.with_column("c", e2).collect()]
+    assert cnts[:6] == [1, 2, 2, 2, 2, 3]
+    e3 = win_min(col("v")).over(["p"], ["o"], rows_between=(-1, 0))
+    mins = [r[3] for r in df.with_column("m", e3).collect()]
+    assert mins[:6] == [1.0, 1.0, 2.0, 4.0, 4.0, 5.0]
+
+
+@_pt.mark.gpu
+@_pt.mark.parametrize("frame", [(-1, 1), (-3, 0), (0, 2), (-5, -1)])
+def test_gpu_bounded_frame_matches_cpu(frame):
+    data = _win_data(8000)
+    e = lambda: win_sum(col("v")).over(["p"], ["o"], rows_between=frame)
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    qg = _df(sg, data).with_column("w", e())
+    assert "GpuWindow" in qg.physical_plan().tree_string()
+    gout = qg.to_pydict()["w"]
+    cout = _df(sc, data).with_column("w", e()).to_pydict()["w"]
+    for i, (g, c) in enumerate(zip(gout, cout)):
+        if c is None or g is None:
+            assert g is None and c is None, (frame, i, g, c)
+        else:
+            assert g == _pt.approx(c, rel=1e-9), (frame, i)
+    ec = lambda: win_count(col("v")).over(["p"], ["o"], rows_between=frame)
+    gc_ = _df(sg, data).with_column("w", ec()).to_pydict()["w"]
+    cc_ = _df(sc, data).with_column("w", ec()).to_pydict()["w"]
+    assert gc_ == cc_
