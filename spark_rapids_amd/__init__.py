@@ -11,8 +11,8 @@ from .config import RapidsConf, help_doc
 from .expr.aggregates import (avg, count, count_star, max_, min_, stddev,
                               sum_, variance)
 from .expr.expressions import (CaseWhen, coalesce, col, date_add, date_sub,
-                               datediff, greatest, isin, least, lit, round_,
-                               when)
+                               datediff, greatest, hour, isin, least, lit,
+                               minute, round_, second, when)
 from .expr.windows import (dense_rank, lag, lead, rank, row_number, win_avg,
                            win_count, win_max, win_min, win_sum)
 from .types import (BOOL, DATE32, FLOAT32, FLOAT64, INT8, INT16, INT32, INT64,

@@ -609,6 +609,26 @@ def isin(e, *values) -> Expression:
     return acc
 
 
+def hour(ts) -> Expression:
+    """hour of a timestamp (micros since epoch, UTC)."""
+    e = _as_expr(ts)
+    return BinaryExpr("pmod", BinaryExpr("int_div", CastExpr(e, INT64),
+                                         Literal(3_600_000_000)),
+                      Literal(24))
+
+
+def minute(ts) -> Expression:
+    e = _as_expr(ts)
+    return BinaryExpr("pmod", BinaryExpr("int_div", CastExpr(e, INT64),
+                                         Literal(60_000_000)), Literal(60))
+
+
+def second(ts) -> Expression:
+    e = _as_expr(ts)
+    return BinaryExpr("pmod", BinaryExpr("int_div", CastExpr(e, INT64),
+                                         Literal(1_000_000)), Literal(60))
+
+
 def date_add(d, days) -> Expression:
     return BinaryExpr("add", _as_expr(d), _as_expr(days))
 

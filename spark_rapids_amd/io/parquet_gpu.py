@@ -193,6 +193,28 @@ class _ChunkDecoder:
 
         mask, valid_idx, n_valid = self._valid_parts(levels, n)
         nulls = mask is not None
+        if encoding == PLAIN and self.phys == "BOOLEAN":
+            # plain booleans are LSB-first bit-packed — same layout as the
+            # validity bitmask, so the mask-expand kernel decodes them
+            nbytes = (n_valid + 7) // 8
+            bits = torch.from_numpy(np.frombuffer(
+                values, dtype=np.uint8, count=nbytes).copy()).cuda()
+            padded = torch.zeros(mask_nbytes(n_valid), dtype=torch.uint8,
+                                 device="cuda")
+            padded[:nbytes] = bits
+            dense = torch.empty(max(n_valid, 1), dtype=torch.uint8,
+                                device="cuda")[:n_valid]
+            if n_valid:
+                self.ext.mask_expand(padded.data_ptr(), dense.data_ptr(),
+                                     False, n_valid, self.s)
+            if not nulls:
+                return Column(self.dtype, n, dense, None, null_count=0)
+            out = torch.zeros(n, dtype=torch.uint8, device="cuda")
+            if n_valid:
+                self.ext.scatter_fixed(1, dense.data_ptr(),
+                                       valid_idx.data_ptr(), out.data_ptr(),
+                                       n_valid, self.s)
+            return Column(self.dtype, n, out, mask, null_count=None)
         if encoding == PLAIN:
             if self.phys not in _PHYS_NP:
                 raise NotImplementedError(f"PLAIN {self.phys}")

@@ -52,3 +52,15 @@ def test_year_month_day(session):
     m = df.select(UnaryExpr("month", col("d")).alias("m")).to_pydict()["m"]
     d = df.select(UnaryExpr("day", col("d")).alias("dd")).to_pydict()["dd"]
     assert y == [2000, 2001] and m == [1, 1] and d == [1, 1]
+
+
+def test_timestamp_fields(session):
+    from spark_rapids_amd import TIMESTAMP, hour, minute, second
+
+    # 2000-01-01 13:45:30 UTC in micros
+    micros = 946_734_330_000_000
+    df = session.create_dataframe({"t": [micros, micros + 61_000_000]},
+                                  dtypes={"t": TIMESTAMP})
+    assert df.select(hour(col("t")).alias("h")).to_pydict()["h"] == [13, 13]
+    assert df.select(minute(col("t")).alias("m")).to_pydict()["m"] == [45, 46]
+    assert df.select(second(col("t")).alias("s")).to_pydict()["s"] == [30, 31]

