@@ -32,6 +32,7 @@ KERNELS = [
     "kernels/strings.hip",
     "kernels/window.hip",
     "kernels/decimal128.hip",
+    "kernels/regex.hip",
 ]
 
 CXXFLAGS = ["-O3", "-std=c++20", "-fPIC", f"--offload-arch={ARCH}",

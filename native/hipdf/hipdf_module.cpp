@@ -100,6 +100,8 @@ void hipdf_scatter_fixed(int, const void*, const void*, void*, int64_t,
 void hipdf_levels_to_mask(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_str_cmp(int, const void*, const void*, const void*, const void*,
                    void*, int64_t, hipStream_t);
+void hipdf_regex_match(const void*, int, const void*, const void*,
+                       const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_str_cmp_scalar(int, const void*, const void*, const void*, int,
                           void*, int64_t, hipStream_t);
 void hipdf_str_find(int, const void*, const void*, const void*, int, void*,
@@ -394,6 +396,13 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
 
+  m.def("regex_match", [](int64_t prog, int nops, int64_t classes,
+                          int64_t offsets, int64_t bytes, int64_t out,
+                          int64_t overflow, int64_t n, int64_t stream) {
+    hipdf_regex_match(P(prog), nops, P(classes), P(offsets), P(bytes),
+                      PM(out), PM(overflow), n, S(stream));
+    check_async();
+  });
   m.def("str_cmp", [](int op, int64_t ao, int64_t ab, int64_t bo, int64_t bb,
                       int64_t out, int64_t n, int64_t stream) {
     hipdf_str_cmp(op, P(ao), P(ab), P(bo), P(bb), PM(out), n, S(stream));

@@ -119,6 +119,10 @@ class Expression:
     def like(self, pattern: str) -> "StringPredicate":
         return StringPredicate("like", self, pattern)
 
+    def rlike(self, pattern: str) -> "StringPredicate":
+        """Java-regex find() semantics (RLike)."""
+        return StringPredicate("rlike", self, pattern)
+
     def substr(self, pos: int, length: int = -1) -> "Substring":
         return Substring(self, pos, length)
 

@@ -396,6 +396,13 @@ def round_half_up(col: Column, scale: int) -> Column:
 
 def str_predicate(op: str, col: Column, pattern: str) -> Column:
     a, av = _vals(col), _valid(col)
+    if op == "rlike":
+        import re
+
+        prog = re.compile(pattern)
+        res = np.array([bool(prog.search(x)) if x is not None else False
+                        for x in a], dtype=np.uint8)
+        return _make(res, av if not av.all() else None, DType.bool_())
     if op == "like":
         import re
 

@@ -229,6 +229,29 @@ def str_predicate(op: str, col: Column, pattern: str) -> Column:
     n = col.size
     s = _stream()
     out = _alloc(n, DType.bool_())
+    if op == "rlike":
+        from .regex_compiler import compile_regex
+
+        prog = compile_regex(pattern)  # tagger pre-checks; raise = bug
+        flat = []
+        for o, a0, a1 in prog.ops:
+            flat.extend([o, a0, a1])
+        prog_t = torch.tensor(flat, dtype=torch.int32).cuda()
+        cls_blob = b"".join(prog.classes) or b"\x00" * 32
+        cls_t = torch.frombuffer(bytearray(cls_blob),
+                                 dtype=torch.uint8).cuda()
+        overflow = torch.zeros(1, dtype=torch.int32, device="cuda")
+        ext.regex_match(prog_t.data_ptr(), len(prog.ops), cls_t.data_ptr(),
+                        col.offsets.data_ptr(), col.data.data_ptr(),
+                        out.data_ptr(), overflow.data_ptr(), n, s)
+        if int(overflow.item()) > 0:
+            # pathological backtracking: redo the whole column on CPU
+            from . import cpu_backend
+
+            host = cpu_backend.str_predicate(op, col.cpu(), pattern)
+            return host.cuda()
+        v = col.validity.clone() if col.validity is not None else None
+        return Column(DType.bool_(), n, out, v, null_count=col._null_count)
     pat = _pattern_tensor(pattern)
     if op == "like":
         ext.str_like(col.offsets.data_ptr(), col.data.data_ptr(),

@@ -110,7 +110,13 @@ class Tagger:
         elif isinstance(e, (IsNull, CaseWhen, Coalesce, Round)):
             pass
         elif isinstance(e, StringPredicate):
-            pass  # contains/starts/ends/like have GPU kernels
+            if e.op == "rlike":
+                from ..ops.regex_compiler import RegexUnsupported, compile_regex
+
+                try:
+                    compile_regex(e.pattern)
+                except RegexUnsupported as ex:
+                    out.append(f"regex not supported on GPU: {ex}")
         elif isinstance(e, Substring):
             pass
         else:
