@@ -97,6 +97,16 @@ class DataFrame:
             desc = list(descending)
         return DataFrame(self.session, L.Sort(self.plan, ks, desc))
 
+    def sample(self, fraction: float, seed: int = 42) -> "DataFrame":
+        """Bernoulli sample via the murmur3 row-position hash (reference
+        analogue: GpuPartitionwiseSampledRDD / GpuPoissonSampler's
+        deterministic per-row draw). Deterministic for a given seed."""
+        from .expr.sample import SampleHash
+        from .expr.expressions import lit
+
+        threshold = int(fraction * 2147483647)
+        return self.filter(SampleHash(seed) < lit(threshold))
+
     def distinct(self) -> "DataFrame":
         """Drop duplicate rows (group-by all columns with no aggregates)."""
         keys = [_col(f.name) for f in self.plan.schema().fields]

@@ -119,6 +119,8 @@ class Tagger:
                     out.append(f"regex not supported on GPU: {ex}")
         elif isinstance(e, Substring):
             pass
+        elif type(e).__name__ == "SampleHash":
+            pass
         else:
             out.append(f"expression {type(e).__name__} not supported on GPU")
         for c in e.children:

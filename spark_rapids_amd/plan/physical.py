@@ -272,7 +272,9 @@ class HashAggregateExec(PhysicalExec):
         nkeys = len(self.group_exprs)
         value_exprs, partial, merge_ops, final = _lower_aggs(self.aggs, in_schema)
 
-        partial_results: List[ColumnBatch] = []
+        from ..memory.spill import SpillableBatch
+
+        partial_handles: List[SpillableBatch] = []
         for batch in source.execute():
             def task(b):
                 sel = None
@@ -292,12 +294,19 @@ class HashAggregateExec(PhysicalExec):
                     return _gb.group_by_aggregate(pre, list(range(nkeys)),
                                                   specs, sel=sel)
                 return ops.group_by_aggregate(pre, list(range(nkeys)), specs)
-            partial_results.append(with_retry_split_single(task, batch))
+            # partial results are spillable between input batches so memory
+            # pressure can evict them (reference: partial aggs held as
+            # SpillableColumnarBatch between batches)
+            partial_handles.append(
+                SpillableBatch(with_retry_split_single(task, batch)))
 
-        if not partial_results:
+        if not partial_handles:
             return
+        partial_results = [h.get() for h in partial_handles]
         merged_in = ops.concat_batches(partial_results) if len(partial_results) > 1 \
             else partial_results[0]
+        for h in partial_handles:
+            h.close()
 
         # distributed merge: keyed -> RCCL all-to-all hash exchange so each
         # rank owns a disjoint key range; keyless -> all-gather partials and
@@ -600,10 +609,15 @@ class SortExec(PhysicalExec):
         self.nulls_last = nulls_last
 
     def execute(self) -> Iterator[ColumnBatch]:
-        batches = list(self.children[0].execute())
-        if not batches:
+        from ..memory.spill import SpillableBatch
+
+        handles = [SpillableBatch(b) for b in self.children[0].execute()]
+        if not handles:
             return
+        batches = [h.get() for h in handles]
         table = ops.concat_batches(batches) if len(batches) > 1 else batches[0]
+        for h in handles:
+            h.close()
         if table.num_rows == 0:
             yield table
             return

@@ -49,7 +49,8 @@ def supported_ops_doc() -> str:
     lines.append("compare: `" + "`, `".join(sorted(ov._GPU_STRING_OK)) + "`; "
                  "unary: `" + "`, `".join(sorted(ov._GPU_STRING_UNARY)) +
                  "`; plus `contains`, `starts_with`, `ends_with`, `like`, "
-                 "`substring` (see native/hipdf/kernels/strings.hip)")
+                 "`substring` (strings.hip) and `rlike` (bytecode regex "
+                 "VM, regex.hip, CPU fallback outside the subset)")
     lines += ["", "## Aggregate functions", "",
               "`sum`, `count`, `count(*)`, `min`, `max`, `avg` "
               "(partial/merge lowering; mean as sum+count)", "",

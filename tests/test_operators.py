@@ -222,3 +222,13 @@ def test_cross_join_and_non_equi(session):
     # non-equi: x * 10 < y
     ne = sorted(a.cross_join(b).filter(col("x") * 10 < col("y")).collect())
     assert ne == [(1, 20)]
+
+
+def test_sample_deterministic(session):
+    df = session.create_dataframe({"x": list(range(10_000))})
+    a = df.sample(0.1, seed=7).count()
+    b = df.sample(0.1, seed=7).count()
+    assert a == b
+    assert 700 < a < 1300  # ~10%
+    c = df.sample(0.5, seed=7).count()
+    assert 4500 < c < 5500
