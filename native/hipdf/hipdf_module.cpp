@@ -95,6 +95,8 @@ void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
                              hipStream_t);
+void hipdf_str_plain_offsets(const void*, int64_t, int64_t, void*, void*,
+                             void*, hipStream_t);
 void hipdf_scatter_fixed(int, const void*, const void*, void*, int64_t,
                          hipStream_t);
 void hipdf_levels_to_mask(const void*, int, void*, int64_t, hipStream_t);
@@ -378,6 +380,13 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("scan_add_offsets_f64", [](int64_t out, int64_t sums, int64_t n,
                                    int64_t stream) {
     hipdf_scan_add_offsets_f64(PM(out), P(sums), n, S(stream));
+    check_async();
+  });
+  m.def("str_plain_offsets", [](int64_t data, int64_t nbytes,
+                                int64_t n_values, int64_t starts,
+                                int64_t lens, int64_t error, int64_t stream) {
+    hipdf_str_plain_offsets(P(data), nbytes, n_values, PM(starts), PM(lens),
+                            PM(error), S(stream));
     check_async();
   });
   m.def("rle_hybrid_decode", [](int64_t data, int64_t nbytes, int bw,

@@ -117,7 +117,44 @@ __global__ void k_levels_to_mask(const int32_t* __restrict__ levels,
   }
 }
 
+// PLAIN byte-array page: [u32 len][bytes]... — walk the lengths (single
+// block, sequential dependency) emitting per-value (start, len); the
+// byte compaction reuses the substr copy kernel.
+__global__ void k_str_plain_offsets(const uint8_t* __restrict__ data,
+                                    int64_t nbytes, int64_t n_values,
+                                    int32_t* __restrict__ starts,
+                                    int64_t* __restrict__ lens,
+                                    int32_t* __restrict__ error) {
+  if (blockIdx.x != 0 || threadIdx.x != 0) return;
+  int64_t p = 0;
+  for (int64_t i = 0; i < n_values; ++i) {
+    if (p + 4 > nbytes) {
+      atomicAdd(error, 1);
+      return;
+    }
+    uint32_t ln = (uint32_t)data[p] | ((uint32_t)data[p + 1] << 8) |
+                  ((uint32_t)data[p + 2] << 16) |
+                  ((uint32_t)data[p + 3] << 24);
+    p += 4;
+    if (p + ln > (uint64_t)nbytes) {
+      atomicAdd(error, 1);
+      return;
+    }
+    starts[i] = (int32_t)p;
+    lens[i] = ln;
+    p += ln;
+  }
+}
+
 extern "C" {
+
+void hipdf_str_plain_offsets(const void* data, int64_t nbytes,
+                             int64_t n_values, void* starts, void* lens,
+                             void* error, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_plain_offsets, dim3(1), dim3(64), 0, stream,
+                     (const uint8_t*)data, nbytes, n_values,
+                     (int32_t*)starts, (int64_t*)lens, (int32_t*)error);
+}
 
 void hipdf_rle_hybrid_decode(const void* data, int64_t nbytes, int bit_width,
                              void* out, int64_t n_values, hipStream_t stream) {

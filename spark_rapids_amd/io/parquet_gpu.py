@@ -215,6 +215,53 @@ class _ChunkDecoder:
                                        valid_idx.data_ptr(), out.data_ptr(),
                                        n_valid, self.s)
             return Column(self.dtype, n, out, mask, null_count=None)
+        if encoding == PLAIN and self.phys == "BYTE_ARRAY":
+            page = torch.from_numpy(np.frombuffer(
+                values, dtype=np.uint8).copy()).cuda()
+            starts = torch.empty(max(n_valid, 1), dtype=torch.int32,
+                                 device="cuda")[:n_valid]
+            lens = torch.empty(max(n_valid, 1), dtype=torch.int64,
+                               device="cuda")[:n_valid]
+            err = torch.zeros(1, dtype=torch.int32, device="cuda")
+            if n_valid:
+                self.ext.str_plain_offsets(page.data_ptr(), page.numel(),
+                                           n_valid, starts.data_ptr(),
+                                           lens.data_ptr(), err.data_ptr(),
+                                           self.s)
+            if int(err.item()) > 0:
+                raise NotImplementedError("corrupt plain byte-array page")
+            from ..ops import gpu_backend as gb
+
+            scanned, total = gb._exclusive_scan_i64(lens) if n_valid                 else (lens, 0)
+            out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
+                                    device="cuda")[:total]
+            if total:
+                self.ext.substr_copy(page.data_ptr(), starts.data_ptr(),
+                                     lens.data_ptr(), scanned.data_ptr(),
+                                     out_bytes.data_ptr(), n_valid, self.s)
+            offs_dense = torch.empty(n_valid + 1, dtype=torch.int32,
+                                     device="cuda")
+            if n_valid:
+                self.ext.narrow_i64_i32(scanned.data_ptr(),
+                                        offs_dense.data_ptr(), n_valid,
+                                        self.s)
+            offs_dense[n_valid] = total
+            dense_col = Column(DType.string(), n_valid, out_bytes, None,
+                               offs_dense, 0)
+            if not nulls:
+                return dense_col
+            # scatter dense -> rows with nulls: build ridx then string-gather
+            ridx = torch.full((n,), -1, dtype=torch.int32, device="cuda")
+            dense_iota = torch.empty(max(n_valid, 1), dtype=torch.int32,
+                                     device="cuda")[:n_valid]
+            if n_valid:
+                self.ext.iota_i32(dense_iota.data_ptr(), n_valid, self.s)
+                self.ext.scatter_fixed(4, dense_iota.data_ptr(),
+                                       valid_idx.data_ptr(), ridx.data_ptr(),
+                                       n_valid, self.s)
+            out = gb._gather_col(dense_col, ridx, n, maybe_negative=True)
+            return Column(self.dtype, n, out.data, mask, out.offsets,
+                          null_count=None)
         if encoding == PLAIN:
             if self.phys not in _PHYS_NP:
                 raise NotImplementedError(f"PLAIN {self.phys}")

@@ -18,14 +18,17 @@ _EXECS = [
     ("Filter", "boolean mask stream compaction", "all basic types"),
     ("Project", "expression evaluation", "all basic types"),
     ("Aggregate", "hash group-by, partial+merge, distributed exchange",
-     "fixed-width keys on GPU; string keys fall back"),
-    ("Join", "hash equi-join inner/left/semi/anti, broadcast build side",
-     "fixed-width keys on GPU"),
-    ("Sort", "stable LSD radix sort", "fixed-width keys on GPU"),
-    ("Window", "ranking / running + partition aggregates / lag / lead",
-     "CPU fallback this round (GPU segmented scans pending)"),
+     "fixed-width + string + decimal128 keys on GPU"),
+    ("Join", "hash equi-join inner/left/semi/anti/full, broadcast or "
+     "shuffled build side", "fixed-width + string + decimal128 keys on GPU"),
+    ("CrossJoin", "cartesian gather maps (+ filter for non-equi)", "all"),
+    ("Sort", "stable LSD radix sort", "fixed-width keys on GPU; "
+     "string/decimal128 sort keys fall back"),
+    ("Window", "ranking / running + bounded + partition aggregates / "
+     "lag / lead", "GPU segmented scans; running/bounded min-max on CPU"),
     ("Limit", "row limit", "all"),
     ("Union", "concat", "all"),
+    ("Sample", "deterministic murmur3 Bernoulli sample", "all"),
 ]
 
 
@@ -53,10 +56,12 @@ def supported_ops_doc() -> str:
                  "VM, regex.hip, CPU fallback outside the subset)")
     lines += ["", "## Aggregate functions", "",
               "`sum`, `count`, `count(*)`, `min`, `max`, `avg` "
-              "(partial/merge lowering; mean as sum+count)", "",
+              "plus `stddev`/`variance` (partial/merge lowering; mean as sum+count, "
+              "variance as sum+sumsq+count)", "",
               "## Window functions", "",
               "`row_number`, `rank`, `dense_rank`, `sum`, `count`, `min`, "
-              "`max`, `avg` (running + whole partition), `lag`, `lead`", ""]
+              "`max`, `avg` (running, bounded ROWS BETWEEN, whole "
+              "partition), `lag`, `lead`", ""]
     return "\n".join(lines)
 
 

@@ -109,7 +109,7 @@ def test_gpu_decode_matches_pyarrow(tmp_path, compression, dict_encode,
     p = str(tmp_path / "t.parquet")
     tbl = _write_file(p, n=20_000, compression=compression,
                       dict_encode=dict_encode, page_version=page_version,
-                      with_strings=dict_encode, row_group_size=7000)
+                      with_strings=True, row_group_size=7000)
     names = [n for n in tbl.schema.names]
     batch = read_parquet_gpu(p, names).cpu()
     for i, name in enumerate(names):
