@@ -79,6 +79,9 @@ void hipdf_join_fill(int, const void*, const void*, const void*, int,
                      void*, void*, int64_t, hipStream_t);
 int64_t part_num_blocks(int64_t);
 int64_t sort_num_blocks(int64_t);
+void hipdf_dec64_mul_div(int, const void*, const void*, const void*,
+                         const void*, void*, void*, int, int, int, int64_t,
+                         hipStream_t);
 void hipdf_i128_arith(int, const void*, const void*, const void*, const void*,
                       void*, void*, int64_t, hipStream_t);
 void hipdf_i128_cmp(int, const void*, const void*, const void*, const void*,
@@ -457,6 +460,14 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
 
+  m.def("dec64_mul_div", [](int is_div, int64_t a, int64_t b, int64_t av,
+                            int64_t bv, int64_t out, int64_t ov,
+                            int out_is_128, int shift, int out_prec,
+                            int64_t n, int64_t stream) {
+    hipdf_dec64_mul_div(is_div, P(a), P(b), P(av), P(bv), PM(out), PM(ov),
+                        out_is_128, shift, out_prec, n, S(stream));
+    check_async();
+  });
   m.def("i128_arith", [](int op, int64_t a, int64_t b, int64_t av, int64_t bv,
                          int64_t out, int64_t ov, int64_t n, int64_t stream) {
     hipdf_i128_arith(op, P(a), P(b), P(av), P(bv), PM(out), PM(ov), n,

@@ -176,7 +176,105 @@ __global__ void k_gb_sum_i128(const int64_t* __restrict__ vals,
   }
 }
 
+// ---- decimal multiply / divide -------------------------------------------
+// Spark DecimalPrecision result scales (reference analogue: GpuMultiply /
+// GpuDivide over cudf fixed-point). Operands are decimal64 (int64 backing,
+// each at its own scale); native __int128 keeps the math exact:
+//   mul: r = HALF_UP((x*y) / 10^shift)         shift = s1+s2-st >= 0
+//   div: r = HALF_UP((x*10^shift) / y)         shift = st+s2-s1 (may be <0)
+// The div numerator is built by chunked long division so x*10^shift never
+// has to fit in 128 bits. NULL on divide-by-zero and on overflow of
+// 10^out_prec (Spark non-ANSI overflow -> null).
+
+typedef unsigned __int128 u128;
+
+__device__ __forceinline__ u128 pow10_128(int p) {
+  u128 r = 1;
+  for (int i = 0; i < p; ++i) r *= 10;
+  return r;
+}
+
+__global__ void k_dec64_mul_div(int is_div, const int64_t* __restrict__ a,
+                                const int64_t* __restrict__ b,
+                                const uint64_t* __restrict__ av,
+                                const uint64_t* __restrict__ bv,
+                                int64_t* __restrict__ out,
+                                uint64_t* __restrict__ ov, int out_is_128,
+                                int shift, int out_prec, int64_t nstripe,
+                                int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  const u128 u128_max = ~(u128)0;
+  for (int64_t s = wave_global; s < nstripe; s += wave_count) {
+    int64_t row = s * WAVE + lane;
+    bool ok = false;
+    if (row < n) {
+      ok = valid_bit(av, row) && valid_bit(bv, row);
+      u128 q = 0;
+      bool neg = false;
+      if (ok) {
+        int64_t xs = a[row], ys = b[row];
+        neg = (xs < 0) != (ys < 0);
+        u128 x = (u128)(xs < 0 ? -(__int128)xs : (__int128)xs);
+        u128 y = (u128)(ys < 0 ? -(__int128)ys : (__int128)ys);
+        if (!is_div) {
+          q = x * y;  // |x|,|y| < 10^19 -> fits
+          if (shift > 0) {
+            u128 d = pow10_128(shift);
+            q = (2 * q + d) / (2 * d);
+          }
+        } else if (ys == 0) {
+          ok = false;
+        } else {
+          u128 den = y;
+          int m = shift;
+          if (m < 0) {
+            den = den * pow10_128(-m);
+            m = 0;
+          }
+          q = x / den;
+          u128 r = x % den;
+          while (m > 0) {
+            int c = m > 18 ? 18 : m;
+            u128 p = pow10_128(c);
+            if (q > u128_max / p) {
+              q = u128_max;  // forces the precision-overflow null below
+              break;
+            }
+            q = q * p + (r * p) / den;
+            r = (r * p) % den;
+            m -= c;
+          }
+          if (2 * r >= den) q += 1;  // HALF_UP on the magnitude
+        }
+        if (ok && q >= pow10_128(out_prec)) ok = false;
+      }
+      __int128 v = ok ? (neg ? -(__int128)q : (__int128)q) : 0;
+      if (out_is_128) {
+        out[2 * row] = (int64_t)(u128)v;
+        out[2 * row + 1] = (int64_t)((u128)v >> 64);
+      } else {
+        out[row] = (int64_t)v;
+      }
+    }
+    uint64_t ballot = __ballot(ok);
+    write_valid_word(ov, s, ballot, lane);
+  }
+}
+
 extern "C" {
+
+void hipdf_dec64_mul_div(int is_div, const void* a, const void* b,
+                         const void* av, const void* bv, void* out, void* ov,
+                         int out_is_128, int shift, int out_prec, int64_t n,
+                         hipStream_t stream) {
+  int64_t nstripe = (n + WAVE - 1) / WAVE;
+  hipLaunchKernelGGL(k_dec64_mul_div, stripe_grid(nstripe), dim3(HIPDF_BLOCK),
+                     0, stream, is_div, (const int64_t*)a, (const int64_t*)b,
+                     (const uint64_t*)av, (const uint64_t*)bv, (int64_t*)out,
+                     (uint64_t*)ov, out_is_128, shift, out_prec, nstripe, n);
+}
 
 void hipdf_i128_arith(int op, const void* a, const void* b, const void* av,
                       const void* bv, void* out, void* ov, int64_t n,

@@ -294,9 +294,12 @@ class Column:
         if self.dtype.id is TypeId.BOOL:
             arr = arr.astype(bool)
         if self.dtype.is_decimal:
+            import decimal
+
             scale = self.dtype.scale
             return [
-                (int(v) / (10 ** scale) if scale else int(v)) if ok else None
+                (decimal.Decimal(int(v)).scaleb(-scale) if scale else int(v))
+                if ok else None
                 for v, ok in zip(arr, valid)
             ]
         return [v.item() if ok else None for v, ok in zip(arr, valid)]

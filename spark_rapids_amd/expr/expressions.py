@@ -10,7 +10,8 @@ from __future__ import annotations
 from typing import List, Optional, Sequence
 
 from ..column import Column, ColumnBatch, Schema
-from ..types import BOOL, DType, FLOAT64, INT32, INT64, STRING, TypeId, promote
+from ..types import (BOOL, DType, FLOAT64, INT32, INT64, STRING, TypeId,
+                     as_decimal, decimal_arith_type, promote)
 from .. import ops
 
 
@@ -198,6 +199,14 @@ class Literal(Expression):
         return repr(self.value)
 
 
+def _decimal_exact(lt: DType, rt: DType) -> bool:
+    """True when mul/div should use exact decimal arithmetic (at least one
+    decimal operand, the other decimal or integral)."""
+    return (lt.is_decimal or rt.is_decimal) \
+        and (lt.is_decimal or lt.is_integral) \
+        and (rt.is_decimal or rt.is_integral)
+
+
 # ops whose result is boolean
 _BOOL_OPS = {"eq", "ne", "lt", "le", "gt", "ge", "and", "or", "eq_null_safe"}
 # ops that force double output (Spark `/`)
@@ -216,10 +225,14 @@ class BinaryExpr(Expression):
 
     def _common(self, lt: DType, rt: DType) -> DType:
         if (lt.is_decimal or rt.is_decimal) and self.op in (
-                "mul", "div", "int_div", "mod", "pmod", "pow"):
+                "int_div", "mod", "pmod", "pow"):
             raise NotImplementedError(
                 f"decimal {self.op} needs scale arithmetic (not implemented "
                 "yet): cast to double first, e.g. col.cast(FLOAT64)")
+        if (lt.is_decimal or rt.is_decimal) and self.op in ("mul", "div"):
+            # mixed decimal/floating (the exact-decimal path handles
+            # decimal/integral): Spark casts the decimal side to double
+            return FLOAT64
         if lt.id is TypeId.NULL:
             return rt
         if rt.id is TypeId.NULL:
@@ -238,9 +251,13 @@ class BinaryExpr(Expression):
         # more than once makes deep expression chains exponential
         if self.op in _BOOL_OPS:
             return BOOL
+        lt, rt = self.left.dtype(schema), self.right.dtype(schema)
+        if self.op in ("mul", "div") and _decimal_exact(lt, rt):
+            # Spark DecimalPrecision rules: scale s1+s2 (mul) /
+            # max(6, s1+p2+1) (div), precision-loss adjustment at 38
+            return decimal_arith_type(self.op, lt, rt)
         if self.op in _DOUBLE_OPS:
             return FLOAT64
-        lt, rt = self.left.dtype(schema), self.right.dtype(schema)
         if self.op == "sub" and lt.is_timelike and rt.is_timelike:
             return INT32  # datediff domain
         it = self._common(lt, rt)
@@ -254,6 +271,15 @@ class BinaryExpr(Expression):
     _SWAP_CMP = {"lt": "gt", "gt": "lt", "le": "ge", "ge": "le"}
 
     def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        if self.op in ("mul", "div"):
+            lt, rt = self.left.dtype(schema), self.right.dtype(schema)
+            if _decimal_exact(lt, rt):
+                # operands keep their own scales; the backend computes the
+                # exact product/quotient at the Spark result scale
+                lcol = ops.cast(self.left.eval(batch, schema), as_decimal(lt))
+                rcol = ops.cast(self.right.eval(batch, schema), as_decimal(rt))
+                return ops.decimal_mul_div(
+                    self.op, lcol, rcol, decimal_arith_type(self.op, lt, rt))
         common = self._in_dtype(schema)
         if self.op in _DOUBLE_OPS and not common.is_decimal:
             common = FLOAT64

@@ -46,6 +46,14 @@ def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
     return backend_for(lhs).binary_op_scalar(op, lhs, scalar, out_dtype)
 
 
+def decimal_mul_div(op: str, lhs: Column, rhs: Column,
+                    out_dtype: DType) -> Column:
+    """Exact decimal multiply/divide at the Spark result scale (operands
+    keep their own scales; reference analogue: GpuMultiply/GpuDivide over
+    cudf fixed-point with Spark's DecimalPrecision typing)."""
+    return backend_for(lhs, rhs).decimal_mul_div(op, lhs, rhs, out_dtype)
+
+
 def unary_op(op: str, col: Column, out_dtype: Optional[DType] = None) -> Column:
     return backend_for(col).unary_op(op, col, out_dtype or col.dtype)
 

@@ -86,6 +86,50 @@ def binary_op(op: str, lhs: Column, rhs: Column, out_dtype: DType) -> Column:
     return _binary_impl(op, a, av, b, bv, lhs.dtype, out_dtype)
 
 
+def decimal_mul_div(op: str, lhs: Column, rhs: Column,
+                    out_dtype: DType) -> Column:
+    """Exact decimal multiply/divide with python-int arithmetic.
+
+    value(lhs)=a/10^s1, value(rhs)=b/10^s2; result at out_dtype.scale with
+    HALF_UP rounding; NULL on divide-by-zero and on overflow of
+    out_dtype.precision (Spark non-ANSI overflow semantics).
+    """
+    a, av = _vals(lhs), _valid(lhs)
+    b, bv = _vals(rhs), _valid(rhs)
+    s1, s2, st = lhs.dtype.scale, rhs.dtype.scale, out_dtype.scale
+    valid = av & bv
+    bound = 10 ** out_dtype.precision
+    res = np.zeros(len(a), dtype=object)
+    for i in range(len(a)):
+        if not valid[i]:
+            continue
+        x, y = int(a[i]), int(b[i])
+        if op == "mul":
+            r = x * y
+            delta = s1 + s2 - st
+            if delta > 0:
+                r = _div_half_up(r, 10 ** delta)
+        else:
+            if y == 0:
+                valid[i] = False
+                continue
+            m = st + s2 - s1
+            num, den = (x * 10 ** m, y) if m >= 0 else (x, y * 10 ** (-m))
+            r = _div_half_up(num, den)
+        if -bound < r < bound:
+            res[i] = r
+        else:
+            valid[i] = False
+    return _make(res, valid if not valid.all() else None, out_dtype)
+
+
+def _div_half_up(num: int, den: int) -> int:
+    """Signed integer division rounding HALF_UP (away from zero on .5)."""
+    sign = -1 if (num < 0) != (den < 0) else 1
+    num, den = abs(num), abs(den)
+    return sign * ((2 * num + den) // (2 * den))
+
+
 def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
     a, av = _vals(lhs), _valid(lhs)
     if scalar is None:

@@ -123,6 +123,33 @@ def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
 _STR_CMP = {"eq": 0, "ne": 1, "lt": 2, "le": 3, "gt": 4, "ge": 5}
 
 
+def decimal_mul_div(op: str, lhs: Column, rhs: Column,
+                    out_dtype: DType) -> Column:
+    """Exact decimal multiply/divide (dec64 operands) on device; see
+    k_dec64_mul_div in decimal128.hip. NULL on div-by-zero / overflow."""
+    if lhs.dtype.id is not TypeId.DECIMAL64 \
+            or rhs.dtype.id is not TypeId.DECIMAL64:
+        raise NotImplementedError(
+            "GPU decimal mul/div supports decimal64 operands (overrides "
+            "keep wider operands on CPU)")
+    n = lhs.size
+    s = _stream()
+    s1, s2, st = lhs.dtype.scale, rhs.dtype.scale, out_dtype.scale
+    shift = (s1 + s2 - st) if op == "mul" else (st + s2 - s1)
+    is128 = out_dtype.id is TypeId.DECIMAL128
+    width = 2 * n if is128 else n
+    out = torch.empty(max(width, 1), dtype=torch.int64,
+                      device="cuda")[:width]
+    ov = _alloc_mask(n)
+    if n:
+        ext.dec64_mul_div(1 if op == "div" else 0, lhs.data.data_ptr(),
+                          rhs.data.data_ptr(), _ptr(lhs.validity),
+                          _ptr(rhs.validity), out.data_ptr(), ov.data_ptr(),
+                          1 if is128 else 0, shift, out_dtype.precision,
+                          n, s)
+    return Column(out_dtype, n, out, ov, null_count=None)
+
+
 def _binary(op, lhs: Column, rhs: Optional[Column], scalar, out_dtype) -> Column:
     n = lhs.size
     s = _stream()

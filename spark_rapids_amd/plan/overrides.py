@@ -13,11 +13,12 @@ from typing import List, Optional
 
 from ..config import RapidsConf
 from ..config import ALLOW_INCOMPAT as _ALLOW_INCOMPAT
-from ..expr.expressions import (Alias, BinaryExpr, CaseWhen, CastExpr,
+from ..expr.expressions import (_decimal_exact,
+                                Alias, BinaryExpr, CaseWhen, CastExpr,
                                 Coalesce, ColumnRef, Expression, IsNull,
                                 Literal, Round, StringPredicate, Substring,
                                 UnaryExpr)
-from ..types import DType, TypeId, TypeSig
+from ..types import DType, TypeId, TypeSig, as_decimal
 from . import logical as L
 from . import physical as P
 
@@ -87,13 +88,23 @@ class Tagger:
             if not self.conf.expr_enabled("Cast"):
                 out.append("expression Cast disabled by conf")
         elif isinstance(e, BinaryExpr):
-            in_t = e._in_dtype(schema)
-            if e.op not in _GPU_BINARY_OPS:
-                out.append(f"binary op {e.op} has no GPU kernel")
-            elif in_t.id is TypeId.STRING and e.op not in _GPU_STRING_OK:
-                out.append(f"binary op {e.op} on string not on GPU yet")
-            elif in_t.id is TypeId.DECIMAL128 and e.op not in _D128_BINARY_OK:
-                out.append(f"binary op {e.op} on decimal128 not on GPU yet")
+            lt, rt = e.left.dtype(schema), e.right.dtype(schema)
+            if e.op in ("mul", "div") and _decimal_exact(lt, rt):
+                # exact decimal mul/div: GPU kernel covers dec64 operands
+                if not (as_decimal(lt).id is TypeId.DECIMAL64
+                        and as_decimal(rt).id is TypeId.DECIMAL64):
+                    out.append(f"decimal {e.op} with >18-digit operands "
+                               "not on GPU yet")
+            else:
+                in_t = e._in_dtype(schema)
+                if e.op not in _GPU_BINARY_OPS:
+                    out.append(f"binary op {e.op} has no GPU kernel")
+                elif in_t.id is TypeId.STRING and e.op not in _GPU_STRING_OK:
+                    out.append(f"binary op {e.op} on string not on GPU yet")
+                elif in_t.id is TypeId.DECIMAL128 \
+                        and e.op not in _D128_BINARY_OK:
+                    out.append(f"binary op {e.op} on decimal128 "
+                               "not on GPU yet")
             if not self.conf.expr_enabled(e.op):
                 out.append(f"expression {e.op} disabled by conf")
         elif isinstance(e, UnaryExpr):

@@ -202,6 +202,51 @@ _PROMOTION_ORDER = [
 ]
 
 
+# integral -> decimal equivalents for mixed decimal arithmetic
+# (reference analogue: Spark DecimalPrecision.scala integral promotion)
+_INT_AS_DECIMAL = {
+    TypeId.INT8: (3, 0),
+    TypeId.INT16: (5, 0),
+    TypeId.INT32: (10, 0),
+    TypeId.INT64: (20, 0),
+}
+
+
+def as_decimal(dt: DType) -> DType:
+    """The decimal type an operand takes in decimal arithmetic."""
+    if dt.is_decimal:
+        return dt
+    if dt.id in _INT_AS_DECIMAL:
+        p, sc = _INT_AS_DECIMAL[dt.id]
+        return DType.decimal(p, sc)
+    raise TypeError(f"{dt} cannot participate in decimal arithmetic")
+
+
+def _adjust_decimal(p: int, s: int) -> DType:
+    """Spark allowPrecisionLoss=true adjustment when precision > 38
+    (DecimalType.adjustPrecisionScale)."""
+    if p <= 38:
+        return DType.decimal(p, s)
+    int_digits = p - s
+    min_scale = min(s, 6)
+    adj_scale = max(38 - int_digits, min_scale)
+    return DType.decimal(38, adj_scale)
+
+
+def decimal_arith_type(op: str, lt: DType, rt: DType) -> DType:
+    """Result type of decimal multiply / divide per Spark's
+    DecimalPrecision rules (sql.decimalOperations.allowPrecisionLoss=true,
+    the default)."""
+    a, b = as_decimal(lt), as_decimal(rt)
+    p1, s1, p2, s2 = a.precision, a.scale, b.precision, b.scale
+    if op == "mul":
+        return _adjust_decimal(p1 + p2 + 1, s1 + s2)
+    if op == "div":
+        s = max(6, s1 + p2 + 1)
+        return _adjust_decimal(p1 - s1 + s2 + s, s)
+    raise ValueError(f"decimal_arith_type: unsupported op {op}")
+
+
 def promote(a: DType, b: DType) -> DType:
     """Common wider type for arithmetic between a and b (non-decimal path)."""
     if a == b:

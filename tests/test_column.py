@@ -48,4 +48,5 @@ def test_batch_ragged_assert():
 def test_decimal_pylist():
     d = DType.decimal(10, 2)
     c = Column.from_pylist([12345, None], d)
-    assert c.to_pylist() == [123.45, None]
+    from decimal import Decimal
+    assert c.to_pylist() == [Decimal("123.45"), None]

@@ -1,0 +1,148 @@
+"""Decimal multiply/divide with Spark DecimalPrecision scale arithmetic.
+
+Reference analogue: GpuMultiply/GpuDivide + DecimalUtil.scala typing
+(allowPrecisionLoss=true). CPU tests check exact HALF_UP semantics against
+python Decimal; gpu tests check the dec64 __int128 kernel against the CPU
+backend.
+"""
+import decimal
+from decimal import Decimal
+
+import pytest
+
+import spark_rapids_amd as sr
+from spark_rapids_amd import DType, col
+from spark_rapids_amd.types import decimal_arith_type
+
+decimal.getcontext().prec = 60
+
+D72 = DType.decimal(7, 2)
+D104 = DType.decimal(10, 4)
+
+
+@pytest.fixture
+def cpu():
+    return sr.Session({"spark.rapids.sql.enabled": False})
+
+
+def test_result_types():
+    assert decimal_arith_type("mul", D72, D72) == DType.decimal(15, 4)
+    # div: s = max(6, 2+7+1) = 10, p = 7-2+2+10 = 17
+    assert decimal_arith_type("div", D72, D72) == DType.decimal(17, 10)
+    # big operands trigger the precision-loss adjustment
+    d38 = DType.decimal(38, 10)
+    t = decimal_arith_type("mul", d38, d38)
+    assert t.precision == 38 and t.scale == 6
+    # int32 operand promotes to decimal(10,0)
+    assert decimal_arith_type("mul", D72, sr.INT32) == DType.decimal(18, 2)
+
+
+def _dec_df(s, a_vals, b_vals, adt=D72, bdt=D72):
+    df = s.create_dataframe({"a": a_vals, "b": b_vals})
+    return df.select(col("a").cast(adt).alias("a"),
+                     col("b").cast(bdt).alias("b"))
+
+
+def test_cpu_mul_exact(cpu):
+    df = _dec_df(cpu, [1.25, -3.10, None, 99999.99], [2.00, 0.07, 4.0, 0.07])
+    out = df.select((col("a") * col("b")).alias("m")).to_pydict()["m"]
+    assert out[0] == Decimal("2.5000")
+    assert out[1] == Decimal("-0.2170")
+    assert out[2] is None
+    assert out[3] == Decimal("6999.9993")
+
+
+def test_cpu_div_half_up_and_null(cpu):
+    df = _dec_df(cpu, [1.00, 1.00, 5.00], [3.00, 0.00, 2.00])
+    out = df.select((col("a") / col("b")).alias("d")).to_pydict()["d"]
+    assert out[0] == Decimal("0.3333333333")  # scale 10
+    assert out[1] is None                     # div by zero -> NULL
+    assert out[2] == Decimal("2.5000000000")
+
+
+def test_cpu_matches_python_decimal(cpu):
+    import numpy as np
+
+    rng = np.random.default_rng(3)
+    a = [round(float(x), 2) for x in rng.uniform(-9999, 9999, 300)]
+    b = [round(float(x), 2) for x in rng.uniform(-99, 99, 300)]
+    b[7] = 0.0
+    df = _dec_df(cpu, a, b)
+    got = df.select((col("a") * col("b")).alias("m"),
+                    (col("a") / col("b")).alias("d")).to_pydict()
+    for i in range(300):
+        da = Decimal(f"{a[i]:.2f}")
+        db = Decimal(f"{b[i]:.2f}")
+        exp_m = (da * db).quantize(Decimal("0.0001"),
+                                   rounding=decimal.ROUND_HALF_UP)
+        assert got["m"][i] == exp_m, (i, a[i], b[i])
+        if b[i] == 0.0:
+            assert got["d"][i] is None
+        else:
+            exp_d = (da / db).quantize(Decimal("0.0000000001"),
+                                       rounding=decimal.ROUND_HALF_UP)
+            assert got["d"][i] == exp_d, (i, a[i], b[i])
+
+
+def test_cpu_decimal_times_int(cpu):
+    df = cpu.create_dataframe({"a": [1.25, 2.50], "q": [3, -4]})
+    df = df.select(col("a").cast(D72).alias("a"), col("q").alias("q"))
+    out = df.select((col("a") * col("q")).alias("m")).to_pydict()["m"]
+    assert out == [Decimal("3.75"), Decimal("-10.00")]
+
+
+def test_cpu_decimal_times_float_is_double(cpu):
+    df = _dec_df(cpu, [1.25], [1.0])
+    out = df.select((col("a") * 2.0).alias("m"))
+    assert out.schema.fields[0].dtype == sr.FLOAT64
+    assert out.to_pydict()["m"] == [2.5]
+
+
+def test_cpu_overflow_is_null(cpu):
+    d = DType.decimal(38, 0)
+    s = cpu
+    df = s.create_dataframe({"a": [1.0], "b": [1.0]})
+    df = df.select(col("a").cast(d).alias("a"), col("b").cast(d).alias("b"))
+    # (38,0)*(38,0) -> adjusted to (38,6): any value >= 10^32 overflows
+    big = s.create_dataframe({"x": [1]})
+    from spark_rapids_amd.expr.expressions import Literal
+
+    lit = Literal(10 ** 20, DType.decimal(38, 0))
+    out = big.select((lit * lit).alias("m")).to_pydict()["m"]
+    assert out == [None]
+
+
+@pytest.mark.gpu
+def test_gpu_matches_cpu_mul_div():
+    import numpy as np
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    rng = np.random.default_rng(11)
+    a = [round(float(x), 2) for x in rng.uniform(-99999, 99999, 20000)]
+    b = [round(float(x), 4) for x in rng.uniform(-50, 50, 20000)]
+    for i in range(0, 20000, 97):
+        b[i] = 0.0
+    for q in (
+        lambda df: df.select((col("a") * col("b")).alias("r")),
+        lambda df: df.select((col("a") / col("b")).alias("r")),
+        lambda df: df.select((col("b") / col("a")).alias("r")),
+        lambda df: df.filter(col("a") * col("b") > col("a")),
+    ):
+        g = q(_dec_df(sg, a, b, D72, D104)).to_pydict()
+        c = q(_dec_df(sc, a, b, D72, D104)).to_pydict()
+        assert g == c
+
+
+@pytest.mark.gpu
+def test_gpu_mul_output_d128_and_placement():
+    sg = sr.Session()
+    # (10,4)*(10,4) -> (21,8): decimal128 output from dec64 operands
+    df = _dec_df(sg, [123456.7891, -1.0], [99999.9999, 3.0], D104, D104)
+    out = df.select((col("a") * col("b")).alias("m"))
+    assert out.schema.fields[0].dtype == DType.decimal(21, 8)
+    tree = out.physical_plan().tree_string()
+    assert "GpuProject" in tree, tree
+    got = out.to_pydict()["m"]
+    assert got[0] == Decimal("123456.7891") * Decimal("99999.9999")
+    assert got[1] == Decimal("-3.00000000")
