@@ -44,8 +44,8 @@ def q3_selective_revenue(t: Dict[str, DataFrame]) -> DataFrame:
                       & (col("ss_discount") >= 0.05)
                       & (col("ss_discount") <= 0.07)
                       & (col("ss_quantity") < 24))
-            .agg(sum_(col("ss_list_price").cast(FLOAT64)
-                      * col("ss_discount"))))
+            # decimal * double -> double (Spark mixed-type rule)
+            .agg(sum_(col("ss_list_price") * col("ss_discount"))))
 
 
 def q4_customer_rollup(t: Dict[str, DataFrame]) -> DataFrame:
@@ -63,8 +63,9 @@ def q5_store_join(t: Dict[str, DataFrame]) -> DataFrame:
             .join(item, on="ss_item_id", right_on=["i_item_id"])
             .filter(col("s_state") < 25)
             .group_by("s_state", "i_category")
-            .agg(sum_(col("ss_sales_price").cast(FLOAT64)
-                      * col("ss_quantity").cast(FLOAT64)),
+            # exact decimal revenue: dec(7,2) * int32 -> dec(18,2),
+            # summed into decimal128 (the dec64_mul_div __int128 kernel)
+            .agg(sum_(col("ss_sales_price") * col("ss_quantity")),
                  count_star()))
 
 

@@ -99,6 +99,25 @@ def decimal_mul_div(op: str, lhs: Column, rhs: Column,
     s1, s2, st = lhs.dtype.scale, rhs.dtype.scale, out_dtype.scale
     valid = av & bv
     bound = 10 ** out_dtype.precision
+    # vectorized fast paths (no int64 overflow possible by the type rules)
+    if a.dtype != object and b.dtype != object:
+        if op == "mul" and lhs.dtype.precision + rhs.dtype.precision + 1 <= 18:
+            r = a.astype(np.int64) * b.astype(np.int64)
+            delta = s1 + s2 - st
+            if delta > 0:
+                r = _round_half_up_div(r, 10 ** delta)
+            return _make(r, valid if not valid.all() else None, out_dtype)
+        m = st + s2 - s1
+        if op == "div" and 0 <= m and lhs.dtype.precision + m <= 18 \
+                and out_dtype.precision <= 18:
+            valid = valid & (b != 0)
+            num = a.astype(np.int64) * (10 ** m)
+            den = np.where(b == 0, 1, b.astype(np.int64))
+            sign = np.where((num < 0) != (den < 0), -1, 1)
+            q = sign * ((2 * np.abs(num) + np.abs(den)) //
+                        (2 * np.abs(den)))
+            valid = valid & (np.abs(q) < bound)
+            return _make(q, valid if not valid.all() else None, out_dtype)
     res = np.zeros(len(a), dtype=object)
     for i in range(len(a)):
         if not valid[i]:
@@ -416,10 +435,11 @@ def cast(col: Column, to: DType) -> Column:
 
 
 def _round_half_up_div(a: np.ndarray, d: int) -> np.ndarray:
-    q = a // d
-    r = a - q * d
-    adj = (np.abs(r) * 2 >= d).astype(np.int64) * np.sign(a)
-    return q + np.where(np.sign(a) < 0, np.minimum(adj, 0), np.maximum(adj, 0))
+    """HALF_UP (round half away from zero) on the magnitude: the floor-based
+    formulation is wrong for negatives (-5/2 must give -3, not -4)."""
+    sign = np.where(a < 0, -1, 1)
+    aa = np.abs(a)
+    return sign * ((2 * aa + d) // (2 * d))
 
 
 def _dec_str(unscaled: int, scale: int) -> str:

@@ -146,3 +146,29 @@ def test_gpu_mul_output_d128_and_placement():
     got = out.to_pydict()["m"]
     assert got[0] == Decimal("123456.7891") * Decimal("99999.9999")
     assert got[1] == Decimal("-3.00000000")
+
+
+def test_cpu_downscale_cast_half_up_negatives(cpu):
+    # regression: floor-based rounding gave -2.25 -> -2.3 via -23? no: -4/2
+    # style errors; HALF_UP must round magnitude away from zero
+    df = cpu.create_dataframe({"a": [2.25, -2.25, 0.05, -0.05]})
+    df = df.select(col("a").cast(DType.decimal(9, 2)).alias("a"))
+    out = df.select(
+        col("a").cast(DType.decimal(9, 1)).alias("r")).to_pydict()["r"]
+    assert out == [Decimal("2.3"), Decimal("-2.3"),
+                   Decimal("0.1"), Decimal("-0.1")]
+
+
+@pytest.mark.gpu
+def test_gpu_downscale_cast_half_up_negatives():
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    vals = [2.25, -2.25, 0.05, -0.05, 123.455, -123.455, None]
+    for s in (sg, sc):
+        pass
+    def q(s):
+        df = s.create_dataframe({"a": vals})
+        df = df.select(col("a").cast(DType.decimal(9, 3)).alias("a"))
+        return df.select(col("a").cast(DType.decimal(9, 1)).alias("r"),
+                         (col("a") * col("a")).alias("sq")).to_pydict()
+    assert q(sg) == q(sc)
