@@ -19,7 +19,7 @@ from .expr.expressions import ColumnRef, Expression, col as _col
 from .memory.semaphore import GpuSemaphore
 from .plan import logical as L
 from .plan.overrides import plan_physical
-from .types import DType
+from .types import DType, INT32
 
 
 class MemTable:
@@ -72,6 +72,28 @@ class DataFrame:
     def group_by(self, *keys: Union[str, Expression]) -> "GroupedData":
         es = [(_col(k) if isinstance(k, str) else k) for k in keys]
         return GroupedData(self, es)
+
+    def rollup(self, *keys: str) -> "GroupedData":
+        """Hierarchical grouping sets: (k1..kn), (k1..kn-1), ..., ().
+        Lowered to Expand + hash aggregate with a spark_grouping_id column
+        (reference analogue: GpuExpandExec under Aggregate for ROLLUP)."""
+        n = len(keys)
+        sets = [tuple(keys[:i]) for i in range(n, -1, -1)]
+        return GroupedData(self, [_col(k) for k in keys],
+                           grouping_sets=sets)
+
+    def cube(self, *keys: str) -> "GroupedData":
+        """All 2^n grouping-set combinations (n <= 10)."""
+        if len(keys) > 10:
+            raise ValueError("cube supports at most 10 keys")
+        import itertools
+
+        sets = []
+        for r in range(len(keys), -1, -1):
+            for combo in itertools.combinations(keys, r):
+                sets.append(combo)
+        return GroupedData(self, [_col(k) for k in keys],
+                           grouping_sets=sets)
 
     def agg(self, *aggs: AggExpr) -> "DataFrame":
         return DataFrame(self.session, L.Aggregate([], list(aggs), self.plan))
@@ -186,13 +208,39 @@ class DataFrame:
 
 
 class GroupedData:
-    def __init__(self, df: DataFrame, keys: List[Expression]):
+    def __init__(self, df: DataFrame, keys: List[Expression],
+                 grouping_sets=None):
         self.df = df
         self.keys = keys
+        self.grouping_sets = grouping_sets
 
     def agg(self, *aggs: AggExpr) -> DataFrame:
+        if self.grouping_sets is None:
+            return DataFrame(self.df.session,
+                             L.Aggregate(self.keys, list(aggs), self.df.plan))
+        from .expr.expressions import Alias, Literal
+
+        child = self.df.plan
+        cs = child.schema()
+        key_names = [k.output_name() for k in self.keys]
+        projections = []
+        for kept in self.grouping_sets:
+            gid = 0
+            for i, k in enumerate(key_names):
+                if k not in kept:
+                    gid |= 1 << (len(key_names) - 1 - i)
+            proj = []
+            for f in cs.fields:
+                if f.name in key_names and f.name not in kept:
+                    proj.append(Alias(Literal(None, f.dtype), f.name))
+                else:
+                    proj.append(Alias(_col(f.name), f.name))
+            proj.append(Alias(Literal(gid, INT32), "spark_grouping_id"))
+            projections.append(proj)
+        expand = L.Expand(projections, child)
+        group = self.keys + [_col("spark_grouping_id")]
         return DataFrame(self.df.session,
-                         L.Aggregate(self.keys, list(aggs), self.df.plan))
+                         L.Aggregate(group, list(aggs), expand))
 
 
 class Session:

@@ -251,6 +251,13 @@ class Column:
 
     @staticmethod
     def nulls(dtype: DType, size: int, device: str = "cpu") -> "Column":
+        if dtype.id is TypeId.STRING:
+            data = torch.zeros(0, dtype=torch.uint8, device=device)
+            offsets = torch.zeros(size + 1, dtype=torch.int32, device=device)
+            validity = torch.zeros(mask_nbytes(size), dtype=torch.uint8,
+                                   device=device)
+            return Column(dtype, size, data, validity, offsets,
+                          null_count=size)
         nwords = 2 * size if dtype.id is TypeId.DECIMAL128 else size
         data = torch.zeros(nwords, dtype=torch_dtype(dtype), device=device)
         validity = torch.zeros(mask_nbytes(size), dtype=torch.uint8, device=device)

@@ -205,6 +205,32 @@ class Limit(LogicalPlan):
         return self.child.schema()
 
 
+class Expand(LogicalPlan):
+    """GROUPING SETS row expansion: each input row is emitted once per
+    projection (all projections share one output schema). Reference
+    analogue: GpuExpandExec (sql-plugin .../GpuExpandExecMeta) backing
+    rollup / cube / grouping sets."""
+
+    def __init__(self, projections: List[List[Expression]],
+                 child: LogicalPlan):
+        assert projections and all(
+            len(p) == len(projections[0]) for p in projections)
+        self.projections = projections
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        cs = self.child.schema()
+        fields = []
+        for i, e in enumerate(self.projections[0]):
+            nullable = any(p[i].nullable(cs) for p in self.projections)
+            fields.append(Field(e.output_name(), e.dtype(cs), nullable))
+        return Schema(fields)
+
+
 class Union(LogicalPlan):
     def __init__(self, plans: List[LogicalPlan]):
         self.plans = plans

@@ -154,6 +154,10 @@ class Tagger:
         elif isinstance(node, L.Project):
             for e in node.exprs:
                 reasons += self.expr_reasons(e, cs)
+        elif isinstance(node, L.Expand):
+            for proj in node.projections:
+                for e in proj:
+                    reasons += self.expr_reasons(e, cs)
         elif isinstance(node, L.Aggregate):
             for e in node.group_exprs:
                 r = _HASH_KEYS.supports(e.dtype(cs))
@@ -296,6 +300,8 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
         return P.FilterExec(device, node.condition, kids[0])
     if isinstance(node, L.Project):
         return P.ProjectExec(device, node.exprs, kids[0], node.schema())
+    if isinstance(node, L.Expand):
+        return P.ExpandExec(device, node.projections, kids[0], node.schema())
     if isinstance(node, L.Aggregate):
         from ..config import BATCH_SIZE_BYTES
 

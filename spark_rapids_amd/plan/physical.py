@@ -669,6 +669,28 @@ class LimitExec(PhysicalExec):
         return f"{self.name()}({self.n})"
 
 
+class ExpandExec(PhysicalExec):
+    """Evaluates every projection over each input batch and concatenates:
+    an N-projection expand emits N output rows per input row."""
+
+    def __init__(self, device: str, projections, child: PhysicalExec,
+                 schema: Schema):
+        super().__init__(device, schema, [child])
+        self.projections = projections
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        in_schema = self.children[0].schema
+        for batch in self.children[0].execute():
+            parts = []
+            for proj in self.projections:
+                cols = [e.eval(batch, in_schema) for e in proj]
+                parts.append(ColumnBatch(cols, batch.num_rows))
+            yield parts[0] if len(parts) == 1 else ops.concat_batches(parts)
+
+    def describe(self):
+        return f"{self.name()}[{len(self.projections)} projections]"
+
+
 class UnionExec(PhysicalExec):
     def __init__(self, device: str, children: List[PhysicalExec], schema: Schema):
         super().__init__(device, schema, children)
