@@ -26,6 +26,7 @@ void hipdf_binary_bool(int, const void*, const void*, int, int, const void*,
 void hipdf_unary(int, int, const void*, const void*, void*, void*, int64_t,
                  hipStream_t);
 void hipdf_cast(int, int, const void*, void*, int64_t, hipStream_t);
+void hipdf_f64_to_i64_rint(const void*, void*, int64_t, hipStream_t);
 void hipdf_decimal_rescale(const void*, void*, int64_t, int, int64_t,
                            hipStream_t);
 void hipdf_if_else(int, const void*, const void*, const void*, const void*,
@@ -88,6 +89,9 @@ void hipdf_i128_cmp(int, const void*, const void*, const void*, const void*,
                     void*, void*, int64_t, hipStream_t);
 void hipdf_i64_to_i128(const void*, void*, int64_t, hipStream_t);
 void hipdf_i128_to_f64(const void*, void*, int64_t, hipStream_t);
+void hipdf_gb_sum_i128_lds(int, const void*, const void*, const void*,
+                           const void*, void*, void*, int, int64_t,
+                           hipStream_t);
 void hipdf_gb_sum_i64_to_i128(const void*, const void*, const void*,
                               const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_gb_sum_i128(const void*, const void*, const void*, const void*,
@@ -187,6 +191,11 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("cast", [](int ft, int tt, int64_t a, int64_t out, int64_t n,
                    int64_t stream) {
     hipdf_cast(ft, tt, P(a), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("f64_to_i64_rint", [](int64_t a, int64_t out, int64_t n,
+                              int64_t stream) {
+    hipdf_f64_to_i64_rint(P(a), PM(out), n, S(stream));
     check_async();
   });
   m.def("decimal_rescale", [](int64_t a, int64_t out, int64_t pow10, bool up,
@@ -486,6 +495,14 @@ PYBIND11_MODULE(hipdf, m) {
   });
   m.def("i128_to_f64", [](int64_t in, int64_t out, int64_t n, int64_t stream) {
     hipdf_i128_to_f64(P(in), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("gb_sum_i128_lds", [](int in_is_64, int64_t vals, int64_t vvalid,
+                              int64_t row_gid, int64_t sel, int64_t acc,
+                              int64_t cnt, int ngroups, int64_t n,
+                              int64_t stream) {
+    hipdf_gb_sum_i128_lds(in_is_64, P(vals), P(vvalid), P(row_gid), P(sel),
+                          PM(acc), PM(cnt), ngroups, n, S(stream));
     check_async();
   });
   m.def("gb_sum_i64_to_i128", [](int64_t vals, int64_t vvalid, int64_t row_gid,

@@ -176,6 +176,60 @@ __global__ void k_gb_sum_i128(const int64_t* __restrict__ vals,
   }
 }
 
+// LDS-staged grouped i128 sum for small group counts: each block
+// accumulates into shared memory (24 B/group: lo, hi, cnt) and flushes
+// once, turning ~n global atomics on a few hot addresses into
+// ngroups-per-block. in_is_64: values are int64 (widened) vs i128 pairs.
+__global__ void k_gb_sum_i128_lds(int in_is_64,
+                                  const int64_t* __restrict__ vals,
+                                  const uint64_t* __restrict__ vvalid,
+                                  const int32_t* __restrict__ row_gid,
+                                  const int32_t* __restrict__ sel,
+                                  int64_t* __restrict__ acc,
+                                  int64_t* __restrict__ cnt, int ngroups,
+                                  int64_t n) {
+  extern __shared__ unsigned long long smem[];
+  unsigned long long* s_lo = smem;
+  unsigned long long* s_hi = smem + ngroups;
+  unsigned long long* s_cnt = smem + 2 * ngroups;
+  for (int g = threadIdx.x; g < 3 * ngroups; g += blockDim.x) smem[g] = 0;
+  __syncthreads();
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
+    if (!valid_bit(vvalid, i)) continue;
+    int32_t g = row_gid[j];
+    uint64_t lo;
+    int64_t hi;
+    if (in_is_64) {
+      int64_t v = vals[i];
+      lo = (uint64_t)v;
+      hi = v < 0 ? -1 : 0;
+    } else {
+      i128 v = load128(vals, i);
+      lo = v.lo;
+      hi = v.hi;
+    }
+    unsigned long long old = atomicAdd(&s_lo[g], (unsigned long long)lo);
+    if (old + lo < old) atomicAdd(&s_hi[g], 1ull);
+    if (hi) atomicAdd(&s_hi[g], (unsigned long long)hi);
+    atomicAdd(&s_cnt[g], 1ull);
+  }
+  __syncthreads();
+  for (int g = threadIdx.x; g < ngroups; g += blockDim.x) {
+    unsigned long long lo = s_lo[g];
+    if (lo) {
+      unsigned long long old =
+          atomicAdd((unsigned long long*)&acc[2 * g], lo);
+      if (old + lo < old)
+        atomicAdd((unsigned long long*)&acc[2 * g + 1], 1ull);
+    }
+    if (s_hi[g])
+      atomicAdd((unsigned long long*)&acc[2 * g + 1], s_hi[g]);
+    if (s_cnt[g]) atomicAdd((unsigned long long*)&cnt[g], s_cnt[g]);
+  }
+}
+
 // ---- decimal multiply / divide -------------------------------------------
 // Spark DecimalPrecision result scales (reference analogue: GpuMultiply /
 // GpuDivide over cudf fixed-point). Operands are decimal64 (int64 backing,
@@ -304,6 +358,20 @@ void hipdf_i128_to_f64(const void* in, void* out, int64_t n,
                        hipStream_t stream) {
   hipLaunchKernelGGL(k_i128_to_f64, flat_grid(n), dim3(HIPDF_BLOCK), 0,
                      stream, (const int64_t*)in, (double*)out, n);
+}
+
+void hipdf_gb_sum_i128_lds(int in_is_64, const void* vals,
+                           const void* vvalid, const void* row_gid,
+                           const void* sel, void* acc, void* cnt,
+                           int ngroups, int64_t n, hipStream_t stream) {
+  size_t shmem = (size_t)ngroups * 24;
+  int64_t blocks = (n + HIPDF_BLOCK - 1) / HIPDF_BLOCK;
+  if (blocks > 2048) blocks = 2048;
+  hipLaunchKernelGGL(k_gb_sum_i128_lds, dim3((uint32_t)blocks),
+                     dim3(HIPDF_BLOCK), shmem, stream, in_is_64,
+                     (const int64_t*)vals, (const uint64_t*)vvalid,
+                     (const int32_t*)row_gid, (const int32_t*)sel,
+                     (int64_t*)acc, (int64_t*)cnt, ngroups, n);
 }
 
 void hipdf_gb_sum_i64_to_i128(const void* vals, const void* vvalid,

@@ -337,6 +337,20 @@ __global__ void k_cast_to_bool(const From* __restrict__ a,
     out[i] = a[i] != (From)0 ? 1 : 0;
 }
 
+// f64 -> i64 with round-to-nearest-even (float -> decimal cast; matches
+// the CPU backend's np.round; plain k_cast truncates like Spark int casts)
+__global__ void k_f64_to_i64_rint(const double* __restrict__ a,
+                                  int64_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    double v = rint(a[i]);
+    if (v != v) out[i] = 0;
+    else if (v >= 9.223372036854775807e18) out[i] = INT64_MAX;
+    else if (v <= -9.223372036854775808e18) out[i] = INT64_MIN;
+    else out[i] = (int64_t)v;
+  }
+}
+
 // decimal rescale: out = in * 10^k (k>0) or round-half-up(in / 10^-k)
 __global__ void k_decimal_rescale(const int64_t* __restrict__ a,
                                   int64_t* __restrict__ out, int64_t pow10,
@@ -516,6 +530,12 @@ void hipdf_cast(int from_t, int to_t, const void* a, void* out, int64_t n,
                          stream, (const From*)a, (To*)out, n);
     });
   });
+}
+
+void hipdf_f64_to_i64_rint(const void* a, void* out, int64_t n,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(k_f64_to_i64_rint, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const double*)a, (int64_t*)out, n);
 }
 
 void hipdf_decimal_rescale(const void* a, void* out, int64_t pow10, int up,

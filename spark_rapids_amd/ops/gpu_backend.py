@@ -424,7 +424,9 @@ def _cast_decimal(col: Column, to: DType, v) -> Column:
     scaled = binary_op_scalar("mul", dbl, float(10 ** to.scale),
                               DType.float64())
     out = _alloc(n, to)
-    ext.cast(6, 4, scaled.data.data_ptr(), out.data_ptr(), n, s)
+    # round-to-nearest (not the truncating int cast): 40.5074 * 10^4 is
+    # 405073.9999.. in binary and must become 405074
+    ext.f64_to_i64_rint(scaled.data.data_ptr(), out.data_ptr(), n, s)
     return Column(to, n, out, v, null_count=col._null_count)
 
 
@@ -814,15 +816,23 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
             cnt = torch.zeros(max(ngroups, 1), dtype=torch.int64,
                               device="cuda")
             if n:
-                if vc.dtype.id is TypeId.DECIMAL128:
-                    ext.gb_sum_i128(vc.data.data_ptr(), _ptr(vc.validity),
-                                    row_gid.data_ptr(), selp, acc.data_ptr(),
-                                    cnt.data_ptr(), n, s)
-                else:
+                in64 = 0 if vc.dtype.id is TypeId.DECIMAL128 else 1
+                if ngroups <= 2048:
+                    # LDS-staged: one global flush per block instead of
+                    # per-row atomics on a few hot accumulators
+                    ext.gb_sum_i128_lds(
+                        in64, vc.data.data_ptr(), _ptr(vc.validity),
+                        row_gid.data_ptr(), selp, acc.data_ptr(),
+                        cnt.data_ptr(), ngroups, n, s)
+                elif in64:
                     ext.gb_sum_i64_to_i128(
                         vc.data.data_ptr(), _ptr(vc.validity),
                         row_gid.data_ptr(), selp, acc.data_ptr(),
                         cnt.data_ptr(), n, s)
+                else:
+                    ext.gb_sum_i128(vc.data.data_ptr(), _ptr(vc.validity),
+                                    row_gid.data_ptr(), selp, acc.data_ptr(),
+                                    cnt.data_ptr(), n, s)
             allocs.append(("sum_d128", out_dtype, False, acc, cnt))
             continue
         acc_is_double = out_dtype.is_floating or (
