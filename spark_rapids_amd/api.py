@@ -96,6 +96,8 @@ class DataFrame:
                            grouping_sets=sets)
 
     def agg(self, *aggs: AggExpr) -> "DataFrame":
+        if any(a.distinct for a in aggs):
+            return GroupedData(self, [])._agg_distinct(list(aggs))
         return DataFrame(self.session, L.Aggregate([], list(aggs), self.plan))
 
     def join(self, other: "DataFrame", on: Union[str, Sequence[str]],
@@ -215,6 +217,10 @@ class GroupedData:
         self.grouping_sets = grouping_sets
 
     def agg(self, *aggs: AggExpr) -> DataFrame:
+        if any(a.distinct for a in aggs):
+            if self.grouping_sets is not None:
+                raise NotImplementedError("distinct aggs with rollup/cube")
+            return self._agg_distinct(list(aggs))
         if self.grouping_sets is None:
             return DataFrame(self.df.session,
                              L.Aggregate(self.keys, list(aggs), self.df.plan))
@@ -241,6 +247,46 @@ class GroupedData:
         group = self.keys + [_col("spark_grouping_id")]
         return DataFrame(self.df.session,
                          L.Aggregate(group, list(aggs), expand))
+
+    def _agg_distinct(self, aggs: List[AggExpr]) -> DataFrame:
+        """Single-distinct-column rewrite (Spark RewriteDistinctAggregates):
+        inner aggregate groups by (keys, d) computing partials of the
+        non-distinct aggs; the outer aggregate merges them and counts/sums
+        the now-unique d values."""
+        from .expr.expressions import Alias
+
+        dists = [a for a in aggs if a.distinct]
+        if len({str(d.child) for d in dists}) != 1:
+            raise NotImplementedError(
+                "multiple DISTINCT columns in one aggregate")
+        d = dists[0].child
+        inner_aggs: List[AggExpr] = []
+        outer: List[Optional[AggExpr]] = []
+        for i, a in enumerate(aggs):
+            if a.distinct:
+                if a.op not in ("count", "sum"):
+                    raise NotImplementedError(f"{a.op}(DISTINCT) unsupported")
+                outer.append(AggExpr(a.op, _col("__dist__"),
+                                     a.output_name()))
+                continue
+            pname = f"__p{i}"
+            if a.op == "count_all":
+                inner_aggs.append(AggExpr("count_all", None, pname))
+                outer.append(AggExpr("sum", _col(pname), a.output_name()))
+            elif a.op == "count":
+                inner_aggs.append(AggExpr("count", a.child, pname))
+                outer.append(AggExpr("sum", _col(pname), a.output_name()))
+            elif a.op in ("sum", "min", "max"):
+                inner_aggs.append(AggExpr(a.op, a.child, pname))
+                outer.append(AggExpr(a.op, _col(pname), a.output_name()))
+            else:
+                raise NotImplementedError(
+                    f"{a.op} mixed with DISTINCT aggs unsupported")
+        inner_keys = self.keys + [Alias(d, "__dist__")]
+        inner = L.Aggregate(inner_keys, inner_aggs, self.df.plan)
+        outer_keys = [_col(k.output_name()) for k in self.keys]
+        return DataFrame(self.df.session,
+                         L.Aggregate(outer_keys, outer, inner))
 
 
 class Session:

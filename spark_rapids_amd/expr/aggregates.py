@@ -16,13 +16,15 @@ from .expressions import Expression, _as_expr
 
 
 class AggExpr:
-    def __init__(self, op: str, child: Optional[Expression], name: Optional[str] = None):
+    def __init__(self, op: str, child: Optional[Expression],
+                 name: Optional[str] = None, distinct: bool = False):
         self.op = op
         self.child = _as_expr(child) if child is not None else None
         self._name = name
+        self.distinct = distinct
 
     def alias(self, name: str) -> "AggExpr":
-        return AggExpr(self.op, self.child, name)
+        return AggExpr(self.op, self.child, name, self.distinct)
 
     _DISPLAY = {"count_all": "count", "mean": "avg"}
 
@@ -32,6 +34,8 @@ class AggExpr:
         disp = self._DISPLAY.get(self.op, self.op)
         if self.child is None:
             return f"{disp}(*)"
+        if self.distinct:
+            return f"{disp}(DISTINCT {self.child})"
         return f"{disp}({self.child})"
 
     def out_dtype(self, schema: Schema) -> DType:
@@ -86,3 +90,14 @@ def stddev(e) -> AggExpr:
 def variance(e) -> AggExpr:
     """var_samp"""
     return AggExpr("variance", e)
+
+
+def count_distinct(e) -> AggExpr:
+    """count(DISTINCT e): lowered to a two-level aggregate (dedupe on
+    (keys, e) then count). Reference analogue: RewriteDistinctAggregates'
+    single-distinct plan executed by two GpuHashAggregates."""
+    return AggExpr("count", e, distinct=True)
+
+
+def sum_distinct(e) -> AggExpr:
+    return AggExpr("sum", e, distinct=True)
