@@ -89,6 +89,11 @@ void hipdf_i128_cmp(int, const void*, const void*, const void*, const void*,
                     void*, void*, int64_t, hipStream_t);
 void hipdf_i64_to_i128(const void*, void*, int64_t, hipStream_t);
 void hipdf_i128_to_f64(const void*, void*, int64_t, hipStream_t);
+void hipdf_gb_collect_count(const void*, const void*, const void*, void*,
+                            int64_t, hipStream_t);
+void hipdf_gb_collect_fill(int, const void*, const void*, const void*,
+                           const void*, const void*, void*, void*, int64_t,
+                           hipStream_t);
 void hipdf_gb_sum_i128_lds(int, const void*, const void*, const void*,
                            const void*, void*, void*, int, int64_t,
                            hipStream_t);
@@ -495,6 +500,20 @@ PYBIND11_MODULE(hipdf, m) {
   });
   m.def("i128_to_f64", [](int64_t in, int64_t out, int64_t n, int64_t stream) {
     hipdf_i128_to_f64(P(in), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("gb_collect_count", [](int64_t vvalid, int64_t row_gid, int64_t sel,
+                               int64_t counts, int64_t n, int64_t stream) {
+    hipdf_gb_collect_count(P(vvalid), P(row_gid), P(sel), PM(counts), n,
+                           S(stream));
+    check_async();
+  });
+  m.def("gb_collect_fill", [](int esize, int64_t vals, int64_t vvalid,
+                              int64_t row_gid, int64_t sel, int64_t offsets,
+                              int64_t cursor, int64_t out, int64_t n,
+                              int64_t stream) {
+    hipdf_gb_collect_fill(esize, P(vals), P(vvalid), P(row_gid), P(sel),
+                          P(offsets), PM(cursor), PM(out), n, S(stream));
     check_async();
   });
   m.def("gb_sum_i128_lds", [](int in_is_64, int64_t vals, int64_t vvalid,

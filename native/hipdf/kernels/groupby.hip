@@ -297,7 +297,69 @@ __global__ void k_mask_from_nonzero(const int64_t* __restrict__ cnt,
   }
 }
 
+// ---- collect_list / collect_set ------------------------------------------
+// Two-pass grouped gather into a LIST column: count valid values per group,
+// host scans counts into offsets, then an atomic per-group cursor places
+// each value (order within a group is unspecified, matching Spark).
+__global__ void k_gb_collect_count(const uint64_t* __restrict__ vvalid,
+                                   const int32_t* __restrict__ row_gid,
+                                   const int32_t* __restrict__ sel,
+                                   int64_t* __restrict__ counts, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
+    if (!valid_bit(vvalid, i)) continue;
+    atomicAdd((unsigned long long*)&counts[row_gid[j]], 1ull);
+  }
+}
+
+__global__ void k_gb_collect_fill(int esize, const uint8_t* __restrict__ vals,
+                                  const uint64_t* __restrict__ vvalid,
+                                  const int32_t* __restrict__ row_gid,
+                                  const int32_t* __restrict__ sel,
+                                  const int64_t* __restrict__ offsets,
+                                  int64_t* __restrict__ cursor,
+                                  uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
+    if (!valid_bit(vvalid, i)) continue;
+    int32_t g = row_gid[j];
+    int64_t pos = offsets[g] +
+        (int64_t)atomicAdd((unsigned long long*)&cursor[g], 1ull);
+    switch (esize) {
+      case 1: out[pos] = vals[i]; break;
+      case 2: ((uint16_t*)out)[pos] = ((const uint16_t*)vals)[i]; break;
+      case 4: ((uint32_t*)out)[pos] = ((const uint32_t*)vals)[i]; break;
+      case 8: ((uint64_t*)out)[pos] = ((const uint64_t*)vals)[i]; break;
+      case 16:
+        ((ulonglong2*)out)[pos] = ((const ulonglong2*)vals)[i];
+        break;
+    }
+  }
+}
+
 extern "C" {
+
+void hipdf_gb_collect_count(const void* vvalid, const void* row_gid,
+                            const void* sel, void* counts, int64_t n,
+                            hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_collect_count, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const uint64_t*)vvalid, (const int32_t*)row_gid,
+                     (const int32_t*)sel, (int64_t*)counts, n);
+}
+
+void hipdf_gb_collect_fill(int esize, const void* vals, const void* vvalid,
+                           const void* row_gid, const void* sel,
+                           const void* offsets, void* cursor, void* out,
+                           int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_collect_fill, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, esize, (const uint8_t*)vals,
+                     (const uint64_t*)vvalid, (const int32_t*)row_gid,
+                     (const int32_t*)sel, (const int64_t*)offsets,
+                     (int64_t*)cursor, (uint8_t*)out, n);
+}
+
 
 void hipdf_gb_build(const void* hashes, const void* keys, int nkeys,
                     const void* sel, void* slot_row, void* row_slot,

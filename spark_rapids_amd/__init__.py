@@ -8,7 +8,8 @@ spark.rapids.* config registry.
 from .api import DataFrame, Session
 from .column import Column, ColumnBatch, Field, Schema
 from .config import RapidsConf, help_doc
-from .expr.aggregates import (avg, count, count_distinct, count_star,
+from .expr.aggregates import (avg, collect_list, collect_set,
+                              count, count_distinct, count_star,
                               max_, min_, stddev, sum_distinct,
                               sum_, variance)
 from .expr.expressions import (CaseWhen, coalesce, col, date_add, date_sub,

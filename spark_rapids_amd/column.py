@@ -99,7 +99,8 @@ def unpack_validity(mask: torch.Tensor, size: int) -> np.ndarray:
 class Column:
     """One column of data; immutable by convention."""
 
-    __slots__ = ("dtype", "size", "data", "validity", "offsets", "_null_count")
+    __slots__ = ("dtype", "size", "data", "validity", "offsets",
+                 "_null_count", "child")
 
     def __init__(
         self,
@@ -109,6 +110,7 @@ class Column:
         validity: Optional[torch.Tensor] = None,
         offsets: Optional[torch.Tensor] = None,
         null_count: Optional[int] = None,
+        child: Optional["Column"] = None,
     ):
         self.dtype = dtype
         self.size = size
@@ -116,8 +118,11 @@ class Column:
         self.validity = validity
         self.offsets = offsets
         self._null_count = null_count
+        self.child = child  # LIST element column
         if dtype.id is TypeId.STRING:
             assert offsets is not None and offsets.numel() == size + 1
+        if dtype.id is TypeId.LIST:
+            assert offsets is not None and child is not None
 
     # ---- properties ---------------------------------------------------
     @property
@@ -162,6 +167,7 @@ class Column:
             None if self.validity is None else self.validity.to(device, non_blocking=non_blocking),
             None if self.offsets is None else self.offsets.to(device, non_blocking=non_blocking),
             self._null_count,
+            None if self.child is None else self.child.to(device, non_blocking=non_blocking),
         )
 
     def cuda(self) -> "Column":
@@ -194,6 +200,19 @@ class Column:
     @staticmethod
     def from_pylist(values: Sequence, dtype: DType, device: str = "cpu") -> "Column":
         n = len(values)
+        if dtype.id is TypeId.LIST:
+            valid = np.array([v is not None for v in values], dtype=bool)
+            flat: list = []
+            offsets = np.zeros(n + 1, dtype=np.int32)
+            for i, v in enumerate(values):
+                if v is not None:
+                    flat.extend(v)
+                offsets[i + 1] = len(flat)
+            child = Column.from_pylist(flat, dtype.children[0])
+            col = Column(dtype, n, torch.zeros(0, dtype=torch.uint8),
+                         make_validity(valid) if not valid.all() else None,
+                         torch.from_numpy(offsets), None, child)
+            return col.to(device) if device != "cpu" else col
         if dtype.id is TypeId.STRING:
             valid = np.array([v is not None for v in values], dtype=bool)
             parts = [(v if v is not None else "").encode("utf-8") for v in values]
@@ -279,6 +298,11 @@ class Column:
 
     def to_pylist(self) -> list:
         valid = self.valid_array()
+        if self.dtype.id is TypeId.LIST:
+            offs = self.offsets.cpu().numpy()
+            elems = self.child.to_pylist()
+            return [list(elems[offs[i]:offs[i + 1]]) if valid[i] else None
+                    for i in range(self.size)]
         if self.dtype.id is TypeId.STRING:
             offs = self.offsets.cpu().numpy()
             raw = self.data.cpu().numpy().tobytes()

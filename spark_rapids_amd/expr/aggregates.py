@@ -52,6 +52,8 @@ class AggExpr:
             return FLOAT64
         if self.op in ("min", "max", "first", "last"):
             return ct
+        if self.op in ("collect_list", "collect_set"):
+            return DType.list_(ct)
         raise NotImplementedError(f"agg {self.op}")
 
     def __str__(self):
@@ -101,3 +103,14 @@ def count_distinct(e) -> AggExpr:
 
 def sum_distinct(e) -> AggExpr:
     return AggExpr("sum", e, distinct=True)
+
+
+def collect_list(e) -> AggExpr:
+    """Gather the group's non-null values into an array (order unspecified,
+    like Spark). Reference analogue: GpuCollectList."""
+    return AggExpr("collect_list", e)
+
+
+def collect_set(e) -> AggExpr:
+    """Gather the group's distinct non-null values (GpuCollectSet)."""
+    return AggExpr("collect_set", e)

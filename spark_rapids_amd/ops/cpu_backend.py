@@ -758,6 +758,15 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
             res = np.bincount(codes[av], minlength=ngroups).astype(np.int64)
             out_cols.append(_make(res, None, out_dtype))
             continue
+        if op in ("collect_list", "collect_set"):
+            lists: list = [[] for _ in range(ngroups)]
+            for g, v, ok in zip(codes, vc.to_pylist(), av):
+                if ok:
+                    lists[g].append(v)
+            if op == "collect_set":
+                lists = [list(dict.fromkeys(l)) for l in lists]
+            out_cols.append(Column.from_pylist(lists, out_dtype))
+            continue
         if batch.columns[vidx].dtype.id is TypeId.DECIMAL128 and \
                 op not in ("sum", "count", "count_all"):
             raise NotImplementedError(

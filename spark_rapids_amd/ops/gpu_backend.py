@@ -835,6 +835,11 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                                     cnt.data_ptr(), n, s)
             allocs.append(("sum_d128", out_dtype, False, acc, cnt))
             continue
+        if op in ("collect_list", "collect_set"):
+            col = _gb_collect(vc, row_gid, selp, n, ngroups, out_dtype,
+                              op == "collect_set", s)
+            allocs.append(("collect", out_dtype, False, col, None))
+            continue
         acc_is_double = out_dtype.is_floating or (
             vc is not None and vc.dtype.is_floating)
         acc = torch.empty(max(ngroups, 1),
@@ -856,6 +861,9 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         ext.gb_agg_multi(desc.data_ptr(), len(blobs), row_gid.data_ptr(),
                          selp, ngroups, n, s)
     for op, out_dtype, acc_is_double, acc, cnt in allocs:
+        if op == "collect":
+            out_cols.append(acc)
+            continue
         if op == "sum_d128":
             ov = _alloc_mask(ngroups)
             ext.mask_from_nonzero(cnt.data_ptr(), ov.data_ptr(), ngroups, s)
@@ -879,6 +887,50 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         out_cols.append(Column(out_dtype, ngroups, out_data, ov,
                                null_count=None))
     return ColumnBatch(out_cols, ngroups)
+
+
+def _gb_collect(vc: Column, row_gid: torch.Tensor, selp, n: int,
+                ngroups: int, out_dtype, is_set: bool, s) -> Column:
+    """collect_list/collect_set: count-per-group + scan + atomic-cursor
+    scatter (k_gb_collect_* in groupby.hip). collect_set dedupes first by
+    grouping the (gid, value) pairs with the generic hash groupby."""
+    elem_dt = out_dtype.children[0]
+    if is_set:
+        if selp:
+            raise NotImplementedError(
+                "collect_set under filter fusion: the single-pass exec "
+                "path materializes filtered batches, so sel is never set")
+        # dedupe the (gid, value) pairs with the generic hash groupby, then
+        # collect the unique pairs using gid as a direct group index
+        gid_col = Column(DType.int32(), n, row_gid[:n], None, null_count=0)
+        pair = ColumnBatch([gid_col, vc], n)
+        uniq = group_by_aggregate(pair, [0, 1], [])
+        ug, uv = uniq.columns[0], uniq.columns[1]
+        return _gb_collect(uv, ug.data[:uniq.num_rows], None,
+                           uniq.num_rows, ngroups, out_dtype, False, s)
+    counts = torch.zeros(max(ngroups, 1), dtype=torch.int64, device="cuda")
+    if n:
+        ext.gb_collect_count(_ptr(vc.validity), row_gid.data_ptr(),
+                             selp or 0, counts.data_ptr(), n, s)
+    scanned, total = _exclusive_scan_i64(counts[:ngroups]) if ngroups         else (counts, 0)
+    esize = 16 if elem_dt.id is TypeId.DECIMAL128 else         torch.empty(0, dtype=torch_dtype(elem_dt)).element_size()
+    nelem = 2 * total if elem_dt.id is TypeId.DECIMAL128 else total
+    out_data = torch.empty(max(nelem, 1), dtype=torch_dtype(elem_dt),
+                           device="cuda")[:nelem]
+    cursor = torch.zeros(max(ngroups, 1), dtype=torch.int64, device="cuda")
+    if n and total:
+        ext.gb_collect_fill(esize, vc.data.data_ptr(), _ptr(vc.validity),
+                            row_gid.data_ptr(), selp or 0,
+                            scanned.data_ptr(), cursor.data_ptr(),
+                            out_data.data_ptr(), n, s)
+    offs = torch.empty(ngroups + 1, dtype=torch.int32, device="cuda")
+    if ngroups:
+        ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), ngroups, s)
+    offs[ngroups] = total
+    child = Column(elem_dt, total, out_data, None, null_count=0)
+    return Column(out_dtype, ngroups, torch.zeros(0, dtype=torch.uint8,
+                                                  device="cuda"),
+                  None, offs, 0, child)
 
 
 # ---------------------------------------------------------------------------

@@ -167,6 +167,12 @@ class Tagger:
             for a in node.aggs:
                 if a.child is not None:
                     t = a.child.dtype(cs)
+                    if a.op in ("collect_list", "collect_set"):
+                        if t.id is TypeId.STRING or t.is_nested:
+                            reasons.append(
+                                f"{a.op} over {t} not on GPU yet")
+                        reasons += self.expr_reasons(a.child, cs)
+                        continue
                     r = _NUMERIC.supports(t)
                     if r and a.op not in ("count", "count_all", "min", "max"):
                         reasons.append(f"agg {a.op}({a.child}): {r}")

@@ -223,6 +223,14 @@ def _lower_aggs(aggs: List[AggExpr], in_schema: Schema):
             partial.append(("count", v, INT64))
             merge.extend(["sum", "sum"])
             final.append(("div", js, js + 1, ct))
+        elif a.op in ("collect_list", "collect_set"):
+            value_exprs.append(a.child)
+            j = len(partial)
+            partial.append((a.op, len(value_exprs) - 1,
+                            a.out_dtype(in_schema)))
+            merge.append("__single_pass__")  # lists are never merged:
+            # HashAggregateExec routes collect aggs to the one-shot path
+            final.append(("col", j))
         elif a.op in ("stddev", "variance"):
             from ..expr.expressions import BinaryExpr, CastExpr
             from ..types import FLOAT64 as F64
@@ -271,6 +279,11 @@ class HashAggregateExec(PhysicalExec):
         in_schema = source.schema
         nkeys = len(self.group_exprs)
         value_exprs, partial, merge_ops, final = _lower_aggs(self.aggs, in_schema)
+        if any(op in ("collect_list", "collect_set") for op, _, _ in partial):
+            yield from self._execute_single_pass(
+                source, fused_condition, in_schema, value_exprs, partial,
+                final)
+            return
 
         from ..memory.spill import SpillableBatch
 
@@ -347,31 +360,81 @@ class HashAggregateExec(PhysicalExec):
                 continue
             merged = ops.group_by_aggregate(bucket, list(range(nkeys)),
                                             merge_specs)
-            out_cols: List[Column] = [merged.columns[i] for i in range(nkeys)]
-            for spec, agg in zip(final, self.aggs):
-                if spec[0] == "col":
-                    c = merged.columns[nkeys + spec[1]]
-                    out_cols.append(ops.cast(c, agg.out_dtype(cs)))
-                elif spec[0] == "div":
-                    s = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
-                    c = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
-                    out_cols.append(ops.binary_op("div", s, c, FLOAT64))
-                elif spec[0] == "var":
-                    # sample variance from (sum, sumsq, count):
-                    # (sumsq - sum^2/n) / (n-1); NULL when n < 2
-                    sm = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
-                    sq = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
-                    cn = ops.cast(merged.columns[nkeys + spec[3]], FLOAT64)
-                    mean_sq = ops.binary_op(
-                        "div", ops.binary_op("mul", sm, sm, FLOAT64), cn,
-                        FLOAT64)
-                    num = ops.binary_op("sub", sq, mean_sq, FLOAT64)
-                    den = ops.binary_op_scalar("sub", cn, 1.0, FLOAT64)
-                    var = ops.binary_op("div", num, den, FLOAT64)
-                    if spec[4]:
-                        var = ops.unary_op("sqrt", var, FLOAT64)
-                    out_cols.append(var)
-            yield ColumnBatch(out_cols, merged.num_rows)
+            yield self._final_project(merged, nkeys, final, cs)
+
+    def _final_project(self, merged: ColumnBatch, nkeys: int, final,
+                       cs: Schema) -> ColumnBatch:
+        out_cols: List[Column] = [merged.columns[i] for i in range(nkeys)]
+        for spec, agg in zip(final, self.aggs):
+            if spec[0] == "col":
+                c = merged.columns[nkeys + spec[1]]
+                out_cols.append(ops.cast(c, agg.out_dtype(cs)))
+            elif spec[0] == "div":
+                s = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
+                c = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
+                out_cols.append(ops.binary_op("div", s, c, FLOAT64))
+            elif spec[0] == "var":
+                # sample variance from (sum, sumsq, count):
+                # (sumsq - sum^2/n) / (n-1); NULL when n < 2
+                sm = ops.cast(merged.columns[nkeys + spec[1]], FLOAT64)
+                sq = ops.cast(merged.columns[nkeys + spec[2]], FLOAT64)
+                cn = ops.cast(merged.columns[nkeys + spec[3]], FLOAT64)
+                mean_sq = ops.binary_op(
+                    "div", ops.binary_op("mul", sm, sm, FLOAT64), cn,
+                    FLOAT64)
+                num = ops.binary_op("sub", sq, mean_sq, FLOAT64)
+                den = ops.binary_op_scalar("sub", cn, 1.0, FLOAT64)
+                var = ops.binary_op("div", num, den, FLOAT64)
+                if spec[4]:
+                    var = ops.unary_op("sqrt", var, FLOAT64)
+                out_cols.append(var)
+        return ColumnBatch(out_cols, merged.num_rows)
+
+    def _execute_single_pass(self, source, fused_condition, in_schema,
+                             value_exprs, partial, final):
+        """One-shot aggregation for list-building aggs (collect_list/set):
+        rows are exchanged by key hash FIRST (distributed), then aggregated
+        once — list partials cannot be merged."""
+        nkeys = len(self.group_exprs)
+        pres: List[ColumnBatch] = []
+        for batch in source.execute():
+            if fused_condition is not None:
+                mask = fused_condition.eval(batch, in_schema)
+                batch = ops.apply_boolean_mask(batch, mask)
+            key_cols = [e.eval(batch, in_schema) for e in self.group_exprs]
+            val_cols = [e.eval(batch, in_schema) for e in value_exprs]
+            pres.append(ColumnBatch(key_cols + val_cols, batch.num_rows))
+        from ..shuffle import dist as _dist
+        dist_on = _dist.ctx().is_multi and not self.input_replicated
+        if not pres:
+            if not dist_on:
+                return
+            pres = []  # still must join the collectives below
+        pre = ops.concat_batches(pres) if len(pres) > 1 else (
+            pres[0] if pres else None)
+        if dist_on:
+            from ..shuffle.exchange import exchange_by_hash, gather_all
+            if pre is None:
+                cs0 = self.children[0].schema
+                cols = [Column.from_pylist([], e.dtype(cs0))
+                        for e in list(self.group_exprs) + list(value_exprs)]
+                if self.gpu:
+                    cols = [c.cuda() for c in cols]
+                pre = ColumnBatch(cols, 0)
+            if nkeys:
+                received = exchange_by_hash(pre, list(range(nkeys)))
+            else:
+                received = gather_all(pre)
+            received = [b for b in received if b is not None and b.num_rows]
+            if not received:
+                return
+            pre = ops.concat_batches(received) if len(received) > 1                 else received[0]
+        if pre is None or pre.num_rows == 0:
+            return
+        specs = [(op, (nkeys + v) if v >= 0 else -1, dt)
+                 for op, v, dt in partial]
+        merged = ops.group_by_aggregate(pre, list(range(nkeys)), specs)
+        yield self._final_project(merged, nkeys, final, in_schema)
 
     def _merge_target_bytes(self) -> int:
         return self.merge_target_bytes

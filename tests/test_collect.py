@@ -1,0 +1,89 @@
+"""collect_list / collect_set aggregates producing LIST columns
+(reference analogue: GpuCollectList/GpuCollectSet over cudf lists;
+single-pass aggregation path in HashAggregateExec)."""
+import numpy as np
+import pytest
+
+import spark_rapids_amd as sr
+from spark_rapids_amd import col, collect_list, collect_set, count_star, sum_
+
+
+@pytest.fixture
+def cpu():
+    return sr.Session({"spark.rapids.sql.enabled": False})
+
+
+def _df(s, n=2000):
+    rng = np.random.default_rng(13)
+    return s.create_dataframe({
+        "k": [int(v) for v in rng.integers(0, 9, n)],
+        "v": [int(v) if v != 3 else None for v in rng.integers(0, 30, n)],
+        "w": [float(v) for v in rng.uniform(0, 1, n)],
+    })
+
+
+def _expected(raw):
+    exp = {}
+    for k, v, w in raw:
+        exp.setdefault(k, []).append(v)
+    return exp
+
+
+def test_collect_list_cpu(cpu):
+    df = _df(cpu)
+    raw = df.collect()
+    exp = _expected(raw)
+    for k, lst, st, c in df.group_by("k").agg(
+            collect_list(col("v")), collect_set(col("v")),
+            count_star()).collect():
+        want = [v for v in exp[k] if v is not None]
+        assert sorted(lst) == sorted(want)
+        assert sorted(st) == sorted(set(want))
+        assert c == len(exp[k])
+
+
+def test_collect_empty_group_is_empty_list(cpu):
+    df = cpu.create_dataframe({"k": [1], "v": [None]})
+    rows = df.group_by("k").agg(collect_list(col("v"))).collect()
+    assert rows == [(1, [])]
+
+
+def test_list_column_roundtrip():
+    from spark_rapids_amd import Column, DType, INT64
+
+    lt = DType.list_(INT64)
+    c = Column.from_pylist([[1, 2], [], None, [5]], lt)
+    assert c.to_pylist() == [[1, 2], [], None, [5]]
+
+
+def test_collect_mixed_with_mean(cpu):
+    df = _df(cpu, 500)
+    rows = df.group_by("k").agg(collect_list(col("v")),
+                                sr.avg(col("w"))).collect()
+    raw = df.collect()
+    for k, lst, m in rows:
+        ws = [r[2] for r in raw if r[0] == k]
+        assert m == pytest.approx(sum(ws) / len(ws))
+
+
+@pytest.mark.gpu
+def test_gpu_collect_matches_cpu():
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+
+    def q(s):
+        rows = (_df(s, 20000).group_by("k")
+                .agg(collect_list(col("v")), collect_set(col("v")),
+                     sum_(col("w")), count_star()).collect())
+        return sorted((k, sorted(l), sorted(st), round(sw, 6), c)
+                      for k, l, st, sw, c in rows)
+
+    assert q(sg) == q(sc)
+
+
+@pytest.mark.gpu
+def test_gpu_collect_placement():
+    s = sr.Session()
+    tree = (_df(s, 10).group_by("k").agg(collect_list(col("v")))
+            .physical_plan().tree_string())
+    assert "GpuHashAggregate" in tree, tree
