@@ -38,7 +38,8 @@ _TYPES = {
     "string": STRING,
 }
 
-_AGG_FUNCS = {"sum", "avg", "count", "min", "max", "stddev", "variance"}
+_AGG_FUNCS = {"sum", "avg", "count", "min", "max", "stddev", "variance",
+              "collect_list", "collect_set"}
 
 
 class SqlError(ValueError):
@@ -152,12 +153,23 @@ class Parser:
         if self.kw("WHERE"):
             where = self.parse_expr()
         group_cols = []
+        group_mode = "plain"
         if self.kw("GROUP"):
             self.expect_kw("BY")
-            while True:
-                group_cols.append(self.next()[1])
-                if not self.op(","):
-                    break
+            if self.kw("ROLLUP") or self.kw("CUBE"):
+                group_mode = "rollup" if \
+                    self.toks[self.i - 1][1].upper() == "ROLLUP" else "cube"
+                self.expect_op("(")
+                while True:
+                    group_cols.append(self.next()[1])
+                    if not self.op(","):
+                        break
+                self.expect_op(")")
+            else:
+                while True:
+                    group_cols.append(self.next()[1])
+                    if not self.op(","):
+                        break
         having = None
         if self.kw("HAVING"):
             having = self.parse_expr()
@@ -197,7 +209,14 @@ class Parser:
                 else:
                     raise SqlError(
                         f"non-aggregate select item {e} not in GROUP BY")
-            df = df.group_by(*keys).agg(*aggs) if keys else df.agg(*aggs)
+            if not keys:
+                df = df.agg(*aggs)
+            elif group_mode == "rollup":
+                df = df.rollup(*keys).agg(*aggs)
+            elif group_mode == "cube":
+                df = df.cube(*keys).agg(*aggs)
+            else:
+                df = df.group_by(*keys).agg(*aggs)
         elif not star:
             exprs = [(e.alias(alias) if alias else e) for e, alias in items]
             # ORDER BY may reference pre-projection columns (Spark allows
@@ -406,6 +425,9 @@ class Parser:
         if name == "count" and self.op("*"):
             self.expect_op(")")
             return A.count_star()
+        distinct = False
+        if name in _AGG_FUNCS and self.kw("DISTINCT"):
+            distinct = True
         args = []
         if not self.op(")"):
             while True:
@@ -416,8 +438,13 @@ class Parser:
         if name in _AGG_FUNCS:
             ctor = {"sum": A.sum_, "avg": A.avg, "count": A.count,
                     "min": A.min_, "max": A.max_, "stddev": A.stddev,
-                    "variance": A.variance}[name]
-            return ctor(args[0])
+                    "variance": A.variance,
+                    "collect_list": A.collect_list,
+                    "collect_set": A.collect_set}[name]
+            agg = ctor(args[0])
+            if distinct:
+                agg = A.AggExpr(agg.op, agg.child, distinct=True)
+            return agg
         if name == "coalesce":
             return Coalesce(*args)
         if name == "round":

@@ -96,3 +96,33 @@ def test_sql_runs_on_gpu(s):
     assert q.collect() == [(1, 40.0, ), (2, 20.0), (3, 50.0)] or True
     out = q.collect()
     assert out[0][0] == 1
+
+
+def test_sql_count_distinct(session):
+    s = session
+    df = s.create_dataframe({"k": ["a", "a", "b"], "c": [1, 1, 2]})
+    s.register("tdist", df)
+    out = sorted(s.sql("SELECT k, COUNT(DISTINCT c) FROM tdist GROUP BY k")
+                 .collect())
+    assert out == [("a", 1), ("b", 1)]
+    assert s.sql("SELECT COUNT(DISTINCT c) FROM tdist").collect() == [(2,)]
+
+
+def test_sql_rollup_cube(session):
+    s = session
+    df = s.create_dataframe({"k": ["a", "a", "b"], "v": [1.0, 2.0, 3.0]})
+    s.register("troll", df)
+    rows = sorted(s.sql("SELECT k, SUM(v) FROM troll GROUP BY ROLLUP(k)")
+                  .collect(), key=repr)
+    assert (None, 1, 6.0) in rows and len(rows) == 3
+    rows = s.sql("SELECT k, COUNT(*) FROM troll GROUP BY CUBE(k)").collect()
+    assert len(rows) == 3
+
+
+def test_sql_collect_list(session):
+    s = session
+    df = s.create_dataframe({"k": ["a", "b"], "c": [1, 2]})
+    s.register("tcoll", df)
+    out = sorted(s.sql("SELECT k, collect_list(c) FROM tcoll GROUP BY k")
+                 .collect())
+    assert out == [("a", [1]), ("b", [2])]
