@@ -869,6 +869,36 @@ def _key_series(col: Column):
     return s
 
 
+def range_key(col: Column, desc: bool, nulls_last: bool) -> Column:
+    """Monotone int64 sort-position proxy (see gpu_backend.range_key)."""
+    a, av = _vals(col), _valid(col)
+    n = len(a)
+    if col.dtype.id is TypeId.STRING:
+        key = np.zeros(n, dtype=np.int64)
+        for i, v in enumerate(a):
+            b = (v or "").encode("utf-8")[:8]
+            u = int.from_bytes(b.ljust(8, b"\0"), "big")
+            key[i] = (u ^ (1 << 63)) - (1 << 63)  # bias to signed order
+    elif col.dtype.id is TypeId.DECIMAL128:
+        key = np.array([int(x) >> 64 if x is not None else 0 for x in a],
+                       dtype=np.int64)
+    elif col.dtype.is_floating:
+        f = a.astype(np.float64)
+        bits = f.view(np.int64).copy()
+        neg = bits < 0
+        bits[neg] = np.int64(-0x8000000000000000) - bits[neg] - 1
+        bits[np.isnan(f)] = np.int64(0x7FFFFFFFFFFFFFFE)  # NaN greatest
+        key = bits
+    else:
+        key = a.astype(np.int64)
+    if desc:
+        key = ~key
+    extreme = np.int64(0x7FFFFFFFFFFFFFFF) if nulls_last \
+        else np.int64(-0x8000000000000000)
+    key = np.where(av, key, extreme)
+    return _make(key, None, DType.int64())
+
+
 def sort_order(batch: ColumnBatch, key_idx: List[int],
                descending: List[bool], nulls_last: List[bool]) -> Column:
     n = batch.num_rows

@@ -976,6 +976,27 @@ def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
 # sort: stable LSD radix over order-preserving u64 keys (native/hipdf sort.hip)
 # ---------------------------------------------------------------------------
 
+def range_key(col: Column, desc: bool, nulls_last: bool) -> Column:
+    """Monotone int64 proxy of the sort position of each row (signed order;
+    ties allowed). Used by external sort to range-partition rows into
+    bounded buckets before the in-core sort of each bucket."""
+    n = col.size
+    s = _stream()
+    keys = torch.empty(max(n, 1), dtype=torch.int64, device="cuda")[:n]
+    if n:
+        ext.make_sort_keys(_ht(col.dtype), col.data.data_ptr(),
+                           _ptr(col.validity), 0, desc, nulls_last, False,
+                           keys.data_ptr(), n, s)
+    kc = Column(DType.int64(), n, keys, None, null_count=0)
+    # make_sort_keys output orders UNSIGNED; bias the top bit for signed use
+    kc = binary_op_scalar("bitxor", kc, -(1 << 63), DType.int64())
+    if col.validity is not None:
+        extreme = (1 << 63) - 1 if nulls_last else -(1 << 63)
+        nm = is_null(col)
+        kc = if_else(nm, Column.full(extreme, DType.int64(), n, "cuda"), kc)
+    return Column(DType.int64(), n, kc.data, None, null_count=0)
+
+
 def sort_order(batch: ColumnBatch, key_idx: List[int], descending: List[bool],
                nulls_last: List[bool]) -> Column:
     n = batch.num_rows

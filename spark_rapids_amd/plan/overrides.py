@@ -333,8 +333,11 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
 
         return WindowExec(device, node.window_exprs, kids[0], node.schema())
     if isinstance(node, L.Sort):
+        from ..config import BATCH_SIZE_BYTES
+
         return P.SortExec(device, node.keys, node.descending,
-                          node.nulls_last, kids[0])
+                          node.nulls_last, kids[0],
+                          target_bytes=conf.get(BATCH_SIZE_BYTES))
     if isinstance(node, L.Limit):
         return P.LimitExec(device, node.n, kids[0])
     if isinstance(node, L.Union):

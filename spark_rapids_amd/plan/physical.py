@@ -18,7 +18,7 @@ from .. import ops
 from ..column import Column, ColumnBatch, Field, Schema
 from ..expr.aggregates import AggExpr
 from ..expr.expressions import ColumnRef, Expression
-from ..types import DType, FLOAT64, INT64
+from ..types import BOOL, DType, FLOAT64, INT64
 from ..memory.retry import with_retry_split
 
 
@@ -665,17 +665,23 @@ class CrossJoinExec(PhysicalExec):
 
 class SortExec(PhysicalExec):
     def __init__(self, device: str, keys: List[str], descending: List[bool],
-                 nulls_last: List[bool], child: PhysicalExec):
+                 nulls_last: List[bool], child: PhysicalExec,
+                 target_bytes: int = 2 << 30):
         super().__init__(device, child.schema, [child])
         self.keys = keys
         self.descending = descending
         self.nulls_last = nulls_last
+        self.target_bytes = target_bytes
 
     def execute(self) -> Iterator[ColumnBatch]:
         from ..memory.spill import SpillableBatch
 
         handles = [SpillableBatch(b) for b in self.children[0].execute()]
         if not handles:
+            return
+        total = sum(h.nbytes for h in handles)
+        if total > self.target_bytes:
+            yield from self._external_sort(handles)
             return
         batches = [h.get() for h in handles]
         table = ops.concat_batches(batches) if len(batches) > 1 else batches[0]
@@ -684,9 +690,76 @@ class SortExec(PhysicalExec):
         if table.num_rows == 0:
             yield table
             return
+        yield self._sort_one(table)
+
+    def _sort_one(self, table: ColumnBatch) -> ColumnBatch:
         kidx = [self.schema.index(k) for k in self.keys]
         order = ops.sort_order(table, kidx, self.descending, self.nulls_last)
-        yield ops.gather(table, order)
+        return ops.gather(table, order)
+
+    def _external_sort(self, handles) -> Iterator[ColumnBatch]:
+        """Out-of-core sort (reference analogue: GpuSortExec's full-sort
+        out-of-core path / GpuOutOfCoreSortIterator): range-partition the
+        input on a monotone int64 proxy of the primary sort key, spill the
+        bucket pieces, then sort each bounded bucket in-core and emit the
+        buckets in key order. Equal proxies always share a bucket, so the
+        per-bucket in-core sort (all keys) makes the global order exact."""
+        import numpy as np
+
+        from ..memory.spill import SpillableBatch
+
+        kidx = [self.schema.index(k) for k in self.keys]
+        k0, d0, n0 = kidx[0], self.descending[0], self.nulls_last[0]
+        total = sum(h.nbytes for h in handles)
+        nbuckets = int(min(64, max(2, -(-total // self.target_bytes))))
+
+        # pass 1: per-chunk range keys + a key sample for the boundaries
+        samples = []
+        key_handles = []
+        for h in handles:
+            batch = h.get()
+            kc = ops.backend_for(*batch.columns).range_key(
+                batch.columns[k0], d0, n0)
+            key_handles.append(SpillableBatch(ColumnBatch([kc], kc.size)))
+            arr = kc.data.cpu().numpy()[:kc.size]
+            if len(arr):
+                stride = max(1, len(arr) // 2048)
+                samples.append(arr[::stride].copy())
+        if not samples:
+            return
+        sample = np.concatenate(samples)
+        qs = np.quantile(sample, [i / nbuckets for i in range(1, nbuckets)],
+                         method="nearest").astype(np.int64)
+        bounds = sorted(set(int(q) for q in qs))
+
+        # pass 2: split every chunk into bucket pieces and spill them
+        pieces: List[List] = [[] for _ in range(len(bounds) + 1)]
+        for h, kh in zip(handles, key_handles):
+            batch = h.get()
+            kc = kh.get().columns[0]
+            for j in range(len(bounds) + 1):
+                mask = None
+                if j > 0:
+                    mask = ops.binary_op_scalar("ge", kc, bounds[j - 1],
+                                                BOOL)
+                if j < len(bounds):
+                    m2 = ops.binary_op_scalar("lt", kc, bounds[j], BOOL)
+                    mask = m2 if mask is None else                         ops.binary_op("and", mask, m2, BOOL)
+                piece = batch if mask is None else                     ops.apply_boolean_mask(batch, mask)
+                if piece.num_rows:
+                    pieces[j].append(SpillableBatch(piece))
+            h.close()
+            kh.close()
+
+        # pass 3: in-core sort per bucket, emitted in key order
+        for j in range(len(bounds) + 1):
+            if not pieces[j]:
+                continue
+            parts = [p.get() for p in pieces[j]]
+            bucket = ops.concat_batches(parts) if len(parts) > 1 else parts[0]
+            for p in pieces[j]:
+                p.close()
+            yield self._sort_one(bucket)
 
     def describe(self):
         ks = ", ".join(f"{k}{' DESC' if d else ''}"
