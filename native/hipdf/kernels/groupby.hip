@@ -78,7 +78,8 @@ __global__ void k_gb_rowgid(const int32_t* __restrict__ row_slot,
 }
 
 // ---- aggregation ---------------------------------------------------------
-enum GbOp : int { GB_SUM = 0, GB_MIN, GB_MAX, GB_COUNT, GB_COUNT_ALL };
+enum GbOp : int { GB_SUM = 0, GB_MIN, GB_MAX, GB_COUNT, GB_COUNT_ALL,
+                  GB_FIRST };
 
 template <typename ACC>
 __device__ __forceinline__ void acc_atomic(int op, ACC* addr, ACC v);
@@ -88,6 +89,8 @@ __device__ __forceinline__ void acc_atomic<int64_t>(int op, int64_t* addr,
                                                     int64_t v) {
   if (op == GB_MIN) atomicMin((long long*)addr, (long long)v);
   else if (op == GB_MAX) atomicMax((long long*)addr, (long long)v);
+  else if (op == GB_FIRST) *addr = v;  // any-value semantics (Spark first
+  // without ordering is unspecified); aligned 8B store cannot tear
   else atomicAdd((unsigned long long*)addr, (unsigned long long)v);
 }
 
@@ -96,6 +99,10 @@ __device__ __forceinline__ void acc_atomic<double>(int op, double* addr,
                                                    double v) {
   if (op == GB_SUM) {
     atomicAdd(addr, v);
+    return;
+  }
+  if (op == GB_FIRST) {
+    *addr = v;
     return;
   }
   unsigned long long* up = (unsigned long long*)addr;

@@ -9,7 +9,7 @@ from .api import DataFrame, Session
 from .column import Column, ColumnBatch, Field, Schema
 from .config import RapidsConf, help_doc
 from .expr.aggregates import (avg, collect_list, collect_set,
-                              count, count_distinct, count_star,
+                              count, count_distinct, count_star, first, last,
                               max_, min_, stddev, sum_distinct,
                               sum_, variance)
 from .expr.expressions import (CaseWhen, coalesce, col, date_add, date_sub,

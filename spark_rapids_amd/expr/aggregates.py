@@ -105,6 +105,19 @@ def sum_distinct(e) -> AggExpr:
     return AggExpr("sum", e, distinct=True)
 
 
+def first(e) -> AggExpr:
+    """First non-null value in the group. Like Spark's first() without an
+    ordering: which value is 'first' is unspecified on the GPU (any
+    non-null value of the group); the CPU backend returns the actual first
+    by input order. Reference analogue: GpuFirst (ignoreNulls=true)."""
+    return AggExpr("first", e)
+
+
+def last(e) -> AggExpr:
+    """Last non-null value (GpuLast); same determinism caveats as first."""
+    return AggExpr("last", e)
+
+
 def collect_list(e) -> AggExpr:
     """Gather the group's non-null values into an array (order unspecified,
     like Spark). Reference analogue: GpuCollectList."""

@@ -758,6 +758,25 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
             res = np.bincount(codes[av], minlength=ngroups).astype(np.int64)
             out_cols.append(_make(res, None, out_dtype))
             continue
+        if op in ("first", "last"):
+            idx = np.nonzero(av)[0]
+            if op == "first":
+                idx = idx[::-1]  # reversed assignment: earliest wins
+            if a.dtype == object or vc.dtype.id in (TypeId.STRING,
+                                                    TypeId.DECIMAL128):
+                res = np.zeros(ngroups, dtype=object)
+            else:
+                res = np.zeros(ngroups, dtype=a.dtype)
+            res[codes[idx]] = a[idx]
+            cnt = np.bincount(codes[av], minlength=ngroups)
+            gv = cnt > 0
+            if vc.dtype.id is TypeId.STRING:
+                out_cols.append(Column.from_pylist(
+                    [v if ok else None for v, ok in zip(res, gv)], out_dtype))
+            else:
+                out_cols.append(_make(res, gv if not gv.all() else None,
+                                      out_dtype))
+            continue
         if op in ("collect_list", "collect_set"):
             lists: list = [[] for _ in range(ngroups)]
             for g, v, ok in zip(codes, vc.to_pylist(), av):

@@ -174,10 +174,16 @@ class Tagger:
                         reasons += self.expr_reasons(a.child, cs)
                         continue
                     r = _NUMERIC.supports(t)
-                    if r and a.op not in ("count", "count_all", "min", "max"):
+                    if t.id is TypeId.STRING and a.op not in (
+                            "count", "count_all"):
+                        reasons.append(
+                            f"agg {a.op} over strings not on GPU yet")
+                    elif r and a.op not in ("count", "count_all", "min",
+                                            "max", "first", "last"):
                         reasons.append(f"agg {a.op}({a.child}): {r}")
                     if t.id is TypeId.DECIMAL128 and a.op not in (
-                            "sum", "count", "count_all"):
+                            "sum", "count", "count_all",
+                            "collect_list", "collect_set"):
                         reasons.append(
                             f"agg {a.op} over decimal128 not supported yet")
                     reasons += self.expr_reasons(a.child, cs)

@@ -201,7 +201,7 @@ def _lower_aggs(aggs: List[AggExpr], in_schema: Schema):
             partial.append(("count", len(value_exprs) - 1, INT64))
             merge.append("sum")
             final.append(("col", j))
-        elif a.op in ("sum", "min", "max"):
+        elif a.op in ("sum", "min", "max", "first", "last"):
             value_exprs.append(a.child)
             j = len(partial)
             partial.append((a.op, len(value_exprs) - 1, a.out_dtype(in_schema)))

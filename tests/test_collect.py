@@ -87,3 +87,34 @@ def test_gpu_collect_placement():
     tree = (_df(s, 10).group_by("k").agg(collect_list(col("v")))
             .physical_plan().tree_string())
     assert "GpuHashAggregate" in tree, tree
+
+
+def test_first_last_cpu(cpu):
+    from spark_rapids_amd import first, last
+
+    df = cpu.create_dataframe({
+        "k": [1, 1, 1, 2, 2],
+        "v": [None, 10, 20, None, None],
+        "s": ["a", None, "c", "d", None]})
+    rows = sorted(df.group_by("k").agg(first(col("v")), last(col("v")),
+                                       first(col("s"))).collect())
+    assert rows == [(1, 10, 20, "a"), (2, None, None, "d")]
+
+
+@pytest.mark.gpu
+def test_gpu_first_any_value():
+    """GPU first() returns SOME non-null value of the group."""
+    from spark_rapids_amd import first
+
+    sg = sr.Session()
+    df = _df(sg, 5000)
+    raw = df.collect()
+    groups = {}
+    for k, v, w in raw:
+        if v is not None:
+            groups.setdefault(k, set()).add(v)
+    for k, f in df.group_by("k").agg(first(col("v"))).collect():
+        if k in groups:
+            assert f in groups[k], (k, f)
+        else:
+            assert f is None
