@@ -89,6 +89,8 @@ void hipdf_i128_cmp(int, const void*, const void*, const void*, const void*,
                     void*, void*, int64_t, hipStream_t);
 void hipdf_i64_to_i128(const void*, void*, int64_t, hipStream_t);
 void hipdf_i128_to_f64(const void*, void*, int64_t, hipStream_t);
+void hipdf_gb_percentile(const void*, const void*, const void*, const void*,
+                         double, void*, int, hipStream_t);
 void hipdf_gb_collect_count(const void*, const void*, const void*, void*,
                             int64_t, hipStream_t);
 void hipdf_gb_collect_fill(int, const void*, const void*, const void*,
@@ -500,6 +502,13 @@ PYBIND11_MODULE(hipdf, m) {
   });
   m.def("i128_to_f64", [](int64_t in, int64_t out, int64_t n, int64_t stream) {
     hipdf_i128_to_f64(P(in), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("gb_percentile", [](int64_t vals, int64_t perm, int64_t starts,
+                            int64_t vcnt, double p, int64_t out, int ngroups,
+                            int64_t stream) {
+    hipdf_gb_percentile(P(vals), P(perm), P(starts), P(vcnt), p, PM(out),
+                        ngroups, S(stream));
     check_async();
   });
   m.def("gb_collect_count", [](int64_t vvalid, int64_t row_gid, int64_t sel,

@@ -346,7 +346,41 @@ __global__ void k_gb_collect_fill(int esize, const uint8_t* __restrict__ vals,
   }
 }
 
+// exact percentile with linear interpolation over per-group sorted values.
+// perm: row order sorted by (gid, value, nulls last); starts: first row of
+// each group in that order (scan of total counts); vcnt: valid counts.
+__global__ void k_gb_percentile(const double* __restrict__ vals,
+                                const int32_t* __restrict__ perm,
+                                const int64_t* __restrict__ starts,
+                                const int64_t* __restrict__ vcnt, double p,
+                                double* __restrict__ out, int32_t ngroups) {
+  for (int g = blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += gridDim.x * blockDim.x) {
+    int64_t c = vcnt[g];
+    if (!c) {
+      out[g] = 0.0;
+      continue;
+    }
+    double pos = p * (double)(c - 1);
+    int64_t lo = (int64_t)pos;
+    int64_t hi = lo + 1 < c ? lo + 1 : c - 1;
+    double frac = pos - (double)lo;
+    double a = vals[perm[starts[g] + lo]];
+    double b = vals[perm[starts[g] + hi]];
+    out[g] = a + (b - a) * frac;
+  }
+}
+
 extern "C" {
+
+void hipdf_gb_percentile(const void* vals, const void* perm,
+                         const void* starts, const void* vcnt, double p,
+                         void* out, int ngroups, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_percentile, flat_grid(ngroups), dim3(HIPDF_BLOCK),
+                     0, stream, (const double*)vals, (const int32_t*)perm,
+                     (const int64_t*)starts, (const int64_t*)vcnt, p,
+                     (double*)out, ngroups);
+}
 
 void hipdf_gb_collect_count(const void* vvalid, const void* row_gid,
                             const void* sel, void* counts, int64_t n,

@@ -8,7 +8,8 @@ spark.rapids.* config registry.
 from .api import DataFrame, Session
 from .column import Column, ColumnBatch, Field, Schema
 from .config import RapidsConf, help_doc
-from .expr.aggregates import (avg, collect_list, collect_set,
+from .expr.aggregates import (approx_percentile, avg, collect_list,
+                              collect_set, percentile,
                               count, count_distinct, count_star, first, last,
                               max_, min_, stddev, sum_distinct,
                               sum_, variance)

@@ -31,6 +31,8 @@ class AggExpr:
     def output_name(self) -> str:
         if self._name:
             return self._name
+        if self.op.startswith("percentile:"):
+            return f"percentile({self.child}, {self.op.split(':', 1)[1]})"
         disp = self._DISPLAY.get(self.op, self.op)
         if self.child is None:
             return f"{disp}(*)"
@@ -54,6 +56,8 @@ class AggExpr:
             return ct
         if self.op in ("collect_list", "collect_set"):
             return DType.list_(ct)
+        if self.op.startswith("percentile:"):
+            return FLOAT64
         raise NotImplementedError(f"agg {self.op}")
 
     def __str__(self):
@@ -116,6 +120,19 @@ def first(e) -> AggExpr:
 def last(e) -> AggExpr:
     """Last non-null value (GpuLast); same determinism caveats as first."""
     return AggExpr("last", e)
+
+
+def percentile(e, p: float) -> AggExpr:
+    """Exact percentile with linear interpolation (Spark percentile(col, p);
+    GpuPercentile analogue). p in [0, 1]."""
+    assert 0.0 <= p <= 1.0
+    return AggExpr(f"percentile:{p}", e)
+
+
+def approx_percentile(e, p: float) -> AggExpr:
+    """Served by the exact implementation (always at least as accurate as
+    the reference's t-digest approx_percentile)."""
+    return percentile(e, p)
 
 
 def collect_list(e) -> AggExpr:

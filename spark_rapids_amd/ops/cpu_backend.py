@@ -777,6 +777,20 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                 out_cols.append(_make(res, gv if not gv.all() else None,
                                       out_dtype))
             continue
+        if op.startswith("percentile:"):
+            pq = float(op.split(":", 1)[1])
+            res = np.zeros(ngroups)
+            gv = np.zeros(ngroups, dtype=bool)
+            af = a[av].astype(np.float64)
+            gc2 = codes[av]
+            for g in range(ngroups):
+                vals_g = af[gc2 == g]
+                if len(vals_g):
+                    res[g] = np.percentile(vals_g, 100.0 * pq)
+                    gv[g] = True
+            out_cols.append(_make(res, gv if not gv.all() else None,
+                                  out_dtype))
+            continue
         if op in ("collect_list", "collect_set"):
             lists: list = [[] for _ in range(ngroups)]
             for g, v, ok in zip(codes, vc.to_pylist(), av):

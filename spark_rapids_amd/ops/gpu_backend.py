@@ -841,6 +841,11 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                               op == "collect_set", s)
             allocs.append(("collect", out_dtype, False, col, None))
             continue
+        if op.startswith("percentile:"):
+            col = _gb_percentile(vc, row_gid, selp, n, ngroups,
+                                 float(op.split(":", 1)[1]), s)
+            allocs.append(("collect", out_dtype, False, col, None))
+            continue
         acc_is_double = out_dtype.is_floating or (
             vc is not None and vc.dtype.is_floating)
         acc = torch.empty(max(ngroups, 1),
@@ -888,6 +893,38 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         out_cols.append(Column(out_dtype, ngroups, out_data, ov,
                                null_count=None))
     return ColumnBatch(out_cols, ngroups)
+
+
+def _gb_percentile(vc: Column, row_gid: torch.Tensor, selp, n: int,
+                   ngroups: int, p: float, s) -> Column:
+    """Exact percentile (linear interpolation): sort (gid, value) with
+    nulls last, then interpolate inside each group's valid window
+    (k_gb_percentile). Reference analogue: GpuPercentile."""
+    if selp:
+        raise NotImplementedError("percentile under filter fusion")
+    vf = cast(vc, DType.float64()) if vc.dtype.id is not TypeId.FLOAT64 \
+        else vc
+    gid_col = Column(DType.int32(), n, row_gid[:n], None, null_count=0)
+    pair = ColumnBatch([gid_col, vf], n)
+    perm = sort_order(pair, [0, 1], [False, False], [True, True])
+    total = torch.zeros(max(ngroups, 1), dtype=torch.int64, device="cuda")
+    vcnt = torch.zeros(max(ngroups, 1), dtype=torch.int64, device="cuda")
+    if n:
+        ext.gb_collect_count(0, row_gid.data_ptr(), 0, total.data_ptr(),
+                             n, s)
+        ext.gb_collect_count(_ptr(vf.validity), row_gid.data_ptr(), 0,
+                             vcnt.data_ptr(), n, s)
+    starts, _ = _exclusive_scan_i64(total[:ngroups]) if ngroups \
+        else (total, 0)
+    out = torch.empty(max(ngroups, 1), dtype=torch.float64,
+                      device="cuda")[:ngroups]
+    if ngroups:
+        ext.gb_percentile(vf.data.data_ptr(), perm.data.data_ptr(),
+                          starts.data_ptr(), vcnt.data_ptr(), p,
+                          out.data_ptr(), ngroups, s)
+    ov = _alloc_mask(ngroups)
+    ext.mask_from_nonzero(vcnt.data_ptr(), ov.data_ptr(), ngroups, s)
+    return Column(DType.float64(), ngroups, out, ov, null_count=None)
 
 
 def _gb_collect(vc: Column, row_gid: torch.Tensor, selp, n: int,

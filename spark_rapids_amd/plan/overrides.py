@@ -167,6 +167,13 @@ class Tagger:
             for a in node.aggs:
                 if a.child is not None:
                     t = a.child.dtype(cs)
+                    if a.op.startswith("percentile:"):
+                        if not (t.is_integral or t.is_floating
+                                or t.is_decimal):
+                            reasons.append(
+                                f"percentile over {t} not supported")
+                        reasons += self.expr_reasons(a.child, cs)
+                        continue
                     if a.op in ("collect_list", "collect_set"):
                         if t.id is TypeId.STRING or t.is_nested:
                             reasons.append(

@@ -435,6 +435,8 @@ class Parser:
                 if not self.op(","):
                     break
             self.expect_op(")")
+        if name in ("percentile", "approx_percentile"):
+            return A.percentile(args[0], float(args[1].value))
         if name in _AGG_FUNCS:
             ctor = {"sum": A.sum_, "avg": A.avg, "count": A.count,
                     "min": A.min_, "max": A.max_, "stddev": A.stddev,

@@ -118,3 +118,48 @@ def test_gpu_first_any_value():
             assert f in groups[k], (k, f)
         else:
             assert f is None
+
+
+def test_percentile_cpu(cpu):
+    from spark_rapids_amd import percentile
+    import numpy as np
+
+    rng = np.random.default_rng(4)
+    k = [int(v) for v in rng.integers(0, 5, 3000)]
+    v = [float(x) if i % 11 else None
+         for i, x in enumerate(rng.uniform(-50, 50, 3000))]
+    df = cpu.create_dataframe({"k": k, "v": v})
+    rows = df.group_by("k").agg(percentile(col("v"), 0.25),
+                                percentile(col("v"), 0.75)).collect()
+    for g, p25, p75 in rows:
+        vals = [x for kk, x in zip(k, v) if kk == g and x is not None]
+        assert p25 == pytest.approx(np.percentile(vals, 25))
+        assert p75 == pytest.approx(np.percentile(vals, 75))
+
+
+def test_sql_percentile(cpu):
+    df = cpu.create_dataframe({"k": [1, 1, 1, 1], "v": [1.0, 2.0, 3.0, 4.0]})
+    cpu.register("tperc", df)
+    out = cpu.sql("SELECT k, percentile(v, 0.5) FROM tperc GROUP BY k")
+    assert out.collect() == [(1, 2.5)]
+
+
+@pytest.mark.gpu
+def test_gpu_percentile_matches_cpu():
+    from spark_rapids_amd import percentile
+    import numpy as np
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+
+    def q(s):
+        df = _df(s, 30000)
+        return sorted(df.group_by("k").agg(
+            percentile(col("v"), 0.5), percentile(col("w"), 0.99),
+            count_star()).collect())
+
+    g, c = q(sg), q(sc)
+    for rg, rc in zip(g, c):
+        assert rg[0] == rc[0] and rg[3] == rc[3]
+        assert rg[1] == pytest.approx(rc[1], rel=1e-12)
+        assert rg[2] == pytest.approx(rc[2], rel=1e-12)
