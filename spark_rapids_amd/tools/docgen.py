@@ -22,8 +22,10 @@ _EXECS = [
     ("Join", "hash equi-join inner/left/semi/anti/full, broadcast or "
      "shuffled build side", "fixed-width + string + decimal128 keys on GPU"),
     ("CrossJoin", "cartesian gather maps (+ filter for non-equi)", "all"),
-    ("Sort", "stable LSD radix sort", "fixed-width keys on GPU; "
-     "string/decimal128 sort keys fall back"),
+    ("Sort", "stable LSD radix sort; out-of-core range-partitioned spill "
+     "buckets", "fixed-width keys on GPU; string/decimal128 sort keys "
+     "fall back"),
+    ("Expand", "grouping-sets projections (rollup / cube)", "all"),
     ("Window", "ranking / running + bounded + partition aggregates / "
      "lag / lead", "GPU segmented scans; running/bounded min-max on CPU"),
     ("Limit", "row limit", "all"),
@@ -55,9 +57,11 @@ def supported_ops_doc() -> str:
                  "`substring` (strings.hip) and `rlike` (bytecode regex "
                  "VM, regex.hip, CPU fallback outside the subset)")
     lines += ["", "## Aggregate functions", "",
-              "`sum`, `count`, `count(*)`, `min`, `max`, `avg` "
-              "plus `stddev`/`variance` (partial/merge lowering; mean as sum+count, "
-              "variance as sum+sumsq+count)", "",
+              "`sum`, `count`, `count(*)`, `min`, `max`, `avg`, `stddev`, "
+              "`variance`, `first`, `last`, `count/sum(DISTINCT)`, "
+              "`collect_list`, `collect_set`, `percentile`/"
+              "`approx_percentile` (partial/merge lowering; collect and "
+              "percentile run the single-pass path)", "",
               "## Window functions", "",
               "`row_number`, `rank`, `dense_rank`, `sum`, `count`, `min`, "
               "`max`, `avg` (running, bounded ROWS BETWEEN, whole "
