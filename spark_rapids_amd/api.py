@@ -330,7 +330,33 @@ class Session:
         return DataFrame(self, L.Scan(table, schema, "memory"))
 
     def from_batches(self, batches: List[ColumnBatch], schema: Schema,
-                     label: str = "memory", replicated: bool = False) -> DataFrame:
+                     label: str = "memory", replicated: bool = False,
+                     coalesce: bool = True) -> DataFrame:
+        """In-memory table. By default the stored layout is coalesced ONCE
+        toward spark.rapids.sql.batchSizeBytes at load time, so every query
+        over the table streams large contiguous batches instead of paying a
+        re-concatenation per scan (288 GB of HBM3E favors few, large
+        resident batches). Pass coalesce=False to keep the caller's exact
+        batch boundaries (multi-batch code paths in tests)."""
+        if coalesce and len(batches) > 1:
+            from .config import BATCH_SIZE_BYTES
+            from . import ops as _ops
+
+            target = self.conf.get(BATCH_SIZE_BYTES)
+            merged: List[ColumnBatch] = []
+            pending: List[ColumnBatch] = []
+            nbytes = 0
+            for b in batches:
+                if pending and nbytes + b.nbytes > target:
+                    merged.append(pending[0] if len(pending) == 1
+                                  else _ops.concat_batches(pending))
+                    pending, nbytes = [], 0
+                pending.append(b)
+                nbytes += b.nbytes
+            if pending:
+                merged.append(pending[0] if len(pending) == 1
+                              else _ops.concat_batches(pending))
+            batches = merged
         return DataFrame(self, L.Scan(MemTable(batches, schema, replicated),
                                       schema, label))
 

@@ -173,6 +173,11 @@ PRUNE_COLUMNS = bool_conf(
     "spark.rapids.sql.optimizer.pruneColumns.enabled", True,
     "Push projections below joins/aggregates so unused columns are never "
     "gathered or transferred (Catalyst-optimizer analogue).")
+PUSH_FILTERS = bool_conf(
+    "spark.rapids.sql.optimizer.pushFilters.enabled", True,
+    "Push filter conjuncts below joins when they reference only one side "
+    "(PushPredicateThroughJoin analogue): dimension-table predicates then "
+    "filter 50 rows instead of the joined fact output.")
 LORE_DUMP_PATH = str_conf(
     "spark.rapids.sql.lore.dumpPath", "",
     "When set, dump every operator's output batches to this directory as "

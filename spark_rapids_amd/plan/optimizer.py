@@ -42,6 +42,102 @@ def _project_to(plan: L.LogicalPlan, needed: Optional[Set[str]]):
     return L.Project([ColumnRef(n) for n in keep], plan)
 
 
+def _split_conjuncts(e: Expression):
+    """Split a predicate on AND. Filtering by `a AND b` equals filtering by
+    `a` then `b` under Kleene logic (the row passes only when both are
+    TRUE), so conjuncts can be pushed independently."""
+    from ..expr.expressions import BinaryExpr
+
+    if isinstance(e, BinaryExpr) and e.op == "and":
+        return _split_conjuncts(e.left) + _split_conjuncts(e.right)
+    return [e]
+
+
+def _and_all(conjs):
+    from ..expr.expressions import BinaryExpr
+
+    out = conjs[0]
+    for c in conjs[1:]:
+        out = BinaryExpr("and", out, c)
+    return out
+
+
+def push_filters(plan: L.LogicalPlan) -> L.LogicalPlan:
+    """Predicate pushdown through joins (Catalyst PushPredicateThroughJoin
+    analogue — the reference inherits this from Spark; here the engine owns
+    the plan). A conjunct whose columns all come from one join side filters
+    that side BEFORE the join: on dimension-filter queries this shrinks the
+    build side and every downstream gather.
+
+    Safety: inner/cross joins push to either side; left/semi/anti joins
+    push only left-side conjuncts (right-side predicates would change the
+    null-extension semantics)."""
+    # rewrite children first
+    kids = [push_filters(c) for c in plan.children]
+    plan = _with_children(plan, kids)
+    if not isinstance(plan, L.Filter):
+        return plan
+    child = plan.child
+    if isinstance(child, (L.Join, L.CrossJoin)):
+        ls = set(child.left.schema().names)
+        rs = set(child.right.schema().names)
+        how = getattr(child, "how", "inner")
+        left_ok = how in ("inner", "left", "semi", "anti", "cross") or \
+            isinstance(child, L.CrossJoin)
+        right_ok = how == "inner" or isinstance(child, L.CrossJoin)
+        push_l, push_r, keep = [], [], []
+        for c in _split_conjuncts(plan.condition):
+            refs: Set[str] = set()
+            _refs(c, refs)
+            if refs and refs <= ls and left_ok:
+                push_l.append(c)
+            elif refs and refs <= rs and right_ok:
+                push_r.append(c)
+            else:
+                keep.append(c)
+        if not push_l and not push_r:
+            return plan
+        left = L.Filter(_and_all(push_l), child.left) if push_l \
+            else child.left
+        right = L.Filter(_and_all(push_r), child.right) if push_r \
+            else child.right
+        if isinstance(child, L.CrossJoin):
+            new_join = L.CrossJoin(left, right)
+        else:
+            new_join = L.Join(left, right, child.left_on, child.right_on,
+                              child.how)
+        # re-run on the pushed filters (stacked joins push further down)
+        new_join = _with_children(
+            new_join, [push_filters(c) for c in new_join.children])
+        if keep:
+            return L.Filter(_and_all(keep), new_join)
+        return new_join
+    if isinstance(child, L.Filter):
+        # merge adjacent filters so conjuncts push as one set
+        merged = L.Filter(_and_all([plan.condition, child.condition]),
+                          child.child)
+        out = push_filters(merged)
+        return out
+    return plan
+
+
+def _with_children(plan: L.LogicalPlan, kids):
+    """Shallow-rebuild a node with new children (nodes are plain objects;
+    mutate the child slots in place on a copy)."""
+    import copy
+
+    if list(plan.children) == list(kids):
+        return plan
+    p = copy.copy(plan)
+    if isinstance(plan, (L.Join, L.CrossJoin)):
+        p.left, p.right = kids
+    elif isinstance(plan, L.Union):
+        p.plans = list(kids)
+    elif hasattr(plan, "child"):
+        p.child = kids[0]
+    return p
+
+
 def prune_columns(plan: L.LogicalPlan,
                   needed: Optional[Set[str]] = None) -> L.LogicalPlan:
     """Return an equivalent plan where children materialize only the columns

@@ -256,7 +256,11 @@ def plan_physical(node: L.LogicalPlan, conf: RapidsConf,
     top = tagger is None
     if tagger is None:
         tagger = Tagger(conf)
-        from ..config import PRUNE_COLUMNS
+        from ..config import PRUNE_COLUMNS, PUSH_FILTERS
+        if conf.get(PUSH_FILTERS):
+            from .optimizer import push_filters
+
+            node = push_filters(node)
         if conf.get(PRUNE_COLUMNS):
             from .optimizer import prune_columns
 
