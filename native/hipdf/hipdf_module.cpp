@@ -91,6 +91,8 @@ void hipdf_i64_to_i128(const void*, void*, int64_t, hipStream_t);
 void hipdf_i128_to_f64(const void*, void*, int64_t, hipStream_t);
 void hipdf_gb_percentile(const void*, const void*, const void*, const void*,
                          double, void*, int, hipStream_t);
+void hipdf_dense_gid(const void*, int, const void*, void*, int64_t,
+                     hipStream_t);
 void hipdf_gb_collect_count(const void*, const void*, const void*, void*,
                             int64_t, hipStream_t);
 void hipdf_gb_collect_fill(int, const void*, const void*, const void*,
@@ -509,6 +511,11 @@ PYBIND11_MODULE(hipdf, m) {
                             int64_t stream) {
     hipdf_gb_percentile(P(vals), P(perm), P(starts), P(vcnt), p, PM(out),
                         ngroups, S(stream));
+    check_async();
+  });
+  m.def("dense_gid", [](int64_t keys, int nkeys, int64_t sel,
+                        int64_t row_gid, int64_t n, int64_t stream) {
+    hipdf_dense_gid(P(keys), nkeys, P(sel), PM(row_gid), n, S(stream));
     check_async();
   });
   m.def("gb_collect_count", [](int64_t vvalid, int64_t row_gid, int64_t sel,

@@ -55,6 +55,51 @@ __global__ void k_gb_build(const int32_t* __restrict__ hashes,
   }
 }
 
+// ---- dense integer-key fast path -----------------------------------------
+// When the (possibly multi-) integer group keys span a small value range,
+// the group id is a direct radix index — no hash table, no probe, no
+// leader gather. gid = sum_i code_i * stride_i with code_i = key_i - min_i
+// (range_i for NULL, so nulls group together). Empty ids are compacted by
+// the host afterwards.
+struct DenseKey {
+  int type;        // HType of the key column
+  int pad;
+  const void* vals;
+  const uint64_t* valid;
+  int64_t kmin;
+  int64_t range;   // #distinct slots for values; NULL takes index `range`
+  int64_t stride;
+};
+
+__device__ __forceinline__ int64_t dense_load_i64(const void* p, int t,
+                                                  int64_t i) {
+  switch (t) {
+    case HT_U8: return ((const uint8_t*)p)[i];
+    case HT_I8: return ((const int8_t*)p)[i];
+    case HT_I16: return ((const int16_t*)p)[i];
+    case HT_I32: return ((const int32_t*)p)[i];
+    default: return ((const int64_t*)p)[i];
+  }
+}
+
+__global__ void k_dense_gid(const DenseKey* __restrict__ keys, int nkeys,
+                            const int32_t* __restrict__ sel,
+                            int32_t* __restrict__ row_gid, int64_t n) {
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
+    int64_t gid = 0;
+    for (int k = 0; k < nkeys; ++k) {
+      const DenseKey& d = keys[k];
+      int64_t code = valid_bit(d.valid, i)
+                         ? dense_load_i64(d.vals, d.type, i) - d.kmin
+                         : d.range;
+      gid += code * d.stride;
+    }
+    row_gid[j] = (int32_t)gid;
+  }
+}
+
 __global__ void k_gb_number(const int32_t* __restrict__ claimed_slots,
                             const int32_t* __restrict__ slot_row,
                             int32_t* __restrict__ slot_gid,
@@ -380,6 +425,13 @@ void hipdf_gb_percentile(const void* vals, const void* perm,
                      0, stream, (const double*)vals, (const int32_t*)perm,
                      (const int64_t*)starts, (const int64_t*)vcnt, p,
                      (double*)out, ngroups);
+}
+
+void hipdf_dense_gid(const void* keys, int nkeys, const void* sel,
+                     void* row_gid, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_dense_gid, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (const DenseKey*)keys, nkeys, (const int32_t*)sel,
+                     (int32_t*)row_gid, n);
 }
 
 void hipdf_gb_collect_count(const void* vvalid, const void* row_gid,

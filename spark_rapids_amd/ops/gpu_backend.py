@@ -773,11 +773,18 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         out += [_empty_col(dt) for _, _, dt in aggs]
         return ColumnBatch(out, 0)
     selp = 0 if sel is None else sel.data_ptr()
+    dense = None
     if not key_idx:
         # global aggregate: single group
         row_gid = torch.zeros(n, dtype=torch.int32, device="cuda")
         ngroups = 1
         leaders = torch.zeros(1, dtype=torch.int32, device="cuda")
+    elif (dense := _try_dense_keys(keys, aggs, selp, n)) is not None:
+        # dense integer-key fast path: gid = radix index over the small
+        # value ranges — no hash table, no probe, no leader gather
+        # (k_dense_gid); empty ids are compacted after aggregation
+        row_gid, ngroups, dense_info = dense
+        leaders = None
     else:
         h = _murmur3_tensor(keys, 42, sel, n)
         cap = max(1024, _next_pow2(2 * n))
@@ -798,11 +805,14 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         row_gid = torch.empty(n, dtype=torch.int32, device="cuda")
         ext.gb_rowgid(row_slot.data_ptr(), slot_gid.data_ptr(),
                       row_gid.data_ptr(), n, s)
-    leaders = leaders[:ngroups]
+    if leaders is not None:
+        leaders = leaders[:ngroups]
     # leaders are ORIGINAL row ids (pre-selection), so gather keys from the
-    # full-length key columns
+    # full-length key columns (hash path; the dense path reconstructs keys
+    # arithmetically after compaction)
     key_batch = _gather_by_idx(ColumnBatch(keys, keys[0].size), leaders,
-                               ngroups, maybe_negative=False) if keys else None
+                               ngroups, maybe_negative=False) \
+        if keys and leaders is not None else None
     out_cols = list(key_batch.columns) if key_batch is not None else []
 
     # fused multi-aggregate: one kernel pass accumulates every agg;
@@ -892,7 +902,95 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         ext.mask_from_nonzero(cnt.data_ptr(), ov.data_ptr(), ngroups, s)
         out_cols.append(Column(out_dtype, ngroups, out_data, ov,
                                null_count=None))
+    if dense is not None:
+        return _compact_dense(out_cols, allocs, keys, dense_info, ngroups,
+                              row_gid, n, s)
     return ColumnBatch(out_cols, ngroups)
+
+
+_DENSE_OK = {TypeId.BOOL, TypeId.INT8, TypeId.INT16, TypeId.INT32,
+             TypeId.INT64, TypeId.DATE32}
+
+
+def _try_dense_keys(keys, aggs, selp, n):
+    """Dense path applicability: all keys are non-null small-range integers
+    and the range product stays tiny relative to n."""
+    if any(k.validity is not None for k in keys):
+        return None
+    if any(k.dtype.id not in _DENSE_OK for k in keys):
+        return None
+    if any(op in ("collect_list", "collect_set") for op, _, _ in aggs):
+        return None  # LIST columns cannot be re-gathered in compaction
+    mins, ranges = [], []
+    for k in keys:
+        kmin, kmax = reduce("min", k), reduce("max", k)
+        if kmin is None:
+            return None
+        rng = int(kmax) - int(kmin) + 1
+        if rng <= 0:
+            return None
+        mins.append(int(kmin))
+        ranges.append(rng)
+    prod = 1
+    for r in ranges:
+        prod *= r
+        if prod > (1 << 24):
+            return None
+    if prod > max(1 << 16, 4 * n):
+        return None
+    strides = [0] * len(keys)
+    acc = 1
+    for i in range(len(keys) - 1, -1, -1):
+        strides[i] = acc
+        acc *= ranges[i]
+    s = _stream()
+    blob = b"".join(
+        struct.pack("<iiqqqqq", _ht(k.dtype), 0, k.data.data_ptr(), 0,
+                    mins[i], ranges[i], strides[i])
+        for i, k in enumerate(keys))
+    desc = torch.frombuffer(bytearray(blob), dtype=torch.uint8).cuda()
+    row_gid = torch.empty(n, dtype=torch.int32, device="cuda")
+    ext.dense_gid(desc.data_ptr(), len(keys), selp, row_gid.data_ptr(), n, s)
+    return row_gid, prod, (mins, ranges, strides)
+
+
+def _compact_dense(out_cols, allocs, keys, dense_info, ngroups, row_gid,
+                   n, s):
+    """Drop empty dense group ids and reconstruct the key columns from the
+    surviving ids (kmin + (g // stride) % range per key)."""
+    mins, ranges, strides = dense_info
+    counts = None
+    for (op, _, _, acc, cnt) in allocs:
+        if op == "count_all":
+            counts = cnt
+            break
+    if counts is None:
+        counts = torch.zeros(max(ngroups, 1), dtype=torch.int64,
+                             device="cuda")
+        if n:
+            ext.gb_collect_count(0, row_gid.data_ptr(), 0,
+                                 counts.data_ptr(), n, s)
+    ccol = Column(DType.int64(), ngroups, counts[:ngroups], None,
+                  null_count=0)
+    live = binary_op_scalar("gt", ccol, 0, DType.bool_())
+    gsel = mask_to_sel(live, ngroups)
+    ncomp = gsel.numel()
+    gcol = Column(DType.int32(), ncomp, gsel, None, null_count=0)
+    g64 = cast(gcol, DType.int64())
+    key_cols = []
+    for i, k in enumerate(keys):
+        code = g64
+        if strides[i] != 1:
+            code = binary_op_scalar("int_div", code, strides[i],
+                                    DType.int64())
+        if i > 0:  # leading key needs no modulo (gid < stride_{i-1}*range)
+            code = binary_op_scalar("mod", code, ranges[i], DType.int64())
+        if mins[i]:
+            code = binary_op_scalar("add", code, mins[i], DType.int64())
+        key_cols.append(cast(code, k.dtype))
+    agg_batch = _gather_by_idx(ColumnBatch(out_cols, ngroups), gsel, ncomp,
+                               maybe_negative=False)
+    return ColumnBatch(key_cols + list(agg_batch.columns), ncomp)
 
 
 def _gb_percentile(vc: Column, row_gid: torch.Tensor, selp, n: int,
