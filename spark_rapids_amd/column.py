@@ -100,7 +100,7 @@ class Column:
     """One column of data; immutable by convention."""
 
     __slots__ = ("dtype", "size", "data", "validity", "offsets",
-                 "_null_count", "child")
+                 "_null_count", "child", "_minmax")
 
     def __init__(
         self,
@@ -119,6 +119,7 @@ class Column:
         self.offsets = offsets
         self._null_count = null_count
         self.child = child  # LIST element column
+        self._minmax = None  # cached (min, max) for dense-key group-by
         if dtype.id is TypeId.STRING:
             assert offsets is not None and offsets.numel() == size + 1
         if dtype.id is TypeId.LIST:

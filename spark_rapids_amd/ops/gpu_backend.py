@@ -923,7 +923,11 @@ def _try_dense_keys(keys, aggs, selp, n):
         return None  # LIST columns cannot be re-gathered in compaction
     mins, ranges = [], []
     for k in keys:
-        kmin, kmax = reduce("min", k), reduce("max", k)
+        if k._minmax is None:
+            # cached on the column: stored tables are re-scanned every
+            # query, and columns are immutable by convention
+            k._minmax = (reduce("min", k), reduce("max", k))
+        kmin, kmax = k._minmax
         if kmin is None:
             return None
         rng = int(kmax) - int(kmin) + 1
