@@ -118,6 +118,13 @@ void hipdf_scatter_fixed(int, const void*, const void*, void*, int64_t,
 void hipdf_levels_to_mask(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_str_cmp(int, const void*, const void*, const void*, const void*,
                    void*, int64_t, hipStream_t);
+void hipdf_regex_extract(const void*, int, const void*, const void*,
+                         const void*, int, void*, void*, void*, int64_t,
+                         hipStream_t);
+void hipdf_regex_replace(const void*, int, const void*, const void*,
+                         const void*, const void*, int, const void*,
+                         const void*, void*, void*, int, void*, int64_t,
+                         hipStream_t);
 void hipdf_regex_match(const void*, int, const void*, const void*,
                        const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_str_cmp_scalar(int, const void*, const void*, const void*, int,
@@ -426,6 +433,25 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
 
+  m.def("regex_extract", [](int64_t prog, int nops, int64_t classes,
+                            int64_t offsets, int64_t bytes, int group,
+                            int64_t out_start, int64_t out_len,
+                            int64_t overflow, int64_t n, int64_t stream) {
+    hipdf_regex_extract(P(prog), nops, P(classes), P(offsets), P(bytes),
+                        group, PM(out_start), PM(out_len), PM(overflow), n,
+                        S(stream));
+    check_async();
+  });
+  m.def("regex_replace", [](int64_t prog, int nops, int64_t classes,
+                            int64_t offsets, int64_t bytes, int64_t repl_ops,
+                            int nrepl, int64_t lit, int64_t out_off,
+                            int64_t out_len, int64_t out_bytes, int mode,
+                            int64_t overflow, int64_t n, int64_t stream) {
+    hipdf_regex_replace(P(prog), nops, P(classes), P(offsets), P(bytes),
+                        P(repl_ops), nrepl, P(lit), P(out_off), PM(out_len),
+                        PM(out_bytes), mode, PM(overflow), n, S(stream));
+    check_async();
+  });
   m.def("regex_match", [](int64_t prog, int nops, int64_t classes,
                           int64_t offsets, int64_t bytes, int64_t out,
                           int64_t overflow, int64_t n, int64_t stream) {

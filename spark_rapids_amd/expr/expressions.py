@@ -124,6 +124,12 @@ class Expression:
         """Java-regex find() semantics (RLike)."""
         return StringPredicate("rlike", self, pattern)
 
+    def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
+        return RegexpExtract(self, pattern, group)
+
+    def regexp_replace(self, pattern: str, replacement: str) -> "RegexpReplace":
+        return RegexpReplace(self, pattern, replacement)
+
     def substr(self, pos: int, length: int = -1) -> "Substring":
         return Substring(self, pos, length)
 
@@ -525,6 +531,58 @@ class Substring(Expression):
 # ---------------------------------------------------------------------------
 # public DSL
 # ---------------------------------------------------------------------------
+
+class RegexpExtract(Expression):
+    """regexp_extract(str, pattern, idx): the capture group's text for the
+    first match; "" when no match or non-participating group (Spark
+    semantics). GPU: capture-group backtracking VM (regex.hip k_regex_extract);
+    reference analogue: GpuRegExpExtract over the transpiled cudf regex."""
+
+    def __init__(self, child: Expression, pattern: str, group: int = 1):
+        self.child = child
+        self.pattern = pattern
+        self.group = group
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return STRING
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.regexp_extract(self.child.eval(batch, schema),
+                                  self.pattern, self.group)
+
+    def __str__(self):
+        return f"regexp_extract({self.child}, {self.pattern!r}, {self.group})"
+
+
+class RegexpReplace(Expression):
+    """regexp_replace(str, pattern, replacement) with $g group references
+    (GpuRegExpReplace analogue; java Matcher.appendReplacement semantics
+    incl. empty-match advancement)."""
+
+    def __init__(self, child: Expression, pattern: str, replacement: str):
+        self.child = child
+        self.pattern = pattern
+        self.replacement = replacement
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return STRING
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.regexp_replace(self.child.eval(batch, schema),
+                                  self.pattern, self.replacement)
+
+    def __str__(self):
+        return (f"regexp_replace({self.child}, {self.pattern!r}, "
+                f"{self.replacement!r})")
+
 
 class Coalesce(Expression):
     def __init__(self, *exprs):

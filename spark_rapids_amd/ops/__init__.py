@@ -46,6 +46,14 @@ def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
     return backend_for(lhs).binary_op_scalar(op, lhs, scalar, out_dtype)
 
 
+def regexp_extract(col: Column, pattern: str, group: int) -> Column:
+    return backend_for(col).regexp_extract(col, pattern, group)
+
+
+def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
+    return backend_for(col).regexp_replace(col, pattern, replacement)
+
+
 def decimal_mul_div(op: str, lhs: Column, rhs: Column,
                     out_dtype: DType) -> Column:
     """Exact decimal multiply/divide at the Spark result scale (operands

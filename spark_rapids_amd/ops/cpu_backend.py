@@ -484,6 +484,48 @@ def str_predicate(op: str, col: Column, pattern: str) -> Column:
     return _make(res, av if not av.all() else None, DType.bool_())
 
 
+def regexp_extract(col: Column, pattern: str, group: int) -> Column:
+    import re as _re
+
+    rx = _re.compile(pattern)
+    out = []
+    for v in col.to_pylist():
+        if v is None:
+            out.append(None)
+            continue
+        m = rx.search(v)
+        if not m or group > rx.groups:
+            out.append("")
+        else:
+            out.append(m.group(group) or "")
+    return Column.from_pylist(out, DType.string())
+
+
+def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
+    import re as _re
+
+    rx = _re.compile(pattern)
+    # java $g refs / \$ escapes -> python \g<g>
+    pyrepl = ""
+    i = 0
+    while i < len(replacement):
+        c = replacement[i]
+        if c == "$" and i + 1 < len(replacement) and \
+                replacement[i + 1].isdigit():
+            pyrepl += f"\\g<{replacement[i + 1]}>"
+            i += 2
+        elif c == "\\" and i + 1 < len(replacement):
+            ch = replacement[i + 1]
+            pyrepl += _re.escape(ch) if ch in "\\$" else "\\" + ch
+            i += 2
+        else:
+            pyrepl += c.replace("\\", "\\\\")
+            i += 1
+    out = [None if v is None else rx.sub(pyrepl, v)
+           for v in col.to_pylist()]
+    return Column.from_pylist(out, DType.string())
+
+
 def substring(col: Column, pos: int, length: int = -1) -> Column:
     a, av = _vals(col), _valid(col)
     out = []

@@ -292,6 +292,136 @@ def str_predicate(op: str, col: Column, pattern: str) -> Column:
     return Column(DType.bool_(), n, out, v, null_count=col._null_count)
 
 
+def _rx_prog_tensors(pattern: str):
+    from .regex_compiler import compile_regex
+
+    prog = compile_regex(pattern)
+    flat = []
+    for o, a0, a1 in prog.ops:
+        flat.extend([o, a0, a1])
+    prog_t = torch.tensor(flat, dtype=torch.int32).cuda()
+    cls_blob = b"".join(prog.classes) or b"\x00" * 32
+    cls_t = torch.frombuffer(bytearray(cls_blob), dtype=torch.uint8).cuda()
+    return prog, prog_t, cls_t
+
+
+def regexp_extract(col: Column, pattern: str, group: int) -> Column:
+    n = col.size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.string())
+    prog, prog_t, cls_t = _rx_prog_tensors(pattern)
+    if group > prog.ngroups:
+        raise ValueError(f"regexp_extract group {group} > "
+                         f"{prog.ngroups} capture groups")
+    starts = torch.empty(n, dtype=torch.int32, device="cuda")
+    lens = torch.empty(n, dtype=torch.int64, device="cuda")
+    overflow = torch.zeros(1, dtype=torch.int32, device="cuda")
+    ext.regex_extract(prog_t.data_ptr(), len(prog.ops), cls_t.data_ptr(),
+                      col.offsets.data_ptr(), col.data.data_ptr(), group,
+                      starts.data_ptr(), lens.data_ptr(),
+                      overflow.data_ptr(), n, s)
+    if int(overflow.item()) > 0:
+        from . import cpu_backend
+
+        return cpu_backend.regexp_extract(col.cpu(), pattern, group).cuda()
+    return _strings_from_spans(col, starts, lens, n, s)
+
+
+def _strings_from_spans(col: Column, starts, lens, n, s) -> Column:
+    """Assemble a string column from absolute (start, len) spans over the
+    source bytes (substr_copy compaction)."""
+    scanned, total = _exclusive_scan_i64(lens)
+    out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
+                            device="cuda")[:total]
+    if total:
+        ext.substr_copy(col.data.data_ptr(), starts.data_ptr(),
+                        lens.data_ptr(), scanned.data_ptr(),
+                        out_bytes.data_ptr(), n, s)
+    offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+    offs[n] = total
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(DType.string(), n, out_bytes, v, offs,
+                  null_count=col._null_count)
+
+
+def _compile_replacement(replacement: str, ngroups: int):
+    """Java appendReplacement template -> (ops int32 triples, literal
+    bytes): $g group refs, backslash escapes."""
+    ops_l = []
+    lit = bytearray()
+    cur = bytearray()
+
+    def flush():
+        nonlocal cur
+        if cur:
+            ops_l.append((0, len(lit), len(cur)))
+            lit.extend(cur)
+            cur = bytearray()
+
+    i = 0
+    while i < len(replacement):
+        c = replacement[i]
+        if c == "$" and i + 1 < len(replacement) and \
+                replacement[i + 1].isdigit():
+            g = int(replacement[i + 1])
+            if g > ngroups:
+                raise ValueError(f"replacement group ${g} out of range")
+            flush()
+            ops_l.append((1, g, 0))
+            i += 2
+        elif c == "\\" and i + 1 < len(replacement):
+            cur.extend(replacement[i + 1].encode("utf-8"))
+            i += 2
+        else:
+            cur.extend(c.encode("utf-8"))
+            i += 1
+    flush()
+    return ops_l, bytes(lit)
+
+
+def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
+    n = col.size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.string())
+    prog, prog_t, cls_t = _rx_prog_tensors(pattern)
+    ops_l, lit = _compile_replacement(replacement, prog.ngroups)
+    flat = []
+    for k, a, b in ops_l:
+        flat.extend([k, a, b])
+    repl_t = torch.tensor(flat or [0, 0, 0], dtype=torch.int32).cuda()
+    lit_t = torch.frombuffer(bytearray(lit or b"\x00"),
+                             dtype=torch.uint8).cuda()
+    lens = torch.empty(n, dtype=torch.int64, device="cuda")
+    overflow = torch.zeros(1, dtype=torch.int32, device="cuda")
+    ext.regex_replace(prog_t.data_ptr(), len(prog.ops), cls_t.data_ptr(),
+                      col.offsets.data_ptr(), col.data.data_ptr(),
+                      repl_t.data_ptr(), len(ops_l), lit_t.data_ptr(),
+                      0, lens.data_ptr(), 0, 0, overflow.data_ptr(), n, s)
+    if int(overflow.item()) > 0:
+        from . import cpu_backend
+
+        return cpu_backend.regexp_replace(col.cpu(), pattern,
+                                          replacement).cuda()
+    scanned, total = _exclusive_scan_i64(lens)
+    out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
+                            device="cuda")[:total]
+    if total:
+        ext.regex_replace(prog_t.data_ptr(), len(prog.ops), cls_t.data_ptr(),
+                          col.offsets.data_ptr(), col.data.data_ptr(),
+                          repl_t.data_ptr(), len(ops_l), lit_t.data_ptr(),
+                          scanned.data_ptr(), lens.data_ptr(),
+                          out_bytes.data_ptr(), 1, overflow.data_ptr(), n, s)
+    offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+    offs[n] = total
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(DType.string(), n, out_bytes, v, offs,
+                  null_count=col._null_count)
+
+
 def substring(col: Column, pos: int, length: int = -1) -> Column:
     n = col.size
     s = _stream()

@@ -26,6 +26,7 @@ OP_JMP = 4     # arg0 = target
 OP_SPLIT = 5   # arg0, arg1 = targets (try arg0 first)
 OP_BOL = 6
 OP_EOL = 7
+OP_SAVE = 8    # arg0 = save slot (2*group / 2*group+1): capture positions
 
 MAX_EXPANSION = 32
 
@@ -77,6 +78,7 @@ class Program:
         self.ops: List[Tuple[int, int, int]] = []
         self.classes: List[bytes] = []
         self.anchored_start = False
+        self.ngroups = 0  # capturing groups (slots 2..2*ngroups+1)
 
     def emit(self, op, a0=0, a1=0) -> int:
         self.ops.append((op, a0, a1))
@@ -170,16 +172,25 @@ class _Parser:
     def _atom(self) -> list:
         ch = self.next()
         if ch == "(":
+            capturing = True
             if self.peek() == "?":
-                # (?: ... ) non-capturing is fine; anything else unsupported
+                # (?: ... ) non-capturing; anything else unsupported
                 self.next()
                 if self.peek() != ":":
                     raise RegexUnsupported("lookaround / named group")
                 self.next()
+                capturing = False
+            gid = 0
+            if capturing:
+                self.prog.ngroups += 1
+                gid = self.prog.ngroups
             inner = self._alternation()
             if self.peek() != ")":
                 raise RegexUnsupported("unbalanced (")
             self.next()
+            if capturing:
+                inner = [(OP_SAVE, 2 * gid, 0)] + \
+                    self._shift(inner, 1) + [(OP_SAVE, 2 * gid + 1, 0)]
             return inner
         if ch == "[":
             return [self._char_class()]

@@ -128,6 +128,13 @@ class Tagger:
                     compile_regex(e.pattern)
                 except RegexUnsupported as ex:
                     out.append(f"regex not supported on GPU: {ex}")
+        elif type(e).__name__ in ("RegexpExtract", "RegexpReplace"):
+            from ..ops.regex_compiler import RegexUnsupported, compile_regex
+
+            try:
+                compile_regex(e.pattern)
+            except RegexUnsupported as ex:
+                out.append(f"regex not supported on GPU: {ex}")
         elif isinstance(e, Substring):
             pass
         elif type(e).__name__ == "SampleHash":

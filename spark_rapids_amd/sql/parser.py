@@ -435,6 +435,15 @@ class Parser:
                 if not self.op(","):
                     break
             self.expect_op(")")
+        if name == "regexp_extract":
+            from ..expr.expressions import RegexpExtract
+
+            idx = int(args[2].value) if len(args) > 2 else 1
+            return RegexpExtract(args[0], args[1].value, idx)
+        if name == "regexp_replace":
+            from ..expr.expressions import RegexpReplace
+
+            return RegexpReplace(args[0], args[1].value, args[2].value)
         if name in ("percentile", "approx_percentile"):
             return A.percentile(args[0], float(args[1].value))
         if name in _AGG_FUNCS:

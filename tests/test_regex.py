@@ -58,3 +58,70 @@ def test_gpu_rlike_plan_and_fallback():
     # unsupported pattern tags the filter onto CPU instead of failing
     q2 = df.filter(col("s").rlike(r"(a)\1"))
     assert "CpuFilter" in q2.physical_plan().tree_string()
+
+
+class TestRegexpExtractReplace:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_extract_cpu(self, cpu):
+        df = cpu.create_dataframe(
+            {"s": ["ab-12-34", "x99y", None, "nope", ""]})
+        out = df.select(
+            col("s").regexp_extract(r"(\d+)-(\d+)", 1).alias("g1"),
+            col("s").regexp_extract(r"(\d+)-(\d+)", 2).alias("g2"),
+            col("s").regexp_extract(r"(\d+)", 1).alias("d")).to_pydict()
+        assert out["g1"] == ["12", "", None, "", ""]
+        assert out["g2"] == ["34", "", None, "", ""]
+        assert out["d"] == ["12", "99", None, "", ""]
+
+    def test_replace_cpu(self, cpu):
+        df = cpu.create_dataframe({"s": ["a1b22c", "", None, "xyz"]})
+        out = df.select(
+            col("s").regexp_replace(r"\d+", "#").alias("r"),
+            col("s").regexp_replace(r"([a-z])(\d)", "$2$1").alias("sw"),
+        ).to_pydict()
+        assert out["r"] == ["a#b#c", "", None, "xyz"]
+        assert out["sw"] == ["1a2b2c", "", None, "xyz"]
+
+    def test_replace_empty_match(self, cpu):
+        df = cpu.create_dataframe({"s": ["ab"]})
+        out = df.select(col("s").regexp_replace("x*", "-")).collect()
+        assert out == [("-a-b-",)]
+
+    def test_sql(self, cpu):
+        df = cpu.create_dataframe({"s": ["k=42"]})
+        cpu.register("trx", df)
+        out = cpu.sql(
+            "SELECT regexp_extract(s, '(\\d+)', 1) FROM trx").collect()
+        assert out == [("42",)]
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        import numpy as np
+
+        rng = np.random.default_rng(6)
+        words = ["item-%d-%d" % (a, b) for a, b in
+                 zip(rng.integers(0, 1000, 4000), rng.integers(0, 99, 4000))]
+        words += ["nomatch", "", "x-y-z"] * 100
+        vals = [w if i % 31 else None for i, w in enumerate(words)]
+
+        def q(s):
+            df = s.create_dataframe({"s": vals})
+            return df.select(
+                col("s").regexp_extract(r"(\d+)-(\d+)", 2).alias("g"),
+                col("s").regexp_replace(r"(\d+)", "<$1>").alias("r"),
+                col("s").regexp_replace(r"-", "_").alias("u")).to_pydict()
+
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        assert q(sg) == q(sc)
+
+    @pytest.mark.gpu
+    def test_gpu_placement(self):
+        sg = sr.Session()
+        df = sg.create_dataframe({"s": ["a1"]})
+        tree = (df.select(col("s").regexp_extract(r"(\d)", 1))
+                .physical_plan().tree_string())
+        assert "GpuProject" in tree, tree
