@@ -67,7 +67,9 @@ void hipdf_gb_rowgid(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_gb_agg(int, int, const void*, const void*, const void*, void*,
                   void*, int, int32_t, int64_t, hipStream_t);
 void hipdf_gb_agg_multi(const void*, int, const void*, const void*,
-                        int32_t, int64_t, hipStream_t);
+                        int32_t, int, int64_t, hipStream_t);
+void hipdf_gb_reduce_reps(int, void*, int, void*, int32_t, int,
+                          hipStream_t);
 void hipdf_gb_acc_init(int, void*, int, int32_t, hipStream_t);
 void hipdf_mask_from_nonzero(const void*, void*, int64_t, hipStream_t);
 void hipdf_join_build(const void*, const void*, int, void*, void*, int64_t,
@@ -355,10 +357,17 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
   m.def("gb_agg_multi", [](int64_t aggs, int naggs, int64_t row_gid,
-                           int64_t sel, int ngroups, int64_t n,
+                           int64_t sel, int ngroups, int nrep, int64_t n,
                            int64_t stream) {
-    hipdf_gb_agg_multi(P(aggs), naggs, P(row_gid), P(sel), ngroups, n,
+    hipdf_gb_agg_multi(P(aggs), naggs, P(row_gid), P(sel), ngroups, nrep, n,
                        S(stream));
+    check_async();
+  });
+  m.def("gb_reduce_reps", [](int op, int64_t acc, int acc_is_double,
+                             int64_t cnt, int ngroups, int nrep,
+                             int64_t stream) {
+    hipdf_gb_reduce_reps(op, PM(acc), acc_is_double, PM(cnt), ngroups, nrep,
+                         S(stream));
     check_async();
   });
   m.def("mask_from_nonzero", [](int64_t cnt, int64_t mask, int64_t n,

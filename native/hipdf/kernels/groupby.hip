@@ -259,7 +259,16 @@ template <bool USE_LDS>
 __global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
                                const int32_t* __restrict__ row_gid,
                                const int32_t* __restrict__ sel,
-                               int32_t ngroups, int64_t n) {
+                               int32_t ngroups, int nrep, int64_t n) {
+  // nrep > 1 (non-LDS path): each wave accumulates into one of nrep
+  // replica accumulator arrays, so a zipf-skewed hot group spreads its
+  // atomics over nrep addresses instead of serializing on one; replicas
+  // are folded by k_gb_reduce_reps afterwards.
+  size_t rep_off = 0;
+  if (!USE_LDS && nrep > 1) {
+    int wave = (int)(((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / 64);
+    rep_off = (size_t)(wave % nrep) * ngroups;
+  }
   extern __shared__ char lds_raw[];
   // LDS layout per agg a: acc[a][ngroups] (8B each) then cnt[a][ngroups]
   if (USE_LDS) {
@@ -288,7 +297,7 @@ __global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
       const AggDesc& d = aggs[a];
       char* base = lds_raw + (size_t)a * ngroups * 16;
       int64_t* cnt_p = USE_LDS ? (int64_t*)(base + (size_t)ngroups * 8)
-                               : d.cnt;
+                               : d.cnt + rep_off;
       if (d.op == GB_COUNT_ALL) {
         atomicAdd((unsigned long long*)&cnt_p[g], 1ull);
         continue;
@@ -297,10 +306,10 @@ __global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
       atomicAdd((unsigned long long*)&cnt_p[g], 1ull);
       if (d.op == GB_COUNT) continue;
       if (d.acc_is_double) {
-        double* acc_p = USE_LDS ? (double*)base : (double*)d.acc;
+        double* acc_p = USE_LDS ? (double*)base : (double*)d.acc + rep_off;
         acc_atomic<double>(d.op, &acc_p[g], load_as_double(d.vals, d.type, i));
       } else {
-        int64_t* acc_p = USE_LDS ? (int64_t*)base : (int64_t*)d.acc;
+        int64_t* acc_p = USE_LDS ? (int64_t*)base : (int64_t*)d.acc + rep_off;
         acc_atomic<int64_t>(d.op, &acc_p[g], load_as_i64(d.vals, d.type, i));
       }
     }
@@ -323,6 +332,29 @@ __global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
                               ((int64_t*)base)[g]);
       }
     }
+  }
+}
+
+// fold replica accumulators back into replica 0
+template <typename ACC>
+__global__ void k_gb_reduce_reps(int op, ACC* __restrict__ acc,
+                                 int64_t* __restrict__ cnt, int32_t ngroups,
+                                 int nrep) {
+  for (int g = blockIdx.x * blockDim.x + threadIdx.x; g < ngroups;
+       g += gridDim.x * blockDim.x) {
+    int64_t c = cnt[g];
+    ACC v = acc[g];
+    for (int r = 1; r < nrep; ++r) {
+      int64_t cr = cnt[(size_t)r * ngroups + g];
+      ACC vr = acc[(size_t)r * ngroups + g];
+      if (op == GB_SUM) v = v + vr;
+      else if (op == GB_MIN) v = (cr && (!c || vr < v)) ? vr : v;
+      else if (op == GB_MAX) v = (cr && (!c || vr > v)) ? vr : v;
+      else if (op == GB_FIRST && c == 0 && cr > 0) v = vr;
+      c += cr;
+    }
+    cnt[g] = c;
+    acc[g] = v;
   }
 }
 
@@ -521,8 +553,8 @@ void hipdf_gb_acc_init(int op, void* acc, int acc_is_double, int32_t ngroups,
 }
 
 void hipdf_gb_agg_multi(const void* aggs, int naggs, const void* row_gid,
-                        const void* sel, int32_t ngroups, int64_t n,
-                        hipStream_t stream) {
+                        const void* sel, int32_t ngroups, int nrep,
+                        int64_t n, hipStream_t stream) {
   size_t lds = (size_t)naggs * ngroups * 16;
   bool use_lds = lds > 0 && lds <= 64 * 1024;
   dim3 grid = flat_grid(n, 4);
@@ -530,12 +562,24 @@ void hipdf_gb_agg_multi(const void* aggs, int naggs, const void* row_gid,
     hipLaunchKernelGGL((k_gb_agg_multi<true>), grid, dim3(HIPDF_BLOCK), lds,
                        stream, (const AggDesc*)aggs, naggs,
                        (const int32_t*)row_gid, (const int32_t*)sel, ngroups,
-                       n);
+                       1, n);
   else
     hipLaunchKernelGGL((k_gb_agg_multi<false>), grid, dim3(HIPDF_BLOCK), 0,
                        stream, (const AggDesc*)aggs, naggs,
                        (const int32_t*)row_gid, (const int32_t*)sel, ngroups,
-                       n);
+                       nrep, n);
+}
+
+void hipdf_gb_reduce_reps(int op, void* acc, int acc_is_double, void* cnt,
+                          int32_t ngroups, int nrep, hipStream_t stream) {
+  if (acc_is_double)
+    hipLaunchKernelGGL((k_gb_reduce_reps<double>), flat_grid(ngroups),
+                       dim3(HIPDF_BLOCK), 0, stream, op, (double*)acc,
+                       (int64_t*)cnt, ngroups, nrep);
+  else
+    hipLaunchKernelGGL((k_gb_reduce_reps<int64_t>), flat_grid(ngroups),
+                       dim3(HIPDF_BLOCK), 0, stream, op, (int64_t*)acc,
+                       (int64_t*)cnt, ngroups, nrep);
 }
 
 void hipdf_mask_from_nonzero(const void* cnt, void* mask, int64_t n,

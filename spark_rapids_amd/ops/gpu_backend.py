@@ -947,6 +947,11 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
 
     # fused multi-aggregate: one kernel pass accumulates every agg;
     # decimal128 sums go through dedicated carry-correct i128 kernels
+    # replica count for the non-LDS atomic path: spreads a skew-hot
+    # group's atomics over nrep accumulator copies (zipf keys would
+    # otherwise serialize millions of atomics on one address)
+    nrep = 1 if ngroups * max(1, len(aggs)) <= 4096 \
+        else max(1, min(64, (1 << 21) // ngroups))
     allocs = []
     blobs = []
     for op, vidx, out_dtype in aggs:
@@ -988,13 +993,15 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
             continue
         acc_is_double = out_dtype.is_floating or (
             vc is not None and vc.dtype.is_floating)
-        acc = torch.empty(max(ngroups, 1),
+        acc = torch.empty(max(nrep * ngroups, 1),
                           dtype=torch.float64 if acc_is_double else torch.int64,
                           device="cuda")
-        cnt = torch.zeros(max(ngroups, 1), dtype=torch.int64, device="cuda")
+        cnt = torch.zeros(max(nrep * ngroups, 1), dtype=torch.int64,
+                          device="cuda")
         t = _ht(vc.dtype) if vc is not None else 4
         if op not in ("count", "count_all"):
-            ext.gb_acc_init(_GB[op], acc.data_ptr(), acc_is_double, ngroups, s)
+            ext.gb_acc_init(_GB[op], acc.data_ptr(), acc_is_double,
+                            nrep * ngroups, s)
         allocs.append((op, out_dtype, acc_is_double, acc, cnt))
         blobs.append(struct.pack(
             "<iiiiqqqq", _GB[op], t, acc_is_double, 0,
@@ -1005,7 +1012,14 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         desc = torch.frombuffer(bytearray(b"".join(blobs)),
                                 dtype=torch.uint8).cuda()
         ext.gb_agg_multi(desc.data_ptr(), len(blobs), row_gid.data_ptr(),
-                         selp, ngroups, n, s)
+                         selp, ngroups, nrep, n, s)
+        if nrep > 1:
+            for op, _, acc_is_double, acc, cnt in allocs:
+                if op in ("sum_d128", "collect"):
+                    continue
+                ext.gb_reduce_reps(_GB.get(op, 0), acc.data_ptr(),
+                                   1 if acc_is_double else 0,
+                                   cnt.data_ptr(), ngroups, nrep, s)
     for op, out_dtype, acc_is_double, acc, cnt in allocs:
         if op == "collect":
             out_cols.append(acc)
