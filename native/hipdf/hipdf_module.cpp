@@ -107,6 +107,9 @@ void hipdf_gb_sum_i64_to_i128(const void*, const void*, const void*,
                               const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_gb_sum_i128(const void*, const void*, const void*, const void*,
                        void*, void*, int64_t, hipStream_t);
+void hipdf_range_bounds(const void*, const void*, const void*, double,
+                        double, int, int, void*, void*, int64_t,
+                        hipStream_t);
 void hipdf_change_flags(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_iota_i32(void*, int64_t, hipStream_t);
 void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
@@ -400,6 +403,14 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
 
+  m.def("range_bounds", [](int64_t vals, int64_t seg_start, int64_t seg_end,
+                           double lo, double hi, int lo_unb, int hi_unb,
+                           int64_t a_idx, int64_t b_idx, int64_t n,
+                           int64_t stream) {
+    hipdf_range_bounds(P(vals), P(seg_start), P(seg_end), lo, hi, lo_unb,
+                       hi_unb, PM(a_idx), PM(b_idx), n, S(stream));
+    check_async();
+  });
   m.def("change_flags", [](int64_t keys, int nkeys, int64_t out, int64_t n,
                            int64_t stream) {
     hipdf_change_flags(P(keys), nkeys, PM(out), n, S(stream));

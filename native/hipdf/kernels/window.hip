@@ -78,7 +78,57 @@ __global__ void k_scan_add_offsets_f64(double* __restrict__ out,
   }
 }
 
+// RANGE frame bounds: for each row, binary-search its segment (ascending
+// order key, cast to double) for the first/last row whose key lies in
+// [key_i + lo, key_i + hi]; unbounded ends snap to the segment edge.
+__global__ void k_range_bounds(const double* __restrict__ vals,
+                               const int32_t* __restrict__ seg_start,
+                               const int32_t* __restrict__ seg_end,
+                               double lo, double hi, int lo_unb, int hi_unb,
+                               int32_t* __restrict__ a_idx,
+                               int32_t* __restrict__ b_idx, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t s0 = seg_start[i], e0 = seg_end[i];
+    double v = vals[i];
+    int32_t a = s0, b = e0;
+    if (!lo_unb) {
+      double target = v + lo;
+      int32_t l = s0, r = e0 + 1;  // first idx with vals[idx] >= target
+      while (l < r) {
+        int32_t m = l + (r - l) / 2;
+        if (vals[m] < target) l = m + 1;
+        else r = m;
+      }
+      a = l;
+    }
+    if (!hi_unb) {
+      double target = v + hi;
+      int32_t l = s0, r = e0 + 1;  // first idx with vals[idx] > target
+      while (l < r) {
+        int32_t m = l + (r - l) / 2;
+        if (vals[m] <= target) l = m + 1;
+        else r = m;
+      }
+      b = l - 1;
+    }
+    a_idx[i] = a;
+    b_idx[i] = b;
+  }
+}
+
 extern "C" {
+
+void hipdf_range_bounds(const void* vals, const void* seg_start,
+                        const void* seg_end, double lo, double hi,
+                        int lo_unb, int hi_unb, void* a_idx, void* b_idx,
+                        int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_range_bounds, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const double*)vals, (const int32_t*)seg_start,
+                     (const int32_t*)seg_end, lo, hi, lo_unb, hi_unb,
+                     (int32_t*)a_idx, (int32_t*)b_idx, n);
+}
+
 
 void hipdf_change_flags(const void* keys, int nkeys, void* out, int64_t n,
                         hipStream_t stream) {

@@ -24,7 +24,7 @@ class WindowSpec:
     def __init__(self, partition_by: Sequence[str] = (),
                  order_by: Sequence[str] = (),
                  descending: Optional[Sequence[bool]] = None,
-                 rows_between=None):
+                 rows_between=None, range_between=None):
         self.partition_by = list(partition_by)
         self.order_by = list(order_by)
         self.descending = list(descending) if descending is not None \
@@ -32,6 +32,10 @@ class WindowSpec:
         # (preceding, following) row offsets, e.g. (-3, 0) = 3 PRECEDING..
         # CURRENT ROW; None = default frame (running / whole partition)
         self.rows_between = rows_between
+        # RANGE frame: (lo, hi) VALUE offsets on the single numeric order
+        # key — frame = rows with key in [cur+lo, cur+hi]; None for
+        # UNBOUNDED on that end (reference: GpuSpecifiedWindowFrame RangeFrame)
+        self.range_between = range_between
 
 
 class WindowFunc:
@@ -47,9 +51,9 @@ class WindowFunc:
     def over(self, partition_by: Sequence[str] = (),
              order_by: Sequence[str] = (),
              descending: Optional[Sequence[bool]] = None,
-             rows_between=None) -> "WindowExpr":
+             rows_between=None, range_between=None) -> "WindowExpr":
         return WindowExpr(self, WindowSpec(partition_by, order_by, descending,
-                                           rows_between))
+                                           rows_between, range_between))
 
 
 class WindowExpr:
@@ -62,6 +66,15 @@ class WindowExpr:
             raise ValueError(f"{func.op} requires order_by")
         if func.op in OFFSETS and not spec.order_by:
             raise ValueError(f"{func.op} requires order_by")
+        if spec.range_between is not None:
+            if func.op not in AGGS:
+                raise ValueError("range_between needs an aggregate function")
+            if len(spec.order_by) != 1:
+                raise ValueError("range_between requires exactly one "
+                                 "order_by column")
+            lo, hi = spec.range_between
+            if lo is not None and hi is not None and lo > hi:
+                raise ValueError("range_between needs lo <= hi")
         if spec.rows_between is not None:
             if func.op not in AGGS:
                 raise ValueError("rows_between needs an aggregate function")

@@ -225,6 +225,17 @@ class Tagger:
                 op = w.func.op
                 if op in ("row_number", "rank", "dense_rank"):
                     continue
+                if w.spec.range_between is not None:
+                    okt = cs.field(w.spec.order_by[0]).dtype
+                    if op not in ("sum", "count", "mean"):
+                        reasons.append(
+                            f"range-frame {op} window not on GPU yet")
+                    elif w.spec.descending[0]:
+                        reasons.append(
+                            "range frame over descending order on CPU")
+                    elif not okt.is_numeric or okt.id is TypeId.DECIMAL128:
+                        reasons.append(
+                            f"range frame over {okt} order key on CPU")
                 vt = w.func.child.dtype(cs) if w.func.child is not None else None
                 if op in ("lag", "lead"):
                     if vt is not None and vt.is_nested:
@@ -239,7 +250,8 @@ class Tagger:
                         reasons.append(f"window {op}({vt}) not on GPU")
                     continue
                 if op in ("min", "max"):
-                    if spec.rows_between is not None:
+                    if spec.rows_between is not None \
+                            or spec.range_between is not None:
                         reasons.append(
                             f"bounded {op} window has no GPU kernel yet")
                     elif spec.order_by:

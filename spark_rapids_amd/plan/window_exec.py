@@ -253,8 +253,9 @@ class WindowExec(PhysicalExec):
                             False, n, s)
         nn_col = Column(DType.bool_(), n, valid_u8, None, null_count=0)
         rb = self.spec.rows_between
-        if rb is not None and op in ("sum", "count", "mean"):
-            lo_off, hi_off = rb
+        rgb = self.spec.range_between
+        if (rb is not None or rgb is not None) \
+                and op in ("sum", "count", "mean"):
             hp_ext = torch.cat([head_pos, torch.tensor(
                 [n], dtype=torch.int32, device="cuda")])
             segid_next = gb.binary_op_scalar(
@@ -264,13 +265,34 @@ class WindowExec(PhysicalExec):
                                                         segid_next.data, n),
                              None, null_count=0)
             seg_end = gb.binary_op_scalar("sub", seg_end, 1, INT32)
-            b_idx = gb.binary_op("min",
-                                 gb.binary_op_scalar("add", iota_col, hi_off,
-                                                     INT32), seg_end, INT32)
-            a_idx = gb.binary_op("max",
-                                 gb.binary_op_scalar("add", iota_col, lo_off,
-                                                     INT32), seg_start_col,
-                                 INT32)
+            if rb is not None:
+                lo_off, hi_off = rb
+                b_idx = gb.binary_op(
+                    "min", gb.binary_op_scalar("add", iota_col, hi_off,
+                                               INT32), seg_end, INT32)
+                a_idx = gb.binary_op(
+                    "max", gb.binary_op_scalar("add", iota_col, lo_off,
+                                               INT32), seg_start_col,
+                    INT32)
+            else:
+                # RANGE frame: per-row binary search of the ascending order
+                # key within the segment (k_range_bounds)
+                lo_v, hi_v = rgb
+                okey = table.columns[cs.index(self.spec.order_by[0])]
+                of = gb.cast(Column(okey.dtype, n, okey.data, None,
+                                    null_count=0), FLOAT64)
+                a_t = torch.empty(n, dtype=torch.int32, device="cuda")
+                b_t = torch.empty(n, dtype=torch.int32, device="cuda")
+                ext.range_bounds(of.data.data_ptr(),
+                                 seg_start_col.data.data_ptr(),
+                                 seg_end.data.data_ptr(),
+                                 float(lo_v if lo_v is not None else 0.0),
+                                 float(hi_v if hi_v is not None else 0.0),
+                                 1 if lo_v is None else 0,
+                                 1 if hi_v is None else 0,
+                                 a_t.data_ptr(), b_t.data_ptr(), n, s)
+                a_idx = _i32col(a_t)
+                b_idx = _i32col(b_t)
             am1 = gb.binary_op_scalar("sub", a_idx, 1, INT32)
             use_f64 = out_dt.is_floating
             work_t = FLOAT64 if use_f64 else INT64
@@ -485,11 +507,44 @@ def _compute(w: WindowExpr, table: ColumnBatch, cs, n, heads, ochange, idx,
     vv = np.where(valid, v, 0.0)
     cnt_f = valid.astype(np.int64)
     rb = w.spec.rows_between
-    if rb is not None:
-        lo_off, hi_off = rb
+    rgb = w.spec.range_between
+    if rb is not None or rgb is not None:
         seg_end = _segment_ends(heads, idx, n)
-        a = np.maximum(idx + lo_off, seg_start)
-        b = np.minimum(idx + hi_off, seg_end)
+        if rb is not None:
+            lo_off, hi_off = rb
+            a = np.maximum(idx + lo_off, seg_start)
+            b = np.minimum(idx + hi_off, seg_end)
+        else:
+            # RANGE frame: per-segment searchsorted on the order key
+            # (descending handled by negation; null keys as -inf peers)
+            lo_v, hi_v = rgb
+            okc = table.columns[cs.index(w.spec.order_by[0])]
+            from ..ops.cpu_backend import _vals as _cvals, _valid as _cvalid
+
+            ov = _cvals(okc).astype(np.float64).copy()
+            ovalid = _cvalid(okc)
+            ov[~ovalid] = -np.inf
+            if w.spec.descending[0]:
+                # frame offsets follow the SORT direction: on the negated
+                # (ascending) axis the same (lo, hi) apply directly
+                ov = -ov
+            a = np.empty(n, dtype=np.int64)
+            b = np.empty(n, dtype=np.int64)
+            starts = np.flatnonzero(heads)
+            bounds = np.append(starts, n)
+            for si in range(len(starts)):
+                s0, e0 = bounds[si], bounds[si + 1]
+                seg = ov[s0:e0]
+                tgt = ov[s0:e0]
+                if lo_v is None:
+                    a[s0:e0] = s0
+                else:
+                    a[s0:e0] = s0 + np.searchsorted(seg, tgt + lo_v, "left")
+                if hi_v is None:
+                    b[s0:e0] = e0 - 1
+                else:
+                    b[s0:e0] = s0 + np.searchsorted(seg, tgt + hi_v,
+                                                    "right") - 1
         empty_frame = b < a
         if op in ("min", "max"):
             fn = min if op == "min" else max
