@@ -124,7 +124,7 @@ __global__ void k_gb_rowgid(const int32_t* __restrict__ row_slot,
 
 // ---- aggregation ---------------------------------------------------------
 enum GbOp : int { GB_SUM = 0, GB_MIN, GB_MAX, GB_COUNT, GB_COUNT_ALL,
-                  GB_FIRST };
+                  GB_FIRST, GB_BITAND, GB_BITOR, GB_BITXOR };
 
 template <typename ACC>
 __device__ __forceinline__ void acc_atomic(int op, ACC* addr, ACC v);
@@ -134,6 +134,12 @@ __device__ __forceinline__ void acc_atomic<int64_t>(int op, int64_t* addr,
                                                     int64_t v) {
   if (op == GB_MIN) atomicMin((long long*)addr, (long long)v);
   else if (op == GB_MAX) atomicMax((long long*)addr, (long long)v);
+  else if (op == GB_BITAND)
+    atomicAnd((unsigned long long*)addr, (unsigned long long)v);
+  else if (op == GB_BITOR)
+    atomicOr((unsigned long long*)addr, (unsigned long long)v);
+  else if (op == GB_BITXOR)
+    atomicXor((unsigned long long*)addr, (unsigned long long)v);
   else if (op == GB_FIRST) *addr = v;  // any-value semantics (Spark first
   // without ordering is unspecified); aligned 8B store cannot tear
   else atomicAdd((unsigned long long*)addr, (unsigned long long)v);
@@ -167,6 +173,7 @@ template <typename ACC>
 __device__ __forceinline__ ACC acc_init(int op) {
   if (op == GB_MIN) return std::numeric_limits<ACC>::max();
   if (op == GB_MAX) return std::numeric_limits<ACC>::lowest();
+  if (op == GB_BITAND) return (ACC)-1;  // all ones identity
   return (ACC)0;
 }
 
@@ -350,6 +357,12 @@ __global__ void k_gb_reduce_reps(int op, ACC* __restrict__ acc,
       if (op == GB_SUM) v = v + vr;
       else if (op == GB_MIN) v = (cr && (!c || vr < v)) ? vr : v;
       else if (op == GB_MAX) v = (cr && (!c || vr > v)) ? vr : v;
+      else if (op == GB_BITAND)
+        v = (ACC)((int64_t)v & (int64_t)vr);
+      else if (op == GB_BITOR)
+        v = (ACC)((int64_t)v | (int64_t)vr);
+      else if (op == GB_BITXOR)
+        v = (ACC)((int64_t)v ^ (int64_t)vr);
       else if (op == GB_FIRST && c == 0 && cr > 0) v = vr;
       c += cr;
     }

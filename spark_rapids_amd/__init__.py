@@ -8,7 +8,8 @@ spark.rapids.* config registry.
 from .api import DataFrame, Session
 from .column import Column, ColumnBatch, Field, Schema
 from .config import RapidsConf, help_doc
-from .expr.aggregates import (approx_percentile, avg, collect_list,
+from .expr.aggregates import (approx_count_distinct, approx_percentile,
+                              avg, bit_and, bit_or, bit_xor, collect_list,
                               collect_set, percentile,
                               count, count_distinct, count_star, first, last,
                               max_, min_, stddev, sum_distinct,

@@ -54,6 +54,8 @@ class AggExpr:
             return FLOAT64
         if self.op in ("min", "max", "first", "last"):
             return ct
+        if self.op in ("bit_and", "bit_or", "bit_xor"):
+            return ct
         if self.op in ("collect_list", "collect_set"):
             return DType.list_(ct)
         if self.op.startswith("percentile:"):
@@ -120,6 +122,25 @@ def first(e) -> AggExpr:
 def last(e) -> AggExpr:
     """Last non-null value (GpuLast); same determinism caveats as first."""
     return AggExpr("last", e)
+
+
+def bit_and(e) -> AggExpr:
+    """Bitwise AND of the group's non-null integers (GpuBitAndAgg)."""
+    return AggExpr("bit_and", e)
+
+
+def bit_or(e) -> AggExpr:
+    return AggExpr("bit_or", e)
+
+
+def bit_xor(e) -> AggExpr:
+    return AggExpr("bit_xor", e)
+
+
+def approx_count_distinct(e) -> AggExpr:
+    """Served by the exact two-level distinct rewrite (always at least as
+    accurate as the reference's HyperLogLog++ approximation)."""
+    return AggExpr("count", e, distinct=True)
 
 
 def percentile(e, p: float) -> AggExpr:

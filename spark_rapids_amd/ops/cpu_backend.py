@@ -800,6 +800,17 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
             res = np.bincount(codes[av], minlength=ngroups).astype(np.int64)
             out_cols.append(_make(res, None, out_dtype))
             continue
+        if op in ("bit_and", "bit_or", "bit_xor"):
+            ident = -1 if op == "bit_and" else 0
+            res = np.full(ngroups, ident, dtype=np.int64)
+            ufunc = {"bit_and": np.bitwise_and, "bit_or": np.bitwise_or,
+                     "bit_xor": np.bitwise_xor}[op]
+            ufunc.at(res, codes[av], a[av].astype(np.int64))
+            cnt = np.bincount(codes[av], minlength=ngroups)
+            gv = cnt > 0
+            out_cols.append(_make(res.astype(out_dtype.numpy_dtype()),
+                                  gv if not gv.all() else None, out_dtype))
+            continue
         if op in ("first", "last"):
             idx = np.nonzero(av)[0]
             if op == "first":

@@ -58,7 +58,7 @@ _UN_OPS = {
 _HK_INT, _HK_LONG, _HK_FLOAT, _HK_DOUBLE = 0, 1, 2, 3
 _RED = {"sum": 0, "min": 1, "max": 2, "count": 3}
 _GB = {"sum": 0, "min": 1, "max": 2, "count": 3, "count_all": 4,
-       "first": 5, "last": 5}
+       "first": 5, "last": 5, "bit_and": 6, "bit_or": 7, "bit_xor": 8}
 _JOIN = {"inner": 0, "left": 1, "semi": 2, "anti": 3, "full": 4}
 
 

@@ -193,8 +193,12 @@ class Tagger:
                         reasons.append(
                             f"agg {a.op} over strings not on GPU yet")
                     elif r and a.op not in ("count", "count_all", "min",
-                                            "max", "first", "last"):
+                                            "max", "first", "last",
+                                            "bit_and", "bit_or", "bit_xor"):
                         reasons.append(f"agg {a.op}({a.child}): {r}")
+                    if a.op in ("bit_and", "bit_or", "bit_xor") \
+                            and not t.is_integral:
+                        reasons.append(f"{a.op} needs an integral input")
                     if t.id is TypeId.DECIMAL128 and a.op not in (
                             "sum", "count", "count_all",
                             "collect_list", "collect_set"):

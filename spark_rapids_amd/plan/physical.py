@@ -201,7 +201,8 @@ def _lower_aggs(aggs: List[AggExpr], in_schema: Schema):
             partial.append(("count", len(value_exprs) - 1, INT64))
             merge.append("sum")
             final.append(("col", j))
-        elif a.op in ("sum", "min", "max", "first", "last"):
+        elif a.op in ("sum", "min", "max", "first", "last",
+                      "bit_and", "bit_or", "bit_xor"):
             value_exprs.append(a.child)
             j = len(partial)
             partial.append((a.op, len(value_exprs) - 1, a.out_dtype(in_schema)))

@@ -163,3 +163,35 @@ def test_gpu_percentile_matches_cpu():
         assert rg[0] == rc[0] and rg[3] == rc[3]
         assert rg[1] == pytest.approx(rc[1], rel=1e-12)
         assert rg[2] == pytest.approx(rc[2], rel=1e-12)
+
+
+def test_bit_aggs_cpu(cpu):
+    from spark_rapids_amd import bit_and, bit_or, bit_xor
+
+    df = cpu.create_dataframe({
+        "k": [1, 1, 2, 2, 3],
+        "v": [0b1100, 0b1010, 7, None, None]})
+    rows = sorted(df.group_by("k").agg(
+        bit_and(col("v")), bit_or(col("v")), bit_xor(col("v"))).collect())
+    assert rows == [(1, 8, 14, 6), (2, 7, 7, 7), (3, None, None, None)]
+
+
+@pytest.mark.gpu
+def test_gpu_bit_aggs_match_cpu():
+    from spark_rapids_amd import bit_and, bit_or, bit_xor
+    import numpy as np
+
+    rng = np.random.default_rng(5)
+    data = {"k": [int(v) for v in rng.integers(0, 50, 30000)],
+            "v": [int(v) if v % 7 else None
+                  for v in rng.integers(0, 2**40, 30000)]}
+
+    def q(s):
+        df = s.create_dataframe(data)
+        return sorted(df.group_by("k").agg(
+            bit_and(col("v")), bit_or(col("v")),
+            bit_xor(col("v")), count_star()).collect())
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    assert q(sg) == q(sc)
