@@ -301,6 +301,10 @@ class Session:
         from .tools import lore
 
         lore.configure(self.conf.get_raw("spark.rapids.sql.lore.dumpPath"))
+        from .config import SHUFFLE_CODEC
+        from .shuffle import dist as _dist
+
+        _dist.set_codec(self.conf.get(SHUFFLE_CODEC))
 
     # ---- conf ----------------------------------------------------------
     def set(self, key: str, value) -> "Session":

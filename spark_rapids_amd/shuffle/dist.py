@@ -29,6 +29,41 @@ class DistContext:
 
 _ctx: Optional[DistContext] = None
 
+# Host-path shuffle compression codec (reference analogue: the nvcomp /
+# lz4 shuffle codecs — GpuCompressedColumnVector). Device buffers stay
+# uncompressed: xGMI moves them at link speed and compressing would force
+# a host round-trip; the codec applies to the gloo (host) transport only.
+_codec: Optional[str] = None
+
+
+def set_codec(name: Optional[str]):
+    global _codec
+    _codec = None if not name or name == "none" else name
+
+
+def _compress(t: torch.Tensor) -> torch.Tensor:
+    if _codec is None or t.is_cuda or t.numel() == 0:
+        return t
+    import pyarrow as pa
+
+    raw = t.numpy().tobytes()
+    comp = pa.Codec(_codec).compress(raw, asbytes=True)
+    import numpy as np
+
+    hdr = np.array([len(raw)], dtype=np.int64).tobytes()
+    return torch.frombuffer(bytearray(hdr + comp), dtype=torch.uint8)
+
+
+def _decompress(t: torch.Tensor) -> torch.Tensor:
+    if _codec is None or t.is_cuda or t.numel() == 0:
+        return t
+    import pyarrow as pa
+
+    b = t.numpy().tobytes()
+    raw_len = int.from_bytes(b[:8], "little")
+    out = pa.Codec(_codec).decompress(b[8:], raw_len, asbytes=True)
+    return torch.frombuffer(bytearray(out), dtype=torch.uint8)
+
 
 def ctx() -> DistContext:
     global _ctx
@@ -44,6 +79,8 @@ def all_to_all_bytes(send: List[torch.Tensor]) -> List[torch.Tensor]:
     c = ctx()
     world = c.world
     assert len(send) == world
+    if c.backend != "nccl":
+        send = [_compress(t) for t in send]
     sizes = torch.tensor([t.numel() for t in send], dtype=torch.int64)
     dev = send[0].device if send else torch.device("cpu")
     use_device = dev.type == "cuda"
@@ -80,7 +117,7 @@ def all_to_all_bytes(send: List[torch.Tensor]) -> List[torch.Tensor]:
     if ops:
         for w in td.batch_isend_irecv(ops):
             w.wait()
-    return recv
+    return [_decompress(t) for t in recv]
 
 
 def all_gather_bytes(buf: torch.Tensor) -> List[torch.Tensor]:
