@@ -301,7 +301,7 @@ class Session:
         from .tools import lore
 
         lore.configure(self.conf.get_raw("spark.rapids.sql.lore.dumpPath"))
-        from .config import SHUFFLE_CODEC
+        from .config import SHUFFLE_COMPRESS as SHUFFLE_CODEC
         from .shuffle import dist as _dist
 
         _dist.set_codec(self.conf.get(SHUFFLE_CODEC))

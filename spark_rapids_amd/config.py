@@ -158,7 +158,9 @@ SHUFFLE_PARTITIONS = int_conf(
     "Default number of shuffle partitions per GPU for exchanges.")
 SHUFFLE_COMPRESS = str_conf(
     "spark.rapids.shuffle.compression.codec", "none",
-    "Compression codec for host-staged shuffle payloads: none, lz4.")
+    "Codec for host-path shuffle payloads (none|zstd|lz4|snappy). Device "
+    "(RCCL/xGMI) shuffles stay uncompressed: the links outrun a host "
+    "compression round-trip.")
 RETRY_MAX_SPLITS = int_conf(
     "spark.rapids.sql.retry.maxSplits", 8,
     "Maximum recursive batch splits attempted by the OOM retry framework "
@@ -173,11 +175,6 @@ PRUNE_COLUMNS = bool_conf(
     "spark.rapids.sql.optimizer.pruneColumns.enabled", True,
     "Push projections below joins/aggregates so unused columns are never "
     "gathered or transferred (Catalyst-optimizer analogue).")
-SHUFFLE_CODEC = str_conf(
-    "spark.rapids.shuffle.compression.codec", "none",
-    "Codec for host-path shuffle payloads (none|zstd|lz4|snappy). Device "
-    "(RCCL/xGMI) shuffles stay uncompressed: the links outrun a host "
-    "compression round-trip.")
 PUSH_FILTERS = bool_conf(
     "spark.rapids.sql.optimizer.pushFilters.enabled", True,
     "Push filter conjuncts below joins when they reference only one side "
