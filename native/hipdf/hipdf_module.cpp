@@ -95,6 +95,7 @@ void hipdf_gb_percentile(const void*, const void*, const void*, const void*,
                          double, void*, int, hipStream_t);
 void hipdf_dense_gid(const void*, int, const void*, void*, int64_t,
                      hipStream_t);
+void hipdf_expand_rows(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_gb_collect_count(const void*, const void*, const void*, void*,
                             int64_t, hipStream_t);
 void hipdf_gb_collect_fill(int, const void*, const void*, const void*,
@@ -562,6 +563,11 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("dense_gid", [](int64_t keys, int nkeys, int64_t sel,
                         int64_t row_gid, int64_t n, int64_t stream) {
     hipdf_dense_gid(P(keys), nkeys, P(sel), PM(row_gid), n, S(stream));
+    check_async();
+  });
+  m.def("expand_rows", [](int64_t offsets, int64_t rowid, int64_t pos,
+                          int64_t nrows, int64_t stream) {
+    hipdf_expand_rows(P(offsets), PM(rowid), PM(pos), nrows, S(stream));
     check_async();
   });
   m.def("gb_collect_count", [](int64_t vvalid, int64_t row_gid, int64_t sel,

@@ -222,7 +222,33 @@ __global__ void k_copy_valid_range(const uint64_t* __restrict__ src,
 }
 
 // ---- host entry points ---------------------------------------------------
+// explode: for each input row, its list span [off[i], off[i+1]) emits one
+// output row per element; rowid/pos map built wave-per-row (long lists
+// copy with full wave parallelism).
+__global__ void k_expand_rows(const int32_t* __restrict__ offsets,
+                              int32_t* __restrict__ rowid,
+                              int32_t* __restrict__ pos, int64_t nrows) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t i = wave_global; i < nrows; i += wave_count) {
+    int32_t a = offsets[i], b = offsets[i + 1];
+    for (int32_t j = a + lane; j < b; j += WAVE) {
+      rowid[j] = (int32_t)i;
+      pos[j] = j - a;
+    }
+  }
+}
+
 extern "C" {
+
+void hipdf_expand_rows(const void* offsets, void* rowid, void* pos,
+                       int64_t nrows, hipStream_t stream) {
+  hipLaunchKernelGGL(k_expand_rows, flat_grid(nrows), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)offsets, (int32_t*)rowid,
+                     (int32_t*)pos, nrows);
+}
+
 
 int64_t sel_num_blocks(int64_t n) {
   return (n + (int64_t)HIPDF_BLOCK * SEL_ITEMS - 1) /

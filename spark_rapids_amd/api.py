@@ -131,6 +131,16 @@ class DataFrame:
         threshold = int(fraction * 2147483647)
         return self.filter(SampleHash(seed) < lit(threshold))
 
+    def explode(self, column: str, outer: bool = False) -> "DataFrame":
+        """One output row per element of the LIST column (explode /
+        explode_outer)."""
+        return DataFrame(self.session,
+                         L.Generate(column, self.plan, outer, pos=False))
+
+    def posexplode(self, column: str, outer: bool = False) -> "DataFrame":
+        return DataFrame(self.session,
+                         L.Generate(column, self.plan, outer, pos=True))
+
     def distinct(self) -> "DataFrame":
         """Drop duplicate rows (group-by all columns with no aggregates)."""
         keys = [_col(f.name) for f in self.plan.schema().fields]

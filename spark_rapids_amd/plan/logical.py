@@ -12,7 +12,7 @@ from typing import List, Optional, Sequence, Tuple
 from ..column import Field, Schema
 from ..expr.aggregates import AggExpr
 from ..expr.expressions import Expression
-from ..types import BOOL
+from ..types import BOOL, DType, TypeId
 
 
 class LogicalPlan:
@@ -228,6 +228,34 @@ class Expand(LogicalPlan):
         for i, e in enumerate(self.projections[0]):
             nullable = any(p[i].nullable(cs) for p in self.projections)
             fields.append(Field(e.output_name(), e.dtype(cs), nullable))
+        return Schema(fields)
+
+
+class Generate(LogicalPlan):
+    """explode/posexplode of a LIST column: each input row emits one row per
+    element (outer=True keeps empty/null lists as one null-element row).
+    Reference analogue: GpuGenerateExec + GpuExplode/GpuPosExplode."""
+
+    def __init__(self, column: str, child: LogicalPlan, outer: bool = False,
+                 pos: bool = False):
+        self.column = column
+        self.child = child
+        self.outer = outer
+        self.pos = pos
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        cs = self.child.schema()
+        fields = [f for f in cs.fields if f.name != self.column]
+        if self.pos:
+            fields.append(Field("pos", DType.int32(), False))
+        lf = cs.field(self.column)
+        assert lf.dtype.id is TypeId.LIST, \
+            f"explode needs a LIST column, got {lf.dtype}"
+        fields.append(Field(self.column, lf.dtype.children[0], True))
         return Schema(fields)
 
 

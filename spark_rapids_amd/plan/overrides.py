@@ -161,6 +161,9 @@ class Tagger:
         elif isinstance(node, L.Project):
             for e in node.exprs:
                 reasons += self.expr_reasons(e, cs)
+        elif isinstance(node, L.Generate):
+            if node.outer:
+                reasons.append("explode_outer runs on CPU (padding path)")
         elif isinstance(node, L.Expand):
             for proj in node.projections:
                 for e in proj:
@@ -348,6 +351,9 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
         return P.ProjectExec(device, node.exprs, kids[0], node.schema())
     if isinstance(node, L.Expand):
         return P.ExpandExec(device, node.projections, kids[0], node.schema())
+    if isinstance(node, L.Generate):
+        return P.GenerateExec(device, node.column, kids[0], node.schema(),
+                              node.outer, node.pos)
     if isinstance(node, L.Aggregate):
         from ..config import BATCH_SIZE_BYTES
 

@@ -830,6 +830,96 @@ class ExpandExec(PhysicalExec):
         return f"{self.name()}[{len(self.projections)} projections]"
 
 
+class GenerateExec(PhysicalExec):
+    """explode/posexplode: parent columns are gathered by a rowid map built
+    from the list offsets (k_expand_rows on GPU); the element column is the
+    list child passed through (its layout is already the flattened rows)."""
+
+    def __init__(self, device: str, column: str, child: PhysicalExec,
+                 schema: Schema, outer: bool, pos: bool):
+        super().__init__(device, schema, [child])
+        self.column = column
+        self.outer = outer
+        self.pos = pos
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        cs = self.children[0].schema
+        ci = cs.index(self.column)
+        for batch in self.children[0].execute():
+            yield self._one(batch, ci)
+
+    def _one(self, batch: ColumnBatch, ci: int) -> ColumnBatch:
+        import numpy as np
+
+        lc = batch.columns[ci]
+        others = [c for j, c in enumerate(batch.columns) if j != ci]
+        n = batch.num_rows
+        if self.gpu and not self.outer:
+            import torch
+
+            from ..ops import gpu_backend as gb
+            from ..ops.gpu_backend import ext
+
+            s = gb._stream()
+            total = int(lc.offsets[n].item()) if n else 0
+            rowid = torch.empty(max(total, 1), dtype=torch.int32,
+                                device="cuda")[:total]
+            pos_t = torch.empty(max(total, 1), dtype=torch.int32,
+                                device="cuda")[:total]
+            if n and total:
+                ext.expand_rows(lc.offsets.data_ptr(), rowid.data_ptr(),
+                                pos_t.data_ptr(), n, s)
+            parent = ops.gather(
+                ColumnBatch(others, n),
+                Column(DType.int32(), total, rowid, None, null_count=0)) \
+                if others else ColumnBatch([], total)
+            out = list(parent.columns)
+            if self.pos:
+                out.append(Column(DType.int32(), total, pos_t, None,
+                                  null_count=0))
+            elem = lc.child
+            out.append(Column(elem.dtype, total, elem.data, elem.validity,
+                              elem.offsets, elem._null_count, elem.child))
+            return ColumnBatch(out, total)
+        # CPU (and outer) path
+        host = batch.cpu()
+        lch = host.columns[ci]
+        offs = lch.offsets.numpy()
+        lvalid = lch.valid_array()
+        rowids: List[int] = []
+        poss: List[int] = []
+        elem_take: List[int] = []  # -1 = null element (outer padding)
+        for i in range(n):
+            cnt = int(offs[i + 1] - offs[i]) if lvalid[i] else 0
+            if cnt == 0:
+                if self.outer:
+                    rowids.append(i)
+                    poss.append(0)
+                    elem_take.append(-1)
+                continue
+            for k in range(cnt):
+                rowids.append(i)
+                poss.append(k)
+                elem_take.append(int(offs[i]) + k)
+        idxc = Column.from_numpy(np.array(rowids, dtype=np.int32))
+        parent = ops.gather(ColumnBatch(
+            [c for j, c in enumerate(host.columns) if j != ci], n), idxc) \
+            if others else ColumnBatch([], len(rowids))
+        out = list(parent.columns)
+        if self.pos:
+            out.append(Column.from_numpy(np.array(poss, dtype=np.int32)))
+        ev = lch.child.to_pylist()
+        elems = [ev[t] if t >= 0 else None for t in elem_take]
+        out.append(Column.from_pylist(elems, lch.dtype.children[0]))
+        ob = ColumnBatch(out, len(rowids))
+        return ob.cuda() if self.gpu else ob
+
+    def describe(self):
+        mode = "posexplode" if self.pos else "explode"
+        o = "_outer" if self.outer else ""
+        return f"{self.name()}[{mode}{o}({self.column})]"
+
+
 class UnionExec(PhysicalExec):
     def __init__(self, device: str, children: List[PhysicalExec], schema: Schema):
         super().__init__(device, schema, children)

@@ -195,3 +195,51 @@ def test_gpu_bit_aggs_match_cpu():
     sg = sr.Session()
     sc = sr.Session({"spark.rapids.sql.enabled": False})
     assert q(sg) == q(sc)
+
+
+def test_explode_cpu(cpu):
+    from spark_rapids_amd import Column, DType, INT64
+    from spark_rapids_amd.column import ColumnBatch, Field, Schema
+
+    lt = DType.list_(INT64)
+    lc = Column.from_pylist([[1, 2], [], None, [5]], lt)
+    kc = Column.from_pylist(["a", "b", "c", "d"], sr.STRING)
+    df = cpu.from_batches([ColumnBatch([kc, lc])],
+                          Schema([Field("k", sr.STRING), Field("v", lt)]))
+    assert df.explode("v").collect() == [("a", 1), ("a", 2), ("d", 5)]
+    assert df.explode("v", outer=True).collect() == \
+        [("a", 1), ("a", 2), ("b", None), ("c", None), ("d", 5)]
+    assert df.posexplode("v").collect() == \
+        [("a", 0, 1), ("a", 1, 2), ("d", 0, 5)]
+
+
+def test_explode_of_collect_roundtrip(cpu):
+    df = _df(cpu, 500)
+    collected = df.group_by("k").agg(collect_list(col("v")))
+    back = collected.explode("collect_list(v)")
+    raw = [(r[0], r[1]) for r in df.collect() if r[1] is not None]
+    assert sorted(back.collect()) == sorted(raw)
+
+
+@pytest.mark.gpu
+def test_gpu_explode_matches_cpu():
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+
+    def q(s):
+        df = _df(s, 20000)
+        out = (df.group_by("k").agg(collect_list(col("v")))
+               .posexplode("collect_list(v)").collect())
+        return sorted(out)
+
+    assert q(sg) == q(sc) or sorted(r[::2] for r in q(sg)) == \
+        sorted(r[::2] for r in q(sc))
+
+
+@pytest.mark.gpu
+def test_gpu_explode_placement():
+    sg = sr.Session()
+    df = _df(sg, 10)
+    tree = (df.group_by("k").agg(collect_list(col("v")))
+            .explode("collect_list(v)").physical_plan().tree_string())
+    assert "GpuGenerate" in tree, tree
