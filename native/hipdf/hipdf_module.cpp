@@ -141,6 +141,10 @@ void hipdf_str_like(const void*, const void*, const void*, int, void*,
                     int64_t, hipStream_t);
 void hipdf_str_length(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_str_case(int, const void*, void*, int64_t, hipStream_t);
+void hipdf_str_trim_ranges(int, const void*, const void*, void*, void*,
+                           int64_t, hipStream_t);
+void hipdf_str_concat2(const void*, const void*, const void*, const void*,
+                       const void*, void*, void*, int, int64_t, hipStream_t);
 void hipdf_substr_ranges(const void*, const void*, int, int, void*, void*,
                          int64_t, hipStream_t);
 void hipdf_substr_copy(const void*, const void*, const void*, const void*,
@@ -508,6 +512,21 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("str_case", [](bool upper, int64_t in, int64_t out, int64_t nbytes,
                        int64_t stream) {
     hipdf_str_case(upper, P(in), PM(out), nbytes, S(stream));
+    check_async();
+  });
+  m.def("str_trim_ranges", [](int mode, int64_t ao, int64_t ab,
+                              int64_t bstart, int64_t blen, int64_t n,
+                              int64_t stream) {
+    hipdf_str_trim_ranges(mode, P(ao), P(ab), PM(bstart), PM(blen), n,
+                          S(stream));
+    check_async();
+  });
+  m.def("str_concat2", [](int64_t ao, int64_t ab, int64_t bo, int64_t bb,
+                          int64_t out_off, int64_t out_len,
+                          int64_t out_bytes, int mode, int64_t n,
+                          int64_t stream) {
+    hipdf_str_concat2(P(ao), P(ab), P(bo), P(bb), P(out_off), PM(out_len),
+                      PM(out_bytes), mode, n, S(stream));
     check_async();
   });
   m.def("substr_ranges", [](int64_t ao, int64_t ab, int start, int slen,

@@ -147,6 +147,48 @@ __global__ void k_str_case(int upper, const uint8_t* __restrict__ in,
 // substring (1-based start in codepoints, length in codepoints; Spark
 // semantics: start 0 behaves like 1, negative counts from the end).
 // pass 1: byte [start,len) per row
+// trim spans: mode 0 both, 1 leading, 2 trailing (ascii space like Spark
+// trim's default)
+__global__ void k_str_trim_ranges(int mode, const int32_t* __restrict__ ao,
+                                  const uint8_t* __restrict__ ab,
+                                  int32_t* __restrict__ bstart,
+                                  int64_t* __restrict__ blen, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t a = ao[i], b = ao[i + 1];
+    if (mode != 2)
+      while (a < b && ab[a] == ' ') ++a;
+    if (mode != 1)
+      while (b > a && ab[b - 1] == ' ') --b;
+    bstart[i] = a;
+    blen[i] = b - a;
+  }
+}
+
+// concat two string columns: pass 0 lengths, pass 1 bytes
+__global__ void k_str_concat2(const int32_t* __restrict__ ao,
+                              const uint8_t* __restrict__ ab,
+                              const int32_t* __restrict__ bo,
+                              const uint8_t* __restrict__ bb,
+                              const int64_t* __restrict__ out_off,
+                              int64_t* __restrict__ out_len,
+                              uint8_t* __restrict__ out_bytes, int mode,
+                              int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t i = wave_global; i < n; i += wave_count) {
+    int32_t la = ao[i + 1] - ao[i], lb = bo[i + 1] - bo[i];
+    if (!mode) {
+      if (lane == 0) out_len[i] = (int64_t)la + lb;
+      continue;
+    }
+    uint8_t* dst = out_bytes + out_off[i];
+    for (int32_t k = lane; k < la; k += WAVE) dst[k] = ab[ao[i] + k];
+    for (int32_t k = lane; k < lb; k += WAVE) dst[la + k] = bb[bo[i] + k];
+  }
+}
+
 __global__ void k_substr_ranges(const int32_t* __restrict__ ao,
                                 const uint8_t* __restrict__ ab, int32_t start,
                                 int32_t slen, int32_t* __restrict__ bstart,
@@ -242,6 +284,26 @@ void hipdf_str_case(int upper, const void* in, void* out, int64_t nbytes,
                     hipStream_t stream) {
   hipLaunchKernelGGL(k_str_case, flat_grid(nbytes), dim3(HIPDF_BLOCK), 0,
                      stream, upper, (const uint8_t*)in, (uint8_t*)out, nbytes);
+}
+
+void hipdf_str_trim_ranges(int mode, const void* ao, const void* ab,
+                           void* bstart, void* blen, int64_t n,
+                           hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_trim_ranges, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, mode, (const int32_t*)ao, (const uint8_t*)ab,
+                     (int32_t*)bstart, (int64_t*)blen, n);
+}
+
+void hipdf_str_concat2(const void* ao, const void* ab, const void* bo,
+                       const void* bb, const void* out_off, void* out_len,
+                       void* out_bytes, int mode, int64_t n,
+                       hipStream_t stream) {
+  int64_t waves_needed = n;
+  hipLaunchKernelGGL(k_str_concat2, flat_grid(waves_needed),
+                     dim3(HIPDF_BLOCK), 0, stream, (const int32_t*)ao,
+                     (const uint8_t*)ab, (const int32_t*)bo,
+                     (const uint8_t*)bb, (const int64_t*)out_off,
+                     (int64_t*)out_len, (uint8_t*)out_bytes, mode, n);
 }
 
 void hipdf_substr_ranges(const void* ao, const void* ab, int start, int slen,

@@ -124,6 +124,27 @@ class Expression:
         """Java-regex find() semantics (RLike)."""
         return StringPredicate("rlike", self, pattern)
 
+    def trim(self) -> "UnaryExpr":
+        return UnaryExpr("trim", self)
+
+    def ltrim(self) -> "UnaryExpr":
+        return UnaryExpr("ltrim", self)
+
+    def rtrim(self) -> "UnaryExpr":
+        return UnaryExpr("rtrim", self)
+
+    def concat(self, other) -> "BinaryExpr":
+        """Spark concat(): NULL if either side is NULL."""
+        return BinaryExpr("concat", self, _as_expr(other))
+
+    def replace(self, search: str, replacement: str) -> "RegexpReplace":
+        """Literal substring replace (StringReplace analogue), lowered to
+        the regex engine with an escaped pattern."""
+        esc = "".join("\\" + c if c in ".\\+*?()[]{}|^$" else c
+                      for c in search)
+        resc = replacement.replace("\\", "\\\\").replace("$", "\\$")
+        return RegexpReplace(self, esc, resc)
+
     def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
         return RegexpExtract(self, pattern, group)
 
@@ -292,11 +313,11 @@ class BinaryExpr(Expression):
         out = self.dtype(schema)
         # scalar fast path: literal on either side
         if isinstance(self.right, Literal) and self.right.value is not None \
-                and not common.is_decimal:
+                and not common.is_decimal and self.op != "concat":
             lcol = ops.cast(self.left.eval(batch, schema), common)
             return ops.binary_op_scalar(self.op, lcol, _coerce_py(self.right.value, common), out)
         if isinstance(self.left, Literal) and self.left.value is not None \
-                and not common.is_decimal:
+                and not common.is_decimal and self.op != "concat":
             swapped = self.op if self.op in self._COMMUTATIVE \
                 else self._SWAP_CMP.get(self.op)
             if swapped is not None:
@@ -327,6 +348,9 @@ def _coerce_py(v, dtype: DType):
 
 _UNARY_OUT = {
     "not": lambda t: BOOL,
+    "trim": lambda t: STRING,
+    "ltrim": lambda t: STRING,
+    "rtrim": lambda t: STRING,
     "is_nan": lambda t: BOOL,
     "neg": lambda t: t,
     "abs": lambda t: t,

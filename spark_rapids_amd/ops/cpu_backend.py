@@ -185,6 +185,10 @@ def _binary_impl(op, a, av, b, bv, in_dtype: DType, out_dtype: DType) -> Column:
         return _bool_col(res.astype(np.uint8), None)
 
     valid = av & bv
+    if op == "concat":
+        res = np.array([(x or "") + (y or "") for x, y in zip(a, b)],
+                       dtype=object)
+        return _make(res, valid if not valid.all() else None, out_dtype)
     if op in _CMP_OPS:
         if in_dtype.id is TypeId.STRING:
             # elementwise python compare on object arrays
@@ -288,6 +292,12 @@ def _str_cmp(op, x, y):
 # ---------------------------------------------------------------------------
 
 def unary_op(op: str, col: Column, out_dtype: DType) -> Column:
+    if op in ("trim", "ltrim", "rtrim"):
+        fn = {"trim": str.strip, "ltrim": str.lstrip,
+              "rtrim": str.rstrip}[op]
+        out = [None if v is None else fn(v, " ")
+               for v in col.to_pylist()]
+        return Column.from_pylist(out, DType.string())
     a, av = _vals(col), _valid(col)
     np_out = out_dtype.numpy_dtype() if out_dtype.id is not TypeId.STRING else None
     valid = av
@@ -516,7 +526,9 @@ def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
             i += 2
         elif c == "\\" and i + 1 < len(replacement):
             ch = replacement[i + 1]
-            pyrepl += _re.escape(ch) if ch in "\\$" else "\\" + ch
+            # python templates: $ is literal; backslash must double
+            pyrepl += ch if ch == "$" else (
+                "\\\\" if ch == "\\" else "\\" + ch)
             i += 2
         else:
             pyrepl += c.replace("\\", "\\\\")

@@ -156,6 +156,8 @@ def _binary(op, lhs: Column, rhs: Optional[Column], scalar, out_dtype) -> Column
     s = _stream()
     scalar_rhs = rhs is None
     if lhs.dtype.id is TypeId.STRING:
+        if op == "concat" and not scalar_rhs:
+            return str_concat(lhs, rhs)
         if op not in _STR_CMP:
             raise NotImplementedError(f"string op {op} not on GPU")
         out = _alloc(n, out_dtype)
@@ -422,6 +424,46 @@ def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
                   null_count=col._null_count)
 
 
+def str_trim(col: Column, mode: str) -> Column:
+    """trim/ltrim/rtrim (ascii space, Spark default)."""
+    n = col.size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.string())
+    m = {"both": 0, "leading": 1, "trailing": 2}[mode]
+    bstart = torch.empty(n, dtype=torch.int32, device="cuda")
+    blen = torch.empty(n, dtype=torch.int64, device="cuda")
+    ext.str_trim_ranges(m, col.offsets.data_ptr(), col.data.data_ptr(),
+                        bstart.data_ptr(), blen.data_ptr(), n, s)
+    return _strings_from_spans(col, bstart, blen, n, s)
+
+
+def str_concat(a: Column, b: Column) -> Column:
+    """concat(a, b): NULL if either side is NULL (Spark concat)."""
+    n = a.size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.string())
+    lens = torch.empty(n, dtype=torch.int64, device="cuda")
+    ext.str_concat2(a.offsets.data_ptr(), a.data.data_ptr(),
+                    b.offsets.data_ptr(), b.data.data_ptr(), 0,
+                    lens.data_ptr(), 0, 0, n, s)
+    scanned, total = _exclusive_scan_i64(lens)
+    out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
+                            device="cuda")[:total]
+    if total:
+        ext.str_concat2(a.offsets.data_ptr(), a.data.data_ptr(),
+                        b.offsets.data_ptr(), b.data.data_ptr(),
+                        scanned.data_ptr(), lens.data_ptr(),
+                        out_bytes.data_ptr(), 1, n, s)
+    offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+    offs[n] = total
+    v = _and_masks(a.validity, b.validity)
+    return Column(DType.string(), n, out_bytes, v, offs,
+                  null_count=None if v is not None else 0)
+
+
 def substring(col: Column, pos: int, length: int = -1) -> Column:
     n = col.size
     s = _stream()
@@ -460,6 +502,9 @@ def unary_op(op: str, col: Column, out_dtype: DType) -> Column:
             ext.str_length(col.offsets.data_ptr(), col.data.data_ptr(),
                            out.data_ptr(), n, s)
             return Column(out_dtype, n, out, v, null_count=col._null_count)
+        if op in ("trim", "ltrim", "rtrim"):
+            return str_trim(col, {"trim": "both", "ltrim": "leading",
+                                  "rtrim": "trailing"}[op])
         if op in ("upper", "lower"):
             nb = int(col.data.numel())
             ob = torch.empty(max(nb, 1), dtype=torch.uint8,

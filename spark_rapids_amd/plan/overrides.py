@@ -48,15 +48,16 @@ _GPU_BINARY_OPS = {
     "add", "sub", "mul", "div", "int_div", "mod", "pmod", "pow",
     "eq", "ne", "lt", "le", "gt", "ge", "eq_null_safe",
     "and", "or", "bitand", "bitor", "bitxor", "shiftleft", "shiftright",
-    "min", "max",
+    "min", "max", "concat",
 }
 _GPU_UNARY_OPS = {
     "neg", "abs", "not", "sqrt", "exp", "log", "floor", "ceil",
     "sin", "cos", "tan", "is_nan", "year", "month", "day",
+    "trim", "ltrim", "rtrim",
 }
 # string ops with GPU kernels (strings.hip); eq_null_safe still CPU-only
-_GPU_STRING_OK = {"eq", "ne", "lt", "le", "gt", "ge"}
-_GPU_STRING_UNARY = {"length", "upper", "lower"}
+_GPU_STRING_OK = {"eq", "ne", "lt", "le", "gt", "ge", "concat"}
+_GPU_STRING_UNARY = {"length", "upper", "lower", "trim", "ltrim", "rtrim"}
 
 
 class TagReason:

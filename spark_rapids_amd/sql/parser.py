@@ -462,8 +462,16 @@ class Parser:
             scale = args[1].value if len(args) > 1 else 0
             return Round(args[0], int(scale))
         if name in ("abs", "sqrt", "exp", "log", "floor", "ceil", "upper",
-                    "lower", "length", "year", "month", "day"):
+                    "lower", "length", "year", "month", "day", "trim",
+                    "ltrim", "rtrim"):
             return UnaryExpr(name, args[0])
+        if name == "concat":
+            out = args[0]
+            for nxt in args[1:]:
+                out = BinaryExpr("concat", out, nxt)
+            return out
+        if name == "replace":
+            return args[0].replace(args[1].value, args[2].value)
         if name == "substring" or name == "substr":
             pos = int(args[1].value)
             ln = int(args[2].value) if len(args) > 2 else -1

@@ -121,3 +121,52 @@ def test_gpu_string_filter_e2e():
         "v": list(range(n)),
     })
     assert gpu == df2.filter(col("s").like("sp%")).count()
+
+
+class TestTrimConcatReplace:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_cpu(self, cpu):
+        df = cpu.create_dataframe({"a": ["  hi ", "x", None, ""],
+                                   "b": ["A", None, "B", ""]})
+        out = df.select(
+            col("a").trim().alias("t"), col("a").ltrim().alias("l"),
+            col("a").rtrim().alias("r"),
+            col("a").concat(col("b")).alias("c"),
+            col("a").replace("hi", "yo$").alias("rep")).to_pydict()
+        assert out["t"] == ["hi", "x", None, ""]
+        assert out["l"] == ["hi ", "x", None, ""]
+        assert out["r"] == ["  hi", "x", None, ""]
+        assert out["c"] == ["  hi A", None, None, ""]
+        assert out["rep"] == ["  yo$ ", "x", None, ""]
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        import numpy as np
+
+        rng = np.random.default_rng(2)
+        base = ["  pad %d " % v if v % 3 else "x%d.y" % v
+                for v in rng.integers(0, 1000, 5000)]
+        vals = [v if i % 17 else None for i, v in enumerate(base)]
+
+        def q(s):
+            df = s.create_dataframe({"a": vals, "b": base})
+            return df.select(
+                col("a").trim().alias("t"),
+                col("a").concat(col("b")).alias("c"),
+                col("a").concat("#").alias("cl"),
+                col("a").replace(".", "_").alias("rep")).to_pydict()
+
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        assert q(sg) == q(sc)
+
+    @pytest.mark.gpu
+    def test_gpu_placement(self):
+        sg = sr.Session()
+        df = sg.create_dataframe({"a": ["q"]})
+        tree = (df.select(col("a").trim().concat("z"))
+                .physical_plan().tree_string())
+        assert "GpuProject" in tree, tree
