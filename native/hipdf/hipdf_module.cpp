@@ -117,6 +117,8 @@ void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
                              hipStream_t);
+void hipdf_str_plain_encode(const void*, const void*, const void*, void*,
+                            void*, int, int64_t, hipStream_t);
 void hipdf_str_plain_offsets(const void*, int64_t, int64_t, void*, void*,
                              void*, hipStream_t);
 void hipdf_scatter_fixed(int, const void*, const void*, void*, int64_t,
@@ -433,6 +435,13 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("scan_add_offsets_f64", [](int64_t out, int64_t sums, int64_t n,
                                    int64_t stream) {
     hipdf_scan_add_offsets_f64(PM(out), P(sums), n, S(stream));
+    check_async();
+  });
+  m.def("str_plain_encode", [](int64_t offsets, int64_t bytes,
+                               int64_t out_off, int64_t out_len, int64_t out,
+                               int mode, int64_t n, int64_t stream) {
+    hipdf_str_plain_encode(P(offsets), P(bytes), P(out_off), PM(out_len),
+                           PM(out), mode, n, S(stream));
     check_async();
   });
   m.def("str_plain_offsets", [](int64_t data, int64_t nbytes,

@@ -146,7 +146,45 @@ __global__ void k_str_plain_offsets(const uint8_t* __restrict__ data,
   }
 }
 
+// inverse of the PLAIN byte-array decode: emit [u32 len][bytes] per row
+// (GPU parquet writer). mode 0: per-row output sizes; mode 1: write.
+__global__ void k_str_plain_encode(const int32_t* __restrict__ offsets,
+                                   const uint8_t* __restrict__ bytes,
+                                   const int64_t* __restrict__ out_off,
+                                   int64_t* __restrict__ out_len,
+                                   uint8_t* __restrict__ out, int mode,
+                                   int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / 64;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / 64;
+  int lane = (int)(threadIdx.x & 63);
+  for (int64_t i = wave_global; i < n; i += wave_count) {
+    int32_t a = offsets[i], b = offsets[i + 1];
+    int32_t len = b - a;
+    if (!mode) {
+      if (lane == 0) out_len[i] = 4 + (int64_t)len;
+      continue;
+    }
+    uint8_t* dst = out + out_off[i];
+    if (lane == 0) {
+      dst[0] = (uint8_t)len;
+      dst[1] = (uint8_t)(len >> 8);
+      dst[2] = (uint8_t)(len >> 16);
+      dst[3] = (uint8_t)(len >> 24);
+    }
+    for (int32_t k = lane; k < len; k += 64) dst[4 + k] = bytes[a + k];
+  }
+}
+
 extern "C" {
+
+void hipdf_str_plain_encode(const void* offsets, const void* bytes,
+                            const void* out_off, void* out_len, void* out,
+                            int mode, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_plain_encode, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)offsets, (const uint8_t*)bytes,
+                     (const int64_t*)out_off, (int64_t*)out_len,
+                     (uint8_t*)out, mode, n);
+}
 
 void hipdf_str_plain_offsets(const void* data, int64_t nbytes,
                              int64_t n_values, void* starts, void* lens,

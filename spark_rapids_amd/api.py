@@ -402,7 +402,31 @@ class Session:
         return DataFrame(self, L.Scan(src, src.schema, f"parquet:{path}"))
 
     def write_parquet(self, df: DataFrame, path: str,
-                      compression: str = "snappy"):
+                      compression: str = "snappy",
+                      gpu_encode: bool = False):
+        """gpu_encode=True keeps the result on the GPU and writes PLAIN
+        pages encoded by device kernels (io/parquet_write.py); the default
+        stages through arrow on the host."""
+        if gpu_encode:
+            from . import ops as _ops
+            from .io.parquet_write import write_parquet_gpu
+
+            exec_ = df.physical_plan()
+            sem = GpuSemaphore.get()
+            with sem.held():
+                # strip the trailing host transfer: encode from device
+                root = exec_
+                while type(root).__name__ == "DeviceTransferExec":
+                    root = root.children[0]
+                batches = list(root.execute())
+                if not batches:
+                    batches = [ColumnBatch(
+                        [Column.from_pylist([], f.dtype).cuda()
+                         for f in df.schema.fields], 0)]
+                batch = batches[0] if len(batches) == 1 \
+                    else _ops.concat_batches(batches)
+                write_parquet_gpu(batch, df.schema, path)
+            return
         from .io.parquet import write_parquet
 
         batch = df.collect_batch()

@@ -221,3 +221,99 @@ def parse_page_header(buf: bytes, pos: int = 0) -> PageHeader:
             r.skip(ctype)
     ph.header_size = r.pos - start
     return ph
+
+
+# ---------------------------------------------------------------------------
+# compact-protocol WRITER (parquet file authoring: PageHeader + footer
+# FileMetaData). Symmetric subset of the parser above.
+# ---------------------------------------------------------------------------
+
+def _w_varint(out: bytearray, v: int):
+    while True:
+        b = v & 0x7F
+        v >>= 7
+        if v:
+            out.append(b | 0x80)
+        else:
+            out.append(b)
+            return
+
+
+def _w_zigzag(out: bytearray, v: int):
+    _w_varint(out, (v << 1) ^ (v >> 63) if v < 0 else v << 1)
+
+
+class StructWriter:
+    """Compact-protocol struct writer: call the typed field methods in
+    ascending field-id order, then bytes()."""
+
+    def __init__(self):
+        self.buf = bytearray()
+        self.last_fid = 0
+
+    def _hdr(self, fid: int, ctype: int):
+        delta = fid - self.last_fid
+        if 0 < delta < 16:
+            self.buf.append((delta << 4) | ctype)
+        else:
+            self.buf.append(ctype)
+            _w_zigzag(self.buf, fid)
+        self.last_fid = fid
+
+    def f_i32(self, fid: int, v: int):
+        self._hdr(fid, CT_I32)
+        _w_zigzag(self.buf, v)
+
+    def f_i64(self, fid: int, v: int):
+        self._hdr(fid, CT_I64)
+        _w_zigzag(self.buf, v)
+
+    def f_bool(self, fid: int, v: bool):
+        self._hdr(fid, CT_TRUE if v else CT_FALSE)
+
+    def f_binary(self, fid: int, b: bytes):
+        self._hdr(fid, CT_BINARY)
+        _w_varint(self.buf, len(b))
+        self.buf.extend(b)
+
+    def f_struct(self, fid: int, body: bytes):
+        self._hdr(fid, CT_STRUCT)
+        self.buf.extend(body)
+
+    def f_list_struct(self, fid: int, items):
+        self._hdr(fid, CT_LIST)
+        n = len(items)
+        if n < 15:
+            self.buf.append((n << 4) | CT_STRUCT)
+        else:
+            self.buf.append(0xF0 | CT_STRUCT)
+            _w_varint(self.buf, n)
+        for it in items:
+            self.buf.extend(it)
+
+    def f_list_binary(self, fid: int, items):
+        self._hdr(fid, CT_LIST)
+        n = len(items)
+        if n < 15:
+            self.buf.append((n << 4) | CT_BINARY)
+        else:
+            self.buf.append(0xF0 | CT_BINARY)
+            _w_varint(self.buf, n)
+        for b in items:
+            _w_varint(self.buf, len(b))
+            self.buf.extend(b)
+
+    def f_list_i32(self, fid: int, items):
+        self._hdr(fid, CT_LIST)
+        n = len(items)
+        if n < 15:
+            self.buf.append((n << 4) | CT_I32)
+        else:
+            self.buf.append(0xF0 | CT_I32)
+            _w_varint(self.buf, n)
+        for v in items:
+            _w_zigzag(self.buf, v)
+
+    def bytes(self) -> bytes:
+        self.buf.append(CT_STOP)
+        return bytes(self.buf)
