@@ -19,7 +19,7 @@ from .expr.expressions import ColumnRef, Expression, col as _col
 from .memory.semaphore import GpuSemaphore
 from .plan import logical as L
 from .plan.overrides import plan_physical
-from .types import DType, INT32
+from .types import DType, INT32, TypeId
 
 
 class MemTable:
@@ -476,10 +476,19 @@ class Session:
 def _infer_list_dtype(v: list) -> DType:
     from .expr.expressions import _infer_literal_dtype
 
+    dt = None
     for x in v:
-        if x is not None:
-            return _infer_literal_dtype(x)
-    return DType.int32()
+        if x is None:
+            continue
+        if dt is None:
+            dt = _infer_literal_dtype(x)
+            if dt.id is not TypeId.INT32:
+                return dt
+        # ints: widen to INT64 if ANY value needs it (not just the first)
+        elif not isinstance(x, bool) and isinstance(x, int) \
+                and not -(2 ** 31) <= x < 2 ** 31:
+            return DType.int64()
+    return dt or DType.int32()
 
 
 def _split_partitions(batch: ColumnBatch, n: int) -> List[ColumnBatch]:
