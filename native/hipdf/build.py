@@ -29,6 +29,7 @@ KERNELS = [
     "kernels/partition.hip",
     "kernels/sort.hip",
     "kernels/decode.hip",
+    "kernels/csv.hip",
     "kernels/strings.hip",
     "kernels/window.hip",
     "kernels/decimal128.hip",

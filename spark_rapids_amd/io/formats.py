@@ -35,19 +35,38 @@ class CsvTable:
         import pyarrow.csv as pacsv
 
         self.files = _expand(path)
+        self.header = header
+        self.delimiter = delimiter
         self.read_opts = pacsv.ReadOptions(autogenerate_column_names=not header)
         self.parse_opts = pacsv.ParseOptions(delimiter=delimiter)
+        # Spark CSV semantics: empty field = null (also what the GPU
+        # decoder produces)
+        self.convert_opts = pacsv.ConvertOptions(strings_can_be_null=True)
         tbl = pacsv.read_csv(self.files[0], read_options=self.read_opts,
-                             parse_options=self.parse_opts)
+                             parse_options=self.parse_opts,
+                             convert_options=self.convert_opts)
         self.schema = Schema([Field(f.name, arrow_to_dtype(f.type), True)
                               for f in tbl.schema])
 
     def partitions(self) -> Iterable[ColumnBatch]:
         import pyarrow.csv as pacsv
 
+        import torch
+
         for f in self.files:
+            if torch.cuda.is_available():
+                try:
+                    from .csv_gpu import read_csv_gpu
+
+                    yield read_csv_gpu(f, self.schema,
+                                       header=self.header,
+                                       delimiter=self.delimiter)
+                    continue
+                except NotImplementedError:
+                    pass  # per-file CPU fallback (quotes / exotic types)
             tbl = pacsv.read_csv(f, read_options=self.read_opts,
-                                 parse_options=self.parse_opts)
+                                 parse_options=self.parse_opts,
+                                 convert_options=self.convert_opts)
             yield arrow_table_to_batch(tbl)
 
 

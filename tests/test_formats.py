@@ -29,3 +29,50 @@ def test_json_lines(tmp_path, session):
     back = session.read_json(str(p))
     assert back.to_pydict()["a"] == [1, 2, 3]
     assert back.to_pydict()["s"] == ["x", None, "z"]
+
+
+@pytest.mark.gpu
+def test_gpu_csv_matches_arrow(tmp_path):
+    import numpy as np
+
+    rng = np.random.default_rng(4)
+    n = 30_000
+    lines = ["i,f,s"]
+    for k in range(n):
+        i = "" if k % 17 == 0 else str(int(rng.integers(-10**12, 10**12)))
+        f = "" if k % 23 == 0 else repr(float(rng.uniform(-1e6, 1e6)))
+        sv = "" if k % 13 == 0 else f"name_{k}"
+        lines.append(f"{i},{f},{sv}")
+    p = str(tmp_path / "t.csv")
+    with open(p, "w") as fh:
+        fh.write("\n".join(lines) + "\n")
+    sg = sr.Session()
+    df = sg.read_csv(p)
+    got = df.to_pydict()
+    import pyarrow.csv as pacsv
+
+    exp = pacsv.read_csv(
+        p, convert_options=pacsv.ConvertOptions(strings_can_be_null=True))
+    assert got["i"] == exp.column("i").to_pylist()
+    ge, ee = got["f"], exp.column("f").to_pylist()
+    for a, b in zip(ge, ee):
+        assert (a is None) == (b is None)
+        if a is not None:
+            assert a == pytest.approx(b, rel=1e-14, abs=1e-300)
+    assert got["s"] == exp.column("s").to_pylist()
+    # a GPU query over the csv scan
+    out = df.filter(col("i") > 0).agg(count_star()).collect()
+    cpuv = sum(1 for v in exp.column("i").to_pylist()
+               if v is not None and v > 0)
+    assert out[0][0] == cpuv
+
+
+@pytest.mark.gpu
+def test_gpu_csv_quoted_falls_back(tmp_path):
+    p = str(tmp_path / "q.csv")
+    with open(p, "w") as fh:
+        fh.write('a,b\n1,"x,y"\n2,plain\n')
+    sg = sr.Session()
+    out = sg.read_csv(p).to_pydict()
+    assert out["a"] == [1, 2]
+    assert out["b"] == ["x,y", "plain"]
