@@ -1,6 +1,7 @@
 """CSV / ORC / JSON scan + write tests (host parse, columnar engine on top)."""
 import pytest
 
+import spark_rapids_amd as sr
 from spark_rapids_amd import col, count_star, sum_
 
 
