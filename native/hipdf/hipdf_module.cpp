@@ -117,6 +117,9 @@ void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
                              hipStream_t);
+void hipdf_json_field(const void*, const void*, const void*, const void*,
+                      int, int, void*, void*, void*, void*, void*, void*,
+                      int64_t, hipStream_t);
 void hipdf_byte_eq(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_csv_parse(const void*, const void*, const void*, int, int, int,
                      void*, void*, void*, void*, void*, void*, int64_t,
@@ -439,6 +442,16 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("scan_add_offsets_f64", [](int64_t out, int64_t sums, int64_t n,
                                    int64_t stream) {
     hipdf_scan_add_offsets_f64(PM(out), P(sums), n, S(stream));
+    check_async();
+  });
+  m.def("json_field", [](int64_t bytes, int64_t row_start, int64_t row_end,
+                         int64_t name, int name_len, int type,
+                         int64_t out_i64, int64_t out_f64, int64_t out_ss,
+                         int64_t out_sl, int64_t valid, int64_t unsupported,
+                         int64_t nrows, int64_t stream) {
+    hipdf_json_field(P(bytes), P(row_start), P(row_end), P(name), name_len,
+                     type, PM(out_i64), PM(out_f64), PM(out_ss), PM(out_sl),
+                     PM(valid), PM(unsupported), nrows, S(stream));
     check_async();
   });
   m.def("byte_eq", [](int64_t bytes, int target, int64_t out, int64_t n,

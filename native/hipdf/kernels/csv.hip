@@ -138,7 +138,159 @@ __global__ void k_csv_parse(const uint8_t* __restrict__ bytes,
   }
 }
 
+// JSON-lines field extraction: find `"name":` at object level (previous
+// non-space char is '{' or ',') inside the row, parse the value.
+// type: 0 int64, 1 float64, 2 string span, 3 bool
+__global__ void k_json_field(const uint8_t* __restrict__ bytes,
+                             const int32_t* __restrict__ row_start,
+                             const int32_t* __restrict__ row_end,
+                             const uint8_t* __restrict__ name, int name_len,
+                             int type, int64_t* __restrict__ out_i64,
+                             double* __restrict__ out_f64,
+                             int32_t* __restrict__ out_ss,
+                             int64_t* __restrict__ out_sl,
+                             uint8_t* __restrict__ valid,
+                             int* __restrict__ unsupported, int64_t nrows) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x;
+       i < nrows; i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t rs = row_start[i], re = row_end[i];
+    if (re > rs && bytes[re - 1] == '\r') --re;
+    int32_t vpos = -1;
+    for (int32_t p = rs; p + name_len + 2 < re; ++p) {
+      if (bytes[p] != '"') continue;
+      bool m = true;
+      for (int k = 0; k < name_len; ++k)
+        if (bytes[p + 1 + k] != name[k]) { m = false; break; }
+      if (!m || bytes[p + 1 + name_len] != '"') continue;
+      // previous non-space must open an object member
+      int32_t q = p - 1;
+      while (q >= rs && (bytes[q] == ' ' || bytes[q] == '\t')) --q;
+      if (q < rs || (bytes[q] != '{' && bytes[q] != ',')) continue;
+      q = p + 2 + name_len;
+      while (q < re && (bytes[q] == ' ' || bytes[q] == '\t')) ++q;
+      if (q < re && bytes[q] == ':') {
+        ++q;
+        while (q < re && (bytes[q] == ' ' || bytes[q] == '\t')) ++q;
+        vpos = q;
+        break;
+      }
+    }
+    bool ok = false;
+    if (type == 2) {
+      out_ss[i] = 0;
+      out_sl[i] = 0;
+    } else if (type == 0) {
+      out_i64[i] = 0;
+    } else if (type == 1) {
+      out_f64[i] = 0.0;
+    } else {
+      out_i64[i] = 0;
+    }
+    if (vpos >= 0 && vpos < re && !(bytes[vpos] == 'n')) {  // null -> null
+      uint8_t c = bytes[vpos];
+      if (type == 2) {
+        if (c == '"') {
+          int32_t e = vpos + 1;
+          bool esc = false;
+          while (e < re && bytes[e] != '"') {
+            if (bytes[e] == '\\') { esc = true; break; }
+            ++e;
+          }
+          if (esc) {
+            atomicAdd(unsupported, 1);
+          } else if (e < re) {
+            out_ss[i] = vpos + 1;
+            out_sl[i] = e - vpos - 1;
+            ok = true;
+          }
+        }
+      } else if (type == 3) {
+        if (c == 't') { out_i64[i] = 1; ok = true; }
+        else if (c == 'f') { out_i64[i] = 0; ok = true; }
+      } else {
+        // numeric: find the value end (",", "}", space)
+        int32_t e = vpos;
+        while (e < re && bytes[e] != ',' && bytes[e] != '}' &&
+               bytes[e] != ' ')
+          ++e;
+        int32_t p = vpos;
+        bool neg = false;
+        if (p < e && (bytes[p] == '-' || bytes[p] == '+')) {
+          neg = bytes[p] == '-';
+          ++p;
+        }
+        if (type == 0) {
+          int64_t v = 0;
+          bool any = false, bad = false;
+          for (; p < e; ++p) {
+            uint8_t d = bytes[p];
+            if (d < '0' || d > '9') { bad = true; break; }
+            v = v * 10 + (d - '0');
+            any = true;
+          }
+          if (any && !bad) { out_i64[i] = neg ? -v : v; ok = true; }
+        } else {
+          double mant = 0.0;
+          int exp10 = 0;
+          bool any = false, bad = false, dot = false;
+          for (; p < e; ++p) {
+            uint8_t d = bytes[p];
+            if (d >= '0' && d <= '9') {
+              mant = mant * 10.0 + (d - '0');
+              if (dot) --exp10;
+              any = true;
+            } else if (d == '.' && !dot) {
+              dot = true;
+            } else if ((d == 'e' || d == 'E') && any) {
+              ++p;
+              bool en = false;
+              if (p < e && (bytes[p] == '-' || bytes[p] == '+')) {
+                en = bytes[p] == '-';
+                ++p;
+              }
+              int ev = 0;
+              for (; p < e; ++p) {
+                if (bytes[p] < '0' || bytes[p] > '9') { bad = true; break; }
+                ev = ev * 10 + (bytes[p] - '0');
+              }
+              exp10 += en ? -ev : ev;
+              break;
+            } else {
+              bad = true;
+              break;
+            }
+          }
+          if (any && !bad) {
+            double v = mant;
+            if (exp10 > 0)
+              for (int k = 0; k < exp10; ++k) v *= 10.0;
+            else
+              for (int k = 0; k < -exp10; ++k) v /= 10.0;
+            out_f64[i] = neg ? -v : v;
+            ok = true;
+          }
+        }
+      }
+    }
+    valid[i] = ok;
+  }
+}
+
 extern "C" {
+
+void hipdf_json_field(const void* bytes, const void* row_start,
+                      const void* row_end, const void* name, int name_len,
+                      int type, void* out_i64, void* out_f64, void* out_ss,
+                      void* out_sl, void* valid, void* unsupported,
+                      int64_t nrows, hipStream_t stream) {
+  hipLaunchKernelGGL(k_json_field, flat_grid(nrows), dim3(HIPDF_BLOCK), 0,
+                     stream, (const uint8_t*)bytes,
+                     (const int32_t*)row_start, (const int32_t*)row_end,
+                     (const uint8_t*)name, name_len, type,
+                     (int64_t*)out_i64, (double*)out_f64, (int32_t*)out_ss,
+                     (int64_t*)out_sl, (uint8_t*)valid, (int*)unsupported,
+                     nrows);
+}
 
 void hipdf_byte_eq(const void* bytes, int target, void* out, int64_t n,
                    hipStream_t stream) {

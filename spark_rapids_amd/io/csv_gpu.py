@@ -114,3 +114,88 @@ def _pack_valid(valid_u8: torch.Tensor, n: int, s) -> torch.Tensor:
     mask = torch.empty(mask_nbytes(n), dtype=torch.uint8, device="cuda")
     ext.mask_from_nonzero(v64.data_ptr(), mask.data_ptr(), n, s)
     return mask
+
+
+def read_json_gpu(path: str, schema: Schema) -> ColumnBatch:
+    """JSON-lines decode with k_json_field (flat objects; string values
+    with escapes or non-scalar columns fall back per file)."""
+    from ..ops import gpu_backend as gb
+    from ..ops.gpu_backend import ext
+
+    for f in schema.fields:
+        if f.dtype.id not in _INT_IDS and not f.dtype.is_floating \
+                and f.dtype.id not in (TypeId.STRING, TypeId.BOOL):
+            raise NotImplementedError(f"gpu json: column type {f.dtype}")
+    with open(path, "rb") as fh:
+        raw = fh.read()
+    if not raw.strip():
+        return ColumnBatch([Column.from_pylist([], f.dtype).cuda()
+                            for f in schema.fields], 0)
+    s = gb._stream()
+    data = torch.frombuffer(bytearray(raw), dtype=torch.uint8).cuda()
+    nb = data.numel()
+    nl = torch.empty(nb, dtype=torch.uint8, device="cuda")
+    ext.byte_eq(data.data_ptr(), ord("\n"), nl.data_ptr(), nb, s)
+    pos = gb.mask_to_sel(Column(DType.bool_(), nb, nl, None, null_count=0),
+                         nb)
+    npos = pos.numel()
+    trailing = not raw.endswith(b"\n")
+    starts = torch.empty(npos + 1, dtype=torch.int32, device="cuda")
+    starts[0] = 0
+    if npos:
+        starts[1:] = pos + 1
+    ends = torch.empty(npos + (1 if trailing else 0), dtype=torch.int32,
+                       device="cuda")
+    if npos:
+        ends[:npos] = pos
+    if trailing:
+        ends[npos] = nb
+    n = ends.numel()
+    row_start, row_end = starts[:n], ends[:n]
+    unsupported = torch.zeros(1, dtype=torch.int32, device="cuda")
+    cols = []
+    for f in schema.fields:
+        nameb = f.name.encode("utf-8")
+        name_t = torch.frombuffer(bytearray(nameb),
+                                  dtype=torch.uint8).cuda()
+        valid_u8 = torch.empty(n, dtype=torch.uint8, device="cuda")
+        if f.dtype.id is TypeId.STRING:
+            ss = torch.empty(n, dtype=torch.int32, device="cuda")
+            sl = torch.empty(n, dtype=torch.int64, device="cuda")
+            ext.json_field(data.data_ptr(), row_start.data_ptr(),
+                           row_end.data_ptr(), name_t.data_ptr(),
+                           len(nameb), 2, 0, 0, ss.data_ptr(),
+                           sl.data_ptr(), valid_u8.data_ptr(),
+                           unsupported.data_ptr(), n, s)
+            scanned, total = gb._exclusive_scan_i64(sl)
+            out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
+                                    device="cuda")[:total]
+            if total:
+                ext.substr_copy(data.data_ptr(), ss.data_ptr(),
+                                sl.data_ptr(), scanned.data_ptr(),
+                                out_bytes.data_ptr(), n, s)
+            offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+            ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+            offs[n] = total
+            cols.append(Column(DType.string(), n, out_bytes,
+                               _pack_valid(valid_u8, n, s), offs,
+                               null_count=None))
+            continue
+        is_f = f.dtype.is_floating
+        is_b = f.dtype.id is TypeId.BOOL
+        out = torch.empty(n, dtype=torch.float64 if is_f else torch.int64,
+                          device="cuda")
+        t = 1 if is_f else (3 if is_b else 0)
+        ext.json_field(data.data_ptr(), row_start.data_ptr(),
+                       row_end.data_ptr(), name_t.data_ptr(), len(nameb),
+                       t, 0 if is_f else out.data_ptr(),
+                       out.data_ptr() if is_f else 0, 0, 0,
+                       valid_u8.data_ptr(), unsupported.data_ptr(), n, s)
+        mask = _pack_valid(valid_u8, n, s)
+        wide = Column(DType.float64() if is_f else DType.int64(), n, out,
+                      mask, null_count=None)
+        cols.append(gb.cast(wide, f.dtype) if wide.dtype != f.dtype
+                    else wide)
+    if int(unsupported.item()) > 0:
+        raise NotImplementedError("gpu json: escaped strings present")
+    return ColumnBatch(cols, n)

@@ -100,7 +100,17 @@ class JsonTable:
     def partitions(self) -> Iterable[ColumnBatch]:
         import pyarrow.json as pajson
 
+        import torch
+
         for f in self.files:
+            if torch.cuda.is_available():
+                try:
+                    from .csv_gpu import read_json_gpu
+
+                    yield read_json_gpu(f, self.schema)
+                    continue
+                except NotImplementedError:
+                    pass
             yield arrow_table_to_batch(pajson.read_json(f))
 
 

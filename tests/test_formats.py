@@ -77,3 +77,34 @@ def test_gpu_csv_quoted_falls_back(tmp_path):
     out = sg.read_csv(p).to_pydict()
     assert out["a"] == [1, 2]
     assert out["b"] == ["x,y", "plain"]
+
+
+@pytest.mark.gpu
+def test_gpu_json_matches_arrow(tmp_path):
+    import json as pyjson
+
+    import numpy as np
+
+    rng = np.random.default_rng(9)
+    n = 10_000
+    rows = []
+    for k in range(n):
+        r = {"i": int(rng.integers(-10**9, 10**9)),
+             "f": float(round(rng.uniform(-100, 100), 6)),
+             "s": f"v{k}", "b": bool(k % 3)}
+        if k % 11 == 0:
+            r["i"] = None
+        if k % 13 == 0:
+            del r["s"]
+        rows.append(r)
+    p = str(tmp_path / "t.json")
+    with open(p, "w") as fh:
+        for r in rows:
+            fh.write(pyjson.dumps(r) + "\n")
+    sg = sr.Session()
+    got = sg.read_json(p).to_pydict()
+    assert got["i"] == [r["i"] for r in rows]
+    assert got["s"] == [r.get("s") for r in rows]
+    assert got["b"] == [r["b"] for r in rows]
+    for a, r in zip(got["f"], rows):
+        assert a == pytest.approx(r["f"], rel=1e-12)
