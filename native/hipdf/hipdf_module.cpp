@@ -117,6 +117,9 @@ void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
                              hipStream_t);
+void hipdf_orc_bool_rle(const void*, int64_t, int64_t, void*, hipStream_t);
+void hipdf_orc_rle_v2(const void*, int64_t, int64_t, int, void*,
+                      hipStream_t);
 void hipdf_json_field(const void*, const void*, const void*, const void*,
                       int, int, void*, void*, void*, void*, void*, void*,
                       int64_t, hipStream_t);
@@ -442,6 +445,16 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("scan_add_offsets_f64", [](int64_t out, int64_t sums, int64_t n,
                                    int64_t stream) {
     hipdf_scan_add_offsets_f64(PM(out), P(sums), n, S(stream));
+    check_async();
+  });
+  m.def("orc_bool_rle", [](int64_t b, int64_t nbytes, int64_t n,
+                           int64_t out, int64_t stream) {
+    hipdf_orc_bool_rle(P(b), nbytes, n, PM(out), S(stream));
+    check_async();
+  });
+  m.def("orc_rle_v2", [](int64_t b, int64_t nbytes, int64_t n, int is_signed,
+                         int64_t out, int64_t stream) {
+    hipdf_orc_rle_v2(P(b), nbytes, n, is_signed, PM(out), S(stream));
     check_async();
   });
   m.def("json_field", [](int64_t bytes, int64_t row_start, int64_t row_end,

@@ -82,7 +82,18 @@ class OrcTable:
     def partitions(self) -> Iterable[ColumnBatch]:
         import pyarrow.orc as paorc
 
+        import torch
+
         for f in self.files:
+            if torch.cuda.is_available():
+                try:
+                    from .orc_gpu import read_orc_gpu
+
+                    yield read_orc_gpu(f, [fl.name
+                                           for fl in self.schema.fields])
+                    continue
+                except NotImplementedError:
+                    pass  # per-file fallback (dict encodings, exotic types)
             yield arrow_table_to_batch(paorc.ORCFile(f).read())
 
 
