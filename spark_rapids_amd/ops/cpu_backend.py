@@ -892,7 +892,8 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         af = a[av].astype(np.float64)
         if op in ("sum", "mean", "m2"):
             s = np.zeros(ngroups)
-            np.add.at(s, gc, af)
+            with np.errstate(invalid="ignore"):  # inf + -inf -> nan (Spark)
+                np.add.at(s, gc, af)
             if op == "sum":
                 res = s
             elif op == "mean":
