@@ -195,6 +195,28 @@ class Parser:
         # assemble plan
         if where is not None:
             df = df.filter(where)
+        from ..expr.windows import WindowExpr as _WE
+
+        if any(isinstance(e, _WE) for e, _ in items):
+            out_names = []
+            for e, alias in items:
+                name = alias or e.output_name()
+                if isinstance(e, _WE):
+                    df = df.with_column(name, e)
+                out_names.append((e, name))
+            sel = []
+            for e, name in out_names:
+                sel.append(ColumnRef(name) if isinstance(e, _WE)
+                           else (e.alias(name)))
+            df = df.select(*sel)
+            if order:
+                df = df.sort(*[n for n, _ in order],
+                             descending=[d for _, d in order])
+            if limit is not None:
+                df = df.limit(limit)
+            if self.peek()[0] != "end":
+                raise SqlError("trailing tokens")
+            return df
         has_aggs = any(isinstance(e, A.AggExpr) for e, _ in items)
         if group_cols or has_aggs:
             keys = group_cols
@@ -266,7 +288,81 @@ class Parser:
         return n
 
     def parse_select_item(self):
-        return self.parse_expr()
+        e = self.parse_expr()
+        if self.kw("OVER"):
+            e = self.parse_over(e)
+        return e
+
+    def parse_over(self, e):
+        """<agg or ranking func> OVER (PARTITION BY .. ORDER BY ..
+        [ROWS|RANGE BETWEEN .. AND ..])"""
+        from ..expr import windows as W
+
+        if isinstance(e, W.WindowFunc):
+            func = e
+        elif isinstance(e, A.AggExpr):
+            if e.op not in ("sum", "count", "min", "max", "mean"):
+                raise SqlError(f"{e.op} cannot be a window function")
+            func = W.WindowFunc(e.op, e.child)
+        else:
+            raise SqlError("OVER needs an aggregate or ranking function")
+        self.expect_op("(")
+        part, order, desc = [], [], []
+        rows_between = None
+        range_between = None
+        if self.kw("PARTITION"):
+            self.expect_kw("BY")
+            while True:
+                part.append(self.next()[1])
+                if not self.op(","):
+                    break
+        if self.kw("ORDER"):
+            self.expect_kw("BY")
+            while True:
+                order.append(self.next()[1])
+                if self.kw("DESC"):
+                    desc.append(True)
+                else:
+                    self.kw("ASC")
+                    desc.append(False)
+                if not self.op(","):
+                    break
+        frame_kind = None
+        if self.kw("ROWS"):
+            frame_kind = "rows"
+        elif self.kw("RANGE"):
+            frame_kind = "range"
+        if frame_kind:
+            self.expect_kw("BETWEEN")
+            lo = self._frame_bound(preceding=True)
+            self.expect_kw("AND")
+            hi = self._frame_bound(preceding=False)
+            if frame_kind == "rows":
+                rows_between = (int(lo) if lo is not None else None,
+                                int(hi) if hi is not None else None)
+                if rows_between[0] is None:
+                    raise SqlError("ROWS UNBOUNDED PRECEDING frame is the "
+                                   "default running frame; omit the clause")
+            else:
+                range_between = (lo, hi)
+        self.expect_op(")")
+        return W.WindowExpr(func, W.WindowSpec(
+            part, order, desc or None, rows_between, range_between))
+
+    def _frame_bound(self, preceding: bool):
+        if self.kw("UNBOUNDED"):
+            if not (self.kw("PRECEDING") or self.kw("FOLLOWING")):
+                raise SqlError("expected PRECEDING/FOLLOWING")
+            return None
+        if self.kw("CURRENT"):
+            self.expect_kw("ROW")
+            return 0
+        tok = self.next()[1]
+        v = float(tok) if "." in tok else int(tok)
+        if self.kw("PRECEDING"):
+            return -v
+        self.expect_kw("FOLLOWING")
+        return v
 
     # expression precedence: OR < AND < NOT < cmp < add < mul < unary
     def parse_expr(self):
@@ -435,6 +531,17 @@ class Parser:
                 if not self.op(","):
                     break
             self.expect_op(")")
+        if name in ("row_number", "rank", "dense_rank"):
+            from ..expr import windows as W
+
+            return {"row_number": W.row_number, "rank": W.rank,
+                    "dense_rank": W.dense_rank}[name]()
+        if name in ("lag", "lead"):
+            from ..expr import windows as W
+
+            off = int(args[1].value) if len(args) > 1 else 1
+            dflt = args[2].value if len(args) > 2 else None
+            return {"lag": W.lag, "lead": W.lead}[name](args[0], off, dflt)
         if name == "regexp_extract":
             from ..expr.expressions import RegexpExtract
 

@@ -126,3 +126,42 @@ def test_sql_collect_list(session):
     out = sorted(s.sql("SELECT k, collect_list(c) FROM tcoll GROUP BY k")
                  .collect())
     assert out == [("a", [1]), ("b", [2])]
+
+
+def test_sql_window_functions(session):
+    s = session
+    df = s.create_dataframe({"k": ["a", "a", "a", "b"], "t": [1, 2, 3, 1],
+                             "v": [10.0, 20.0, 30.0, 5.0]})
+    s.register("twin", df)
+    out = s.sql(
+        "SELECT k, t, row_number() OVER (PARTITION BY k ORDER BY t) rn, "
+        "SUM(v) OVER (PARTITION BY k ORDER BY t) rs "
+        "FROM twin ORDER BY k, t").collect()
+    assert out == [("a", 1, 1, 10.0), ("a", 2, 2, 30.0),
+                   ("a", 3, 3, 60.0), ("b", 1, 1, 5.0)]
+
+
+def test_sql_window_frames(session):
+    s = session
+    df = s.create_dataframe({"k": [1, 1, 1, 1], "t": [1.0, 2.0, 3.0, 10.0],
+                             "v": [1.0, 2.0, 3.0, 4.0]})
+    s.register("tfr", df)
+    rows = s.sql("SELECT SUM(v) OVER (PARTITION BY k ORDER BY t "
+                 "ROWS BETWEEN 1 PRECEDING AND CURRENT ROW) m FROM tfr"
+                 ).collect()
+    assert [r[0] for r in rows] == [1.0, 3.0, 5.0, 7.0]
+    rows = s.sql("SELECT SUM(v) OVER (PARTITION BY k ORDER BY t "
+                 "RANGE BETWEEN 1.0 PRECEDING AND 1.0 FOLLOWING) m FROM tfr"
+                 ).collect()
+    assert [r[0] for r in rows] == [3.0, 6.0, 5.0, 4.0]
+
+
+def test_sql_lag_lead(session):
+    s = session
+    df = s.create_dataframe({"k": [1, 1, 1], "t": [1, 2, 3],
+                             "v": [10, 20, 30]})
+    s.register("tlag", df)
+    rows = s.sql("SELECT lag(v) OVER (PARTITION BY k ORDER BY t) l, "
+                 "lead(v) OVER (PARTITION BY k ORDER BY t) r FROM tlag"
+                 ).collect()
+    assert rows == [(None, 20), (10, 30), (20, None)]
