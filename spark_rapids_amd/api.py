@@ -15,7 +15,7 @@ import numpy as np
 from .column import Column, ColumnBatch, Field, Schema
 from .config import RapidsConf, CONCURRENT_GPU_TASKS
 from .expr.aggregates import AggExpr
-from .expr.expressions import ColumnRef, Expression, col as _col
+from .expr.expressions import Expression, col as _col
 from .memory.semaphore import GpuSemaphore
 from .plan import logical as L
 from .plan.overrides import plan_physical

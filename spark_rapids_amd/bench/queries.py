@@ -8,12 +8,12 @@ table + repartition paths).
 """
 from __future__ import annotations
 
-from typing import Callable, Dict, List
+from typing import Dict, List
 
-from ..api import DataFrame, Session
-from ..expr.aggregates import avg, count_star, max_, min_, sum_
+from ..api import DataFrame
+from ..expr.aggregates import avg, count_star, sum_
 from ..expr.expressions import col, lit
-from ..types import FLOAT64, INT64
+from ..types import FLOAT64
 
 
 def q1_pricing_summary(t: Dict[str, DataFrame]) -> DataFrame:

@@ -13,12 +13,12 @@ from __future__ import annotations
 import argparse
 import json
 import time
-from typing import Callable, Dict, List
+from typing import Dict, List
 
 from ..api import Session
 from ..expr.aggregates import avg, count_star, max_, min_, sum_
-from ..expr.expressions import col, lit
-from ..expr.windows import rank, row_number, win_sum
+from ..expr.expressions import col
+from ..expr.windows import rank
 from . import datagen
 
 

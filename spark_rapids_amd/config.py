@@ -8,8 +8,8 @@ own docs (docs/configs.md) and that every layer reads through one object.
 from __future__ import annotations
 
 import threading
-from dataclasses import dataclass, field
-from typing import Any, Callable, Dict, List, Optional
+from dataclasses import dataclass
+from typing import Any, Callable, Dict, Optional
 
 
 @dataclass

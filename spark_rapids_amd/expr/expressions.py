@@ -7,7 +7,7 @@ on the CPU backend or on the GPU backend depending on where the batch lives.
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 from ..column import Column, ColumnBatch, Schema
 from ..types import (BOOL, DType, FLOAT64, INT32, INT64, STRING, TypeId,

@@ -10,7 +10,7 @@ the default when no order is given).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence
+from typing import Optional, Sequence
 
 from ..types import DType, FLOAT64, INT32, INT64
 from .expressions import Expression, _as_expr

@@ -15,7 +15,6 @@ from __future__ import annotations
 
 from typing import List, Optional
 
-import numpy as np
 import torch
 
 from ..column import Column, ColumnBatch, Schema, mask_nbytes

@@ -8,7 +8,7 @@ varints, binary, bool, nested struct skip).
 """
 from __future__ import annotations
 
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Optional, Tuple
 
 CT_STOP = 0

@@ -20,7 +20,7 @@ from typing import List, Optional, Sequence, Tuple
 import numpy as np
 import torch
 
-from ..column import Column, ColumnBatch, make_validity, torch_dtype
+from ..column import Column, ColumnBatch, make_validity
 from ..types import DType, TypeId
 
 

@@ -7,7 +7,7 @@ Exchange (inserted by the planner for distributed runs).
 """
 from __future__ import annotations
 
-from typing import List, Optional, Sequence, Tuple
+from typing import List, Optional, Sequence
 
 from ..column import Field, Schema
 from ..expr.aggregates import AggExpr

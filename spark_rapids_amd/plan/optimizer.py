@@ -11,7 +11,6 @@ from __future__ import annotations
 from typing import Optional, Set
 
 from ..expr.expressions import ColumnRef, Expression
-from ..expr.windows import WindowExpr
 from . import logical as L
 
 

@@ -11,8 +11,6 @@ from __future__ import annotations
 from typing import List, Sequence
 
 import numpy as np
-import torch
-
 from .. import ops
 from ..column import Column, ColumnBatch, Field, Schema
 from . import dist
