@@ -17,7 +17,8 @@ from .expr.aggregates import (approx_count_distinct, approx_percentile,
 from .expr.expressions import (CaseWhen, coalesce, col, date_add, date_sub,
                                datediff, greatest, hour, isin, least, lit,
                                minute, round_, second, when)
-from .expr.windows import (dense_rank, lag, lead, rank, row_number, win_avg,
+from .expr.windows import (dense_rank, lag, lead, nth_value, ntile,
+                           rank, row_number, win_avg,
                            win_count, win_max, win_min, win_sum)
 from .types import (BOOL, DATE32, FLOAT32, FLOAT64, INT8, INT16, INT32, INT64,
                     STRING, TIMESTAMP, DType)

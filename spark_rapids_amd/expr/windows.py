@@ -15,9 +15,9 @@ from typing import Optional, Sequence
 from ..types import DType, FLOAT64, INT32, INT64
 from .expressions import Expression, _as_expr
 
-RANKING = {"row_number", "rank", "dense_rank"}
+RANKING = {"row_number", "rank", "dense_rank", "ntile"}
 AGGS = {"sum", "count", "min", "max", "mean"}
-OFFSETS = {"lag", "lead"}
+OFFSETS = {"lag", "lead", "nth_value"}
 
 
 class WindowSpec:
@@ -108,6 +108,20 @@ class WindowExpr:
 
     def nullable(self, schema) -> bool:
         return self.func.op not in RANKING
+
+
+def ntile(n: int) -> WindowFunc:
+    """Bucket 1..n by position within the ordered partition (GpuNTile):
+    bucket = (rn-1)*n // size + 1 with Spark's remainder-to-front split."""
+    assert n >= 1
+    return WindowFunc("ntile", None, offset=n)
+
+
+def nth_value(e, k: int) -> WindowFunc:
+    """k-th value (1-based) of the ordered partition (GpuNthValue);
+    NULL when the partition has fewer than k rows."""
+    assert k >= 1
+    return WindowFunc("nth_value", e, offset=k)
 
 
 def row_number() -> WindowFunc:

@@ -231,7 +231,7 @@ class Tagger:
                     reasons.append(f"window key {k}: {r}")
             for w in node.window_exprs:
                 op = w.func.op
-                if op in ("row_number", "rank", "dense_rank"):
+                if op in ("row_number", "rank", "dense_rank", "ntile"):
                     continue
                 if w.spec.range_between is not None:
                     okt = cs.field(w.spec.order_by[0]).dtype
@@ -245,6 +245,10 @@ class Tagger:
                         reasons.append(
                             f"range frame over {okt} order key on CPU")
                 vt = w.func.child.dtype(cs) if w.func.child is not None else None
+                if op == "nth_value":
+                    if vt is not None and vt.is_nested:
+                        reasons.append(f"nth_value over {vt} not on GPU")
+                    continue
                 if op in ("lag", "lead"):
                     if vt is not None and vt.is_nested:
                         reasons.append(f"lag/lead over {vt} not on GPU")

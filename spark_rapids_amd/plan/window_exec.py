@@ -220,8 +220,39 @@ class WindowExec(PhysicalExec):
             dr = gb.binary_op("sub", ri32, at_start, INT32)
             return gb.binary_op_scalar("add", dr, 1, INT32)
 
+        if op == "ntile":
+            nt = w.func.offset
+            hp_ext = torch.cat([head_pos, torch.tensor(
+                [n], dtype=torch.int32, device="cuda")])
+            segid_next = gb.binary_op_scalar("add", gb.cast(
+                Column(INT32, n, seg_id.data, None, null_count=0), INT32),
+                1, INT32)
+            seg_end = _i32col(self._gather_i32(hp_ext, segid_next.data, n))
+            size = gb.binary_op("sub", seg_end, seg_start_col, INT32)
+            size = gb.cast(gb.binary_op_scalar("add", size, 1, INT32),
+                           INT64)
+            rn0 = gb.cast(gb.binary_op("sub", iota_col, seg_start_col,
+                                       INT32), INT64)
+            num = gb.binary_op_scalar("mul", rn0, nt, INT64)
+            t = gb.binary_op("int_div", num, size, INT64)
+            return gb.cast(gb.binary_op_scalar("add", t, 1, INT64), INT32)
+
         # value-based functions
         vc = w.func.child.eval(table, cs)
+        if op == "nth_value":
+            hp_ext = torch.cat([head_pos, torch.tensor(
+                [n], dtype=torch.int32, device="cuda")])
+            segid_next = gb.binary_op_scalar("add", gb.cast(
+                Column(INT32, n, seg_id.data, None, null_count=0), INT32),
+                1, INT32)
+            seg_end = _i32col(self._gather_i32(hp_ext, segid_next.data, n))
+            seg_end = gb.binary_op_scalar("sub", seg_end, 1, INT32)
+            src = gb.binary_op_scalar("add", seg_start_col,
+                                      w.func.offset - 1, INT32)
+            ok = gb.binary_op("le", src, seg_end, DType.bool_())
+            neg1 = Column.full(-1, INT32, n, "cuda")
+            srcm = gb.if_else(ok, src, neg1)
+            return gb.gather(ColumnBatch([vc], n), srcm).columns[0]
         if op in ("lag", "lead"):
             k = w.func.offset if op == "lag" else -w.func.offset
             src = gb.binary_op_scalar("sub", iota_col, k, INT32)
@@ -486,10 +517,27 @@ def _compute(w: WindowExpr, table: ColumnBatch, cs, n, heads, ochange, idx,
     if op == "dense_rank":
         dr = np.cumsum(ochange)
         return _make((dr - dr[seg_start] + 1).astype(np.int32), None, out_dt)
+    if op == "ntile":
+        nt = w.func.offset
+        seg_end = _segment_ends(heads, idx, n)
+        size = (seg_end - seg_start + 1).astype(np.int64)
+        rn0 = (idx - seg_start).astype(np.int64)
+        return _make((rn0 * nt // size + 1).astype(np.int32), None, out_dt)
 
     vc = w.func.child.eval(table, cs)
     v = _vals(vc).astype(np.float64) if vc.dtype.is_numeric else _vals(vc)
     valid = _valid(vc)
+    if op == "nth_value":
+        seg_end = _segment_ends(heads, idx, n)
+        src = seg_start + w.func.offset - 1
+        ok = src <= seg_end
+        srcc = np.where(ok, src, 0)
+        raw = _vals(vc)
+        outv = np.where(ok, raw[srcc], 0)
+        outvalid = np.where(ok, valid[srcc], False)
+        return _make(outv.astype(out_dt.numpy_dtype())
+                     if out_dt.is_fixed_width else outv,
+                     outvalid if not outvalid.all() else None, out_dt)
     if op in ("lag", "lead"):
         k = w.func.offset if op == "lag" else -w.func.offset
         src = idx - k

@@ -277,3 +277,38 @@ class TestRangeFrames:
             assert (x is None) == (y is None)
             if x is not None:
                 assert x == pytest.approx(y, rel=1e-9)
+
+
+def test_ntile_nth_value_cpu(cpu_session):
+    from spark_rapids_amd import ntile, nth_value
+
+    df = cpu_session.create_dataframe({
+        "k": ["a"] * 7 + ["b"] * 3,
+        "t": list(range(7)) + [0, 1, 2],
+        "v": [float(x) for x in range(10)]})
+    out = (df.with_column("nt", ntile(3).over(["k"], ["t"]))
+           .with_column("n2", nth_value(col("v"), 2).over(["k"], ["t"]))
+           .with_column("n9", nth_value(col("v"), 9).over(["k"], ["t"]))
+           .collect())
+    nts = [r[3] for r in out]
+    assert nts == [1, 1, 1, 2, 2, 3, 3, 1, 2, 3]  # 3/2/2 split like Spark
+    assert [r[4] for r in out] == [1.0] * 7 + [8.0] * 3
+    assert all(r[5] is None for r in out)  # k=9 beyond every partition
+
+
+@pytest.mark.gpu
+def test_gpu_ntile_nth_matches_cpu():
+    from spark_rapids_amd import ntile, nth_value
+
+    def q(s):
+        df = _rand_df(s, 4000) if False else s.create_dataframe({
+            "k": [int(v) % 11 for v in range(4000)],
+            "t": [int(v) // 11 for v in range(4000)],
+            "v": [float(v) * 0.5 for v in range(4000)]})
+        return (df.with_column("nt", ntile(4).over(["k"], ["t"]))
+                .with_column("nv", nth_value(col("v"), 3).over(["k"], ["t"]))
+                .collect())
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    assert sorted(q(sg)) == sorted(q(sc))
