@@ -15,7 +15,8 @@ from .expr.aggregates import (approx_count_distinct, approx_percentile,
                               max_, min_, stddev, sum_distinct,
                               sum_, variance)
 from .expr.expressions import (CaseWhen, coalesce, col, date_add, date_sub,
-                               datediff, greatest, hour, isin, least, lit,
+                               datediff, dayofweek, greatest, hour, isin,
+                               least, lit, quarter,
                                minute, round_, second, when)
 from .expr.windows import (dense_rank, lag, lead, nth_value, ntile,
                            rank, row_number, win_avg,

@@ -749,6 +749,27 @@ def date_sub(d, days) -> Expression:
     return BinaryExpr("sub", _as_expr(d), _as_expr(days))
 
 
+def dayofweek(d) -> Expression:
+    """Spark dayofweek: 1 = Sunday .. 7 = Saturday (1970-01-01 was a
+    Thursday, day-number 4 in this scheme)."""
+    e = _as_expr(d)
+    days = CastExpr(CastExpr(e, DType.date32()), INT64)
+    return BinaryExpr("add",
+                      BinaryExpr("pmod", BinaryExpr("add", days,
+                                                    Literal(4)),
+                                 Literal(7)), Literal(1))
+
+
+def quarter(d) -> Expression:
+    """quarter 1..4 from the month."""
+    e = _as_expr(d)
+    m = UnaryExpr("month", e)
+    return BinaryExpr("add",
+                      BinaryExpr("int_div",
+                                 BinaryExpr("sub", m, Literal(1)),
+                                 Literal(3)), Literal(1))
+
+
 def datediff(end, start) -> Expression:
     from ..types import DATE32
 
