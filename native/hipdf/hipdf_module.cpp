@@ -117,6 +117,7 @@ void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
                              hipStream_t);
+void hipdf_pq_delta_i64(const void*, int64_t, int64_t, void*, hipStream_t);
 void hipdf_orc_bool_rle(const void*, int64_t, int64_t, void*, hipStream_t);
 void hipdf_orc_rle_v2(const void*, int64_t, int64_t, int, void*,
                       hipStream_t);
@@ -445,6 +446,11 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("scan_add_offsets_f64", [](int64_t out, int64_t sums, int64_t n,
                                    int64_t stream) {
     hipdf_scan_add_offsets_f64(PM(out), P(sums), n, S(stream));
+    check_async();
+  });
+  m.def("pq_delta_i64", [](int64_t b, int64_t nbytes, int64_t n,
+                           int64_t out, int64_t stream) {
+    hipdf_pq_delta_i64(P(b), nbytes, n, PM(out), S(stream));
     check_async();
   });
   m.def("orc_bool_rle", [](int64_t b, int64_t nbytes, int64_t n,

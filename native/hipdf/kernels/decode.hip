@@ -117,6 +117,67 @@ __global__ void k_levels_to_mask(const int32_t* __restrict__ levels,
   }
 }
 
+// parquet DELTA_BINARY_PACKED (encoding 5): header(block_size,
+// miniblocks/block, total, first zigzag) then per block min_delta +
+// per-miniblock bit widths + LSB-first packed deltas. Sequential walker,
+// one launch per page (pages decode concurrently on the stream).
+__device__ __forceinline__ uint64_t pq_uvarint(const uint8_t* b,
+                                               int64_t* p) {
+  uint64_t out = 0;
+  int sh = 0;
+  while (true) {
+    uint8_t v = b[(*p)++];
+    out |= (uint64_t)(v & 0x7F) << sh;
+    if (!(v & 0x80)) return out;
+    sh += 7;
+  }
+}
+
+__global__ void k_pq_delta_i64(const uint8_t* __restrict__ b,
+                               int64_t nbytes, int64_t n,
+                               int64_t* __restrict__ out) {
+  if (blockIdx.x || threadIdx.x) return;
+  int64_t p = 0;
+  uint64_t block_size = pq_uvarint(b, &p);
+  uint64_t mini_per_block = pq_uvarint(b, &p);
+  uint64_t total = pq_uvarint(b, &p);
+  (void)total;
+  uint64_t fz = pq_uvarint(b, &p);
+  int64_t cur = (int64_t)(fz >> 1) ^ -(int64_t)(fz & 1);
+  uint64_t per_mini = block_size / mini_per_block;
+  int64_t k = 0;
+  if (k < n) out[k++] = cur;
+  while (k < n && p < nbytes) {
+    uint64_t mdz = pq_uvarint(b, &p);
+    int64_t min_delta = (int64_t)(mdz >> 1) ^ -(int64_t)(mdz & 1);
+    int64_t widths_at = p;
+    p += mini_per_block;
+    for (uint64_t m = 0; m < mini_per_block && k < n; ++m) {
+      int w = b[widths_at + m];
+      // LSB-first bit unpacking (parquet packing order)
+      int64_t bit = 0;
+      for (uint64_t i = 0; i < per_mini && k < n; ++i) {
+        uint64_t v = 0;
+        for (int got = 0; got < w; ++got, ++bit)
+          v |= (uint64_t)((b[p + (bit >> 3)] >> (bit & 7)) & 1) << got;
+        cur += min_delta + (int64_t)v;
+        out[k++] = cur;
+      }
+      p += ((int64_t)w * per_mini + 7) / 8;
+    }
+  }
+}
+
+extern "C" {
+
+void hipdf_pq_delta_i64(const void* b, int64_t nbytes, int64_t n, void* out,
+                        hipStream_t stream) {
+  hipLaunchKernelGGL(k_pq_delta_i64, dim3(1), dim3(64), 0, stream,
+                     (const uint8_t*)b, nbytes, n, (int64_t*)out);
+}
+
+}  // extern "C"
+
 // ---- ORC stream decoders (reference analogue: libcudf ORC reader fed by
 // GpuOrcScan; SURVEY.md §2.3 ORC row). Sequential single-thread walkers:
 // one launch per (stripe, column, stream) so stripes and columns decode

@@ -25,6 +25,7 @@ from ..types import DType, TypeId
 from . import thrift_compact as tc
 
 PLAIN, PLAIN_DICTIONARY, RLE, RLE_DICTIONARY = 0, 2, 3, 8
+DELTA_BINARY_PACKED = 5
 DATA_PAGE, DICTIONARY_PAGE, DATA_PAGE_V2 = 0, 2, 3
 
 _PHYS_NP = {
@@ -262,6 +263,32 @@ class _ChunkDecoder:
             out = gb._gather_col(dense_col, ridx, n, maybe_negative=True)
             return Column(self.dtype, n, out.data, mask, out.offsets,
                           null_count=None)
+        if encoding == DELTA_BINARY_PACKED and self.phys in ("INT32",
+                                                             "INT64"):
+            page = torch.from_numpy(np.frombuffer(
+                values, dtype=np.uint8).copy()).cuda()
+            vals = torch.empty(max(n_valid, 1), dtype=torch.int64,
+                               device="cuda")[:n_valid]
+            if n_valid:
+                self.ext.pq_delta_i64(page.data_ptr(), page.numel(),
+                                      n_valid, vals.data_ptr(), self.s)
+            tdt = torch_dtype(self.dtype)
+            if tdt != torch.int64:
+                from ..ops import gpu_backend as gb
+
+                wide = Column(DType.int64(), n_valid, vals, None,
+                              null_count=0)
+                dense = gb.cast(wide, self.dtype).data
+            else:
+                dense = vals
+            if not nulls:
+                return Column(self.dtype, n, dense, None, null_count=0)
+            out = torch.zeros(n, dtype=tdt, device="cuda")
+            if n_valid:
+                self.ext.scatter_fixed(self.dtype.itemsize, dense.data_ptr(),
+                                       valid_idx.data_ptr(), out.data_ptr(),
+                                       n_valid, self.s)
+            return Column(self.dtype, n, out, mask, null_count=None)
         if encoding == PLAIN:
             if self.phys not in _PHYS_NP:
                 raise NotImplementedError(f"PLAIN {self.phys}")

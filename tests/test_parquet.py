@@ -203,3 +203,27 @@ def test_gpu_writer_own_reader_roundtrip(tmp_path):
     assert batch.columns[0].to_pylist() == \
         [int(v) if v % 9 else None for v in range(n)]
     assert batch.columns[1].to_pylist() == [f"x{v}" for v in range(n)]
+
+
+@pytest.mark.gpu
+def test_gpu_delta_binary_packed(tmp_path):
+    from spark_rapids_amd.io.parquet_gpu import read_parquet_gpu
+
+    rng = np.random.default_rng(3)
+    n = 30_000
+    tbl = pa.table({
+        "a": pa.array((np.arange(n, dtype=np.int64) * 3 + 7),
+                      pa.int64()),
+        "b": pa.array(rng.integers(-10**6, 10**6, n).astype(np.int32),
+                      mask=rng.random(n) < 0.1),
+        "c": pa.array(rng.integers(-10**14, 10**14, n).astype(np.int64)),
+    })
+    p = str(tmp_path / "delta.parquet")
+    pq.write_table(tbl, p, use_dictionary=False, data_page_version="2.0",
+                   column_encoding={"a": "DELTA_BINARY_PACKED",
+                                    "b": "DELTA_BINARY_PACKED",
+                                    "c": "DELTA_BINARY_PACKED"})
+    batch = read_parquet_gpu(p, ["a", "b", "c"]).cpu()
+    for i, name in enumerate(["a", "b", "c"]):
+        assert batch.columns[i].to_pylist() == \
+            tbl.column(name).to_pylist(), name
