@@ -227,10 +227,10 @@ class WindowExec(PhysicalExec):
             segid_next = gb.binary_op_scalar("add", gb.cast(
                 Column(INT32, n, seg_id.data, None, null_count=0), INT32),
                 1, INT32)
-            seg_end = _i32col(self._gather_i32(hp_ext, segid_next.data, n))
-            size = gb.binary_op("sub", seg_end, seg_start_col, INT32)
-            size = gb.cast(gb.binary_op_scalar("add", size, 1, INT32),
-                           INT64)
+            nxt_head = _i32col(self._gather_i32(hp_ext, segid_next.data, n))
+            # next segment head minus this segment start IS the size
+            size = gb.cast(gb.binary_op("sub", nxt_head, seg_start_col,
+                                        INT32), INT64)
             rn0 = gb.cast(gb.binary_op("sub", iota_col, seg_start_col,
                                        INT32), INT64)
             num = gb.binary_op_scalar("mul", rn0, nt, INT64)
