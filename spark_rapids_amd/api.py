@@ -315,6 +315,10 @@ class Session:
         from .shuffle import dist as _dist
 
         _dist.set_codec(self.conf.get(SHUFFLE_CODEC))
+        from .config import FILECACHE
+        from .io import filecache as _fc
+
+        _fc.configure(self.conf.get(FILECACHE))
 
     # ---- conf ----------------------------------------------------------
     def set(self, key: str, value) -> "Session":
@@ -455,6 +459,12 @@ class Session:
 
         src = OrcTable(path)
         return DataFrame(self, L.Scan(src, src.schema, f"orc:{path}"))
+
+    def read_hive_text(self, path: str, header: bool = False,
+                       delimiter: str = "\x01") -> DataFrame:
+        """Hive-style delimited text (ctrl-A separated by default): the
+        CSV scan with the Hive delimiter (GpuHiveTextFileFormat analogue)."""
+        return self.read_csv(path, header=header, delimiter=delimiter)
 
     def read_json(self, path: str) -> DataFrame:
         from .io.formats import JsonTable

@@ -175,6 +175,11 @@ PRUNE_COLUMNS = bool_conf(
     "spark.rapids.sql.optimizer.pruneColumns.enabled", True,
     "Push projections below joins/aggregates so unused columns are never "
     "gathered or transferred (Catalyst-optimizer analogue).")
+FILECACHE = bool_conf(
+    "spark.rapids.filecache.enabled", False,
+    "Cache decoded scan batches per (file, mtime) in host memory so "
+    "repeated scans skip IO + decode (reference analogue: the filecache "
+    "layer). Best for dimension tables read by many queries.")
 PUSH_FILTERS = bool_conf(
     "spark.rapids.sql.optimizer.pushFilters.enabled", True,
     "Push filter conjuncts below joins when they reference only one side "

@@ -152,6 +152,16 @@ class ParquetTable:
             self.schema = Schema(keep)
 
     def _read_one(self, path: str) -> ColumnBatch:
+        from .filecache import get_cached, put_cached
+
+        cached = get_cached(path, self.columns)
+        if cached is not None:
+            return cached
+        batch = self._read_one_uncached(path)
+        put_cached(path, self.columns, batch)
+        return batch
+
+    def _read_one_uncached(self, path: str) -> ColumnBatch:
         if self.reader == "GPU_DECODE":
             try:
                 from .parquet_gpu import read_parquet_gpu

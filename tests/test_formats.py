@@ -183,3 +183,12 @@ def test_gpu_orc_query(tmp_path):
            .agg(count_star(), sum_(col("i32"))).collect())
     vals = [v for v in t.column("i32").to_pylist() if v is not None and v > 0]
     assert out[0][0] == len(vals) and out[0][1] == sum(vals)
+
+
+def test_hive_text(tmp_path, session):
+    p = str(tmp_path / "h.txt")
+    with open(p, "w") as f:
+        f.write("1\x01alpha\x012.5\n2\x01beta\x013.5\n")
+    df = session.read_hive_text(p)
+    rows = df.collect()
+    assert len(rows) == 2 and rows[0][1] == "alpha" and rows[1][2] == 3.5

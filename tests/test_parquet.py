@@ -227,3 +227,26 @@ def test_gpu_delta_binary_packed(tmp_path):
     for i, name in enumerate(["a", "b", "c"]):
         assert batch.columns[i].to_pylist() == \
             tbl.column(name).to_pylist(), name
+
+
+def test_filecache_roundtrip(tmp_path):
+    p = str(tmp_path / "fc.parquet")
+    _write_file(p, n=2000)
+    s = sr.Session({"spark.rapids.sql.enabled": False,
+                    "spark.rapids.filecache.enabled": True})
+    df1 = s.read_parquet(p).agg(count_star()).collect()
+    from spark_rapids_amd.io import filecache as fc
+
+    assert len(fc._cache) == 1
+    df2 = s.read_parquet(p).agg(count_star()).collect()
+    assert df1 == df2
+    # mtime bump invalidates
+    import os
+    import time
+
+    time.sleep(0.01)
+    os.utime(p)
+    s.read_parquet(p).agg(count_star()).collect()
+    assert len(fc._cache) == 2
+    sr.Session({"spark.rapids.sql.enabled": False})  # off -> cleared
+    assert len(fc._cache) == 0
