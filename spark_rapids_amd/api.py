@@ -165,7 +165,11 @@ class DataFrame:
         return self.plan.schema()
 
     def physical_plan(self):
-        return plan_physical(self.plan, self.session.conf)
+        # cached per DataFrame: repeated collect() of the same frame
+        # (dashboards, benchmark steps) skips the tag/convert pass
+        if getattr(self, "_phys", None) is None:
+            self._phys = plan_physical(self.plan, self.session.conf)
+        return self._phys
 
     def collect_batch(self) -> ColumnBatch:
         exec_ = self.physical_plan()

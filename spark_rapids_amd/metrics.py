@@ -27,9 +27,10 @@ def instrument(exec_) -> None:
     seen = set()
 
     def _wrap(e):
-        if id(e) in seen:
+        if id(e) in seen or getattr(e, "_instrumented", False):
             return
         seen.add(id(e))
+        e._instrumented = True
         orig = e.execute
         e.metrics = {"opTimeMs": 0.0, "numOutputRows": 0,
                      "numOutputBatches": 0}
