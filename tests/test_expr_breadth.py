@@ -64,3 +64,18 @@ def test_timestamp_fields(session):
     assert df.select(hour(col("t")).alias("h")).to_pydict()["h"] == [13, 13]
     assert df.select(minute(col("t")).alias("m")).to_pydict()["m"] == [45, 46]
     assert df.select(second(col("t")).alias("s")).to_pydict()["s"] == [30, 31]
+
+
+def test_dayofweek_quarter(session):
+    import datetime
+
+    from spark_rapids_amd import DATE32, dayofweek, quarter
+
+    days = [0, 1, 3, 100, 19000, 7305]
+    df = session.create_dataframe({"d": days}, dtypes={"d": DATE32})
+    out = df.select(dayofweek(col("d")).alias("w"),
+                    quarter(col("d")).alias("q")).to_pydict()
+    for dv, w, q in zip(days, out["w"], out["q"]):
+        date = datetime.date(1970, 1, 1) + datetime.timedelta(days=dv)
+        assert w == (date.weekday() + 1) % 7 + 1
+        assert q == (date.month - 1) // 3 + 1
