@@ -79,3 +79,18 @@ def test_dayofweek_quarter(session):
         date = datetime.date(1970, 1, 1) + datetime.timedelta(days=dv)
         assert w == (date.weekday() + 1) % 7 + 1
         assert q == (date.month - 1) // 3 + 1
+
+
+def test_udf_compiler_traces_to_expressions(session):
+    from spark_rapids_amd.tools.udf import UdfFallback, compile_udf
+
+    df = session.create_dataframe({"x": [1.0, 2.0, None], "y": [10, 20, 30]})
+    expr = compile_udf(lambda a, b: (a + b) * 2.0 - 1.0, "x", "y")
+    out = df.with_column("z", expr).to_pydict()["z"]
+    assert out == [21.0, 43.0, None]
+
+    cond = compile_udf(lambda a: (a > 1.5) & a.is_not_null(), "x")
+    assert df.filter(cond).count() == 1
+
+    with pytest.raises(UdfFallback):
+        compile_udf(lambda a: float(a) + 1, "x")  # materializes -> fallback
