@@ -470,6 +470,19 @@ class Session:
         CSV scan with the Hive delimiter (GpuHiveTextFileFormat analogue)."""
         return self.read_csv(path, header=header, delimiter=delimiter)
 
+    def read_avro(self, path: str) -> DataFrame:
+        """Avro object-container scan (flat records; host decode)."""
+        from .io.avro import AvroTable
+
+        t = AvroTable(path)
+        return DataFrame(self, L.Scan(t, t.schema, "avro"))
+
+    def write_avro(self, df: DataFrame, path: str,
+                   codec: str = "deflate"):
+        from .io.avro import write_avro
+
+        write_avro(df.collect_batch(), df.schema, path, codec)
+
     def read_json(self, path: str) -> DataFrame:
         from .io.formats import JsonTable
 

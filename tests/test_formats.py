@@ -192,3 +192,28 @@ def test_hive_text(tmp_path, session):
     df = session.read_hive_text(p)
     rows = df.collect()
     assert len(rows) == 2 and rows[0][1] == "alpha" and rows[1][2] == 3.5
+
+
+def test_avro_roundtrip(tmp_path, session):
+    df = session.create_dataframe({
+        "i": [1, 2, None, 4],
+        "f": [1.5, None, 2.5, -3.5],
+        "s": ["x", "", None, "zz"],
+        "b": [True, False, None, True]})
+    p = str(tmp_path / "t.avro")
+    session.write_avro(df, p)
+    back = session.read_avro(p).to_pydict()
+    assert back["i"] == [1, 2, None, 4]
+    assert back["f"] == [1.5, None, 2.5, -3.5]
+    assert back["s"] == ["x", "", None, "zz"]
+    assert back["b"] == [True, False, None, True]
+
+
+def test_avro_null_codec_and_bytes(tmp_path, session):
+    df = session.create_dataframe({"i": list(range(500)),
+                                   "s": [f"v{v}" for v in range(500)]})
+    p = str(tmp_path / "u.avro")
+    session.write_avro(df, p, codec="null")
+    out = session.read_avro(p)
+    assert out.count() == 500
+    assert out.to_pydict()["s"][123] == "v123"
