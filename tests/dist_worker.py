@@ -97,6 +97,43 @@ def main():
     gexp = int(sum(rc[v % 97] if rc[v % 97] else 1 for v in all_rows.tolist()))
     assert left_join[0][0] == gexp, (left_join, gexp)
 
+    # single-pass aggregates (collect_list / percentile / distinct): rows
+    # are exchanged by key hash BEFORE aggregation — every rank must hold
+    # each key's complete value set
+    from spark_rapids_amd import collect_list, count_distinct, percentile
+
+    outc = df.group_by("c").agg(collect_list(col("k")),
+                                percentile(col("v"), 0.5)).collect()
+    outd = dict((r[0], r[1]) for r in
+                df.group_by("c").agg(count_distinct(col("k"))).collect())
+    exp_by_c = {}
+    for c in range(5):
+        m = all_rows % 5 == c
+        ks = (all_rows[m] % 97).tolist()
+        exp_by_c[c] = (sorted(ks), len(set(ks)),
+                       float(np.percentile(all_rows[m].astype(float), 50)))
+    seen_cs = []
+    for c, lst, p50 in outc:
+        el, ec, ep = exp_by_c[c]
+        assert sorted(lst) == el, (c, len(lst), len(el))
+        assert outd[c] == ec and abs(p50 - ep) < 1e-9, (c, p50, exp_by_c[c])
+        seen_cs.append(c)
+    gath = [None] * world
+    td.all_gather_object(gath, sorted(seen_cs))
+    if rank == 0:
+        flat = [c for cs in gath for c in cs]
+        assert sorted(flat) == [0, 1, 2, 3, 4], flat
+
+    # rollup across ranks: grand total row must count every rank's rows
+    ro = df.rollup("c").agg(count_star()).collect()
+    grand = [r for r in ro if r[1] == 1]
+    gath2 = [None] * world
+    td.all_gather_object(gath2, [tuple(r) for r in ro])
+    if rank == 0:
+        allro = [r for rs in gath2 for r in rs]
+        g = [r for r in allro if r[1] == 1]
+        assert sum(r[2] for r in g) == n_total, g
+
     td.barrier()
     if rank == 0:
         print("DIST_OK")
