@@ -176,3 +176,38 @@ def test_gpu_external_sort_matches_cpu():
     for r1, r2 in zip(g, c):
         assert r1[1] == r2[1]
         assert r1[0] == r2[0] or (r1[0] != r1[0] and r2[0] != r2[0])
+
+
+def test_async_writer_throttles_and_raises(tmp_path):
+    import time
+
+    from spark_rapids_amd.io.async_write import AsyncWriter
+
+    w = AsyncWriter(max_pending=1)
+    done = []
+    for i in range(4):
+        w.submit(lambda i=i: (time.sleep(0.01), done.append(i)))
+    w.close()
+    assert done == [0, 1, 2, 3]
+
+    w2 = AsyncWriter()
+    w2.submit(lambda: (_ for _ in ()).throw(IOError("disk full")))
+    import pytest as _pytest
+
+    with _pytest.raises(IOError):
+        w2.close()
+
+
+@pytest.mark.gpu
+def test_no_leak_across_query():
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, sum_
+    from spark_rapids_amd.tools.memwatch import assert_no_leak
+
+    s = sr.Session()
+    df = s.create_dataframe({"k": list(range(1000)),
+                             "v": [float(v) for v in range(1000)]})
+    df.group_by("k").agg(sum_(col("v"))).collect()  # warm caches
+    with assert_no_leak(tolerance_bytes=16 << 20):
+        for _ in range(5):
+            df.group_by("k").agg(sum_(col("v"))).collect()
