@@ -22,6 +22,8 @@ _EXECS = [
     ("Join", "hash equi-join inner/left/semi/anti/full, broadcast or "
      "shuffled build side", "fixed-width + string + decimal128 keys on GPU"),
     ("CrossJoin", "cartesian gather maps (+ filter for non-equi)", "all"),
+    ("Generate", "explode / posexplode (+_outer) over LIST columns",
+     "outer pads on CPU"),
     ("Sort", "stable LSD radix sort; out-of-core range-partitioned spill "
      "buckets", "fixed-width keys on GPU; string/decimal128 sort keys "
      "fall back"),
@@ -65,7 +67,8 @@ def supported_ops_doc() -> str:
               "## Window functions", "",
               "`row_number`, `rank`, `dense_rank`, `sum`, `count`, `min`, "
               "`max`, `avg` (running, bounded ROWS BETWEEN, whole "
-              "partition), `lag`, `lead`", ""]
+              "partition, ROWS and RANGE frames), `lag`, `lead`, `ntile`, "
+              "`nth_value`; SQL OVER clause", ""]
     return "\n".join(lines)
 
 
