@@ -175,6 +175,11 @@ PRUNE_COLUMNS = bool_conf(
     "spark.rapids.sql.optimizer.pruneColumns.enabled", True,
     "Push projections below joins/aggregates so unused columns are never "
     "gathered or transferred (Catalyst-optimizer analogue).")
+JOIN_SUBPARTITION_BYTES = int_conf(
+    "spark.rapids.sql.join.subPartition.targetBytes", 1 << 30,
+    "Build sides larger than this are hash-split into buckets and joined "
+    "bucket-by-bucket (GpuSubPartitionHashJoin analogue), bounding the "
+    "peak size of any single hash table and its gather maps.")
 FILECACHE = bool_conf(
     "spark.rapids.filecache.enabled", False,
     "Cache decoded scan batches per (file, mtime) in host memory so "

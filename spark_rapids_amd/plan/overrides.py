@@ -367,12 +367,14 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
                                    input_replicated=L.is_replicated(node.child),
                                    merge_target_bytes=conf.get(BATCH_SIZE_BYTES))
     if isinstance(node, L.Join):
-        from ..config import BROADCAST_THRESHOLD
+        from ..config import BROADCAST_THRESHOLD, JOIN_SUBPARTITION_BYTES
 
         return P.HashJoinExec(device, kids[0], kids[1], node.left_on,
                               node.right_on, node.how, node.schema(),
                               right_replicated=L.is_replicated(node.right),
-                              broadcast_threshold=conf.get(BROADCAST_THRESHOLD))
+                              broadcast_threshold=conf.get(BROADCAST_THRESHOLD),
+                              sub_partition_bytes=conf.get(
+                                  JOIN_SUBPARTITION_BYTES))
     if isinstance(node, L.MapBatches):
         return P.MapBatchesExec(node.fn, _ensure_device(kids[0], "cpu"),
                                 node.schema())

@@ -455,13 +455,15 @@ class HashJoinExec(PhysicalExec):
     def __init__(self, device: str, left: PhysicalExec, right: PhysicalExec,
                  left_on: List[str], right_on: List[str], how: str,
                  schema: Schema, right_replicated: bool = True,
-                 broadcast_threshold: int = 512 << 20):
+                 broadcast_threshold: int = 512 << 20,
+                 sub_partition_bytes: int = 1 << 30):
         super().__init__(device, schema, [left, right])
         self.left_on = left_on
         self.right_on = right_on
         self.how = how
         self.right_replicated = right_replicated
         self.broadcast_threshold = broadcast_threshold
+        self.sub_partition_bytes = sub_partition_bytes
         self._strategy = "local"
 
     def _local_or_empty(self, batches: List[ColumnBatch], schema: Schema):
@@ -528,6 +530,14 @@ class HashJoinExec(PhysicalExec):
                              _np.zeros(rtable.num_rows, dtype=bool))
         lsource = lbatches_override if lbatches_override is not None \
             else left.execute()
+        same_types = all(
+            left.schema.fields[a].dtype.id == right.schema.fields[b].dtype.id
+            for a, b in zip(lkidx, rkidx))
+        if rtable is not None and same_types \
+                and rtable.nbytes > self.sub_partition_bytes:
+            yield from self._execute_subpartitioned(
+                rtable, lsource, lkidx, rkidx, left.schema)
+            return
         for lbatch in lsource:
             if lbatch.num_rows == 0:
                 continue
@@ -553,6 +563,76 @@ class HashJoinExec(PhysicalExec):
             extra = self._unmatched_right(rtable, right_matched, left.schema)
             if extra is not None and extra.num_rows:
                 yield extra
+
+    def _execute_subpartitioned(self, rtable, lsource, lkidx, rkidx,
+                                left_schema) -> Iterator[ColumnBatch]:
+        """Sub-partitioned join (GpuSubPartitionHashJoin analogue): both
+        sides are hash-split on the join keys and joined bucket-by-bucket;
+        a key lands in exactly one bucket, so per-bucket inner/left/semi/
+        anti/full results concatenate to the exact join. Left pieces are
+        spill-registered between buckets."""
+        from ..memory.spill import SpillableBatch
+
+        nb = int(min(64, max(2, -(-rtable.nbytes //
+                                  self.sub_partition_bytes))))
+        rparted, roffs = ops.hash_partition(rtable, rkidx, nb)
+        rbuckets = [_slice_rows(rparted, roffs[i], roffs[i + 1])
+                    for i in range(nb)]
+        lpieces: List[List] = [[] for _ in range(nb)]
+        for lbatch in lsource:
+            if lbatch.num_rows == 0:
+                continue
+            parted, offs = ops.hash_partition(lbatch, lkidx, nb)
+            for i in range(nb):
+                piece = _slice_rows(parted, offs[i], offs[i + 1])
+                if piece.num_rows:
+                    lpieces[i].append(SpillableBatch(piece))
+        for i in range(nb):
+            rb = rbuckets[i]
+            parts = [h.get() for h in lpieces[i]]
+            for h in lpieces[i]:
+                h.close()
+            lb = ops.concat_batches(parts) if len(parts) > 1 else (
+                parts[0] if parts else None)
+            if lb is None or lb.num_rows == 0:
+                if self.how == "full" and rb.num_rows:
+                    extra = self._unmatched_right(
+                        rb, self._fresh_matched(rb), left_schema)
+                    if extra is not None and extra.num_rows:
+                        yield extra
+                continue
+            if rb.num_rows == 0:
+                if self.how == "anti":
+                    yield lb
+                elif self.how in ("left", "full"):
+                    yield self._left_with_null_right(lb)
+                continue
+            right_matched = self._fresh_matched(rb) \
+                if self.how == "full" else None
+            lmap, rmap = ops.join_gather_maps(lb, rb, lkidx, rkidx,
+                                              self.how, right_matched)
+            if self.how in ("semi", "anti"):
+                out = ops.gather(lb, lmap)
+                if out.num_rows:
+                    yield out
+            else:
+                lout = ops.gather(lb, lmap)
+                rout = ops.gather(rb, rmap)
+                if lout.num_rows:
+                    yield ColumnBatch(lout.columns + rout.columns,
+                                      lout.num_rows)
+            if self.how == "full":
+                extra = self._unmatched_right(rb, right_matched, left_schema)
+                if extra is not None and extra.num_rows:
+                    yield extra
+
+    def _fresh_matched(self, rb: ColumnBatch):
+        import numpy as _np
+        import torch as _torch
+
+        return (_torch.zeros(rb.num_rows, dtype=_torch.uint8,
+                             device="cuda") if self.gpu
+                else _np.zeros(rb.num_rows, dtype=bool))
 
     def _unmatched_right(self, rtable: ColumnBatch, right_matched,
                          left_schema: Schema):

@@ -211,3 +211,53 @@ def test_no_leak_across_query():
     with assert_no_leak(tolerance_bytes=16 << 20):
         for _ in range(5):
             df.group_by("k").agg(sum_(col("v"))).collect()
+
+
+def test_subpartitioned_join_matches_plain():
+    """Tiny subPartition threshold forces the bucketed join path."""
+    import numpy as np
+    import spark_rapids_amd as sr
+
+    rng = np.random.default_rng(31)
+    lk = [int(v) for v in rng.integers(0, 200, 5000)]
+    rk = [int(v) for v in rng.integers(0, 250, 800)]
+    data_l = {"k": lk, "v": [float(v) for v in range(5000)]}
+    data_r = {"k": rk, "w": [float(v) for v in range(800)]}
+    small = sr.Session({"spark.rapids.sql.enabled": False,
+                        "spark.rapids.sql.join.subPartition.targetBytes": 256})
+    plain = sr.Session({"spark.rapids.sql.enabled": False})
+
+    for how in ("inner", "left", "semi", "anti", "full"):
+        def q(s):
+            l = s.create_dataframe(data_l)
+            r = s.create_dataframe(data_r)
+            return sorted(l.join(r, on="k", how=how).collect(), key=repr)
+
+        assert q(small) == q(plain), how
+
+
+@pytest.mark.gpu
+def test_gpu_subpartitioned_join_matches_cpu():
+    import numpy as np
+    import spark_rapids_amd as sr
+
+    rng = np.random.default_rng(7)
+    data_l = {"k": [int(v) for v in rng.integers(0, 500, 100000)],
+              "v": [float(v) for v in range(100000)]}
+    data_r = {"k": [int(v) for v in rng.integers(0, 600, 20000)],
+              "w": [float(v) for v in range(20000)]}
+    sg = sr.Session({"spark.rapids.sql.join.subPartition.targetBytes": 4096})
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+
+    def q(s, how):
+        l = s.create_dataframe(data_l)
+        r = s.create_dataframe(data_r)
+        from spark_rapids_amd import col, count_star, sum_
+
+        return (l.join(r, on="k", how=how)
+                .agg(count_star(), sum_(col("v"))).collect())
+
+    for how in ("inner", "left", "full"):
+        g, c = q(sg, how), q(sc, how)
+        assert g[0][0] == c[0][0], how
+        assert g[0][1] == pytest.approx(c[0][1], rel=1e-12), how
