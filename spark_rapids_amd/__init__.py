@@ -16,7 +16,8 @@ from .expr.aggregates import (approx_count_distinct, approx_percentile,
                               sum_, variance)
 from .expr.expressions import (CaseWhen, coalesce, col, date_add, date_sub,
                                datediff, dayofweek, greatest, hour, isin,
-                               least, lit, quarter,
+                               least, lit, quarter, to_date,
+                               unix_timestamp,
                                minute, round_, second, when)
 from .expr.windows import (dense_rank, lag, lead, nth_value, ntile,
                            rank, row_number, win_avg,

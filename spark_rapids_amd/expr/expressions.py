@@ -749,6 +749,25 @@ def date_sub(d, days) -> Expression:
     return BinaryExpr("sub", _as_expr(d), _as_expr(days))
 
 
+def to_date(ts) -> Expression:
+    """timestamp (micros) -> date32 days, floored for pre-epoch values."""
+    e = _as_expr(ts)
+    us = CastExpr(e, INT64)
+    rem = BinaryExpr("pmod", us, Literal(86_400_000_000))
+    days = BinaryExpr("int_div", BinaryExpr("sub", us, rem),
+                      Literal(86_400_000_000))
+    return CastExpr(days, DType.date32())
+
+
+def unix_timestamp(ts) -> Expression:
+    """timestamp -> whole seconds since epoch (floored)."""
+    e = _as_expr(ts)
+    us = CastExpr(e, INT64)
+    rem = BinaryExpr("pmod", us, Literal(1_000_000))
+    return BinaryExpr("int_div", BinaryExpr("sub", us, rem),
+                      Literal(1_000_000))
+
+
 def dayofweek(d) -> Expression:
     """Spark dayofweek: 1 = Sunday .. 7 = Saturday (1970-01-01 was a
     Thursday, day-number 4 in this scheme)."""

@@ -94,3 +94,19 @@ def test_udf_compiler_traces_to_expressions(session):
 
     with pytest.raises(UdfFallback):
         compile_udf(lambda a: float(a) + 1, "x")  # materializes -> fallback
+
+
+def test_to_date_unix_timestamp(session):
+    import datetime
+
+    from spark_rapids_amd import TIMESTAMP, to_date, unix_timestamp
+
+    us = [0, 1, 86_400_000_000, -1, 1_600_000_123_456_789]
+    df = session.create_dataframe({"t": us}, dtypes={"t": TIMESTAMP})
+    out = df.select(to_date(col("t")).alias("d"),
+                    unix_timestamp(col("t")).alias("u")).to_pydict()
+    for u, d, sec in zip(us, out["d"], out["u"]):
+        dt = datetime.datetime(1970, 1, 1) + \
+            datetime.timedelta(microseconds=u)
+        assert d == (dt.date() - datetime.date(1970, 1, 1)).days
+        assert sec == u // 1_000_000
