@@ -1186,6 +1186,38 @@ def _compact_dense(out_cols, allocs, keys, dense_info, ngroups, row_gid,
     return ColumnBatch(key_cols + list(agg_batch.columns), ncomp)
 
 
+def _gb_collect_str(vc: Column, row_gid: torch.Tensor, selp, n: int,
+                    ngroups: int, out_dtype, s) -> Column:
+    """collect_list over strings: stable-sort rows by group id, then one
+    string gather in group order; offsets come from the per-group valid
+    counts."""
+    if selp:
+        raise NotImplementedError("string collect under filter fusion")
+    # drop null values first (collect skips nulls) while keeping gids
+    gid_col = Column(DType.int32(), n, row_gid[:n], None, null_count=0)
+    if vc.validity is not None:
+        nn = unary_op("not", is_null(vc), DType.bool_())
+        kept = apply_boolean_mask(ColumnBatch([gid_col, vc], n), nn)
+        gid_col, vc = kept.columns[0], kept.columns[1]
+        n = kept.num_rows
+    pair = ColumnBatch([gid_col, vc], n)
+    perm = sort_order(pair, [0], [False], [False])
+    child = _gather_col(vc, perm.data, n, maybe_negative=False)
+    counts = torch.zeros(max(ngroups, 1), dtype=torch.int64, device="cuda")
+    if n:
+        ext.gb_collect_count(0, gid_col.data.data_ptr(), 0,
+                             counts.data_ptr(), n, s)
+    scanned, total = _exclusive_scan_i64(counts[:ngroups]) if ngroups \
+        else (counts, 0)
+    offs = torch.empty(ngroups + 1, dtype=torch.int32, device="cuda")
+    if ngroups:
+        ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), ngroups, s)
+    offs[ngroups] = total
+    return Column(out_dtype, ngroups, torch.zeros(0, dtype=torch.uint8,
+                                                  device="cuda"),
+                  None, offs, 0, child)
+
+
 def _gb_percentile(vc: Column, row_gid: torch.Tensor, selp, n: int,
                    ngroups: int, p: float, s) -> Column:
     """Exact percentile (linear interpolation): sort (gid, value) with
@@ -1224,6 +1256,8 @@ def _gb_collect(vc: Column, row_gid: torch.Tensor, selp, n: int,
     scatter (k_gb_collect_* in groupby.hip). collect_set dedupes first by
     grouping the (gid, value) pairs with the generic hash groupby."""
     elem_dt = out_dtype.children[0]
+    if elem_dt.id is TypeId.STRING and not is_set:
+        return _gb_collect_str(vc, row_gid, selp, n, ngroups, out_dtype, s)
     if is_set:
         if selp:
             raise NotImplementedError(

@@ -186,7 +186,8 @@ class Tagger:
                         reasons += self.expr_reasons(a.child, cs)
                         continue
                     if a.op in ("collect_list", "collect_set"):
-                        if t.id is TypeId.STRING or t.is_nested:
+                        if t.is_nested or (t.id is TypeId.STRING
+                                           and a.op == "collect_set"):
                             reasons.append(
                                 f"{a.op} over {t} not on GPU yet")
                         reasons += self.expr_reasons(a.child, cs)

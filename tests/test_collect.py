@@ -243,3 +243,22 @@ def test_gpu_explode_placement():
     tree = (df.group_by("k").agg(collect_list(col("v")))
             .explode("collect_list(v)").physical_plan().tree_string())
     assert "GpuGenerate" in tree, tree
+
+
+@pytest.mark.gpu
+def test_gpu_collect_strings_matches_cpu():
+    words = ["alpha", "b", "", "gamma-long-string", None]
+
+    def q(s):
+        import numpy as np
+
+        rng = np.random.default_rng(3)
+        df = s.create_dataframe({
+            "k": [int(v) for v in rng.integers(0, 9, 8000)],
+            "s": [words[v % 5] for v in rng.integers(0, 5, 8000)]})
+        rows = df.group_by("k").agg(collect_list(col("s"))).collect()
+        return sorted((k, sorted(l)) for k, l in rows)
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    assert q(sg) == q(sc)
