@@ -470,6 +470,20 @@ class Session:
         CSV scan with the Hive delimiter (GpuHiveTextFileFormat analogue)."""
         return self.read_csv(path, header=header, delimiter=delimiter)
 
+    def read_delta(self, path: str) -> DataFrame:
+        """Delta Lake table scan: replay the _delta_log to the live file
+        set, then scan with the parquet reader (GPU page decode)."""
+        from .io.delta import live_files
+        from .io.parquet import ParquetTable
+
+        files = live_files(path)
+        if not files:
+            raise FileNotFoundError(f"delta table has no live files: {path}")
+        t = ParquetTable.__new__(ParquetTable)
+        t.__init__(files[0])
+        t.files = files
+        return DataFrame(self, L.Scan(t, t.schema, "delta"))
+
     def read_avro(self, path: str) -> DataFrame:
         """Avro object-container scan (flat records; host decode)."""
         from .io.avro import AvroTable

@@ -217,3 +217,49 @@ def test_avro_null_codec_and_bytes(tmp_path, session):
     out = session.read_avro(p)
     assert out.count() == 500
     assert out.to_pydict()["s"][123] == "v123"
+
+
+def _make_delta(tmp_path, session):
+    import json
+    import os
+
+    root = str(tmp_path / "dtab")
+    log = os.path.join(root, "_delta_log")
+    os.makedirs(log)
+    # three parquet files; one later removed
+    for i in range(3):
+        df = session.create_dataframe({"a": [i * 10 + k for k in range(5)],
+                                       "s": [f"f{i}"] * 5})
+        session.write_parquet(df, os.path.join(root, f"part-{i}.parquet"))
+    with open(os.path.join(log, "00000000000000000000.json"), "w") as f:
+        f.write(json.dumps({"metaData": {"configuration": {}}}) + "\n")
+        for i in range(3):
+            f.write(json.dumps({"add": {"path": f"part-{i}.parquet"}}) + "\n")
+    with open(os.path.join(log, "00000000000000000001.json"), "w") as f:
+        f.write(json.dumps({"remove": {"path": "part-1.parquet"}}) + "\n")
+    return root
+
+
+def test_delta_log_replay(tmp_path, session):
+    root = _make_delta(tmp_path, session)
+    df = session.read_delta(root)
+    rows = sorted(df.collect())
+    assert len(rows) == 10  # files 0 and 2 live, file 1 removed
+    assert {r[1] for r in rows} == {"f0", "f2"}
+    out = df.agg(count_star(), sum_(col("a"))).collect()
+    assert out[0][0] == 10
+    assert out[0][1] == sum(range(0, 5)) + sum(range(20, 25))
+
+
+def test_delta_deletion_vectors_unsupported(tmp_path, session):
+    import json
+    import os
+
+    root = str(tmp_path / "dv")
+    os.makedirs(os.path.join(root, "_delta_log"))
+    with open(os.path.join(root, "_delta_log",
+                           "00000000000000000000.json"), "w") as f:
+        f.write(json.dumps({"add": {"path": "x.parquet",
+                                    "deletionVector": {"id": 1}}}) + "\n")
+    with pytest.raises(NotImplementedError):
+        session.read_delta(root)
