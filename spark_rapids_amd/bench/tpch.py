@@ -145,6 +145,33 @@ def q5(t):
             .sort("revenue", descending=True))
 
 
+def q3(t):
+    """TPC-H Q3 shipping priority (top unshipped orders by revenue)."""
+    li, o, c = t["lineitem"], t["orders"], t["customer"]
+    rev = col("l_extendedprice") * (lit(1.0) - col("l_discount"))
+    return (li.filter(col("l_shipdate") > 9204)       # > 1995-03-15
+            .join(o, on="l_orderkey", right_on=["o_orderkey"])
+            .filter(col("o_orderdate") < 9204)
+            .join(c, on="o_custkey", right_on=["c_custkey"])
+            .filter(col("c_nationkey") < 5)           # BUILDING-segment proxy
+            .group_by("l_orderkey", "o_orderdate")
+            .agg(sum_(rev).alias("revenue"))
+            .sort("revenue", descending=True)
+            .limit(10))
+
+
+def q6(t):
+    """TPC-H Q6 forecasting revenue change (selective filter + agg)."""
+    li = t["lineitem"]
+    return (li.filter((col("l_shipdate") >= 8766)      # [1994-01-01,
+                      & (col("l_shipdate") < 9131)     #  1995-01-01)
+                      & (col("l_discount") >= 0.05)
+                      & (col("l_discount") <= 0.07)
+                      & (col("l_quantity") < 24.0))
+            .agg(sum_(col("l_extendedprice") * col("l_discount"))
+                 .alias("revenue")))
+
+
 def run(sf: float, partitions: int, gpu: bool, iters: int = 3):
     session = Session({"spark.rapids.sql.enabled": gpu})
     t0 = time.perf_counter()
@@ -152,7 +179,7 @@ def run(sf: float, partitions: int, gpu: bool, iters: int = 3):
     gen_s = time.perf_counter() - t0
     out = {"sf": sf, "gen_seconds": round(gen_s, 2), "gpu": gpu,
            "queries": {}}
-    for name, fn in (("q1", q1), ("q5", q5)):
+    for name, fn in (("q1", q1), ("q3", q3), ("q5", q5), ("q6", q6)):
         fn(tables).collect()  # warmup
         times = []
         for _ in range(iters):
