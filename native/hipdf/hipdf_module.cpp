@@ -154,6 +154,11 @@ void hipdf_str_like(const void*, const void*, const void*, int, void*,
                     int64_t, hipStream_t);
 void hipdf_str_length(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_str_case(int, const void*, void*, int64_t, hipStream_t);
+void hipdf_str_split_count(const void*, const void*, const void*, int,
+                           void*, int64_t, hipStream_t);
+void hipdf_str_split_fill(const void*, const void*, const void*, int,
+                          const void*, const void*, void*, void*, int64_t,
+                          hipStream_t);
 void hipdf_str_trim_ranges(int, const void*, const void*, void*, void*,
                            int64_t, hipStream_t);
 void hipdf_str_concat2(const void*, const void*, const void*, const void*,
@@ -572,6 +577,21 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("str_case", [](bool upper, int64_t in, int64_t out, int64_t nbytes,
                        int64_t stream) {
     hipdf_str_case(upper, P(in), PM(out), nbytes, S(stream));
+    check_async();
+  });
+  m.def("str_split_count", [](int64_t ao, int64_t ab, int64_t delim,
+                              int dlen, int64_t counts, int64_t n,
+                              int64_t stream) {
+    hipdf_str_split_count(P(ao), P(ab), P(delim), dlen, PM(counts), n,
+                          S(stream));
+    check_async();
+  });
+  m.def("str_split_fill", [](int64_t ao, int64_t ab, int64_t delim,
+                             int dlen, int64_t part_off, int64_t counts,
+                             int64_t out_ss, int64_t out_sl, int64_t n,
+                             int64_t stream) {
+    hipdf_str_split_fill(P(ao), P(ab), P(delim), dlen, P(part_off),
+                         P(counts), PM(out_ss), PM(out_sl), n, S(stream));
     check_async();
   });
   m.def("str_trim_ranges", [](int mode, int64_t ao, int64_t ab,

@@ -147,6 +147,78 @@ __global__ void k_str_case(int upper, const uint8_t* __restrict__ in,
 // substring (1-based start in codepoints, length in codepoints; Spark
 // semantics: start 0 behaves like 1, negative counts from the end).
 // pass 1: byte [start,len) per row
+// split by a literal delimiter -> per-part (start, len) spans.
+// Java limit-0 semantics: trailing empty parts are dropped ("a,,".split
+// -> ["a"], ",,".split -> [], "".split -> [""]).
+__global__ void k_str_split_count(const int32_t* __restrict__ ao,
+                                  const uint8_t* __restrict__ ab,
+                                  const uint8_t* __restrict__ delim,
+                                  int dlen, int64_t* __restrict__ counts,
+                                  int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t s0 = ao[i], s1 = ao[i + 1];
+    if (s1 == s0) {
+      counts[i] = 1;  // "" -> [""]
+      continue;
+    }
+    int part = 0;
+    int last_nonempty = -1;
+    int32_t start = s0;
+    for (int32_t p = s0; p + dlen <= s1;) {
+      bool m = true;
+      for (int k = 0; k < dlen; ++k)
+        if (ab[p + k] != delim[k]) { m = false; break; }
+      if (m) {
+        if (p > start) last_nonempty = part;
+        ++part;
+        p += dlen;
+        start = p;
+      } else {
+        ++p;
+      }
+    }
+    if (s1 > start) last_nonempty = part;
+    counts[i] = last_nonempty + 1;
+  }
+}
+
+__global__ void k_str_split_fill(const int32_t* __restrict__ ao,
+                                 const uint8_t* __restrict__ ab,
+                                 const uint8_t* __restrict__ delim, int dlen,
+                                 const int64_t* __restrict__ part_off,
+                                 const int64_t* __restrict__ counts,
+                                 int32_t* __restrict__ out_ss,
+                                 int64_t* __restrict__ out_sl, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t s0 = ao[i], s1 = ao[i + 1];
+    int64_t want = counts[i];
+    int64_t w = part_off[i];
+    int64_t part = 0;
+    int32_t start = s0;
+    for (int32_t p = s0; part < want && p + dlen <= s1;) {
+      bool m = true;
+      for (int k = 0; k < dlen; ++k)
+        if (ab[p + k] != delim[k]) { m = false; break; }
+      if (m) {
+        out_ss[w] = start;
+        out_sl[w] = p - start;
+        ++w;
+        ++part;
+        p += dlen;
+        start = p;
+      } else {
+        ++p;
+      }
+    }
+    if (part < want) {
+      out_ss[w] = start;
+      out_sl[w] = s1 - start;
+    }
+  }
+}
+
 // trim spans: mode 0 both, 1 leading, 2 trailing (ascii space like Spark
 // trim's default)
 __global__ void k_str_trim_ranges(int mode, const int32_t* __restrict__ ao,
@@ -284,6 +356,25 @@ void hipdf_str_case(int upper, const void* in, void* out, int64_t nbytes,
                     hipStream_t stream) {
   hipLaunchKernelGGL(k_str_case, flat_grid(nbytes), dim3(HIPDF_BLOCK), 0,
                      stream, upper, (const uint8_t*)in, (uint8_t*)out, nbytes);
+}
+
+void hipdf_str_split_count(const void* ao, const void* ab,
+                           const void* delim, int dlen, void* counts,
+                           int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_split_count, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)ao, (const uint8_t*)ab,
+                     (const uint8_t*)delim, dlen, (int64_t*)counts, n);
+}
+
+void hipdf_str_split_fill(const void* ao, const void* ab, const void* delim,
+                          int dlen, const void* part_off, const void* counts,
+                          void* out_ss, void* out_sl, int64_t n,
+                          hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_split_fill, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)ao, (const uint8_t*)ab,
+                     (const uint8_t*)delim, dlen, (const int64_t*)part_off,
+                     (const int64_t*)counts, (int32_t*)out_ss,
+                     (int64_t*)out_sl, n);
 }
 
 void hipdf_str_trim_ranges(int mode, const void* ao, const void* ab,

@@ -145,6 +145,15 @@ class Expression:
         resc = replacement.replace("\\", "\\\\").replace("$", "\\$")
         return RegexpReplace(self, esc, resc)
 
+    def split(self, delimiter: str) -> "StrSplit":
+        """Split into a LIST of strings (java limit-0 semantics: trailing
+        empty parts dropped)."""
+        return StrSplit(self, delimiter)
+
+    def size(self) -> "ArraySize":
+        """Element count of a LIST value (null list -> null)."""
+        return ArraySize(self)
+
     def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
         return RegexpExtract(self, pattern, group)
 
@@ -555,6 +564,49 @@ class Substring(Expression):
 # ---------------------------------------------------------------------------
 # public DSL
 # ---------------------------------------------------------------------------
+
+class StrSplit(Expression):
+    """split(str, delim) -> array<string>. GPU kernel handles literal
+    delimiters (k_str_split_* in strings.hip); regex delimiters run on
+    the CPU via re.split (GpuStringSplit analogue)."""
+
+    def __init__(self, child: Expression, delimiter: str):
+        self.child = child
+        self.delimiter = delimiter
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return DType.list_(STRING)
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.str_split(self.child.eval(batch, schema), self.delimiter)
+
+    def __str__(self):
+        return f"split({self.child}, {self.delimiter!r})"
+
+
+class ArraySize(Expression):
+    """size(array) -> int32; null array -> null (GpuSize analogue)."""
+
+    def __init__(self, child: Expression):
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return INT32
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.array_size(self.child.eval(batch, schema))
+
+    def __str__(self):
+        return f"size({self.child})"
+
 
 class RegexpExtract(Expression):
     """regexp_extract(str, pattern, idx): the capture group's text for the

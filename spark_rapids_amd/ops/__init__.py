@@ -46,6 +46,14 @@ def binary_op_scalar(op: str, lhs: Column, scalar, out_dtype: DType) -> Column:
     return backend_for(lhs).binary_op_scalar(op, lhs, scalar, out_dtype)
 
 
+def str_split(col: Column, delimiter: str) -> Column:
+    return backend_for(col).str_split(col, delimiter)
+
+
+def array_size(col: Column) -> Column:
+    return backend_for(col).array_size(col)
+
+
 def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return backend_for(col).regexp_extract(col, pattern, group)
 

@@ -494,6 +494,32 @@ def str_predicate(op: str, col: Column, pattern: str) -> Column:
     return _make(res, av if not av.all() else None, DType.bool_())
 
 
+def str_split(col: Column, delimiter: str) -> Column:
+    import re as _re
+
+    rx = _re.compile(delimiter)
+    out = []
+    for v in col.to_pylist():
+        if v is None:
+            out.append(None)
+            continue
+        if v == "":
+            out.append([""])
+            continue
+        parts = rx.split(v)
+        while parts and parts[-1] == "":
+            parts.pop()
+        out.append(parts)
+    return Column.from_pylist(out, DType.list_(DType.string()))
+
+
+def array_size(col: Column) -> Column:
+    valid = _valid(col)
+    offs = col.offsets.numpy()
+    sizes = (offs[1:] - offs[:-1]).astype(np.int32)
+    return _make(sizes, valid if not valid.all() else None, DType.int32())
+
+
 def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     import re as _re
 

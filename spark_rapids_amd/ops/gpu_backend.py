@@ -464,6 +464,60 @@ def str_concat(a: Column, b: Column) -> Column:
                   null_count=None if v is not None else 0)
 
 
+def str_split(col: Column, delimiter: str) -> Column:
+    """Literal-delimiter split into LIST<STRING> (see StrSplit)."""
+    n = col.size
+    s = _stream()
+    lt = DType.list_(DType.string())
+    if n == 0:
+        return Column.from_pylist([], lt).cuda()
+    d = _pattern_tensor(delimiter)
+    counts = torch.empty(n, dtype=torch.int64, device="cuda")
+    ext.str_split_count(col.offsets.data_ptr(), col.data.data_ptr(),
+                        d.data_ptr(), d.numel(), counts.data_ptr(), n, s)
+    part_off, total = _exclusive_scan_i64(counts)
+    ss = torch.empty(max(total, 1), dtype=torch.int32, device="cuda")[:total]
+    sl = torch.empty(max(total, 1), dtype=torch.int64, device="cuda")[:total]
+    if total:
+        ext.str_split_fill(col.offsets.data_ptr(), col.data.data_ptr(),
+                           d.data_ptr(), d.numel(), part_off.data_ptr(),
+                           counts.data_ptr(), ss.data_ptr(), sl.data_ptr(),
+                           n, s)
+    # child strings from (start, len) spans
+    scanned, nbytes = _exclusive_scan_i64(sl) if total else (sl, 0)
+    out_bytes = torch.empty(max(nbytes, 1), dtype=torch.uint8,
+                            device="cuda")[:nbytes]
+    if nbytes:
+        ext.substr_copy(col.data.data_ptr(), ss.data_ptr(), sl.data_ptr(),
+                        scanned.data_ptr(), out_bytes.data_ptr(), total, s)
+    coffs = torch.empty(total + 1, dtype=torch.int32, device="cuda")
+    if total:
+        ext.narrow_i64_i32(scanned.data_ptr(), coffs.data_ptr(), total, s)
+    coffs[total] = nbytes
+    child = Column(DType.string(), total, out_bytes, None, coffs, 0)
+    loffs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    ext.narrow_i64_i32(part_off.data_ptr(), loffs.data_ptr(), n, s)
+    loffs[n] = total
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(lt, n, torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                  v, loffs, col._null_count, child)
+
+
+def array_size(col: Column) -> Column:
+    n = col.size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.int32())
+    out = torch.empty(n, dtype=torch.int32, device="cuda")
+    # sizes = offsets[i+1] - offsets[i] via the int32 sub kernel on views
+    a = Column(DType.int32(), n, col.offsets[1:], None, null_count=0)
+    b = Column(DType.int32(), n, col.offsets[:n], None, null_count=0)
+    diff = _binary("sub", a, b, None, DType.int32())
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(DType.int32(), n, diff.data, v,
+                  null_count=col._null_count)
+
+
 def substring(col: Column, pos: int, length: int = -1) -> Column:
     n = col.size
     s = _stream()

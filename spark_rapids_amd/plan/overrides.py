@@ -129,6 +129,11 @@ class Tagger:
                     compile_regex(e.pattern)
                 except RegexUnsupported as ex:
                     out.append(f"regex not supported on GPU: {ex}")
+        elif type(e).__name__ == "StrSplit":
+            if any(c in e.delimiter for c in ".\\+*?()[]{}|^$"):
+                out.append("regex split delimiters run on CPU")
+        elif type(e).__name__ == "ArraySize":
+            pass
         elif type(e).__name__ in ("RegexpExtract", "RegexpReplace"):
             from ..ops.regex_compiler import RegexUnsupported, compile_regex
 

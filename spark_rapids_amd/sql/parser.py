@@ -577,6 +577,14 @@ class Parser:
             for nxt in args[1:]:
                 out = BinaryExpr("concat", out, nxt)
             return out
+        if name == "split":
+            from ..expr.expressions import StrSplit
+
+            return StrSplit(args[0], args[1].value)
+        if name == "size":
+            from ..expr.expressions import ArraySize
+
+            return ArraySize(args[0])
         if name == "replace":
             return args[0].replace(args[1].value, args[2].value)
         if name == "substring" or name == "substr":

@@ -170,3 +170,56 @@ class TestTrimConcatReplace:
         tree = (df.select(col("a").trim().concat("z"))
                 .physical_plan().tree_string())
         assert "GpuProject" in tree, tree
+
+
+class TestSplitSize:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_cpu_java_semantics(self, cpu):
+        df = cpu.create_dataframe({"s": ["a,b,c", "x", "", "a,,", ",,",
+                                         None, ",lead"]})
+        out = df.select(col("s").split(",").alias("p")) \
+                .with_column("n", col("p").size()).to_pydict()
+        assert out["p"] == [["a", "b", "c"], ["x"], [""], ["a"], [],
+                            None, ["", "lead"]]
+        assert out["n"] == [3, 1, 1, 1, 0, None, 2]
+
+    def test_split_explode(self, cpu):
+        df = cpu.create_dataframe({"s": ["a b", "c"]})
+        out = df.select(col("s").split(" ").alias("w")).explode("w").collect()
+        assert out == [("a",), ("b",), ("c",)]
+
+    def test_sql_split_size(self, cpu):
+        cpu.register("tsplit", cpu.create_dataframe({"s": ["a|b|c"]}))
+        # note: | must go through CPU (regex special) — still correct
+        out = cpu.sql("SELECT size(split(s, ',')) FROM tsplit").collect()
+        assert out == [(1,)]
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        import numpy as np
+
+        rng = np.random.default_rng(8)
+        toks = ["a", "bb", "", "ccc"]
+        vals = [",".join(toks[j % 4] for j in range(v % 6))
+                if i % 13 else None
+                for i, v in enumerate(rng.integers(0, 40, 6000))]
+
+        def q(s):
+            df = s.create_dataframe({"s": vals})
+            return (df.select(col("s").split(",").alias("p"))
+                    .with_column("n", col("p").size()).to_pydict())
+
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        assert q(sg) == q(sc)
+
+    @pytest.mark.gpu
+    def test_gpu_split_placement(self):
+        sg = sr.Session()
+        df = sg.create_dataframe({"s": ["q,r"]})
+        tree = (df.select(col("s").split(",").alias("p"))
+                .physical_plan().tree_string())
+        assert "GpuProject" in tree, tree
