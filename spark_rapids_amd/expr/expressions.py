@@ -150,6 +150,11 @@ class Expression:
         empty parts dropped)."""
         return StrSplit(self, delimiter)
 
+    def element_at(self, index: int) -> "ElementAt":
+        """1-based element of a LIST value; NULL beyond the list length
+        (Spark element_at)."""
+        return ElementAt(self, index)
+
     def size(self) -> "ArraySize":
         """Element count of a LIST value (null list -> null)."""
         return ArraySize(self)
@@ -586,6 +591,29 @@ class StrSplit(Expression):
 
     def __str__(self):
         return f"split({self.child}, {self.delimiter!r})"
+
+
+class ElementAt(Expression):
+    """element_at(array, k) with 1-based k (negative k counts from the
+    end); NULL when |k| exceeds the length (GpuElementAt analogue)."""
+
+    def __init__(self, child: Expression, index: int):
+        assert index != 0, "element_at is 1-based (Spark semantics)"
+        self.child = child
+        self.index = index
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return self.child.dtype(schema).children[0]
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.element_at(self.child.eval(batch, schema), self.index)
+
+    def __str__(self):
+        return f"element_at({self.child}, {self.index})"
 
 
 class ArraySize(Expression):

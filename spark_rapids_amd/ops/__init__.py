@@ -54,6 +54,10 @@ def array_size(col: Column) -> Column:
     return backend_for(col).array_size(col)
 
 
+def element_at(col: Column, index: int) -> Column:
+    return backend_for(col).element_at(col, index)
+
+
 def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return backend_for(col).regexp_extract(col, pattern, group)
 

@@ -520,6 +520,18 @@ def array_size(col: Column) -> Column:
     return _make(sizes, valid if not valid.all() else None, DType.int32())
 
 
+def element_at(col: Column, index: int) -> Column:
+    out = []
+    for v in col.to_pylist():
+        if v is None:
+            out.append(None)
+        elif index > 0:
+            out.append(v[index - 1] if index <= len(v) else None)
+        else:
+            out.append(v[index] if -index <= len(v) else None)
+    return Column.from_pylist(out, col.dtype.children[0])
+
+
 def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     import re as _re
 

@@ -503,6 +503,32 @@ def str_split(col: Column, delimiter: str) -> Column:
                   v, loffs, col._null_count, child)
 
 
+def element_at(col: Column, index: int) -> Column:
+    """Gather child elements at per-row offset +- index with bounds
+    nullification (reuses the negative-index gather)."""
+    n = col.size
+    s = _stream()
+    elem_dt = col.dtype.children[0]
+    if n == 0:
+        return _empty_col(elem_dt)
+    starts = Column(DType.int32(), n, col.offsets[:n], None, null_count=0)
+    ends = Column(DType.int32(), n, col.offsets[1:], None, null_count=0)
+    if index > 0:
+        idx = binary_op_scalar("add", starts, index - 1, DType.int32())
+        ok = binary_op("lt", idx, ends, DType.bool_())
+    else:
+        idx = binary_op_scalar("add", ends, index, DType.int32())
+        ok = binary_op("ge", idx, starts, DType.bool_())
+    if col.validity is not None:
+        # null list -> null element: fold into the gather index (-1 rows
+        # nullify) so string elements need no typed if_else
+        nn = unary_op("not", is_null(col), DType.bool_())
+        ok = binary_op("and", ok, nn, DType.bool_())
+    neg1 = Column.full(-1, DType.int32(), n, "cuda")
+    srcm = if_else(ok, idx, neg1)
+    return _gather_col(col.child, srcm.data, n, maybe_negative=True)
+
+
 def array_size(col: Column) -> Column:
     n = col.size
     s = _stream()

@@ -132,7 +132,7 @@ class Tagger:
         elif type(e).__name__ == "StrSplit":
             if any(c in e.delimiter for c in ".\\+*?()[]{}|^$"):
                 out.append("regex split delimiters run on CPU")
-        elif type(e).__name__ == "ArraySize":
+        elif type(e).__name__ in ("ArraySize", "ElementAt"):
             pass
         elif type(e).__name__ in ("RegexpExtract", "RegexpReplace"):
             from ..ops.regex_compiler import RegexUnsupported, compile_regex

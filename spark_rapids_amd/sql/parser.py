@@ -581,6 +581,10 @@ class Parser:
             from ..expr.expressions import StrSplit
 
             return StrSplit(args[0], args[1].value)
+        if name == "element_at":
+            from ..expr.expressions import ElementAt
+
+            return ElementAt(args[0], int(args[1].value))
         if name == "size":
             from ..expr.expressions import ArraySize
 

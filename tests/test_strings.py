@@ -223,3 +223,36 @@ class TestSplitSize:
         tree = (df.select(col("s").split(",").alias("p"))
                 .physical_plan().tree_string())
         assert "GpuProject" in tree, tree
+
+
+class TestElementAt:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_cpu(self, cpu):
+        df = cpu.create_dataframe({"s": ["a,b,c", "x", None, ""]})
+        out = (df.select(col("s").split(",").alias("p"))
+               .select(col("p").element_at(1).alias("f"),
+                       col("p").element_at(-1).alias("l"),
+                       col("p").element_at(5).alias("m")).to_pydict())
+        assert out["f"] == ["a", "x", None, ""]
+        assert out["l"] == ["c", "x", None, ""]
+        assert out["m"] == [None] * 4
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        vals = [",".join(f"t{j}" for j in range(i % 5)) if i % 7 else None
+                for i in range(3000)]
+
+        def q(s):
+            df = s.create_dataframe({"s": vals})
+            return (df.select(col("s").split(",").alias("p"))
+                    .select(col("p").element_at(1).alias("a"),
+                            col("p").element_at(2).alias("b"),
+                            col("p").element_at(-1).alias("z"))
+                    .to_pydict())
+
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        assert q(sg) == q(sc)
