@@ -108,6 +108,8 @@ void hipdf_gb_sum_i64_to_i128(const void*, const void*, const void*,
                               const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_gb_sum_i128(const void*, const void*, const void*, const void*,
                        void*, void*, int64_t, hipStream_t);
+void hipdf_win_minmax(int, const void*, int, const void*, const void*,
+                      int, void*, int64_t, hipStream_t);
 void hipdf_range_bounds(const void*, const void*, const void*, double,
                         double, int, int, void*, void*, int64_t,
                         hipStream_t);
@@ -426,6 +428,13 @@ PYBIND11_MODULE(hipdf, m) {
     check_async();
   });
 
+  m.def("win_minmax", [](int is_double, int64_t level_ptrs, int nlevels,
+                         int64_t a_idx, int64_t b_idx, int is_min,
+                         int64_t out, int64_t n, int64_t stream) {
+    hipdf_win_minmax(is_double, P(level_ptrs), nlevels, P(a_idx), P(b_idx),
+                     is_min, PM(out), n, S(stream));
+    check_async();
+  });
   m.def("range_bounds", [](int64_t vals, int64_t seg_start, int64_t seg_end,
                            double lo, double hi, int lo_unb, int hi_unb,
                            int64_t a_idx, int64_t b_idx, int64_t n,

@@ -117,7 +117,50 @@ __global__ void k_range_bounds(const double* __restrict__ vals,
   }
 }
 
+// sparse-table range min/max query: levels[l][i] covers [i, i+2^l-1];
+// answer = op(levels[k][a], levels[k][b-2^k+1]) with k = floor(log2(len)).
+// Frames never cross partition boundaries, so the table builds globally.
+template <typename T>
+__global__ void k_win_minmax(const int64_t* __restrict__ level_ptrs,
+                             int nlevels, const int32_t* __restrict__ a_idx,
+                             const int32_t* __restrict__ b_idx, int is_min,
+                             T* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t a = a_idx[i], b = b_idx[i];
+    if (b < a) {
+      out[i] = (T)0;
+      continue;
+    }
+    int64_t len = (int64_t)b - a + 1;
+    int k = 63 - __clzll((unsigned long long)len);
+    if (k >= nlevels) k = nlevels - 1;
+    const T* lv = (const T*)level_ptrs[k];
+    T x = lv[a];
+    T y = lv[b - (1ll << k) + 1];
+    out[i] = is_min ? (x < y ? x : y) : (x > y ? x : y);
+  }
+}
+
 extern "C" {
+
+void hipdf_win_minmax(int is_double, const void* level_ptrs, int nlevels,
+                      const void* a_idx, const void* b_idx, int is_min,
+                      void* out, int64_t n, hipStream_t stream) {
+  if (is_double)
+    hipLaunchKernelGGL((k_win_minmax<double>), flat_grid(n),
+                       dim3(HIPDF_BLOCK), 0, stream,
+                       (const int64_t*)level_ptrs, nlevels,
+                       (const int32_t*)a_idx, (const int32_t*)b_idx, is_min,
+                       (double*)out, n);
+  else
+    hipLaunchKernelGGL((k_win_minmax<int64_t>), flat_grid(n),
+                       dim3(HIPDF_BLOCK), 0, stream,
+                       (const int64_t*)level_ptrs, nlevels,
+                       (const int32_t*)a_idx, (const int32_t*)b_idx, is_min,
+                       (int64_t*)out, n);
+}
+
 
 void hipdf_range_bounds(const void* vals, const void* seg_start,
                         const void* seg_end, double lo, double hi,

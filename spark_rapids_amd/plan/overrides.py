@@ -241,7 +241,7 @@ class Tagger:
                     continue
                 if w.spec.range_between is not None:
                     okt = cs.field(w.spec.order_by[0]).dtype
-                    if op not in ("sum", "count", "mean"):
+                    if op not in ("sum", "count", "mean", "min", "max"):
                         reasons.append(
                             f"range-frame {op} window not on GPU yet")
                     elif w.spec.descending[0]:
@@ -268,14 +268,8 @@ class Tagger:
                         reasons.append(f"window {op}({vt}) not on GPU")
                     continue
                 if op in ("min", "max"):
-                    if spec.rows_between is not None \
-                            or spec.range_between is not None:
-                        reasons.append(
-                            f"bounded {op} window has no GPU kernel yet")
-                    elif spec.order_by:
-                        reasons.append(
-                            f"running {op} window has no GPU kernel yet")
-                    elif vt is not None and not vt.is_numeric:
+                    if vt is not None and (not vt.is_numeric
+                                           or vt.id is TypeId.DECIMAL128):
                         reasons.append(f"window {op}({vt}) not on GPU")
                     continue
                 reasons.append(f"window function {op} not on GPU")

@@ -285,8 +285,9 @@ class WindowExec(PhysicalExec):
         nn_col = Column(DType.bool_(), n, valid_u8, None, null_count=0)
         rb = self.spec.rows_between
         rgb = self.spec.range_between
-        if (rb is not None or rgb is not None) \
-                and op in ("sum", "count", "mean"):
+        framed_minmax = running and op in ("min", "max")
+        if (rb is not None or rgb is not None or framed_minmax) \
+                and op in ("sum", "count", "mean", "min", "max"):
             hp_ext = torch.cat([head_pos, torch.tensor(
                 [n], dtype=torch.int32, device="cuda")])
             segid_next = gb.binary_op_scalar(
@@ -305,7 +306,7 @@ class WindowExec(PhysicalExec):
                     "max", gb.binary_op_scalar("add", iota_col, lo_off,
                                                INT32), seg_start_col,
                     INT32)
-            else:
+            elif rgb is not None:
                 # RANGE frame: per-row binary search of the ascending order
                 # key within the segment (k_range_bounds)
                 lo_v, hi_v = rgb
@@ -324,7 +325,13 @@ class WindowExec(PhysicalExec):
                                  a_t.data_ptr(), b_t.data_ptr(), n, s)
                 a_idx = _i32col(a_t)
                 b_idx = _i32col(b_t)
+            else:
+                # running min/max: frame = [segment start, current row]
+                a_idx, b_idx = seg_start_col, iota_col
             am1 = gb.binary_op_scalar("sub", a_idx, 1, INT32)
+            if op in ("min", "max"):
+                return self._framed_minmax(op, vc, nn_col, a_idx, b_idx,
+                                           am1, n, s)
             use_f64 = out_dt.is_floating
             work_t = FLOAT64 if use_f64 else INT64
             v64 = gb.cast(Column(vc.dtype, n, vc.data, None, null_count=0),
@@ -426,6 +433,69 @@ class WindowExec(PhysicalExec):
                 return gb.binary_op("div", gb.cast(val, FLOAT64), cf, FLOAT64)
             return gb.cast(val, out_dt) if work_t != out_dt else val
         raise NotImplementedError(f"gpu window {op}")
+
+    def _framed_minmax(self, op, vc, nn_col, a_idx, b_idx, am1, n, s):
+        """Bounded / running min-max via a sparse table: log2(n) build
+        passes with the elementwise min/max kernel, then one range-query
+        kernel gather per row (k_win_minmax). NULL values enter as
+        +-infinity sentinels; output validity is the frame's valid count."""
+        import torch
+
+        from ..column import mask_nbytes
+        from ..ops import gpu_backend as gb
+        from ..ops.gpu_backend import ext
+        from ..types import FLOAT64, INT64, DType
+
+        is_f = vc.dtype.is_floating
+        work_t = FLOAT64 if is_f else INT64
+        v64 = gb.cast(Column(vc.dtype, n, vc.data, None, null_count=0),
+                      work_t)
+        if is_f:
+            sent = float("inf") if op == "min" else float("-inf")
+        else:
+            sent = (1 << 63) - 1 if op == "min" else -(1 << 63)
+        vz = gb.if_else(nn_col, v64,
+                        Column.full(sent, work_t, n, "cuda"))
+        levels = [vz.data]
+        span = 1
+        while span < n:
+            prev = levels[-1]
+            shifted = torch.empty_like(prev)
+            shifted[: n - span] = prev[span:]
+            shifted[n - span:] = prev[n - span:]
+            nxt = gb.binary_op(op, Column(work_t, n, prev, None,
+                                          null_count=0),
+                               Column(work_t, n, shifted, None,
+                                      null_count=0), work_t)
+            levels.append(nxt.data)
+            span *= 2
+        ptrs = torch.tensor([t.data_ptr() for t in levels],
+                            dtype=torch.int64).cuda()
+        out = torch.empty(n, dtype=torch.float64 if is_f else torch.int64,
+                          device="cuda")
+        ext.win_minmax(1 if is_f else 0, ptrs.data_ptr(), len(levels),
+                       a_idx.data.data_ptr(), b_idx.data.data_ptr(),
+                       1 if op == "min" else 0, out.data_ptr(), n, s)
+        # validity: number of valid values in the frame > 0
+        nn64 = gb.cast(nn_col, INT64)
+        excl_n, _ = gb._exclusive_scan_i64(nn64.data)
+        incl_n = gb.binary_op("add", Column(INT64, n, excl_n, None,
+                                            null_count=0), nn64, INT64)
+        cnt_b = Column(INT64, n, self._gather_i32(incl_n.data, b_idx.data,
+                                                  n), None, null_count=0)
+        cnt_a = Column(INT64, n, self._gather_i32(incl_n.data, am1.data, n),
+                       None, null_count=0)
+        cnt_r = gb.binary_op("sub", cnt_b, cnt_a, INT64)
+        cnt_r = gb.binary_op_scalar("max", cnt_r, 0, INT64)
+        ov = torch.empty(mask_nbytes(n), dtype=torch.uint8, device="cuda")
+        ext.mask_from_nonzero(cnt_r.data.data_ptr(), ov.data_ptr(), n, s)
+        from ..types import TypeId
+
+        if vc.dtype == work_t or vc.dtype.id is TypeId.DECIMAL64:
+            # decimal64 backing is the same scaled int64 — no rescale
+            return Column(vc.dtype, n, out, ov, null_count=None)
+        res = Column(work_t, n, out, ov, null_count=None)
+        return gb.cast(res, vc.dtype)
 
     def _running_sum_i64(self, vz: "torch.Tensor", heads, seg_start_col, n):
         import torch

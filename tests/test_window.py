@@ -312,3 +312,37 @@ def test_gpu_ntile_nth_matches_cpu():
     sg = sr.Session()
     sc = sr.Session({"spark.rapids.sql.enabled": False})
     assert sorted(q(sg)) == sorted(q(sc))
+
+
+@pytest.mark.gpu
+def test_gpu_minmax_windows_match_cpu():
+    from spark_rapids_amd import win_max, win_min
+
+    rng = np.random.default_rng(23)
+    n = 20000
+    data = {"p": [int(v) for v in rng.integers(0, 40, n)],
+            "t": [int(v) for v in rng.integers(0, 10**6, n)],
+            "v": [float(v) if v % 11 else None
+                  for v in rng.integers(-1000, 1000, n)],
+            "i": [int(v) for v in rng.integers(-50, 50, n)]}
+
+    def q(s):
+        df = s.create_dataframe(data)
+        return (df
+                .with_column("rmin", win_min(col("v")).over(["p"], ["t"]))
+                .with_column("bmax", win_max(col("v")).over(
+                    ["p"], ["t"], rows_between=(-3, 1)))
+                .with_column("imin", win_min(col("i")).over(
+                    ["p"], ["t"], rows_between=(-5, 0)))
+                .with_column("gmax", win_max(col("v")).over(["p"]))
+                .to_pydict())
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    g, c = q(sg), q(sc)
+    for k in ("rmin", "bmax", "imin", "gmax"):
+        assert len(g[k]) == len(c[k])
+        for a, b in zip(g[k], c[k]):
+            assert (a is None) == (b is None), k
+            if a is not None:
+                assert a == pytest.approx(b), k
