@@ -28,8 +28,9 @@ _EXECS = [
      "buckets", "fixed-width keys on GPU; string/decimal128 sort keys "
      "fall back"),
     ("Expand", "grouping-sets projections (rollup / cube)", "all"),
-    ("Window", "ranking / running + bounded + partition aggregates / "
-     "lag / lead", "GPU segmented scans; running/bounded min-max on CPU"),
+    ("Window", "ranking / running + bounded (ROWS and RANGE) + partition "
+     "aggregates / lag / lead / ntile / nth_value",
+     "GPU segmented scans + sparse-table min-max"),
     ("Limit", "row limit", "all"),
     ("Union", "concat", "all"),
     ("Sample", "deterministic murmur3 Bernoulli sample", "all"),

@@ -263,3 +263,28 @@ def test_delta_deletion_vectors_unsupported(tmp_path, session):
                                     "deletionVector": {"id": 1}}}) + "\n")
     with pytest.raises(NotImplementedError):
         session.read_delta(root)
+
+
+@pytest.mark.gpu
+def test_gpu_orc_multi_stripe(tmp_path):
+    import pyarrow.orc as paorc
+
+    rng = np.random.default_rng(5)
+    n = 60_000
+    t = pa.table({
+        "a": pa.array([int(v) if i % 9 else None for i, v in
+                       enumerate(rng.integers(-10**9, 10**9, n))],
+                      pa.int64()),
+        "s": pa.array([f"r{v}" for v in range(n)]),
+    })
+    p = str(tmp_path / "ms.orc")
+    paorc.write_table(t, p, stripe_size=64 * 1024,
+                      dictionary_key_size_threshold=0.0)
+    sg = sr.Session()
+    meta_stripes = len(__import__(
+        "spark_rapids_amd.io.orc_meta", fromlist=["read_meta"]
+    ).read_meta(p).stripes)
+    assert meta_stripes > 1, meta_stripes
+    got = sg.read_orc(p).to_pydict()
+    assert got["a"] == t.column("a").to_pylist()
+    assert got["s"] == t.column("s").to_pylist()
