@@ -156,6 +156,10 @@ void hipdf_str_like(const void*, const void*, const void*, int, void*,
                     int64_t, hipStream_t);
 void hipdf_str_length(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_str_case(int, const void*, void*, int64_t, hipStream_t);
+void hipdf_str_initcap(const void*, const void*, void*, int64_t,
+                       hipStream_t);
+void hipdf_str_reverse(const void*, const void*, void*, int64_t,
+                       hipStream_t);
 void hipdf_str_split_count(const void*, const void*, const void*, int,
                            void*, int64_t, hipStream_t);
 void hipdf_str_split_fill(const void*, const void*, const void*, int,
@@ -586,6 +590,16 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("str_case", [](bool upper, int64_t in, int64_t out, int64_t nbytes,
                        int64_t stream) {
     hipdf_str_case(upper, P(in), PM(out), nbytes, S(stream));
+    check_async();
+  });
+  m.def("str_initcap", [](int64_t ao, int64_t ab, int64_t out, int64_t n,
+                          int64_t stream) {
+    hipdf_str_initcap(P(ao), P(ab), PM(out), n, S(stream));
+    check_async();
+  });
+  m.def("str_reverse", [](int64_t ao, int64_t ab, int64_t out, int64_t n,
+                          int64_t stream) {
+    hipdf_str_reverse(P(ao), P(ab), PM(out), n, S(stream));
     check_async();
   });
   m.def("str_split_count", [](int64_t ao, int64_t ab, int64_t delim,

@@ -147,6 +147,49 @@ __global__ void k_str_case(int upper, const uint8_t* __restrict__ in,
 // substring (1-based start in codepoints, length in codepoints; Spark
 // semantics: start 0 behaves like 1, negative counts from the end).
 // pass 1: byte [start,len) per row
+// initcap (ASCII): uppercase the first letter of each whitespace-split
+// word, lowercase the rest. reverse: reverse UTF-8 codepoint order in
+// place (same byte length).
+__global__ void k_str_initcap(const int32_t* __restrict__ ao,
+                              const uint8_t* __restrict__ ab,
+                              uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t s0 = ao[i], s1 = ao[i + 1];
+    bool word_start = true;
+    for (int32_t p = s0; p < s1; ++p) {
+      uint8_t c = ab[p];
+      if (c == ' ') {
+        out[p] = c;
+        word_start = true;
+      } else {
+        if (word_start && c >= 'a' && c <= 'z') c -= 32;
+        else if (!word_start && c >= 'A' && c <= 'Z') c += 32;
+        out[p] = c;
+        word_start = false;
+      }
+    }
+  }
+}
+
+__global__ void k_str_reverse(const int32_t* __restrict__ ao,
+                              const uint8_t* __restrict__ ab,
+                              uint8_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t s0 = ao[i], s1 = ao[i + 1];
+    int32_t w = s1;
+    int32_t p = s0;
+    while (p < s1) {
+      int32_t q = p + 1;  // span one UTF-8 codepoint
+      while (q < s1 && (ab[q] & 0xC0) == 0x80) ++q;
+      w -= q - p;
+      for (int32_t k = 0; k < q - p; ++k) out[w + k] = ab[p + k];
+      p = q;
+    }
+  }
+}
+
 // split by a literal delimiter -> per-part (start, len) spans.
 // Java limit-0 semantics: trailing empty parts are dropped ("a,,".split
 // -> ["a"], ",,".split -> [], "".split -> [""]).
@@ -356,6 +399,20 @@ void hipdf_str_case(int upper, const void* in, void* out, int64_t nbytes,
                     hipStream_t stream) {
   hipLaunchKernelGGL(k_str_case, flat_grid(nbytes), dim3(HIPDF_BLOCK), 0,
                      stream, upper, (const uint8_t*)in, (uint8_t*)out, nbytes);
+}
+
+void hipdf_str_initcap(const void* ao, const void* ab, void* out,
+                       int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_initcap, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)ao, (const uint8_t*)ab,
+                     (uint8_t*)out, n);
+}
+
+void hipdf_str_reverse(const void* ao, const void* ab, void* out, int64_t n,
+                       hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_reverse, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)ao, (const uint8_t*)ab,
+                     (uint8_t*)out, n);
 }
 
 void hipdf_str_split_count(const void* ao, const void* ab,

@@ -171,6 +171,14 @@ class Expression:
     def length(self) -> "UnaryExpr":
         return UnaryExpr("length", self)
 
+    def initcap(self) -> "UnaryExpr":
+        """Capitalize each space-separated word (ASCII on GPU)."""
+        return UnaryExpr("initcap", self)
+
+    def reverse(self) -> "UnaryExpr":
+        """Reverse codepoint order."""
+        return UnaryExpr("reverse", self)
+
     def upper(self) -> "UnaryExpr":
         return UnaryExpr("upper", self)
 
@@ -363,6 +371,8 @@ def _coerce_py(v, dtype: DType):
 _UNARY_OUT = {
     "not": lambda t: BOOL,
     "trim": lambda t: STRING,
+    "initcap": lambda t: STRING,
+    "reverse": lambda t: STRING,
     "ltrim": lambda t: STRING,
     "rtrim": lambda t: STRING,
     "is_nan": lambda t: BOOL,

@@ -292,6 +292,15 @@ def _str_cmp(op, x, y):
 # ---------------------------------------------------------------------------
 
 def unary_op(op: str, col: Column, out_dtype: DType) -> Column:
+    if op == "initcap":
+        out = [None if v is None else
+               " ".join(w[:1].upper() + w[1:].lower() if w else w
+                        for w in v.split(" "))
+               for v in col.to_pylist()]
+        return Column.from_pylist(out, DType.string())
+    if op == "reverse":
+        out = [None if v is None else v[::-1] for v in col.to_pylist()]
+        return Column.from_pylist(out, DType.string())
     if op in ("trim", "ltrim", "rtrim"):
         fn = {"trim": str.strip, "ltrim": str.lstrip,
               "rtrim": str.rstrip}[op]

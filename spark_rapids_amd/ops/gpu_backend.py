@@ -585,6 +585,16 @@ def unary_op(op: str, col: Column, out_dtype: DType) -> Column:
         if op in ("trim", "ltrim", "rtrim"):
             return str_trim(col, {"trim": "both", "ltrim": "leading",
                                   "rtrim": "trailing"}[op])
+        if op in ("initcap", "reverse"):
+            nb = int(col.data.numel())
+            ob = torch.empty(max(nb, 1), dtype=torch.uint8,
+                             device="cuda")[:nb]
+            if n:
+                fn = ext.str_initcap if op == "initcap" else ext.str_reverse
+                fn(col.offsets.data_ptr(), col.data.data_ptr(),
+                   ob.data_ptr(), n, s)
+            return Column(out_dtype, n, ob, v, col.offsets.clone(),
+                          null_count=col._null_count)
         if op in ("upper", "lower"):
             nb = int(col.data.numel())
             ob = torch.empty(max(nb, 1), dtype=torch.uint8,

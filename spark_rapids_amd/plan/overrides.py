@@ -57,7 +57,8 @@ _GPU_UNARY_OPS = {
 }
 # string ops with GPU kernels (strings.hip); eq_null_safe still CPU-only
 _GPU_STRING_OK = {"eq", "ne", "lt", "le", "gt", "ge", "concat"}
-_GPU_STRING_UNARY = {"length", "upper", "lower", "trim", "ltrim", "rtrim"}
+_GPU_STRING_UNARY = {"length", "upper", "lower", "trim", "ltrim",
+                     "rtrim", "initcap", "reverse"}
 
 
 class TagReason:
@@ -113,7 +114,7 @@ class Tagger:
             if in_t.id is TypeId.STRING:
                 if e.op not in _GPU_STRING_UNARY:
                     out.append(f"unary op {e.op} on string not on GPU yet")
-                elif e.op in ("upper", "lower") and \
+                elif e.op in ("upper", "lower", "initcap") and \
                         not self.conf.get(_ALLOW_INCOMPAT):
                     out.append(f"{e.op} on GPU is ASCII-only "
                                "(spark.rapids.sql.incompatibleOps.enabled)")

@@ -108,6 +108,7 @@ class Parser:
     # -- grammar ---------------------------------------------------------
     def parse_query(self):
         self.expect_kw("SELECT")
+        distinct = self.kw("DISTINCT")
         star = False
         items = []  # (expr_or_aggexpr, alias)
         if self.op("*"):
@@ -195,6 +196,20 @@ class Parser:
         # assemble plan
         if where is not None:
             df = df.filter(where)
+        if distinct and not star:
+            # SELECT DISTINCT a, b ... -> project then dedupe
+            exprs = [(e.alias(alias) if alias else e) for e, alias in items]
+            df = df.select(*exprs).distinct()
+            if order:
+                df = df.sort(*[n for n, _ in order],
+                             descending=[d for _, d in order])
+            if limit is not None:
+                df = df.limit(limit)
+            if self.peek()[0] != "end":
+                raise SqlError("trailing tokens")
+            return df
+        if distinct and star:
+            df = df.distinct()
         from ..expr.windows import WindowExpr as _WE
 
         if any(isinstance(e, _WE) for e, _ in items):
@@ -570,7 +585,7 @@ class Parser:
             return Round(args[0], int(scale))
         if name in ("abs", "sqrt", "exp", "log", "floor", "ceil", "upper",
                     "lower", "length", "year", "month", "day", "trim",
-                    "ltrim", "rtrim"):
+                    "ltrim", "rtrim", "initcap", "reverse"):
             return UnaryExpr(name, args[0])
         if name == "concat":
             out = args[0]

@@ -256,3 +256,43 @@ class TestElementAt:
         sg = sr.Session()
         sc = sr.Session({"spark.rapids.sql.enabled": False})
         assert q(sg) == q(sc)
+
+
+class TestInitcapReverse:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_cpu(self, cpu):
+        df = cpu.create_dataframe({"s": ["hello wORLD", "", None, "émile x"]})
+        out = df.select(col("s").initcap().alias("i"),
+                        col("s").reverse().alias("r")).to_pydict()
+        assert out["i"][0] == "Hello World"
+        assert out["r"][0] == "DLROw olleh"
+        assert out["r"][3] == "x elimé"  # codepoint-order reverse
+
+    def test_sql_distinct(self, cpu):
+        cpu.register("tdst", cpu.create_dataframe({"a": [1, 1, 2]}))
+        assert sorted(cpu.sql("SELECT DISTINCT a FROM tdst").collect()) == \
+            [(1,), (2,)]
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        vals = [f"word{v} second{v}" if v % 7 else None for v in range(4000)]
+        vals += ["émile unicode-pass", ""]
+
+        def q(s):
+            df = s.create_dataframe({"s": vals})
+            return df.select(col("s").reverse().alias("r"),
+                             col("s").initcap().alias("i")).to_pydict()
+
+        sg = sr.Session({"spark.rapids.sql.incompatibleOps.enabled": True})
+        sc = sr.Session({"spark.rapids.sql.enabled": False,
+                         "spark.rapids.sql.incompatibleOps.enabled": True})
+        g, c = q(sg), q(sc)
+        assert g["r"] == c["r"]
+        # initcap on pure-ASCII rows must match; the unicode row is the
+        # documented ASCII-only incompat (é is not uppercased on GPU)
+        for a, b, v in zip(g["i"], c["i"], vals):
+            if v is not None and v.isascii():
+                assert a == b, v
