@@ -393,6 +393,14 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
                           node.nulls_last, kids[0],
                           target_bytes=conf.get(BATCH_SIZE_BYTES))
     if isinstance(node, L.Limit):
+        child = node.children[0]
+        if isinstance(child, L.Sort) and node.n <= 10_000_000:
+            # fuse ORDER BY + LIMIT: sort only per-batch heads
+            inner = _ensure_device(
+                _convert(child.children[0], conf, tagger, gpu_wanted),
+                kids[0].device)
+            return P.TopNExec(kids[0].device, child.keys, child.descending,
+                              child.nulls_last, node.n, inner)
         return P.LimitExec(device, node.n, kids[0])
     if isinstance(node, L.Union):
         return P.UnionExec(device, kids, node.schema())

@@ -864,6 +864,41 @@ class MapBatchesExec(PhysicalExec):
                 yield out
 
 
+class TopNExec(PhysicalExec):
+    """Fused ORDER BY + LIMIT n (GpuTakeOrderedAndProjectExec analogue):
+    each input batch is sorted and truncated to n rows before the final
+    merge sort, so the full input is never globally sorted."""
+
+    def __init__(self, device: str, keys: List[str], descending: List[bool],
+                 nulls_last: List[bool], n: int, child: PhysicalExec):
+        super().__init__(device, child.schema, [child])
+        self.keys = keys
+        self.descending = descending
+        self.nulls_last = nulls_last
+        self.n = n
+
+    def _sort_head(self, batch: ColumnBatch) -> ColumnBatch:
+        kidx = [self.schema.index(k) for k in self.keys]
+        order = ops.sort_order(batch, kidx, self.descending,
+                               self.nulls_last)
+        taken = ops.gather(batch, order)
+        return _slice_rows(taken, 0, min(self.n, taken.num_rows)) \
+            if taken.num_rows > self.n else taken
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        heads = [self._sort_head(b)
+                 for b in self.children[0].execute() if b.num_rows]
+        if not heads:
+            return
+        merged = ops.concat_batches(heads) if len(heads) > 1 else heads[0]
+        yield self._sort_head(merged)
+
+    def describe(self):
+        ks = ", ".join(f"{k}{' DESC' if d else ''}"
+                       for k, d in zip(self.keys, self.descending))
+        return f"{self.name()}[top {self.n} by {ks}]"
+
+
 class LimitExec(PhysicalExec):
     def __init__(self, device: str, n: int, child: PhysicalExec):
         super().__init__(device, child.schema, [child])

@@ -261,3 +261,46 @@ def test_gpu_subpartitioned_join_matches_cpu():
         g, c = q(sg, how), q(sc, how)
         assert g[0][0] == c[0][0], how
         assert g[0][1] == pytest.approx(c[0][1], rel=1e-12), how
+
+
+def test_topn_fused_matches_full_sort():
+    import numpy as np
+    import spark_rapids_amd as sr
+
+    rng = np.random.default_rng(4)
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({
+        "a": [int(v) if i % 17 else None
+              for i, v in enumerate(rng.integers(0, 10**6, 30000))],
+        "b": [float(v) for v in rng.uniform(0, 1, 30000)]},
+        num_partitions=5, coalesce=False) if False else s.create_dataframe({
+        "a": [int(v) if i % 17 else None
+              for i, v in enumerate(rng.integers(0, 10**6, 30000))],
+        "b": [float(v) for v in rng.uniform(0, 1, 30000)]})
+    top = df.sort("a", descending=True).limit(25)
+    assert "TopN" in top.physical_plan().tree_string()
+    got = [r[0] for r in top.collect()]
+    allv = sorted((r[0] for r in df.collect() if r[0] is not None),
+                  reverse=True)
+    assert got == allv[:25]  # DESC: nulls last by default
+
+    asc = df.sort("a").limit(5).collect()
+    assert all(r[0] is None for r in asc)  # asc: nulls first
+
+
+@pytest.mark.gpu
+def test_gpu_topn_matches_cpu():
+    import numpy as np
+    import spark_rapids_amd as sr
+
+    rng = np.random.default_rng(9)
+    data = {"a": [int(v) for v in rng.integers(0, 10**9, 200000)],
+            "b": [int(v) for v in rng.integers(0, 100, 200000)]}
+
+    def q(s):
+        df = s.create_dataframe(data)
+        return df.sort("a", "b", descending=[True, False]).limit(50).collect()
+
+    g = q(sr.Session())
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert g == c
