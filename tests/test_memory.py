@@ -272,10 +272,6 @@ def test_topn_fused_matches_full_sort():
     df = s.create_dataframe({
         "a": [int(v) if i % 17 else None
               for i, v in enumerate(rng.integers(0, 10**6, 30000))],
-        "b": [float(v) for v in rng.uniform(0, 1, 30000)]},
-        num_partitions=5, coalesce=False) if False else s.create_dataframe({
-        "a": [int(v) if i % 17 else None
-              for i, v in enumerate(rng.integers(0, 10**6, 30000))],
         "b": [float(v) for v in rng.uniform(0, 1, 30000)]})
     top = df.sort("a", descending=True).limit(25)
     assert "TopN" in top.physical_plan().tree_string()
