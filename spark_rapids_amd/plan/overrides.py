@@ -391,7 +391,8 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
 
         return P.SortExec(device, node.keys, node.descending,
                           node.nulls_last, kids[0],
-                          target_bytes=conf.get(BATCH_SIZE_BYTES))
+                          target_bytes=conf.get(BATCH_SIZE_BYTES),
+                          input_replicated=L.is_replicated(node.child))
     if isinstance(node, L.Limit):
         child = node.children[0]
         if isinstance(child, L.Sort) and node.n <= 10_000_000:

@@ -749,16 +749,22 @@ class CrossJoinExec(PhysicalExec):
 class SortExec(PhysicalExec):
     def __init__(self, device: str, keys: List[str], descending: List[bool],
                  nulls_last: List[bool], child: PhysicalExec,
-                 target_bytes: int = 2 << 30):
+                 target_bytes: int = 2 << 30,
+                 input_replicated: bool = False):
         super().__init__(device, child.schema, [child])
         self.keys = keys
         self.descending = descending
         self.nulls_last = nulls_last
         self.target_bytes = target_bytes
+        self.input_replicated = input_replicated
 
     def execute(self) -> Iterator[ColumnBatch]:
         from ..memory.spill import SpillableBatch
+        from ..shuffle import dist as _dist
 
+        if _dist.ctx().is_multi and not self.input_replicated:
+            yield from self._execute_distributed()
+            return
         handles = [SpillableBatch(b) for b in self.children[0].execute()]
         if not handles:
             return
@@ -779,6 +785,61 @@ class SortExec(PhysicalExec):
         kidx = [self.schema.index(k) for k in self.keys]
         order = ops.sort_order(table, kidx, self.descending, self.nulls_last)
         return ops.gather(table, order)
+
+    def _execute_distributed(self) -> Iterator[ColumnBatch]:
+        """Distributed global ORDER BY (GpuRangePartitioner analogue):
+        sample the monotone int64 sort-key proxy on every rank, agree on
+        world-1 range boundaries, exchange rows by range over RCCL, then
+        sort locally — rank r then holds the r-th globally ordered range.
+        Every rank runs the same collective sequence (sample all-gather +
+        one all-to-all) regardless of data."""
+        import numpy as np
+        import torch
+
+        from ..shuffle import dist as _dist
+        from ..shuffle.exchange import exchange_by_ranges
+
+        c = _dist.ctx()
+        batches = [b for b in self.children[0].execute() if b.num_rows]
+        table = ops.concat_batches(batches) if len(batches) > 1 else (
+            batches[0] if batches else None)
+        kidx = [self.schema.index(k) for k in self.keys]
+        k0, d0, n0 = kidx[0], self.descending[0], self.nulls_last[0]
+        # sample the range-key proxy (empty ranks contribute a pad that is
+        # filtered by the sample-count header)
+        if table is not None and table.num_rows:
+            kc = ops.backend_for(*table.columns).range_key(
+                table.columns[k0], d0, n0)
+            arr = kc.data.cpu().numpy()[:kc.size]
+            stride = max(1, len(arr) // 4096)
+            sample = np.ascontiguousarray(arr[::stride][:4096],
+                                          dtype=np.int64)
+        else:
+            sample = np.zeros(0, dtype=np.int64)
+        import torch.distributed as td
+
+        gathered: List = [None] * c.world
+        td.all_gather_object(gathered, sample)
+        allsamp = np.concatenate([g for g in gathered if len(g)]) \
+            if any(len(g) for g in gathered) else np.zeros(1, np.int64)
+        qs = np.quantile(allsamp, [r / c.world for r in range(1, c.world)],
+                         method="nearest").astype(np.int64)
+        bounds = list(np.maximum.accumulate(qs))  # monotone cut points
+        if table is None:
+            # participate in the exchange with an empty batch
+            cols = [Column.from_pylist([], f.dtype) for f in self.schema.fields]
+            if self.gpu:
+                cols = [col.cuda() for col in cols]
+            table = ColumnBatch(cols, 0)
+            kc = ops.backend_for(*table.columns).range_key(
+                table.columns[k0], d0, n0)
+        received = exchange_by_ranges(table, kc, bounds)
+        received = [b for b in received if b.num_rows]
+        if not received:
+            return
+        mine = ops.concat_batches(received) if len(received) > 1 \
+            else received[0]
+        yield self._sort_one(mine)
 
     def _external_sort(self, handles) -> Iterator[ColumnBatch]:
         """Out-of-core sort (reference analogue: GpuSortExec's full-sort

@@ -45,3 +45,29 @@ def gather_all(batch: ColumnBatch) -> List[ColumnBatch]:
     schema = batch_schema(batch)
     bufs = dist.all_gather_bytes(serialize_batch(batch))
     return [deserialize_batch(b, schema) for b in bufs]
+
+
+def exchange_by_ranges(batch: ColumnBatch, range_key: Column,
+                       bounds: Sequence[int]) -> List[ColumnBatch]:
+    """Range-partition rows by a monotone int64 key against world-1 cut
+    points and exchange (distributed global sort: rank r receives the
+    r-th key range). Reference analogue: GpuRangePartitioner."""
+    from ..types import BOOL
+
+    c = dist.ctx()
+    schema = batch_schema(batch)
+    send = []
+    for j in range(c.world):
+        mask = None
+        if j > 0:
+            mask = ops.binary_op_scalar("ge", range_key, int(bounds[j - 1]),
+                                        BOOL)
+        if j < len(bounds):
+            m2 = ops.binary_op_scalar("lt", range_key, int(bounds[j]), BOOL)
+            mask = m2 if mask is None else ops.binary_op("and", mask, m2,
+                                                         BOOL)
+        piece = batch if mask is None else ops.apply_boolean_mask(batch,
+                                                                  mask)
+        send.append(serialize_batch(piece))
+    recv = dist.all_to_all_bytes(send)
+    return [deserialize_batch(b, schema) for b in recv]

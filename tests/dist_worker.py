@@ -134,6 +134,18 @@ def main():
         g = [r for r in allro if r[1] == 1]
         assert sum(r[2] for r in g) == n_total, g
 
+    # distributed global ORDER BY: rank r holds the r-th sorted range and
+    # the rank-order concatenation is the full global sort
+    sorted_rows = df.sort("v", descending=True).collect()
+    my_vals = [r[1] for r in sorted_rows]
+    assert my_vals == sorted(my_vals, reverse=True)
+    gs = [None] * world
+    td.all_gather_object(gs, my_vals)
+    if rank == 0:
+        flat = [v for vs in gs for v in vs]
+        assert flat == sorted(flat, reverse=True), "ranges out of order"
+        assert len(flat) == n_total
+
     td.barrier()
     if rank == 0:
         print("DIST_OK")
