@@ -107,6 +107,7 @@ class Parser:
 
     # -- grammar ---------------------------------------------------------
     def parse_query(self):
+        self.explain_only = self.kw("EXPLAIN")
         self.expect_kw("SELECT")
         distinct = self.kw("DISTINCT")
         star = False
@@ -614,4 +615,8 @@ class Parser:
 
 
 def parse_sql(session, text: str):
-    return Parser(text, session).parse_query()
+    p = Parser(text, session)
+    df = p.parse_query()
+    if getattr(p, "explain_only", False):
+        return df.physical_plan().tree_string()
+    return df

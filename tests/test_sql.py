@@ -165,3 +165,9 @@ def test_sql_lag_lead(session):
                  "lead(v) OVER (PARTITION BY k ORDER BY t) r FROM tlag"
                  ).collect()
     assert rows == [(None, 20), (10, 30), (20, None)]
+
+
+def test_sql_explain(session):
+    session.register("texp", session.create_dataframe({"a": [1]}))
+    out = session.sql("EXPLAIN SELECT a FROM texp WHERE a > 0")
+    assert isinstance(out, str) and "Filter" in out
