@@ -180,6 +180,11 @@ JOIN_SUBPARTITION_BYTES = int_conf(
     "Build sides larger than this are hash-split into buckets and joined "
     "bucket-by-bucket (GpuSubPartitionHashJoin analogue), bounding the "
     "peak size of any single hash table and its gather maps.")
+CPU_BRIDGE = bool_conf(
+    "spark.rapids.sql.cpuBridge.enabled", True,
+    "Evaluate CPU-only expressions inside GPU projections via a host "
+    "round-trip instead of demoting the whole project "
+    "(GpuCpuBridgeExpression analogue).")
 FILECACHE = bool_conf(
     "spark.rapids.filecache.enabled", False,
     "Cache decoded scan batches per (file, mtime) in host memory so "

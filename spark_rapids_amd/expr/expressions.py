@@ -646,6 +646,38 @@ class ArraySize(Expression):
         return f"size({self.child})"
 
 
+class CpuBridge(Expression):
+    """Evaluate a CPU-only expression inside a GPU projection via a host
+    round-trip, so one unsupported expression no longer demotes the whole
+    project (reference analogue: GpuCpuBridgeExpression /
+    GpuCpuBridgeOptimizer)."""
+
+    def __init__(self, child: Expression):
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return self.child.dtype(schema)
+
+    def nullable(self, schema: Schema) -> bool:
+        return self.child.nullable(schema)
+
+    def output_name(self) -> str:
+        return self.child.output_name()
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        if batch.columns and batch.columns[0].is_cuda:
+            out = self.child.eval(batch.cpu(), schema)
+            return out.cuda()
+        return self.child.eval(batch, schema)
+
+    def __str__(self):
+        return f"cpu_bridge({self.child})"
+
+
 class RegexpExtract(Expression):
     """regexp_extract(str, pattern, idx): the capture group's text for the
     first match; "" when no match or non-participating group (Spark

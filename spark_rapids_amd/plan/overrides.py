@@ -130,6 +130,8 @@ class Tagger:
                     compile_regex(e.pattern)
                 except RegexUnsupported as ex:
                     out.append(f"regex not supported on GPU: {ex}")
+        elif type(e).__name__ == "CpuBridge":
+            return  # bridged subtree runs on host by design
         elif type(e).__name__ == "StrSplit":
             if any(c in e.delimiter for c in ".\\+*?()[]{}|^$"):
                 out.append("regex split delimiters run on CPU")
@@ -354,6 +356,26 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
     if isinstance(node, L.Filter):
         return P.FilterExec(device, node.condition, kids[0])
     if isinstance(node, L.Project):
+        from ..config import CPU_BRIDGE
+
+        if not on_gpu and gpu_wanted and conf.get(CPU_BRIDGE) \
+                and conf.exec_enabled("Project"):
+            from ..expr.expressions import CpuBridge
+
+            cs2 = node.child.schema()
+            wrapped = []
+            any_clean = False
+            for e in node.exprs:
+                if tagger.expr_reasons(e, cs2):
+                    wrapped.append(CpuBridge(e))
+                else:
+                    wrapped.append(e)
+                    any_clean = True
+            if any_clean:
+                inner = _ensure_device(
+                    _convert(node.children[0], conf, tagger, gpu_wanted),
+                    "cuda")
+                return P.ProjectExec("cuda", wrapped, inner, node.schema())
         return P.ProjectExec(device, node.exprs, kids[0], node.schema())
     if isinstance(node, L.Expand):
         return P.ExpandExec(device, node.projections, kids[0], node.schema())

@@ -131,3 +131,43 @@ def test_column_pruning_under_join(session):
     res = q.collect()
     session.conf.set("spark.rapids.sql.optimizer.pruneColumns.enabled", False)
     assert sorted(res) == sorted(q.collect())
+
+
+def test_cpu_bridge_expression_plan():
+    import torch
+
+    orig = torch.cuda.is_available
+    torch.cuda.is_available = lambda: True
+    try:
+        import spark_rapids_amd as sr
+
+        s = sr.Session({"spark.rapids.sql.incompatibleOps.enabled": False})
+        df = s.create_dataframe({"a": ["x"], "b": [1]})
+        tree = (df.select(col("a").upper().alias("u"),
+                          (col("b") + 1).alias("c"))
+                .physical_plan().tree_string())
+        assert "GpuProject" in tree and "cpu_bridge" in tree, tree
+        off = sr.Session({"spark.rapids.sql.incompatibleOps.enabled": False,
+                          "spark.rapids.sql.cpuBridge.enabled": False})
+        df2 = off.create_dataframe({"a": ["x"], "b": [1]})
+        tree2 = (df2.select(col("a").upper().alias("u"))
+                 .physical_plan().tree_string())
+        assert "CpuProject" in tree2, tree2
+    finally:
+        torch.cuda.is_available = orig
+
+
+@pytest.mark.gpu
+def test_cpu_bridge_results_match():
+    import spark_rapids_amd as sr
+
+    s = sr.Session({"spark.rapids.sql.incompatibleOps.enabled": False})
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    vals = ["MiXeD case", "ABC", None, ""]
+
+    def q(sess):
+        df = sess.create_dataframe({"a": vals, "b": list(range(4))})
+        return (df.select(col("a").upper().alias("u"),
+                          (col("b") * 2).alias("d")).to_pydict())
+
+    assert q(s) == q(sc)
