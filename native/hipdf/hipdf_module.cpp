@@ -142,6 +142,9 @@ void hipdf_str_cmp(int, const void*, const void*, const void*, const void*,
 void hipdf_regex_extract(const void*, int, const void*, const void*,
                          const void*, int, void*, void*, void*, int64_t,
                          hipStream_t);
+void hipdf_regex_extract_all(const void*, int, const void*, const void*,
+                             const void*, int, const void*, void*, void*,
+                             void*, int, void*, int64_t, hipStream_t);
 void hipdf_regex_replace(const void*, int, const void*, const void*,
                          const void*, const void*, int, const void*,
                          const void*, void*, void*, int, void*, int64_t,
@@ -543,6 +546,17 @@ PYBIND11_MODULE(hipdf, m) {
     hipdf_regex_extract(P(prog), nops, P(classes), P(offsets), P(bytes),
                         group, PM(out_start), PM(out_len), PM(overflow), n,
                         S(stream));
+    check_async();
+  });
+  m.def("regex_extract_all", [](int64_t prog, int nops, int64_t classes,
+                                int64_t offsets, int64_t bytes, int group,
+                                int64_t part_off, int64_t counts,
+                                int64_t out_ss, int64_t out_sl, int mode,
+                                int64_t overflow, int64_t n,
+                                int64_t stream) {
+    hipdf_regex_extract_all(P(prog), nops, P(classes), P(offsets), P(bytes),
+                            group, P(part_off), PM(counts), PM(out_ss),
+                            PM(out_sl), mode, PM(overflow), n, S(stream));
     check_async();
   });
   m.def("regex_replace", [](int64_t prog, int nops, int64_t classes,

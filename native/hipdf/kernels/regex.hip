@@ -336,7 +336,60 @@ __global__ void k_regex_replace(const int32_t* __restrict__ prog, int nops,
   }
 }
 
+// regexp_extract_all: every non-overlapping match's group span.
+// mode 0: per-row match count; mode 1: (start, len) spans at part_off.
+__global__ void k_regex_extract_all(const int32_t* __restrict__ prog,
+                                    int nops,
+                                    const uint8_t* __restrict__ classes,
+                                    const int32_t* __restrict__ offsets,
+                                    const uint8_t* __restrict__ bytes,
+                                    int group,
+                                    const int64_t* __restrict__ part_off,
+                                    int64_t* __restrict__ counts,
+                                    int32_t* __restrict__ out_ss,
+                                    int64_t* __restrict__ out_sl, int mode,
+                                    int32_t* __restrict__ overflow,
+                                    int64_t n) {
+  for (int64_t row = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; row < n;
+       row += (int64_t)gridDim.x * blockDim.x) {
+    int32_t begin = offsets[row];
+    int32_t len = offsets[row + 1] - begin;
+    const uint8_t* s = bytes + begin;
+    int32_t saves[RX_SLOTS];
+    int64_t steps = 0;
+    int32_t pos = 0;
+    int64_t cnt = 0;
+    int64_t w = mode ? part_off[row] : 0;
+    bool blown = false;
+    while (pos <= len) {
+      int32_t e = rx_find_from(prog, classes, s, len, pos, saves, &steps);
+      if (e == -2) {
+        blown = true;
+        break;
+      }
+      if (e < 0) break;
+      int32_t ms = saves[0], me = saves[1];
+      if (mode) {
+        int32_t gs = saves[2 * group], ge = saves[2 * group + 1];
+        if (gs >= 0 && ge >= gs) {
+          out_ss[w] = begin + gs;
+          out_sl[w] = ge - gs;
+        } else {
+          out_ss[w] = begin;
+          out_sl[w] = 0;
+        }
+        ++w;
+      }
+      ++cnt;
+      pos = me > ms ? me : ms + 1;
+    }
+    if (blown) atomicAdd(overflow, 1);
+    if (!mode) counts[row] = blown ? 0 : cnt;
+  }
+}
+
 extern "C" {
+
 
 void hipdf_regex_extract(const void* prog, int nops, const void* classes,
                          const void* offsets, const void* bytes, int group,
@@ -347,6 +400,21 @@ void hipdf_regex_extract(const void* prog, int nops, const void* classes,
                      (const uint8_t*)classes, (const int32_t*)offsets,
                      (const uint8_t*)bytes, group, (int32_t*)out_start,
                      (int64_t*)out_len, (int32_t*)overflow, n);
+}
+
+void hipdf_regex_extract_all(const void* prog, int nops,
+                             const void* classes, const void* offsets,
+                             const void* bytes, int group,
+                             const void* part_off, void* counts,
+                             void* out_ss, void* out_sl, int mode,
+                             void* overflow, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_regex_extract_all, flat_grid(n), dim3(HIPDF_BLOCK),
+                     0, stream, (const int32_t*)prog, nops,
+                     (const uint8_t*)classes, (const int32_t*)offsets,
+                     (const uint8_t*)bytes, group,
+                     (const int64_t*)part_off, (int64_t*)counts,
+                     (int32_t*)out_ss, (int64_t*)out_sl, mode,
+                     (int32_t*)overflow, n);
 }
 
 void hipdf_regex_replace(const void* prog, int nops, const void* classes,

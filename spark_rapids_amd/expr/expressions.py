@@ -162,6 +162,10 @@ class Expression:
     def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
         return RegexpExtract(self, pattern, group)
 
+    def regexp_extract_all(self, pattern: str,
+                           group: int = 1) -> "RegexpExtractAll":
+        return RegexpExtractAll(self, pattern, group)
+
     def regexp_replace(self, pattern: str, replacement: str) -> "RegexpReplace":
         return RegexpReplace(self, pattern, replacement)
 
@@ -702,6 +706,31 @@ class RegexpExtract(Expression):
 
     def __str__(self):
         return f"regexp_extract({self.child}, {self.pattern!r}, {self.group})"
+
+
+class RegexpExtractAll(Expression):
+    """regexp_extract_all: every match's group text as array<string>
+    (GpuRegExpExtractAll analogue; shares the capture-group VM)."""
+
+    def __init__(self, child: Expression, pattern: str, group: int = 1):
+        self.child = child
+        self.pattern = pattern
+        self.group = group
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return DType.list_(STRING)
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.regexp_extract_all(self.child.eval(batch, schema),
+                                      self.pattern, self.group)
+
+    def __str__(self):
+        return (f"regexp_extract_all({self.child}, {self.pattern!r}, "
+                f"{self.group})")
 
 
 class RegexpReplace(Expression):

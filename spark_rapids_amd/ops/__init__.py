@@ -62,6 +62,10 @@ def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return backend_for(col).regexp_extract(col, pattern, group)
 
 
+def regexp_extract_all(col: Column, pattern: str, group: int) -> Column:
+    return backend_for(col).regexp_extract_all(col, pattern, group)
+
+
 def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
     return backend_for(col).regexp_replace(col, pattern, replacement)
 

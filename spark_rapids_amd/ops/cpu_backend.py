@@ -558,6 +558,23 @@ def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return Column.from_pylist(out, DType.string())
 
 
+def regexp_extract_all(col: Column, pattern: str, group: int) -> Column:
+    import re as _re
+
+    rx = _re.compile(pattern)
+    out = []
+    for v in col.to_pylist():
+        if v is None:
+            out.append(None)
+            continue
+        vals = []
+        for m in rx.finditer(v):
+            g = m.group(group) if group <= rx.groups else None
+            vals.append(g if g is not None else "")
+        out.append(vals)
+    return Column.from_pylist(out, DType.list_(DType.string()))
+
+
 def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
     import re as _re
 

@@ -383,6 +383,56 @@ def _compile_replacement(replacement: str, ngroups: int):
     return ops_l, bytes(lit)
 
 
+def regexp_extract_all(col: Column, pattern: str, group: int) -> Column:
+    """All matches' group text per row as LIST<STRING>."""
+    n = col.size
+    s = _stream()
+    lt = DType.list_(DType.string())
+    if n == 0:
+        return Column.from_pylist([], lt).cuda()
+    prog, prog_t, cls_t = _rx_prog_tensors(pattern)
+    if group > prog.ngroups:
+        raise ValueError(f"group {group} > {prog.ngroups}")
+    counts = torch.empty(n, dtype=torch.int64, device="cuda")
+    overflow = torch.zeros(1, dtype=torch.int32, device="cuda")
+    ext.regex_extract_all(prog_t.data_ptr(), len(prog.ops),
+                          cls_t.data_ptr(), col.offsets.data_ptr(),
+                          col.data.data_ptr(), group, 0, counts.data_ptr(),
+                          0, 0, 0, overflow.data_ptr(), n, s)
+    if int(overflow.item()) > 0:
+        from . import cpu_backend
+
+        return cpu_backend.regexp_extract_all(col.cpu(), pattern,
+                                              group).cuda()
+    part_off, total = _exclusive_scan_i64(counts)
+    ss = torch.empty(max(total, 1), dtype=torch.int32, device="cuda")[:total]
+    sl = torch.empty(max(total, 1), dtype=torch.int64, device="cuda")[:total]
+    if total:
+        ext.regex_extract_all(prog_t.data_ptr(), len(prog.ops),
+                              cls_t.data_ptr(), col.offsets.data_ptr(),
+                              col.data.data_ptr(), group,
+                              part_off.data_ptr(), counts.data_ptr(),
+                              ss.data_ptr(), sl.data_ptr(), 1,
+                              overflow.data_ptr(), n, s)
+    scanned, nbytes = _exclusive_scan_i64(sl) if total else (sl, 0)
+    out_bytes = torch.empty(max(nbytes, 1), dtype=torch.uint8,
+                            device="cuda")[:nbytes]
+    if nbytes:
+        ext.substr_copy(col.data.data_ptr(), ss.data_ptr(), sl.data_ptr(),
+                        scanned.data_ptr(), out_bytes.data_ptr(), total, s)
+    coffs = torch.empty(total + 1, dtype=torch.int32, device="cuda")
+    if total:
+        ext.narrow_i64_i32(scanned.data_ptr(), coffs.data_ptr(), total, s)
+    coffs[total] = nbytes
+    child = Column(DType.string(), total, out_bytes, None, coffs, 0)
+    loffs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    ext.narrow_i64_i32(part_off.data_ptr(), loffs.data_ptr(), n, s)
+    loffs[n] = total
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(lt, n, torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                  v, loffs, col._null_count, child)
+
+
 def regexp_replace(col: Column, pattern: str, replacement: str) -> Column:
     n = col.size
     s = _stream()

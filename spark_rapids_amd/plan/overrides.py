@@ -137,7 +137,8 @@ class Tagger:
                 out.append("regex split delimiters run on CPU")
         elif type(e).__name__ in ("ArraySize", "ElementAt"):
             pass
-        elif type(e).__name__ in ("RegexpExtract", "RegexpReplace"):
+        elif type(e).__name__ in ("RegexpExtract", "RegexpReplace",
+                                  "RegexpExtractAll"):
             from ..ops.regex_compiler import RegexUnsupported, compile_regex
 
             try:

@@ -563,6 +563,11 @@ class Parser:
 
             idx = int(args[2].value) if len(args) > 2 else 1
             return RegexpExtract(args[0], args[1].value, idx)
+        if name == "regexp_extract_all":
+            from ..expr.expressions import RegexpExtractAll
+
+            idx = int(args[2].value) if len(args) > 2 else 1
+            return RegexpExtractAll(args[0], args[1].value, idx)
         if name == "regexp_replace":
             from ..expr.expressions import RegexpReplace
 

@@ -125,3 +125,36 @@ class TestRegexpExtractReplace:
         tree = (df.select(col("s").regexp_extract(r"(\d)", 1))
                 .physical_plan().tree_string())
         assert "GpuProject" in tree, tree
+
+
+class TestRegexpExtractAll:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_cpu(self, cpu):
+        df = cpu.create_dataframe({"s": ["a1 b22 c333", "none", None, ""]})
+        out = (df.select(col("s").regexp_extract_all(r"(\d+)").alias("m"),
+                         col("s").regexp_extract_all(r"([a-z])(\d+)", 2)
+                         .alias("g2")).to_pydict())
+        assert out["m"] == [["1", "22", "333"], [], None, []]
+        assert out["g2"] == [["1", "22", "333"], [], None, []]
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        import numpy as np
+
+        rng = np.random.default_rng(12)
+        vals = [" ".join(f"k{v}v{v*3}" for v in rng.integers(0, 99, v % 5))
+                if i % 11 else None
+                for i, v in enumerate(rng.integers(0, 30, 4000))]
+
+        def q(s):
+            df = s.create_dataframe({"s": vals})
+            return (df.select(
+                col("s").regexp_extract_all(r"v(\d+)").alias("m"))
+                .to_pydict())
+
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        assert q(sg) == q(sc)
