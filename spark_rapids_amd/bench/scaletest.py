@@ -83,6 +83,20 @@ def _queries() -> List:
         return (t["fact"].join(hot, on="ss_item_id", how="semi")
                 .agg(count_star()))
 
+    def window_sort_heavy(t):
+        # TPC-DS q67 shape: rollup-style agg, rank over partition by
+        # category ordered by revenue, keep top ranks, global sort
+        agged = (t["fact"].group_by("ss_store_id", "ss_item_id")
+                 .agg(sum_(col("ss_sales_price")).alias("rev")))
+        from ..types import FLOAT64
+
+        agged = agged.with_column("revd", col("rev").cast(FLOAT64))
+        ranked = agged.with_column(
+            "rk", rank().over(["ss_store_id"], ["revd"],
+                              descending=[True]))
+        return (ranked.filter(col("rk") <= 100)
+                .sort("ss_store_id", "rk").limit(1000))
+
     def multi_key_agg(t):
         return (t["fact"].group_by("ss_store_id", "ss_promo")
                 .agg(sum_(col("ss_sales_price")), min_(col("ss_discount")),
@@ -97,6 +111,7 @@ def _queries() -> List:
         ("big_sort", big_sort),
         ("semi_anti", semi_anti),
         ("multi_key_agg", multi_key_agg),
+        ("window_sort_heavy", window_sort_heavy),
     ]
 
 
