@@ -176,7 +176,7 @@ __global__ void k_json_field(const uint8_t* __restrict__ bytes,
       }
     }
     bool ok = false;
-    if (type == 2) {
+    if (type == 2 || type == 4) {
       out_ss[i] = 0;
       out_sl[i] = 0;
     } else if (type == 0) {
@@ -188,7 +188,32 @@ __global__ void k_json_field(const uint8_t* __restrict__ bytes,
     }
     if (vpos >= 0 && vpos < re && !(bytes[vpos] == 'n')) {  // null -> null
       uint8_t c = bytes[vpos];
-      if (type == 2) {
+      if (type == 4) {  // any scalar as raw text (get_json_object)
+        if (c == '"') {
+          int32_t e = vpos + 1;
+          bool esc = false;
+          while (e < re && bytes[e] != '"') {
+            if (bytes[e] == '\\') { esc = true; break; }
+            ++e;
+          }
+          if (esc) atomicAdd(unsupported, 1);
+          else if (e < re) {
+            out_ss[i] = vpos + 1;
+            out_sl[i] = e - vpos - 1;
+            ok = true;
+          }
+        } else if (c == '{' || c == '[') {
+          atomicAdd(unsupported, 1);  // nested -> CPU fallback
+        } else {
+          int32_t e = vpos;
+          while (e < re && bytes[e] != ',' && bytes[e] != '}' &&
+                 bytes[e] != ' ' && bytes[e] != '\t')
+            ++e;
+          out_ss[i] = vpos;
+          out_sl[i] = e - vpos;
+          ok = e > vpos;
+        }
+      } else if (type == 2) {
         if (c == '"') {
           int32_t e = vpos + 1;
           bool esc = false;

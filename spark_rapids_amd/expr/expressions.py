@@ -162,6 +162,9 @@ class Expression:
     def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
         return RegexpExtract(self, pattern, group)
 
+    def get_json_object(self, path: str) -> "GetJsonObject":
+        return GetJsonObject(self, path)
+
     def regexp_extract_all(self, pattern: str,
                            group: int = 1) -> "RegexpExtractAll":
         return RegexpExtractAll(self, pattern, group)
@@ -648,6 +651,30 @@ class ArraySize(Expression):
 
     def __str__(self):
         return f"size({self.child})"
+
+
+class GetJsonObject(Expression):
+    """get_json_object(col, '$.key'): top-level scalar extraction runs on
+    the GPU via the JSON field kernel (csv.hip k_json_field); nested paths
+    fall back to the CPU json parser (GpuGetJsonObject analogue)."""
+
+    def __init__(self, child: Expression, path: str):
+        self.child = child
+        self.path = path
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return STRING
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.get_json_object(self.child.eval(batch, schema),
+                                   self.path)
+
+    def __str__(self):
+        return f"get_json_object({self.child}, {self.path!r})"
 
 
 class CpuBridge(Expression):

@@ -62,6 +62,10 @@ def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return backend_for(col).regexp_extract(col, pattern, group)
 
 
+def get_json_object(col: Column, path: str) -> Column:
+    return backend_for(col).get_json_object(col, path)
+
+
 def regexp_extract_all(col: Column, pattern: str, group: int) -> Column:
     return backend_for(col).regexp_extract_all(col, pattern, group)
 

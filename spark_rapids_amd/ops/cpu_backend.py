@@ -558,6 +558,40 @@ def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return Column.from_pylist(out, DType.string())
 
 
+def get_json_object(col: Column, path: str) -> Column:
+    import json as _json
+
+    assert path.startswith("$"), "json path must start with $"
+    keys = [k for k in path[1:].lstrip(".").split(".") if k]
+    out = []
+    for v in col.to_pylist():
+        if v is None:
+            out.append(None)
+            continue
+        try:
+            cur = _json.loads(v)
+        except ValueError:
+            out.append(None)
+            continue
+        for k in keys:
+            if isinstance(cur, dict) and k in cur:
+                cur = cur[k]
+            else:
+                cur = None
+                break
+        if cur is None:
+            out.append(None)
+        elif isinstance(cur, str):
+            out.append(cur)
+        elif isinstance(cur, bool):
+            out.append("true" if cur else "false")
+        elif isinstance(cur, (dict, list)):
+            out.append(_json.dumps(cur, separators=(",", ":")))
+        else:
+            out.append(_json.dumps(cur))
+    return Column.from_pylist(out, DType.string())
+
+
 def regexp_extract_all(col: Column, pattern: str, group: int) -> Column:
     import re as _re
 

@@ -383,6 +383,45 @@ def _compile_replacement(replacement: str, ngroups: int):
     return ops_l, bytes(lit)
 
 
+def get_json_object(col: Column, path: str) -> Column:
+    """Top-level scalar key extraction ($.key) via k_json_field over the
+    string column's own row spans."""
+    keys = [k for k in path[1:].lstrip(".").split(".") if k]
+    if len(keys) != 1:
+        raise NotImplementedError("gpu get_json_object: nested path")
+    n = col.size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.string())
+    nameb = keys[0].encode("utf-8")
+    name_t = torch.frombuffer(bytearray(nameb), dtype=torch.uint8).cuda()
+    ss = torch.empty(n, dtype=torch.int32, device="cuda")
+    sl = torch.empty(n, dtype=torch.int64, device="cuda")
+    valid_u8 = torch.empty(n, dtype=torch.uint8, device="cuda")
+    unsupported = torch.zeros(1, dtype=torch.int32, device="cuda")
+    row_start = col.offsets[:n]
+    row_end = col.offsets[1:]
+    ext.json_field(col.data.data_ptr(), row_start.data_ptr(),
+                   row_end.data_ptr(), name_t.data_ptr(), len(nameb), 4,
+                   0, 0, ss.data_ptr(), sl.data_ptr(), valid_u8.data_ptr(),
+                   unsupported.data_ptr(), n, s)
+    if int(unsupported.item()) > 0:
+        from . import cpu_backend
+
+        return cpu_backend.get_json_object(col.cpu(), path).cuda()
+    out = _strings_from_spans(col, ss, sl, n, s)
+    from ..column import mask_nbytes
+
+    v64 = torch.empty(n, dtype=torch.int64, device="cuda")
+    ext.cast(0, 4, valid_u8.data_ptr(), v64.data_ptr(), n, s)
+    mask = torch.empty(mask_nbytes(n), dtype=torch.uint8, device="cuda")
+    ext.mask_from_nonzero(v64.data_ptr(), mask.data_ptr(), n, s)
+    if col.validity is not None:
+        mask = _and_masks(col.validity, mask)
+    return Column(DType.string(), n, out.data, mask, out.offsets,
+                  null_count=None)
+
+
 def regexp_extract_all(col: Column, pattern: str, group: int) -> Column:
     """All matches' group text per row as LIST<STRING>."""
     n = col.size

@@ -130,6 +130,10 @@ class Tagger:
                     compile_regex(e.pattern)
                 except RegexUnsupported as ex:
                     out.append(f"regex not supported on GPU: {ex}")
+        elif type(e).__name__ == "GetJsonObject":
+            keys = [k for k in e.path[1:].lstrip(".").split(".") if k]
+            if len(keys) != 1:
+                out.append("nested json paths run on CPU")
         elif type(e).__name__ == "CpuBridge":
             return  # bridged subtree runs on host by design
         elif type(e).__name__ == "StrSplit":

@@ -563,6 +563,10 @@ class Parser:
 
             idx = int(args[2].value) if len(args) > 2 else 1
             return RegexpExtract(args[0], args[1].value, idx)
+        if name == "get_json_object":
+            from ..expr.expressions import GetJsonObject
+
+            return GetJsonObject(args[0], args[1].value)
         if name == "regexp_extract_all":
             from ..expr.expressions import RegexpExtractAll
 

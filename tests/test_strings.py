@@ -296,3 +296,39 @@ class TestInitcapReverse:
         for a, b, v in zip(g["i"], c["i"], vals):
             if v is not None and v.isascii():
                 assert a == b, v
+
+
+class TestGetJsonObject:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_cpu(self, cpu):
+        df = cpu.create_dataframe({"j": ['{"a": 1, "b": "x"}',
+                                         '{"b": "y"}', None, "bad"]})
+        out = df.select(col("j").get_json_object("$.a").alias("a"),
+                        col("j").get_json_object("$.b").alias("b")
+                        ).to_pydict()
+        assert out["a"] == ["1", None, None, None]
+        assert out["b"] == ["x", "y", None, None]
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        import json
+
+        rows = [json.dumps({"k": i, "s": f"v{i}", "f": i / 4,
+                            "b": bool(i % 2)})
+                if i % 9 else None for i in range(4000)]
+
+        def q(s):
+            df = s.create_dataframe({"j": rows})
+            return df.select(
+                col("j").get_json_object("$.k").alias("k"),
+                col("j").get_json_object("$.s").alias("s"),
+                col("j").get_json_object("$.b").alias("b"),
+                col("j").get_json_object("$.missing").alias("m")
+            ).to_pydict()
+
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        assert q(sg) == q(sc)
