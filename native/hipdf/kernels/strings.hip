@@ -147,6 +147,44 @@ __global__ void k_str_case(int upper, const uint8_t* __restrict__ in,
 // substring (1-based start in codepoints, length in codepoints; Spark
 // semantics: start 0 behaves like 1, negative counts from the end).
 // pass 1: byte [start,len) per row
+// concat_ws: join n string columns with a separator, skipping NULL
+// values (result is never null — all-null rows give ""). Two-pass.
+struct StrColDesc {
+  const int32_t* off;
+  const uint8_t* bytes;
+  const uint64_t* valid;
+};
+
+__global__ void k_str_concat_ws(const StrColDesc* __restrict__ cols,
+                                int ncols, const uint8_t* __restrict__ sep,
+                                int seplen,
+                                const int64_t* __restrict__ out_off,
+                                int64_t* __restrict__ out_len,
+                                uint8_t* __restrict__ out, int mode,
+                                int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t w = mode ? out_off[i] : 0;
+    int64_t len = 0;
+    bool first = true;
+    for (int c = 0; c < ncols; ++c) {
+      const StrColDesc& d = cols[c];
+      if (d.valid && !((d.valid[i >> 6] >> (i & 63)) & 1ull)) continue;
+      if (!first) {
+        if (mode)
+          for (int k = 0; k < seplen; ++k) out[w + len + k] = sep[k];
+        len += seplen;
+      }
+      int32_t a = d.off[i], b = d.off[i + 1];
+      if (mode)
+        for (int32_t k = a; k < b; ++k) out[w + len + (k - a)] = d.bytes[k];
+      len += b - a;
+      first = false;
+    }
+    if (!mode) out_len[i] = len;
+  }
+}
+
 // initcap (ASCII): uppercase the first letter of each whitespace-split
 // word, lowercase the rest. reverse: reverse UTF-8 codepoint order in
 // place (same byte length).
@@ -399,6 +437,16 @@ void hipdf_str_case(int upper, const void* in, void* out, int64_t nbytes,
                     hipStream_t stream) {
   hipLaunchKernelGGL(k_str_case, flat_grid(nbytes), dim3(HIPDF_BLOCK), 0,
                      stream, upper, (const uint8_t*)in, (uint8_t*)out, nbytes);
+}
+
+void hipdf_str_concat_ws(const void* cols, int ncols, const void* sep,
+                         int seplen, const void* out_off, void* out_len,
+                         void* out, int mode, int64_t n,
+                         hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_concat_ws, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const StrColDesc*)cols, ncols,
+                     (const uint8_t*)sep, seplen, (const int64_t*)out_off,
+                     (int64_t*)out_len, (uint8_t*)out, mode, n);
 }
 
 void hipdf_str_initcap(const void* ao, const void* ab, void* out,

@@ -653,6 +653,37 @@ class ArraySize(Expression):
         return f"size({self.child})"
 
 
+class ConcatWs(Expression):
+    """concat_ws(sep, c1, c2, ...): join non-null values with sep; rows
+    with every value null give "" (never NULL — Spark concat_ws)."""
+
+    def __init__(self, sep: str, *exprs: Expression):
+        self.sep = sep
+        self.exprs = [_as_expr(e) for e in exprs]
+
+    @property
+    def children(self):
+        return tuple(self.exprs)
+
+    def dtype(self, schema: Schema) -> DType:
+        return STRING
+
+    def nullable(self, schema: Schema) -> bool:
+        return False
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        cols = [e.eval(batch, schema) for e in self.exprs]
+        return ops.concat_ws(self.sep, cols)
+
+    def __str__(self):
+        args = ", ".join(str(e) for e in self.exprs)
+        return f"concat_ws({self.sep!r}, {args})"
+
+
+def concat_ws(sep: str, *exprs) -> ConcatWs:
+    return ConcatWs(sep, *exprs)
+
+
 class GetJsonObject(Expression):
     """get_json_object(col, '$.key'): top-level scalar extraction runs on
     the GPU via the JSON field kernel (csv.hip k_json_field); nested paths

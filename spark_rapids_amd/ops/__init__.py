@@ -62,6 +62,10 @@ def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return backend_for(col).regexp_extract(col, pattern, group)
 
 
+def concat_ws(sep: str, cols: Sequence[Column]) -> Column:
+    return backend_for(*cols).concat_ws(sep, list(cols))
+
+
 def get_json_object(col: Column, path: str) -> Column:
     return backend_for(col).get_json_object(col, path)
 

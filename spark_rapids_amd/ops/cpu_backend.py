@@ -558,6 +558,13 @@ def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return Column.from_pylist(out, DType.string())
 
 
+def concat_ws(sep: str, cols) -> Column:
+    lists = [c.to_pylist() for c in cols]
+    out = [sep.join(v for v in row if v is not None)
+           for row in zip(*lists)]
+    return Column.from_pylist(out, DType.string())
+
+
 def get_json_object(col: Column, path: str) -> Column:
     import json as _json
 

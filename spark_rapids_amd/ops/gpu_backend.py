@@ -383,6 +383,33 @@ def _compile_replacement(replacement: str, ngroups: int):
     return ops_l, bytes(lit)
 
 
+def concat_ws(sep: str, cols) -> Column:
+    n = cols[0].size
+    s = _stream()
+    if n == 0:
+        return _empty_col(DType.string())
+    blob = b"".join(struct.pack("<qqq", c.offsets.data_ptr(),
+                                c.data.data_ptr(), _ptr(c.validity))
+                    for c in cols)
+    desc = torch.frombuffer(bytearray(blob), dtype=torch.uint8).cuda()
+    sep_t = _pattern_tensor(sep or "\x00")
+    seplen = len(sep.encode("utf-8"))
+    lens = torch.empty(n, dtype=torch.int64, device="cuda")
+    ext.str_concat_ws(desc.data_ptr(), len(cols), sep_t.data_ptr(), seplen,
+                      0, lens.data_ptr(), 0, 0, n, s)
+    scanned, total = _exclusive_scan_i64(lens)
+    out = torch.empty(max(total, 1), dtype=torch.uint8,
+                      device="cuda")[:total]
+    if total:
+        ext.str_concat_ws(desc.data_ptr(), len(cols), sep_t.data_ptr(),
+                          seplen, scanned.data_ptr(), lens.data_ptr(),
+                          out.data_ptr(), 1, n, s)
+    offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+    offs[n] = total
+    return Column(DType.string(), n, out, None, offs, 0)
+
+
 def get_json_object(col: Column, path: str) -> Column:
     """Top-level scalar key extraction ($.key) via k_json_field over the
     string column's own row spans."""

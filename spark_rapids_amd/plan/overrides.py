@@ -130,6 +130,8 @@ class Tagger:
                     compile_regex(e.pattern)
                 except RegexUnsupported as ex:
                     out.append(f"regex not supported on GPU: {ex}")
+        elif type(e).__name__ == "ConcatWs":
+            pass
         elif type(e).__name__ == "GetJsonObject":
             keys = [k for k in e.path[1:].lstrip(".").split(".") if k]
             if len(keys) != 1:

@@ -597,6 +597,10 @@ class Parser:
                     "lower", "length", "year", "month", "day", "trim",
                     "ltrim", "rtrim", "initcap", "reverse"):
             return UnaryExpr(name, args[0])
+        if name == "concat_ws":
+            from ..expr.expressions import ConcatWs
+
+            return ConcatWs(args[0].value, *args[1:])
         if name == "concat":
             out = args[0]
             for nxt in args[1:]:

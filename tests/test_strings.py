@@ -332,3 +332,45 @@ class TestGetJsonObject:
         sg = sr.Session()
         sc = sr.Session({"spark.rapids.sql.enabled": False})
         assert q(sg) == q(sc)
+
+
+class TestConcatWs:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def test_cpu(self, cpu):
+        from spark_rapids_amd import concat_ws
+
+        df = cpu.create_dataframe({"a": ["x", None, None, ""],
+                                   "b": ["1", "2", None, "z"]})
+        out = df.select(concat_ws("-", col("a"), col("b")).alias("c"),
+                        concat_ws("", col("a"), col("b")).alias("e")
+                        ).to_pydict()
+        assert out["c"] == ["x-1", "2", "", "-z"]
+        assert out["e"] == ["x1", "2", "", "z"]
+
+    def test_sql(self, cpu):
+        cpu.register("tcw", cpu.create_dataframe({"a": ["p"], "b": ["q"]}))
+        out = cpu.sql("SELECT concat_ws('.', a, b) FROM tcw").collect()
+        assert out == [("p.q",)]
+
+    @pytest.mark.gpu
+    def test_gpu_matches_cpu(self):
+        from spark_rapids_amd import concat_ws
+        import numpy as np
+
+        rng = np.random.default_rng(14)
+        n = 6000
+        a = [f"left{v}" if v % 3 else None for v in rng.integers(0, 30, n)]
+        b = [f"mid{v}" if v % 5 else None for v in rng.integers(0, 30, n)]
+        c = [f"r{v}" if v % 2 else "" for v in rng.integers(0, 30, n)]
+
+        def q(s):
+            df = s.create_dataframe({"a": a, "b": b, "c": c})
+            return df.select(concat_ws("|", col("a"), col("b"),
+                                       col("c")).alias("j")).to_pydict()
+
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        assert q(sg) == q(sc)
