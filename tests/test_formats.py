@@ -288,3 +288,34 @@ def test_gpu_orc_multi_stripe(tmp_path):
     got = sg.read_orc(p).to_pydict()
     assert got["a"] == t.column("a").to_pylist()
     assert got["s"] == t.column("s").to_pylist()
+
+
+def test_delta_checkpoint_replay(tmp_path, session):
+    import json
+    import os
+
+    import pyarrow.parquet as apq
+
+    root = str(tmp_path / "dckpt")
+    log = os.path.join(root, "_delta_log")
+    os.makedirs(log)
+    for i in range(3):
+        df = session.create_dataframe({"a": [i * 10 + k for k in range(4)]})
+        session.write_parquet(df, os.path.join(root, f"p{i}.parquet"))
+    # checkpoint at version 1 holds adds for p0/p1 and a remove of p0
+    ck = pa.table({
+        "add": [{"path": "p0.parquet"}, {"path": "p1.parquet"}, None],
+        "remove": [None, None, {"path": "p0.parquet"}],
+    })
+    apq.write_table(ck, os.path.join(
+        log, f"{1:020d}.checkpoint.parquet"))
+    with open(os.path.join(log, "_last_checkpoint"), "w") as f:
+        f.write(json.dumps({"version": 1}))
+    # later JSON commit adds p2
+    with open(os.path.join(log, f"{2:020d}.json"), "w") as f:
+        f.write(json.dumps({"add": {"path": "p2.parquet"}}) + "\n")
+    # an older JSON commit that must be IGNORED (pre-checkpoint)
+    with open(os.path.join(log, f"{0:020d}.json"), "w") as f:
+        f.write(json.dumps({"add": {"path": "ghost.parquet"}}) + "\n")
+    rows = sorted(r[0] for r in session.read_delta(root).collect())
+    assert rows == sorted(list(range(10, 14)) + list(range(20, 24)))
