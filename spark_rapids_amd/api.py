@@ -141,6 +141,26 @@ class DataFrame:
         return DataFrame(self.session,
                          L.Generate(column, self.plan, outer, pos=True))
 
+    def drop(self, *names: str) -> "DataFrame":
+        """Project away the named columns."""
+        keep = [f.name for f in self.schema.fields if f.name not in names]
+        return self.select(*[_col(n) for n in keep])
+
+    def with_column_renamed(self, old: str, new: str) -> "DataFrame":
+        exprs = [(_col(f.name).alias(new) if f.name == old else _col(f.name))
+                 for f in self.schema.fields]
+        return self.select(*exprs)
+
+    def union_by_name(self, other: "DataFrame") -> "DataFrame":
+        """Union matching the other frame's columns by NAME (Spark
+        unionByName)."""
+        names = [f.name for f in self.schema.fields]
+        other_names = {f.name for f in other.schema.fields}
+        missing = [n for n in names if n not in other_names]
+        if missing:
+            raise ValueError(f"unionByName: missing columns {missing}")
+        return self.union(other.select(*[_col(n) for n in names]))
+
     def distinct(self) -> "DataFrame":
         """Drop duplicate rows (group-by all columns with no aggregates)."""
         keys = [_col(f.name) for f in self.plan.schema().fields]

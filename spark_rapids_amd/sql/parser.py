@@ -628,6 +628,21 @@ class Parser:
 
 
 def parse_sql(session, text: str):
+    stripped = text.lstrip()
+    up = stripped.upper()
+    if up.startswith("CREATE "):
+        # CREATE [OR REPLACE] [TEMP|TEMPORARY] VIEW <name> AS <query>
+        import re as _re
+
+        m = _re.match(r"CREATE\s+(?:OR\s+REPLACE\s+)?"
+                      r"(?:TEMP(?:ORARY)?\s+)?VIEW\s+(\w+)\s+AS\s+(.*)",
+                      stripped, _re.I | _re.S)
+        if not m:
+            raise SqlError("unsupported CREATE statement")
+        name, body = m.group(1), m.group(2)
+        df = parse_sql(session, body)
+        session.register(name, df)
+        return df
     p = Parser(text, session)
     df = p.parse_query()
     if getattr(p, "explain_only", False):

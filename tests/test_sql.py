@@ -171,3 +171,23 @@ def test_sql_explain(session):
     session.register("texp", session.create_dataframe({"a": [1]}))
     out = session.sql("EXPLAIN SELECT a FROM texp WHERE a > 0")
     assert isinstance(out, str) and "Filter" in out
+
+
+def test_create_temp_view(session):
+    session.register("base", session.create_dataframe(
+        {"a": [1, 2, 3], "b": [1.0, 2.0, 3.0]}))
+    session.sql("CREATE OR REPLACE TEMP VIEW v2 AS "
+                "SELECT a, b FROM base WHERE a >= 2")
+    assert session.sql("SELECT COUNT(*) FROM v2").collect() == [(2,)]
+
+
+def test_dataframe_ergonomics(session):
+    df = session.create_dataframe({"a": [1], "b": ["x"], "c": [0.5]})
+    assert df.drop("b").schema.names == ["a", "c"]
+    assert df.with_column_renamed("b", "z").schema.names == ["a", "z", "c"]
+    other = session.create_dataframe({"c": [9.0], "a": [3], "b": ["y"]})
+    assert df.union_by_name(other).count() == 2
+    import pytest as _p
+
+    with _p.raises(ValueError):
+        df.union_by_name(session.create_dataframe({"a": [1]}))
