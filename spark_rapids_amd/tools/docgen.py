@@ -57,8 +57,11 @@ def supported_ops_doc() -> str:
     lines.append("compare: `" + "`, `".join(sorted(ov._GPU_STRING_OK)) + "`; "
                  "unary: `" + "`, `".join(sorted(ov._GPU_STRING_UNARY)) +
                  "`; plus `contains`, `starts_with`, `ends_with`, `like`, "
-                 "`substring` (strings.hip) and `rlike` (bytecode regex "
-                 "VM, regex.hip, CPU fallback outside the subset)")
+                 "`substring`, `split`, `concat_ws`, `element_at`/`size` "
+                 "over arrays, `get_json_object` (top-level keys) — "
+                 "strings.hip/csv.hip — and `rlike`, `regexp_extract`, "
+                 "`regexp_extract_all`, `regexp_replace` (capture-group "
+                 "regex VM, regex.hip, CPU fallback outside the subset)")
     lines += ["", "## Aggregate functions", "",
               "`sum`, `count`, `count(*)`, `min`, `max`, `avg`, `stddev`, "
               "`variance`, `first`, `last`, `count/sum(DISTINCT)`, "
