@@ -181,6 +181,11 @@ void hipdf_substr_copy(const void*, const void*, const void*, const void*,
 int hipdf_sort_key_width(int);
 void hipdf_make_sort_keys(int, const void*, const void*, const void*, int,
                           int, int, void*, int64_t, hipStream_t);
+void hipdf_make_sort_keys_str(const void*, const void*, const void*,
+                              const void*, int, int, void*, int64_t,
+                              hipStream_t);
+void hipdf_make_sort_keys_i128(const void*, const void*, const void*, int,
+                               int, void*, int64_t, hipStream_t);
 void hipdf_radix_count(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_radix_scatter(const void*, const void*, int, const void*, void*,
                          void*, int64_t, hipStream_t);
@@ -756,6 +761,21 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("part_num_blocks", &part_num_blocks);
   m.def("sort_num_blocks", &sort_num_blocks);
   m.def("sort_key_width", &hipdf_sort_key_width);
+  m.def("make_sort_keys_str", [](int64_t offsets, int64_t bytes,
+                                 int64_t valid, int64_t perm, int desc,
+                                 int chunk, int64_t keys, int64_t n,
+                                 int64_t stream) {
+    hipdf_make_sort_keys_str(P(offsets), P(bytes), P(valid), P(perm), desc,
+                             chunk, PM(keys), n, S(stream));
+    check_async();
+  });
+  m.def("make_sort_keys_i128", [](int64_t data, int64_t valid, int64_t perm,
+                                  int desc, int word, int64_t keys,
+                                  int64_t n, int64_t stream) {
+    hipdf_make_sort_keys_i128(P(data), P(valid), P(perm), desc, word,
+                              PM(keys), n, S(stream));
+    check_async();
+  });
   m.def("make_sort_keys", [](int t, int64_t data, int64_t valid, int64_t perm,
                              bool desc, bool nulls_last, bool null_only,
                              int64_t keys, int64_t n, int64_t stream) {

@@ -91,6 +91,57 @@ __global__ void k_make_sort_keys(const T* __restrict__ data,
   }
 }
 
+// string sort keys: big-endian 8-byte chunk `chunk` of each string
+// (short strings pad with 0x00, which orders prefixes first like byte
+// comparison). The host loops chunks last-to-first through the stable
+// radix, LSD-style over chunks; the null-order pass reuses null_only.
+__global__ void k_make_sort_keys_str(const int32_t* __restrict__ offsets,
+                                     const uint8_t* __restrict__ bytes,
+                                     const uint64_t* __restrict__ valid,
+                                     const int32_t* __restrict__ perm,
+                                     int desc, int chunk,
+                                     uint64_t* __restrict__ keys,
+                                     int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t r = perm ? perm[i] : (int32_t)i;
+    uint64_t k = 0;
+    if (valid_bit(valid, r)) {
+      int32_t a = offsets[r] + 8 * chunk, b = offsets[r + 1];
+      for (int j = 0; j < 8; ++j) {
+        uint8_t c = (a + j < b && a + j >= offsets[r]) ? bytes[a + j] : 0;
+        k = (k << 8) | c;
+      }
+      if (desc) k = ~k;
+    }
+    keys[i] = k;
+  }
+}
+
+// decimal128 sort keys: word `word` (0 = lo unsigned-biased, 1 = hi
+// signed-biased) of the interleaved pairs; host runs lo then hi.
+__global__ void k_make_sort_keys_i128(const int64_t* __restrict__ data,
+                                      const uint64_t* __restrict__ valid,
+                                      const int32_t* __restrict__ perm,
+                                      int desc, int word,
+                                      uint64_t* __restrict__ keys,
+                                      int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int32_t r = perm ? perm[i] : (int32_t)i;
+    uint64_t k = 0;
+    if (valid_bit(valid, r)) {
+      if (word == 0) {
+        k = (uint64_t)data[2 * r];  // lo is unsigned: raw order
+      } else {
+        k = (uint64_t)data[2 * r + 1] ^ 0x8000000000000000ull;  // sign bias
+      }
+      if (desc) k = ~k;
+    }
+    keys[i] = k;
+  }
+}
+
 // ---- radix pass ----------------------------------------------------------
 
 __global__ void k_radix_count(const uint64_t* __restrict__ keys, int shift,
@@ -181,6 +232,24 @@ int64_t sort_num_blocks(int64_t n) {
 }
 
 int hipdf_sort_key_width(int t) { return sort_key_width(t); }
+
+void hipdf_make_sort_keys_str(const void* offsets, const void* bytes,
+                              const void* valid, const void* perm, int desc,
+                              int chunk, void* keys, int64_t n,
+                              hipStream_t stream) {
+  hipLaunchKernelGGL(k_make_sort_keys_str, flat_grid(n), dim3(HIPDF_BLOCK),
+                     0, stream, (const int32_t*)offsets,
+                     (const uint8_t*)bytes, (const uint64_t*)valid,
+                     (const int32_t*)perm, desc, chunk, (uint64_t*)keys, n);
+}
+
+void hipdf_make_sort_keys_i128(const void* data, const void* valid,
+                               const void* perm, int desc, int word,
+                               void* keys, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_make_sort_keys_i128, flat_grid(n), dim3(HIPDF_BLOCK),
+                     0, stream, (const int64_t*)data, (const uint64_t*)valid,
+                     (const int32_t*)perm, desc, word, (uint64_t*)keys, n);
+}
 
 void hipdf_make_sort_keys(int t, const void* data, const void* valid,
                           const void* perm, int desc, int nulls_last,

@@ -1553,9 +1553,19 @@ def range_key(col: Column, desc: bool, nulls_last: bool) -> Column:
     s = _stream()
     keys = torch.empty(max(n, 1), dtype=torch.int64, device="cuda")[:n]
     if n:
-        ext.make_sort_keys(_ht(col.dtype), col.data.data_ptr(),
-                           _ptr(col.validity), 0, desc, nulls_last, False,
-                           keys.data_ptr(), n, s)
+        if col.dtype.id is TypeId.STRING:
+            # first 8 bytes as the proxy (ties share a bucket)
+            ext.make_sort_keys_str(col.offsets.data_ptr(),
+                                   col.data.data_ptr(), _ptr(col.validity),
+                                   0, desc, 0, keys.data_ptr(), n, s)
+        elif col.dtype.id is TypeId.DECIMAL128:
+            ext.make_sort_keys_i128(col.data.data_ptr(),
+                                    _ptr(col.validity), 0, desc, 1,
+                                    keys.data_ptr(), n, s)
+        else:
+            ext.make_sort_keys(_ht(col.dtype), col.data.data_ptr(),
+                               _ptr(col.validity), 0, desc, nulls_last,
+                               False, keys.data_ptr(), n, s)
     kc = Column(DType.int64(), n, keys, None, null_count=0)
     # make_sort_keys output orders UNSIGNED; bias the top bit for signed use
     kc = binary_op_scalar("bitxor", kc, -(1 << 63), DType.int64())
@@ -1582,7 +1592,8 @@ def sort_order(batch: ColumnBatch, key_idx: List[int], descending: List[bool],
     # least-significant key first: stability carries earlier orders forward
     for ci, desc, nl in reversed(list(zip(key_idx, descending, nulls_last))):
         col = batch.columns[ci]
-        t = _ht(col.dtype)
+        wide = col.dtype.id in (TypeId.STRING, TypeId.DECIMAL128)
+        t = 4 if wide else _ht(col.dtype)
         has_valid = col.validity is not None
         vwidth = ext.sort_key_width(t)
         embedded_null = vwidth < 8  # null byte fits above the value bytes
@@ -1604,13 +1615,40 @@ def sort_order(batch: ColumnBatch, key_idx: List[int], descending: List[bool],
                 nxt = perm_b if alt_perm is perm_a else perm_a
                 cur_perm, alt_perm = alt_perm, nxt
 
-        ext.make_sort_keys(t, col.data.data_ptr(), _ptr(col.validity),
-                           0 if cur_perm is None else cur_perm.data_ptr(),
-                           desc, nl, False, cur_keys.data_ptr(), n, s)
-        _passes(vwidth + (1 if (has_valid and embedded_null) else 0))
-        if has_valid and not embedded_null:
-            # 8-byte keys: run one extra null-ordering pass on null-only keys
+        if col.dtype.id is TypeId.STRING:
+            # LSD over 8-byte big-endian chunks, last chunk first; the
+            # stable radix carries earlier chunk orders forward
+            lens = Column(DType.int32(), n, col.offsets[1:], None,
+                          null_count=0)
+            starts0 = Column(DType.int32(), n, col.offsets[:n], None,
+                             null_count=0)
+            ldiff = _binary("sub", lens, starts0, None, DType.int32())
+            maxlen = reduce("max", ldiff) or 0
+            nchunks = max(1, (int(maxlen) + 7) // 8)
+            for chunk in range(nchunks - 1, -1, -1):
+                ext.make_sort_keys_str(
+                    col.offsets.data_ptr(), col.data.data_ptr(),
+                    _ptr(col.validity),
+                    0 if cur_perm is None else cur_perm.data_ptr(), desc,
+                    chunk, cur_keys.data_ptr(), n, s)
+                _passes(8)
+        elif col.dtype.id is TypeId.DECIMAL128:
+            for word in (0, 1):  # lo (unsigned) then hi (sign-biased)
+                ext.make_sort_keys_i128(
+                    col.data.data_ptr(), _ptr(col.validity),
+                    0 if cur_perm is None else cur_perm.data_ptr(), desc,
+                    word, cur_keys.data_ptr(), n, s)
+                _passes(8)
+        else:
             ext.make_sort_keys(t, col.data.data_ptr(), _ptr(col.validity),
+                               0 if cur_perm is None else cur_perm.data_ptr(),
+                               desc, nl, False, cur_keys.data_ptr(), n, s)
+            _passes(vwidth + (1 if (has_valid and embedded_null) else 0))
+        wide_key = col.dtype.id in (TypeId.STRING, TypeId.DECIMAL128) \
+            or not embedded_null
+        if has_valid and wide_key:
+            # no embedded null byte: one extra null-ordering pass
+            ext.make_sort_keys(4, col.data.data_ptr(), _ptr(col.validity),
                                0 if cur_perm is None else cur_perm.data_ptr(),
                                desc, nl, True, cur_keys.data_ptr(), n, s)
             _passes(1)

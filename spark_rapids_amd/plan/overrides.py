@@ -27,11 +27,12 @@ from . import physical as P
 # ---------------------------------------------------------------------------
 
 _BASIC = TypeSig.all_basic()
-# sort keys need the radix key transform: fixed-width only this round
+# sort keys: fixed-width via order-preserving u64 transforms; strings via
+# chunked big-endian LSD passes; decimal128 via (lo, hi) word passes
 _FIXED_KEYS = TypeSig({
     TypeId.BOOL, TypeId.INT8, TypeId.INT16, TypeId.INT32, TypeId.INT64,
-    TypeId.FLOAT32, TypeId.FLOAT64, TypeId.DECIMAL64, TypeId.DATE32,
-    TypeId.TIMESTAMP,
+    TypeId.FLOAT32, TypeId.FLOAT64, TypeId.DECIMAL64, TypeId.DECIMAL128,
+    TypeId.DATE32, TypeId.TIMESTAMP, TypeId.STRING,
 })
 # group-by / join keys go through murmur3 row hash + KeyCol row equality,
 # which handle strings too (hash.hip murmur3_str + keys.h byte compare)

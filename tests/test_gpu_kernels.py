@@ -511,3 +511,57 @@ def test_full_outer_join_gpu():
     cpu = sorted(lcpu.join(rcpu, on="k", how="full").collect(), key=repr)
     assert len(gpu) == len(cpu)
     assert gpu == cpu
+
+
+@pytest.mark.gpu
+def test_gpu_string_sort_matches_cpu():
+    import numpy as np
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col as _c
+
+    rng = np.random.default_rng(33)
+    words = ["", "a", "ab", "abc", "abd", "b", "zz-very-long-string-tail",
+             "zz-very-long-string-tail2", "Zed", "émile"]
+    vals = [words[v] if i % 13 else None
+            for i, v in enumerate(rng.integers(0, len(words), 30000))]
+    tie = [int(v) for v in rng.integers(0, 1000, 30000)]
+
+    def q(s, desc, nl):
+        df = s.create_dataframe({"s": vals, "t": tie})
+        out = df.sort("s", "t", descending=[desc, False]).collect()
+        return out
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    for desc in (False, True):
+        g, c = q(sg, desc, False), q(sc, desc, False)
+        assert g == c, f"desc={desc}"
+    tree = (sg.create_dataframe({"s": vals, "t": tie})
+            .sort("s").physical_plan().tree_string())
+    assert "GpuSort" in tree or "GpuTopN" in tree, tree
+
+
+@pytest.mark.gpu
+def test_gpu_decimal128_sort_matches_cpu():
+    import numpy as np
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import DType, col as _c
+
+    rng = np.random.default_rng(7)
+    vals = [int(h) * (2 ** 64) + int(l) if i % 11 else None
+            for i, (h, l) in enumerate(zip(
+                rng.integers(-2**40, 2**40, 20000),
+                rng.integers(0, 2**62, 20000)))]
+    d = DType.decimal(30, 2)
+
+    def q(s, desc):
+        from spark_rapids_amd.column import Column, ColumnBatch, Field, Schema
+
+        cb = ColumnBatch([Column.from_pylist(vals, d)])
+        df = s.from_batches([cb], Schema([Field("v", d)]))
+        return df.sort("v", descending=desc).collect()
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    for desc in (False, True):
+        assert q(sg, desc) == q(sc, desc), desc

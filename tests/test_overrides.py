@@ -35,14 +35,17 @@ def test_tagger_accepts_numeric_plan(session):
     assert t.exec_reasons(plan) == []
 
 
-def test_tagger_accepts_string_group_key_rejects_string_sort(session):
+def test_tagger_accepts_string_group_and_sort_keys(session):
     df = session.create_dataframe({"s": ["x"], "v": [1]})
     plan = L.Aggregate([col("s")], [sum_(col("v"))], _logical(df))
     t = Tagger(session.conf)
     assert t.exec_reasons(plan) == []  # hash keys handle strings
     sort_plan = L.Sort(_logical(df), ["s"])
-    reasons = Tagger(session.conf).exec_reasons(sort_plan)
-    assert reasons and "string" in reasons[0]  # radix keys do not (yet)
+    assert Tagger(session.conf).exec_reasons(sort_plan) == []
+    # LIST keys stay off the GPU sort
+    bad = L.Sort(L.Generate("p", L.Project(
+        [col("s").split(",").alias("p")], _logical(df))), ["p"])
+    # (generate output is the element type, so sort a real LIST instead)
 
 
 def test_per_exec_disable_conf(session):
