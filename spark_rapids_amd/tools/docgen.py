@@ -24,9 +24,9 @@ _EXECS = [
     ("CrossJoin", "cartesian gather maps (+ filter for non-equi)", "all"),
     ("Generate", "explode / posexplode (+_outer) over LIST columns",
      "outer pads on CPU"),
-    ("Sort", "stable LSD radix sort; out-of-core range-partitioned spill "
-     "buckets", "fixed-width keys on GPU; string/decimal128 sort keys "
-     "fall back"),
+    ("Sort", "stable LSD radix sort (incl. string + decimal128 keys); "
+     "out-of-core range-partitioned spill buckets; distributed "
+     "range-partitioned global ORDER BY; fused TopN", "all basic types"),
     ("Expand", "grouping-sets projections (rollup / cube)", "all"),
     ("Window", "ranking / running + bounded (ROWS and RANGE) + partition "
      "aggregates / lag / lead / ntile / nth_value",
