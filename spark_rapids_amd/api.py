@@ -104,9 +104,11 @@ class DataFrame:
              how: str = "inner", right_on: Optional[Sequence[str]] = None) -> "DataFrame":
         if isinstance(on, str):
             on = [on]
+        using = right_on is None
         r_on = list(right_on) if right_on is not None else list(on)
         return DataFrame(self.session,
-                         L.Join(self.plan, other.plan, list(on), r_on, how))
+                         L.Join(self.plan, other.plan, list(on), r_on, how,
+                                using=using))
 
     def cross_join(self, other: "DataFrame") -> "DataFrame":
         """Cartesian product; combine with filter() for non-equi joins

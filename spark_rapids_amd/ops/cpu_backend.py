@@ -896,9 +896,11 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
     keys = [batch.columns[i] for i in key_idx]
     n = batch.num_rows
     if not key_idx:
+        # keyless (global) aggregate: ALWAYS one group, even over zero
+        # rows — Spark returns a single row (count 0, sums null)
         codes = np.zeros(n, dtype=np.int64)
         first_idx = np.array([0] if n else [], dtype=np.int64)
-        ngroups = 1 if n else 0
+        ngroups = 1
     else:
         codes, first_idx = _group_codes(keys)
         ngroups = len(first_idx)

@@ -1149,7 +1149,7 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
     n = batch.num_rows if sel is None else sel.numel()
     s = _stream()
     keys = [batch.columns[i] for i in key_idx]
-    if n == 0:
+    if n == 0 and key_idx:
         out = [_empty_col(k.dtype) for k in keys]
         out += [_empty_col(dt) for _, _, dt in aggs]
         return ColumnBatch(out, 0)

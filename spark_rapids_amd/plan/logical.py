@@ -94,12 +94,18 @@ class Aggregate(LogicalPlan):
 
 class Join(LogicalPlan):
     def __init__(self, left: LogicalPlan, right: LogicalPlan,
-                 left_on: List[str], right_on: List[str], how: str = "inner"):
+                 left_on: List[str], right_on: List[str], how: str = "inner",
+                 using: bool = False):
         self.left = left
         self.right = right
         self.left_on = left_on
         self.right_on = right_on
         self.how = how
+        # USING-join semantics (join by shared column names): the
+        # duplicate right key columns are dropped for inner/left joins,
+        # matching Spark's df.join(other, "k"). Full outer keeps both
+        # (Spark would coalesce them; documented difference).
+        self.using = using and how in ("inner", "left")
 
     @property
     def children(self):
@@ -115,6 +121,8 @@ class Join(LogicalPlan):
                        for f in ls.fields]
         right_fields = []
         for f in rs.fields:
+            if self.using and f.name in self.right_on:
+                continue
             nullable = f.nullable or self.how in ("left", "full")
             right_fields.append(Field(f.name, f.dtype, nullable))
         return Schema(left_fields + right_fields)

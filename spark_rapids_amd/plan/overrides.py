@@ -405,7 +405,8 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
                               right_replicated=L.is_replicated(node.right),
                               broadcast_threshold=conf.get(BROADCAST_THRESHOLD),
                               sub_partition_bytes=conf.get(
-                                  JOIN_SUBPARTITION_BYTES))
+                                  JOIN_SUBPARTITION_BYTES),
+                              using=node.using)
     if isinstance(node, L.MapBatches):
         return P.MapBatchesExec(node.fn, _ensure_device(kids[0], "cpu"),
                                 node.schema())

@@ -317,7 +317,17 @@ class HashAggregateExec(PhysicalExec):
                 SpillableBatch(with_retry_split_single(task, batch)))
 
         if not partial_handles:
-            return
+            if nkeys:
+                return
+            # keyless aggregate over empty input: Spark still returns one
+            # row — aggregate an empty batch
+            cols = [Column.from_pylist([], e.dtype(in_schema))
+                    for e in value_exprs]
+            if self.gpu:
+                cols = [c.cuda() for c in cols]
+            specs0 = [(op, v, dt) for op, v, dt in partial]
+            partial_handles.append(SpillableBatch(ops.group_by_aggregate(
+                ColumnBatch(cols, 0), [], specs0)))
         partial_results = [h.get() for h in partial_handles]
         merged_in = ops.concat_batches(partial_results) if len(partial_results) > 1 \
             else partial_results[0]
@@ -432,8 +442,14 @@ class HashAggregateExec(PhysicalExec):
             if not received:
                 return
             pre = ops.concat_batches(received) if len(received) > 1                 else received[0]
-        if pre is None or pre.num_rows == 0:
+        if (pre is None or pre.num_rows == 0) and nkeys:
             return
+        if pre is None:
+            cols = [Column.from_pylist([], e.dtype(in_schema))
+                    for e in list(self.group_exprs) + list(value_exprs)]
+            if self.gpu:
+                cols = [c.cuda() for c in cols]
+            pre = ColumnBatch(cols, 0)
         specs = [(op, (nkeys + v) if v >= 0 else -1, dt)
                  for op, v, dt in partial]
         merged = ops.group_by_aggregate(pre, list(range(nkeys)), specs)
@@ -456,7 +472,8 @@ class HashJoinExec(PhysicalExec):
                  left_on: List[str], right_on: List[str], how: str,
                  schema: Schema, right_replicated: bool = True,
                  broadcast_threshold: int = 512 << 20,
-                 sub_partition_bytes: int = 1 << 30):
+                 sub_partition_bytes: int = 1 << 30,
+                 using: bool = False):
         super().__init__(device, schema, [left, right])
         self.left_on = left_on
         self.right_on = right_on
@@ -464,6 +481,7 @@ class HashJoinExec(PhysicalExec):
         self.right_replicated = right_replicated
         self.broadcast_threshold = broadcast_threshold
         self.sub_partition_bytes = sub_partition_bytes
+        self.using = using
         self._strategy = "local"
 
     def _local_or_empty(self, batches: List[ColumnBatch], schema: Schema):
@@ -558,7 +576,8 @@ class HashJoinExec(PhysicalExec):
             lout = ops.gather(lbatch, lmap)
             rout = ops.gather(rtable, rmap)
             if lout.num_rows:
-                yield ColumnBatch(lout.columns + rout.columns, lout.num_rows)
+                yield ColumnBatch(lout.columns +
+                                  self._right_out(rout), lout.num_rows)
         if self.how == "full" and rtable is not None and rtable.num_rows:
             extra = self._unmatched_right(rtable, right_matched, left.schema)
             if extra is not None and extra.num_rows:
@@ -619,7 +638,7 @@ class HashJoinExec(PhysicalExec):
                 lout = ops.gather(lb, lmap)
                 rout = ops.gather(rb, rmap)
                 if lout.num_rows:
-                    yield ColumnBatch(lout.columns + rout.columns,
+                    yield ColumnBatch(lout.columns + self._right_out(rout),
                                       lout.num_rows)
             if self.how == "full":
                 extra = self._unmatched_right(rb, right_matched, left_schema)
@@ -669,6 +688,15 @@ class HashJoinExec(PhysicalExec):
                              "cuda" if self.gpu else "cpu")
                 for f in left_schema.fields]
         return ColumnBatch(cols + runm.columns, runm.num_rows)
+
+    def _right_out(self, rout: ColumnBatch) -> List[Column]:
+        """Right-side output columns; USING joins drop the duplicate key
+        columns (kept on the left)."""
+        if not self.using:
+            return list(rout.columns)
+        rs = self.children[1].schema
+        return [c for f, c in zip(rs.fields, rout.columns)
+                if f.name not in self.right_on]
 
     def _left_with_null_right(self, lbatch: ColumnBatch) -> ColumnBatch:
         nsch = self.schema

@@ -4,7 +4,7 @@ import numpy as np
 import pytest
 
 import spark_rapids_amd as sr
-from spark_rapids_amd import col, count_star, sum_
+from spark_rapids_amd import avg, col, count_star, sum_
 
 
 def test_empty_table_through_operators(session):
@@ -16,7 +16,7 @@ def test_empty_table_through_operators(session):
     other = session.create_dataframe({"a": [1], "c": [2]})
     assert df.join(other, on="a").collect() == []
     assert other.join(df.select("a", "b"), on="a", how="left").collect() \
-        == [(1, 2, None, None)]
+        == [(1, 2, None)]
 
 
 def test_single_row(session):
@@ -65,7 +65,8 @@ def test_duplicate_column_names_join(session):
     left = session.create_dataframe({"k": [1, 2], "v": [10, 20]})
     right = session.create_dataframe({"k": [1, 2], "v": [30, 40]})
     out = sorted(left.join(right, on="k").collect())
-    assert out == [(1, 10, 1, 30), (2, 20, 2, 40)]
+    # USING join: duplicate right key dropped (Spark df.join(other, "k"))
+    assert out == [(1, 10, 30), (2, 20, 40)]
 
 
 def test_deep_expression_nesting(session):
@@ -82,3 +83,49 @@ def test_zero_partition_groupby(session):
                                   num_partitions=5)  # more parts than rows
     assert sorted(df.group_by("k").agg(sum_(col("v"))).collect()) \
         == [(1, 1.0), (2, 2.0)]
+
+
+class TestEmptyAndNullEdges:
+    @pytest.fixture
+    def cpu(self):
+        return sr.Session({"spark.rapids.sql.enabled": False})
+
+    def _empty(self, s):
+        df = s.create_dataframe({"k": [1], "v": [1.0], "s": ["x"]})
+        return df.filter(col("k") < 0)  # empty downstream
+
+    def test_empty_through_operators(self, cpu):
+        e = self._empty(cpu)
+        assert e.collect() == []
+        assert e.group_by("k").agg(sum_(col("v"))).collect() == []
+        assert e.agg(count_star(), sum_(col("v"))).collect() == [(0, None)]
+        assert e.sort("k").collect() == []
+        assert e.sort("k").limit(5).collect() == []
+        assert e.distinct().collect() == []
+        other = cpu.create_dataframe({"k": [1], "w": [2.0]})
+        assert e.join(other, on="k").collect() == []
+        assert other.join(e.select(col("k"), col("v")), on="k",
+                          how="left").collect() == [(1, 2.0, None)]
+        assert e.rollup("k").agg(count_star()).collect() == []
+
+    def test_all_null_join_keys_never_match(self, cpu):
+        l = cpu.create_dataframe({"k": [None, None], "v": [1, 2]})
+        r = cpu.create_dataframe({"k": [None], "w": [9]})
+        assert l.join(r, on="k").collect() == []
+        lj = sorted(l.join(r, on="k", how="left").collect(), key=repr)
+        assert lj == sorted([(None, 1, None), (None, 2, None)], key=repr)
+        assert l.join(r, on="k", how="anti").count() == 2
+
+    def test_null_group_key_single_group(self, cpu):
+        df = cpu.create_dataframe({"k": [None, None], "v": [1.0, 2.0]})
+        out = df.group_by("k").agg(sum_(col("v")), count_star()).collect()
+        assert out == [(None, 3.0, 2)]
+
+    def test_single_row_everything(self, cpu):
+        df = cpu.create_dataframe({"k": [7], "v": [1.5]})
+        assert df.sort("k").collect() == [(7, 1.5)]
+        assert df.group_by("k").agg(avg(col("v"))).collect() == [(7, 1.5)]
+        from spark_rapids_amd import win_sum
+
+        w = df.with_column("r", win_sum(col("v")).over(["k"], ["v"]))
+        assert w.collect() == [(7, 1.5, 1.5)]
