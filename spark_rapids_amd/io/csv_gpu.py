@@ -1,5 +1,7 @@
-"""GPU CSV decode (reference analogue: GpuCsvScan over cudf's CSV reader,
-SURVEY.md §2.3 CSV row).
+"""GPU CSV decode (reference analogue: GpuCsvScan —
+sql-plugin/src/main/scala/com/nvidia/spark/rapids/GpuBatchScanExec.scala
+GpuCsvScan + CSVPartitionReader — over cudf's CSV reader; SURVEY.md §2.3
+CSV row).
 
 The whole file is staged to device memory once; newline positions come
 from the byte-compare + stream-compaction kernels, and one k_csv_parse
