@@ -159,6 +159,8 @@ void hipdf_str_like(const void*, const void*, const void*, int, void*,
                     int64_t, hipStream_t);
 void hipdf_str_length(const void*, const void*, void*, int64_t, hipStream_t);
 void hipdf_str_case(int, const void*, void*, int64_t, hipStream_t);
+void hipdf_i64_to_str(const void*, const void*, void*, void*, int,
+                      int64_t, hipStream_t);
 void hipdf_str_concat_ws(const void*, int, const void*, int, const void*,
                          void*, void*, int, int64_t, hipStream_t);
 void hipdf_str_initcap(const void*, const void*, void*, int64_t,
@@ -611,6 +613,12 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("str_case", [](bool upper, int64_t in, int64_t out, int64_t nbytes,
                        int64_t stream) {
     hipdf_str_case(upper, P(in), PM(out), nbytes, S(stream));
+    check_async();
+  });
+  m.def("i64_to_str", [](int64_t vals, int64_t out_off, int64_t out_len,
+                         int64_t out, int mode, int64_t n, int64_t stream) {
+    hipdf_i64_to_str(P(vals), P(out_off), PM(out_len), PM(out), mode, n,
+                     S(stream));
     check_async();
   });
   m.def("str_concat_ws", [](int64_t cols, int ncols, int64_t sep,

@@ -147,6 +147,33 @@ __global__ void k_str_case(int upper, const uint8_t* __restrict__ in,
 // substring (1-based start in codepoints, length in codepoints; Spark
 // semantics: start 0 behaves like 1, negative counts from the end).
 // pass 1: byte [start,len) per row
+// int64 -> decimal string (cast long/int to string). mode 0: lengths,
+// mode 1: write digits.
+__global__ void k_i64_to_str(const int64_t* __restrict__ vals,
+                             const int64_t* __restrict__ out_off,
+                             int64_t* __restrict__ out_len,
+                             uint8_t* __restrict__ out, int mode,
+                             int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t v = vals[i];
+    uint64_t u = v < 0 ? (uint64_t)(-(v + 1)) + 1 : (uint64_t)v;
+    int digits = 1;
+    for (uint64_t t = u; t >= 10; t /= 10) ++digits;
+    int len = digits + (v < 0 ? 1 : 0);
+    if (!mode) {
+      out_len[i] = len;
+      continue;
+    }
+    uint8_t* dst = out + out_off[i];
+    if (v < 0) dst[0] = '-';
+    for (int d = digits - 1; d >= 0; --d) {
+      dst[(v < 0 ? 1 : 0) + d] = '0' + (uint8_t)(u % 10);
+      u /= 10;
+    }
+  }
+}
+
 // concat_ws: join n string columns with a separator, skipping NULL
 // values (result is never null — all-null rows give ""). Two-pass.
 struct StrColDesc {
@@ -437,6 +464,13 @@ void hipdf_str_case(int upper, const void* in, void* out, int64_t nbytes,
                     hipStream_t stream) {
   hipLaunchKernelGGL(k_str_case, flat_grid(nbytes), dim3(HIPDF_BLOCK), 0,
                      stream, upper, (const uint8_t*)in, (uint8_t*)out, nbytes);
+}
+
+void hipdf_i64_to_str(const void* vals, const void* out_off, void* out_len,
+                      void* out, int mode, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_i64_to_str, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int64_t*)vals, (const int64_t*)out_off,
+                     (int64_t*)out_len, (uint8_t*)out, mode, n);
 }
 
 void hipdf_str_concat_ws(const void* cols, int ncols, const void* sep,

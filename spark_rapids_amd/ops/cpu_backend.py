@@ -393,7 +393,12 @@ def cast(col: Column, to: DType) -> Column:
             if not av[i]:
                 continue
             try:
-                res[i] = float(v) if to.is_floating else int(float(v))
+                if to.is_decimal:
+                    res[i] = int(round(float(v) * (10 ** to.scale)))
+                elif to.is_floating:
+                    res[i] = float(v)
+                else:
+                    res[i] = int(float(v))
             except (ValueError, TypeError):
                 valid[i] = False
         return _make(res, valid if not valid.all() else None, to)

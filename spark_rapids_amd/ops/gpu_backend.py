@@ -742,11 +742,72 @@ def cast(col: Column, to: DType) -> Column:
     v = col.validity.clone() if col.validity is not None else None
     if col.dtype.is_decimal or to.is_decimal:
         return _cast_decimal(col, to, v)
-    if col.dtype.id is TypeId.STRING or to.id is TypeId.STRING:
-        raise NotImplementedError("string casts not on GPU yet")
+    if col.dtype.id is TypeId.STRING:
+        return _cast_string_to(col, to, v)
+    if to.id is TypeId.STRING:
+        return _cast_to_string(col, v)
     out = _alloc(n, to)
     ext.cast(_ht(col.dtype), _ht(to), col.data.data_ptr(), out.data_ptr(), n, s)
     return Column(to, n, out, v, null_count=col._null_count)
+
+
+def _cast_string_to(col: Column, to: DType, v) -> Column:
+    """string -> numeric cast (CastStrings analogue): trim, parse as f64
+    with the CSV field parser over each row's full span, then the generic
+    saturating cast; unparsable -> NULL (Spark non-ANSI)."""
+    if not (to.is_floating or to.is_integral or to.is_decimal):
+        raise NotImplementedError(f"gpu cast string -> {to}")
+    n = col.size
+    s = _stream()
+    trimmed = str_trim(col, "both")
+    f64 = torch.empty(max(n, 1), dtype=torch.float64, device="cuda")[:n]
+    valid_u8 = torch.empty(max(n, 1), dtype=torch.uint8, device="cuda")[:n]
+    unsupported = torch.zeros(1, dtype=torch.int32, device="cuda")
+    if n:
+        ext.csv_parse(trimmed.data.data_ptr(),
+                      trimmed.offsets[:n].data_ptr(),
+                      trimmed.offsets[1:].data_ptr(), 0, 0, 1,
+                      0, f64.data_ptr(), 0, 0, valid_u8.data_ptr(),
+                      unsupported.data_ptr(), n, s)
+    from ..column import mask_nbytes
+
+    v64 = torch.empty(max(n, 1), dtype=torch.int64, device="cuda")[:n]
+    if n:
+        ext.cast(0, 4, valid_u8.data_ptr(), v64.data_ptr(), n, s)
+    pmask = torch.empty(mask_nbytes(n), dtype=torch.uint8, device="cuda")
+    if n:
+        ext.mask_from_nonzero(v64.data_ptr(), pmask.data_ptr(), n, s)
+    mask = _and_masks(v, pmask) if v is not None else pmask
+    wide = Column(DType.float64(), n, f64, mask, null_count=None)
+    if to.id is TypeId.FLOAT64:
+        return wide
+    return cast(wide, to)
+
+
+def _cast_to_string(col: Column, v) -> Column:
+    """integral/date -> string; floats/decimals fall back (format parity
+    with python repr is CPU-side)."""
+    if not col.dtype.is_integral and col.dtype.id is not TypeId.BOOL:
+        raise NotImplementedError(f"gpu cast {col.dtype} -> string")
+    n = col.size
+    s = _stream()
+    i64 = cast(Column(col.dtype, n, col.data, None, null_count=0),
+               DType.int64()) if col.dtype.id is not TypeId.INT64         else Column(DType.int64(), n, col.data, None, null_count=0)
+    lens = torch.empty(max(n, 1), dtype=torch.int64, device="cuda")[:n]
+    if n:
+        ext.i64_to_str(i64.data.data_ptr(), 0, lens.data_ptr(), 0, 0, n, s)
+    scanned, total = _exclusive_scan_i64(lens) if n else (lens, 0)
+    out = torch.empty(max(total, 1), dtype=torch.uint8,
+                      device="cuda")[:total]
+    if total:
+        ext.i64_to_str(i64.data.data_ptr(), scanned.data_ptr(),
+                       lens.data_ptr(), out.data_ptr(), 1, n, s)
+    offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    if n:
+        ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+    offs[n] = total
+    return Column(DType.string(), n, out, v, offs,
+                  null_count=None if v is not None else 0)
 
 
 def _cast_decimal(col: Column, to: DType, v) -> Column:

@@ -1,6 +1,8 @@
 """coalesce / round / greatest / least / isin / date arithmetic tests."""
 import pytest
 
+import spark_rapids_amd as sr
+
 from spark_rapids_amd import (DATE32, Session, coalesce, col, date_add,
                               datediff, greatest, isin, least, lit, round_)
 
@@ -110,3 +112,52 @@ def test_to_date_unix_timestamp(session):
             datetime.timedelta(microseconds=u)
         assert d == (dt.date() - datetime.date(1970, 1, 1)).days
         assert sec == u // 1_000_000
+
+
+def test_string_casts(session):
+    from spark_rapids_amd import DType
+
+    df = session.create_dataframe({"s": ["1.25", " 42 ", "x", None, "",
+                                         "-7e2"]})
+    out = df.select(col("s").cast(DType.decimal(9, 2)).alias("d"),
+                    col("s").cast(sr.INT64).alias("i"),
+                    col("s").cast(sr.FLOAT64).alias("f")).to_pydict()
+    import decimal
+
+    assert out["d"] == [decimal.Decimal("1.25"), decimal.Decimal("42.00"),
+                        None, None, None, decimal.Decimal("-700.00")]
+    assert out["i"] == [1, 42, None, None, None, -700]
+    assert out["f"] == [1.25, 42.0, None, None, None, -700.0]
+
+
+def test_int_to_string_cast(session):
+    df = session.create_dataframe({"i": [0, -5, 123456789, None]})
+    out = df.select(col("i").cast(sr.STRING).alias("s")).to_pydict()
+    assert out["s"] == ["0", "-5", "123456789", None]
+
+
+@pytest.mark.gpu
+def test_gpu_string_casts_match_cpu():
+    import numpy as np
+
+    rng = np.random.default_rng(3)
+    vals = [f"{v:.3f}" if v % 7 else ("junk" if v % 5 else None)
+            for v in rng.uniform(-10**6, 10**6, 8000)]
+    ints = [int(v) for v in rng.integers(-10**12, 10**12, 8000)]
+
+    def q(s):
+        from spark_rapids_amd import DType
+
+        df = s.create_dataframe({"s": vals, "i": ints})
+        return df.select(col("s").cast(sr.FLOAT64).alias("f"),
+                         col("s").cast(sr.INT64).alias("n"),
+                         col("s").cast(DType.decimal(15, 2)).alias("d"),
+                         col("i").cast(sr.STRING).alias("t")).to_pydict()
+
+    g = q(sr.Session())
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert g["t"] == c["t"] and g["n"] == c["n"] and g["d"] == c["d"]
+    for a, b in zip(g["f"], c["f"]):
+        assert (a is None) == (b is None)
+        if a is not None:
+            assert a == pytest.approx(b, rel=1e-12)
