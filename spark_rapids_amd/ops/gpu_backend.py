@@ -740,12 +740,14 @@ def cast(col: Column, to: DType) -> Column:
     n = col.size
     s = _stream()
     v = col.validity.clone() if col.validity is not None else None
-    if col.dtype.is_decimal or to.is_decimal:
-        return _cast_decimal(col, to, v)
     if col.dtype.id is TypeId.STRING:
+        # must run before the decimal branch: string -> decimal parses as
+        # f64 first (truncating through int64 would drop the fraction)
         return _cast_string_to(col, to, v)
     if to.id is TypeId.STRING:
         return _cast_to_string(col, v)
+    if col.dtype.is_decimal or to.is_decimal:
+        return _cast_decimal(col, to, v)
     out = _alloc(n, to)
     ext.cast(_ht(col.dtype), _ht(to), col.data.data_ptr(), out.data_ptr(), n, s)
     return Column(to, n, out, v, null_count=col._null_count)
