@@ -207,3 +207,22 @@ def test_docgen_runs(tmp_path, monkeypatch):
     text = open(os.path.join(str(tmp_path), "docs",
                              "supported_ops.md")).read()
     assert "Expand" in text and "regexp_extract" in text
+
+
+def test_config_surface_documented():
+    """Every registered spark.rapids.* key renders into docs/configs.md."""
+    from spark_rapids_amd.config import _REGISTRY, help_doc
+
+    doc = help_doc()
+    public = [k for k, e in _REGISTRY.items()
+              if not getattr(e, "internal", False)]
+    assert len(public) >= 25
+    for k in public:
+        assert k in doc, k
+
+
+def test_expression_nullability():
+    s = Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({"a": [1, None], "b": [1.0, 2.0]})
+    sch = df.select((col("a") + col("b")).alias("x")).schema
+    assert sch.fields[0].nullable
