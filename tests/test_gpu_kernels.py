@@ -565,3 +565,11 @@ def test_gpu_decimal128_sort_matches_cpu():
     sc = sr.Session({"spark.rapids.sql.enabled": False})
     for desc in (False, True):
         assert q(sg, desc) == q(sc, desc), desc
+
+
+@pytest.mark.gpu
+def test_graft_smoke_entry():
+    """The driver's smoke() contract runs end-to-end on this box."""
+    import __graft_entry__
+
+    __graft_entry__.smoke()
