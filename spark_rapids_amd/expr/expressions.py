@@ -508,11 +508,15 @@ class CaseWhen(Expression):
     def dtype(self, schema: Schema) -> DType:
         t = self.branches[0][1].dtype(schema)
         for _, v in self.branches[1:]:
-            t = promote(t, v.dtype(schema))
+            vt = v.dtype(schema)
+            if t.id is TypeId.NULL:
+                t = vt
+            elif vt.id is not TypeId.NULL:
+                t = promote(t, vt)
         if self.else_expr is not None:
             et = self.else_expr.dtype(schema)
             if et.id is not TypeId.NULL:
-                t = promote(t, et)
+                t = promote(t, et) if t.id is not TypeId.NULL else et
         return t
 
     def eval(self, batch: ColumnBatch, schema: Schema) -> Column:

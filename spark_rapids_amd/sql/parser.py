@@ -588,8 +588,17 @@ class Parser:
             if distinct:
                 agg = A.AggExpr(agg.op, agg.child, distinct=True)
             return agg
-        if name == "coalesce":
+        if name in ("coalesce", "nvl", "ifnull"):
             return Coalesce(*args)
+        if name == "nullif":
+            from ..expr.expressions import CaseWhen, Literal
+
+            return CaseWhen([(BinaryExpr("eq_null_safe", args[0], args[1]),
+                              Literal(None))], args[0])
+        if name in ("greatest", "least"):
+            from ..expr.expressions import greatest as _g, least as _l
+
+            return (_g if name == "greatest" else _l)(*args)
         if name == "round":
             scale = args[1].value if len(args) > 1 else 0
             return Round(args[0], int(scale))

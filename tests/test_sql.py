@@ -191,3 +191,11 @@ def test_dataframe_ergonomics(session):
 
     with _p.raises(ValueError):
         df.union_by_name(session.create_dataframe({"a": [1]}))
+
+
+def test_sql_null_functions(session):
+    session.register("tnull", session.create_dataframe(
+        {"a": [1, None, 3], "b": [9, 9, 3]}))
+    out = session.sql("SELECT nvl(a, 0) x, nullif(a, b) y, "
+                      "greatest(a, b) g, least(a, b) l FROM tnull").collect()
+    assert out == [(1, 1, 9, 1), (0, None, 9, 9), (3, None, 3, 3)]
