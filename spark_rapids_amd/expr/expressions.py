@@ -162,6 +162,16 @@ class Expression:
     def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
         return RegexpExtract(self, pattern, group)
 
+    def lpad(self, width: int, fill: str = " ") -> "PadExpr":
+        return PadExpr(self, width, fill, left=True)
+
+    def rpad(self, width: int, fill: str = " ") -> "PadExpr":
+        return PadExpr(self, width, fill, left=False)
+
+    def locate(self, substr: str, pos: int = 1) -> "LocateExpr":
+        """1-based position of substr (0 when absent) — Spark locate/instr."""
+        return LocateExpr(self, substr, pos)
+
     def get_json_object(self, path: str) -> "GetJsonObject":
         return GetJsonObject(self, path)
 
@@ -655,6 +665,78 @@ class ArraySize(Expression):
 
     def __str__(self):
         return f"size({self.child})"
+
+
+class PadExpr(Expression):
+    """lpad/rpad to a fixed width (GpuStringLPad/RPad analogue; CPU
+    evaluation this round — tagged off the GPU by the overrides pass)."""
+
+    def __init__(self, child: Expression, width: int, fill: str, left: bool):
+        self.child = child
+        self.width = width
+        self.fill = fill or " "
+        self.left = left
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return STRING
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        from .. import ops as _ops
+
+        host = c if not c.is_cuda else c.cpu()
+        vals = host.to_pylist()
+        out = []
+        for v in vals:
+            if v is None:
+                out.append(None)
+            elif self.left:
+                out.append(v.rjust(self.width, self.fill[0])[: self.width]
+                           if len(v) < self.width else v[: self.width])
+            else:
+                out.append(v.ljust(self.width, self.fill[0])[: self.width]
+                           if len(v) < self.width else v[: self.width])
+        res = Column.from_pylist(out, STRING)
+        return res.cuda() if c.is_cuda else res
+
+    def __str__(self):
+        side = "lpad" if self.left else "rpad"
+        return f"{side}({self.child}, {self.width}, {self.fill!r})"
+
+
+class LocateExpr(Expression):
+    """locate(substr, str, pos): 1-based find, 0 when absent (GpuLocate)."""
+
+    def __init__(self, child: Expression, substr: str, pos: int = 1):
+        self.child = child
+        self.substr = substr
+        self.pos = pos
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return INT32
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        host = c if not c.is_cuda else c.cpu()
+        out = []
+        for v in host.to_pylist():
+            if v is None:
+                out.append(None)
+            else:
+                out.append(v.find(self.substr, max(self.pos - 1, 0)) + 1)
+        res = Column.from_pylist(out, INT32)
+        return res.cuda() if c.is_cuda else res
+
+    def __str__(self):
+        return f"locate({self.substr!r}, {self.child}, {self.pos})"
 
 
 class ConcatWs(Expression):

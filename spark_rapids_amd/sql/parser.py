@@ -627,6 +627,19 @@ class Parser:
             from ..expr.expressions import ArraySize
 
             return ArraySize(args[0])
+        if name in ("lpad", "rpad"):
+            from ..expr.expressions import PadExpr
+
+            fill = args[2].value if len(args) > 2 else " "
+            return PadExpr(args[0], int(args[1].value), fill,
+                           left=name == "lpad")
+        if name in ("locate", "instr"):
+            from ..expr.expressions import LocateExpr
+
+            if name == "locate":  # locate(substr, str[, pos])
+                pos = int(args[2].value) if len(args) > 2 else 1
+                return LocateExpr(args[1], args[0].value, pos)
+            return LocateExpr(args[0], args[1].value, 1)
         if name == "replace":
             return args[0].replace(args[1].value, args[2].value)
         if name == "substring" or name == "substr":

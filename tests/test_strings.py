@@ -374,3 +374,17 @@ class TestConcatWs:
         sg = sr.Session()
         sc = sr.Session({"spark.rapids.sql.enabled": False})
         assert q(sg) == q(sc)
+
+
+def test_pad_and_locate():
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({"s": ["ab", "abcdef", None]})
+    out = df.select(col("s").lpad(4, "0").alias("l"),
+                    col("s").rpad(4, "*").alias("r"),
+                    col("s").locate("b").alias("p")).to_pydict()
+    assert out["l"] == ["00ab", "abcd", None]
+    assert out["r"] == ["ab**", "abcd", None]
+    assert out["p"] == [2, 2, None]
+    s.register("tpad", df)
+    assert s.sql("SELECT instr(s, 'cd') FROM tpad").collect() == \
+        [(0,), (3,), (None,)]
