@@ -720,6 +720,12 @@ def concat_batches(batches: List[ColumnBatch]) -> ColumnBatch:
     cols = []
     for i in range(ncols):
         dtype = batches[0].columns[i].dtype
+        if dtype.id is TypeId.LIST:
+            vals = []
+            for b in batches:
+                vals.extend(b.columns[i].to_pylist())
+            cols.append(Column.from_pylist(vals, dtype))
+            continue
         if dtype.id is TypeId.STRING:
             vals = []
             for b in batches:

@@ -1026,6 +1026,10 @@ def concat_batches(batches: List[ColumnBatch]) -> ColumnBatch:
     out_cols = []
     for ci in range(ncols):
         ins = [b.columns[ci] for b in batches]
+        if ins[0].dtype.id is TypeId.LIST:
+            raise NotImplementedError(
+                "concat of LIST columns on GPU (explode or collect before "
+                "unioning, or keep the union on the CPU)")
         dtype = ins[0].dtype
         any_valid = any(c.validity is not None for c in ins)
         out_valid = None

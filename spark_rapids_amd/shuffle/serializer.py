@@ -36,6 +36,10 @@ def serialize_batch(batch: ColumnBatch) -> torch.Tensor:
     header = [n]
     sections: List[torch.Tensor] = []
     for c in batch.columns:
+        if c.dtype.id is TypeId.LIST:
+            raise NotImplementedError(
+                "LIST columns cannot be shuffled yet (collect/explode "
+                "happen after the exchange by design)")
         data_b = _as_bytes(c.data)
         header.extend([1 if c.validity is not None else 0, data_b.numel()])
         sections.append(data_b)
