@@ -703,6 +703,12 @@ def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> C
 def _gather_idx(batch: ColumnBatch, idx: np.ndarray, row_ok: np.ndarray) -> ColumnBatch:
     cols = []
     for c in batch.columns:
+        if c.dtype.id is TypeId.LIST:
+            vals_py = c.to_pylist()
+            out = [vals_py[i] if ok and 0 <= i < len(vals_py) else None
+                   for i, ok in zip(idx, row_ok)]
+            cols.append(Column.from_pylist(out, c.dtype))
+            continue
         a, av = _vals(c), _valid(c)
         if len(a) == 0:
             vals = np.zeros(len(idx), dtype=object if c.dtype.id is TypeId.STRING

@@ -262,3 +262,15 @@ def test_gpu_collect_strings_matches_cpu():
     sg = sr.Session()
     sc = sr.Session({"spark.rapids.sql.enabled": False})
     assert q(sg) == q(sc)
+
+
+def test_list_columns_through_sort_filter_union(cpu):
+    df = (cpu.create_dataframe({"k": [2, 1, 1, 3]})
+          .group_by("k").agg(collect_list(col("k"))))
+    assert df.sort("k").collect() == [(1, [1, 1]), (2, [2]), (3, [3])]
+    assert df.filter(col("k") > 1).sort("k").collect() == \
+        [(2, [2]), (3, [3])]
+    other = (cpu.create_dataframe({"k": [9]})
+             .group_by("k").agg(collect_list(col("k"))))
+    assert sorted(df.union(other).collect()) == \
+        [(1, [1, 1]), (2, [2]), (3, [3]), (9, [9])]
