@@ -155,6 +155,8 @@ class Column:
             n += self.validity.numel()
         if self.offsets is not None:
             n += self.offsets.numel() * 4
+        if self.child is not None:
+            n += self.child.nbytes
         return n
 
     # ---- movement -----------------------------------------------------
