@@ -317,17 +317,24 @@ class HashAggregateExec(PhysicalExec):
                 SpillableBatch(with_retry_split_single(task, batch)))
 
         if not partial_handles:
-            if nkeys:
+            from ..shuffle import dist as _d0
+
+            keyed_local = nkeys and not (_d0.ctx().is_multi
+                                         and not self.input_replicated)
+            if keyed_local:
                 return
-            # keyless aggregate over empty input: Spark still returns one
-            # row — aggregate an empty batch
+            # empty input still aggregates an empty batch: keyless Spark
+            # semantics need the one-row result, and a distributed keyed
+            # rank must still JOIN the exchange collectives (an early
+            # return here would deadlock the other ranks)
             cols = [Column.from_pylist([], e.dtype(in_schema))
-                    for e in value_exprs]
+                    for e in list(self.group_exprs) + list(value_exprs)]
             if self.gpu:
                 cols = [c.cuda() for c in cols]
-            specs0 = [(op, v, dt) for op, v, dt in partial]
+            specs0 = [(op, (nkeys + v) if v >= 0 else -1, dt)
+                      for op, v, dt in partial]
             partial_handles.append(SpillableBatch(ops.group_by_aggregate(
-                ColumnBatch(cols, 0), [], specs0)))
+                ColumnBatch(cols, 0), list(range(nkeys)), specs0)))
         partial_results = [h.get() for h in partial_handles]
         merged_in = ops.concat_batches(partial_results) if len(partial_results) > 1 \
             else partial_results[0]

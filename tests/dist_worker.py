@@ -134,6 +134,23 @@ def main():
         g = [r for r in allro if r[1] == 1]
         assert sum(r[2] for r in g) == n_total, g
 
+    # a rank with ZERO rows must still join every collective (keyed agg
+    # exchange would otherwise deadlock)
+    import spark_rapids_amd as _sr
+
+    empty_on_1 = s.create_dataframe(
+        {"k": [1, 2, 3] if rank == 0 else [],
+         "v": [1.0, 2.0, 3.0] if rank == 0 else []},
+        dtypes={"k": _sr.INT32, "v": _sr.FLOAT64})
+    ek = sorted(empty_on_1.group_by("k").agg(sum_(col("v"))).collect())
+    gathered_e = [None] * world
+    td.all_gather_object(gathered_e, ek)
+    if rank == 0:
+        alle = sorted(r for rs in gathered_e for r in rs)
+        assert alle == [(1, 1.0), (2, 2.0), (3, 3.0)], alle
+    tot_e = empty_on_1.agg(count_star()).collect()
+    assert tot_e == [(3,)], tot_e
+
     # distributed global ORDER BY: rank r holds the r-th sorted range and
     # the rank-order concatenation is the full global sort
     sorted_rows = df.sort("v", descending=True).collect()
