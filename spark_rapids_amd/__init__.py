@@ -14,7 +14,8 @@ from .expr.aggregates import (approx_count_distinct, approx_percentile,
                               count, count_distinct, count_star, first, last,
                               max_, min_, stddev, sum_distinct,
                               sum_, variance)
-from .expr.expressions import (CaseWhen, coalesce, col, concat_ws,
+from .expr.expressions import (CaseWhen, ascii_, coalesce, col, concat_ws,
+                               repeat_str, substring_index, translate,
                                date_add, date_sub,
                                datediff, dayofweek, greatest, hour, isin,
                                least, lit, quarter, to_date,

@@ -667,6 +667,68 @@ class ArraySize(Expression):
         return f"size({self.child})"
 
 
+class HostStringFn(Expression):
+    """A scalar string function evaluated on the host (rows through a
+    python callable); the overrides pass tags it off the GPU, so inside a
+    GPU project it executes via CpuBridge. Used for the long tail of
+    string builtins (repeat, substring_index, translate, ...) until they
+    earn kernels."""
+
+    def __init__(self, name: str, child: Expression, fn, out_dtype=None):
+        self.name = name
+        self.child = child
+        self.fn = fn
+        self._out = out_dtype or STRING
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return self._out
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        host = c if not c.is_cuda else c.cpu()
+        out = [None if v is None else self.fn(v) for v in host.to_pylist()]
+        res = Column.from_pylist(out, self._out)
+        return res.cuda() if c.is_cuda else res
+
+    def __str__(self):
+        return f"{self.name}({self.child})"
+
+
+def repeat_str(e, n: int) -> HostStringFn:
+    return HostStringFn("repeat", _as_expr(e), lambda v: v * n)
+
+
+def substring_index(e, delim: str, count: int) -> HostStringFn:
+    """Spark substring_index: text before the count-th delimiter
+    (negative count: after the count-th from the right)."""
+
+    def fn(v, d=delim, c=count):
+        parts = v.split(d)
+        if c > 0:
+            return d.join(parts[:c])
+        if c < 0:
+            return d.join(parts[c:])
+        return ""
+
+    return HostStringFn("substring_index", _as_expr(e), fn)
+
+
+def translate(e, src: str, repl: str) -> HostStringFn:
+    table = {ord(a): (repl[i] if i < len(repl) else None)
+             for i, a in enumerate(src)}
+    return HostStringFn("translate", _as_expr(e),
+                        lambda v: v.translate(table))
+
+
+def ascii_(e) -> HostStringFn:
+    return HostStringFn("ascii", _as_expr(e),
+                        lambda v: ord(v[0]) if v else 0, INT32)
+
+
 class PadExpr(Expression):
     """lpad/rpad to a fixed width (GpuStringLPad/RPad analogue; CPU
     evaluation this round — tagged off the GPU by the overrides pass)."""

@@ -147,8 +147,9 @@ class Tagger:
                 out.append("regex split delimiters run on CPU")
         elif type(e).__name__ in ("ArraySize", "ElementAt"):
             pass
-        elif type(e).__name__ in ("PadExpr", "LocateExpr"):
-            out.append(f"{type(e).__name__} runs on CPU this round")
+        elif type(e).__name__ in ("PadExpr", "LocateExpr", "HostStringFn"):
+            out.append(f"{getattr(e, 'name', type(e).__name__)} runs on "
+                       "CPU this round")
         elif type(e).__name__ in ("RegexpExtract", "RegexpReplace",
                                   "RegexpExtractAll"):
             from ..ops.regex_compiler import RegexUnsupported, compile_regex

@@ -531,6 +531,17 @@ class Parser:
         self.expect_kw("END")
         return CaseWhen(branches, else_e)
 
+    @staticmethod
+    def _int_arg(e) -> int:
+        from ..expr.expressions import Literal, UnaryExpr
+
+        if isinstance(e, Literal):
+            return int(e.value)
+        if isinstance(e, UnaryExpr) and e.op == "neg" and \
+                isinstance(e.child, Literal):
+            return -int(e.child.value)
+        raise SqlError(f"expected an integer literal, got {e}")
+
     def parse_func(self):
         name = self.next()[1].lower()
         self.expect_op("(")
@@ -555,13 +566,13 @@ class Parser:
         if name in ("lag", "lead"):
             from ..expr import windows as W
 
-            off = int(args[1].value) if len(args) > 1 else 1
+            off = self._int_arg(args[1]) if len(args) > 1 else 1
             dflt = args[2].value if len(args) > 2 else None
             return {"lag": W.lag, "lead": W.lead}[name](args[0], off, dflt)
         if name == "regexp_extract":
             from ..expr.expressions import RegexpExtract
 
-            idx = int(args[2].value) if len(args) > 2 else 1
+            idx = self._int_arg(args[2]) if len(args) > 2 else 1
             return RegexpExtract(args[0], args[1].value, idx)
         if name == "get_json_object":
             from ..expr.expressions import GetJsonObject
@@ -570,7 +581,7 @@ class Parser:
         if name == "regexp_extract_all":
             from ..expr.expressions import RegexpExtractAll
 
-            idx = int(args[2].value) if len(args) > 2 else 1
+            idx = self._int_arg(args[2]) if len(args) > 2 else 1
             return RegexpExtractAll(args[0], args[1].value, idx)
         if name == "regexp_replace":
             from ..expr.expressions import RegexpReplace
@@ -622,29 +633,46 @@ class Parser:
         if name == "element_at":
             from ..expr.expressions import ElementAt
 
-            return ElementAt(args[0], int(args[1].value))
+            return ElementAt(args[0], self._int_arg(args[1]))
         if name == "size":
             from ..expr.expressions import ArraySize
 
             return ArraySize(args[0])
+        if name == "repeat":
+            from ..expr.expressions import repeat_str
+
+            return repeat_str(args[0], self._int_arg(args[1]))
+        if name == "substring_index":
+            from ..expr.expressions import substring_index
+
+            return substring_index(args[0], args[1].value,
+                                   self._int_arg(args[2]))
+        if name == "translate":
+            from ..expr.expressions import translate
+
+            return translate(args[0], args[1].value, args[2].value)
+        if name == "ascii":
+            from ..expr.expressions import ascii_
+
+            return ascii_(args[0])
         if name in ("lpad", "rpad"):
             from ..expr.expressions import PadExpr
 
             fill = args[2].value if len(args) > 2 else " "
-            return PadExpr(args[0], int(args[1].value), fill,
+            return PadExpr(args[0], self._int_arg(args[1]), fill,
                            left=name == "lpad")
         if name in ("locate", "instr"):
             from ..expr.expressions import LocateExpr
 
             if name == "locate":  # locate(substr, str[, pos])
-                pos = int(args[2].value) if len(args) > 2 else 1
+                pos = self._int_arg(args[2]) if len(args) > 2 else 1
                 return LocateExpr(args[1], args[0].value, pos)
             return LocateExpr(args[0], args[1].value, 1)
         if name == "replace":
             return args[0].replace(args[1].value, args[2].value)
         if name == "substring" or name == "substr":
-            pos = int(args[1].value)
-            ln = int(args[2].value) if len(args) > 2 else -1
+            pos = self._int_arg(args[1])
+            ln = self._int_arg(args[2]) if len(args) > 2 else -1
             return Substring(args[0], pos, ln)
         raise SqlError(f"unknown function {name}")
 

@@ -388,3 +388,24 @@ def test_pad_and_locate():
     s.register("tpad", df)
     assert s.sql("SELECT instr(s, 'cd') FROM tpad").collect() == \
         [(0,), (3,), (None,)]
+
+
+def test_host_string_fns():
+    from spark_rapids_amd import (ascii_, repeat_str, substring_index,
+                                  translate)
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({"s": ["a.b.c", "xyz", None]})
+    out = df.select(substring_index(col("s"), ".", 2).alias("si"),
+                    substring_index(col("s"), ".", -1).alias("sn"),
+                    translate(col("s"), "abc", "ABC").alias("tr"),
+                    repeat_str(col("s"), 2).alias("rp"),
+                    ascii_(col("s")).alias("a")).to_pydict()
+    assert out["si"] == ["a.b", "xyz", None]
+    assert out["sn"] == ["c", "xyz", None]
+    assert out["tr"] == ["A.B.C", "xyz", None]
+    assert out["rp"] == ["a.b.ca.b.c", "xyzxyz", None]
+    assert out["a"] == [97, 120, None]
+    s.register("thsf", df)
+    assert s.sql("SELECT substring_index(s, '.', -1) FROM thsf").collect() \
+        == [("c",), ("xyz",), (None,)]
