@@ -544,6 +544,11 @@ def _infer_list_dtype(v: list) -> DType:
         if x is None:
             continue
         if dt is None:
+            if isinstance(x, list):
+                elem = next((e for e in x if e is not None), 0)
+                from .expr.expressions import _infer_literal_dtype as _ild
+
+                return DType.list_(_ild(elem))
             dt = _infer_literal_dtype(x)
             if dt.id is not TypeId.INT32:
                 return dt
