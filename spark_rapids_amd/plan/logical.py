@@ -76,6 +76,12 @@ class Project(LogicalPlan):
 class Aggregate(LogicalPlan):
     def __init__(self, group_exprs: List[Expression], aggs: List[AggExpr],
                  child: LogicalPlan):
+        cs = child.schema()
+        for e in group_exprs:
+            if e.dtype(cs).is_nested:
+                raise NotImplementedError(
+                    f"grouping by nested type {e.dtype(cs)} ({e}) is not "
+                    "supported — explode the array first")
         self.group_exprs = group_exprs
         self.aggs = aggs
         self.child = child

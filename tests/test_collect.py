@@ -274,3 +274,9 @@ def test_list_columns_through_sort_filter_union(cpu):
              .group_by("k").agg(collect_list(col("k"))))
     assert sorted(df.union(other).collect()) == \
         [(1, [1, 1]), (2, [2]), (3, [3]), (9, [9])]
+
+
+def test_nested_group_key_clear_error(cpu):
+    df = cpu.create_dataframe({"p": [["a"], ["b"]]})
+    with pytest.raises(NotImplementedError, match="nested"):
+        df.group_by("p").agg(count_star())
