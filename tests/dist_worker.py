@@ -151,6 +151,19 @@ def main():
     tot_e = empty_on_1.agg(count_star()).collect()
     assert tot_e == [(3,)], tot_e
 
+    # shuffled join + sub-partitioned build in one plan: tiny broadcast
+    # threshold forces the hash-exchange strategy, tiny subPartition
+    # bytes forces the bucketed join of the exchanged build side
+    s3 = Session({"spark.rapids.sql.enabled": False,
+                  "spark.rapids.sql.join.broadcastThreshold": 1,
+                  "spark.rapids.sql.join.subPartition.targetBytes": 512})
+    df3 = s3.create_dataframe({"k": (rows % 97),
+                               "v": rows.astype(np.float64)})
+    right3 = s3.create_dataframe({"k": right_rows % 97,
+                                  "r": right_rows * 100})
+    j4 = df3.join(right3, on="k").agg(count_star()).collect()
+    assert j4[0][0] == exp, (j4, exp, "shuffled+subpartitioned")
+
     # distributed global ORDER BY: rank r holds the r-th sorted range and
     # the rank-order concatenation is the full global sort
     sorted_rows = df.sort("v", descending=True).collect()
