@@ -226,3 +226,25 @@ def test_expression_nullability():
     df = s.create_dataframe({"a": [1, None], "b": [1.0, 2.0]})
     sch = df.select((col("a") + col("b")).alias("x")).schema
     assert sch.fields[0].nullable
+
+
+def test_lore_replay_roundtrip(tmp_path):
+    """Dump an operator's output, then replay it as a DataFrame."""
+    import glob
+    import os
+
+    import spark_rapids_amd as sr
+    from spark_rapids_amd.tools import lore
+
+    s = sr.Session({"spark.rapids.sql.enabled": False,
+                    "spark.rapids.sql.lore.dumpPath": str(tmp_path)})
+    df = s.create_dataframe({"a": [3, 1, 2], "b": [1.0, 2.0, 3.0]})
+    expected = df.filter(col("a") > 1).collect()
+    df.filter(col("a") > 1).collect()
+    dirs = [d for d in glob.glob(os.path.join(str(tmp_path), "*Filter*"))
+            if os.path.isdir(d)]
+    assert dirs, os.listdir(str(tmp_path))
+    clean = sr.Session({"spark.rapids.sql.enabled": False})
+    lore.configure("")
+    replayed = sorted(lore.replay(clean, dirs[0]).collect())
+    assert replayed == sorted(expected)
