@@ -172,3 +172,35 @@ def test_gpu_downscale_cast_half_up_negatives():
         return df.select(col("a").cast(DType.decimal(9, 1)).alias("r"),
                          (col("a") * col("a")).alias("sq")).to_pydict()
     assert q(sg) == q(sc)
+
+
+def test_d128_boundary_values(cpu):
+    """int128 boundary arithmetic on the CPU reference (add/sub near
+    +-2^126, compare ordering)."""
+    from spark_rapids_amd import Column, DType
+    from spark_rapids_amd.column import ColumnBatch, Field, Schema
+
+    d = DType.decimal(38, 0)
+    big = 2 ** 126
+    vals = [big, -big, big - 1, -(big - 1), 0, None]
+    cb = ColumnBatch([Column.from_pylist(vals, d),
+                      Column.from_pylist([1] * 6, d)])
+    df = cpu.from_batches([cb], Schema([Field("v", d), Field("o", d)]))
+    out = df.select((col("v") + col("o")).alias("p"),
+                    (col("v") - col("o")).alias("m")).to_pydict()
+    assert int(out["p"][0]) == big + 1
+    assert int(out["m"][1]) == -big - 1
+    assert out["p"][5] is None
+    srt = [r[0] for r in df.sort("v").collect()]
+    assert srt[0] is None  # nulls first asc
+    nn = [int(v) for v in srt[1:]]
+    assert nn == sorted(nn)
+
+
+def test_decimal_div_precision_loss_adjustment(cpu):
+    """allowPrecisionLoss path: (38,10)/(38,10) adjusts to scale 6."""
+    from spark_rapids_amd.types import decimal_arith_type
+
+    d = DType.decimal(38, 10)
+    t = decimal_arith_type("div", d, d)
+    assert t.precision == 38 and t.scale == 6
