@@ -90,6 +90,8 @@ void hipdf_i128_arith(int, const void*, const void*, const void*, const void*,
 void hipdf_i128_cmp(int, const void*, const void*, const void*, const void*,
                     void*, void*, int64_t, hipStream_t);
 void hipdf_i64_to_i128(const void*, void*, int64_t, hipStream_t);
+void hipdf_i128_rescale(const void*, const void*, void*, void*, int, int,
+                        int, int64_t, hipStream_t);
 void hipdf_i128_to_f64(const void*, void*, int64_t, hipStream_t);
 void hipdf_gb_percentile(const void*, const void*, const void*, const void*,
                          double, void*, int, hipStream_t);
@@ -703,6 +705,12 @@ PYBIND11_MODULE(hipdf, m) {
     hipdf_i128_cmp(op, P(a), P(b), P(av), P(bv), PM(out), PM(ov), n,
                    S(stream));
     check_async();
+  });
+  m.def("i128_rescale", [](int64_t in, int64_t iv, int64_t out, int64_t ov,
+                           int shift, int out_prec, int out_is_64, int64_t n,
+                           int64_t stream) {
+    hipdf_i128_rescale(P(in), P(iv), PM(out), PM(ov), shift, out_prec,
+                       out_is_64, n, S(stream));
   });
   m.def("i64_to_i128", [](int64_t in, int64_t out, int64_t n, int64_t stream) {
     hipdf_i64_to_i128(P(in), PM(out), n, S(stream));

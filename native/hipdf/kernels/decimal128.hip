@@ -317,7 +317,70 @@ __global__ void k_dec64_mul_div(int is_div, const int64_t* __restrict__ a,
   }
 }
 
+// rescale int128 decimal values by 10^|shift|: shift>0 multiplies (null on
+// u128 overflow), shift<0 divides with HALF_UP on the magnitude. Result is
+// nulled when |v| >= 10^out_prec (Spark non-ANSI cast overflow -> null;
+// out_prec<=38 so the bound fits u128). out_is_64 narrows to int64 words
+// (decimal128 -> decimal64 cast), otherwise writes (lo,hi) pairs.
+__global__ void k_i128_rescale(const int64_t* __restrict__ in,
+                               const uint64_t* __restrict__ iv,
+                               int64_t* __restrict__ out,
+                               uint64_t* __restrict__ ov, int shift,
+                               int out_prec, int out_is_64, int64_t nstripe,
+                               int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  const u128 u128_max = ~(u128)0;
+  for (int64_t s = wave_global; s < nstripe; s += wave_count) {
+    int64_t row = s * WAVE + lane;
+    bool ok = false;
+    if (row < n) {
+      ok = valid_bit(iv, row);
+      u128 q = 0;
+      bool neg = false;
+      if (ok) {
+        i128 v = load128(in, row);
+        neg = v.hi < 0;
+        i128 m = neg ? neg128(v) : v;
+        q = ((u128)(uint64_t)m.hi << 64) | m.lo;
+        if (shift > 0) {
+          u128 p = pow10_128(shift);
+          if (q > u128_max / p) ok = false;
+          else q *= p;
+        } else if (shift < 0) {
+          u128 d = pow10_128(-shift);
+          u128 r = q % d;
+          q /= d;
+          if (2 * r >= d) q += 1;
+        }
+        if (ok && out_prec <= 38 && q >= pow10_128(out_prec)) ok = false;
+        if (ok && out_is_64 && q > (u128)0x7fffffffffffffffULL) ok = false;
+      }
+      __int128 w = ok ? (neg ? -(__int128)q : (__int128)q) : 0;
+      if (out_is_64) {
+        out[row] = (int64_t)w;
+      } else {
+        out[2 * row] = (int64_t)(u128)w;
+        out[2 * row + 1] = (int64_t)((u128)w >> 64);
+      }
+    }
+    uint64_t ballot = __ballot(ok);
+    write_valid_word(ov, s, ballot, lane);
+  }
+}
+
 extern "C" {
+
+void hipdf_i128_rescale(const void* in, const void* iv, void* out, void* ov,
+                        int shift, int out_prec, int out_is_64, int64_t n,
+                        hipStream_t stream) {
+  int64_t nstripe = (n + WAVE - 1) / WAVE;
+  hipLaunchKernelGGL(k_i128_rescale, stripe_grid(nstripe), dim3(HIPDF_BLOCK),
+                     0, stream, (const int64_t*)in, (const uint64_t*)iv,
+                     (int64_t*)out, (uint64_t*)ov, shift, out_prec, out_is_64,
+                     nstripe, n);
+}
 
 void hipdf_dec64_mul_div(int is_div, const void* a, const void* b,
                          const void* av, const void* bv, void* out, void* ov,

@@ -11,7 +11,7 @@ from typing import Optional, Sequence
 
 from ..column import Column, ColumnBatch, Schema
 from ..types import (BOOL, DType, FLOAT64, INT32, INT64, STRING, TypeId,
-                     as_decimal, decimal_arith_type, promote)
+                     _adjust_decimal, as_decimal, decimal_arith_type, promote)
 from .. import ops
 
 
@@ -328,7 +328,9 @@ class BinaryExpr(Expression):
             return INT32  # datediff domain
         it = self._common(lt, rt)
         if it.is_decimal and self.op in ("add", "sub"):
-            return DType.decimal(min(it.precision + 1, 38), it.scale)
+            # Spark add/sub result: max(p1-s1,p2-s2)+max(s1,s2)+1 at
+            # scale max(s1,s2); promote() already produced the first part.
+            return _adjust_decimal(it.precision + 1, it.scale)
         return it
 
     # ops where `scalar OP col` can run through the col-scalar kernel
@@ -350,6 +352,11 @@ class BinaryExpr(Expression):
         if self.op in _DOUBLE_OPS and not common.is_decimal:
             common = FLOAT64
         out = self.dtype(schema)
+        if out.is_decimal and out.id is TypeId.DECIMAL128 \
+                and common.id is TypeId.DECIMAL64:
+            # result crosses into decimal128 (e.g. (18,0)+(18,0) -> (19,0)):
+            # widen the operands so the 128-bit kernel runs end to end
+            common = DType(TypeId.DECIMAL128, common.precision, common.scale)
         # scalar fast path: literal on either side
         if isinstance(self.right, Literal) and self.right.value is not None \
                 and not common.is_decimal and self.op != "concat":

@@ -403,22 +403,12 @@ def cast(col: Column, to: DType) -> Column:
                 valid[i] = False
         return _make(res, valid if not valid.all() else None, to)
     if src.id is TypeId.DECIMAL128 or to.id is TypeId.DECIMAL128:
-        if src.is_decimal and to.id is TypeId.DECIMAL128:
-            shift = to.scale - src.scale
-            vals = np.array([int(v) * (10 ** shift) if shift >= 0 else
-                             int(v) // (10 ** -shift) for v in a],
-                            dtype=object)
-            return _make(vals, av if not av.all() else None, to)
+        if src.is_decimal and to.is_decimal:
+            vals, valid = _rescale_exact(a, av, src, to)
+            return _make(vals, valid if not valid.all() else None, to)
         if src.id is TypeId.DECIMAL128 and to.is_floating:
             f = np.array([float(int(v)) / (10 ** src.scale) for v in a])
             return _make(f.astype(to.numpy_dtype()),
-                         av if not av.all() else None, to)
-        if src.id is TypeId.DECIMAL128 and to.id is TypeId.DECIMAL64:
-            shift = to.scale - src.scale
-            vals = np.array([int(v) * (10 ** shift) if shift >= 0 else
-                             int(v) // (10 ** -shift) for v in a],
-                            dtype=object)
-            return _make(np.array([int(v) for v in vals], dtype=np.int64),
                          av if not av.all() else None, to)
         if to.id is TypeId.DECIMAL128:
             scaled = np.array([int(round(float(v) * (10 ** to.scale)))
@@ -426,9 +416,8 @@ def cast(col: Column, to: DType) -> Column:
             return _make(scaled, av if not av.all() else None, to)
         raise NotImplementedError(f"cast {src} -> {to}")
     if src.is_decimal and to.is_decimal:
-        shift = to.scale - src.scale
-        res = a * (10 ** shift) if shift >= 0 else _round_half_up_div(a, 10 ** (-shift))
-        return _make(res, av if not av.all() else None, to)
+        vals, valid = _rescale_exact(a, av, src, to)
+        return _make(vals, valid if not valid.all() else None, to)
     if src.is_decimal:
         f = a.astype(np.float64) / (10 ** src.scale)
         res = f.astype(to.numpy_dtype())
@@ -456,6 +445,36 @@ def cast(col: Column, to: DType) -> Column:
     with np.errstate(invalid="ignore", over="ignore"):
         res = a.astype(to.numpy_dtype())
     return _make(res, av if not av.all() else None, to)
+
+
+def _rescale_exact(a, av, src: DType, to: DType):
+    """Exact decimal->decimal rescale in Python ints (no int64 overflow),
+    HALF_UP on down-shift, NULL where the result exceeds the target
+    precision or an int64 backing (Spark non-ANSI overflow -> null)."""
+    shift = to.scale - src.scale
+    bound = 10 ** to.precision
+    up = 10 ** shift if shift >= 0 else None
+    down = 10 ** -shift if shift < 0 else None
+    valid = av.copy()
+    out = []
+    for i, v in enumerate(a):
+        if not av[i]:
+            out.append(0)
+            continue
+        x = int(v)
+        if up is not None:
+            x *= up
+        else:
+            sign = -1 if x < 0 else 1
+            x = sign * ((2 * abs(x) + down) // (2 * down))
+        if abs(x) >= bound or (to.id is TypeId.DECIMAL64
+                               and abs(x) > 0x7FFFFFFFFFFFFFFF):
+            valid[i] = False
+            x = 0
+        out.append(x)
+    if to.id is TypeId.DECIMAL128:
+        return np.array(out, dtype=object), valid
+    return np.array(out, dtype=np.int64), valid
 
 
 def _round_half_up_div(a: np.ndarray, d: int) -> np.ndarray:

@@ -816,13 +816,31 @@ def _cast_decimal(col: Column, to: DType, v) -> Column:
     n = col.size
     s = _stream()
     if col.dtype.id is TypeId.DECIMAL128 or to.id is TypeId.DECIMAL128:
-        if col.dtype.id is TypeId.DECIMAL64 and to.id is TypeId.DECIMAL128:
-            src = col
-            if to.scale != col.dtype.scale:
-                src = _cast_decimal(col, DType.decimal(18, to.scale), v)
-            out = _alloc(n, to)
-            ext.i64_to_i128(src.data.data_ptr(), out.data_ptr(), n, s)
-            return Column(to, n, out, v, null_count=col._null_count)
+        if col.dtype.is_decimal and to.is_decimal:
+            # Widen to int128 FIRST, then rescale exactly in 128-bit with
+            # HALF_UP down-shift and null-on-overflow (Spark non-ANSI).
+            # Rescaling inside int64 before widening silently overflows
+            # (decimal(18,0) + decimal(18,10)).
+            if col.dtype.id is TypeId.DECIMAL64:
+                wide = torch.empty(max(2 * n, 1), dtype=torch.int64,
+                                   device="cuda")[:2 * n]
+                if n:
+                    ext.i64_to_i128(col.data.data_ptr(), wide.data_ptr(),
+                                    n, s)
+                src128 = wide
+            else:
+                src128 = col.data
+            shift = to.scale - col.dtype.scale
+            out_is_64 = to.id is TypeId.DECIMAL64
+            width = n if out_is_64 else 2 * n
+            out = torch.empty(max(width, 1), dtype=torch.int64,
+                              device="cuda")[:width]
+            ov = _alloc_mask(n)
+            if n:
+                ext.i128_rescale(src128.data_ptr(), _ptr(v), out.data_ptr(),
+                                 ov.data_ptr(), shift, to.precision,
+                                 1 if out_is_64 else 0, n, s)
+            return Column(to, n, out, ov, null_count=None)
         if col.dtype.id is TypeId.DECIMAL128 and to.is_floating:
             dbl = torch.empty(max(n, 1), dtype=torch.float64,
                               device="cuda")[:n]
@@ -834,10 +852,14 @@ def _cast_decimal(col: Column, to: DType, v) -> Column:
             scaled = Column(DType.float64(), n, scaled.data, v,
                             null_count=col._null_count)
             return cast(scaled, to) if to.id is not TypeId.FLOAT64 else scaled
-        if col.dtype.id is TypeId.DECIMAL128 and to.id is TypeId.DECIMAL128 \
-                and to.scale == col.dtype.scale:
-            return Column(to, n, col.data.clone(), v,
-                          null_count=col._null_count)
+        if col.dtype.is_integral and to.id is TypeId.DECIMAL128:
+            # integral -> decimal128: treat as decimal(scale 0) and rescale
+            i64 = col if col.dtype.id is TypeId.INT64 \
+                else cast(Column(col.dtype, n, col.data, None, null_count=0),
+                          DType.int64())
+            as_dec = Column(DType.decimal(18, 0), n, i64.data, v,
+                            null_count=col._null_count)
+            return _cast_decimal(as_dec, to, v)
         raise NotImplementedError(f"gpu cast {col.dtype} -> {to}")
     if col.dtype.is_decimal and to.is_decimal:
         shift = to.scale - col.dtype.scale

@@ -252,8 +252,13 @@ def promote(a: DType, b: DType) -> DType:
     if a == b:
         return a
     if a.is_decimal and b.is_decimal:
-        return DType.decimal(max(a.precision, b.precision),
-                             max(a.scale, b.scale))
+        # Spark DecimalPrecision widening (add/sub/compare common type):
+        # scale = max(s1, s2); integral digits = max(p1-s1, p2-s2); routing
+        # to DECIMAL128 when precision > 18 so the rescale of the
+        # smaller-scale operand cannot overflow int64.
+        s = max(a.scale, b.scale)
+        p = max(a.precision - a.scale, b.precision - b.scale) + s
+        return _adjust_decimal(p, s)
     if a.is_decimal or b.is_decimal:
         # decimal + integral -> decimal with enough precision; handled by caller
         d = a if a.is_decimal else b

@@ -204,3 +204,85 @@ def test_decimal_div_precision_loss_adjustment(cpu):
     d = DType.decimal(38, 10)
     t = decimal_arith_type("div", d, d)
     assert t.precision == 38 and t.scale == 6
+
+
+# ---- ADVICE.md (round 1, high): Spark DecimalPrecision widening ----------
+
+def test_promote_widening_add_sub_compare():
+    from spark_rapids_amd.types import promote
+    # decimal(18,0) vs decimal(18,10): common = (28,10) -> DECIMAL128
+    t = promote(DType.decimal(18, 0), DType.decimal(18, 10))
+    assert (t.precision, t.scale) == (28, 10)
+    from spark_rapids_amd.types import TypeId
+    assert t.id is TypeId.DECIMAL128
+
+
+def test_add_large_scale_mismatch_exact(cpu):
+    # ADVICE repro: cast(1e17 as decimal(18,0)) + cast(1 as decimal(18,10))
+    df = cpu.create_dataframe({"a": [100000000000000000], "b": [1]})
+    df = df.select(col("a").cast(DType.decimal(18, 0)).alias("a"),
+                   col("b").cast(DType.decimal(18, 10)).alias("b"))
+    out = df.select((col("a") + col("b")).alias("s")).to_pydict()["s"]
+    assert out[0] == Decimal("100000000000000001.0000000000")
+
+
+def test_compare_large_scale_mismatch(cpu):
+    df = cpu.create_dataframe({"a": [100000000000000000], "b": [1]})
+    df = df.select(col("a").cast(DType.decimal(18, 0)).alias("a"),
+                   col("b").cast(DType.decimal(18, 10)).alias("b"))
+    out = df.select((col("a") > col("b")).alias("g")).to_pydict()["g"]
+    assert out[0] is True
+
+
+def test_add_sub_result_dtype_spark_rules(cpu):
+    # (p1,s1)=(18,0), (p2,s2)=(18,10): add -> (29,10)
+    df = cpu.create_dataframe({"a": [1], "b": [1]})
+    df = df.select(col("a").cast(DType.decimal(18, 0)).alias("a"),
+                   col("b").cast(DType.decimal(18, 10)).alias("b"))
+    out = df.select((col("a") + col("b")).alias("s"))
+    t = out.schema.field("s").dtype
+    assert (t.precision, t.scale) == (29, 10)
+
+
+def test_rescale_overflow_is_null(cpu):
+    # value too big for the target precision -> NULL, not garbage
+    df = cpu.create_dataframe({"a": [999999999, 1]})
+    df = df.select(col("a").cast(DType.decimal(9, 0)).alias("a"))
+    out = df.select(col("a").cast(DType.decimal(5, 2)).alias("c")) \
+        .to_pydict()["c"]
+    assert out[0] is None
+    assert out[1] == Decimal("1.00")
+
+
+@pytest.mark.gpu
+def test_gpu_add_large_scale_mismatch_exact():
+    sg = sr.Session()
+    df = sg.create_dataframe({"a": [100000000000000000, None, -7],
+                              "b": [1, 5, 23]})
+    df = df.select(col("a").cast(DType.decimal(18, 0)).alias("a"),
+                   col("b").cast(DType.decimal(18, 10)).alias("b"))
+    out = df.select((col("a") + col("b")).alias("s"),
+                    (col("a") > col("b")).alias("g")).to_pydict()
+    assert out["s"][0] == Decimal("100000000000000001.0000000000")
+    assert out["s"][1] is None
+    assert out["s"][2] == Decimal("-6.9999999977")
+    assert out["g"][0] is True
+    assert out["g"][1] is None
+    assert out["g"][2] is False
+
+
+@pytest.mark.gpu
+def test_gpu_rescale_roundtrip_and_overflow():
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+
+    def q(s):
+        df = s.create_dataframe(
+            {"a": [12345678901234567, -999995, 0, None, 55]})
+        df = df.select(col("a").cast(DType.decimal(17, 0)).alias("a"))
+        return df.select(
+            col("a").cast(DType.decimal(27, 10)).alias("up128"),
+            col("a").cast(DType.decimal(7, 1)).alias("narrow"),
+        ).to_pydict()
+
+    assert q(sg) == q(sc)
