@@ -82,7 +82,9 @@ __global__ void k_str_find(int mode, const int32_t* __restrict__ ao,
   }
 }
 
-// SQL LIKE: % any-run, _ one char (byte-approx: one codepoint via lead byte)
+// SQL LIKE: % any-run, _ one char (byte-approx: one codepoint via lead
+// byte); backslash escapes the next pattern char (Spark's default LIKE
+// escape: \% and \_ match literally).
 __device__ bool like_match(const uint8_t* s, int32_t sl, const uint8_t* p,
                            int32_t pl) {
   int32_t si = 0, pi = 0, star_p = -1, star_s = 0;
@@ -90,16 +92,26 @@ __device__ bool like_match(const uint8_t* s, int32_t sl, const uint8_t* p,
     if (pi < pl && p[pi] == '%') {
       star_p = ++pi;
       star_s = si;
-    } else if (pi < pl && (p[pi] == '_' || p[pi] == s[si])) {
-      if (p[pi] == '_') {
-        // skip one UTF-8 codepoint
+      continue;
+    }
+    if (pi < pl && p[pi] == '\\' && pi + 1 < pl) {
+      if (p[pi + 1] == s[si]) {
         ++si;
-        while (si < sl && (s[si] & 0xC0) == 0x80) ++si;
-      } else {
-        ++si;
+        pi += 2;
+        continue;
       }
+    } else if (pi < pl && p[pi] == '_') {
+      // skip one UTF-8 codepoint
+      ++si;
+      while (si < sl && (s[si] & 0xC0) == 0x80) ++si;
       ++pi;
-    } else if (star_p >= 0) {
+      continue;
+    } else if (pi < pl && p[pi] == s[si]) {
+      ++si;
+      ++pi;
+      continue;
+    }
+    if (star_p >= 0) {
       pi = star_p;
       si = ++star_s;
     } else {

@@ -121,6 +121,24 @@ class DataFrame:
             desc = [descending] * len(ks)
         else:
             desc = list(descending)
+        cs = self.plan.schema()
+        missing = [k for k in ks if k not in cs.names]
+        if missing:
+            # Spark allows ORDER BY on input columns the projection dropped:
+            # re-plan as project+passthrough -> sort -> drop (ADVICE.md
+            # round 1: SELECT rank() OVER (...) FROM t ORDER BY g).
+            if isinstance(self.plan, L.Project) \
+                    and all(m in self.plan.child.schema().names
+                            for m in missing):
+                widened = L.Project(
+                    list(self.plan.exprs) + [_col(m) for m in missing],
+                    self.plan.child)
+                sorted_ = L.Sort(widened, ks, desc)
+                final = L.Project([_col(n) for n in cs.names], sorted_)
+                return DataFrame(self.session, final)
+            raise ValueError(
+                f"ORDER BY column(s) {missing} are neither in the select "
+                f"list nor available from the input (have: {cs.names})")
         return DataFrame(self.session, L.Sort(self.plan, ks, desc))
 
     def sample(self, fraction: float, seed: int = 42) -> "DataFrame":

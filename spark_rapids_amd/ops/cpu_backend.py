@@ -501,21 +501,46 @@ def round_half_up(col: Column, scale: int) -> Column:
     return _make(r, av if not av.all() else None, col.dtype)
 
 
+def _like_to_regex(pattern: str) -> str:
+    """SQL LIKE -> python regex with Spark's backslash escape: \\% and \\_
+    match the literal char."""
+    import re
+
+    out = []
+    i = 0
+    while i < len(pattern):
+        c = pattern[i]
+        if c == "\\" and i + 1 < len(pattern):
+            out.append(re.escape(pattern[i + 1]))
+            i += 2
+            continue
+        if c == "%":
+            out.append(".*")
+        elif c == "_":
+            out.append(".")
+        else:
+            out.append(re.escape(c))
+        i += 1
+    return "".join(out)
+
+
 def str_predicate(op: str, col: Column, pattern: str) -> Column:
     a, av = _vals(col), _valid(col)
     if op == "rlike":
         import re
 
-        prog = re.compile(pattern)
+        # re.ASCII: Java regex (Spark's engine) treats \d \w \s as
+        # ASCII-only by default, python re as unicode (ADVICE.md round 1).
+        # Remaining divergences (possessive quantifiers, \p classes) are
+        # routed away by the tagger before this point.
+        prog = re.compile(pattern, re.ASCII)
         res = np.array([bool(prog.search(x)) if x is not None else False
                         for x in a], dtype=np.uint8)
         return _make(res, av if not av.all() else None, DType.bool_())
     if op == "like":
         import re
 
-        rx = re.escape(pattern).replace(r"\%", ".*").replace("%", ".*") \
-            .replace(r"\_", ".").replace("_", ".")
-        prog = re.compile(f"^{rx}$", re.DOTALL)
+        prog = re.compile(f"^{_like_to_regex(pattern)}$", re.DOTALL)
         res = np.array([bool(prog.match(x)) if x is not None else False
                         for x in a], dtype=np.uint8)
     else:

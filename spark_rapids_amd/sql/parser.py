@@ -26,7 +26,7 @@ from ..types import (BOOL, DATE32, DType, FLOAT32, FLOAT64, INT16, INT32,
 _TOKEN_RE = re.compile(r"""
     \s*(?:
       (?P<num>\d+\.\d*(?:[eE][+-]?\d+)?|\.\d+|\d+(?:[eE][+-]?\d+)?)
-    | (?P<str>'(?:[^']|'')*')
+    | (?P<str>'(?:[^'\\]|''|\\.)*')
     | (?P<name>[A-Za-z_][A-Za-z_0-9]*)
     | (?P<op><=|>=|<>|!=|=|<|>|\+|-|\*|/|%|\(|\)|,|\.)
     )""", re.VERBOSE)
@@ -46,6 +46,44 @@ class SqlError(ValueError):
     pass
 
 
+_ESCAPES = {"0": "\0", "'": "'", '"': '"', "b": "\b", "n": "\n", "r": "\r",
+            "t": "\t", "Z": "\x1a", "\\": "\\"}
+
+
+def _unescape_sql_string(s: str) -> str:
+    """Spark ParserUtils.unescapeSQLString: backslash escapes (\\n, \\t,
+    \\uXXXX, ...) are processed; \\% and \\_ KEEP the backslash so LIKE
+    sees them as its own escape; '' collapses to ' (ADVICE.md round 1)."""
+    out = []
+    i = 0
+    while i < len(s):
+        c = s[i]
+        if c == "'" and i + 1 < len(s) and s[i + 1] == "'":
+            out.append("'")
+            i += 2
+            continue
+        if c == "\\" and i + 1 < len(s):
+            nx = s[i + 1]
+            if nx in ("%", "_"):
+                out.append("\\" + nx)
+            elif nx == "u" and i + 5 < len(s):
+                try:
+                    out.append(chr(int(s[i + 2:i + 6], 16)))
+                    i += 6
+                    continue
+                except ValueError:
+                    out.append(nx)
+            elif nx in _ESCAPES:
+                out.append(_ESCAPES[nx])
+            else:
+                out.append(nx)
+            i += 2
+            continue
+        out.append(c)
+        i += 1
+    return "".join(out)
+
+
 def tokenize(text: str) -> List[tuple]:
     out = []
     pos = 0
@@ -59,7 +97,7 @@ def tokenize(text: str) -> List[tuple]:
         if m.group("num") is not None:
             out.append(("num", m.group("num")))
         elif m.group("str") is not None:
-            out.append(("str", m.group("str")[1:-1].replace("''", "'")))
+            out.append(("str", _unescape_sql_string(m.group("str")[1:-1])))
         elif m.group("name") is not None:
             out.append(("name", m.group("name")))
         else:
@@ -433,6 +471,11 @@ class Parser:
             if pat[0] != "str":
                 raise SqlError("LIKE needs a string literal")
             return StringPredicate("like", e, pat[1])
+        if self.kw("RLIKE") or self.kw("REGEXP"):
+            pat = self.next()
+            if pat[0] != "str":
+                raise SqlError("RLIKE needs a string literal")
+            return StringPredicate("rlike", e, pat[1])
         return e
 
     def parse_add(self):

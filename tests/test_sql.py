@@ -199,3 +199,46 @@ def test_sql_null_functions(session):
     out = session.sql("SELECT nvl(a, 0) x, nullif(a, b) y, "
                       "greatest(a, b) g, least(a, b) l FROM tnull").collect()
     assert out == [(1, 1, 9, 1), (0, None, 9, 9), (3, None, 3, 3)]
+
+
+# ---- ADVICE.md (round 1, low) fixes --------------------------------------
+
+def test_string_literal_escapes(s):
+    df = s.create_dataframe({"x": ["a\nb", "anb", "a\\nb"]})
+    s.register("esc_t", df)
+    out = s.sql("SELECT x FROM esc_t WHERE x = 'a\\nb'").to_pydict()
+    assert out["x"] == ["a\nb"]
+
+
+def test_like_escaped_percent(s):
+    df = s.create_dataframe({"x": ["a%c", "abc", "a_c"]})
+    s.register("like_t", df)
+    out = s.sql(r"SELECT x FROM like_t WHERE x LIKE 'a\%c'").to_pydict()
+    assert out["x"] == ["a%c"]
+    out = s.sql(r"SELECT x FROM like_t WHERE x LIKE 'a\_c'").to_pydict()
+    assert out["x"] == ["a_c"]
+    out = s.sql("SELECT x FROM like_t WHERE x LIKE 'a%c'").to_pydict()
+    assert out["x"] == ["a%c", "abc", "a_c"]
+
+
+def test_order_by_dropped_column(s):
+    df = s.create_dataframe({"g": [2, 1, 1], "v": [1.0, 2.0, 3.0]})
+    s.register("obt", df)
+    out = s.sql("SELECT rank() OVER (PARTITION BY g ORDER BY v) AS r "
+                  "FROM obt ORDER BY g").to_pydict()
+    assert out == {"r": [1, 2, 1]}
+
+
+def test_order_by_truly_unknown_column_raises(s):
+    from spark_rapids_amd import col
+    df = s.create_dataframe({"a": [1, 2]})
+    with pytest.raises(ValueError, match="ORDER BY"):
+        df.select(col("a").alias("b")).sort("zz")
+
+
+def test_rlike_ascii_digit_class(s):
+    # Java regex \d is ASCII-only; Arabic-Indic digit must NOT match
+    df = s.create_dataframe({"x": ["7", "٣", "x"]})
+    s.register("rl_t", df)
+    out = s.sql(r"SELECT x FROM rl_t WHERE x RLIKE '^\\d$'").to_pydict()
+    assert out["x"] == ["7"]
