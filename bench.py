@@ -1,15 +1,24 @@
 #!/usr/bin/env python3
-"""Flagship benchmark: NDS-like power run on N MI355X GPUs (weak scaling).
+"""Flagship benchmark: NDS (TPC-DS-derived) power run on N MI355X GPUs.
 
 Driver contract: `python bench.py --gpus N --steps K --warmup W` (launched
-under torch.distributed.run for N>1, one rank per GPU over RCCL). Each rank
-owns a fixed-size synthetic partition of an NDS-style star schema
-(BASELINE.json metric: NDS power-run wall-clock + speedup vs CPU Spark; the
-CPU baseline here is this engine's own CPU backend on identical data — the
-reference's in-repo baseline number is its 3.0x default operator speedup,
-tools/generated_files/operatorsScore.csv).
+under torch.distributed.run for N>1, one rank per GPU over RCCL).
 
-One step = the full power-run query suite executed on-GPU via hipdf kernels.
+What one step is: the full 11-query NDS-shaped power run (bench/nds_queries
+.py — string-keyed dimension joins, decimal aggregation, rollup, windows,
+top-N sorts) executed end to end, every query SCANNING ITS INPUT FROM
+PARQUET ON DISK through the engine's GPU decode reader. Data is a
+deterministic TPC-DS-shaped star schema (bench/nds.py) staged to local disk
+before the timed region; the host file cache stays at its default (off) so
+every step re-reads and re-decodes.
+
+BASELINE.json metric: NDS power-run wall-clock + speedup vs CPU Spark. The
+CPU baseline here is this engine's own CPU backend on multiple worker
+processes over the same on-disk data (there is no JVM Spark in the image);
+vs_baseline divides that measured speedup by the reference's default 3.0x
+operator score — the reference repo publishes no absolute
+wall-clock numbers in-tree.
+
 Timing: barrier + torch.cuda.synchronize on both sides, MAX over ranks.
 """
 from __future__ import annotations
@@ -25,10 +34,8 @@ sys.path.insert(0, os.path.dirname(os.path.abspath(__file__)))
 import torch
 
 from spark_rapids_amd import Session
-from spark_rapids_amd.api import MemTable
-from spark_rapids_amd.bench import datagen
-from spark_rapids_amd.bench.queries import POWER_RUN, run_power
-from spark_rapids_amd.plan import logical as L
+from spark_rapids_amd.bench import nds
+from spark_rapids_amd.bench.nds_queries import POWER_RUN, run_power
 
 
 def _dist_env():
@@ -38,48 +45,44 @@ def _dist_env():
     return rank, world, local_rank
 
 
-def _make_tables(session, rows: int, seed: int, device: str,
-                 partitions: int):
-    per = rows // partitions
-    fact_batches = [datagen.gen_fact_partition(per, seed * 1000 + p)
-                    for p in range(partitions)]
-    if device == "cuda":
-        fact_batches = [b.cuda() for b in fact_batches]
-    items = datagen.gen_items()
-    stores = datagen.gen_stores()
-    if device == "cuda":
-        items, stores = items.cuda(), stores.cuda()
-    return {
-        "store_sales": session.from_batches(fact_batches,
-                                            datagen.fact_schema(), "store_sales"),
-        # dimension tables are generated identically on every rank
-        "item": session.from_batches([items], datagen.item_schema(), "item",
-                                     replicated=True),
-        "store": session.from_batches([stores], datagen.store_schema(),
-                                      "store", replicated=True),
-    }
+def _default_data_dir(rows: int, world: int) -> str:
+    base = os.environ.get("TMPDIR", "/tmp")
+    return os.path.join(base, f"nds_data_{rows}x{world}")
 
 
-def _cpu_worker(rows: int, seed: int, steps: int) -> float:
+def _open_tables(session: Session, paths):
+    tables = {"store_sales": session.read_parquet(paths["store_sales"])}
+    for dim in ("date_dim", "item", "store", "customer"):
+        tables[dim] = session.read_parquet(paths[dim], replicated=True)
+    return tables
+
+
+def _cpu_worker(data_dir: str, my_fact_files, steps: int) -> float:
     session = Session({"spark.rapids.sql.enabled": False})
-    tables = _make_tables(session, rows, seed=seed, device="cpu",
-                          partitions=1)
+    paths = {name: os.path.join(data_dir, name)
+             for name in ("date_dim", "item", "store", "customer")}
+    tables = {"store_sales": session.read_parquet(list(my_fact_files))}
+    for dim, p in paths.items():
+        tables[dim] = session.read_parquet(p, replicated=True)
     t0 = time.perf_counter()
     for _ in range(steps):
         run_power(tables)
     return time.perf_counter() - t0
 
 
-def _cpu_baseline(rows_total: int, procs: int, steps: int) -> float:
-    """Run the power suite on `procs` CPU worker processes, each owning
-    rows_total/procs rows (the multi-core CPU Spark analogue). Returns
-    wall-clock seconds per step (max over workers)."""
+def _cpu_baseline(data_dir: str, procs: int, steps: int) -> float:
+    """Power run on `procs` CPU worker processes, each scanning a disjoint
+    shard of the on-disk fact files (multi-core CPU Spark analogue).
+    Returns wall-clock seconds per step (max over workers)."""
     import concurrent.futures as cf
+    import glob as g
 
-    per = max(rows_total // procs, 1)
+    fact_files = sorted(
+        g.glob(os.path.join(data_dir, "store_sales", "*.parquet")))
     t0 = time.perf_counter()
     with cf.ProcessPoolExecutor(procs) as pool:
-        futs = [pool.submit(_cpu_worker, per, 1000 + i, steps)
+        futs = [pool.submit(_cpu_worker, data_dir,
+                            fact_files[i::procs] or fact_files[:1], steps)
                 for i in range(procs)]
         [f.result() for f in futs]
     return (time.perf_counter() - t0) / steps
@@ -88,17 +91,18 @@ def _cpu_baseline(rows_total: int, procs: int, steps: int) -> float:
 def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
-    ap.add_argument("--steps", type=int, default=5)
-    ap.add_argument("--warmup", type=int, default=2)
+    ap.add_argument("--steps", type=int, default=3)
+    ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--rows-per-gpu", type=int, default=20_000_000)
-    ap.add_argument("--partitions", type=int, default=8)
+    ap.add_argument("--partitions", type=int, default=8,
+                    help="fact parquet files per rank")
+    ap.add_argument("--data-dir", default=None)
+    ap.add_argument("--queries", default=None,
+                    help="comma list to restrict the suite (debug)")
     ap.add_argument("--cpu-baseline-steps", type=int, default=1,
                     help="0 disables the CPU-backend baseline measurement")
     ap.add_argument("--cpu-baseline-procs", type=int,
-                    default=min(16, os.cpu_count() or 1),
-                    help="worker processes for the CPU baseline (models the "
-                    "multi-core CPU Spark executor the reference compares "
-                    "against)")
+                    default=min(16, os.cpu_count() or 1))
     args = ap.parse_args()
 
     rank, world, local_rank = _dist_env()
@@ -112,10 +116,18 @@ def main():
         torch.distributed.init_process_group(
             backend="nccl" if use_gpu else "gloo")
 
-    device = "cuda" if use_gpu else "cpu"
+    data_dir = args.data_dir or _default_data_dir(args.rows_per_gpu, world)
+    os.makedirs(data_dir, exist_ok=True)
+    stage_t0 = time.perf_counter()
+    paths = nds.stage(data_dir, args.rows_per_gpu, rank, world,
+                      partitions=args.partitions)
+    if distributed:
+        torch.distributed.barrier()
+    stage_s = time.perf_counter() - stage_t0
+
     session = Session({"spark.rapids.sql.enabled": use_gpu})
-    tables = _make_tables(session, args.rows_per_gpu, seed=rank + 1,
-                          device=device, partitions=args.partitions)
+    tables = _open_tables(session, paths)
+    queries = args.queries.split(",") if args.queries else None
 
     def barrier_sync():
         if distributed:
@@ -125,39 +137,39 @@ def main():
 
     # ---- warmup ----
     for _ in range(args.warmup):
-        run_power(tables)
+        run_power(tables, queries)
     barrier_sync()
 
     # ---- timed region ----
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        run_power(tables)
+        run_power(tables, queries)
     barrier_sync()
     elapsed = time.perf_counter() - t0
 
     # MAX over ranks
     if distributed:
         t = torch.tensor([elapsed], dtype=torch.float64,
-                         device=device if use_gpu else "cpu")
+                         device="cuda" if use_gpu else "cpu")
         torch.distributed.all_reduce(t, op=torch.distributed.ReduceOp.MAX)
         elapsed = float(t.item())
 
+    n_queries = len(queries) if queries else len(POWER_RUN)
     ms_per_step = elapsed / args.steps * 1000.0
     # whole-job aggregate: fact rows scanned per second across all queries
-    rows_per_step = args.rows_per_gpu * len(POWER_RUN) * n_gpus
+    rows_per_step = args.rows_per_gpu * n_queries * n_gpus
     value = rows_per_step * args.steps / elapsed
 
-    # ---- CPU baseline (rank 0, once, same per-GPU data size, multi-process
-    # to model the multi-core CPU Spark executor) ----
     speedup = None
     if rank == 0 and args.cpu_baseline_steps > 0:
-        tcpu = _cpu_baseline(args.rows_per_gpu, args.cpu_baseline_procs,
+        tcpu = _cpu_baseline(data_dir, args.cpu_baseline_procs,
                              args.cpu_baseline_steps)
         speedup = tcpu / (elapsed / args.steps)
 
     if rank == 0:
         result = {
-            "metric": "NDS-like power-run fact rows/s (speedup vs CPU in config)",
+            "metric": "NDS power-run fact rows/s from on-disk parquet "
+                      "(speedup vs CPU backend in config)",
             "value": value,
             "unit": "rows/s",
             "n_gpus": n_gpus,
@@ -167,21 +179,30 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": (speedup / 3.0) if speedup is not None else None,
-            "dtype": "decimal(7,2)+fp64",
-            "data": "synthetic",
+            "dtype": "decimal(7,2)+fp64+string-keys",
+            "data": "synthetic TPC-DS-shaped star schema staged to parquet "
+                    "on local disk; every step re-scans from disk",
             "config": {
-                "model": "nds_like_power_run_5q",
+                "model": "nds_power_run_11q",
                 "global_batch": args.rows_per_gpu * n_gpus,
                 "seq_len": None,
                 "parallelism": f"dp{n_gpus}",
                 "rows_per_gpu": args.rows_per_gpu,
-                "queries": [q for q, _ in POWER_RUN],
+                "sf_equivalent_per_gpu":
+                    round(nds.sf_equivalent(args.rows_per_gpu), 1),
+                "queries": [q for q, _ in POWER_RUN] if not queries
+                           else queries,
+                "dims": "item(102k, string brand/category/id) "
+                        "store(1k, string state/name) "
+                        "customer(1M, distinct string ids) date_dim(1826)",
+                "on_disk_input": data_dir,
+                "staging_s_untimed": round(stage_s, 1),
                 "speedup_vs_cpu_backend": speedup,
                 "cpu_baseline_procs": args.cpu_baseline_procs,
-                "baseline_definition": "vs_baseline = speedup / 3.0: speedup "
-                "= wall-clock of this engine's CPU backend on "
-                f"{args.cpu_baseline_procs} worker processes (multi-core CPU "
-                "Spark analogue, same data) / GPU wall-clock; 3.0x is the "
+                "baseline_definition": "vs_baseline = speedup / 3.0: "
+                "speedup = wall-clock of this engine's CPU backend on "
+                f"{args.cpu_baseline_procs} worker processes scanning the "
+                "same on-disk parquet / GPU wall-clock; 3.0x is the "
                 "reference's default operator speedup score (BASELINE.md)",
             },
         }

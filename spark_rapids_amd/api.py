@@ -435,7 +435,11 @@ class Session:
 
         return parse_sql(self, query)
 
-    def read_parquet(self, path: str, columns=None) -> DataFrame:
+    def read_parquet(self, path: str, columns=None,
+                     replicated: bool = False) -> DataFrame:
+        """replicated=True marks the table as fully present on every rank
+        (dimension tables): scans read all files instead of sharding
+        files[rank::world], and joins against it skip the exchange."""
         import torch
 
         from .config import PARQUET_MT_THREADS, PARQUET_READER_TYPE
@@ -446,7 +450,8 @@ class Session:
             reader = "GPU_DECODE" if (torch.cuda.is_available()
                                       and self.conf.sql_enabled) else "CPU"
         src = ParquetTable(path, columns=columns, reader=reader,
-                           prefetch_threads=self.conf.get(PARQUET_MT_THREADS))
+                           prefetch_threads=self.conf.get(PARQUET_MT_THREADS),
+                           replicated=replicated)
         return DataFrame(self, L.Scan(src, src.schema, f"parquet:{path}"))
 
     def write_parquet(self, df: DataFrame, path: str,

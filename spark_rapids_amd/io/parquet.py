@@ -137,10 +137,19 @@ class ParquetTable:
 
     def __init__(self, path: str, num_partitions: int = 0,
                  columns: Optional[List[str]] = None, reader: str = "CPU",
-                 prefetch_threads: int = 4):
-        self.files = sorted(glob.glob(path)) if any(ch in path for ch in "*?") \
-            else ([os.path.join(path, f) for f in sorted(os.listdir(path))
-                   if f.endswith(".parquet")] if os.path.isdir(path) else [path])
+                 prefetch_threads: int = 4, replicated: bool = False):
+        # replicated=True: every rank scans ALL files (dimension tables in
+        # a distributed star-schema query; broadcast-join build sides skip
+        # the exchange, plan/logical.py is_replicated)
+        self.replicated = replicated
+        if isinstance(path, (list, tuple)):
+            self.files = list(path)
+        else:
+            self.files = sorted(glob.glob(path)) \
+                if any(ch in path for ch in "*?") \
+                else ([os.path.join(path, f) for f in sorted(os.listdir(path))
+                       if f.endswith(".parquet")]
+                      if os.path.isdir(path) else [path])
         if not self.files:
             raise FileNotFoundError(path)
         self.columns = columns
@@ -182,7 +191,7 @@ class ParquetTable:
         from ..shuffle import dist
 
         c = dist.ctx()
-        if c.is_multi and len(self.files) > 1:
+        if c.is_multi and len(self.files) > 1 and not self.replicated:
             return self.files[c.rank::c.world]
         return self.files
 
