@@ -197,6 +197,9 @@ def main():
                         "customer(1M, distinct string ids) date_dim(1826)",
                 "on_disk_input": data_dir,
                 "staging_s_untimed": round(stage_s, 1),
+                "scan_stats": __import__(
+                    "spark_rapids_amd.io.parquet", fromlist=["SCAN_STATS"]
+                ).SCAN_STATS,
                 "speedup_vs_cpu_backend": speedup,
                 "cpu_baseline_procs": args.cpu_baseline_procs,
                 "baseline_definition": "vs_baseline = speedup / 3.0: "

@@ -7,10 +7,11 @@ Reference analogue: GpuParquetScan + libcudf's parquet decode kernels
 (CpuCompressionConfig, GpuParquetScan.scala:1549) is the model for the
 decompress-on-host/decode-on-device split used here.
 
-Scope this round: flat schemas; INT32/INT64/FLOAT/DOUBLE (+date32/
-timestamp/decimal on those physical types) PLAIN or dictionary; BYTE_ARRAY
-strings when dictionary-encoded. Anything else raises NotImplementedError
-and the caller falls back to the CPU (hybrid) reader per file.
+Scope: flat schemas; INT32/INT64/FLOAT/DOUBLE/BOOLEAN (+date32/timestamp/
+decimal on those physical types, including width-widening INT32-physical
+decimals) PLAIN, dictionary or DELTA_BINARY_PACKED; BYTE_ARRAY strings
+PLAIN or dictionary-encoded. Anything else raises NotImplementedError and
+the caller falls back to the CPU (hybrid) reader per file.
 """
 from __future__ import annotations
 
@@ -329,11 +330,23 @@ class _ChunkDecoder:
             return self._gather_dict(ridx, n, mask)
         raise NotImplementedError(f"encoding {encoding}")
 
+    _RAW_HT = {torch.uint8: 0, torch.int8: 1, torch.int16: 2,
+               torch.int32: 3, torch.int64: 4, torch.float32: 5,
+               torch.float64: 6}
+
     def _cast_raw(self, dense: torch.Tensor, tdt) -> torch.Tensor:
         # same-width reinterpret (e.g. int32 -> date32 backing)
         if dense.element_size() == torch.tensor([], dtype=tdt).element_size():
             return dense.view(tdt)
-        raise NotImplementedError("width-changing parquet cast")
+        # width-changing RAW widen (e.g. INT32-physical decimal into its
+        # int64 backing: parquet stores the unscaled value, so this must
+        # NOT rescale like a value-level decimal cast would)
+        n = dense.numel()
+        out = torch.empty(max(n, 1), dtype=tdt, device="cuda")[:n]
+        if n:
+            self.ext.cast(self._RAW_HT[dense.dtype], self._RAW_HT[tdt],
+                          dense.data_ptr(), out.data_ptr(), n, self.s)
+        return out
 
     def _gather_dict(self, ridx: torch.Tensor, n: int, mask) -> Column:
         from ..ops import gpu_backend as gb

@@ -250,3 +250,28 @@ def test_filecache_roundtrip(tmp_path):
     assert len(fc._cache) == 2
     sr.Session({"spark.rapids.sql.enabled": False})  # off -> cleared
     assert len(fc._cache) == 0
+
+
+@pytest.mark.gpu
+def test_gpu_decode_nds_staging_no_fallback(tmp_path):
+    """The NDS bench staging (INT32-physical decimals, dictionary strings,
+    PLAIN facts) must decode fully on the device — zero CPU fallbacks —
+    and match the CPU reader byte for byte."""
+    import spark_rapids_amd as sr
+    from spark_rapids_amd.bench import nds
+    from spark_rapids_amd.io import parquet as iop
+
+    d = str(tmp_path / "nds")
+    paths = nds.stage(d, rows=50_000, rank=0, world=1, partitions=2)
+
+    sg = sr.Session()  # GPU_DECODE
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    before = dict(iop.SCAN_STATS)
+    for name in ("store_sales", "date_dim", "item", "store", "customer"):
+        g = sg.read_parquet(paths[name]).collect()
+        c = sc.read_parquet(paths[name]).collect()
+        assert g == c, f"{name}: first diff " + str(
+            next((a, b) for a, b in zip(g, c) if a != b))
+    assert iop.SCAN_STATS["fallback_files"] == before["fallback_files"], \
+        iop.SCAN_STATS["last_fallback"]
+    assert iop.SCAN_STATS["gpu_files"] > before["gpu_files"]
