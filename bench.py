@@ -99,6 +99,8 @@ def main():
     ap.add_argument("--data-dir", default=None)
     ap.add_argument("--queries", default=None,
                     help="comma list to restrict the suite (debug)")
+    ap.add_argument("--per-query", action="store_true",
+                    help="print per-query wall times to stderr (untimed)")
     ap.add_argument("--cpu-baseline-steps", type=int, default=1,
                     help="0 disables the CPU-backend baseline measurement")
     ap.add_argument("--cpu-baseline-procs", type=int,
@@ -139,6 +141,19 @@ def main():
     for _ in range(args.warmup):
         run_power(tables, queries)
     barrier_sync()
+
+    if args.per_query and rank == 0:
+        for qname in (queries or [q for q, _ in POWER_RUN]):
+            if use_gpu:
+                torch.cuda.synchronize()
+            tq = time.perf_counter()
+            run_power(tables, [qname])
+            if use_gpu:
+                torch.cuda.synchronize()
+            print(f"[per-query] {qname}: "
+                  f"{(time.perf_counter() - tq) * 1000:.1f} ms",
+                  file=sys.stderr)
+        barrier_sync()
 
     # ---- timed region ----
     t0 = time.perf_counter()

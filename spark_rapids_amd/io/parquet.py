@@ -165,6 +165,18 @@ class ParquetTable:
             keep = [f for f in self.schema.fields if f.name in columns]
             self.schema = Schema(keep)
 
+    def with_columns(self, names: List[str]) -> "ParquetTable":
+        """Narrowed view for column pruning: only these columns are read
+        and decoded (parquet stores column chunks separately on disk)."""
+        t = ParquetTable.__new__(ParquetTable)
+        t.replicated = self.replicated
+        t.files = self.files
+        t.columns = list(names)
+        t.reader = self.reader
+        t.prefetch_threads = self.prefetch_threads
+        t.schema = Schema([f for f in self.schema.fields if f.name in names])
+        return t
+
     def _read_one(self, path: str) -> ColumnBatch:
         from .filecache import get_cached, put_cached
 

@@ -142,6 +142,14 @@ def prune_columns(plan: L.LogicalPlan,
     """Return an equivalent plan where children materialize only the columns
     the ancestors reference. needed=None means every output column."""
     if isinstance(plan, L.Scan):
+        # push the needed-column set into sources that can skip IO+decode
+        # for unused columns (parquet/orc: per-column chunks on disk)
+        if needed is not None and hasattr(plan.source, "with_columns"):
+            fields = plan.schema().fields
+            keep = [f.name for f in fields if f.name in needed]
+            if keep and len(keep) < len(fields):
+                src = plan.source.with_columns(keep)
+                return L.Scan(src, src.schema, plan.label)
         return _project_to(plan, needed)
     if isinstance(plan, L.Project):
         cs = plan.child.schema()

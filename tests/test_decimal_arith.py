@@ -265,7 +265,7 @@ def test_gpu_add_large_scale_mismatch_exact():
                     (col("a") > col("b")).alias("g")).to_pydict()
     assert out["s"][0] == Decimal("100000000000000001.0000000000")
     assert out["s"][1] is None
-    assert out["s"][2] == Decimal("-6.9999999977")
+    assert out["s"][2] == Decimal("16.0000000000")
     assert out["g"][0] is True
     assert out["g"][1] is None
     assert out["g"][2] is False
