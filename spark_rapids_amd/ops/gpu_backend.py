@@ -815,32 +815,37 @@ def _cast_to_string(col: Column, v) -> Column:
 def _cast_decimal(col: Column, to: DType, v) -> Column:
     n = col.size
     s = _stream()
-    if col.dtype.id is TypeId.DECIMAL128 or to.id is TypeId.DECIMAL128:
-        if col.dtype.is_decimal and to.is_decimal:
-            # Widen to int128 FIRST, then rescale exactly in 128-bit with
-            # HALF_UP down-shift and null-on-overflow (Spark non-ANSI).
-            # Rescaling inside int64 before widening silently overflows
-            # (decimal(18,0) + decimal(18,10)).
-            if col.dtype.id is TypeId.DECIMAL64:
-                wide = torch.empty(max(2 * n, 1), dtype=torch.int64,
-                                   device="cuda")[:2 * n]
-                if n:
-                    ext.i64_to_i128(col.data.data_ptr(), wide.data_ptr(),
-                                    n, s)
-                src128 = wide
-            else:
-                src128 = col.data
-            shift = to.scale - col.dtype.scale
-            out_is_64 = to.id is TypeId.DECIMAL64
-            width = n if out_is_64 else 2 * n
-            out = torch.empty(max(width, 1), dtype=torch.int64,
-                              device="cuda")[:width]
-            ov = _alloc_mask(n)
+    if col.dtype.is_decimal and to.is_decimal:
+        if to.scale == col.dtype.scale and to.id is col.dtype.id \
+                and to.precision >= col.dtype.precision:
+            # pure widening retype: no rescale, no overflow possible
+            return Column(to, n, col.data.clone(), v,
+                          null_count=col._null_count)
+        # Widen to int128 FIRST, then rescale exactly in 128-bit with
+        # HALF_UP down-shift and null-on-overflow (Spark non-ANSI).
+        # Rescaling inside int64 before widening silently overflows
+        # (decimal(18,0) + decimal(18,10)); the same
+        # kernel checks the target precision for narrowing user casts.
+        if col.dtype.id is TypeId.DECIMAL64:
+            wide = torch.empty(max(2 * n, 1), dtype=torch.int64,
+                               device="cuda")[:2 * n]
             if n:
-                ext.i128_rescale(src128.data_ptr(), _ptr(v), out.data_ptr(),
-                                 ov.data_ptr(), shift, to.precision,
-                                 1 if out_is_64 else 0, n, s)
-            return Column(to, n, out, ov, null_count=None)
+                ext.i64_to_i128(col.data.data_ptr(), wide.data_ptr(), n, s)
+            src128 = wide
+        else:
+            src128 = col.data
+        shift = to.scale - col.dtype.scale
+        out_is_64 = to.id is TypeId.DECIMAL64
+        width = n if out_is_64 else 2 * n
+        out = torch.empty(max(width, 1), dtype=torch.int64,
+                          device="cuda")[:width]
+        ov = _alloc_mask(n)
+        if n:
+            ext.i128_rescale(src128.data_ptr(), _ptr(v), out.data_ptr(),
+                             ov.data_ptr(), shift, to.precision,
+                             1 if out_is_64 else 0, n, s)
+        return Column(to, n, out, ov, null_count=None)
+    if col.dtype.id is TypeId.DECIMAL128 or to.id is TypeId.DECIMAL128:
         if col.dtype.id is TypeId.DECIMAL128 and to.is_floating:
             dbl = torch.empty(max(n, 1), dtype=torch.float64,
                               device="cuda")[:n]
@@ -861,15 +866,6 @@ def _cast_decimal(col: Column, to: DType, v) -> Column:
                             null_count=col._null_count)
             return _cast_decimal(as_dec, to, v)
         raise NotImplementedError(f"gpu cast {col.dtype} -> {to}")
-    if col.dtype.is_decimal and to.is_decimal:
-        shift = to.scale - col.dtype.scale
-        out = _alloc(n, to)
-        if shift == 0:
-            out.copy_(col.data)
-        else:
-            ext.decimal_rescale(col.data.data_ptr(), out.data_ptr(),
-                                10 ** abs(shift), shift > 0, n, s)
-        return Column(to, n, out, v, null_count=col._null_count)
     if col.dtype.is_decimal:
         # decimal -> float/int: via double divide
         dbl = _alloc(n, DType.float64())
