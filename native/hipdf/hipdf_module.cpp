@@ -119,6 +119,8 @@ void hipdf_change_flags(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_iota_i32(void*, int64_t, hipStream_t);
 void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
+void hipdf_rle_hybrid_batch(const void*, const void*, int, void*,
+                            hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
                              hipStream_t);
 void hipdf_pq_delta_i64(const void*, int64_t, int64_t, void*, hipStream_t);
@@ -533,6 +535,10 @@ PYBIND11_MODULE(hipdf, m) {
     hipdf_str_plain_offsets(P(data), nbytes, n_values, PM(starts), PM(lens),
                             PM(error), S(stream));
     check_async();
+  });
+  m.def("rle_hybrid_batch", [](int64_t base, int64_t descs, int nstreams,
+                               int64_t out, int64_t stream) {
+    hipdf_rle_hybrid_batch(P(base), P(descs), nstreams, PM(out), S(stream));
   });
   m.def("rle_hybrid_decode", [](int64_t data, int64_t nbytes, int bw,
                                 int64_t out, int64_t n, int64_t stream) {

@@ -10,11 +10,12 @@
 #include "hipdf_common.h"
 
 // decode an RLE/bit-packed hybrid stream of n_values values of bit_width
-// bits into out[0..n_values)
-__global__ void k_rle_hybrid_decode(const uint8_t* __restrict__ data,
-                                    int64_t nbytes, int bit_width,
-                                    int32_t* __restrict__ out,
-                                    int64_t n_values) {
+// bits into out[0..n_values); one workgroup per stream (function-scope
+// __shared__ is per block, so the batched kernel reuses this body)
+__device__ void rle_decode_one(const uint8_t* __restrict__ data,
+                               int64_t nbytes, int bit_width,
+                               int32_t* __restrict__ out,
+                               int64_t n_values) {
   __shared__ int64_t s_pos;       // byte position in stream
   __shared__ int64_t s_out;       // values emitted
   __shared__ int64_t s_run_len;   // current run length (values)
@@ -90,6 +91,25 @@ __global__ void k_rle_hybrid_decode(const uint8_t* __restrict__ data,
       if (threadIdx.x == 0) s_out = base + emit;
     }
   }
+}
+
+__global__ void k_rle_hybrid_decode(const uint8_t* __restrict__ data,
+                                    int64_t nbytes, int bit_width,
+                                    int32_t* __restrict__ out,
+                                    int64_t n_values) {
+  rle_decode_one(data, nbytes, bit_width, out, n_values);
+}
+
+// batched variant: one workgroup per RLE stream (per parquet page), so a
+// whole column chunk's pages decode in one launch instead of serial
+// per-page launches. descs: 5 int64 per stream
+// [src_off, nbytes, out_off, n_values, bit_width] into a single base
+// buffer (the decompressed chunk).
+__global__ void k_rle_hybrid_batch(const uint8_t* __restrict__ base,
+                                   const int64_t* __restrict__ descs,
+                                   int32_t* __restrict__ out) {
+  const int64_t* d = descs + 5 * (int64_t)blockIdx.x;
+  rle_decode_one(base + d[0], d[1], (int)d[4], out + d[2], d[3]);
 }
 
 // out[idx[j]] = vals[j]
@@ -440,6 +460,14 @@ void hipdf_rle_hybrid_decode(const void* data, int64_t nbytes, int bit_width,
   hipLaunchKernelGGL(k_rle_hybrid_decode, dim3(1), dim3(HIPDF_BLOCK), 0,
                      stream, (const uint8_t*)data, nbytes, bit_width,
                      (int32_t*)out, n_values);
+}
+
+void hipdf_rle_hybrid_batch(const void* base, const void* descs,
+                            int nstreams, void* out, hipStream_t stream) {
+  if (nstreams <= 0) return;
+  hipLaunchKernelGGL(k_rle_hybrid_batch, dim3((uint32_t)nstreams),
+                     dim3(HIPDF_BLOCK), 0, stream, (const uint8_t*)base,
+                     (const int64_t*)descs, (int32_t*)out);
 }
 
 void hipdf_scatter_fixed(int esize, const void* vals, const void* idx,

@@ -53,6 +53,48 @@ def _decompress(codec, payload: bytes, usize: int) -> bytes:
     return out.to_pybytes() if hasattr(out, "to_pybytes") else bytes(out)
 
 
+class _Page:
+    __slots__ = ("n", "levels", "values", "encoding", "all_valid")
+
+    def __init__(self, n, levels, values, encoding, all_valid):
+        self.n = n
+        self.levels = levels
+        self.values = values
+        self.encoding = encoding
+        self.all_valid = all_valid
+
+
+def _rle_all_valid(levels: bytes, n: int, max_def: int) -> bool:
+    """Host-side check that a def-level RLE stream encodes `n` copies of
+    max_def (no nulls). All-valid pages are written as a single short RLE
+    run, so this reads a handful of bytes and lets the decoder skip the
+    level decode + validity-mask pipeline entirely."""
+    pos = 0
+    seen = 0
+    nb = len(levels)
+    byte_per_val = 1  # def levels for flat schemas: bit width 1..7 -> 1B
+    while seen < n:
+        if pos >= nb:
+            return False
+        h = 0
+        shift = 0
+        while pos < nb:
+            b = levels[pos]
+            pos += 1
+            h |= (b & 0x7F) << shift
+            if not (b & 0x80):
+                break
+            shift += 7
+        if h & 1:
+            return False  # bit-packed run: has a mix -> not trivially valid
+        count = h >> 1
+        if pos >= nb or levels[pos] != max_def:
+            return False
+        pos += byte_per_val
+        seen += count
+    return True
+
+
 class _ChunkDecoder:
     """Decodes one column chunk (one column of one row group) on the GPU."""
 
@@ -70,9 +112,39 @@ class _ChunkDecoder:
         self.dict_str: Optional[tuple] = None  # (offsets cuda, bytes cuda)
 
     def decode(self) -> Column:
+        """Chunk-granular decode: page headers are parsed host-side first;
+        the hot paths then run ONE upload + ONE kernel sequence for the
+        whole column chunk instead of a launch storm per page (the round-1
+        profile showed per-page rle_hybrid_decode + a 6-kernel validity
+        chain per page as 60%+ of NDS scan kernel time)."""
+        pages = self._parse_pages()
+        if not pages:
+            return Column.nulls(self.dtype, 0, "cuda")
+        fast = self._decode_chunk(pages)
+        if fast is not None:
+            return fast
+        page_cols = []
+        for pg in pages:
+            levels = None
+            if not pg.all_valid and self.max_def > 0:
+                levels = self._decode_levels(pg.levels, pg.n)
+            page_cols.append(self._materialize(pg.values, pg.n, levels,
+                                               pg.encoding))
+        if len(page_cols) == 1:
+            return page_cols[0]
+        from ..ops import gpu_backend
+
+        return gpu_backend.concat_batches(
+            [ColumnBatch([c]) for c in page_cols]).columns[0]
+
+    def _parse_pages(self):
+        """Parse headers + decompress payloads for every page of the chunk;
+        load the dictionary; mark pages whose def-levels are trivially
+        all-valid (a single RLE run of max_def — the overwhelmingly common
+        case, checked host-side in a few bytes)."""
         pos = 0
-        page_cols: List[Column] = []
         decoded = 0
+        pages = []
         while decoded < self.num_values and pos < len(self.raw):
             ph = tc.parse_page_header(self.raw, pos)
             pos += ph.header_size
@@ -85,23 +157,120 @@ class _ChunkDecoder:
             elif ph.type == DATA_PAGE:
                 data = _decompress(self.codec, payload,
                                    ph.uncompressed_page_size)
-                col = self._decode_data_page(data, ph.data_page)
-                page_cols.append(col)
-                decoded += col.size
+                hdr = ph.data_page
+                n = hdr.num_values
+                levels = b""
+                vpos = 0
+                if self.max_def > 0:
+                    (lvl_len,) = pystruct.unpack_from("<I", data, 0)
+                    levels = data[4:4 + lvl_len]
+                    vpos = 4 + lvl_len
+                pages.append(_Page(n, levels, data[vpos:], hdr.encoding,
+                                   self.max_def == 0
+                                   or _rle_all_valid(levels, n,
+                                                     self.max_def)))
+                decoded += n
             elif ph.type == DATA_PAGE_V2:
-                col = self._decode_data_page_v2(payload, ph)
-                page_cols.append(col)
-                decoded += col.size
+                hdr = ph.data_page_v2
+                n = hdr.num_values
+                lvl_len = (hdr.def_levels_byte_length
+                           + hdr.rep_levels_byte_length)
+                levels = payload[hdr.rep_levels_byte_length:lvl_len]
+                vals = payload[lvl_len:]
+                if hdr.is_compressed and self.codec is not None:
+                    vals = _decompress(self.codec, vals,
+                                       ph.uncompressed_page_size - lvl_len)
+                all_valid = self.max_def == 0 or hdr.num_nulls == 0
+                # v2 levels carry no 4-byte length prefix and are a pure
+                # RLE stream like v1's
+                pages.append(_Page(n, levels, vals, hdr.encoding, all_valid))
+                decoded += n
             else:
                 raise NotImplementedError(f"page type {ph.type}")
-        if not page_cols:
-            return Column.nulls(self.dtype, 0, "cuda")
-        if len(page_cols) == 1:
-            return page_cols[0]
-        from ..ops import gpu_backend
+        return pages
 
-        return gpu_backend.concat_batches(
-            [ColumnBatch([c]) for c in page_cols]).columns[0]
+    # -- chunk-granular fast paths --------------------------------------
+
+    def _decode_chunk(self, pages) -> Optional[Column]:
+        encs = {p.encoding for p in pages}
+        total = sum(p.n for p in pages)
+        all_valid = all(p.all_valid for p in pages)
+        if encs == {PLAIN} and self.phys in _PHYS_NP:
+            return self._chunk_plain_fixed(pages, total, all_valid)
+        if encs <= {PLAIN_DICTIONARY, RLE_DICTIONARY} and all_valid:
+            return self._chunk_dict(pages, total)
+        if encs == {PLAIN} and self.phys == "BYTE_ARRAY" and all_valid:
+            # concat value sections: one plain byte-array decode for the
+            # whole chunk
+            blob = b"".join(p.values for p in pages)
+            return self._materialize(blob, total, None, PLAIN)
+        return None
+
+    def _chunk_validity(self, pages, total):
+        """(mask, valid_idx, n_valid) for the whole chunk: batch-decode all
+        pages' def-level RLE streams in ONE launch, then run the validity
+        chain once over the contiguous levels array."""
+        levels_blob = b"".join(p.levels for p in pages)
+        base = torch.from_numpy(np.frombuffer(
+            levels_blob, dtype=np.uint8).copy()).cuda()
+        descs = np.empty((len(pages), 5), dtype=np.int64)
+        src = out = 0
+        for i, p in enumerate(pages):
+            descs[i] = (src, len(p.levels), out, p.n, 1)
+            src += len(p.levels)
+            out += p.n
+        dt = torch.from_numpy(descs).cuda()
+        levels = torch.empty(total, dtype=torch.int32, device="cuda")
+        self.ext.rle_hybrid_batch(base.data_ptr(), dt.data_ptr(),
+                                  len(pages), levels.data_ptr(), self.s)
+        return self._valid_parts(levels, total)
+
+    def _chunk_plain_fixed(self, pages, total, all_valid) -> Column:
+        from ..column import torch_dtype
+
+        np_dt = _PHYS_NP[self.phys]
+        blob = b"".join(p.values for p in pages)
+        dense_np = np.frombuffer(blob, dtype=np_dt,
+                                 count=len(blob) // np_dt.itemsize)
+        dense = torch.from_numpy(dense_np.copy()).cuda()
+        tdt = torch_dtype(self.dtype)
+        if dense.dtype != tdt:
+            dense = self._cast_raw(dense, tdt)
+        if all_valid:
+            return Column(self.dtype, total, dense[:total], None,
+                          null_count=0)
+        mask, valid_idx, n_valid = self._chunk_validity(pages, total)
+        if mask is None:
+            return Column(self.dtype, total, dense[:total], None,
+                          null_count=0)
+        out = torch.zeros(total, dtype=tdt, device="cuda")
+        if n_valid:
+            self.ext.scatter_fixed(self.dtype.itemsize, dense.data_ptr(),
+                                   valid_idx.data_ptr(), out.data_ptr(),
+                                   n_valid, self.s)
+        return Column(self.dtype, total, out, mask, null_count=None)
+
+    def _chunk_dict(self, pages, total) -> Optional[Column]:
+        if self.dict_fixed is None and self.dict_str is None:
+            return None
+        # batch-decode every page's index stream in one launch; first
+        # value byte of each page is the bit width
+        blob = b"".join(p.values[1:] for p in pages)
+        base = torch.from_numpy(np.frombuffer(
+            blob, dtype=np.uint8).copy()).cuda() if blob else \
+            torch.zeros(1, dtype=torch.uint8, device="cuda")
+        descs = np.empty((len(pages), 5), dtype=np.int64)
+        src = out = 0
+        for i, p in enumerate(pages):
+            nb = len(p.values) - 1
+            descs[i] = (src, nb, out, p.n, p.values[0] if p.values else 0)
+            src += nb
+            out += p.n
+        dt = torch.from_numpy(descs).cuda()
+        ridx = torch.empty(total, dtype=torch.int32, device="cuda")
+        self.ext.rle_hybrid_batch(base.data_ptr(), dt.data_ptr(),
+                                  len(pages), ridx.data_ptr(), self.s)
+        return self._gather_dict(ridx, total, None)
 
     # -- dictionary ------------------------------------------------------
     def _load_dict(self, data: bytes, count: int):
@@ -109,55 +278,46 @@ class _ChunkDecoder:
             vals = np.frombuffer(data, dtype=_PHYS_NP[self.phys], count=count)
             self.dict_fixed = torch.from_numpy(vals.copy()).cuda()
         elif self.phys == "BYTE_ARRAY":
-            offsets = np.empty(count + 1, dtype=np.int32)
-            chunks = []
-            p = 0
-            total = 0
-            mv = memoryview(data)
-            for i in range(count):
-                (ln,) = pystruct.unpack_from("<I", mv, p)
-                p += 4
-                chunks.append(bytes(mv[p: p + ln]))
-                p += ln
-                offsets[i] = total
-                total += ln
-            offsets[count] = total
-            blob = b"".join(chunks)
-            self.dict_str = (
-                torch.from_numpy(offsets).cuda(),
-                torch.from_numpy(
-                    np.frombuffer(blob, dtype=np.uint8).copy()).cuda()
-                if blob else torch.zeros(0, dtype=torch.uint8, device="cuda"),
-            )
+            # device-side parse of the length-prefixed records: the 1M-entry
+            # customer dictionary took ~300ms in a python loop here
+            self.dict_str = self._byte_array_dense(data, count)
         else:
             raise NotImplementedError(f"dict for {self.phys}")
 
+    def _byte_array_dense(self, data: bytes, count: int):
+        """Parse parquet length-prefixed BYTE_ARRAY records on device ->
+        (offsets int32 [count+1] cuda, compact bytes cuda)."""
+        from ..ops import gpu_backend as gb
+
+        page = torch.from_numpy(np.frombuffer(data, dtype=np.uint8).copy()) \
+            .cuda() if data else torch.zeros(1, dtype=torch.uint8,
+                                             device="cuda")
+        starts = torch.empty(max(count, 1), dtype=torch.int32,
+                             device="cuda")[:count]
+        lens = torch.empty(max(count, 1), dtype=torch.int64,
+                           device="cuda")[:count]
+        err = torch.zeros(1, dtype=torch.int32, device="cuda")
+        if count:
+            self.ext.str_plain_offsets(page.data_ptr(), len(data), count,
+                                       starts.data_ptr(), lens.data_ptr(),
+                                       err.data_ptr(), self.s)
+        if count and int(err.item()) > 0:
+            raise NotImplementedError("corrupt byte-array records")
+        scanned, total = gb._exclusive_scan_i64(lens) if count else (lens, 0)
+        out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
+                                device="cuda")[:total]
+        if total:
+            self.ext.substr_copy(page.data_ptr(), starts.data_ptr(),
+                                 lens.data_ptr(), scanned.data_ptr(),
+                                 out_bytes.data_ptr(), count, self.s)
+        offs = torch.empty(count + 1, dtype=torch.int32, device="cuda")
+        if count:
+            self.ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(),
+                                    count, self.s)
+        offs[count] = total
+        return offs, out_bytes
+
     # -- data pages ------------------------------------------------------
-    def _decode_data_page(self, data: bytes, hdr) -> Column:
-        n = hdr.num_values
-        pos = 0
-        levels = None
-        if self.max_def > 0:
-            (lvl_len,) = pystruct.unpack_from("<I", data, 0)
-            pos = 4 + lvl_len
-            levels = self._decode_levels(data[4:pos], n)
-        values = data[pos:]
-        return self._materialize(values, n, levels, hdr.encoding)
-
-    def _decode_data_page_v2(self, payload: bytes, ph) -> Column:
-        hdr = ph.data_page_v2
-        n = hdr.num_values
-        lvl_len = hdr.def_levels_byte_length + hdr.rep_levels_byte_length
-        levels = None
-        if self.max_def > 0 and hdr.def_levels_byte_length:
-            levels = self._decode_levels(
-                payload[hdr.rep_levels_byte_length: lvl_len], n)
-        vals = payload[lvl_len:]
-        if hdr.is_compressed and self.codec is not None:
-            vals = _decompress(self.codec, vals,
-                               ph.uncompressed_page_size - lvl_len)
-        return self._materialize(vals, n, levels, hdr.encoding)
-
     def _decode_levels(self, rle: bytes, n: int) -> torch.Tensor:
         dev = torch.from_numpy(
             np.frombuffer(rle, dtype=np.uint8).copy()).cuda()
@@ -218,36 +378,9 @@ class _ChunkDecoder:
                                        n_valid, self.s)
             return Column(self.dtype, n, out, mask, null_count=None)
         if encoding == PLAIN and self.phys == "BYTE_ARRAY":
-            page = torch.from_numpy(np.frombuffer(
-                values, dtype=np.uint8).copy()).cuda()
-            starts = torch.empty(max(n_valid, 1), dtype=torch.int32,
-                                 device="cuda")[:n_valid]
-            lens = torch.empty(max(n_valid, 1), dtype=torch.int64,
-                               device="cuda")[:n_valid]
-            err = torch.zeros(1, dtype=torch.int32, device="cuda")
-            if n_valid:
-                self.ext.str_plain_offsets(page.data_ptr(), page.numel(),
-                                           n_valid, starts.data_ptr(),
-                                           lens.data_ptr(), err.data_ptr(),
-                                           self.s)
-            if int(err.item()) > 0:
-                raise NotImplementedError("corrupt plain byte-array page")
             from ..ops import gpu_backend as gb
 
-            scanned, total = gb._exclusive_scan_i64(lens) if n_valid                 else (lens, 0)
-            out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
-                                    device="cuda")[:total]
-            if total:
-                self.ext.substr_copy(page.data_ptr(), starts.data_ptr(),
-                                     lens.data_ptr(), scanned.data_ptr(),
-                                     out_bytes.data_ptr(), n_valid, self.s)
-            offs_dense = torch.empty(n_valid + 1, dtype=torch.int32,
-                                     device="cuda")
-            if n_valid:
-                self.ext.narrow_i64_i32(scanned.data_ptr(),
-                                        offs_dense.data_ptr(), n_valid,
-                                        self.s)
-            offs_dense[n_valid] = total
+            offs_dense, out_bytes = self._byte_array_dense(values, n_valid)
             dense_col = Column(DType.string(), n_valid, out_bytes, None,
                                offs_dense, 0)
             if not nulls:
