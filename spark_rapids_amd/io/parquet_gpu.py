@@ -285,36 +285,35 @@ class _ChunkDecoder:
             raise NotImplementedError(f"dict for {self.phys}")
 
     def _byte_array_dense(self, data: bytes, count: int):
-        """Parse parquet length-prefixed BYTE_ARRAY records on device ->
-        (offsets int32 [count+1] cuda, compact bytes cuda)."""
-        from ..ops import gpu_backend as gb
-
-        page = torch.from_numpy(np.frombuffer(data, dtype=np.uint8).copy()) \
-            .cuda() if data else torch.zeros(1, dtype=torch.uint8,
-                                             device="cuda")
-        starts = torch.empty(max(count, 1), dtype=torch.int32,
-                             device="cuda")[:count]
-        lens = torch.empty(max(count, 1), dtype=torch.int64,
-                           device="cuda")[:count]
-        err = torch.zeros(1, dtype=torch.int32, device="cuda")
-        if count:
-            self.ext.str_plain_offsets(page.data_ptr(), len(data), count,
-                                       starts.data_ptr(), lens.data_ptr(),
-                                       err.data_ptr(), self.s)
-        if count and int(err.item()) > 0:
+        """Parse parquet length-prefixed BYTE_ARRAY records -> (offsets
+        int32 [count+1] cuda, compact bytes cuda). The serial record walk
+        runs on the host in C++ (see byte_array_offsets_host); the GPU
+        does the parallel payload compaction."""
+        if count == 0:
+            return (torch.zeros(1, dtype=torch.int32, device="cuda"),
+                    torch.zeros(0, dtype=torch.uint8, device="cuda"))
+        arr = np.frombuffer(data, dtype=np.uint8)
+        starts_h = np.empty(count, dtype=np.int32)
+        lens_h = np.empty(count, dtype=np.int64)
+        total = self.ext.byte_array_offsets_host(
+            arr.ctypes.data, len(data), count, starts_h.ctypes.data,
+            lens_h.ctypes.data)
+        if total < 0:
             raise NotImplementedError("corrupt byte-array records")
-        scanned, total = gb._exclusive_scan_i64(lens) if count else (lens, 0)
+        offs_h = np.empty(count + 1, dtype=np.int64)
+        offs_h[0] = 0
+        np.cumsum(lens_h, out=offs_h[1:])
+        page = torch.from_numpy(arr.copy()).cuda()
+        starts = torch.from_numpy(starts_h).cuda()
+        lens = torch.from_numpy(lens_h).cuda()
+        scanned = torch.from_numpy(offs_h[:-1]).cuda()
         out_bytes = torch.empty(max(total, 1), dtype=torch.uint8,
                                 device="cuda")[:total]
         if total:
             self.ext.substr_copy(page.data_ptr(), starts.data_ptr(),
                                  lens.data_ptr(), scanned.data_ptr(),
                                  out_bytes.data_ptr(), count, self.s)
-        offs = torch.empty(count + 1, dtype=torch.int32, device="cuda")
-        if count:
-            self.ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(),
-                                    count, self.s)
-        offs[count] = total
+        offs = torch.from_numpy(offs_h.astype(np.int32)).cuda()
         return offs, out_bytes
 
     # -- data pages ------------------------------------------------------
