@@ -270,6 +270,11 @@ class GroupedData:
         self.keys = keys
         self.grouping_sets = grouping_sets
 
+    # grouping-set aggs that re-aggregate exactly from finest-level
+    # partials (distributive): coarser rollup/cube levels are computed
+    # from the small base result instead of Expand-replicating the input
+    _HIERARCHICAL_OK = {"sum", "count", "count_all", "min", "max"}
+
     def agg(self, *aggs: AggExpr) -> DataFrame:
         if any(a.distinct for a in aggs):
             if self.grouping_sets is not None:
@@ -278,6 +283,8 @@ class GroupedData:
         if self.grouping_sets is None:
             return DataFrame(self.df.session,
                              L.Aggregate(self.keys, list(aggs), self.df.plan))
+        if all(a.op in self._HIERARCHICAL_OK for a in aggs):
+            return self._agg_hierarchical(list(aggs))
         from .expr.expressions import Alias, Literal
 
         child = self.df.plan
@@ -301,6 +308,58 @@ class GroupedData:
         group = self.keys + [_col("spark_grouping_id")]
         return DataFrame(self.df.session,
                          L.Aggregate(group, list(aggs), expand))
+
+    def _agg_hierarchical(self, aggs: List[AggExpr]) -> DataFrame:
+        """Rollup/cube via hierarchical re-aggregation (distributive aggs
+        only): aggregate ONCE at the full key set, then each coarser
+        grouping set re-aggregates that small result — instead of Spark's
+        Expand path which replicates the whole input n_sets times
+        (GpuExpandExec feeding GpuHashAggregateExec). Counts re-aggregate
+        as sums; decimal re-sums are cast back to the finest sum dtype
+        (same null-on-overflow bound Spark applies)."""
+        from .expr.expressions import Alias, Literal
+
+        key_names = [k.output_name() for k in self.keys]
+        agg_names = [a.output_name() for a in aggs]
+        base = L.Aggregate(self.keys, list(aggs), self.df.plan)
+        bs = base.schema()
+        key_dtype = {k: bs.field(k).dtype for k in key_names}
+        agg_dtype = {n: bs.field(n).dtype for n in agg_names}
+        levels = []
+        for kept in self.grouping_sets:
+            gid = 0
+            for i, k in enumerate(key_names):
+                if k not in kept:
+                    gid |= 1 << (len(key_names) - 1 - i)
+            gid_lit = Alias(Literal(gid, INT32), "spark_grouping_id")
+            if set(kept) == set(key_names):
+                proj = [Alias(_col(k), k) for k in key_names] + [gid_lit] \
+                    + [Alias(_col(n), n) for n in agg_names]
+                levels.append(L.Project(proj, base))
+                continue
+            # grouping by the constant gid keeps empty-input semantics:
+            # zero input rows -> zero output rows even for the () set
+            pre = L.Project([Alias(_col(k), k) for k in kept] + [gid_lit]
+                            + [Alias(_col(n), n) for n in agg_names], base)
+            re_aggs = [AggExpr("sum" if a.op in ("count", "count_all")
+                               else a.op, _col(n)).alias(n)
+                       for a, n in zip(aggs, agg_names)]
+            g2 = L.Aggregate([_col(k) for k in kept]
+                             + [_col("spark_grouping_id")], re_aggs, pre)
+            g2s = g2.schema()
+            proj = []
+            for k in key_names:
+                proj.append(Alias(_col(k), k) if k in kept
+                            else Alias(Literal(None, key_dtype[k]), k))
+            proj.append(Alias(_col("spark_grouping_id"),
+                              "spark_grouping_id"))
+            for n in agg_names:
+                e = _col(n)
+                if g2s.field(n).dtype != agg_dtype[n]:
+                    e = e.cast(agg_dtype[n])
+                proj.append(Alias(e, n))
+            levels.append(L.Project(proj, g2))
+        return DataFrame(self.df.session, L.Union(levels))
 
     def _agg_distinct(self, aggs: List[AggExpr]) -> DataFrame:
         """Single-distinct-column rewrite (Spark RewriteDistinctAggregates):

@@ -74,3 +74,38 @@ def test_gpu_expand_placement():
     tree = (_df(s, 10).rollup("b").agg(count_star())
             .physical_plan().tree_string())
     assert "GpuExpand" in tree, tree
+
+
+def test_rollup_hierarchical_decimal_and_counts():
+    """The hierarchical grouping-set path must match brute-force totals,
+    including decimal sums (re-agg casts back to the finest sum dtype)."""
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, sum_, count_star, DType
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({
+        "a": ["x", "x", "y", "y", "y"],
+        "b": [1, 2, 1, 1, None],
+        "v": [10, 20, 30, 40, 50],
+    })
+    df = df.with_column("d", col("v").cast(DType.decimal(7, 2)))
+    out = df.rollup("a", "b").agg(sum_(col("d")).alias("s"),
+                                  count_star().alias("c")).to_pydict()
+    rows = {(a, b, g): (str(sv), c) for a, b, g, sv, c in
+            zip(out["a"], out["b"], out["spark_grouping_id"], out["s"],
+                out["c"])}
+    assert rows[("x", 1, 0)] == ("10.00", 1)
+    assert rows[("y", None, 0)] == ("50.00", 1)   # null key, finest level
+    assert rows[("x", None, 1)] == ("30.00", 2)
+    assert rows[("y", None, 1)] == ("120.00", 3)
+    assert rows[(None, None, 3)] == ("150.00", 5)
+
+
+def test_rollup_empty_input_no_rows():
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, sum_
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({"a": [1], "v": [1.0]}).filter(col("a") > 5)
+    out = df.rollup("a").agg(sum_(col("v"))).collect()
+    assert out == []
