@@ -37,6 +37,37 @@ _PHYS_NP = {
 }
 
 
+_PINNED = __import__("threading").local()
+
+
+def _upload(data, torch_dtype=torch.uint8) -> torch.Tensor:
+    """bytes/ndarray -> device tensor through a reusable pinned staging
+    buffer (one host memcpy + async DMA, instead of frombuffer().copy()
+    + a pageable H2D which stages internally anyway). Per-thread buffer:
+    the multi-file prefetch pool uploads concurrently."""
+    if isinstance(data, np.ndarray):
+        arr = data.view(np.uint8).reshape(-1)
+    else:
+        arr = np.frombuffer(data, dtype=np.uint8)
+    nbytes = arr.nbytes
+    if nbytes == 0:
+        return torch.zeros(0, dtype=torch_dtype, device="cuda")
+    buf = getattr(_PINNED, "buf", None)
+    if buf is None or buf.numel() < nbytes:
+        cap = max(nbytes, 1 << 20)
+        cap = 1 << (cap - 1).bit_length()
+        buf = torch.empty(cap, dtype=torch.uint8, pin_memory=True)
+        _PINNED.buf = buf
+    buf.numpy()[:nbytes] = arr
+    dev = torch.empty(nbytes, dtype=torch.uint8, device="cuda")
+    # blocking copy: the pinned buffer is reused by the next upload on
+    # this thread, so the DMA must complete before returning
+    dev.copy_(buf[:nbytes])
+    itemsize = torch.empty(0, dtype=torch_dtype).element_size()
+    return dev.view(torch_dtype) if itemsize == 1 else \
+        dev[: (nbytes // itemsize) * itemsize].view(torch_dtype)
+
+
 def _codec(name: str):
     import pyarrow as pa
 
@@ -211,8 +242,7 @@ class _ChunkDecoder:
         pages' def-level RLE streams in ONE launch, then run the validity
         chain once over the contiguous levels array."""
         levels_blob = b"".join(p.levels for p in pages)
-        base = torch.from_numpy(np.frombuffer(
-            levels_blob, dtype=np.uint8).copy()).cuda()
+        base = _upload(levels_blob)
         descs = np.empty((len(pages), 5), dtype=np.int64)
         src = out = 0
         for i, p in enumerate(pages):
@@ -229,10 +259,10 @@ class _ChunkDecoder:
         from ..column import torch_dtype
 
         np_dt = _PHYS_NP[self.phys]
-        blob = b"".join(p.values for p in pages)
-        dense_np = np.frombuffer(blob, dtype=np_dt,
-                                 count=len(blob) // np_dt.itemsize)
-        dense = torch.from_numpy(dense_np.copy()).cuda()
+        blob = pages[0].values if len(pages) == 1 \
+            else b"".join(p.values for p in pages)
+        dense = _upload(blob, torch.from_numpy(
+            np.empty(0, dtype=np_dt)).dtype)
         tdt = torch_dtype(self.dtype)
         if dense.dtype != tdt:
             dense = self._cast_raw(dense, tdt)
@@ -256,8 +286,7 @@ class _ChunkDecoder:
         # batch-decode every page's index stream in one launch; first
         # value byte of each page is the bit width
         blob = b"".join(p.values[1:] for p in pages)
-        base = torch.from_numpy(np.frombuffer(
-            blob, dtype=np.uint8).copy()).cuda() if blob else \
+        base = _upload(blob) if blob else \
             torch.zeros(1, dtype=torch.uint8, device="cuda")
         descs = np.empty((len(pages), 5), dtype=np.int64)
         src = out = 0
@@ -303,7 +332,7 @@ class _ChunkDecoder:
         offs_h = np.empty(count + 1, dtype=np.int64)
         offs_h[0] = 0
         np.cumsum(lens_h, out=offs_h[1:])
-        page = torch.from_numpy(arr.copy()).cuda()
+        page = _upload(arr)
         starts = torch.from_numpy(starts_h).cuda()
         lens = torch.from_numpy(lens_h).cuda()
         scanned = torch.from_numpy(offs_h[:-1]).cuda()
