@@ -212,6 +212,9 @@ class DataFrame:
         return self._phys
 
     def collect_batch(self) -> ColumnBatch:
+        from .plan.physical import new_execution
+
+        new_execution()
         exec_ = self.physical_plan()
         from .metrics import instrument
 
@@ -321,7 +324,7 @@ class GroupedData:
 
         key_names = [k.output_name() for k in self.keys]
         agg_names = [a.output_name() for a in aggs]
-        base = L.Aggregate(self.keys, list(aggs), self.df.plan)
+        base = L.Cached(L.Aggregate(self.keys, list(aggs), self.df.plan))
         bs = base.schema()
         key_dtype = {k: bs.field(k).dtype for k in key_names}
         agg_dtype = {n: bs.field(n).dtype for n in agg_names}

@@ -44,6 +44,24 @@ class Scan(LogicalPlan):
         return f"Scan({self.label})"
 
 
+class Cached(LogicalPlan):
+    """A subplan shared by several parents (e.g. the finest-level
+    aggregate feeding every rollup/cube grouping set): the physical plan
+    converts it to ONE exec that materializes once per execution instead
+    of re-running the subtree per consumer. Reference analogue:
+    ReusedExchangeExec / AQE stage reuse."""
+
+    def __init__(self, child: LogicalPlan):
+        self.child = child
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        return self.child.schema()
+
+
 class Filter(LogicalPlan):
     def __init__(self, condition: Expression, child: LogicalPlan):
         self.condition = condition

@@ -71,6 +71,15 @@ def push_filters(plan: L.LogicalPlan) -> L.LogicalPlan:
     Safety: inner/cross joins push to either side; left/semi/anti joins
     push only left-side conjuncts (right-side predicates would change the
     null-extension semantics)."""
+    if isinstance(plan, L.Cached):
+        # shared subplan: rewrite ONCE and return the same node to every
+        # parent, else each consumer gets its own copy and the physical
+        # CachedExec sharing is lost
+        memo = getattr(plan, "_pushed_memo", None)
+        if memo is None:
+            memo = L.Cached(push_filters(plan.child))
+            plan._pushed_memo = memo
+        return memo
     # rewrite children first
     kids = [push_filters(c) for c in plan.children]
     plan = _with_children(plan, kids)
@@ -141,6 +150,14 @@ def prune_columns(plan: L.LogicalPlan,
                   needed: Optional[Set[str]] = None) -> L.LogicalPlan:
     """Return an equivalent plan where children materialize only the columns
     the ancestors reference. needed=None means every output column."""
+    if isinstance(plan, L.Cached):
+        memo = getattr(plan, "_pruned_memo", None)
+        if memo is None:
+            # consumers may need different subsets; cache the full output
+            # (it is small by construction) and let them project
+            memo = L.Cached(prune_columns(plan.child, None))
+            plan._pruned_memo = memo
+        return memo
     if isinstance(plan, L.Scan):
         # push the needed-column set into sources that can skip IO+decode
         # for unused columns (parquet/orc: per-column chunks on disk)

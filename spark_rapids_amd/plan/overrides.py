@@ -364,6 +364,16 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
             return P.CoalesceBatchesExec(scan, conf.get(BATCH_SIZE_BYTES))
         return scan
 
+    if isinstance(node, L.Cached):
+        # one physical exec per logical node, however many parents
+        # reference it (hierarchical rollup levels share their base)
+        memo = getattr(node, "_phys_memo", None)
+        if memo is None:
+            inner = _convert(node.child, conf, tagger, gpu_wanted)
+            memo = P.CachedExec(inner.device, inner, node.schema())
+            node._phys_memo = memo
+        return memo
+
     kids = [_ensure_device(_convert(c, conf, tagger, gpu_wanted), device)
             for c in node.children]
 

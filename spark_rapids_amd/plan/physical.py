@@ -22,6 +22,16 @@ from ..types import BOOL, DType, FLOAT64, INT64
 from ..memory.retry import with_retry_split
 
 
+# bumped by the API layer at every root execution: CachedExec instances
+# recompute when the epoch moves, replay within the same execution
+EXECUTION_EPOCH = 0
+
+
+def new_execution():
+    global EXECUTION_EPOCH
+    EXECUTION_EPOCH += 1
+
+
 class PhysicalExec:
     def __init__(self, device: str, schema: Schema,
                  children: Sequence["PhysicalExec"] = ()):  # noqa: D401
@@ -1129,6 +1139,25 @@ class GenerateExec(PhysicalExec):
         mode = "posexplode" if self.pos else "explode"
         o = "_outer" if self.outer else ""
         return f"{self.name()}[{mode}{o}({self.column})]"
+
+
+class CachedExec(PhysicalExec):
+    """Materializes its child once per root execution and replays the
+    batches to every consumer (hierarchical rollup levels share the
+    finest-level aggregate; reference analogue: ReusedExchangeExec).
+    The epoch check makes repeated executions of a cached physical plan
+    (benchmark steps) recompute instead of serving stale results."""
+
+    def __init__(self, device: str, child: PhysicalExec, schema: Schema):
+        super().__init__(device, schema, [child])
+        self._epoch = -1
+        self._batches = None
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        if self._epoch != EXECUTION_EPOCH or self._batches is None:
+            self._batches = list(self.children[0].execute())
+            self._epoch = EXECUTION_EPOCH
+        yield from self._batches
 
 
 class UnionExec(PhysicalExec):

@@ -69,11 +69,18 @@ def test_gpu_rollup_matches_cpu():
 
 
 @pytest.mark.gpu
-def test_gpu_expand_placement():
+def test_gpu_rollup_plan_shapes():
     s = sr.Session()
+    # distributive aggs: hierarchical re-aggregation (Cached + Union)
     tree = (_df(s, 10).rollup("b").agg(count_star())
             .physical_plan().tree_string())
-    assert "GpuExpand" in tree, tree
+    assert "GpuCached" in tree and "GpuUnion" in tree, tree
+    # non-distributive aggs (avg) still take the Expand path
+    from spark_rapids_amd import avg, col
+
+    tree2 = (_df(s, 10).rollup("b").agg(avg(col("v")))
+             .physical_plan().tree_string())
+    assert "GpuExpand" in tree2, tree2
 
 
 def test_rollup_hierarchical_decimal_and_counts():
