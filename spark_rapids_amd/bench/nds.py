@@ -199,6 +199,7 @@ _FACT_OPTS = dict(
     use_dictionary=False,
     store_decimal_as_integer=True,   # decimal(7,2) -> INT32 physical
     data_page_size=1 << 20,
+    row_group_size=1 << 23,      # one row group per staged file
 )
 # dims: dictionary-encoded strings (the GPU string decode path); the large
 # dictionary limit keeps even c_customer_id (1M distinct) dictionary-coded
