@@ -538,16 +538,33 @@ def torch_dtype_of(dt: DType):
     return torch_dtype(dt)
 
 
-def read_parquet_gpu(path: str, columns: List[str]) -> ColumnBatch:
+# footer/metadata cache keyed by (path, mtime): the reference's FileCache
+# caches parquet footers for exactly this reason — every query of a power
+# run re-opens the same files (SURVEY.md §2.3 "File cache" row)
+_META_CACHE: dict = {}
+
+
+def _file_meta(path: str):
+    import os
+
     import pyarrow.parquet as pq
 
+    key = (path, os.path.getmtime(path))
+    hit = _META_CACHE.get(key)
+    if hit is None:
+        pf = pq.ParquetFile(path)
+        hit = (pf.metadata, pf.schema_arrow, pf.schema)
+        if len(_META_CACHE) > 8192:
+            _META_CACHE.clear()
+        _META_CACHE[key] = hit
+    return hit
+
+
+def read_parquet_gpu(path: str, columns: List[str]) -> ColumnBatch:
     from ..ops.gpu_backend import ext, _stream
     from .parquet import arrow_to_dtype
 
-    pf = pq.ParquetFile(path)
-    md = pf.metadata
-    arrow_schema = pf.schema_arrow
-    pq_schema = pf.schema
+    md, arrow_schema, pq_schema = _file_meta(path)
     name_to_idx = {md.row_group(0).column(j).path_in_schema: j
                    for j in range(md.num_columns)} if md.num_row_groups else {}
     s = _stream()
