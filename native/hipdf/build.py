@@ -34,6 +34,7 @@ KERNELS = [
     "kernels/window.hip",
     "kernels/decimal128.hip",
     "kernels/regex.hip",
+    "pool.hip",
 ]
 
 CXXFLAGS = ["-O3", "-std=c++20", "-fPIC", f"--offload-arch={ARCH}",

@@ -211,6 +211,30 @@ static void check_async() {
 #define PM(x) reinterpret_cast<void*>(x)
 #define S(x) reinterpret_cast<hipStream_t>(x)
 
+// ---- device memory pool (pool.hip) --------------------------------------
+extern "C" {
+int hipdf_pool_init(double, size_t);
+int hipdf_pool_active();
+typedef int (*hipdf_failure_cb)(size_t, int);
+void hipdf_pool_set_failure_cb(hipdf_failure_cb);
+size_t hipdf_pool_used();
+size_t hipdf_pool_reserved();
+size_t hipdf_pool_high_watermark();
+int hipdf_pool_selftest();
+}
+
+static py::object g_spill_cb;
+
+static int spill_cb_trampoline(size_t needed, int retry) {
+  py::gil_scoped_acquire gil;
+  try {
+    return g_spill_cb((size_t)needed, retry).cast<int>();
+  } catch (...) {
+    PyErr_Clear();
+    return 0;
+  }
+}
+
 // Host-side walk of parquet length-prefixed BYTE_ARRAY records. The chain
 // pos -> len -> pos is inherently serial, so it runs on the CPU (one
 // dependent L1 load per record, ~2ns) instead of a single GPU thread
@@ -235,6 +259,18 @@ static int64_t byte_array_offsets_walk(const uint8_t* p, int64_t nbytes,
 }
 
 PYBIND11_MODULE(hipdf, m) {
+  m.def("pool_init", [](double fraction, size_t bytes) {
+    return hipdf_pool_init(fraction, bytes);
+  });
+  m.def("pool_active", []() { return hipdf_pool_active() != 0; });
+  m.def("pool_used", []() { return hipdf_pool_used(); });
+  m.def("pool_reserved", []() { return hipdf_pool_reserved(); });
+  m.def("pool_high_watermark", []() { return hipdf_pool_high_watermark(); });
+  m.def("pool_selftest", []() { return hipdf_pool_selftest(); });
+  m.def("pool_set_spill_cb", [](py::object f) {
+    g_spill_cb = f;
+    hipdf_pool_set_failure_cb(f.is_none() ? nullptr : &spill_cb_trampoline);
+  });
   m.def("byte_array_offsets_host",
         [](int64_t data, int64_t nbytes, int64_t count, int64_t starts,
            int64_t lens) -> int64_t {

@@ -425,6 +425,16 @@ class Session:
         from .io import filecache as _fc
 
         _fc.configure(self.conf.get(FILECACHE))
+        from .config import MEM_POOL_FRACTION, MEM_POOL_MODE
+
+        if self.conf.sql_enabled \
+                and str(self.conf.get(MEM_POOL_MODE)).upper() == "HIPDF":
+            from .config import MEM_SPILL_WATERMARK
+            from .memory import device_pool
+            from .memory import spill as _spill
+
+            device_pool.activate(self.conf.get(MEM_POOL_FRACTION))
+            _spill.configure_watermark(self.conf.get(MEM_SPILL_WATERMARK))
 
     # ---- conf ----------------------------------------------------------
     def set(self, key: str, value) -> "Session":

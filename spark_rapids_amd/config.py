@@ -122,6 +122,18 @@ DECIMAL_ENABLED = bool_conf(
 MEM_POOL_FRACTION = float_conf(
     "spark.rapids.memory.gpu.allocFraction", 0.9,
     "Fraction of free device memory the pool may grow to.")
+MEM_POOL_MODE = str_conf(
+    "spark.rapids.memory.gpu.pool", "HIPDF",
+    "Device memory pool: HIPDF installs the hipdf sub-allocator (hipMalloc "
+    "slab + spill-before-OOM failure callback, RMM-pool analogue) as "
+    "torch's CUDA allocator; TORCH keeps torch's caching allocator. The "
+    "HIPDF pool can only install before the first device allocation of "
+    "the process.")
+MEM_SPILL_WATERMARK = float_conf(
+    "spark.rapids.memory.gpu.spillWatermark", 0.85,
+    "Pool-usage fraction above which spillable batches are proactively "
+    "moved to host (reference analogue: spill from the RMM event handler "
+    "before allocations fail).")
 PINNED_POOL_SIZE = bytes_conf(
     "spark.rapids.memory.pinnedPool.size", 8 << 30,
     "Size of the pinned host memory pool used for spill and H2D/D2H staging.")

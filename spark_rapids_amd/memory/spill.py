@@ -148,6 +148,16 @@ def _spill_dir() -> str:
     return d
 
 
+# pool-usage fraction for proactive spill; set by Session from
+# spark.rapids.memory.gpu.spillWatermark (None until a session configures)
+_watermark: Optional[float] = None
+
+
+def configure_watermark(w: Optional[float]):
+    global _watermark
+    _watermark = w
+
+
 class SpillStore:
     """Registry of live spillables; spills lowest-priority first."""
 
@@ -157,6 +167,12 @@ class SpillStore:
     def register(self, h: SpillableBatch):
         with _store_lock:
             self._handles[h.id] = h
+        if h.state == DEVICE and _watermark is not None:
+            # proactive spill: keep pool usage under the watermark so the
+            # failure callback is the backstop, not the steady state
+            from . import device_pool
+
+            device_pool.maybe_spill(_watermark)
 
     def unregister(self, h: SpillableBatch):
         with _store_lock:

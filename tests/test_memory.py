@@ -1,8 +1,11 @@
 """Spill framework + OOM retry + semaphore unit tests (CPU-side semantics;
 reference analogues: SpillFrameworkSuite, WithRetrySuite, GpuSemaphoreSuite)."""
+import os
 import threading
 
 import pytest
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 from spark_rapids_amd import Column, ColumnBatch, INT64, Session, col
 from spark_rapids_amd.memory.retry import (GpuRetryOOM, GpuSplitAndRetryOOM,
@@ -300,3 +303,27 @@ def test_gpu_topn_matches_cpu():
     g = q(sr.Session())
     c = q(sr.Session({"spark.rapids.sql.enabled": False}))
     assert g == c
+
+
+def test_pool_arena_selftest():
+    """Host-memory run of the device-pool arena logic (alloc/free,
+    coalescing, exhaustion) — native/hipdf/pool.hip."""
+    import hipdf
+
+    assert hipdf.pool_selftest() == 0
+
+
+@pytest.mark.gpu
+def test_pool_spill_before_oom():
+    """hipdf pool as the torch allocator: exhaustion spills registered
+    batches via the failure callback before the allocation fails
+    (VERDICT round 1 #5). Subprocess: the allocator must install before
+    the process's first device allocation."""
+    import subprocess
+    import sys
+
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "tests", "pool_probe.py")],
+        capture_output=True, text=True, timeout=300, cwd=REPO)
+    assert r.returncode == 0, r.stdout[-2000:] + r.stderr[-2000:]
+    assert "POOL_OK" in r.stdout, r.stdout
