@@ -223,12 +223,15 @@ size_t hipdf_pool_high_watermark();
 int hipdf_pool_selftest();
 }
 
-static py::object g_spill_cb;
+// heap-allocated and intentionally leaked: a static py::object would run
+// its destructor at .so unload AFTER Py_Finalize and crash the exit
+static py::object* g_spill_cb = nullptr;
 
 static int spill_cb_trampoline(size_t needed, int retry) {
+  if (!g_spill_cb || !Py_IsInitialized()) return 0;
   py::gil_scoped_acquire gil;
   try {
-    return g_spill_cb((size_t)needed, retry).cast<int>();
+    return (*g_spill_cb)((size_t)needed, retry).cast<int>();
   } catch (...) {
     PyErr_Clear();
     return 0;
@@ -268,7 +271,8 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("pool_high_watermark", []() { return hipdf_pool_high_watermark(); });
   m.def("pool_selftest", []() { return hipdf_pool_selftest(); });
   m.def("pool_set_spill_cb", [](py::object f) {
-    g_spill_cb = f;
+    if (!g_spill_cb) g_spill_cb = new py::object();
+    *g_spill_cb = f;
     hipdf_pool_set_failure_cb(f.is_none() ? nullptr : &spill_cb_trampoline);
   });
   m.def("byte_array_offsets_host",
