@@ -34,6 +34,7 @@ KERNELS = [
     "kernels/window.hip",
     "kernels/decimal128.hip",
     "kernels/regex.hip",
+    "kernels/cast_str.hip",
     "pool.hip",
 ]
 

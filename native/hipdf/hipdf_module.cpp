@@ -211,6 +211,14 @@ static void check_async() {
 #define PM(x) reinterpret_cast<void*>(x)
 #define S(x) reinterpret_cast<hipStream_t>(x)
 
+// ---- exact string<->decimal casts (kernels/cast_str.hip) -----------------
+extern "C" {
+void hipdf_str_to_dec(const void*, const void*, const void*, int, int, int,
+                      void*, void*, int64_t, hipStream_t);
+void hipdf_dec_to_str(const void*, int, int, const void*, void*, void*, int,
+                      int64_t, hipStream_t);
+}
+
 // ---- device memory pool (pool.hip) --------------------------------------
 extern "C" {
 int hipdf_pool_init(double, size_t);
@@ -262,6 +270,18 @@ static int64_t byte_array_offsets_walk(const uint8_t* p, int64_t nbytes,
 }
 
 PYBIND11_MODULE(hipdf, m) {
+  m.def("str_to_dec", [](int64_t ao, int64_t ab, int64_t av, int out_kind,
+                         int out_scale, int out_prec, int64_t out,
+                         int64_t ov, int64_t n, int64_t stream) {
+    hipdf_str_to_dec(P(ao), P(ab), P(av), out_kind, out_scale, out_prec,
+                     PM(out), PM(ov), n, S(stream));
+  });
+  m.def("dec_to_str", [](int64_t vals, int in_is_128, int scale,
+                         int64_t out_off, int64_t out_len, int64_t out,
+                         int mode, int64_t n, int64_t stream) {
+    hipdf_dec_to_str(P(vals), in_is_128, scale, P(out_off), PM(out_len),
+                     PM(out), mode, n, S(stream));
+  });
   m.def("pool_init", [](double fraction, size_t bytes) {
     return hipdf_pool_init(fraction, bytes);
   });

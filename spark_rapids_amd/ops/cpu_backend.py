@@ -387,21 +387,7 @@ def cast(col: Column, to: DType) -> Column:
                 out.append(str(v))
         return Column.from_pylist(out, to)
     if src.id is TypeId.STRING:
-        res = np.zeros(len(a), dtype=to.numpy_dtype())
-        valid = av.copy()
-        for i, v in enumerate(a):
-            if not av[i]:
-                continue
-            try:
-                if to.is_decimal:
-                    res[i] = int(round(float(v) * (10 ** to.scale)))
-                elif to.is_floating:
-                    res[i] = float(v)
-                else:
-                    res[i] = int(float(v))
-            except (ValueError, TypeError):
-                valid[i] = False
-        return _make(res, valid if not valid.all() else None, to)
+        return _cast_string_exact(a, av, to)
     if src.id is TypeId.DECIMAL128 or to.id is TypeId.DECIMAL128:
         if src.is_decimal and to.is_decimal:
             vals, valid = _rescale_exact(a, av, src, to)
@@ -445,6 +431,86 @@ def cast(col: Column, to: DType) -> Column:
     with np.errstate(invalid="ignore", over="ignore"):
         res = a.astype(to.numpy_dtype())
     return _make(res, av if not av.all() else None, to)
+
+
+_INT_BOUNDS = {TypeId.INT8: 1 << 7, TypeId.INT16: 1 << 15,
+               TypeId.INT32: 1 << 31, TypeId.INT64: 1 << 63}
+
+
+def _cast_string_exact(a, av, to: DType) -> Column:
+    """string -> numeric with Spark semantics, exact via python decimal
+    (matches the device k_str_to_dec kernel): HALF_UP to the target scale
+    for decimals, truncate-toward-zero for ints (no scientific notation
+    for int targets), NULL on garbage/overflow."""
+    import decimal as pydec
+
+    ctx = pydec.Context(prec=50)
+    n = len(a)
+    valid = av.copy()
+    if to.is_decimal:
+        vals = [0] * n
+        q = pydec.Decimal(1).scaleb(-to.scale)
+        bound = 10 ** to.precision
+        for i, v in enumerate(a):
+            if not av[i]:
+                continue
+            try:
+                d = ctx.create_decimal(str(v).strip())
+                if not d.is_finite():
+                    raise pydec.InvalidOperation
+                u = int(d.quantize(q, rounding=pydec.ROUND_HALF_UP,
+                                   context=ctx).scaleb(to.scale, ctx))
+                if abs(u) >= bound:
+                    raise pydec.InvalidOperation
+                vals[i] = u
+            except (pydec.InvalidOperation, ValueError, ArithmeticError):
+                valid[i] = False
+        arr = np.array(vals, dtype=object) \
+            if to.id is TypeId.DECIMAL128 else np.array(vals, dtype=np.int64)
+        return _make(arr, valid if not valid.all() else None, to)
+    if to.is_integral:
+        res = np.zeros(n, dtype=to.numpy_dtype())
+        lim = _INT_BOUNDS[to.id]
+        for i, v in enumerate(a):
+            if not av[i]:
+                continue
+            sv = str(v).strip()
+            try:
+                if "e" in sv or "E" in sv:
+                    raise ValueError  # Spark: no exponent in int literals
+                d = ctx.create_decimal(sv)
+                if not d.is_finite():
+                    raise ValueError
+                u = int(d.to_integral_value(rounding=pydec.ROUND_DOWN))
+                if not (-lim <= u < lim):
+                    raise ValueError
+                res[i] = u
+            except (pydec.InvalidOperation, ValueError, ArithmeticError):
+                valid[i] = False
+        return _make(res, valid if not valid.all() else None, to)
+    if to.id is TypeId.BOOL:
+        res = np.zeros(n, dtype=np.uint8)
+        yes = {"true", "t", "yes", "y", "1"}
+        no = {"false", "f", "no", "n", "0"}
+        for i, v in enumerate(a):
+            if not av[i]:
+                continue
+            sv = str(v).strip().lower()
+            if sv in yes:
+                res[i] = 1
+            elif sv not in no:
+                valid[i] = False
+        return _make(res, valid if not valid.all() else None, to)
+    # float target
+    res = np.zeros(n, dtype=to.numpy_dtype())
+    for i, v in enumerate(a):
+        if not av[i]:
+            continue
+        try:
+            res[i] = float(v)
+        except (ValueError, TypeError):
+            valid[i] = False
+    return _make(res, valid if not valid.all() else None, to)
 
 
 def _rescale_exact(a, av, src: DType, to: DType):

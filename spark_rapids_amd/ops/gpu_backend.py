@@ -753,14 +753,42 @@ def cast(col: Column, to: DType) -> Column:
     return Column(to, n, out, v, null_count=col._null_count)
 
 
+_STR_TO_KIND = {TypeId.INT8: 2, TypeId.INT16: 3, TypeId.INT32: 4,
+                TypeId.INT64: 5}
+
+
 def _cast_string_to(col: Column, to: DType, v) -> Column:
-    """string -> numeric cast (CastStrings analogue): trim, parse as f64
-    with the CSV field parser over each row's full span, then the generic
-    saturating cast; unparsable -> NULL (Spark non-ANSI)."""
-    if not (to.is_floating or to.is_integral or to.is_decimal):
-        raise NotImplementedError(f"gpu cast string -> {to}")
+    """string -> numeric cast (CastStrings analogue). Integral and decimal
+    targets parse EXACTLY on device (k_str_to_dec: u128 digit accumulate,
+    HALF_UP for decimals, truncate-toward-zero for ints, null on
+    overflow/garbage); float targets ride the CSV f64 parser."""
     n = col.size
     s = _stream()
+    if to.is_decimal or to.is_integral:
+        if to.id is TypeId.DECIMAL128:
+            kind, scale, prec = 1, to.scale, to.precision
+            width = 2 * n
+        elif to.id is TypeId.DECIMAL64:
+            kind, scale, prec = 0, to.scale, to.precision
+            width = n
+        else:
+            kind, scale, prec = _STR_TO_KIND[to.id], 0, 39
+            width = n
+        out64 = torch.empty(max(width, 1), dtype=torch.int64,
+                            device="cuda")[:width]
+        ov = _alloc_mask(n)
+        if n:
+            ext.str_to_dec(col.offsets.data_ptr(), col.data.data_ptr(),
+                           _ptr(v), kind, scale, prec, out64.data_ptr(),
+                           ov.data_ptr(), n, s)
+        if to.is_decimal or to.id is TypeId.INT64:
+            return Column(to, n, out64, ov, null_count=None)
+        narrow = _alloc(n, to)
+        if n:
+            ext.cast(4, _ht(to), out64.data_ptr(), narrow.data_ptr(), n, s)
+        return Column(to, n, narrow, ov, null_count=None)
+    if not to.is_floating:
+        raise NotImplementedError(f"gpu cast string -> {to}")
     trimmed = str_trim(col, "both")
     f64 = torch.empty(max(n, 1), dtype=torch.float64, device="cuda")[:n]
     valid_u8 = torch.empty(max(n, 1), dtype=torch.uint8, device="cuda")[:n]
@@ -787,12 +815,32 @@ def _cast_string_to(col: Column, to: DType, v) -> Column:
 
 
 def _cast_to_string(col: Column, v) -> Column:
-    """integral/date -> string; floats/decimals fall back (format parity
-    with python repr is CPU-side)."""
-    if not col.dtype.is_integral and col.dtype.id is not TypeId.BOOL:
-        raise NotImplementedError(f"gpu cast {col.dtype} -> string")
+    """integral/decimal/date -> string on device; floats fall back (Java
+    shortest-roundtrip float formatting parity is CPU-side, mirroring the
+    reference's castFloatToString incompat gate)."""
     n = col.size
     s = _stream()
+    if col.dtype.is_decimal:
+        is128 = 1 if col.dtype.id is TypeId.DECIMAL128 else 0
+        lens = torch.empty(max(n, 1), dtype=torch.int64, device="cuda")[:n]
+        if n:
+            ext.dec_to_str(col.data.data_ptr(), is128, col.dtype.scale,
+                           0, lens.data_ptr(), 0, 0, n, s)
+        scanned, total = _exclusive_scan_i64(lens) if n else (lens, 0)
+        out = torch.empty(max(total, 1), dtype=torch.uint8,
+                          device="cuda")[:total]
+        if total:
+            ext.dec_to_str(col.data.data_ptr(), is128, col.dtype.scale,
+                           scanned.data_ptr(), lens.data_ptr(),
+                           out.data_ptr(), 1, n, s)
+        offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+        if n:
+            ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+        offs[n] = total
+        return Column(DType.string(), n, out, v, offs,
+                      null_count=None if v is not None else 0)
+    if not col.dtype.is_integral and col.dtype.id is not TypeId.BOOL:
+        raise NotImplementedError(f"gpu cast {col.dtype} -> string")
     i64 = cast(Column(col.dtype, n, col.data, None, null_count=0),
                DType.int64()) if col.dtype.id is not TypeId.INT64         else Column(DType.int64(), n, col.data, None, null_count=0)
     lens = torch.empty(max(n, 1), dtype=torch.int64, device="cuda")[:n]

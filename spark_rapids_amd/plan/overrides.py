@@ -86,11 +86,20 @@ class Tagger:
             pass
         elif isinstance(e, CastExpr):
             src = e.child.dtype(schema)
-            if src.id is TypeId.STRING or e.to.id is TypeId.STRING:
-                # parse/format kernels exist (csv_parse / i64_to_str) but
-                # bit-exact parity with the CPU double parser is still
-                # being verified — round 2 enables them
-                out.append(f"cast {src} -> {e.to} not supported on GPU yet")
+            if src.id is TypeId.STRING:
+                # exact device parser (k_str_to_dec) covers integral,
+                # decimal and bool-free numeric targets; float targets
+                # parse as f64 (csv_parse)
+                if not (e.to.is_integral or e.to.is_decimal
+                        or e.to.is_floating):
+                    out.append(f"cast string -> {e.to} not on GPU yet")
+            elif e.to.id is TypeId.STRING:
+                # ints and decimals format on device (i64_to_str /
+                # k_dec_to_str); float->string needs Java shortest-
+                # round-trip parity, bool/date literal forms are host-side
+                if not (src.is_integral or src.is_decimal) \
+                        or src.id is TypeId.BOOL:
+                    out.append(f"cast {src} -> string not on GPU yet")
             if not self.conf.expr_enabled("Cast"):
                 out.append("expression Cast disabled by conf")
         elif isinstance(e, BinaryExpr):

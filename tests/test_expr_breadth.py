@@ -126,7 +126,8 @@ def test_string_casts(session):
 
     assert out["d"] == [decimal.Decimal("1.25"), decimal.Decimal("42.00"),
                         None, None, None, decimal.Decimal("-700.00")]
-    assert out["i"] == [1, 42, None, None, None, -700]
+    # Spark UTF8String.toLong rejects scientific notation: '-7e2' -> NULL
+    assert out["i"] == [1, 42, None, None, None, None]
     assert out["f"] == [1.25, 42.0, None, None, None, -700.0]
 
 
