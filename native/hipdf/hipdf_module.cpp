@@ -92,6 +92,9 @@ void hipdf_i128_cmp(int, const void*, const void*, const void*, const void*,
 void hipdf_i64_to_i128(const void*, void*, int64_t, hipStream_t);
 void hipdf_i128_rescale(const void*, const void*, void*, void*, int, int,
                         int, int64_t, hipStream_t);
+void hipdf_dec_mul_div_wide(int, const void*, const void*, const void*,
+                            const void*, int, int, void*, void*, int, int,
+                            int, int64_t, hipStream_t);
 void hipdf_i128_to_f64(const void*, void*, int64_t, hipStream_t);
 void hipdf_gb_percentile(const void*, const void*, const void*, const void*,
                          double, void*, int, hipStream_t);
@@ -801,6 +804,14 @@ PYBIND11_MODULE(hipdf, m) {
     hipdf_i128_cmp(op, P(a), P(b), P(av), P(bv), PM(out), PM(ov), n,
                    S(stream));
     check_async();
+  });
+  m.def("dec_mul_div_wide", [](int is_div, int64_t a, int64_t b, int64_t av,
+                               int64_t bv, int a128, int b128, int64_t out,
+                               int64_t ov, int out128, int shift,
+                               int out_prec, int64_t n, int64_t stream) {
+    hipdf_dec_mul_div_wide(is_div, P(a), P(b), P(av), P(bv), a128, b128,
+                           PM(out), PM(ov), out128, shift, out_prec, n,
+                           S(stream));
   });
   m.def("i128_rescale", [](int64_t in, int64_t iv, int64_t out, int64_t ov,
                            int shift, int out_prec, int out_is_64, int64_t n,

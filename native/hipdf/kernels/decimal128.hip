@@ -370,7 +370,170 @@ __global__ void k_i128_rescale(const int64_t* __restrict__ in,
   }
 }
 
+// ---- wide decimal multiply / divide (decimal128 operands) ----------------
+// mul: 256-bit product of the u128 magnitudes, then HALF_UP removal of
+// 10^shift one digit at a time (the last removed digit decides rounding).
+// div: schoolbook long division producing one decimal digit per step with
+// an overflow-safe u192 remainder (r*10 can exceed u128 when the divisor
+// is close to 10^38).
+
+struct u256 {
+  uint64_t w[4];  // little-endian limbs
+};
+
+__device__ __forceinline__ u256 mul_u128(u128 a, u128 b) {
+  uint64_t a0 = (uint64_t)a, a1 = (uint64_t)(a >> 64);
+  uint64_t b0 = (uint64_t)b, b1 = (uint64_t)(b >> 64);
+  u128 p00 = (u128)a0 * b0;
+  u128 p01 = (u128)a0 * b1;
+  u128 p10 = (u128)a1 * b0;
+  u128 p11 = (u128)a1 * b1;
+  u256 r;
+  r.w[0] = (uint64_t)p00;
+  u128 mid = (u128)(uint64_t)(p00 >> 64) + (uint64_t)p01 + (uint64_t)p10;
+  r.w[1] = (uint64_t)mid;
+  u128 hi = (u128)(uint64_t)(mid >> 64) + (uint64_t)(p01 >> 64) +
+            (uint64_t)(p10 >> 64) + (uint64_t)p11;
+  r.w[2] = (uint64_t)hi;
+  r.w[3] = (uint64_t)(hi >> 64) + (uint64_t)(p11 >> 64);
+  return r;
+}
+
+// v /= 10, returns the remainder digit
+__device__ __forceinline__ int div10_u256(u256* v) {
+  u128 rem = 0;
+  for (int i = 3; i >= 0; --i) {
+    u128 cur = (rem << 64) | v->w[i];
+    v->w[i] = (uint64_t)(cur / 10);
+    rem = cur % 10;
+  }
+  return (int)rem;
+}
+
+__global__ void k_dec_mul_div_wide(int is_div, const int64_t* __restrict__ a,
+                                   const int64_t* __restrict__ b,
+                                   const uint64_t* __restrict__ av,
+                                   const uint64_t* __restrict__ bv,
+                                   int a_is_128, int b_is_128,
+                                   int64_t* __restrict__ out,
+                                   uint64_t* __restrict__ ov, int out_is_128,
+                                   int shift, int out_prec, int64_t nstripe,
+                                   int64_t n) {
+  int64_t wave_global = ((int64_t)blockIdx.x * blockDim.x + threadIdx.x) / WAVE;
+  int64_t wave_count = ((int64_t)gridDim.x * blockDim.x) / WAVE;
+  int lane = lane_id();
+  for (int64_t s = wave_global; s < nstripe; s += wave_count) {
+    int64_t row = s * WAVE + lane;
+    bool ok = false;
+    if (row < n) {
+      ok = valid_bit(av, row) && valid_bit(bv, row);
+      u128 q = 0;
+      bool neg = false;
+      if (ok) {
+        u128 x, y;
+        bool xn, yn;
+        if (a_is_128) {
+          i128 v = load128(a, row);
+          xn = v.hi < 0;
+          i128 m = xn ? neg128(v) : v;
+          x = ((u128)(uint64_t)m.hi << 64) | m.lo;
+        } else {
+          int64_t v = a[row];
+          xn = v < 0;
+          x = (u128)(xn ? -(__int128)v : (__int128)v);
+        }
+        if (b_is_128) {
+          i128 v = load128(b, row);
+          yn = v.hi < 0;
+          i128 m = yn ? neg128(v) : v;
+          y = ((u128)(uint64_t)m.hi << 64) | m.lo;
+        } else {
+          int64_t v = b[row];
+          yn = v < 0;
+          y = (u128)(yn ? -(__int128)v : (__int128)v);
+        }
+        neg = xn != yn;
+        if (!is_div) {
+          u256 p = mul_u128(x, y);
+          int last = 0;
+          for (int k = 0; k < shift; ++k) last = div10_u256(&p);
+          if (p.w[2] | p.w[3]) {
+            ok = false;
+          } else {
+            q = ((u128)p.w[1] << 64) | p.w[0];
+            if (last >= 5) q += 1;
+          }
+        } else if (y == 0) {
+          ok = false;
+        } else {
+          int m = shift;
+          if (m < 0) {
+            // denominator gains the digits instead; y*10^-m must fit
+            for (int k = 0; k < -m && ok; ++k) {
+              if (y > ~(u128)0 / 10) ok = false;
+              else y *= 10;
+            }
+            m = 0;
+          }
+          if (ok) {
+            q = x / y;
+            u128 r = x % y;
+            const u128 q_lim = (~(u128)0 - 9) / 10;
+            for (int k = 0; k < m && ok; ++k) {
+              if (q > q_lim) { ok = false; break; }
+              // u192 rem10 = r * 10 (can exceed u128)
+              uint64_t rl = (uint64_t)r, rh = (uint64_t)(r >> 64);
+              u128 lo10 = (u128)rl * 10;
+              u128 hi10 = (u128)rh * 10 + (uint64_t)(lo10 >> 64);
+              u128 lo = ((u128)(uint64_t)hi10 << 64) | (uint64_t)lo10;
+              uint32_t hi = (uint32_t)(hi10 >> 64);
+              int digit = 0;
+              while (hi || lo >= y) {
+                if (lo < y) --hi;
+                lo -= y;
+                ++digit;
+              }
+              q = q * 10 + digit;
+              r = lo;
+            }
+            if (ok && 2 * r >= y) {
+              // HALF_UP on the true remainder of the last digit
+              if (q == ~(u128)0) ok = false;
+              else q += 1;
+            }
+          }
+        }
+        if (ok && out_prec <= 38 && q >= pow10_128(out_prec)) ok = false;
+        if (ok && !out_is_128 && q > (u128)0x7fffffffffffffffULL) ok = false;
+      }
+      __int128 w = ok ? (neg ? -(__int128)q : (__int128)q) : 0;
+      if (out_is_128) {
+        out[2 * row] = (int64_t)(u128)w;
+        out[2 * row + 1] = (int64_t)((u128)w >> 64);
+      } else {
+        out[row] = (int64_t)w;
+      }
+    }
+    uint64_t ballot = __ballot(ok);
+    write_valid_word(ov, s, ballot, lane);
+  }
+}
+
 extern "C" {
+
+void hipdf_dec_mul_div_wide(int is_div, const void* a, const void* b,
+                            const void* av, const void* bv, int a_is_128,
+                            int b_is_128, void* out, void* ov,
+                            int out_is_128, int shift, int out_prec,
+                            int64_t n, hipStream_t stream) {
+  int64_t nstripe = (n + WAVE - 1) / WAVE;
+  hipLaunchKernelGGL(k_dec_mul_div_wide, stripe_grid(nstripe),
+                     dim3(HIPDF_BLOCK), 0, stream, is_div,
+                     (const int64_t*)a, (const int64_t*)b,
+                     (const uint64_t*)av, (const uint64_t*)bv, a_is_128,
+                     b_is_128, (int64_t*)out, (uint64_t*)ov, out_is_128,
+                     shift, out_prec, nstripe, n);
+}
 
 void hipdf_i128_rescale(const void* in, const void* iv, void* out, void* ov,
                         int shift, int out_prec, int out_is_64, int64_t n,

@@ -126,13 +126,10 @@ _STR_CMP = {"eq": 0, "ne": 1, "lt": 2, "le": 3, "gt": 4, "ge": 5}
 
 def decimal_mul_div(op: str, lhs: Column, rhs: Column,
                     out_dtype: DType) -> Column:
-    """Exact decimal multiply/divide (dec64 operands) on device; see
-    k_dec64_mul_div in decimal128.hip. NULL on div-by-zero / overflow."""
-    if lhs.dtype.id is not TypeId.DECIMAL64 \
-            or rhs.dtype.id is not TypeId.DECIMAL64:
-        raise NotImplementedError(
-            "GPU decimal mul/div supports decimal64 operands (overrides "
-            "keep wider operands on CPU)")
+    """Exact decimal multiply/divide on device. dec64 operands ride
+    k_dec64_mul_div (__int128 intermediates); any decimal128 operand
+    routes to k_dec_mul_div_wide (256-bit product / u192-remainder long
+    division). NULL on div-by-zero / overflow."""
     n = lhs.size
     s = _stream()
     s1, s2, st = lhs.dtype.scale, rhs.dtype.scale, out_dtype.scale
@@ -142,12 +139,21 @@ def decimal_mul_div(op: str, lhs: Column, rhs: Column,
     out = torch.empty(max(width, 1), dtype=torch.int64,
                       device="cuda")[:width]
     ov = _alloc_mask(n)
-    if n:
+    a128 = lhs.dtype.id is TypeId.DECIMAL128
+    b128 = rhs.dtype.id is TypeId.DECIMAL128
+    if n and not (a128 or b128):
         ext.dec64_mul_div(1 if op == "div" else 0, lhs.data.data_ptr(),
                           rhs.data.data_ptr(), _ptr(lhs.validity),
                           _ptr(rhs.validity), out.data_ptr(), ov.data_ptr(),
                           1 if is128 else 0, shift, out_dtype.precision,
                           n, s)
+    elif n:
+        ext.dec_mul_div_wide(1 if op == "div" else 0, lhs.data.data_ptr(),
+                             rhs.data.data_ptr(), _ptr(lhs.validity),
+                             _ptr(rhs.validity), 1 if a128 else 0,
+                             1 if b128 else 0, out.data_ptr(),
+                             ov.data_ptr(), 1 if is128 else 0, shift,
+                             out_dtype.precision, n, s)
     return Column(out_dtype, n, out, ov, null_count=None)
 
 

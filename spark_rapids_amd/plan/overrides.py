@@ -105,11 +105,7 @@ class Tagger:
         elif isinstance(e, BinaryExpr):
             lt, rt = e.left.dtype(schema), e.right.dtype(schema)
             if e.op in ("mul", "div") and _decimal_exact(lt, rt):
-                # exact decimal mul/div: GPU kernel covers dec64 operands
-                if not (as_decimal(lt).id is TypeId.DECIMAL64
-                        and as_decimal(rt).id is TypeId.DECIMAL64):
-                    out.append(f"decimal {e.op} with >18-digit operands "
-                               "not on GPU yet")
+                pass  # exact kernels cover dec64 and dec128 operands
             else:
                 in_t = e._in_dtype(schema)
                 if e.op not in _GPU_BINARY_OPS:

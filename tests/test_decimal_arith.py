@@ -286,3 +286,64 @@ def test_gpu_rescale_roundtrip_and_overflow():
         ).to_pydict()
 
     assert q(sg) == q(sc)
+
+
+# ---- wide (decimal128-operand) multiply / divide -------------------------
+
+def _wide_ops(s):
+    df = s.create_dataframe({
+        "a": ["123456789012345678901234.5678", "-0.0001",
+              "99999999999999999999999999999999.99", "3.14", None],
+        "b": ["2.5", "4000.77", "2.0", "-0.000001", "9"],
+    })
+    df = df.select(col("a").cast(DType.decimal(36, 4)).alias("a"),
+                   col("b").cast(DType.decimal(10, 6)).alias("b"))
+    return df.select((col("a") * col("b")).alias("m"),
+                     (col("a") / col("b")).alias("d")).to_pydict()
+
+
+def test_cpu_wide_mul_div_exact(cpu):
+    out = _wide_ops(cpu)
+    # scales from Spark DecimalPrecision (allowPrecisionLoss)
+    assert out["m"][0] == Decimal("308641972530864197253086.419500")
+    assert out["d"][0] == Decimal("49382715604938271560493.827120")
+    assert out["d"][1] == Decimal("0.000000")  # rounds to zero at scale 6
+    assert out["d"][2] == Decimal("49999999999999999999999999999999.995000")
+    assert out["m"][4] is None and out["d"][4] is None
+
+
+@pytest.mark.gpu
+def test_gpu_wide_mul_div_matches_cpu():
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    g, c = _wide_ops(sg), _wide_ops(sc)
+    assert g == c, [(i, a, b) for i, (a, b) in
+                    enumerate(zip(g["m"], c["m"])) if a != b] + \
+        [(i, a, b) for i, (a, b) in enumerate(zip(g["d"], c["d"]))
+         if a != b]
+
+
+@pytest.mark.gpu
+def test_gpu_wide_mul_div_fuzz():
+    import numpy as np
+
+    rng = np.random.default_rng(17)
+    xs = [str(rng.integers(-10**17, 10**17)) + "." +
+          str(rng.integers(0, 10**6)).zfill(6) for _ in range(4000)]
+    ys = [str(rng.integers(-10**9, 10**9)) + "." +
+          str(rng.integers(0, 100)).zfill(2) for _ in range(4000)]
+
+    def q(s):
+        df = s.create_dataframe({"x": xs, "y": ys})
+        df = df.select(col("x").cast(DType.decimal(24, 6)).alias("x"),
+                       col("y").cast(DType.decimal(22, 2)).alias("y"))
+        return df.select((col("x") * col("y")).alias("m"),
+                         (col("x") / col("y")).alias("d")).to_pydict()
+
+    sg = sr.Session()
+    sc = sr.Session({"spark.rapids.sql.enabled": False})
+    g, c = q(sg), q(sc)
+    for k in ("m", "d"):
+        bad = [(i, a, b) for i, (a, b) in enumerate(zip(g[k], c[k]))
+               if a != b]
+        assert not bad, (k, bad[:5])
