@@ -214,6 +214,14 @@ static void check_async() {
 #define PM(x) reinterpret_cast<void*>(x)
 #define S(x) reinterpret_cast<hipStream_t>(x)
 
+extern "C" {
+void hipdf_str_pad(int, const void*, const void*, const void*, int32_t,
+                   int32_t, int32_t, const void*, void*, void*, int,
+                   int64_t, hipStream_t);
+void hipdf_str_locate(const void*, const void*, const void*, int32_t,
+                      int32_t, void*, int64_t, hipStream_t);
+}
+
 // ---- exact string<->decimal casts (kernels/cast_str.hip) -----------------
 extern "C" {
 void hipdf_str_to_dec(const void*, const void*, const void*, int, int, int,
@@ -273,6 +281,19 @@ static int64_t byte_array_offsets_walk(const uint8_t* p, int64_t nbytes,
 }
 
 PYBIND11_MODULE(hipdf, m) {
+  m.def("str_pad", [](int left, int64_t ao, int64_t ab, int64_t fill,
+                      int fill_nb, int fill_cps, int width, int64_t out_off,
+                      int64_t out_len, int64_t out, int mode, int64_t n,
+                      int64_t stream) {
+    hipdf_str_pad(left, P(ao), P(ab), P(fill), fill_nb, fill_cps, width,
+                  P(out_off), PM(out_len), PM(out), mode, n, S(stream));
+  });
+  m.def("str_locate", [](int64_t ao, int64_t ab, int64_t needle,
+                         int needle_nb, int pos, int64_t out, int64_t n,
+                         int64_t stream) {
+    hipdf_str_locate(P(ao), P(ab), P(needle), needle_nb, pos, PM(out), n,
+                     S(stream));
+  });
   m.def("str_to_dec", [](int64_t ao, int64_t ab, int64_t av, int out_kind,
                          int out_scale, int out_prec, int64_t out,
                          int64_t ov, int64_t n, int64_t stream) {

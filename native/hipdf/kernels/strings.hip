@@ -131,6 +131,103 @@ __global__ void k_str_like(const int32_t* __restrict__ ao,
     out[i] = (uint8_t)like_match(ab + ao[i], ao[i + 1] - ao[i], pat, plen);
 }
 
+// lpad/rpad to `width` codepoints, cycling the fill string (Spark
+// lpad('hi',5,'xy') = 'xyxhi'); longer inputs truncate to width
+// codepoints. mode 0: byte lengths; mode 1: write (GpuStringLPad/RPad).
+__device__ __forceinline__ int32_t cp_count(const uint8_t* s, int32_t nb) {
+  int32_t c = 0;
+  for (int32_t i = 0; i < nb; ++i) c += (s[i] & 0xC0) != 0x80;
+  return c;
+}
+
+// byte length of the first `cps` codepoints
+__device__ __forceinline__ int32_t cp_prefix_bytes(const uint8_t* s,
+                                                   int32_t nb,
+                                                   int32_t cps) {
+  int32_t seen = 0;
+  for (int32_t i = 0; i < nb; ++i) {
+    if ((s[i] & 0xC0) != 0x80) {
+      if (seen == cps) return i;
+      ++seen;
+    }
+  }
+  return nb;
+}
+
+__global__ void k_str_pad(int left, const int32_t* __restrict__ ao,
+                          const uint8_t* __restrict__ ab,
+                          const uint8_t* __restrict__ fill, int32_t fill_nb,
+                          int32_t fill_cps, int32_t width,
+                          const int64_t* __restrict__ out_off,
+                          int64_t* __restrict__ out_len,
+                          uint8_t* __restrict__ out, int mode, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* s = ab + ao[i];
+    int32_t nb = ao[i + 1] - ao[i];
+    int32_t cps = cp_count(s, nb);
+    int64_t len;
+    int32_t pad_cps = 0;
+    if (cps >= width || fill_nb == 0) {
+      len = cp_prefix_bytes(s, nb, width);
+    } else {
+      pad_cps = width - cps;
+      int32_t full = pad_cps / fill_cps;
+      int32_t part = pad_cps % fill_cps;
+      len = nb + (int64_t)full * fill_nb +
+            cp_prefix_bytes(fill, fill_nb, part);
+    }
+    if (mode == 0) {
+      out_len[i] = len;
+      continue;
+    }
+    uint8_t* w = out + out_off[i];
+    if (cps >= width || fill_nb == 0) {
+      for (int64_t k = 0; k < len; ++k) w[k] = s[k];
+      continue;
+    }
+    int32_t pad_bytes = (int32_t)(len - nb);
+    uint8_t* dst = left ? w : w + nb;
+    for (int32_t k = 0; k < pad_bytes; ++k) {
+      // cycle the fill string byte-wise (whole codepoints by
+      // construction of pad_bytes)
+      dst[k] = fill[k % fill_nb];
+    }
+    uint8_t* sdst = left ? w + pad_bytes : w;
+    for (int32_t k = 0; k < nb; ++k) sdst[k] = s[k];
+  }
+}
+
+// locate(needle, s, pos): 1-based CODEPOINT index of the first match at or
+// after codepoint `pos`; 0 when absent; empty needle -> pos (Spark).
+__global__ void k_str_locate(const int32_t* __restrict__ ao,
+                             const uint8_t* __restrict__ ab,
+                             const uint8_t* __restrict__ needle,
+                             int32_t needle_nb, int32_t pos,
+                             int32_t* __restrict__ out, int64_t n) {
+  for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    const uint8_t* s = ab + ao[i];
+    int32_t nb = ao[i + 1] - ao[i];
+    if (needle_nb == 0) {
+      out[i] = pos <= cp_count(s, nb) + 1 ? pos : 0;
+      continue;
+    }
+    int32_t cp = 0;
+    int32_t found = 0;
+    for (int32_t b = 0; b + needle_nb <= nb && !found; ++b) {
+      if ((s[b] & 0xC0) == 0x80) continue;
+      ++cp;  // s[b] starts codepoint number `cp` (1-based)
+      if (cp < pos) continue;
+      bool eq = true;
+      for (int32_t k = 0; k < needle_nb; ++k)
+        if (s[b + k] != needle[k]) { eq = false; break; }
+      if (eq) found = cp;
+    }
+    out[i] = found;
+  }
+}
+
 // codepoint length
 __global__ void k_str_length(const int32_t* __restrict__ ao,
                              const uint8_t* __restrict__ ab,
@@ -464,6 +561,26 @@ void hipdf_str_like(const void* ao, const void* ab, const void* pat, int plen,
   hipLaunchKernelGGL(k_str_like, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
                      (const int32_t*)ao, (const uint8_t*)ab,
                      (const uint8_t*)pat, (int32_t)plen, (uint8_t*)out, n);
+}
+
+void hipdf_str_pad(int left, const void* ao, const void* ab,
+                   const void* fill, int32_t fill_nb, int32_t fill_cps,
+                   int32_t width, const void* out_off, void* out_len,
+                   void* out, int mode, int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_pad, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     left, (const int32_t*)ao, (const uint8_t*)ab,
+                     (const uint8_t*)fill, fill_nb, fill_cps, width,
+                     (const int64_t*)out_off, (int64_t*)out_len,
+                     (uint8_t*)out, mode, n);
+}
+
+void hipdf_str_locate(const void* ao, const void* ab, const void* needle,
+                      int32_t needle_nb, int32_t pos, void* out, int64_t n,
+                      hipStream_t stream) {
+  hipLaunchKernelGGL(k_str_locate, flat_grid(n), dim3(HIPDF_BLOCK), 0,
+                     stream, (const int32_t*)ao, (const uint8_t*)ab,
+                     (const uint8_t*)needle, needle_nb, pos, (int32_t*)out,
+                     n);
 }
 
 void hipdf_str_length(const void* ao, const void* ab, void* out, int64_t n,

@@ -743,7 +743,9 @@ class PadExpr(Expression):
     def __init__(self, child: Expression, width: int, fill: str, left: bool):
         self.child = child
         self.width = width
-        self.fill = fill or " "
+        # empty fill is meaningful (Spark UTF8String.lpad: empty pad just
+        # truncates to width); only None defaults to a space
+        self.fill = " " if fill is None else fill
         self.left = left
 
     @property
@@ -755,22 +757,7 @@ class PadExpr(Expression):
 
     def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
         c = self.child.eval(batch, schema)
-        from .. import ops as _ops
-
-        host = c if not c.is_cuda else c.cpu()
-        vals = host.to_pylist()
-        out = []
-        for v in vals:
-            if v is None:
-                out.append(None)
-            elif self.left:
-                out.append(v.rjust(self.width, self.fill[0])[: self.width]
-                           if len(v) < self.width else v[: self.width])
-            else:
-                out.append(v.ljust(self.width, self.fill[0])[: self.width]
-                           if len(v) < self.width else v[: self.width])
-        res = Column.from_pylist(out, STRING)
-        return res.cuda() if c.is_cuda else res
+        return ops.str_pad(c, self.width, self.fill, self.left)
 
     def __str__(self):
         side = "lpad" if self.left else "rpad"
@@ -794,15 +781,7 @@ class LocateExpr(Expression):
 
     def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
         c = self.child.eval(batch, schema)
-        host = c if not c.is_cuda else c.cpu()
-        out = []
-        for v in host.to_pylist():
-            if v is None:
-                out.append(None)
-            else:
-                out.append(v.find(self.substr, max(self.pos - 1, 0)) + 1)
-        res = Column.from_pylist(out, INT32)
-        return res.cuda() if c.is_cuda else res
+        return ops.str_locate(c, self.substr, self.pos)
 
     def __str__(self):
         return f"locate({self.substr!r}, {self.child}, {self.pos})"

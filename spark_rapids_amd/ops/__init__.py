@@ -153,3 +153,11 @@ def join_gather_maps(left: ColumnBatch, right: ColumnBatch,
 def concat_batches(batches: Sequence[ColumnBatch]) -> ColumnBatch:
     assert batches
     return backend_for(*batches[0].columns).concat_batches(list(batches))
+
+
+def str_pad(col: Column, width: int, fill: str, left: bool) -> Column:
+    return backend_for(col).str_pad(col, width, fill, left)
+
+
+def str_locate(col: Column, substr: str, pos: int = 1) -> Column:
+    return backend_for(col).str_locate(col, substr, pos)

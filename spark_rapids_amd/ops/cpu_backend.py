@@ -1284,3 +1284,33 @@ def sort_order(batch: ColumnBatch, key_idx: List[int],
         keys.append(null_rank)
     order = np.lexsort(tuple(keys)) if keys else np.arange(n)
     return _make(order.astype(np.int32), None, DType.int32())
+
+
+def str_pad(col: Column, width: int, fill: str, left: bool) -> Column:
+    """lpad/rpad with Spark semantics: cycle the fill string, truncate to
+    width codepoints; empty fill with short input -> truncate."""
+    a, av = _vals(col), _valid(col)
+    out = []
+    for v, ok in zip(a, av):
+        if not ok:
+            out.append(None)
+            continue
+        if len(v) >= width or not fill:
+            out.append(v[:width])
+        else:
+            pad = (fill * ((width - len(v)) // len(fill) + 1))[: width - len(v)]
+            out.append(pad + v if left else v + pad)
+    return Column.from_pylist(out, DType.string())
+
+
+def str_locate(col: Column, substr: str, pos: int = 1) -> Column:
+    a, av = _vals(col), _valid(col)
+    out = []
+    for v, ok in zip(a, av):
+        if not ok:
+            out.append(None)
+        elif substr == "":
+            out.append(pos if pos <= len(v) + 1 else 0)
+        else:
+            out.append(v.find(substr, max(pos - 1, 0)) + 1)
+    return Column.from_pylist(out, DType.int32())

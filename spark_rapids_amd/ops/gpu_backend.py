@@ -1796,3 +1796,46 @@ def sort_order(batch: ColumnBatch, key_idx: List[int], descending: List[bool],
     if perm is None:
         perm = torch.arange(n, dtype=torch.int32, device="cuda")
     return Column(DType.int32(), n, perm.clone(), None, null_count=0)
+
+
+def str_pad(col: Column, width: int, fill: str, left: bool) -> Column:
+    n = col.size
+    s = _stream()
+    v = col.validity.clone() if col.validity is not None else None
+    fb = fill.encode("utf-8")
+    fill_t = _pattern_tensor(fill) if fb else torch.zeros(
+        1, dtype=torch.uint8, device="cuda")
+    lens = torch.empty(max(n, 1), dtype=torch.int64, device="cuda")[:n]
+    if n:
+        ext.str_pad(1 if left else 0, col.offsets.data_ptr(),
+                    col.data.data_ptr(), fill_t.data_ptr(), len(fb),
+                    len(fill), width, 0, lens.data_ptr(), 0, 0, n, s)
+    scanned, total = _exclusive_scan_i64(lens) if n else (lens, 0)
+    out = torch.empty(max(total, 1), dtype=torch.uint8,
+                      device="cuda")[:total]
+    if total:
+        ext.str_pad(1 if left else 0, col.offsets.data_ptr(),
+                    col.data.data_ptr(), fill_t.data_ptr(), len(fb),
+                    len(fill), width, scanned.data_ptr(), lens.data_ptr(),
+                    out.data_ptr(), 1, n, s)
+    offs = torch.empty(n + 1, dtype=torch.int32, device="cuda")
+    if n:
+        ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n, s)
+    offs[n] = total
+    return Column(DType.string(), n, out, v, offs,
+                  null_count=col._null_count)
+
+
+def str_locate(col: Column, substr: str, pos: int = 1) -> Column:
+    n = col.size
+    s = _stream()
+    v = col.validity.clone() if col.validity is not None else None
+    nb = substr.encode("utf-8")
+    needle = _pattern_tensor(substr) if nb else torch.zeros(
+        1, dtype=torch.uint8, device="cuda")
+    out = _alloc(n, DType.int32())
+    if n:
+        ext.str_locate(col.offsets.data_ptr(), col.data.data_ptr(),
+                       needle.data_ptr(), len(nb), pos, out.data_ptr(),
+                       n, s)
+    return Column(DType.int32(), n, out, v, null_count=col._null_count)

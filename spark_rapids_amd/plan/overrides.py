@@ -152,7 +152,9 @@ class Tagger:
                 out.append("regex split delimiters run on CPU")
         elif type(e).__name__ in ("ArraySize", "ElementAt"):
             pass
-        elif type(e).__name__ in ("PadExpr", "LocateExpr", "HostStringFn"):
+        elif type(e).__name__ in ("PadExpr", "LocateExpr"):
+            pass  # lpad/rpad/locate/instr device kernels (k_str_pad/locate)
+        elif type(e).__name__ == "HostStringFn":
             out.append(f"{getattr(e, 'name', type(e).__name__)} runs on "
                        "CPU this round")
         elif type(e).__name__ in ("RegexpExtract", "RegexpReplace",

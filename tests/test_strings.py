@@ -409,3 +409,66 @@ def test_host_string_fns():
     s.register("thsf", df)
     assert s.sql("SELECT substring_index(s, '.', -1) FROM thsf").collect() \
         == [("c",), ("xyz",), (None,)]
+
+
+# ---- lpad/rpad/locate on device (round 2) --------------------------------
+
+def _pad_cases():
+    return ["hi", "hello world", "", "ab", None, "über", "xyzt"]
+
+
+def test_cpu_pad_cycles_fill():
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col
+    from spark_rapids_amd.expr.expressions import PadExpr
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({"x": _pad_cases()})
+    out = df.select(
+        PadExpr(col("x"), 5, "xy", True).alias("l"),
+        PadExpr(col("x"), 5, "xy", False).alias("r"),
+        PadExpr(col("x"), 3, "", True).alias("e")).to_pydict()
+    # Spark lpad('hi',5,'xy') = 'xyxhi' (fill cycles)
+    assert out["l"][0] == "xyxhi"
+    assert out["r"][0] == "hixyx"
+    assert out["l"][1] == "hello"          # truncate to width
+    assert out["l"][4] is None
+    assert out["e"][0] == "hi"             # empty fill: truncate only
+
+
+def test_cpu_locate_codepoints():
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col
+    from spark_rapids_amd.expr.expressions import LocateExpr
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({"x": ["hello", "ababab", None, "aüba"]})
+    out = df.select(LocateExpr(col("x"), "ab", 2).alias("p"),
+                    LocateExpr(col("x"), "b", 1).alias("q"),
+                    LocateExpr(col("x"), "", 3).alias("e")).to_pydict()
+    assert out["p"] == [0, 3, None, 0]
+    assert out["q"] == [0, 2, None, 3]     # codepoint index, not byte
+    assert out["e"] == [3, 3, None, 3]
+
+
+@pytest.mark.gpu
+def test_gpu_pad_locate_match_cpu():
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col
+    from spark_rapids_amd.expr.expressions import LocateExpr, PadExpr
+
+    cases = _pad_cases() * 500
+
+    def q(s):
+        df = s.create_dataframe({"x": cases})
+        return df.select(
+            PadExpr(col("x"), 7, "xy", True).alias("l"),
+            PadExpr(col("x"), 7, "·", False).alias("r"),
+            PadExpr(col("x"), 2, " ", True).alias("t"),
+            LocateExpr(col("x"), "b", 1).alias("p1"),
+            LocateExpr(col("x"), "l", 3).alias("p3"),
+        ).to_pydict()
+
+    g = q(sr.Session())
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert g == c
