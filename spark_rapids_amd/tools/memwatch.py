@@ -13,6 +13,14 @@ import torch
 def allocated_bytes() -> int:
     if not torch.cuda.is_available():
         return 0
+    from ..memory import device_pool
+
+    if device_pool.is_active():
+        # torch's stats APIs don't cover a pluggable allocator; the hipdf
+        # pool tracks its own usage
+        import hipdf
+
+        return int(hipdf.pool_used())
     return int(torch.cuda.memory_allocated())
 
 
