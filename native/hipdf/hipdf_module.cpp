@@ -239,6 +239,7 @@ void hipdf_pool_set_failure_cb(hipdf_failure_cb);
 size_t hipdf_pool_used();
 size_t hipdf_pool_reserved();
 size_t hipdf_pool_high_watermark();
+size_t hipdf_pool_overflow();
 int hipdf_pool_selftest();
 }
 
@@ -251,6 +252,10 @@ static int spill_cb_trampoline(size_t needed, int retry) {
   py::gil_scoped_acquire gil;
   try {
     return (*g_spill_cb)((size_t)needed, retry).cast<int>();
+  } catch (const std::exception& e) {
+    fprintf(stderr, "[hipdf pool] spill callback raised: %s\n", e.what());
+    PyErr_Clear();
+    return 0;
   } catch (...) {
     PyErr_Clear();
     return 0;
@@ -313,6 +318,7 @@ PYBIND11_MODULE(hipdf, m) {
   m.def("pool_used", []() { return hipdf_pool_used(); });
   m.def("pool_reserved", []() { return hipdf_pool_reserved(); });
   m.def("pool_high_watermark", []() { return hipdf_pool_high_watermark(); });
+  m.def("pool_overflow", []() { return hipdf_pool_overflow(); });
   m.def("pool_selftest", []() { return hipdf_pool_selftest(); });
   m.def("pool_set_spill_cb", [](py::object f) {
     if (!g_spill_cb) g_spill_cb = new py::object();

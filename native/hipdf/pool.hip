@@ -142,6 +142,11 @@ size_t hipdf_pool_high_watermark() {
   return g_device.high_watermark;
 }
 
+size_t hipdf_pool_overflow() {
+  std::lock_guard<std::mutex> lk(g_mu);
+  return g_overflow_bytes;
+}
+
 void* hipdf_pool_alloc(size_t n) {
   {
     std::lock_guard<std::mutex> lk(g_mu);
