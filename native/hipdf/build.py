@@ -35,6 +35,7 @@ KERNELS = [
     "kernels/decimal128.hip",
     "kernels/regex.hip",
     "kernels/cast_str.hip",
+    "kernels/datetime.hip",
     "pool.hip",
 ]
 

@@ -222,6 +222,15 @@ void hipdf_str_locate(const void*, const void*, const void*, int32_t,
                       int32_t, void*, int64_t, hipStream_t);
 }
 
+extern "C" {
+void hipdf_tz_convert(const void*, const void*, const void*, int, int,
+                      void*, int64_t, hipStream_t);
+void hipdf_date_format(const void*, const void*, int, int, void*, int64_t,
+                       hipStream_t);
+void hipdf_ts_parse(const void*, const void*, const void*, const void*, int,
+                    int, void*, void*, int64_t, hipStream_t);
+}
+
 // ---- exact string<->decimal casts (kernels/cast_str.hip) -----------------
 extern "C" {
 void hipdf_str_to_dec(const void*, const void*, const void*, int, int, int,
@@ -286,6 +295,22 @@ static int64_t byte_array_offsets_walk(const uint8_t* p, int64_t nbytes,
 }
 
 PYBIND11_MODULE(hipdf, m) {
+  m.def("tz_convert", [](int64_t ts, int64_t trans, int64_t offs,
+                         int n_trans, int to_utc, int64_t out, int64_t n,
+                         int64_t stream) {
+    hipdf_tz_convert(P(ts), P(trans), P(offs), n_trans, to_utc, PM(out), n,
+                     S(stream));
+  });
+  m.def("date_format", [](int64_t ts, int64_t tokens, int ntok, int width,
+                          int64_t out, int64_t n, int64_t stream) {
+    hipdf_date_format(P(ts), P(tokens), ntok, width, PM(out), n, S(stream));
+  });
+  m.def("ts_parse", [](int64_t ao, int64_t ab, int64_t av, int64_t tokens,
+                       int ntok, int width, int64_t out, int64_t ov,
+                       int64_t n, int64_t stream) {
+    hipdf_ts_parse(P(ao), P(ab), P(av), P(tokens), ntok, width, PM(out),
+                   PM(ov), n, S(stream));
+  });
   m.def("str_pad", [](int left, int64_t ao, int64_t ab, int64_t fill,
                       int fill_nb, int fill_cps, int width, int64_t out_off,
                       int64_t out_len, int64_t out, int mode, int64_t n,

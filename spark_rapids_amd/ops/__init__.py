@@ -161,3 +161,15 @@ def str_pad(col: Column, width: int, fill: str, left: bool) -> Column:
 
 def str_locate(col: Column, substr: str, pos: int = 1) -> Column:
     return backend_for(col).str_locate(col, substr, pos)
+
+
+def tz_convert(col: Column, zone: str, to_utc: bool) -> Column:
+    return backend_for(col).tz_convert(col, zone, to_utc)
+
+
+def date_format(col: Column, tokens, width: int) -> Column:
+    return backend_for(col).date_format(col, tokens, width)
+
+
+def ts_parse(col: Column, tokens, width: int) -> Column:
+    return backend_for(col).ts_parse(col, tokens, width)

@@ -1314,3 +1314,115 @@ def str_locate(col: Column, substr: str, pos: int = 1) -> Column:
         else:
             out.append(v.find(substr, max(pos - 1, 0)) + 1)
     return Column.from_pylist(out, DType.int32())
+
+
+def tz_convert(col: Column, zone: str, to_utc: bool) -> Column:
+    from ..tools import tzdb
+
+    us = _vals(col).astype(np.int64)
+    av = _valid(col)
+    sec = np.floor_divide(us, 1_000_000)
+    if not to_utc:
+        out = us + tzdb.offset_at(zone, sec).astype(np.int64) * 1_000_000
+    else:
+        o0 = tzdb.offset_at(zone, sec).astype(np.int64)
+        o1 = tzdb.offset_at(zone, sec - o0).astype(np.int64)
+        out = us - o1 * 1_000_000
+    return _make(out, av if not av.all() else None, DType.timestamp())
+
+
+def _civil_from_days_np(days):
+    z = days.astype(np.int64) + 719468
+    era = np.floor_divide(z, 146097)
+    doe = z - era * 146097
+    yoe = (doe - doe // 1460 + doe // 36524 - doe // 146096) // 365
+    y = yoe + era * 400
+    doy = doe - (365 * yoe + yoe // 4 - yoe // 100)
+    mp = (5 * doy + 2) // 153
+    d = doy - (153 * mp + 2) // 5 + 1
+    m = np.where(mp < 10, mp + 3, mp - 9)
+    return y + (m <= 2), m, d
+
+
+def date_format(col: Column, tokens, width: int) -> Column:
+    us = _vals(col).astype(np.int64)
+    av = _valid(col)
+    sec = np.floor_divide(us, 1_000_000)
+    days = np.floor_divide(sec, 86_400)
+    tod = sec - days * 86_400
+    y, m, d = _civil_from_days_np(days)
+    hh, mi, ss = tod // 3600, tod // 60 % 60, tod % 60
+    out = []
+    from ..expr.datetime import (DT_DD, DT_HH, DT_LIT, DT_MI, DT_MM,
+                                 DT_SS, DT_YYYY)
+
+    for i in range(len(us)):
+        if not av[i]:
+            out.append(None)
+            continue
+        parts = []
+        for kind, arg in tokens:
+            if kind == DT_LIT:
+                parts.append(chr(arg))
+            elif kind == DT_YYYY:
+                parts.append("%04d" % max(y[i], 0))
+            elif kind == DT_MM:
+                parts.append("%02d" % m[i])
+            elif kind == DT_DD:
+                parts.append("%02d" % d[i])
+            elif kind == DT_HH:
+                parts.append("%02d" % hh[i])
+            elif kind == DT_MI:
+                parts.append("%02d" % mi[i])
+            elif kind == DT_SS:
+                parts.append("%02d" % ss[i])
+        out.append("".join(parts))
+    return Column.from_pylist(out, DType.string())
+
+
+def ts_parse(col: Column, tokens, width: int) -> Column:
+    from datetime import datetime
+
+    from ..expr.datetime import (DT_DD, DT_HH, DT_LIT, DT_MI, DT_MM,
+                                 DT_SS, DT_YYYY)
+
+    a, av = _vals(col), _valid(col)
+    out = np.zeros(len(a), dtype=np.int64)
+    valid = av.copy()
+    for i, s in enumerate(a):
+        if not av[i]:
+            continue
+        if s is None or len(s) != width:
+            valid[i] = False
+            continue
+        p = 0
+        f = {"y": 1970, "M": 1, "d": 1, "H": 0, "m": 0, "s": 0}
+        ok = True
+        for kind, arg in tokens:
+            if kind == DT_LIT:
+                if s[p] != chr(arg):
+                    ok = False
+                    break
+                p += 1
+            else:
+                w = 4 if kind == DT_YYYY else 2
+                seg = s[p:p + w]
+                if not seg.isdigit():
+                    ok = False
+                    break
+                v = int(seg)
+                key = {DT_YYYY: "y", DT_MM: "M", DT_DD: "d", DT_HH: "H",
+                       DT_MI: "m", DT_SS: "s"}[kind]
+                f[key] = v
+                p += w
+        if ok:
+            try:
+                dt = datetime(f["y"], f["M"], f["d"], f["H"], f["m"],
+                              f["s"])
+                epoch = datetime(1970, 1, 1)
+                out[i] = int((dt - epoch).total_seconds()) * 1_000_000
+            except ValueError:
+                ok = False
+        valid[i] = ok
+    return _make(out, valid if not valid.all() else None,
+                 DType.timestamp())

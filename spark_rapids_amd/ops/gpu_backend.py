@@ -1839,3 +1839,68 @@ def str_locate(col: Column, substr: str, pos: int = 1) -> Column:
                        needle.data_ptr(), len(nb), pos, out.data_ptr(),
                        n, s)
     return Column(DType.int32(), n, out, v, null_count=col._null_count)
+
+
+_TZ_DEV_CACHE: dict = {}
+
+
+def _tz_device(zone: str):
+    hit = _TZ_DEV_CACHE.get(zone)
+    if hit is None:
+        from ..tools import tzdb
+
+        trans, offs = tzdb.load(zone)
+        hit = (torch.from_numpy(trans.copy()).cuda(),
+               torch.from_numpy(offs.copy()).cuda())
+        _TZ_DEV_CACHE[zone] = hit
+    return hit
+
+
+def tz_convert(col: Column, zone: str, to_utc: bool) -> Column:
+    n = col.size
+    s = _stream()
+    v = col.validity.clone() if col.validity is not None else None
+    trans, offs = _tz_device(zone)
+    out = _alloc(n, DType.timestamp())
+    if n:
+        ext.tz_convert(col.data.data_ptr(), trans.data_ptr(),
+                       offs.data_ptr(), trans.numel(), 1 if to_utc else 0,
+                       out.data_ptr(), n, s)
+    return Column(DType.timestamp(), n, out, v,
+                  null_count=col._null_count)
+
+
+def _token_tensor(tokens):
+    flat = []
+    for kind, arg in tokens:
+        flat.extend([kind, arg])
+    return torch.tensor(flat, dtype=torch.int32).cuda()
+
+
+def date_format(col: Column, tokens, width: int) -> Column:
+    n = col.size
+    s = _stream()
+    v = col.validity.clone() if col.validity is not None else None
+    tok = _token_tensor(tokens)
+    out = torch.empty(max(n * width, 1), dtype=torch.uint8,
+                      device="cuda")[:n * width]
+    if n:
+        ext.date_format(col.data.data_ptr(), tok.data_ptr(), len(tokens),
+                        width, out.data_ptr(), n, s)
+    offs = torch.arange(0, (n + 1) * width, width, dtype=torch.int32,
+                        device="cuda")
+    return Column(DType.string(), n, out, v, offs,
+                  null_count=col._null_count)
+
+
+def ts_parse(col: Column, tokens, width: int) -> Column:
+    n = col.size
+    s = _stream()
+    tok = _token_tensor(tokens)
+    out = _alloc(n, DType.timestamp())
+    ov = _alloc_mask(n)
+    if n:
+        ext.ts_parse(col.offsets.data_ptr(), col.data.data_ptr(),
+                     _ptr(col.validity), tok.data_ptr(), len(tokens),
+                     width, out.data_ptr(), ov.data_ptr(), n, s)
+    return Column(DType.timestamp(), n, out, ov, null_count=None)
