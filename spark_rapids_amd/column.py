@@ -118,12 +118,15 @@ class Column:
         self.validity = validity
         self.offsets = offsets
         self._null_count = null_count
-        self.child = child  # LIST element column
+        self.child = child  # LIST element column / tuple for STRUCT
         self._minmax = None  # cached (min, max) for dense-key group-by
         if dtype.id is TypeId.STRING:
             assert offsets is not None and offsets.numel() == size + 1
         if dtype.id is TypeId.LIST:
             assert offsets is not None and child is not None
+        if dtype.id is TypeId.STRUCT:
+            assert isinstance(child, tuple) and \
+                len(child) == len(dtype.children)
 
     # ---- properties ---------------------------------------------------
     @property
@@ -155,7 +158,9 @@ class Column:
             n += self.validity.numel()
         if self.offsets is not None:
             n += self.offsets.numel() * 4
-        if self.child is not None:
+        if isinstance(self.child, tuple):
+            n += sum(c.nbytes for c in self.child)
+        elif self.child is not None:
             n += self.child.nbytes
         return n
 
@@ -163,6 +168,12 @@ class Column:
     def to(self, device: str, non_blocking: bool = False) -> "Column":
         if device == self.device:
             return self
+        child = self.child
+        if isinstance(child, tuple):
+            child = tuple(c.to(device, non_blocking=non_blocking)
+                          for c in child)
+        elif child is not None:
+            child = child.to(device, non_blocking=non_blocking)
         return Column(
             self.dtype,
             self.size,
@@ -170,7 +181,7 @@ class Column:
             None if self.validity is None else self.validity.to(device, non_blocking=non_blocking),
             None if self.offsets is None else self.offsets.to(device, non_blocking=non_blocking),
             self._null_count,
-            None if self.child is None else self.child.to(device, non_blocking=non_blocking),
+            child,
         )
 
     def cuda(self) -> "Column":
@@ -203,6 +214,19 @@ class Column:
     @staticmethod
     def from_pylist(values: Sequence, dtype: DType, device: str = "cpu") -> "Column":
         n = len(values)
+        if dtype.id is TypeId.STRUCT:
+            valid = np.array([v is not None for v in values], dtype=bool)
+            kids = []
+            for name, cdt in zip(dtype.field_names, dtype.children):
+                kid_vals = [None if v is None else
+                            (v.get(name) if isinstance(v, dict)
+                             else v[dtype.field_names.index(name)])
+                            for v in values]
+                kids.append(Column.from_pylist(kid_vals, cdt))
+            col = Column(dtype, n, torch.zeros(0, dtype=torch.uint8),
+                         make_validity(valid) if not valid.all() else None,
+                         None, None, tuple(kids))
+            return col.to(device) if device != "cpu" else col
         if dtype.id is TypeId.LIST:
             valid = np.array([v is not None for v in values], dtype=bool)
             flat: list = []
@@ -301,6 +325,12 @@ class Column:
 
     def to_pylist(self) -> list:
         valid = self.valid_array()
+        if self.dtype.id is TypeId.STRUCT:
+            kid_lists = [c.to_pylist() for c in self.child]
+            names = self.dtype.field_names
+            return [dict(zip(names, row)) if valid[i] else None
+                    for i, row in enumerate(zip(*kid_lists))] if kid_lists \
+                else [{} if v else None for v in valid]
         if self.dtype.id is TypeId.LIST:
             offs = self.offsets.cpu().numpy()
             elems = self.child.to_pylist()

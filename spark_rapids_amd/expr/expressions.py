@@ -1137,3 +1137,90 @@ def datediff(end, start) -> Expression:
 
     return BinaryExpr("sub", CastExpr(_as_expr(end), DATE32),
                       CastExpr(_as_expr(start), DATE32))
+
+
+class CreateNamedStruct(Expression):
+    """named_struct(n1, e1, n2, e2, ...) -> STRUCT column (reference
+    analogue: GpuCreateNamedStruct). The struct itself is never null."""
+
+    def __init__(self, names, exprs):
+        self.names = list(names)
+        self.exprs = [_as_expr(e) for e in exprs]
+
+    @property
+    def children(self):
+        return tuple(self.exprs)
+
+    def dtype(self, schema: Schema) -> DType:
+        return DType.struct_(
+            [(n, e.dtype(schema)) for n, e in zip(self.names, self.exprs)])
+
+    def nullable(self, schema: Schema) -> bool:
+        return False
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        import torch
+
+        kids = tuple(e.eval(batch, schema) for e in self.exprs)
+        dev = kids[0].device if kids else batch.device
+        return Column(self.dtype(schema), batch.num_rows,
+                      torch.zeros(0, dtype=torch.uint8, device=dev),
+                      None, None, 0, kids)
+
+    def output_name(self) -> str:
+        return "named_struct(" + ", ".join(self.names) + ")"
+
+    def __str__(self):
+        inner = ", ".join(f"{n}: {e}" for n, e in
+                          zip(self.names, self.exprs))
+        return f"named_struct({inner})"
+
+
+class GetStructField(Expression):
+    """struct.field access (reference analogue: GpuGetStructField)."""
+
+    def __init__(self, child, name: str):
+        self.child = _as_expr(child)
+        self.name = name
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def _field_index(self, schema) -> int:
+        st = self.child.dtype(schema)
+        if st.id is not TypeId.STRUCT:
+            raise TypeError(f"getField on non-struct {st}")
+        if self.name not in st.field_names:
+            raise KeyError(
+                f"struct has no field {self.name!r} (has "
+                f"{list(st.field_names)})")
+        return st.field_names.index(self.name)
+
+    def dtype(self, schema: Schema) -> DType:
+        st = self.child.dtype(schema)
+        return st.children[self._field_index(schema)]
+
+    def nullable(self, schema: Schema) -> bool:
+        return True
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        kid = c.child[self._field_index(schema)]
+        if c.validity is None:
+            return kid
+        return ops.backend_for(c).and_parent_validity(kid, c)
+
+    def output_name(self) -> str:
+        return f"{self.child}.{self.name}"
+
+    def __str__(self):
+        return self.output_name()
+
+
+def named_struct(**fields) -> CreateNamedStruct:
+    return CreateNamedStruct(list(fields), list(fields.values()))
+
+
+def get_field(struct_expr, name: str) -> GetStructField:
+    return GetStructField(struct_expr, name)

@@ -813,7 +813,7 @@ def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> C
 def _gather_idx(batch: ColumnBatch, idx: np.ndarray, row_ok: np.ndarray) -> ColumnBatch:
     cols = []
     for c in batch.columns:
-        if c.dtype.id is TypeId.LIST:
+        if c.dtype.id in (TypeId.LIST, TypeId.STRUCT):
             vals_py = c.to_pylist()
             out = [vals_py[i] if ok and 0 <= i < len(vals_py) else None
                    for i, ok in zip(idx, row_ok)]
@@ -836,7 +836,7 @@ def concat_batches(batches: List[ColumnBatch]) -> ColumnBatch:
     cols = []
     for i in range(ncols):
         dtype = batches[0].columns[i].dtype
-        if dtype.id is TypeId.LIST:
+        if dtype.id in (TypeId.LIST, TypeId.STRUCT):
             vals = []
             for b in batches:
                 vals.extend(b.columns[i].to_pylist())
@@ -1426,3 +1426,15 @@ def ts_parse(col: Column, tokens, width: int) -> Column:
         valid[i] = ok
     return _make(out, valid if not valid.all() else None,
                  DType.timestamp())
+
+
+def and_parent_validity(kid: Column, parent: Column) -> Column:
+    """Null out child rows where the parent (struct) row is null."""
+    pv = parent.valid_array()
+    kv = kid.valid_array()
+    both = pv & kv
+    if both.all():
+        return kid
+    vals = kid.to_pylist()
+    out = [v if ok else None for v, ok in zip(vals, both)]
+    return Column.from_pylist(out, kid.dtype)

@@ -102,6 +102,11 @@ def _empty_col(dtype: DType) -> Column:
     if dtype.id is TypeId.STRING:
         return Column(dtype, 0, torch.zeros(0, dtype=torch.uint8, device="cuda"),
                       None, torch.zeros(1, dtype=torch.int32, device="cuda"), 0)
+    if dtype.id is TypeId.STRUCT:
+        return Column(dtype, 0,
+                      torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                      None, None, 0,
+                      tuple(_empty_col(c) for c in dtype.children))
     return Column(dtype, 0, torch.zeros(0, dtype=torch_dtype(dtype),
                                         device="cuda"), None, None, 0)
 
@@ -1027,6 +1032,17 @@ def _gather_col(c: Column, idx: torch.Tensor, n_out: int,
     s = _stream()
     if n_out == 0:
         return _empty_col(c.dtype)
+    if c.dtype.id is TypeId.STRUCT:
+        kids = tuple(_gather_col(k, idx, n_out, maybe_negative)
+                     for k in c.child)
+        ov = None
+        if c.validity is not None or maybe_negative:
+            ov = _alloc_mask(n_out)
+            ext.gather_validity(_ptr(c.validity), c.validity is not None,
+                                idx.data_ptr(), ov.data_ptr(), n_out, s)
+        return Column(c.dtype, n_out,
+                      torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                      ov, None, None if ov is not None else 0, kids)
     if c.dtype.id is TypeId.STRING:
         lens = torch.empty(n_out, dtype=torch.int64, device="cuda")
         ext.gather_str_lens(c.offsets.data_ptr(), idx.data_ptr(),
@@ -1103,6 +1119,29 @@ def concat_batches(batches: List[ColumnBatch]) -> ColumnBatch:
                 "concat of LIST columns on GPU (explode or collect before "
                 "unioning, or keep the union on the CPU)")
         dtype = ins[0].dtype
+        if dtype.id is TypeId.STRUCT:
+            any_valid = any(c.validity is not None for c in ins)
+            out_valid = None
+            if any_valid:
+                out_valid = torch.zeros(mask_nbytes(total),
+                                        dtype=torch.uint8, device="cuda")
+                off = 0
+                for c in ins:
+                    if c.size:
+                        ext.copy_valid_range(_ptr(c.validity),
+                                             c.validity is not None, off,
+                                             out_valid.data_ptr(), c.size,
+                                             s)
+                    off += c.size
+            kids = tuple(
+                concat_batches([ColumnBatch([c.child[k]], c.size)
+                                for c in ins]).columns[0]
+                for k in range(len(dtype.children)))
+            out_cols.append(Column(
+                dtype, total, torch.zeros(0, dtype=torch.uint8,
+                                          device="cuda"),
+                out_valid, None, None if any_valid else 0, kids))
+            continue
         any_valid = any(c.validity is not None for c in ins)
         out_valid = None
         if any_valid:
@@ -1904,3 +1943,15 @@ def ts_parse(col: Column, tokens, width: int) -> Column:
                      _ptr(col.validity), tok.data_ptr(), len(tokens),
                      width, out.data_ptr(), ov.data_ptr(), n, s)
     return Column(DType.timestamp(), n, out, ov, null_count=None)
+
+
+def and_parent_validity(kid: Column, parent: Column) -> Column:
+    """Null out child rows where the parent (struct) row is null."""
+    if parent.validity is None:
+        return kid
+    n = kid.size
+    s = _stream()
+    mask = _and_masks(kid.validity, parent.validity) \
+        if kid.validity is not None else parent.validity.clone()
+    return Column(kid.dtype, n, kid.data, mask, kid.offsets,
+                  null_count=None, child=kid.child)

@@ -156,6 +156,8 @@ class Tagger:
             pass  # lpad/rpad/locate/instr device kernels (k_str_pad/locate)
         elif type(e).__name__ in ("DateFormat", "ToTimestamp", "TzConvert"):
             pass  # k_date_format / k_ts_parse / k_tz_convert (tzdb table)
+        elif type(e).__name__ in ("CreateNamedStruct", "GetStructField"):
+            pass  # struct columns: child-wise device columns
         elif type(e).__name__ == "HostStringFn":
             out.append(f"{getattr(e, 'name', type(e).__name__)} runs on "
                        "CPU this round")

@@ -72,6 +72,7 @@ class DType:
     precision: int = 0
     scale: int = 0
     children: tuple = ()  # tuple[DType, ...] for LIST/STRUCT
+    field_names: tuple = ()  # STRUCT member names (parallel to children)
 
     # ---- constructors -------------------------------------------------
     @staticmethod
@@ -126,7 +127,17 @@ class DType:
 
     @staticmethod
     def struct(*fields: "DType") -> "DType":
-        return DType(TypeId.STRUCT, children=tuple(fields))
+        names = tuple(f"c{i}" for i in range(len(fields)))
+        return DType(TypeId.STRUCT, children=tuple(fields),
+                     field_names=names)
+
+    @staticmethod
+    def struct_(fields) -> "DType":
+        """struct from (name, DType) pairs."""
+        fields = list(fields)
+        return DType(TypeId.STRUCT,
+                     children=tuple(t for _, t in fields),
+                     field_names=tuple(n for n, _ in fields))
 
     # ---- predicates ---------------------------------------------------
     @property
@@ -174,7 +185,9 @@ class DType:
         if self.id is TypeId.LIST:
             return f"array<{self.children[0]}>"
         if self.id is TypeId.STRUCT:
-            return f"struct<{', '.join(str(c) for c in self.children)}>"
+            inner = ", ".join(f"{n}:{c}" for n, c in
+                              zip(self.field_names, self.children))
+            return f"struct<{inner}>"
         return self.id.value
 
 
