@@ -1082,7 +1082,7 @@ def _gather_by_idx(batch: ColumnBatch, idx: torch.Tensor, n_out: int,
     cols: List[Optional[Column]] = [None] * len(batch.columns)
     fixed: List[tuple] = []  # (position, column)
     for ci, c in enumerate(batch.columns):
-        if c.dtype.id is TypeId.STRING or n_out == 0:
+        if c.dtype.id in (TypeId.STRING, TypeId.STRUCT) or n_out == 0:
             cols[ci] = _gather_col(c, idx, n_out, maybe_negative)
         else:
             fixed.append((ci, c))
