@@ -62,6 +62,12 @@ def arrow_to_dtype(t) -> DType:
         return STRING
     if pa.types.is_decimal(t):
         return DType.decimal(t.precision, t.scale)
+    if pa.types.is_list(t) or pa.types.is_large_list(t):
+        return DType.list_(arrow_to_dtype(t.value_type))
+    if pa.types.is_struct(t):
+        return DType.struct_([(t.field(i).name,
+                               arrow_to_dtype(t.field(i).type))
+                              for i in range(t.num_fields)])
     raise NotImplementedError(f"parquet type {t}")
 
 
@@ -71,10 +77,14 @@ def _arrow_array_to_column(arr, dtype: DType) -> Column:
     import pyarrow as pa
 
     arr = arr.combine_chunks() if isinstance(arr, pa.ChunkedArray) else arr
-    if arr.null_count == len(arr):
+    if arr.null_count == len(arr) and not dtype.is_nested:
         return Column.nulls(dtype, len(arr))
     if isinstance(arr, pa.Array) and arr.offset != 0:
         arr = pa.concat_arrays([arr])  # rebase offset
+    if dtype.is_nested:
+        # LIST/STRUCT adoption via python values (hybrid-reader nested
+        # path; device-decoded nested columns are a later round)
+        return Column.from_pylist(arr.to_pylist(), dtype)
     n = len(arr)
     bufs = arr.buffers()
     validity = None
@@ -133,7 +143,66 @@ def parquet_schema(path: str) -> Schema:
 
 # observability for GPU-vs-fallback scan routing (the bench and GPU tests
 # assert the device decode path actually ran — VERDICT: no silent fallback)
-SCAN_STATS = {"gpu_files": 0, "fallback_files": 0, "last_fallback": None}
+SCAN_STATS = {"gpu_files": 0, "fallback_files": 0, "last_fallback": None,
+              "rg_skipped": 0, "rg_scanned": 0}
+
+
+def _stat_overlaps(op: str, value, mn, mx) -> bool:
+    """Can any row in [mn, mx] satisfy `col op value`? Conservative:
+    comparison errors keep the row group."""
+    try:
+        if isinstance(mn, bytes):
+            mn = mn.decode("utf-8", "ignore")
+        if isinstance(mx, bytes):
+            mx = mx.decode("utf-8", "ignore")
+        if isinstance(value, (int, float)) and not isinstance(mn, str):
+            mn, mx, value = float(mn), float(mx), float(value)
+        elif isinstance(value, str) != isinstance(mn, str):
+            return True
+        if op == "eq":
+            return mn <= value <= mx
+        if op == "lt":
+            return mn < value
+        if op == "le":
+            return mn <= value
+        if op == "gt":
+            return mx > value
+        if op == "ge":
+            return mx >= value
+    except TypeError:
+        return True
+    return True
+
+
+def _rg_keep(path: str, triples) -> set:
+    """Row groups whose min/max stats can satisfy every conjunct."""
+    from .parquet_gpu import _file_meta
+
+    md, _, _ = _file_meta(path)
+    if md.num_row_groups == 0:
+        return set()
+    name_to_idx = {md.row_group(0).column(j).path_in_schema: j
+                   for j in range(md.num_columns)}
+    keep = set()
+    for rg in range(md.num_row_groups):
+        rgmd = md.row_group(rg)
+        ok = True
+        for name, op, value in triples:
+            j = name_to_idx.get(name)
+            if j is None:
+                continue
+            st = rgmd.column(j).statistics
+            if st is None or not st.has_min_max:
+                continue
+            if not _stat_overlaps(op, value, st.min, st.max):
+                ok = False
+                break
+        if ok:
+            keep.add(rg)
+            SCAN_STATS["rg_scanned"] += 1
+        else:
+            SCAN_STATS["rg_skipped"] += 1
+    return keep
 
 
 class ParquetTable:
@@ -165,16 +234,32 @@ class ParquetTable:
             keep = [f for f in self.schema.fields if f.name in columns]
             self.schema = Schema(keep)
 
-    def with_columns(self, names: List[str]) -> "ParquetTable":
-        """Narrowed view for column pruning: only these columns are read
-        and decoded (parquet stores column chunks separately on disk)."""
+    def _clone(self) -> "ParquetTable":
         t = ParquetTable.__new__(ParquetTable)
         t.replicated = self.replicated
         t.files = self.files
-        t.columns = list(names)
+        t.columns = self.columns
         t.reader = self.reader
         t.prefetch_threads = self.prefetch_threads
+        t.schema = self.schema
+        t.rg_predicate = getattr(self, "rg_predicate", None)
+        return t
+
+    def with_columns(self, names: List[str]) -> "ParquetTable":
+        """Narrowed view for column pruning: only these columns are read
+        and decoded (parquet stores column chunks separately on disk)."""
+        t = self._clone()
+        t.columns = list(names)
         t.schema = Schema([f for f in self.schema.fields if f.name in names])
+        return t
+
+    def with_predicate(self, triples) -> "ParquetTable":
+        """Attach simple (col, op, literal) conjuncts for row-group
+        min/max skipping (reference analogue: predicate pushdown + row
+        group filtering in GpuParquetScan.scala:107). The Filter node
+        stays above the scan; the stats only SKIP row groups."""
+        t = self._clone()
+        t.rg_predicate = list(triples)
         return t
 
     def _read_one(self, path: str) -> ColumnBatch:
@@ -188,12 +273,17 @@ class ParquetTable:
         return batch
 
     def _read_one_uncached(self, path: str) -> ColumnBatch:
+        keep = None
+        pred = getattr(self, "rg_predicate", None)
+        if pred:
+            keep = _rg_keep(path, pred)
         if self.reader == "GPU_DECODE":
             try:
                 from .parquet_gpu import read_parquet_gpu
 
                 out = read_parquet_gpu(path,
-                                       [f.name for f in self.schema.fields])
+                                       [f.name for f in self.schema.fields],
+                                       keep_rgs=keep)
                 SCAN_STATS["gpu_files"] += 1
                 return out
             except NotImplementedError as e:
@@ -203,6 +293,19 @@ class ParquetTable:
                 SCAN_STATS["last_fallback"] = f"{path}: {e}"
         import pyarrow.parquet as pq
 
+        if keep is not None:
+            pf = pq.ParquetFile(path)
+            if len(keep) == 0:
+                sch = pf.schema_arrow
+                import pyarrow as pa
+
+                tbl = pa.table({n: pa.array([], type=sch.field(n).type)
+                                for n in (self.columns
+                                          or sch.names)})
+            else:
+                tbl = pf.read_row_groups(sorted(keep),
+                                         columns=self.columns)
+            return arrow_table_to_batch(tbl)
         tbl = pq.read_table(path, columns=self.columns)
         return arrow_table_to_batch(tbl)
 
@@ -242,11 +345,34 @@ def write_parquet(batch: ColumnBatch, schema: Schema, path: str,
     pq.write_table(tbl, path, compression=compression)
 
 
+def _dtype_to_arrow(dtype: DType):
+    import pyarrow as pa
+
+    if dtype.id is TypeId.LIST:
+        return pa.list_(_dtype_to_arrow(dtype.children[0]))
+    if dtype.id is TypeId.STRUCT:
+        return pa.struct([(n, _dtype_to_arrow(t)) for n, t in
+                          zip(dtype.field_names, dtype.children)])
+    if dtype.id is TypeId.STRING:
+        return pa.string()
+    if dtype.is_decimal:
+        return pa.decimal128(dtype.precision, dtype.scale)
+    if dtype.id is TypeId.DATE32:
+        return pa.date32()
+    if dtype.id is TypeId.TIMESTAMP:
+        return pa.timestamp("us")
+    if dtype.id is TypeId.BOOL:
+        return pa.bool_()
+    return pa.from_numpy_dtype(dtype.numpy_dtype())
+
+
 def _column_to_arrow(c: Column, dtype: DType):
     import pyarrow as pa
 
     valid = c.valid_array()
     mask = None if valid.all() else ~valid
+    if dtype.is_nested:
+        return pa.array(c.to_pylist(), type=_dtype_to_arrow(dtype))
     if dtype.id is TypeId.STRING:
         return pa.array(c.to_pylist(), type=pa.string())
     if dtype.id is TypeId.BOOL:

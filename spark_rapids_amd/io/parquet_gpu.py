@@ -560,7 +560,8 @@ def _file_meta(path: str):
     return hit
 
 
-def read_parquet_gpu(path: str, columns: List[str]) -> ColumnBatch:
+def read_parquet_gpu(path: str, columns: List[str],
+                     keep_rgs=None) -> ColumnBatch:
     from ..ops.gpu_backend import ext, _stream
     from .parquet import arrow_to_dtype
 
@@ -572,6 +573,8 @@ def read_parquet_gpu(path: str, columns: List[str]) -> ColumnBatch:
     rg_batches = []
     with open(path, "rb") as f:
         for rg in range(md.num_row_groups):
+            if keep_rgs is not None and rg not in keep_rgs:
+                continue  # pruned by row-group min/max statistics
             rgmd = md.row_group(rg)
             cols = []
             for name in columns:

@@ -126,7 +126,36 @@ def push_filters(plan: L.LogicalPlan) -> L.LogicalPlan:
                           child.child)
         out = push_filters(merged)
         return out
+    if isinstance(child, L.Scan) and hasattr(child.source, "with_predicate"):
+        triples = _simple_conjuncts(plan.condition)
+        if triples:
+            # row-group min/max skipping; the Filter stays for exactness
+            src = child.source.with_predicate(triples)
+            return L.Filter(plan.condition,
+                            L.Scan(src, child.schema(), child.label))
     return plan
+
+
+_STAT_OPS = {"lt", "le", "gt", "ge", "eq"}
+_STAT_SWAP = {"lt": "gt", "le": "ge", "gt": "lt", "ge": "le", "eq": "eq"}
+
+
+def _simple_conjuncts(cond):
+    """(col, op, literal) triples usable against row-group statistics."""
+    from ..expr.expressions import BinaryExpr, ColumnRef, Literal
+
+    out = []
+    for c in _split_conjuncts(cond):
+        if not (isinstance(c, BinaryExpr) and c.op in _STAT_OPS):
+            continue
+        l, r = c.left, c.right
+        if isinstance(l, ColumnRef) and isinstance(r, Literal) \
+                and r.value is not None:
+            out.append((l.name, c.op, r.value))
+        elif isinstance(r, ColumnRef) and isinstance(l, Literal) \
+                and l.value is not None:
+            out.append((r.name, _STAT_SWAP[c.op], l.value))
+    return out
 
 
 def _with_children(plan: L.LogicalPlan, kids):

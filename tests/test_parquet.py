@@ -275,3 +275,66 @@ def test_gpu_decode_nds_staging_no_fallback(tmp_path):
     assert iop.SCAN_STATS["fallback_files"] == before["fallback_files"], \
         iop.SCAN_STATS["last_fallback"]
     assert iop.SCAN_STATS["gpu_files"] > before["gpu_files"]
+
+
+def test_row_group_stats_pruning(tmp_path):
+    """Selective predicates skip row groups via min/max statistics; the
+    result still matches a full scan (GpuParquetScan predicate pushdown
+    analogue)."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col
+    from spark_rapids_amd.io import parquet as iop
+
+    # sorted column -> disjoint row-group ranges
+    n = 40_000
+    t = pa.table({
+        "k": pa.array(np.arange(n, dtype=np.int64)),
+        "v": pa.array(np.arange(n, dtype=np.float64) * 0.5),
+        "s": pa.array([f"row{i:06d}" for i in range(n)]),
+    })
+    f = str(tmp_path / "sorted.parquet")
+    pq.write_table(t, f, row_group_size=5_000)
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.read_parquet(f)
+    before = dict(iop.SCAN_STATS)
+    out = df.filter(col("k") >= 37_000).to_pydict()
+    skipped = iop.SCAN_STATS["rg_skipped"] - before["rg_skipped"]
+    assert skipped == 7, (skipped, iop.SCAN_STATS)
+    assert sorted(out["k"]) == list(range(37_000, n))
+    # string stats too
+    before = dict(iop.SCAN_STATS)
+    out2 = df.filter(col("s") < "row001000").to_pydict()
+    assert iop.SCAN_STATS["rg_skipped"] - before["rg_skipped"] == 7
+    assert len(out2["s"]) == 1000
+    # non-selective predicate keeps everything
+    assert len(df.filter(col("v") >= 0.0).to_pydict()["v"]) == n
+
+
+def test_nested_parquet_roundtrip(tmp_path):
+    """LIST and STRUCT columns round-trip through parquet (hybrid reader
+    adoption; device decode of nested stays a fallback)."""
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import Column
+    from spark_rapids_amd.column import ColumnBatch, Field, Schema
+    from spark_rapids_amd.types import DType, INT32, INT64, STRING
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    st = DType.struct_([("a", INT32), ("b", STRING)])
+    lt = DType.list_(INT64)
+    rows_s = [{"a": 1, "b": "x"}, None, {"a": 3, "b": None}]
+    rows_l = [[1, 2], [], None]
+    cb = ColumnBatch([Column.from_pylist([1, 2, 3], INT32),
+                      Column.from_pylist(rows_s, st),
+                      Column.from_pylist(rows_l, lt)], 3)
+    df = s.from_batches([cb], Schema([Field("k", INT32), Field("st", st),
+                                      Field("ls", lt)]))
+    path = str(tmp_path / "nested.parquet")
+    s.write_parquet(df, path)
+    back = s.read_parquet(path).to_pydict()
+    assert back["st"] == rows_s
+    assert back["ls"] == [[1, 2], [], None]
