@@ -421,6 +421,9 @@ class Session:
         from .shuffle import dist as _dist
 
         _dist.set_codec(self.conf.get(SHUFFLE_CODEC))
+        from .config import SHUFFLE_WAVE_BYTES
+
+        _dist.set_wave_bytes(self.conf.get(SHUFFLE_WAVE_BYTES))
         from .config import FILECACHE
         from .io import filecache as _fc
 

@@ -134,6 +134,11 @@ MEM_SPILL_WATERMARK = float_conf(
     "Pool-usage fraction above which spillable batches are proactively "
     "moved to host (reference analogue: spill from the RMM event handler "
     "before allocations fail).")
+SHUFFLE_WAVE_BYTES = bytes_conf(
+    "spark.rapids.shuffle.wave.bytes", 1 << 30,
+    "Bytes budget per exchange wave: a shuffle whose serialized send "
+    "buffers exceed this is split into multiple all-to-all rounds so the "
+    "exchange memory stays bounded (bounce-buffer windowing analogue).")
 PINNED_POOL_SIZE = bytes_conf(
     "spark.rapids.memory.pinnedPool.size", 8 << 30,
     "Size of the pinned host memory pool used for spill and H2D/D2H staging.")

@@ -36,6 +36,18 @@ _ctx: Optional[DistContext] = None
 _codec: Optional[str] = None
 
 
+_wave_bytes = 1 << 30
+
+
+def set_wave_bytes(n: int):
+    global _wave_bytes
+    _wave_bytes = int(n)
+
+
+def wave_bytes() -> int:
+    return _wave_bytes
+
+
 def set_codec(name: Optional[str]):
     global _codec
     _codec = None if not name or name == "none" else name

@@ -29,14 +29,47 @@ def batch_schema(batch: ColumnBatch) -> Schema:
 
 def exchange_by_hash(batch: ColumnBatch, key_idx: Sequence[int]) -> List[ColumnBatch]:
     """Hash-partition rows across the world by key and exchange; returns the
-    batches received from every rank (caller concatenates)."""
+    batches received from every rank (caller concatenates).
+
+    The exchange runs in WAVES bounded by spark.rapids.shuffle.wave.bytes:
+    every rank agrees (all-reduce max) on the wave count, then sends row
+    slices of each partition per wave, so a skewed or oversized exchange
+    never materializes more than ~wave_bytes of serialized send buffer at
+    once (reference analogue: bounce-buffer windowing in
+    BufferSendState.scala / GpuShuffleCoalesceExec batching)."""
+    import torch
+
     c = dist.ctx()
     schema = batch_schema(batch)
     parted, offsets = ops.hash_partition(batch, list(key_idx), c.world)
-    send = [serialize_batch(_slice_batch(parted, offsets[r], offsets[r + 1]))
-            for r in range(c.world)]
-    recv = dist.all_to_all_bytes(send)
-    return [deserialize_batch(b, schema) for b in recv]
+    budget = dist.wave_bytes()
+    n_rows = max(parted.num_rows, 1)
+    bytes_per_row = max(parted.nbytes // n_rows, 1)
+    max_part = max((offsets[r + 1] - offsets[r] for r in range(c.world)),
+                   default=0)
+    rows_per_wave = max(int(budget // bytes_per_row), 1)
+    my_waves = (max_part + rows_per_wave - 1) // rows_per_wave or 1
+    t = torch.tensor([my_waves], dtype=torch.int64,
+                     device="cuda" if parted.is_cuda else "cpu")
+    import torch.distributed as td
+
+    td.all_reduce(t, op=td.ReduceOp.MAX)
+    nwaves = int(t.item())
+
+    received: List[ColumnBatch] = []
+    for w in range(nwaves):
+        send = []
+        for r in range(c.world):
+            lo, hi = offsets[r], offsets[r + 1]
+            wl = min(lo + w * rows_per_wave, hi)
+            wh = min(wl + rows_per_wave, hi)
+            send.append(serialize_batch(_slice_batch(parted, wl, wh)))
+        recv = dist.all_to_all_bytes(send)
+        for b in recv:
+            piece = deserialize_batch(b, schema)
+            if piece.num_rows or (nwaves == 1 and not received):
+                received.append(piece)
+    return received
 
 
 def gather_all(batch: ColumnBatch) -> List[ColumnBatch]:

@@ -6,9 +6,13 @@ is a single device-resident uint8 tensor so RCCL all-to-all moves it over
 xGMI without touching the host; every section is 8-byte aligned so
 deserialization is zero-copy tensor views into the received buffer.
 
-Layout: [int64 num_rows][per col: int64 has_validity, int64 data_bytes]
-        then per column: data (8B padded) [validity][offsets (strings)].
-Schema is known on both sides, so the header carries only sizes.
+Layout: [int64 num_rows] then a recursive per-column walk that both sides
+replay against the shared schema:
+  flat column:  header [has_validity, data_bytes];
+                sections data (8B padded) [validity] [offsets for STRING]
+  LIST column:  header [has_validity, child_rows];
+                sections offsets (8B padded) [validity]; then the child
+  STRUCT:       header [has_validity]; sections [validity]; then children
 """
 from __future__ import annotations
 
@@ -30,62 +34,108 @@ def _as_bytes(t: torch.Tensor) -> torch.Tensor:
     return t.contiguous().view(torch.uint8).view(-1)
 
 
+def _pad_to8(sections: List[torch.Tensor], t: torch.Tensor, dev):
+    sections.append(t)
+    if t.numel() % 8:
+        sections.append(torch.zeros(8 - t.numel() % 8, dtype=torch.uint8,
+                                    device=dev))
+
+
+def _ser_col(c: Column, header: List[int], sections: List[torch.Tensor],
+             dev):
+    has_valid = c.validity is not None
+    if c.dtype.id is TypeId.LIST:
+        child = c.child
+        header.extend([1 if has_valid else 0, child.size])
+        _pad_to8(sections, _as_bytes(c.offsets), dev)
+        if has_valid:
+            sections.append(_as_bytes(c.validity))
+        _ser_col(child, header, sections, dev)
+        return
+    if c.dtype.id is TypeId.STRUCT:
+        header.append(1 if has_valid else 0)
+        if has_valid:
+            sections.append(_as_bytes(c.validity))
+        for kid in c.child:
+            _ser_col(kid, header, sections, dev)
+        return
+    data_b = _as_bytes(c.data)
+    header.extend([1 if has_valid else 0, data_b.numel()])
+    _pad_to8(sections, data_b, dev)
+    if has_valid:
+        sections.append(_as_bytes(c.validity))
+    if c.dtype.id is TypeId.STRING:
+        _pad_to8(sections, _as_bytes(c.offsets), dev)
+
+
 def serialize_batch(batch: ColumnBatch) -> torch.Tensor:
     n = batch.num_rows
-    dev = batch.columns[0].data.device if batch.columns else torch.device("cpu")
-    header = [n]
+    dev = batch.columns[0].data.device if batch.columns else \
+        torch.device("cpu")
+    header: List[int] = [n]
     sections: List[torch.Tensor] = []
     for c in batch.columns:
-        if c.dtype.id is TypeId.LIST:
-            raise NotImplementedError(
-                "LIST columns cannot be shuffled yet (collect/explode "
-                "happen after the exchange by design)")
-        data_b = _as_bytes(c.data)
-        header.extend([1 if c.validity is not None else 0, data_b.numel()])
-        sections.append(data_b)
-        if data_b.numel() % 8:
-            sections.append(torch.zeros(8 - data_b.numel() % 8,
-                                        dtype=torch.uint8, device=dev))
-        if c.validity is not None:
-            sections.append(_as_bytes(c.validity))
-        if c.dtype.id is TypeId.STRING:
-            ob = _as_bytes(c.offsets)
-            sections.append(ob)
-            if ob.numel() % 8:
-                sections.append(torch.zeros(8 - ob.numel() % 8,
-                                            dtype=torch.uint8, device=dev))
+        _ser_col(c, header, sections, dev)
     ht = torch.tensor(header, dtype=torch.int64).view(torch.uint8).view(-1)
+    nht = torch.tensor([ht.numel() + 8], dtype=torch.int64) \
+        .view(torch.uint8).view(-1)
     sections.insert(0, ht.to(dev))
+    sections.insert(0, nht.to(dev))
     return torch.cat(sections) if len(sections) > 1 else sections[0]
 
 
+class _Reader:
+    def __init__(self, buf: torch.Tensor, header: List[int], off: int):
+        self.buf = buf
+        self.header = header
+        self.h = 1  # header[0] is num_rows
+        self.off = off
+
+    def take_header(self) -> int:
+        v = self.header[self.h]
+        self.h += 1
+        return v
+
+    def take(self, nbytes: int, pad: bool) -> torch.Tensor:
+        t = self.buf[self.off:self.off + nbytes]
+        self.off += _pad8(nbytes) if pad else nbytes
+        return t
+
+
+def _deser_col(dtype: DType, n: int, r: _Reader) -> Column:
+    if dtype.id is TypeId.LIST:
+        has_valid = r.take_header() != 0
+        child_rows = r.take_header()
+        offsets = r.take((n + 1) * 4, pad=True).view(torch.int32)
+        validity = r.take(mask_nbytes(n), pad=False) if has_valid else None
+        child = _deser_col(dtype.children[0], child_rows, r)
+        return Column(dtype, n, torch.zeros(0, dtype=torch.uint8,
+                                            device=r.buf.device),
+                      validity, offsets, None if has_valid else 0, child)
+    if dtype.id is TypeId.STRUCT:
+        has_valid = r.take_header() != 0
+        validity = r.take(mask_nbytes(n), pad=False) if has_valid else None
+        kids = tuple(_deser_col(cd, n, r) for cd in dtype.children)
+        return Column(dtype, n, torch.zeros(0, dtype=torch.uint8,
+                                            device=r.buf.device),
+                      validity, None, None if has_valid else 0, kids)
+    has_valid = r.take_header() != 0
+    data_bytes = r.take_header()
+    data = r.take(data_bytes, pad=True)
+    if dtype.id is not TypeId.STRING:
+        data = data.view(torch_dtype(dtype))
+    validity = r.take(mask_nbytes(n), pad=False) if has_valid else None
+    offsets = None
+    if dtype.id is TypeId.STRING:
+        offsets = r.take((n + 1) * 4, pad=True).view(torch.int32)
+    return Column(dtype, n, data, validity, offsets,
+                  null_count=None if has_valid else 0)
+
+
 def deserialize_batch(buf: torch.Tensor, schema: Schema) -> ColumnBatch:
-    ncols = len(schema.fields)
-    hdr_bytes = 8 * (1 + 2 * ncols)
-    header = buf[:hdr_bytes].view(torch.int64)
-    header = header.cpu().tolist()
+    (hdr_bytes,) = buf[:8].view(torch.int64).cpu().tolist()
+    header = buf[8:hdr_bytes].view(torch.int64).cpu().tolist()
     n = header[0]
-    off = hdr_bytes
-    cols: List[Column] = []
-    for i, f in enumerate(schema.fields):
-        has_valid = header[1 + 2 * i] != 0
-        data_bytes = header[2 + 2 * i]
-        data = buf[off:off + data_bytes]
-        off += _pad8(data_bytes)
-        if f.dtype.id is TypeId.STRING:
-            data = data.view(torch.uint8)
-        else:
-            data = data.view(torch_dtype(f.dtype))
-        validity = None
-        if has_valid:
-            vb = mask_nbytes(n)
-            validity = buf[off:off + vb]
-            off += vb
-        offsets = None
-        if f.dtype.id is TypeId.STRING:
-            ob = (n + 1) * 4
-            offsets = buf[off:off + ob].view(torch.int32)
-            off += _pad8(ob)
-        cols.append(Column(f.dtype, n, data, validity, offsets,
-                           null_count=None if has_valid else 0))
+    r = _Reader(buf, header, hdr_bytes)
+    cols = [_deser_col(f.dtype, n, r) for f in schema.fields]
     return ColumnBatch(cols, n)

@@ -176,6 +176,32 @@ def main():
         assert flat == sorted(flat, reverse=True), "ranges out of order"
         assert len(flat) == n_total
 
+    # bounded-wave exchange: a tiny wave budget splits the shuffle into
+    # many all-to-all rounds; results must be identical
+    from spark_rapids_amd.shuffle import dist as _d
+
+    baseline = sorted(df.group_by("k").agg(sum_(col("v"))).collect())
+    _d.set_wave_bytes(2048)
+    try:
+        waved = sorted(df.group_by("k").agg(sum_(col("v"))).collect())
+    finally:
+        _d.set_wave_bytes(1 << 30)
+    assert waved == baseline, (len(waved), len(baseline))
+
+    # LIST columns survive the exchange (serializer nested walk): the
+    # distributed sort range-exchanges every column, including lv
+    lists_df = s.create_dataframe({
+        "k": rows,
+        "lv": [[int(v), int(v) + 1] if v % 3 else None
+               for v in rows.tolist()],
+    })
+    srt = lists_df.sort("k").collect()
+    for k, lv in srt:
+        assert lv == ([k, k + 1] if k % 3 else None), (k, lv)
+    t2 = torch.tensor([len(srt)], dtype=torch.int64)
+    td.all_reduce(t2)
+    assert int(t2.item()) == n_total
+
     td.barrier()
     if rank == 0:
         print("DIST_OK")

@@ -126,3 +126,23 @@ def test_two_process_nds_parquet_sharded(tmp_path):
                        timeout=300)
     assert r.returncode == 0, r.stdout[-3000:] + r.stderr[-3000:]
     assert "DIST_OK" in r.stdout
+
+
+def test_serializer_nested_roundtrip():
+    """LIST and STRUCT columns serialize for shuffle (recursive walk)."""
+    from spark_rapids_amd.types import DType, INT32, INT64, STRING
+
+    st = DType.struct_([("a", INT32), ("b", STRING)])
+    lt = DType.list_(INT64)
+    cols = [
+        Column.from_pylist([[1, 2], None, [], [7]], lt),
+        Column.from_pylist([{"a": 1, "b": "x"}, None,
+                            {"a": None, "b": "yy"}, {"a": 4, "b": None}],
+                           st),
+        Column.from_pylist([1.5, None, 2.5, 3.5], FLOAT64),
+    ]
+    b = ColumnBatch(cols, 4)
+    buf = serializer.serialize_batch(b)
+    out = serializer.deserialize_batch(buf, batch_schema(b))
+    for c, g in zip(b.columns, out.columns):
+        assert c.to_pylist() == g.to_pylist()
