@@ -324,10 +324,46 @@ class ParquetTable:
             for f in files:
                 yield self._read_one(f)
             return
+        import torch
+
+        use_streams = torch.cuda.is_available() \
+            and self.reader == "GPU_DECODE"
+        if not use_streams:
+            with cf.ThreadPoolExecutor(self.prefetch_threads) as pool:
+                futures = [pool.submit(self._read_one, f) for f in files]
+                for fut in futures:
+                    yield fut.result()
+            return
+        # copy/compute overlap: each prefetch worker decodes on its own
+        # HIP stream (uploads + decode kernels), so file IO, H2D and
+        # device decode of file k+1 overlap compute on file k; the
+        # consumer stream waits on the worker's recorded event before
+        # touching the batch (reference analogue: the multithreaded
+        # reader pool feeding the main task stream, §2.7 pipeline row)
+        side_streams = [torch.cuda.Stream()
+                        for _ in range(self.prefetch_threads)]
+        tl = __import__("threading").local()
+        counter = __import__("itertools").count()
+
+        def read_on_stream(f):
+            st = getattr(tl, "stream", None)
+            if st is None:
+                st = side_streams[next(counter) % len(side_streams)]
+                tl.stream = st
+            with torch.cuda.stream(st):
+                batch = self._read_one(f)
+                ev = torch.cuda.Event()
+                ev.record(st)
+            return batch, ev
+
         with cf.ThreadPoolExecutor(self.prefetch_threads) as pool:
-            futures = [pool.submit(self._read_one, f) for f in files]
+            futures = [pool.submit(read_on_stream, f) for f in files]
             for fut in futures:
-                yield fut.result()
+                batch, ev = fut.result()
+                # order the consumer (current/default) stream after the
+                # producer stream's work without a host sync
+                ev.wait(torch.cuda.current_stream())
+                yield batch
 
 
 def write_parquet(batch: ColumnBatch, schema: Schema, path: str,

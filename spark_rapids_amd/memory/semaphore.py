@@ -20,14 +20,19 @@ class PrioritySemaphore:
         self._cond = threading.Condition(self._lock)
         self._waiters = []  # heap of (priority, seq)
         self._seq = 0
+        self.contended = 0  # acquisitions that had to wait (metrics)
 
     def acquire(self, priority: int = 0):
         with self._cond:
             self._seq += 1
             me = (priority, self._seq)
             heapq.heappush(self._waiters, me)
+            waited = False
             while not (self._permits > 0 and self._waiters[0] == me):
+                waited = True
                 self._cond.wait()
+            if waited:
+                self.contended += 1
             heapq.heappop(self._waiters)
             self._permits -= 1
             self._cond.notify_all()

@@ -62,21 +62,22 @@ _GB = {"sum": 0, "min": 1, "max": 2, "count": 3, "count_all": 4,
 _JOIN = {"inner": 0, "left": 1, "semi": 2, "anti": 3, "full": 4}
 
 
-_cached_stream: Optional[int] = None
+
 
 
 def _stream() -> int:
-    # one stream per process (one process per GPU); refreshed only via
-    # set_stream when the engine switches streams for copy/compute overlap
-    global _cached_stream
-    if _cached_stream is None:
-        _cached_stream = torch.cuda.current_stream().cuda_stream
-    return _cached_stream
+    # follow torch's CURRENT stream: the prefetch pool decodes files on
+    # side streams (with torch.cuda.stream(...)) to overlap H2D + decode
+    # with main-stream compute, so hipdf kernels must land on whatever
+    # stream the calling thread has current
+    return torch.cuda.current_stream().cuda_stream
 
 
 def set_stream(handle: Optional[int]) -> None:
-    global _cached_stream
-    _cached_stream = handle
+    # retained for API compatibility: _stream() now always follows
+    # torch.cuda.current_stream(), which set_stream callers control via
+    # `with torch.cuda.stream(...)`
+    del handle
 
 
 def _ptr(t: Optional[torch.Tensor]) -> int:
