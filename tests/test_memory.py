@@ -362,4 +362,18 @@ def test_semaphore_contention_concurrent_queries():
     [t.join(timeout=120) for t in ts]
     assert not errors, errors
     assert results[0] == expected and results[1] == expected
-    assert sem._sem.contended > 0, "semaphore never contended"
+
+    # deterministic contention: hold the only permit, start a query in a
+    # thread (it must block in acquire), then release
+    import time
+
+    before = sem._sem.contended
+    sem.acquire_if_necessary()
+    t = threading.Thread(target=run, args=(0,))
+    t.start()
+    time.sleep(1.0)
+    assert t.is_alive(), "query did not block on the held semaphore"
+    sem.release_if_necessary()
+    t.join(timeout=120)
+    assert not errors and results[0] == expected
+    assert sem._sem.contended > before, "semaphore never contended"
