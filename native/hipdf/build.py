@@ -92,6 +92,25 @@ def build(verbose: bool = True) -> str:
             os.path.getmtime(o) > os.path.getmtime(out) for o in link_inputs):
         subprocess.run([HIPCC, "-shared", "-fPIC", "-o", out] + link_inputs +
                        [f"--offload-arch={ARCH}"], check=True)
+    # C ABI artifacts: libhipdf.so (kernel objects only, no python symbols)
+    # + the compiled C consumer test (a non-python host can
+    # link the same surface the pybind module uses)
+    libso = os.path.join(REPO, "libhipdf.so")
+    if not os.path.exists(libso) or any(
+            os.path.getmtime(o) > os.path.getmtime(libso) for o in objs):
+        subprocess.run([HIPCC, "-shared", "-fPIC", "-o", libso] + objs +
+                       [f"--offload-arch={ARCH}"], check=True)
+    ctest_src = os.path.join(HERE, "tests", "c_api_test.cpp")
+    ctest_bin = os.path.join(REPO, "c_api_test")
+    if os.path.exists(ctest_src) and (
+            not os.path.exists(ctest_bin)
+            or os.path.getmtime(ctest_src) > os.path.getmtime(ctest_bin)
+            or os.path.getmtime(libso) > os.path.getmtime(ctest_bin)):
+        subprocess.run([HIPCC, f"-I{os.path.join(HERE, 'include')}",
+                        ctest_src, f"-L{REPO}", "-lhipdf",
+                        f"-Wl,-rpath,$ORIGIN", "-o", ctest_bin,
+                        f"--offload-arch={ARCH}"], check=True)
+
     # import check in a fresh interpreter (catches missing symbols at link)
     subprocess.run([sys.executable, "-c",
                     f"import sys; sys.path.insert(0, {REPO!r}); import hipdf"],

@@ -377,3 +377,18 @@ def test_semaphore_contention_concurrent_queries():
     t.join(timeout=120)
     assert not errors and results[0] == expected
     assert sem._sem.contended > before, "semaphore never contended"
+
+
+@pytest.mark.gpu
+def test_c_api_consumer():
+    """The compiled non-python C consumer (native/hipdf/tests/
+    c_api_test.cpp) links libhipdf.so and runs filter+gather+groupby on
+    device (VERDICT round 1 #10: the JNI-able native boundary)."""
+    import subprocess
+
+    binpath = os.path.join(REPO, "c_api_test")
+    assert os.path.exists(binpath), "c_api_test not built (build() makes it)"
+    r = subprocess.run([binpath], capture_output=True, text=True,
+                       timeout=120, cwd=REPO)
+    assert r.returncode == 0, r.stdout + r.stderr
+    assert "C_API_OK" in r.stdout, r.stdout
