@@ -289,8 +289,14 @@ class Column:
             data = one.repeat(size).to(device) if device != "cpu" \
                 else one.repeat(size)
             return Column(dtype, size, data, None, null_count=0)
-        if dtype.id is TypeId.DECIMAL64 and isinstance(value, float):
-            value = int(round(value * (10 ** dtype.scale)))
+        if dtype.id is TypeId.DECIMAL64:
+            import decimal as _dec
+
+            if isinstance(value, _dec.Decimal):
+                value = int(value.scaleb(dtype.scale)
+                            .to_integral_value(_dec.ROUND_HALF_UP))
+            elif isinstance(value, float):
+                value = int(round(value * (10 ** dtype.scale)))
         data = torch.full((size,), value, dtype=torch_dtype(dtype),
                           device=device)
         return Column(dtype, size, data, None, null_count=0)
