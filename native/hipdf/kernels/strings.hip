@@ -445,10 +445,22 @@ __global__ void k_str_trim_ranges(int mode, const int32_t* __restrict__ ao,
   for (int64_t i = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; i < n;
        i += (int64_t)gridDim.x * blockDim.x) {
     int32_t a = ao[i], b = ao[i + 1];
-    if (mode != 2)
-      while (a < b && ab[a] == ' ') ++a;
-    if (mode != 1)
-      while (b > a && ab[b - 1] == ' ') --b;
+    // modes 0..2: spark trim()/ltrim()/rtrim() strip only 0x20; mode 3:
+    // full ASCII whitespace both sides (string->numeric cast semantics)
+    if (mode == 3) {
+      while (a < b && (ab[a] == ' ' || ab[a] == '\t' || ab[a] == '\r' ||
+                       ab[a] == '\n' || ab[a] == '\f' || ab[a] == '\v'))
+        ++a;
+      while (b > a && (ab[b - 1] == ' ' || ab[b - 1] == '\t' ||
+                       ab[b - 1] == '\r' || ab[b - 1] == '\n' ||
+                       ab[b - 1] == '\f' || ab[b - 1] == '\v'))
+        --b;
+    } else {
+      if (mode != 2)
+        while (a < b && ab[a] == ' ') ++a;
+      if (mode != 1)
+        while (b > a && ab[b - 1] == ' ') --b;
+    }
     bstart[i] = a;
     blen[i] = b - a;
   }

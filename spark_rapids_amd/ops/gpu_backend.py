@@ -558,7 +558,7 @@ def str_trim(col: Column, mode: str) -> Column:
     s = _stream()
     if n == 0:
         return _empty_col(DType.string())
-    m = {"both": 0, "leading": 1, "trailing": 2}[mode]
+    m = {"both": 0, "leading": 1, "trailing": 2, "cast": 3}[mode]
     bstart = torch.empty(n, dtype=torch.int32, device="cuda")
     blen = torch.empty(n, dtype=torch.int64, device="cuda")
     ext.str_trim_ranges(m, col.offsets.data_ptr(), col.data.data_ptr(),
