@@ -153,6 +153,12 @@ def main():
             print(f"[per-query] {qname}: "
                   f"{(time.perf_counter() - tq) * 1000:.1f} ms",
                   file=sys.stderr)
+        # per-exec wall breakdown of one representative query
+        fn = dict(POWER_RUN)["q3"]
+        dfq = fn(tables)
+        dfq.collect()
+        for m in dfq.metrics():
+            print(f"[q3-metrics] {m}", file=sys.stderr)
         barrier_sync()
 
     # ---- timed region ----
@@ -215,6 +221,10 @@ def main():
                 "scan_stats": __import__(
                     "spark_rapids_amd.io.parquet", fromlist=["SCAN_STATS"]
                 ).SCAN_STATS,
+                "scan_phases": {
+                    k: round(v, 3) for k, v in __import__(
+                        "spark_rapids_amd.io.parquet",
+                        fromlist=["PHASE_STATS"]).PHASE_STATS.items()},
                 "speedup_vs_cpu_backend": speedup,
                 "cpu_baseline_procs": args.cpu_baseline_procs,
                 "baseline_definition": "vs_baseline = speedup / 3.0: "

@@ -562,10 +562,14 @@ def _file_meta(path: str):
 
 def read_parquet_gpu(path: str, columns: List[str],
                      keep_rgs=None) -> ColumnBatch:
-    from ..ops.gpu_backend import ext, _stream
-    from .parquet import arrow_to_dtype
+    import time as _time
 
+    from ..ops.gpu_backend import ext, _stream
+    from .parquet import PHASE_STATS, arrow_to_dtype
+
+    t0 = _time.perf_counter()
     md, arrow_schema, pq_schema = _file_meta(path)
+    PHASE_STATS["meta_s"] += _time.perf_counter() - t0
     name_to_idx = {md.row_group(0).column(j).path_in_schema: j
                    for j in range(md.num_columns)} if md.num_row_groups else {}
     s = _stream()
@@ -587,12 +591,16 @@ def read_parquet_gpu(path: str, columns: List[str],
                 start = cmd.dictionary_page_offset \
                     if cmd.dictionary_page_offset is not None \
                     else cmd.data_page_offset
+                t1 = _time.perf_counter()
                 f.seek(start)
                 raw = f.read(cmd.total_compressed_size)
+                t2 = _time.perf_counter()
+                PHASE_STATS["io_s"] += t2 - t1
                 dec = _ChunkDecoder(raw, cmd.physical_type, dtype, max_def,
                                     _codec(cmd.compression), cmd.num_values,
                                     ext, s)
                 cols.append(dec.decode())
+                PHASE_STATS["decode_s"] += _time.perf_counter() - t2
             rg_batches.append(ColumnBatch(cols, rgmd.num_rows))
     if len(rg_batches) == 1:
         return rg_batches[0]
