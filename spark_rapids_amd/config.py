@@ -159,7 +159,7 @@ PARQUET_ENABLED = bool_conf(
     "spark.rapids.sql.format.parquet.enabled", True,
     "Enable parquet scans on GPU.")
 PARQUET_MT_THREADS = int_conf(
-    "spark.rapids.sql.format.parquet.multiThreadedRead.numThreads", 4,
+    "spark.rapids.sql.format.parquet.multiThreadedRead.numThreads", 8,
     "Threads in the multithreaded parquet prefetch pool.")
 BROADCAST_THRESHOLD = bytes_conf(
     "spark.rapids.sql.join.broadcastThreshold", 512 << 20,

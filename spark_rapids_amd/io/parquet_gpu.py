@@ -570,12 +570,18 @@ def read_parquet_gpu(path: str, columns: List[str],
     t0 = _time.perf_counter()
     md, arrow_schema, pq_schema = _file_meta(path)
     PHASE_STATS["meta_s"] += _time.perf_counter() - t0
+    import mmap as _mmap
     name_to_idx = {md.row_group(0).column(j).path_in_schema: j
                    for j in range(md.num_columns)} if md.num_row_groups else {}
     s = _stream()
 
     rg_batches = []
     with open(path, "rb") as f:
+        # zero-copy chunk access: page-cache-resident column chunks are
+        # sliced as memoryviews instead of read() into fresh bytes (the
+        # read copies alone were ~0.8 s/step at 20M rows)
+        mm = _mmap.mmap(f.fileno(), 0, prot=_mmap.PROT_READ)
+        mv = memoryview(mm)
         for rg in range(md.num_row_groups):
             if keep_rgs is not None and rg not in keep_rgs:
                 continue  # pruned by row-group min/max statistics
@@ -592,8 +598,7 @@ def read_parquet_gpu(path: str, columns: List[str],
                     if cmd.dictionary_page_offset is not None \
                     else cmd.data_page_offset
                 t1 = _time.perf_counter()
-                f.seek(start)
-                raw = f.read(cmd.total_compressed_size)
+                raw = mv[start:start + cmd.total_compressed_size]
                 t2 = _time.perf_counter()
                 PHASE_STATS["io_s"] += t2 - t1
                 dec = _ChunkDecoder(raw, cmd.physical_type, dtype, max_def,
