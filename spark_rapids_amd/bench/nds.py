@@ -198,7 +198,10 @@ _FACT_OPTS = dict(
     compression=None,            # measured path is GPU decode, not zstd
     use_dictionary=False,
     store_decimal_as_integer=True,   # decimal(7,2) -> INT32 physical
-    data_page_size=1 << 20,
+    data_page_size=8 << 20,
+    # pyarrow otherwise caps pages at ~20k rows, shattering a 2.5M-row
+    # chunk into 125 tiny pages whose header parse dominates the scan
+    max_rows_per_page=1 << 20,
     row_group_size=1 << 23,      # one row group per staged file
 )
 # dims: dictionary-encoded strings (the GPU string decode path); the large
@@ -208,7 +211,8 @@ _DIM_OPTS = dict(
     use_dictionary=True,
     store_decimal_as_integer=True,
     dictionary_pagesize_limit=1 << 26,
-    data_page_size=1 << 20,
+    data_page_size=8 << 20,
+    max_rows_per_page=1 << 20,
 )
 
 
