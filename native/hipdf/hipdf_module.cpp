@@ -122,6 +122,8 @@ void hipdf_change_flags(const void*, int, void*, int64_t, hipStream_t);
 void hipdf_iota_i32(void*, int64_t, hipStream_t);
 void hipdf_scan_block_f64(const void*, void*, void*, int64_t, hipStream_t);
 void hipdf_scan_add_offsets_f64(void*, const void*, int64_t, hipStream_t);
+void hipdf_rle_expand(const void*, const void*, int64_t, int, void*,
+                      int64_t, hipStream_t);
 void hipdf_rle_hybrid_batch(const void*, const void*, int, void*,
                             hipStream_t);
 void hipdf_rle_hybrid_decode(const void*, int64_t, int, void*, int64_t,
@@ -271,6 +273,51 @@ static int spill_cb_trampoline(size_t needed, int retry) {
   }
 }
 
+// Host-side walk of an RLE/bit-packed hybrid stream's RUN HEADERS: one
+// varint per run (cheap), emitting [kind, src_off_abs, out_off, count]
+// int64 records for the fully parallel k_rle_expand kernel. Returns the
+// number of runs, or -1 when max_runs would overflow / stream malformed.
+static int64_t rle_walk(const uint8_t* data, int64_t nbytes, int bit_width,
+                        int64_t n_values, int64_t src_base, int64_t out_base,
+                        int64_t* runs, int64_t max_runs) {
+  int64_t pos = 0, emitted = 0, nruns = 0;
+  int byte_per_val = (bit_width + 7) / 8;
+  while (emitted < n_values) {
+    if (pos >= nbytes) return -1;
+    uint64_t h = 0;
+    int shift = 0;
+    while (pos < nbytes) {
+      uint8_t b = data[pos++];
+      h |= (uint64_t)(b & 0x7F) << shift;
+      if (!(b & 0x80)) break;
+      shift += 7;
+    }
+    if (nruns >= max_runs) return -1;
+    int64_t* r = runs + 4 * nruns;
+    if ((h & 1) == 0) {
+      int64_t cnt = (int64_t)(h >> 1);
+      if (cnt <= 0) return -1;
+      r[0] = (int64_t)bit_width << 1;  // kind 0 | bw
+      r[1] = src_base + pos;
+      r[2] = out_base + emitted;
+      r[3] = cnt;
+      pos += byte_per_val;
+      emitted += cnt;
+    } else {
+      int64_t groups = (int64_t)(h >> 1);
+      if (groups <= 0) return -1;
+      r[0] = ((int64_t)bit_width << 1) | 1;  // kind 1 | bw
+      r[1] = src_base + pos;
+      r[2] = out_base + emitted;
+      r[3] = groups * 8;
+      pos += groups * bit_width;
+      emitted += groups * 8;
+    }
+    ++nruns;
+  }
+  return nruns;
+}
+
 // Host-side walk of parquet length-prefixed BYTE_ARRAY records. The chain
 // pos -> len -> pos is inherently serial, so it runs on the CPU (one
 // dependent L1 load per record, ~2ns) instead of a single GPU thread
@@ -349,6 +396,18 @@ PYBIND11_MODULE(hipdf, m) {
     if (!g_spill_cb) g_spill_cb = new py::object();
     *g_spill_cb = f;
     hipdf_pool_set_failure_cb(f.is_none() ? nullptr : &spill_cb_trampoline);
+  });
+  m.def("rle_walk_host",
+        [](int64_t data, int64_t nbytes, int bw, int64_t n_values,
+           int64_t src_base, int64_t out_base, int64_t runs,
+           int64_t max_runs) -> int64_t {
+          return rle_walk((const uint8_t*)data, nbytes, bw, n_values,
+                          src_base, out_base, (int64_t*)runs, max_runs);
+        },
+        py::call_guard<py::gil_scoped_release>());
+  m.def("rle_expand", [](int64_t base, int64_t runs, int64_t nruns, int bw,
+                         int64_t out, int64_t n, int64_t stream) {
+    hipdf_rle_expand(P(base), P(runs), nruns, bw, PM(out), n, S(stream));
   });
   m.def("byte_array_offsets_host",
         [](int64_t data, int64_t nbytes, int64_t count, int64_t starts,

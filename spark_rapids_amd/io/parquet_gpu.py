@@ -237,22 +237,64 @@ class _ChunkDecoder:
             return self._materialize(blob, total, None, PLAIN)
         return None
 
-    def _chunk_validity(self, pages, total):
-        """(mask, valid_idx, n_valid) for the whole chunk: batch-decode all
-        pages' def-level RLE streams in ONE launch, then run the validity
-        chain once over the contiguous levels array."""
-        levels_blob = b"".join(p.levels for p in pages)
-        base = _upload(levels_blob)
-        descs = np.empty((len(pages), 5), dtype=np.int64)
+    def _rle_streams_to_i32(self, streams, total) -> torch.Tensor:
+        """Decode several RLE/bit-packed hybrid streams into one int32
+        tensor. Fast path: host walk of the run headers (one varint per
+        run, C++) + fully parallel k_rle_expand — a 2.5M-value stream is
+        decoded by the whole chip instead of one workgroup. Falls back to
+        the per-stream workgroup kernel when a walk fails."""
+        runs_list = []
+        parts = []
         src = out = 0
-        for i, p in enumerate(pages):
-            descs[i] = (src, len(p.levels), out, p.n, 1)
-            src += len(p.levels)
-            out += p.n
-        dt = torch.from_numpy(descs).cuda()
-        levels = torch.empty(total, dtype=torch.int32, device="cuda")
-        self.ext.rle_hybrid_batch(base.data_ptr(), dt.data_ptr(),
-                                  len(pages), levels.data_ptr(), self.s)
+        ok = True
+        for data, n, bw in streams:
+            arr = np.frombuffer(data, dtype=np.uint8)
+            max_runs = len(arr) + 2  # every run consumes >= 1 header byte
+            runs = np.empty((max_runs, 4), dtype=np.int64)
+            cnt = self.ext.rle_walk_host(arr.ctypes.data, len(arr), bw, n,
+                                         src, out, runs.ctypes.data,
+                                         max_runs)
+            if cnt < 0:
+                ok = False
+                break
+            runs_list.append(runs[:cnt])
+            parts.append(data)
+            src += len(arr)
+            out += n
+        out_t = torch.empty(max(total, 1), dtype=torch.int32,
+                            device="cuda")[:total]
+        if ok and total:
+            blob = b"".join(parts) + b"\x00" * 8  # bitpack tail overread
+            base = _upload(blob)
+            all_runs = np.concatenate(runs_list) if len(runs_list) > 1 \
+                else runs_list[0]
+            rt = _upload(all_runs).view(torch.int64)
+            self.ext.rle_expand(base.data_ptr(), rt.data_ptr(),
+                                len(all_runs), 0, out_t.data_ptr(), total,
+                                self.s)
+            return out_t
+        # fallback: per-stream workgroup decode
+        blob = b"".join(d for d, _, _ in streams)
+        base = _upload(blob)
+        descs = np.empty((len(streams), 5), dtype=np.int64)
+        src = out = 0
+        for i, (data, n, bw) in enumerate(streams):
+            descs[i] = (src, len(data), out, n, bw)
+            src += len(data)
+            out += n
+        dt = _upload(descs).view(torch.int64)
+        if total:
+            self.ext.rle_hybrid_batch(base.data_ptr(), dt.data_ptr(),
+                                      len(streams), out_t.data_ptr(),
+                                      self.s)
+        return out_t
+
+    def _chunk_validity(self, pages, total):
+        """(mask, valid_idx, n_valid) for the whole chunk: decode all
+        pages' def-level RLE streams at once, then run the validity
+        chain once over the contiguous levels array."""
+        levels = self._rle_streams_to_i32(
+            [(p.levels, p.n, 1) for p in pages], total)
         return self._valid_parts(levels, total)
 
     def _chunk_plain_fixed(self, pages, total, all_valid) -> Column:
@@ -283,22 +325,10 @@ class _ChunkDecoder:
     def _chunk_dict(self, pages, total) -> Optional[Column]:
         if self.dict_fixed is None and self.dict_str is None:
             return None
-        # batch-decode every page's index stream in one launch; first
-        # value byte of each page is the bit width
-        blob = b"".join(p.values[1:] for p in pages)
-        base = _upload(blob) if blob else \
-            torch.zeros(1, dtype=torch.uint8, device="cuda")
-        descs = np.empty((len(pages), 5), dtype=np.int64)
-        src = out = 0
-        for i, p in enumerate(pages):
-            nb = len(p.values) - 1
-            descs[i] = (src, nb, out, p.n, p.values[0] if p.values else 0)
-            src += nb
-            out += p.n
-        dt = torch.from_numpy(descs).cuda()
-        ridx = torch.empty(total, dtype=torch.int32, device="cuda")
-        self.ext.rle_hybrid_batch(base.data_ptr(), dt.data_ptr(),
-                                  len(pages), ridx.data_ptr(), self.s)
+        # decode every page's index stream (first value byte = bit width)
+        ridx = self._rle_streams_to_i32(
+            [(p.values[1:], p.n, p.values[0] if len(p.values) else 0)
+             for p in pages], total)
         return self._gather_dict(ridx, total, None)
 
     # -- dictionary ------------------------------------------------------
