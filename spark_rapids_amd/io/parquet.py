@@ -147,7 +147,9 @@ SCAN_STATS = {"gpu_files": 0, "fallback_files": 0, "last_fallback": None,
               "rg_skipped": 0, "rg_scanned": 0}
 # cumulative host-side phase timers for the GPU decode path (seconds,
 # summed across prefetch threads — can exceed wall when overlapped)
-PHASE_STATS = {"meta_s": 0.0, "io_s": 0.0, "decode_s": 0.0}
+PHASE_STATS = {"meta_s": 0.0, "io_s": 0.0, "decode_s": 0.0,
+               "parse_s": 0.0, "fast_s": 0.0, "upload_s": 0.0,
+               "upload_bytes": 0}
 
 
 def _stat_overlaps(op: str, value, mn, mx) -> bool:

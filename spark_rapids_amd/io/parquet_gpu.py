@@ -58,11 +58,18 @@ def _upload(data, torch_dtype=torch.uint8) -> torch.Tensor:
         cap = 1 << (cap - 1).bit_length()
         buf = torch.empty(cap, dtype=torch.uint8, pin_memory=True)
         _PINNED.buf = buf
+    import time as _t
+
+    from .parquet import PHASE_STATS
+
+    t0 = _t.perf_counter()
     buf.numpy()[:nbytes] = arr
     dev = torch.empty(nbytes, dtype=torch.uint8, device="cuda")
     # blocking copy: the pinned buffer is reused by the next upload on
     # this thread, so the DMA must complete before returning
     dev.copy_(buf[:nbytes])
+    PHASE_STATS["upload_s"] += _t.perf_counter() - t0
+    PHASE_STATS["upload_bytes"] += nbytes
     itemsize = torch.empty(0, dtype=torch_dtype).element_size()
     return dev.view(torch_dtype) if itemsize == 1 else \
         dev[: (nbytes // itemsize) * itemsize].view(torch_dtype)
@@ -148,10 +155,18 @@ class _ChunkDecoder:
         whole column chunk instead of a launch storm per page (the round-1
         profile showed per-page rle_hybrid_decode + a 6-kernel validity
         chain per page as 60%+ of NDS scan kernel time)."""
+        import time as _t
+
+        from .parquet import PHASE_STATS
+
+        t0 = _t.perf_counter()
         pages = self._parse_pages()
+        PHASE_STATS["parse_s"] += _t.perf_counter() - t0
         if not pages:
             return Column.nulls(self.dtype, 0, "cuda")
+        t0 = _t.perf_counter()
         fast = self._decode_chunk(pages)
+        PHASE_STATS["fast_s"] += _t.perf_counter() - t0
         if fast is not None:
             return fast
         page_cols = []
