@@ -397,6 +397,21 @@ PYBIND11_MODULE(hipdf, m) {
     *g_spill_cb = f;
     hipdf_pool_set_failure_cb(f.is_none() ? nullptr : &spill_cb_trampoline);
   });
+  m.def("host_register", [](int64_t ptr, int64_t nbytes) -> int {
+    // pin an existing host range (the parquet file mmap) so H2D copies
+    // from it are direct DMA instead of a staged bounce
+    return (int)hipHostRegister((void*)ptr, (size_t)nbytes,
+                                hipHostRegisterDefault);
+  });
+  m.def("host_unregister", [](int64_t ptr) {
+    hipHostUnregister((void*)ptr);
+  });
+  m.def("memcpy_h2d", [](int64_t dst, int64_t src, int64_t nbytes,
+                         int64_t stream) -> int {
+    return (int)hipMemcpyAsync((void*)dst, (const void*)src,
+                               (size_t)nbytes, hipMemcpyHostToDevice,
+                               S(stream));
+  });
   m.def("rle_walk_host",
         [](int64_t data, int64_t nbytes, int bw, int64_t n_values,
            int64_t src_base, int64_t out_base, int64_t runs,
