@@ -60,16 +60,13 @@ def _file_mmap(path: str):
         f.close()
         arr = np.frombuffer(memoryview(mm), dtype=np.uint8)
         base = arr.ctypes.data
-        registered = False
-        try:
-            from ..ops.gpu_backend import ext
-
-            registered = ext.host_register(base, arr.nbytes) == 0
-        except Exception:  # noqa: BLE001 - no GPU / already registered
-            registered = False
+        # NOTE: hipHostRegister on the file-backed mapping looked
+        # attractive (direct DMA from page cache) but the GPU faults on
+        # DMA from MAP_PRIVATE file pages on this stack — uploads bounce
+        # through the per-thread pinned buffer instead
         if len(_MMAP_CACHE) > 512:
             _MMAP_CACHE.clear()
-        hit = (mm, memoryview(mm), base, arr.nbytes, registered)
+        hit = (mm, memoryview(mm), base, arr.nbytes, False)
         _MMAP_CACHE[key] = hit
         return hit
 
