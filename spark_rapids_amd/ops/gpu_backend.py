@@ -801,7 +801,7 @@ def _cast_string_to(col: Column, to: DType, v) -> Column:
         return Column(to, n, narrow, ov, null_count=None)
     if not to.is_floating:
         raise NotImplementedError(f"gpu cast string -> {to}")
-    trimmed = str_trim(col, "both")
+    trimmed = str_trim(col, "cast")  # full ASCII whitespace (Spark casts)
     f64 = torch.empty(max(n, 1), dtype=torch.float64, device="cuda")[:n]
     valid_u8 = torch.empty(max(n, 1), dtype=torch.uint8, device="cuda")[:n]
     unsupported = torch.zeros(1, dtype=torch.int32, device="cuda")
