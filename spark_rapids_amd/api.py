@@ -590,19 +590,32 @@ class Session:
         CSV scan with the Hive delimiter (GpuHiveTextFileFormat analogue)."""
         return self.read_csv(path, header=header, delimiter=delimiter)
 
-    def read_delta(self, path: str) -> DataFrame:
+    def read_delta(self, path: str, version=None) -> DataFrame:
         """Delta Lake table scan: replay the _delta_log to the live file
-        set, then scan with the parquet reader (GPU page decode)."""
+        set (optionally at an older `version` — time travel), then scan
+        with the parquet reader (GPU page decode)."""
         from .io.delta import live_files
         from .io.parquet import ParquetTable
 
-        files = live_files(path)
+        files = live_files(path, version=version)
         if not files:
             raise FileNotFoundError(f"delta table has no live files: {path}")
         t = ParquetTable.__new__(ParquetTable)
         t.__init__(files[0])
         t.files = files
         return DataFrame(self, L.Scan(t, t.schema, "delta"))
+
+    def delta_table(self, path: str):
+        """Writer-side Delta handle (append/overwrite/delete/update/merge/
+        optimize/history)."""
+        from .io.delta_write import DeltaTable
+
+        return DeltaTable(self, path)
+
+    def write_delta(self, df: DataFrame, path: str, mode: str = "error"):
+        from .io.delta_write import DeltaTable
+
+        return DeltaTable.create(self, path, df, mode=mode)
 
     def read_avro(self, path: str) -> DataFrame:
         """Avro object-container scan (flat records; host decode)."""

@@ -16,7 +16,9 @@ import os
 from typing import Dict, List
 
 
-def live_files(table_path: str) -> List[str]:
+def live_files(table_path: str, version=None) -> List[str]:
+    """Live data files at `version` (None = latest) — time travel reads
+    replay the log only up to that commit."""
     log_dir = os.path.join(table_path, "_delta_log")
     if not os.path.isdir(log_dir):
         raise FileNotFoundError(f"not a Delta table (no _delta_log): "
@@ -24,6 +26,11 @@ def live_files(table_path: str) -> List[str]:
     adds: Dict[str, dict] = {}
     start_version = 0
     ckpt = os.path.join(log_dir, "_last_checkpoint")
+    if version is not None and os.path.exists(ckpt):
+        # time travel may predate the checkpoint: replay json from 0
+        info = json.loads(open(ckpt).read())
+        if int(info["version"]) > version:
+            ckpt = "/nonexistent"
     if os.path.exists(ckpt):
         import pyarrow.parquet as pq
 
@@ -47,9 +54,11 @@ def live_files(table_path: str) -> List[str]:
                     adds.pop(rm["path"], None)
         start_version = v + 1
     for f in sorted(glob.glob(os.path.join(log_dir, "*.json"))):
-        version = int(os.path.basename(f).split(".")[0])
-        if version < start_version:
+        v_f = int(os.path.basename(f).split(".")[0])
+        if v_f < start_version:
             continue
+        if version is not None and v_f > version:
+            break
         with open(f) as fh:
             for line in fh:
                 if not line.strip():
