@@ -605,6 +605,24 @@ class Session:
         t.files = files
         return DataFrame(self, L.Scan(t, t.schema, "delta"))
 
+    def read_iceberg(self, path: str, snapshot_id=None) -> DataFrame:
+        """Iceberg table scan: metadata.json -> manifest list -> manifests
+        -> live parquet files, v2 position deletes applied (reference
+        analogue: the iceberg/ GPU scan bridges)."""
+        import torch
+
+        from .config import PARQUET_MT_THREADS, PARQUET_READER_TYPE
+        from .io.iceberg import IcebergTable
+
+        reader = str(self.conf.get(PARQUET_READER_TYPE)).upper()
+        if reader == "AUTO":
+            reader = "GPU_DECODE" if (torch.cuda.is_available()
+                                      and self.conf.sql_enabled) else "CPU"
+        src = IcebergTable(path, snapshot_id=snapshot_id, reader=reader,
+                           prefetch_threads=self.conf.get(
+                               PARQUET_MT_THREADS))
+        return DataFrame(self, L.Scan(src, src.schema, f"iceberg:{path}"))
+
     def delta_table(self, path: str):
         """Writer-side Delta handle (append/overwrite/delete/update/merge/
         optimize/history)."""

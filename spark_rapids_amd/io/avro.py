@@ -11,6 +11,7 @@ the standard ["null", T] nullable unions.
 from __future__ import annotations
 
 import json
+import os
 import struct
 import zlib
 from typing import List, Optional, Tuple
@@ -215,3 +216,195 @@ class AvroTable:
         yield self._first
         for f in self.files[1:]:
             yield read_avro(f)[1]
+
+
+# ---- generic (nested) record layer ---------------------------------------
+# Iceberg manifests are deeply nested avro records (records in records,
+# arrays, maps, fixed); this generic reader/writer works on python dicts
+# and backs io/iceberg.py (reference analogue: the iceberg-core manifest
+# reading the reference reaches through its bridge classes).
+
+def _read_generic(b, p, ft, named):
+    if isinstance(ft, str) and ft in named:
+        ft = named[ft]
+    if isinstance(ft, list):
+        branch, p = _zigzag_dec(b, p)
+        t = ft[branch]
+        if t == "null":
+            return None, p
+        return _read_generic(b, p, t, named)
+    if isinstance(ft, dict):
+        t = ft["type"]
+        if t == "record":
+            named[ft["name"]] = ft
+            out = {}
+            for fd in ft["fields"]:
+                out[fd["name"]], p = _read_generic(b, p, fd["type"], named)
+            return out, p
+        if t == "array":
+            items = []
+            while True:
+                cnt, p = _zigzag_dec(b, p)
+                if cnt == 0:
+                    break
+                if cnt < 0:
+                    _, p = _zigzag_dec(b, p)  # byte size, unused
+                    cnt = -cnt
+                for _ in range(cnt):
+                    v, p = _read_generic(b, p, ft["items"], named)
+                    items.append(v)
+            return items, p
+        if t == "map":
+            out = {}
+            while True:
+                cnt, p = _zigzag_dec(b, p)
+                if cnt == 0:
+                    break
+                if cnt < 0:
+                    _, p = _zigzag_dec(b, p)
+                    cnt = -cnt
+                for _ in range(cnt):
+                    klen, p = _zigzag_dec(b, p)
+                    k = b[p:p + klen].decode()
+                    p += klen
+                    out[k], p = _read_generic(b, p, ft["values"], named)
+            return out, p
+        if t == "fixed":
+            named[ft["name"]] = ft
+            n = ft["size"]
+            return bytes(b[p:p + n]), p + n
+        if t == "enum":
+            named[ft["name"]] = ft
+            idx, p = _zigzag_dec(b, p)
+            return ft["symbols"][idx], p
+        # logical types wrap a primitive
+        return _read_generic(b, p, t, named)
+    return _read_value(b, p, ft)
+
+
+def read_avro_records(path: str):
+    """-> (schema_json, list[dict]) for arbitrary nested record schemas."""
+    with open(path, "rb") as f:
+        raw = f.read()
+    assert raw[:4] == _MAGIC, "not an avro container file"
+    p = 4
+    meta = {}
+    while True:
+        cnt, p = _zigzag_dec(raw, p)
+        if cnt == 0:
+            break
+        if cnt < 0:
+            _, p = _zigzag_dec(raw, p)
+            cnt = -cnt
+        for _ in range(cnt):
+            klen, p = _zigzag_dec(raw, p)
+            key = raw[p:p + klen].decode()
+            p += klen
+            vlen, p = _zigzag_dec(raw, p)
+            meta[key] = raw[p:p + vlen]
+            p += vlen
+    sync = raw[p:p + 16]
+    p += 16
+    codec = meta.get("avro.codec", b"null").decode()
+    schema_json = json.loads(meta["avro.schema"])
+    records = []
+    while p < len(raw):
+        nrec, p = _zigzag_dec(raw, p)
+        nbytes, p = _zigzag_dec(raw, p)
+        block = raw[p:p + nbytes]
+        p += nbytes
+        assert raw[p:p + 16] == sync, "avro: sync marker mismatch"
+        p += 16
+        if codec == "deflate":
+            block = zlib.decompress(block, -15)
+        elif codec != "null":
+            raise NotImplementedError(f"avro codec {codec}")
+        q = 0
+        for _ in range(nrec):
+            v, q = _read_generic(block, q, schema_json, {})
+            records.append(v)
+    return schema_json, records
+
+
+def _write_generic(out: bytearray, v, ft, named):
+    if isinstance(ft, str) and ft in named:
+        ft = named[ft]
+    if isinstance(ft, list):
+        if v is None and "null" in ft:
+            out += _zigzag_enc(ft.index("null"))
+            return
+        branch = next(i for i, t in enumerate(ft) if t != "null")
+        out += _zigzag_enc(branch)
+        _write_generic(out, v, ft[branch], named)
+        return
+    if isinstance(ft, dict):
+        t = ft["type"]
+        if t == "record":
+            named[ft["name"]] = ft
+            for fd in ft["fields"]:
+                _write_generic(out, v.get(fd["name"]), fd["type"], named)
+            return
+        if t == "array":
+            if v:
+                out += _zigzag_enc(len(v))
+                for item in v:
+                    _write_generic(out, item, ft["items"], named)
+            out += _zigzag_enc(0)
+            return
+        if t == "map":
+            if v:
+                out += _zigzag_enc(len(v))
+                for k, mv in v.items():
+                    kb = k.encode()
+                    out += _zigzag_enc(len(kb)) + kb
+                    _write_generic(out, mv, ft["values"], named)
+            out += _zigzag_enc(0)
+            return
+        if t == "fixed":
+            named[ft["name"]] = ft
+            out += v
+            return
+        _write_generic(out, v, t, named)
+        return
+    if ft == "boolean":
+        out.append(1 if v else 0)
+    elif ft in ("int", "long"):
+        out += _zigzag_enc(int(v))
+    elif ft == "float":
+        out += struct.pack("<f", v)
+    elif ft == "double":
+        out += struct.pack("<d", v)
+    elif ft == "string":
+        vb = v.encode()
+        out += _zigzag_enc(len(vb)) + vb
+    elif ft == "bytes":
+        out += _zigzag_enc(len(v)) + v
+    elif ft == "null":
+        pass
+    else:
+        raise NotImplementedError(f"avro write type {ft}")
+
+
+def write_avro_records(schema_json, records, path: str):
+    """Write arbitrary nested records (python dicts) as an avro container
+    file (null codec)."""
+    body = bytearray()
+    for r in records:
+        _write_generic(body, r, schema_json, {})
+    sync = os.urandom(16) if hasattr(os, "urandom") else b"\x01" * 16
+    out = bytearray(_MAGIC)
+    meta = {"avro.schema": json.dumps(schema_json).encode(),
+            "avro.codec": b"null"}
+    out += _zigzag_enc(len(meta))
+    for k, v in meta.items():
+        kb = k.encode()
+        out += _zigzag_enc(len(kb)) + kb
+        out += _zigzag_enc(len(v)) + v
+    out += _zigzag_enc(0)
+    out += sync
+    out += _zigzag_enc(len(records))
+    out += _zigzag_enc(len(body))
+    out += body
+    out += sync
+    with open(path, "wb") as f:
+        f.write(out)
