@@ -59,6 +59,12 @@ void hipdf_murmur3_col(int, int, const void*, const void*, const void*,
 void hipdf_murmur3_str(const void*, const void*, const void*, const void*,
                        void*, int64_t, hipStream_t);
 void hipdf_pmod_part(const void*, int, void*, int64_t, hipStream_t);
+void hipdf_gb_hll(const void*, const void*, const void*, const void*,
+                  void*, int, int64_t, hipStream_t);
+void hipdf_xxhash64_col(int, int, const void*, const void*, const void*,
+                        void*, int64_t, hipStream_t);
+void hipdf_xxhash64_str(const void*, const void*, const void*, const void*,
+                        void*, int64_t, hipStream_t);
 void hipdf_gb_build(const void*, const void*, int, const void*, void*,
                     void*, void*, void*, int64_t, int64_t, hipStream_t);
 void hipdf_gb_number(const void*, const void*, void*, const void*, void*,
@@ -382,6 +388,24 @@ PYBIND11_MODULE(hipdf, m) {
                          int mode, int64_t n, int64_t stream) {
     hipdf_dec_to_str(P(vals), in_is_128, scale, P(out_off), PM(out_len),
                      PM(out), mode, n, S(stream));
+  });
+  m.def("gb_hll", [](int64_t hashes, int64_t valid, int64_t row_gid,
+                     int64_t sel, int64_t regs, int p, int64_t n,
+                     int64_t stream) {
+    hipdf_gb_hll(P(hashes), P(valid), P(row_gid), P(sel), PM(regs), p, n,
+                 S(stream));
+  });
+  m.def("xxhash64_col", [](int kind, int t, int64_t a, int64_t av,
+                           int64_t sel, int64_t seeds, int64_t n,
+                           int64_t stream) {
+    hipdf_xxhash64_col(kind, t, P(a), P(av), P(sel), PM(seeds), n,
+                       S(stream));
+  });
+  m.def("xxhash64_str", [](int64_t offsets, int64_t bytes, int64_t av,
+                           int64_t sel, int64_t seeds, int64_t n,
+                           int64_t stream) {
+    hipdf_xxhash64_str(P(offsets), P(bytes), P(av), P(sel), PM(seeds), n,
+                       S(stream));
   });
   m.def("pool_init", [](double fraction, size_t bytes) {
     return hipdf_pool_init(fraction, bytes);

@@ -461,7 +461,56 @@ __global__ void k_gb_percentile(const double* __restrict__ vals,
   }
 }
 
+
+// ---- HyperLogLog++ register update (approx_count_distinct) ---------------
+// One uint8 register array of 2^p entries per group; rho = leading-zero
+// count of the suffix of the 64-bit hash + 1, kept as a byte max
+// (reference analogue: HyperLogLogPlusPlusHostUDF in spark-rapids-jni).
+
+__device__ __forceinline__ void atomic_max_u8(uint8_t* addr, uint8_t val) {
+  uint32_t* base = (uint32_t*)((uintptr_t)addr & ~(uintptr_t)3);
+  int shift = (int)((uintptr_t)addr & 3) * 8;
+  uint32_t old = *base;
+  while (true) {
+    uint8_t cur = (uint8_t)((old >> shift) & 0xFF);
+    if (cur >= val) return;
+    uint32_t updated = (old & ~(0xFFu << shift)) |
+                       ((uint32_t)val << shift);
+    uint32_t seen = atomicCAS(base, old, updated);
+    if (seen == old) return;
+    old = seen;
+  }
+}
+
+__global__ void k_gb_hll(const int64_t* __restrict__ hashes,
+                         const uint64_t* __restrict__ valid,
+                         const int32_t* __restrict__ row_gid,
+                         const int32_t* __restrict__ sel,
+                         uint8_t* __restrict__ regs, int p, int64_t n) {
+  int64_t m = (int64_t)1 << p;
+  for (int64_t j = (int64_t)blockIdx.x * blockDim.x + threadIdx.x; j < n;
+       j += (int64_t)gridDim.x * blockDim.x) {
+    int64_t i = sel ? (int64_t)sel[j] : j;
+    if (!valid_bit(valid, i)) continue;
+    uint64_t h = (uint64_t)hashes[j];
+    uint32_t idx = (uint32_t)(h >> (64 - p));
+    uint64_t w = h << p;
+    int rho = w == 0 ? (64 - p + 1) : (__clzll((long long)w) + 1);
+    atomic_max_u8(&regs[(int64_t)row_gid[j] * m + idx], (uint8_t)rho);
+  }
+}
+
 extern "C" {
+
+void hipdf_gb_hll(const void* hashes, const void* valid,
+                  const void* row_gid, const void* sel, void* regs, int p,
+                  int64_t n, hipStream_t stream) {
+  hipLaunchKernelGGL(k_gb_hll, flat_grid(n), dim3(HIPDF_BLOCK), 0, stream,
+                     (const int64_t*)hashes, (const uint64_t*)valid,
+                     (const int32_t*)row_gid, (const int32_t*)sel,
+                     (uint8_t*)regs, p, n);
+}
+
 
 void hipdf_gb_percentile(const void* vals, const void* perm,
                          const void* starts, const void* vcnt, double p,

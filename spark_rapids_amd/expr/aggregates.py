@@ -33,6 +33,8 @@ class AggExpr:
             return self._name
         if self.op.startswith("percentile:"):
             return f"percentile({self.child}, {self.op.split(':', 1)[1]})"
+        if self.op.startswith("hll:"):
+            return f"approx_count_distinct({self.child})"
         disp = self._DISPLAY.get(self.op, self.op)
         if self.child is None:
             return f"{disp}(*)"
@@ -60,6 +62,8 @@ class AggExpr:
             return DType.list_(ct)
         if self.op.startswith("percentile:"):
             return FLOAT64
+        if self.op.startswith("hll:"):
+            return INT64
         raise NotImplementedError(f"agg {self.op}")
 
     def __str__(self):
@@ -137,10 +141,18 @@ def bit_xor(e) -> AggExpr:
     return AggExpr("bit_xor", e)
 
 
-def approx_count_distinct(e) -> AggExpr:
-    """Served by the exact two-level distinct rewrite (always at least as
-    accurate as the reference's HyperLogLog++ approximation)."""
-    return AggExpr("count", e, distinct=True)
+def approx_count_distinct(e, rsd: float = 0.05) -> AggExpr:
+    """HyperLogLog++ sketch (k_gb_hll over xxHash64, Spark precision
+    p = ceil(2*log2(1.106/rsd))). Estimates use the standard HLL
+    small-range correction (no empirical bias tables), so values can
+    differ slightly from Spark's — the same class of divergence the
+    reference gates behind incompatibleOps. For exact counts use
+    count_distinct()."""
+    import math
+
+    assert 0.0 < rsd < 1.0
+    p = max(4, math.ceil(2.0 * math.log2(1.106 / rsd)))
+    return AggExpr(f"hll:{p}", e)
 
 
 def percentile(e, p: float) -> AggExpr:

@@ -1076,6 +1076,10 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                 out_cols.append(_make(res, gv if not gv.all() else None,
                                       out_dtype))
             continue
+        if op.startswith("hll:"):
+            est = hll_groups(vc, codes, ngroups, int(op.split(":", 1)[1]))
+            out_cols.append(_make(est, None, out_dtype))
+            continue
         if op.startswith("percentile:"):
             pq = float(op.split(":", 1)[1])
             res = np.zeros(ngroups)
@@ -1438,3 +1442,147 @@ def and_parent_validity(kid: Column, parent: Column) -> Column:
     vals = kid.to_pylist()
     out = [v if ok else None for v, ok in zip(vals, both)]
     return Column.from_pylist(out, kid.dtype)
+
+
+# ---- xxHash64 (canonical; CPU mirror of hash.hip) ------------------------
+
+_XXP1 = 0x9E3779B185EBCA87
+_XXP2 = 0xC2B2AE3D27D4EB4F
+_XXP3 = 0x165667B19E3779F9
+_XXP4 = 0x85EBCA77C2B2AE63
+_XXP5 = 0x27D4EB2F165667C5
+_M64 = (1 << 64) - 1
+
+
+def _rotl64(x, r):
+    return ((x << r) | (x >> (64 - r))) & _M64
+
+
+def _xx_round(acc, inp):
+    acc = (acc + inp * _XXP2) & _M64
+    return (_rotl64(acc, 31) * _XXP1) & _M64
+
+
+def _xx_avalanche(h):
+    h ^= h >> 33
+    h = (h * _XXP2) & _M64
+    h ^= h >> 29
+    h = (h * _XXP3) & _M64
+    h ^= h >> 32
+    return h
+
+
+def xxh64_long(v: int, seed: int) -> int:
+    h = (seed + _XXP5 + 8) & _M64
+    h ^= _xx_round(0, v & _M64)
+    h = (_rotl64(h, 27) * _XXP1 + _XXP4) & _M64
+    return _xx_avalanche(h)
+
+
+def xxh64_bytes(data: bytes, seed: int) -> int:
+    import struct as st
+
+    n = len(data)
+    p = 0
+    if n >= 32:
+        v1 = (seed + _XXP1 + _XXP2) & _M64
+        v2 = (seed + _XXP2) & _M64
+        v3 = seed & _M64
+        v4 = (seed - _XXP1) & _M64
+        while p + 32 <= n:
+            v1 = _xx_round(v1, st.unpack_from("<Q", data, p)[0])
+            v2 = _xx_round(v2, st.unpack_from("<Q", data, p + 8)[0])
+            v3 = _xx_round(v3, st.unpack_from("<Q", data, p + 16)[0])
+            v4 = _xx_round(v4, st.unpack_from("<Q", data, p + 24)[0])
+            p += 32
+        h = (_rotl64(v1, 1) + _rotl64(v2, 7) + _rotl64(v3, 12)
+             + _rotl64(v4, 18)) & _M64
+        for v in (v1, v2, v3, v4):
+            h = ((h ^ _xx_round(0, v)) * _XXP1 + _XXP4) & _M64
+    else:
+        h = (seed + _XXP5) & _M64
+    h = (h + n) & _M64
+    while p + 8 <= n:
+        h ^= _xx_round(0, st.unpack_from("<Q", data, p)[0])
+        h = (_rotl64(h, 27) * _XXP1 + _XXP4) & _M64
+        p += 8
+    if p + 4 <= n:
+        h ^= (st.unpack_from("<I", data, p)[0] * _XXP1) & _M64
+        h = (_rotl64(h, 23) * _XXP2 + _XXP3) & _M64
+        p += 4
+    while p < n:
+        h ^= (data[p] * _XXP5) & _M64
+        h = (_rotl64(h, 11) * _XXP1) & _M64
+        p += 1
+    return _xx_avalanche(h)
+
+
+def _xxhash64_rows(cols, seed: int = 42) -> np.ndarray:
+    import math
+    import struct as st
+
+    n = cols[0].size
+    out = np.full(n, seed, dtype=np.uint64)
+    for c in cols:
+        av = _valid(c)
+        if c.dtype.id is TypeId.STRING:
+            vals = c.to_pylist()
+            for i in range(n):
+                if av[i]:
+                    out[i] = xxh64_bytes(vals[i].encode(), int(out[i]))
+        else:
+            a = _vals(c)
+            for i in range(n):
+                if not av[i]:
+                    continue
+                v = a[i]
+                if c.dtype.is_floating:
+                    d = float(v)
+                    if math.isnan(d):
+                        d = math.nan
+                    if d == 0.0:
+                        d = 0.0
+                    bits = st.unpack("<Q", st.pack("<d", d))[0]
+                else:
+                    bits = int(v) & _M64
+                out[i] = xxh64_long(bits, int(out[i]))
+    return out
+
+
+def xxhash64(cols, seed: int = 42) -> Column:
+    h = _xxhash64_rows(cols, seed).astype(np.int64)
+    return _make(h, None, DType.int64())
+
+
+def _hll_estimate_host(regs_np) -> np.ndarray:
+    """Standard HLL estimate with the small-range linear-counting
+    correction (no empirical bias tables — estimates differ slightly from
+    Spark's HLL++ constants; documented divergence, like the reference's
+    incompat-gated approximations)."""
+    g, m = regs_np.shape
+    alpha = 0.7213 / (1 + 1.079 / m)
+    E = alpha * m * m / np.power(2.0, -regs_np.astype(np.float64)).sum(1)
+    zeros = (regs_np == 0).sum(1)
+    lc = m * np.log(np.maximum(m / np.maximum(zeros, 1), 1.0))
+    small = (E <= 2.5 * m) & (zeros > 0)
+    return np.rint(np.where(small, lc, E)).astype(np.int64)
+
+
+def hll_groups(vc: Column, codes: np.ndarray, ngroups: int,
+               p: int) -> np.ndarray:
+    """HLL register build + estimate (CPU mirror of k_gb_hll)."""
+    m = 1 << p
+    regs = np.zeros((ngroups, m), dtype=np.uint8)
+    av = _valid(vc)
+    hashes = _xxhash64_rows([vc])
+    for i in range(vc.size):
+        if not av[i]:
+            continue
+        h = int(hashes[i])
+        idx = h >> (64 - p)
+        w = (h << p) & _M64
+        rho = (64 - p + 1) if w == 0 else (64 - w.bit_length() + 1)
+        g = codes[i]
+        if regs[g, idx] < rho:
+            regs[g, idx] = rho
+    return _hll_estimate_host(regs)

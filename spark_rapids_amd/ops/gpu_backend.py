@@ -1421,6 +1421,11 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                                  float(op.split(":", 1)[1]), s)
             allocs.append(("collect", out_dtype, False, col, None))
             continue
+        if op.startswith("hll:"):
+            col = _gb_hll(vc, row_gid, selp, n, ngroups,
+                          int(op.split(":", 1)[1]), s)
+            allocs.append(("collect", out_dtype, False, col, None))
+            continue
         acc_is_double = out_dtype.is_floating or (
             vc is not None and vc.dtype.is_floating)
         acc = torch.empty(max(nrep * ngroups, 1),
@@ -1956,3 +1961,54 @@ def and_parent_validity(kid: Column, parent: Column) -> Column:
         if kid.validity is not None else parent.validity.clone()
     return Column(kid.dtype, n, kid.data, mask, kid.offsets,
                   null_count=None, child=kid.child)
+
+
+def _xxhash64_tensor(cols: List[Column], seed: int = 42,
+                     sel: Optional[torch.Tensor] = None,
+                     n_out: Optional[int] = None) -> torch.Tensor:
+    """Column-chained xxHash64 (Hash.xxhash64 analogue; fixed-width
+    values widened to 8 bytes, floats normalized, strings over bytes)."""
+    n = cols[0].size if sel is None else (n_out if n_out is not None
+                                          else sel.numel())
+    s = _stream()
+    seeds = torch.full((max(n, 1),), seed, dtype=torch.int64,
+                       device="cuda")[:n]
+    selp = 0 if sel is None else sel.data_ptr()
+    for c in cols:
+        if n == 0:
+            break
+        if c.dtype.id is TypeId.STRING:
+            ext.xxhash64_str(c.offsets.data_ptr(), c.data.data_ptr(),
+                             _ptr(c.validity), selp, seeds.data_ptr(), n, s)
+        else:
+            ext.xxhash64_col(_HASH_KIND[c.dtype.id], _ht(c.dtype),
+                             c.data.data_ptr(), _ptr(c.validity), selp,
+                             seeds.data_ptr(), n, s)
+    return seeds
+
+
+def xxhash64(cols: List[Column], seed: int = 42) -> Column:
+    t = _xxhash64_tensor(cols, seed)
+    return Column(DType.int64(), cols[0].size, t, None, null_count=0)
+
+
+def _gb_hll(vc: Column, row_gid: torch.Tensor, selp, n: int, ngroups: int,
+            p: int, s) -> Column:
+    import numpy as np
+
+    sel_t = None
+    if selp:
+        raise NotImplementedError("approx_count_distinct under filter "
+                                  "fusion")
+    from .cpu_backend import _hll_estimate_host
+
+    hashes = _xxhash64_tensor([vc], 42)
+    m = 1 << p
+    regs = torch.zeros(max(ngroups * m, 1), dtype=torch.uint8,
+                       device="cuda")
+    if n:
+        ext.gb_hll(hashes.data_ptr(), _ptr(vc.validity),
+                   row_gid.data_ptr(), 0, regs.data_ptr(), p, n, s)
+    regs_np = regs.cpu().numpy().reshape(ngroups, m)
+    est = _hll_estimate_host(regs_np)
+    return Column.from_numpy(est, DType.int64()).cuda()

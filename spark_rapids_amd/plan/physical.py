@@ -235,7 +235,8 @@ def _lower_aggs(aggs: List[AggExpr], in_schema: Schema):
             merge.extend(["sum", "sum"])
             final.append(("div", js, js + 1, ct))
         elif a.op in ("collect_list", "collect_set") \
-                or a.op.startswith("percentile:"):
+                or a.op.startswith("percentile:") \
+                or a.op.startswith("hll:"):
             value_exprs.append(a.child)
             j = len(partial)
             partial.append((a.op, len(value_exprs) - 1,
@@ -292,7 +293,8 @@ class HashAggregateExec(PhysicalExec):
         nkeys = len(self.group_exprs)
         value_exprs, partial, merge_ops, final = _lower_aggs(self.aggs, in_schema)
         if any(op in ("collect_list", "collect_set")
-               or op.startswith("percentile:") for op, _, _ in partial):
+               or op.startswith("percentile:")
+               or op.startswith("hll:") for op, _, _ in partial):
             yield from self._execute_single_pass(
                 source, fused_condition, in_schema, value_exprs, partial,
                 final)
