@@ -134,6 +134,16 @@ MEM_SPILL_WATERMARK = float_conf(
     "Pool-usage fraction above which spillable batches are proactively "
     "moved to host (reference analogue: spill from the RMM event handler "
     "before allocations fail).")
+OPTIMIZER_ENABLED = bool_conf(
+    "spark.rapids.sql.optimizer.enabled", False,
+    "Cost-based optimizer (CostBasedOptimizer analogue): estimate CPU vs "
+    "GPU cost from scan cardinalities, per-operator costs and transfer "
+    "rates; when the GPU estimate (incl. H2D transfer) exceeds the CPU "
+    "estimate the plan stays on the CPU. Off by default like the "
+    "reference.")
+OPTIMIZER_EXPLAIN = bool_conf(
+    "spark.rapids.sql.optimizer.explain", False,
+    "Print the cost-based optimizer's estimates for every plan.")
 SHUFFLE_WAVE_BYTES = bytes_conf(
     "spark.rapids.shuffle.wave.bytes", 1 << 30,
     "Bytes budget per exchange wave: a shuffle whose serialized send "

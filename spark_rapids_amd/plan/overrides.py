@@ -325,6 +325,19 @@ def plan_physical(node: L.LogicalPlan, conf: RapidsConf,
 
             node = prune_columns(node)
     gpu_wanted = conf.sql_enabled and _gpu_available()
+    if top and gpu_wanted:
+        from ..config import OPTIMIZER_ENABLED, OPTIMIZER_EXPLAIN
+
+        if conf.get(OPTIMIZER_ENABLED):
+            from .costing import evaluate
+
+            keep, note = evaluate(node)
+            if conf.get(OPTIMIZER_EXPLAIN):
+                print(f"!CBO {note} -> "
+                      f"{'GPU' if keep else 'CPU'}")
+            if not keep:
+                tagger.notes.append(TagReason("plan", [note]))
+                gpu_wanted = False
     exec_ = _convert(node, conf, tagger, gpu_wanted)
     if top:
         exec_ = _ensure_device(exec_, "cpu")  # results surface on host

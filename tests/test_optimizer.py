@@ -79,3 +79,52 @@ def test_pushdown_conf_off(cpu):
     q = fact.join(dim, on="id", right_on=["did"]).filter(col("cat") < 1)
     tree = q.physical_plan().tree_string()
     assert tree.index("Filter") < tree.index("HashJoin")
+
+
+def test_cost_based_optimizer_veto_small_plan():
+    """CBO (default off) vetoes GPU placement for tiny inputs when
+    enabled (CostBasedOptimizer analogue)."""
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, sum_
+
+    s = sr.Session({"spark.rapids.sql.optimizer.enabled": True})
+    df = s.create_dataframe({"k": [1, 2], "v": [1.0, 2.0]})
+    q = df.group_by("k").agg(sum_(col("v")))
+    tree = q.physical_plan().tree_string()
+    assert "Gpu" not in tree, tree
+    assert sorted(q.collect()) == [(1, 1.0), (2, 2.0)]
+    # default off: no veto machinery in the way
+    s2 = sr.Session({"spark.rapids.sql.enabled": False})
+    assert sorted(s2.create_dataframe({"k": [1]}).collect()) == [(1,)]
+
+
+def test_cost_based_optimizer_estimates():
+    from spark_rapids_amd.plan.costing import estimate_rows, evaluate
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, sum_
+
+    s = sr.Session({"spark.rapids.sql.enabled": False})
+    df = s.create_dataframe({"k": list(range(1000)),
+                             "v": [0.0] * 1000})
+    plan = df.filter(col("k") > 10).group_by("k") \
+        .agg(sum_(col("v"))).plan
+    assert estimate_rows(plan) > 0
+    keep, note = evaluate(plan)
+    assert "est cpu" in note
+
+
+@pytest.mark.gpu
+def test_gpu_cost_based_optimizer_veto():
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, sum_
+
+    s = sr.Session({"spark.rapids.sql.optimizer.enabled": True})
+    tiny = s.create_dataframe({"k": [1, 2], "v": [1.0, 2.0]})
+    q = tiny.group_by("k").agg(sum_(col("v")))
+    assert "Gpu" not in q.physical_plan().tree_string()
+    assert sorted(q.collect()) == [(1, 1.0), (2, 2.0)]
+    # a large plan stays on GPU despite the optimizer
+    big = s.create_dataframe({"k": list(range(200_000)) * 5,
+                              "v": [0.5] * 1_000_000})
+    q2 = big.group_by("k").agg(sum_(col("v")))
+    assert "Gpu" in q2.physical_plan().tree_string()
