@@ -524,9 +524,12 @@ class Session:
         if reader == "AUTO":
             reader = "GPU_DECODE" if (torch.cuda.is_available()
                                       and self.conf.sql_enabled) else "CPU"
+        from .config import BATCH_SIZE_BYTES
+
         src = ParquetTable(path, columns=columns, reader=reader,
                            prefetch_threads=self.conf.get(PARQUET_MT_THREADS),
-                           replicated=replicated)
+                           replicated=replicated,
+                           chunk_bytes=self.conf.get(BATCH_SIZE_BYTES))
         return DataFrame(self, L.Scan(src, src.schema, f"parquet:{path}"))
 
     def write_parquet(self, df: DataFrame, path: str,

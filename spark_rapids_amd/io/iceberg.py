@@ -138,7 +138,7 @@ class IcebergTable:
 
         files = self._pq._my_files()
         for f in files:
-            batch = self._pq._read_one(f)
+            (batch,) = self._pq._read_one(f)
             dels = self.deletes.get(f)
             if dels:
                 import numpy as np

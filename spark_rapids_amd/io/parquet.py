@@ -216,7 +216,14 @@ class ParquetTable:
 
     def __init__(self, path: str, num_partitions: int = 0,
                  columns: Optional[List[str]] = None, reader: str = "CPU",
-                 prefetch_threads: int = 4, replicated: bool = False):
+                 prefetch_threads: int = 4, replicated: bool = False,
+                 chunk_bytes: int = 0):
+        # chunk_bytes > 0: a file whose decoded size exceeds the budget is
+        # read ROW GROUP by row group as separate batches instead of one
+        # concatenated file batch (the chunked-reader analogue:
+        # ParquetChunkedReader / GpuParquetScan.scala:3401). The coalesce
+        # exec above re-merges toward batchSizeBytes.
+        self.chunk_bytes = chunk_bytes
         # replicated=True: every rank scans ALL files (dimension tables in
         # a distributed star-schema query; broadcast-join build sides skip
         # the exchange, plan/logical.py is_replicated)
@@ -248,6 +255,7 @@ class ParquetTable:
         t.prefetch_threads = self.prefetch_threads
         t.schema = self.schema
         t.rg_predicate = getattr(self, "rg_predicate", None)
+        t.chunk_bytes = getattr(self, "chunk_bytes", 0)
         return t
 
     def with_columns(self, names: List[str]) -> "ParquetTable":
@@ -267,28 +275,60 @@ class ParquetTable:
         t.rg_predicate = list(triples)
         return t
 
-    def _read_one(self, path: str) -> ColumnBatch:
+    def _read_one(self, path: str) -> List[ColumnBatch]:
         from .filecache import get_cached, put_cached
 
         cached = get_cached(path, self.columns)
         if cached is not None:
-            return cached
-        batch = self._read_one_uncached(path)
-        put_cached(path, self.columns, batch)
-        return batch
+            return cached if isinstance(cached, list) else [cached]
+        batches = self._read_one_uncached(path)
+        put_cached(path, self.columns, batches)
+        return batches
 
-    def _read_one_uncached(self, path: str) -> ColumnBatch:
+    def _rg_chunks(self, path: str, keep):
+        """Split a file's surviving row groups into chunked reads whose
+        estimated decoded bytes respect chunk_bytes (the chunked-reader
+        analogue). Returns None for the single-read fast path."""
+        budget = getattr(self, "chunk_bytes", 0)
+        if not budget:
+            return None
+        from .parquet_gpu import _file_meta
+
+        md, _, _ = _file_meta(path)
+        rgs = [r for r in range(md.num_row_groups)
+               if keep is None or r in keep]
+        if len(rgs) <= 1:
+            return None
+        sizes = [md.row_group(r).total_byte_size for r in rgs]
+        if sum(sizes) <= budget:
+            return None
+        chunks, cur, cur_b = [], [], 0
+        for r, sz in zip(rgs, sizes):
+            if cur and cur_b + sz > budget:
+                chunks.append(cur)
+                cur, cur_b = [], 0
+            cur.append(r)
+            cur_b += sz
+        if cur:
+            chunks.append(cur)
+        return chunks
+
+    def _read_one_uncached(self, path: str) -> List[ColumnBatch]:
         keep = None
         pred = getattr(self, "rg_predicate", None)
         if pred:
             keep = _rg_keep(path, pred)
+        chunks = self._rg_chunks(path, keep)
         if self.reader == "GPU_DECODE":
             try:
                 from .parquet_gpu import read_parquet_gpu
 
-                out = read_parquet_gpu(path,
-                                       [f.name for f in self.schema.fields],
-                                       keep_rgs=keep)
+                cols = [f.name for f in self.schema.fields]
+                if chunks is not None:
+                    out = [read_parquet_gpu(path, cols, keep_rgs=set(ch))
+                           for ch in chunks]
+                else:
+                    out = [read_parquet_gpu(path, cols, keep_rgs=keep)]
                 SCAN_STATS["gpu_files"] += 1
                 return out
             except NotImplementedError as e:
@@ -298,6 +338,11 @@ class ParquetTable:
                 SCAN_STATS["last_fallback"] = f"{path}: {e}"
         import pyarrow.parquet as pq
 
+        if chunks is not None:
+            pf = pq.ParquetFile(path)
+            return [arrow_table_to_batch(
+                pf.read_row_groups(ch, columns=self.columns))
+                for ch in chunks]
         if keep is not None:
             pf = pq.ParquetFile(path)
             if len(keep) == 0:
@@ -310,9 +355,9 @@ class ParquetTable:
             else:
                 tbl = pf.read_row_groups(sorted(keep),
                                          columns=self.columns)
-            return arrow_table_to_batch(tbl)
+            return [arrow_table_to_batch(tbl)]
         tbl = pq.read_table(path, columns=self.columns)
-        return arrow_table_to_batch(tbl)
+        return [arrow_table_to_batch(tbl)]
 
     def _my_files(self) -> List[str]:
         # distributed scans shard files round-robin across ranks
@@ -327,7 +372,7 @@ class ParquetTable:
         files = self._my_files()
         if len(files) <= 1 or self.prefetch_threads <= 1:
             for f in files:
-                yield self._read_one(f)
+                yield from self._read_one(f)
             return
         import torch
 
@@ -337,7 +382,7 @@ class ParquetTable:
             with cf.ThreadPoolExecutor(self.prefetch_threads) as pool:
                 futures = [pool.submit(self._read_one, f) for f in files]
                 for fut in futures:
-                    yield fut.result()
+                    yield from fut.result()
             return
         # copy/compute overlap: each prefetch worker decodes on its own
         # HIP stream (uploads + decode kernels), so file IO, H2D and
@@ -356,19 +401,19 @@ class ParquetTable:
                 st = side_streams[next(counter) % len(side_streams)]
                 tl.stream = st
             with torch.cuda.stream(st):
-                batch = self._read_one(f)
+                batches = self._read_one(f)
                 ev = torch.cuda.Event()
                 ev.record(st)
-            return batch, ev
+            return batches, ev
 
         with cf.ThreadPoolExecutor(self.prefetch_threads) as pool:
             futures = [pool.submit(read_on_stream, f) for f in files]
             for fut in futures:
-                batch, ev = fut.result()
+                batches, ev = fut.result()
                 # order the consumer (current/default) stream after the
                 # producer stream's work without a host sync
                 ev.wait(torch.cuda.current_stream())
-                yield batch
+                yield from batches
 
 
 def write_parquet(batch: ColumnBatch, schema: Schema, path: str,

@@ -338,3 +338,34 @@ def test_nested_parquet_roundtrip(tmp_path):
     back = s.read_parquet(path).to_pydict()
     assert back["st"] == rows_s
     assert back["ls"] == [[1, 2], [], None]
+
+
+def test_chunked_read_respects_batch_bytes(tmp_path):
+    """A file larger than batchSizeBytes reads as multiple row-group
+    chunks (ParquetChunkedReader analogue); the query result is
+    unchanged."""
+    import numpy as np
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    import spark_rapids_amd as sr
+
+    n = 200_000
+    f = str(tmp_path / "big.parquet")
+    pq.write_table(pa.table({
+        "k": pa.array(np.arange(n, dtype=np.int64)),
+        "v": pa.array(np.arange(n, dtype=np.float64)),
+    }), f, row_group_size=20_000)
+
+    s = sr.Session({"spark.rapids.sql.enabled": False,
+                    "spark.rapids.sql.batchSizeBytes": 512 * 1024})
+    df = s.read_parquet(f)
+    src = df.plan.source
+    batches = list(src.partitions())
+    assert len(batches) > 1, len(batches)  # chunked
+    assert sum(b.num_rows for b in batches) == n
+    (total,) = df.agg(sr.sum_(sr.col("k"))).collect()[0]
+    assert total == n * (n - 1) // 2
+    # big budget: one batch per file
+    s2 = sr.Session({"spark.rapids.sql.enabled": False})
+    assert len(list(s2.read_parquet(f).plan.source.partitions())) == 1
