@@ -52,4 +52,7 @@ def put_cached(path: str, columns, batch: ColumnBatch):
     if k is None:
         return
     with _lock:
-        _cache[k] = batch.cpu()
+        if isinstance(batch, list):
+            _cache[k] = [b.cpu() for b in batch]
+        else:
+            _cache[k] = batch.cpu()
