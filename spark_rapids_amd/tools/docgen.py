@@ -154,6 +154,8 @@ _EXPR_ROWS = [
     ("Sum(DISTINCT)", "rewrite", "GPU", ""),
     ("Percentile", "gb_percentile", "GPU", "exact sorted-groups kernel"),
     ("ApproximatePercentile", "exact rewrite", "GPU", "exact result (superset of t-digest accuracy)"),
+    ("ApproxCountDistinct/HyperLogLogPlusPlus", "k_gb_hll + xxHash64", "GPU", "HLL++ sketch, Spark precision formula; estimates use standard HLL correction (no bias tables)"),
+    ("XxHash64", "k_xxhash64_col/str", "GPU", "canonical XXH64, column-chained seeds"),
     ("BitAndAgg/BitOrAgg/BitXorAgg", "bit_and/or/xor", "GPU", ""),
     ("PivotFirst", "-", "CPU", "via host fallback"),
     # windows
