@@ -272,9 +272,6 @@ class Tagger:
                     if op not in ("sum", "count", "mean", "min", "max"):
                         reasons.append(
                             f"range-frame {op} window not on GPU yet")
-                    elif w.spec.descending[0]:
-                        reasons.append(
-                            "range frame over descending order on CPU")
                     elif not okt.is_numeric or okt.id is TypeId.DECIMAL128:
                         reasons.append(
                             f"range frame over {okt} order key on CPU")

@@ -313,9 +313,22 @@ class WindowExec(PhysicalExec):
                 okey = table.columns[cs.index(self.spec.order_by[0])]
                 of = gb.cast(Column(okey.dtype, n, okey.data, None,
                                     null_count=0), FLOAT64)
+                of_t = of.data
+                if okey.validity is not None:
+                    ov_u8 = torch.empty(n, dtype=torch.uint8, device="cuda")
+                    ext.mask_expand(okey.validity.data_ptr(),
+                                    ov_u8.data_ptr(), False, n, s)
+                    of_t = torch.where(
+                        ov_u8.bool(), of_t,
+                        torch.full_like(of_t, float("-inf")))
+                if self.spec.descending[0]:
+                    # frame offsets follow the SORT direction: negate to an
+                    # ascending axis and the same (lo, hi) apply directly
+                    # (matches the CPU searchsorted path above)
+                    of_t = torch.neg(of_t)
                 a_t = torch.empty(n, dtype=torch.int32, device="cuda")
                 b_t = torch.empty(n, dtype=torch.int32, device="cuda")
-                ext.range_bounds(of.data.data_ptr(),
+                ext.range_bounds(of_t.data_ptr(),
                                  seg_start_col.data.data_ptr(),
                                  seg_end.data.data_ptr(),
                                  float(lo_v if lo_v is not None else 0.0),

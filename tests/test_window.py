@@ -278,6 +278,58 @@ class TestRangeFrames:
             if x is not None:
                 assert x == pytest.approx(y, rel=1e-9)
 
+    def test_range_desc_null_keys_cpu(self, cpu_session):
+        # null order keys sort last under descending and form a peer
+        # group with an unbounded-feeling key (-inf on the negated axis)
+        df = cpu_session.create_dataframe({
+            "p": [1, 1, 1, 1], "t": [3.0, 2.0, None, None],
+            "v": [1.0, 2.0, 4.0, 8.0]})
+        out = df.with_column("r", win_sum(col("v")).over(
+            partition_by=["p"], order_by=["t"], descending=[True],
+            range_between=(-1.0, 0.0))).to_pydict()["r"]
+        # t=3 -> keys in [3,4] = {3}: 1; t=2 -> [2,3] = {2,3}: 3;
+        # nulls are peers of each other only: 12, 12
+        assert out == [1.0, 3.0, 12.0, 12.0]
+
+    @pytest.mark.gpu
+    def test_gpu_range_desc_matches_cpu(self):
+        sg = sr.Session()
+        sc = sr.Session({"spark.rapids.sql.enabled": False})
+        rng = np.random.default_rng(23)
+        n = 3000
+        data = {
+            "p": [int(v) for v in rng.integers(0, 7, n)],
+            "t": [float(v) if i % 11 else None
+                  for i, v in enumerate(rng.uniform(0, 50, n))],
+            "v": [float(v) if i % 13 else None
+                  for i, v in enumerate(rng.uniform(0, 10, n))]}
+
+        def q(s):
+            df = s.create_dataframe(data)
+            df = df.with_column("r", win_sum(col("v")).over(
+                partition_by=["p"], order_by=["t"], descending=[True],
+                range_between=(-2.0, 2.0)))
+            df = df.with_column("c", win_count(col("v")).over(
+                partition_by=["p"], order_by=["t"], descending=[True],
+                range_between=(None, 0.0)))
+            df = df.with_column("m", win_max(col("v")).over(
+                partition_by=["p"], order_by=["t"], descending=[True],
+                range_between=(-3.0, 0.0)))
+            return df.to_pydict()
+
+        qg = sr.Session().create_dataframe(data).with_column(
+            "r", win_sum(col("v")).over(
+                partition_by=["p"], order_by=["t"], descending=[True],
+                range_between=(-2.0, 2.0)))
+        assert "GpuWindow" in qg.physical_plan().tree_string()
+        g, c = q(sg), q(sc)
+        assert g["c"] == c["c"]
+        assert g["m"] == c["m"]
+        for x, y in zip(g["r"], c["r"]):
+            assert (x is None) == (y is None)
+            if x is not None:
+                assert x == pytest.approx(y, rel=1e-9)
+
 
 def test_ntile_nth_value_cpu(cpu_session):
     from spark_rapids_amd import ntile, nth_value
