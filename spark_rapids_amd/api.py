@@ -101,14 +101,19 @@ class DataFrame:
         return DataFrame(self.session, L.Aggregate([], list(aggs), self.plan))
 
     def join(self, other: "DataFrame", on: Union[str, Sequence[str]],
-             how: str = "inner", right_on: Optional[Sequence[str]] = None) -> "DataFrame":
+             how: str = "inner", right_on: Optional[Sequence[str]] = None,
+             condition=None) -> "DataFrame":
+        """Equi-join, optionally with an extra non-equi `condition`
+        (reference: conditional/mixed hash joins compiled to an AST).
+        The condition is an expression over left columns followed by
+        right columns; duplicate names resolve to the left side."""
         if isinstance(on, str):
             on = [on]
         using = right_on is None
         r_on = list(right_on) if right_on is not None else list(on)
         return DataFrame(self.session,
                          L.Join(self.plan, other.plan, list(on), r_on, how,
-                                using=using))
+                                using=using, condition=condition))
 
     def cross_join(self, other: "DataFrame") -> "DataFrame":
         """Cartesian product; combine with filter() for non-equi joins

@@ -119,12 +119,15 @@ class Aggregate(LogicalPlan):
 class Join(LogicalPlan):
     def __init__(self, left: LogicalPlan, right: LogicalPlan,
                  left_on: List[str], right_on: List[str], how: str = "inner",
-                 using: bool = False):
+                 using: bool = False, condition=None):
         self.left = left
         self.right = right
         self.left_on = left_on
         self.right_on = right_on
         self.how = how
+        # optional non-equi predicate over (left ++ right) columns applied
+        # to matched pairs (reference: conditional/mixed hash joins)
+        self.condition = condition
         # USING-join semantics (join by shared column names): the
         # duplicate right key columns are dropped for inner/left joins,
         # matching Spark's df.join(other, "k"). Full outer keeps both

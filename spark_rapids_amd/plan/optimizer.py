@@ -113,7 +113,8 @@ def push_filters(plan: L.LogicalPlan) -> L.LogicalPlan:
             new_join = L.CrossJoin(left, right)
         else:
             new_join = L.Join(left, right, child.left_on, child.right_on,
-                              child.how, using=child.using)
+                              child.how, using=child.using,
+                              condition=child.condition)
         # re-run on the pushed filters (stacked joins push further down)
         new_join = _with_children(
             new_join, [push_filters(c) for c in new_join.children])
@@ -226,14 +227,17 @@ def prune_columns(plan: L.LogicalPlan,
         if needed is None:
             lneed = rneed = None
         else:
-            lneed = (needed & lnames) | set(plan.left_on)
-            rneed = (needed & rnames) | set(plan.right_on)
+            cond_refs: Set[str] = set()
+            if plan.condition is not None:
+                _refs(plan.condition, cond_refs)
+            lneed = ((needed | cond_refs) & lnames) | set(plan.left_on)
+            rneed = ((needed | cond_refs) & rnames) | set(plan.right_on)
         left = prune_columns(plan.left, lneed)
         left = _project_to(left, lneed)
         right = prune_columns(plan.right, rneed)
         right = _project_to(right, rneed)
         return L.Join(left, right, plan.left_on, plan.right_on, plan.how,
-                      using=plan.using)
+                      using=plan.using, condition=plan.condition)
     if isinstance(plan, L.Sort):
         child_needed = None if needed is None else set(needed) | set(plan.keys)
         return L.Sort(prune_columns(plan.child, child_needed), plan.keys,

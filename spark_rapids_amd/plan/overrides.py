@@ -255,6 +255,11 @@ class Tagger:
                     reasons.append(f"join key {k}: {r}")
             if node.how not in ("inner", "left", "semi", "anti", "full"):
                 reasons.append(f"join type {node.how} not on GPU")
+            if node.condition is not None:
+                from ..column import Schema as _Schema
+
+                pair = _Schema(list(ls.fields) + list(rs.fields))
+                reasons += self.expr_reasons(node.condition, pair)
         elif isinstance(node, L.MapBatches):
             reasons.append("python map_batches runs on CPU (UDF bridge)")
         elif isinstance(node, L.Window):
@@ -443,7 +448,7 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
                               broadcast_threshold=conf.get(BROADCAST_THRESHOLD),
                               sub_partition_bytes=conf.get(
                                   JOIN_SUBPARTITION_BYTES),
-                              using=node.using)
+                              using=node.using, condition=node.condition)
     if isinstance(node, L.MapBatches):
         return P.MapBatchesExec(node.fn, _ensure_device(kids[0], "cpu"),
                                 node.schema())

@@ -492,7 +492,7 @@ class HashJoinExec(PhysicalExec):
                  schema: Schema, right_replicated: bool = True,
                  broadcast_threshold: int = 512 << 20,
                  sub_partition_bytes: int = 1 << 30,
-                 using: bool = False):
+                 using: bool = False, condition=None):
         super().__init__(device, schema, [left, right])
         self.left_on = left_on
         self.right_on = right_on
@@ -501,6 +501,14 @@ class HashJoinExec(PhysicalExec):
         self.broadcast_threshold = broadcast_threshold
         self.sub_partition_bytes = sub_partition_bytes
         self.using = using
+        # extra non-equi predicate over (left ++ right) columns, applied to
+        # the candidate pairs of the equi probe (reference analogue: the
+        # AST-compiled join condition of ConditionalHashJoinIterator /
+        # mixed joins, GpuHashJoin.scala:1653 — there the compiled AST
+        # runs during the probe; here the condition is evaluated
+        # vectorized over the gathered pair batch and the gather maps are
+        # compacted, an equivalent device-side dataflow)
+        self.condition = condition
         self._strategy = "local"
 
     def _local_or_empty(self, batches: List[ColumnBatch], schema: Schema):
@@ -585,6 +593,10 @@ class HashJoinExec(PhysicalExec):
                     else:
                         yield self._left_with_null_right(lbatch)
                 continue
+            if self.condition is not None:
+                yield from self._emit_conditional(lbatch, rtable, lkidx,
+                                                  rkidx, right_matched)
+                continue
             lmap, rmap = ops.join_gather_maps(lbatch, rtable, lkidx, rkidx,
                                               self.how, right_matched)
             if self.how in ("semi", "anti"):
@@ -647,22 +659,95 @@ class HashJoinExec(PhysicalExec):
                 continue
             right_matched = self._fresh_matched(rb) \
                 if self.how == "full" else None
-            lmap, rmap = ops.join_gather_maps(lb, rb, lkidx, rkidx,
-                                              self.how, right_matched)
-            if self.how in ("semi", "anti"):
-                out = ops.gather(lb, lmap)
-                if out.num_rows:
-                    yield out
+            if self.condition is not None:
+                yield from self._emit_conditional(lb, rb, lkidx, rkidx,
+                                                  right_matched)
             else:
-                lout = ops.gather(lb, lmap)
-                rout = ops.gather(rb, rmap)
-                if lout.num_rows:
-                    yield ColumnBatch(lout.columns + self._right_out(rout),
-                                      lout.num_rows)
+                lmap, rmap = ops.join_gather_maps(lb, rb, lkidx, rkidx,
+                                                  self.how, right_matched)
+                if self.how in ("semi", "anti"):
+                    out = ops.gather(lb, lmap)
+                    if out.num_rows:
+                        yield out
+                else:
+                    lout = ops.gather(lb, lmap)
+                    rout = ops.gather(rb, rmap)
+                    if lout.num_rows:
+                        yield ColumnBatch(
+                            lout.columns + self._right_out(rout),
+                            lout.num_rows)
             if self.how == "full":
                 extra = self._unmatched_right(rb, right_matched, left_schema)
                 if extra is not None and extra.num_rows:
                     yield extra
+
+    def _pair_schema(self) -> Schema:
+        """Namespace the join condition resolves against: left fields
+        followed by ALL right fields (duplicate names resolve left-first;
+        rename before joining for unambiguous references, as in Spark)."""
+        ls, rs = self.children[0].schema, self.children[1].schema
+        return Schema(list(ls.fields) + list(rs.fields))
+
+    def _conditional_pairs(self, lb: ColumnBatch, rb: ColumnBatch,
+                           lkidx, rkidx):
+        """Equi-probe pair maps compacted by the join condition."""
+        lmap, rmap = ops.join_gather_maps(lb, rb, lkidx, rkidx, "inner",
+                                          None)
+        if lmap.size == 0:
+            return lmap, rmap
+        lout = ops.gather(lb, lmap)
+        rout = ops.gather(rb, rmap)
+        pair = ColumnBatch(list(lout.columns) + list(rout.columns),
+                           lout.num_rows)
+        mask = self.condition.eval(pair, self._pair_schema())
+        maps = ops.apply_boolean_mask(
+            ColumnBatch([lmap, rmap], lmap.size), mask)
+        return maps.columns[0], maps.columns[1]
+
+    def _left_hit_mask(self, n: int, lmap_f: Column,
+                       invert: bool = False) -> Column:
+        import torch as _torch
+
+        dev = "cuda" if self.gpu else "cpu"
+        hit = _torch.zeros(n, dtype=_torch.bool, device=dev)
+        if lmap_f.size:
+            hit[lmap_f.data[:lmap_f.size].long()] = True
+        if invert:
+            hit = ~hit
+        return Column(DType.bool_(), n, hit.to(_torch.uint8), None,
+                      null_count=0)
+
+    def _emit_conditional(self, lb: ColumnBatch, rb: ColumnBatch,
+                          lkidx, rkidx, right_matched
+                          ) -> Iterator[ColumnBatch]:
+        """Reconstruct each join type from condition-filtered pairs."""
+        lmap_f, rmap_f = self._conditional_pairs(lb, rb, lkidx, rkidx)
+        how = self.how
+        if how in ("semi", "anti"):
+            mask = self._left_hit_mask(lb.num_rows, lmap_f,
+                                       invert=(how == "anti"))
+            out = ops.apply_boolean_mask(lb, mask)
+            if out.num_rows:
+                yield out
+            return
+        if how == "full" and right_matched is not None and rmap_f.size:
+            import torch as _torch
+
+            idx = rmap_f.data[:rmap_f.size].long()
+            if isinstance(right_matched, _torch.Tensor):
+                right_matched[idx] = 1
+            else:
+                right_matched[idx.cpu().numpy()] = True
+        if lmap_f.size:
+            lout = ops.gather(lb, lmap_f)
+            rout = ops.gather(rb, rmap_f)
+            yield ColumnBatch(list(lout.columns) + self._right_out(rout),
+                              lout.num_rows)
+        if how in ("left", "full"):
+            miss = self._left_hit_mask(lb.num_rows, lmap_f, invert=True)
+            lrest = ops.apply_boolean_mask(lb, miss)
+            if lrest.num_rows:
+                yield self._left_with_null_right(lrest)
 
     def _fresh_matched(self, rb: ColumnBatch):
         import numpy as _np

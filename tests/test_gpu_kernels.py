@@ -573,3 +573,39 @@ def test_graft_smoke_entry():
     import __graft_entry__
 
     __graft_entry__.smoke()
+
+
+@pytest.mark.parametrize("how", ["inner", "left", "semi", "anti", "full"])
+def test_conditional_join_gpu_matches_cpu(how):
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col
+
+    rng = np.random.default_rng(31)
+    nl, nr = 20_000, 6_000
+    data_l = {"k": [int(v) if i % 17 else None
+                    for i, v in enumerate(rng.integers(0, 800, nl))],
+              "a": [float(v) for v in rng.uniform(0, 100, nl)]}
+    data_r = {"k": [int(v) if i % 13 else None
+                    for i, v in enumerate(rng.integers(0, 1000, nr))],
+              "b": [float(v) for v in rng.uniform(0, 100, nr)]}
+
+    def q(s):
+        l = s.create_dataframe(data_l)
+        r = s.create_dataframe(data_r)
+        df = l.join(r, on="k", how=how, condition=col("a") < col("b"))
+        return sorted(df.collect(), key=repr)
+
+    sg = sr.Session()
+    qg = sg.create_dataframe(data_l).join(
+        sg.create_dataframe(data_r), on="k", how=how,
+        condition=col("a") < col("b"))
+    assert "GpuHashJoin" in qg.physical_plan().tree_string()
+    g = q(sg)
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert len(g) == len(c), how
+    for gr, cr in zip(g, c):
+        for x, y in zip(gr, cr):
+            if isinstance(y, float) and x is not None:
+                assert x == pytest.approx(y, rel=1e-9), (how, gr, cr)
+            else:
+                assert x == y, (how, gr, cr)

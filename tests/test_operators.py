@@ -232,3 +232,75 @@ def test_sample_deterministic(session):
     assert 700 < a < 1300  # ~10%
     c = df.sample(0.5, seed=7).count()
     assert 4500 < c < 5500
+
+
+class TestConditionalJoins:
+    """Non-equi join conditions over the equi probe (reference:
+    ConditionalHashJoinIterator / mixed joins — AST-compiled condition;
+    here evaluated vectorized over candidate pairs)."""
+
+    def _sides(self, s):
+        left = s.create_dataframe({
+            "k": [1, 1, 2, 2, 3, None],
+            "a": [10, 20, 30, 40, 50, 60]})
+        right = s.create_dataframe({
+            "k": [1, 1, 2, 4],
+            "b": [15, 25, 100, 7]})
+        return left, right
+
+    def test_inner_conditional(self, session):
+        l, r = self._sides(session)
+        out = sorted(l.join(r, on="k", condition=col("a") < col("b"))
+                     .select("a", "b").collect())
+        # k=1 pairs: (10,15)(10,25)(20,25); k=2: none (30,40 < 100 both!)
+        assert out == [(10, 15), (10, 25), (20, 25), (30, 100), (40, 100)]
+
+    def test_left_conditional(self, session):
+        l, r = self._sides(session)
+        out = sorted(l.join(r, on="k", how="left",
+                            condition=col("a") < col("b"))
+                     .select("a", "b").collect())
+        assert out == [(10, 15), (10, 25), (20, 25), (30, 100), (40, 100),
+                       (50, None), (60, None)]
+
+    def test_semi_anti_conditional(self, session):
+        l, r = self._sides(session)
+        semi = sorted(l.join(r, on="k", how="semi",
+                             condition=col("a") < col("b"))
+                      .to_pydict()["a"])
+        anti = sorted(l.join(r, on="k", how="anti",
+                             condition=col("a") < col("b"))
+                      .to_pydict()["a"])
+        assert semi == [10, 20, 30, 40]
+        assert anti == [50, 60]
+
+    def test_full_conditional(self, session):
+        l, r = self._sides(session)
+        out = sorted(l.join(r, on="k", how="full",
+                            condition=col("a") < col("b")).collect(),
+                     key=repr)
+        pairs = [(row[1], row[3]) for row in out]
+        # matched pairs + unmatched left (null b) + unmatched right (b=7
+        # under k=4 never matches; b=15/25/100 all matched)
+        assert sorted(p for p in pairs if None not in p) == \
+            [(10, 15), (10, 25), (20, 25), (30, 100), (40, 100)]
+        assert sorted(p[0] for p in pairs if p[1] is None) == [50, 60]
+        assert sorted(p[1] for p in pairs if p[0] is None) == [7]
+
+    def test_conditional_pruning_keeps_condition_columns(self, session):
+        # column pruning must retain condition-referenced columns that the
+        # final projection drops
+        l, r = self._sides(session)
+        out = sorted(l.join(r, on="k", condition=col("a") < col("b"))
+                     .select("k").to_pydict()["k"])
+        assert out == [1, 1, 1, 2, 2]
+
+    def test_conditional_subpartitioned(self):
+        s = Session({"spark.rapids.sql.enabled": False,
+                     "spark.rapids.sql.join.subPartition.targetBytes": 1})
+        l, r = self._sides(s)
+        out = sorted(l.join(r, on="k", how="left",
+                            condition=col("a") < col("b"))
+                     .select("a", "b").collect())
+        assert out == [(10, 15), (10, 25), (20, 25), (30, 100), (40, 100),
+                       (50, None), (60, None)]
