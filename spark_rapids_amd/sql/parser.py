@@ -187,8 +187,13 @@ class Parser:
                 break
             right = self.parse_table()
             self.expect_kw("ON")
-            lk, rk = self.parse_join_keys()
-            df = df.join(right, on=lk, right_on=rk, how=how)
+            lk, rk, cond = self.parse_join_keys()
+            if not lk:
+                raise ValueError(
+                    "JOIN ON needs at least one equality conjunct "
+                    "(a = b); pure non-equi joins: use CROSS JOIN + WHERE")
+            df = df.join(right, on=lk, right_on=rk, how=how,
+                         condition=cond)
         where = None
         if self.kw("WHERE"):
             where = self.parse_expr()
@@ -324,16 +329,35 @@ class Parser:
         return df
 
     def parse_join_keys(self):
+        """AND-separated conjuncts: `a = b` pairs become equi keys, any
+        other comparison becomes the non-equi join condition (conditional
+        hash join)."""
         lk, rk = [], []
+        cond = None
         while True:
-            a = self.parse_qualified_name()
-            self.expect_op("=")
-            b = self.parse_qualified_name()
-            lk.append(a)
-            rk.append(b)
+            mark = self.i
+            eq = None
+            if self.peek()[0] == "name":
+                try:
+                    a = self.parse_qualified_name()
+                    if self.op("="):
+                        b = self.parse_qualified_name()
+                        nxt = self.peek()
+                        # a = b must end the conjunct (not `a = b + 1`)
+                        if nxt[0] != "op" or nxt[1] in (")", ","):
+                            eq = (a, b)
+                except (ValueError, IndexError):
+                    eq = None
+            if eq is not None:
+                lk.append(eq[0])
+                rk.append(eq[1])
+            else:
+                self.i = mark
+                c = self.parse_expr()
+                cond = c if cond is None else (cond & c)
             if not self.kw("AND"):
                 break
-        return lk, rk
+        return lk, rk, cond
 
     def parse_qualified_name(self) -> str:
         n = self.next()[1]

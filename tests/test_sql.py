@@ -242,3 +242,17 @@ def test_rlike_ascii_digit_class(s):
     s.register("rl_t", df)
     out = s.sql(r"SELECT x FROM rl_t WHERE x RLIKE '^\\d$'").to_pydict()
     assert out["x"] == ["7"]
+
+
+def test_sql_conditional_join(s):
+    l = s.create_dataframe({"k": [1, 1, 2], "a": [10, 20, 30]})
+    r = s.create_dataframe({"k": [1, 2], "b": [15, 5]})
+    s.register("cj_l", l)
+    s.register("cj_r", r)
+    out = s.sql("SELECT a, b FROM cj_l JOIN cj_r ON cj_l.k = cj_r.k "
+                "AND a < b").to_pydict()
+    assert sorted(zip(out["a"], out["b"])) == [(10, 15)]
+    out2 = s.sql("SELECT a, b FROM cj_l LEFT JOIN cj_r ON cj_l.k = cj_r.k "
+                 "AND a < b").to_pydict()
+    assert sorted(zip(out2["a"], out2["b"]), key=repr) == \
+        [(10, 15), (20, None), (30, None)]
