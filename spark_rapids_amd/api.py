@@ -116,9 +116,18 @@ class DataFrame:
                                 using=using, condition=condition))
 
     def cross_join(self, other: "DataFrame") -> "DataFrame":
-        """Cartesian product; combine with filter() for non-equi joins
-        (broadcast-nested-loop-join analogue)."""
+        """Cartesian product; combine with filter() for non-equi INNER
+        joins."""
         return DataFrame(self.session, L.CrossJoin(self.plan, other.plan))
+
+    def join_nl(self, other: "DataFrame", condition,
+                how: str = "inner") -> "DataFrame":
+        """Nested-loop join on an arbitrary condition, no equality keys
+        (reference: GpuBroadcastNestedLoopJoinExec). Supports
+        inner/left/semi/anti/full; pairs are tested in bounded chunks."""
+        return DataFrame(self.session,
+                         L.NestedLoopJoin(self.plan, other.plan, [], [],
+                                          how, condition=condition))
 
     def sort(self, *keys: str, descending: Union[bool, List[bool]] = False) -> "DataFrame":
         ks = list(keys)

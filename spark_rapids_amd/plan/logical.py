@@ -158,6 +158,17 @@ class Join(LogicalPlan):
         return f"Join({self.how})"
 
 
+class NestedLoopJoin(Join):
+    """Join on an arbitrary (non-equi) condition with no equality keys:
+    every (left, right) row pair is tested, chunked to bound memory
+    (reference analogue: GpuBroadcastNestedLoopJoinExec with a compiled
+    AST condition). Same constructor shape as Join so the optimizer can
+    rebuild either with type(plan)(...)."""
+
+    def name(self) -> str:
+        return f"NestedLoopJoin({self.how})"
+
+
 class MapBatches(LogicalPlan):
     """CPU python-function operator (UDF bridge); never places on GPU."""
 

@@ -112,9 +112,10 @@ def push_filters(plan: L.LogicalPlan) -> L.LogicalPlan:
         if isinstance(child, L.CrossJoin):
             new_join = L.CrossJoin(left, right)
         else:
-            new_join = L.Join(left, right, child.left_on, child.right_on,
-                              child.how, using=child.using,
-                              condition=child.condition)
+            new_join = type(child)(left, right, child.left_on,
+                                   child.right_on, child.how,
+                                   using=child.using,
+                                   condition=child.condition)
         # re-run on the pushed filters (stacked joins push further down)
         new_join = _with_children(
             new_join, [push_filters(c) for c in new_join.children])
@@ -236,8 +237,9 @@ def prune_columns(plan: L.LogicalPlan,
         left = _project_to(left, lneed)
         right = prune_columns(plan.right, rneed)
         right = _project_to(right, rneed)
-        return L.Join(left, right, plan.left_on, plan.right_on, plan.how,
-                      using=plan.using, condition=plan.condition)
+        return type(plan)(left, right, plan.left_on, plan.right_on,
+                          plan.how, using=plan.using,
+                          condition=plan.condition)
     if isinstance(plan, L.Sort):
         child_needed = None if needed is None else set(needed) | set(plan.keys)
         return L.Sort(prune_columns(plan.child, child_needed), plan.keys,

@@ -439,6 +439,11 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
                                    kids[0], node.schema(),
                                    input_replicated=L.is_replicated(node.child),
                                    merge_target_bytes=conf.get(BATCH_SIZE_BYTES))
+    if isinstance(node, L.NestedLoopJoin):
+        return P.NestedLoopJoinExec(
+            device, kids[0], kids[1], [], [], node.how, node.schema(),
+            right_replicated=L.is_replicated(node.right),
+            condition=node.condition)
     if isinstance(node, L.Join):
         from ..config import BROADCAST_THRESHOLD, JOIN_SUBPARTITION_BYTES
 

@@ -815,6 +815,112 @@ class HashJoinExec(PhysicalExec):
         return f"{self.name()}({self.how}, {pairs})"
 
 
+class NestedLoopJoinExec(HashJoinExec):
+    """Broadcast nested-loop join: all (left, right) row pairs streamed
+    through the join condition in bounded chunks (reference analogue:
+    GpuBroadcastNestedLoopJoinExec — compiled-AST condition over cross
+    blocks). Supports inner/left/semi/anti/full; the build side is
+    all-gathered when sharded (nested loops cannot co-partition)."""
+
+    PAIR_CHUNK = 1 << 22  # max candidate pairs materialized at once
+
+    def _cross_maps(self, m: int, nr: int):
+        import numpy as _np
+
+        total = m * nr
+        if self.gpu:
+            import torch as _torch
+
+            from ..ops import gpu_backend as _gb
+
+            iota = _torch.empty(total, dtype=_torch.int32, device="cuda")
+            _gb.ext.iota_i32(iota.data_ptr(), total, _gb._stream())
+            icol = Column(DType.int32(), total, iota, None, null_count=0)
+            lmap = _gb.binary_op_scalar("int_div", icol, nr, DType.int32())
+            rmap = _gb.binary_op_scalar("mod", icol, nr, DType.int32())
+        else:
+            lmap = Column.from_numpy(
+                _np.repeat(_np.arange(m, dtype=_np.int32), nr))
+            rmap = Column.from_numpy(
+                _np.tile(_np.arange(nr, dtype=_np.int32), m))
+        return lmap, rmap
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        import torch as _torch
+
+        left, right = self.children
+        rbatches = list(right.execute())
+        from ..shuffle import dist as _dist
+        if _dist.ctx().is_multi and not self.right_replicated:
+            from ..shuffle.exchange import gather_all
+
+            self._strategy = "broadcast"
+            local = self._local_or_empty(rbatches, right.schema)
+            rbatches = [b for b in gather_all(local) if b.num_rows]
+        rtable = None
+        if rbatches:
+            rtable = ops.concat_batches(rbatches) if len(rbatches) > 1 \
+                else rbatches[0]
+        nr = rtable.num_rows if rtable is not None else 0
+        right_matched = self._fresh_matched(rtable) \
+            if self.how == "full" and nr else None
+        dev = "cuda" if self.gpu else "cpu"
+        for lbatch in left.execute():
+            nl = lbatch.num_rows
+            if nl == 0:
+                continue
+            if nr == 0:
+                if self.how == "anti":
+                    yield lbatch
+                elif self.how in ("left", "full"):
+                    yield self._left_with_null_right(lbatch)
+                continue
+            rows_per = max(1, self.PAIR_CHUNK // nr)
+            hit = _torch.zeros(nl, dtype=_torch.bool, device=dev)
+            for s0 in range(0, nl, rows_per):
+                e0 = min(nl, s0 + rows_per)
+                chunk = _slice_rows(lbatch, s0, e0)
+                lmap, rmap = self._cross_maps(e0 - s0, nr)
+                lout = ops.gather(chunk, lmap)
+                rout = ops.gather(rtable, rmap)
+                pair = ColumnBatch(list(lout.columns) + list(rout.columns),
+                                   lout.num_rows)
+                mask = self.condition.eval(pair, self._pair_schema())
+                fmaps = ops.apply_boolean_mask(
+                    ColumnBatch([lmap, rmap], lmap.size), mask)
+                lmap_f, rmap_f = fmaps.columns
+                if lmap_f.size:
+                    hit[s0:e0][lmap_f.data[:lmap_f.size].long()] = True
+                    if right_matched is not None:
+                        idx = rmap_f.data[:rmap_f.size].long()
+                        if isinstance(right_matched, _torch.Tensor):
+                            right_matched[idx] = 1
+                        else:
+                            right_matched[idx.cpu().numpy()] = True
+                    if self.how in ("inner", "left", "full"):
+                        out = ops.apply_boolean_mask(pair, mask)
+                        lcols = out.columns[:len(lbatch.columns)]
+                        rcols = ColumnBatch(
+                            out.columns[len(lbatch.columns):], out.num_rows)
+                        yield ColumnBatch(
+                            list(lcols) + self._right_out(rcols),
+                            out.num_rows)
+            if self.how in ("semi", "anti", "left", "full"):
+                want = hit if self.how == "semi" else ~hit
+                mcol = Column(DType.bool_(), nl, want.to(_torch.uint8),
+                              None, null_count=0)
+                rest = ops.apply_boolean_mask(lbatch, mcol)
+                if rest.num_rows:
+                    if self.how in ("left", "full"):
+                        yield self._left_with_null_right(rest)
+                    else:
+                        yield rest
+        if self.how == "full" and right_matched is not None:
+            extra = self._unmatched_right(rtable, right_matched, left.schema)
+            if extra is not None and extra.num_rows:
+                yield extra
+
+
 class CrossJoinExec(PhysicalExec):
     """Cartesian product; output bounded by maxOutputRows to keep an
     accidental unfiltered cross join from exploding memory. Gather maps are

@@ -609,3 +609,34 @@ def test_conditional_join_gpu_matches_cpu(how):
                 assert x == pytest.approx(y, rel=1e-9), (how, gr, cr)
             else:
                 assert x == y, (how, gr, cr)
+
+
+@pytest.mark.parametrize("how", ["inner", "left", "semi", "anti", "full"])
+def test_nested_loop_join_gpu_matches_cpu(how):
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col
+
+    rng = np.random.default_rng(37)
+    data_l = {"a": [float(v) if i % 19 else None
+                    for i, v in enumerate(rng.uniform(0, 100, 2000))]}
+    data_r = {"b": [float(v) for v in rng.uniform(0, 100, 500)]}
+
+    def q(s):
+        l = s.create_dataframe(data_l)
+        r = s.create_dataframe(data_r)
+        return sorted(l.join_nl(r, (col("a") < col("b") + 0.5)
+                        & (col("b") < col("a") + 0.5), how).collect(), key=repr)
+
+    sg = sr.Session()
+    qg = sg.create_dataframe(data_l).join_nl(
+        sg.create_dataframe(data_r), (col("a") < col("b") + 0.5) & (col("b") < col("a") + 0.5), how)
+    assert "GpuNestedLoopJoin" in qg.physical_plan().tree_string()
+    g = q(sg)
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert len(g) == len(c), how
+    for gr, cr in zip(g, c):
+        for x, y in zip(gr, cr):
+            if isinstance(y, float) and x is not None:
+                assert x == pytest.approx(y, rel=1e-9), (how, gr, cr)
+            else:
+                assert x == y, (how, gr, cr)

@@ -304,3 +304,50 @@ class TestConditionalJoins:
                      .select("a", "b").collect())
         assert out == [(10, 15), (10, 25), (20, 25), (30, 100), (40, 100),
                        (50, None), (60, None)]
+
+
+class TestNestedLoopJoin:
+    def _sides(self, s):
+        left = s.create_dataframe({"a": [1, 5, 9, None]})
+        right = s.create_dataframe({"b": [2, 6, 7]})
+        return left, right
+
+    def test_nl_inner(self, session):
+        l, r = self._sides(session)
+        out = sorted(l.join_nl(r, col("a") < col("b")).collect(), key=repr)
+        assert out == [(1, 2), (1, 6), (1, 7), (5, 6), (5, 7)]
+
+    def test_nl_left_semi_anti(self, session):
+        l, r = self._sides(session)
+        lo = sorted(l.join_nl(r, col("a") < col("b"), "left").collect(),
+                    key=repr)
+        assert lo == [(1, 2), (1, 6), (1, 7), (5, 6), (5, 7),
+                      (9, None), (None, None)]
+        semi = sorted(l.join_nl(r, col("a") < col("b"), "semi")
+                      .to_pydict()["a"])
+        anti = l.join_nl(r, col("a") < col("b"), "anti").to_pydict()["a"]
+        assert semi == [1, 5]
+        assert sorted(anti, key=repr) == [9, None]
+
+    def test_nl_full(self, session):
+        l, r = self._sides(session)
+        out = sorted(l.join_nl(r, col("a") > col("b"), "full").collect(),
+                     key=repr)
+        # pairs: 5>2, 9>2, 9>6, 9>7; unmatched left: 1, None;
+        # unmatched right: none (2,6,7 all matched by 9)
+        assert out == [(1, None), (5, 2), (9, 2), (9, 6), (9, 7),
+                       (None, None)]
+
+    def test_nl_chunked(self, session):
+        import spark_rapids_amd.plan.physical as P
+        old = P.NestedLoopJoinExec.PAIR_CHUNK
+        P.NestedLoopJoinExec.PAIR_CHUNK = 4
+        try:
+            l = session.create_dataframe({"a": list(range(20))})
+            r = session.create_dataframe({"b": [5, 15]})
+            out = sorted(l.join_nl(r, col("a") < col("b")).collect())
+            exp = sorted([(a, b) for a in range(20) for b in (5, 15)
+                          if a < b])
+            assert out == exp
+        finally:
+            P.NestedLoopJoinExec.PAIR_CHUNK = old
