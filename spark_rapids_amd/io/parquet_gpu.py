@@ -27,6 +27,7 @@ from . import thrift_compact as tc
 
 PLAIN, PLAIN_DICTIONARY, RLE, RLE_DICTIONARY = 0, 2, 3, 8
 DELTA_BINARY_PACKED = 5
+DELTA_LENGTH_BYTE_ARRAY, DELTA_BYTE_ARRAY = 6, 7
 DATA_PAGE, DICTIONARY_PAGE, DATA_PAGE_V2 = 0, 2, 3
 
 _PHYS_NP = {
@@ -416,6 +417,75 @@ class _ChunkDecoder:
         else:
             raise NotImplementedError(f"dict for {self.phys}")
 
+    def _string_dense_to_rows(self, offs_dense, out_bytes, n, mask,
+                              valid_idx, n_valid) -> Column:
+        """Dense (valid-only) string buffers -> row-level string column,
+        scattering through a null mask when present."""
+        from ..ops import gpu_backend as gb
+
+        dense_col = Column(DType.string(), n_valid, out_bytes, None,
+                           offs_dense, 0)
+        if mask is None:
+            return dense_col
+        # scatter dense -> rows with nulls: build ridx then string-gather
+        ridx = torch.full((n,), -1, dtype=torch.int32, device="cuda")
+        dense_iota = torch.empty(max(n_valid, 1), dtype=torch.int32,
+                                 device="cuda")[:n_valid]
+        if n_valid:
+            self.ext.iota_i32(dense_iota.data_ptr(), n_valid, self.s)
+            self.ext.scatter_fixed(4, dense_iota.data_ptr(),
+                                   valid_idx.data_ptr(), ridx.data_ptr(),
+                                   n_valid, self.s)
+        out = gb._gather_col(dense_col, ridx, n, maybe_negative=True)
+        return Column(self.dtype, n, out.data, mask, out.offsets,
+                      null_count=None)
+
+    def _delta_byte_array_dense(self, values: bytes, count: int,
+                                encoding: int):
+        """Host decode of DELTA_LENGTH_BYTE_ARRAY / DELTA_BYTE_ARRAY
+        string pages -> (offsets int64 [count+1], dense payload bytes).
+        The length/prefix vectors are serial varint+bitpack chains
+        (pq_delta_walk in C++); reconstruction for DELTA_BYTE_ARRAY is a
+        serial prefix-chain memcpy pass (delta_ba_concat). Reference
+        analogue: GpuParquetScan's cudf delta decoders."""
+        if count == 0:
+            return np.zeros(1, dtype=np.int64), b""
+        arr = np.frombuffer(values, dtype=np.uint8)
+        if encoding == DELTA_LENGTH_BYTE_ARRAY:
+            lens = np.empty(count, dtype=np.int64)
+            consumed = self.ext.pq_delta_walk_host(
+                arr.ctypes.data, len(arr), lens.ctypes.data, count)
+            if consumed < 0 or lens.min() < 0:
+                raise NotImplementedError("corrupt DELTA_LENGTH stream")
+            offs = np.zeros(count + 1, dtype=np.int64)
+            np.cumsum(lens, out=offs[1:])
+            total = int(offs[-1])
+            if consumed + total > len(arr):
+                raise NotImplementedError("DELTA_LENGTH payload overrun")
+            return offs, np.ascontiguousarray(
+                arr[consumed:consumed + total])
+        pre = np.empty(count, dtype=np.int64)
+        c1 = self.ext.pq_delta_walk_host(arr.ctypes.data, len(arr),
+                                         pre.ctypes.data, count)
+        if c1 < 0:
+            raise NotImplementedError("corrupt DELTA_BYTE_ARRAY prefixes")
+        suf = np.empty(count, dtype=np.int64)
+        c2 = self.ext.pq_delta_walk_host(arr.ctypes.data + c1,
+                                         len(arr) - c1, suf.ctypes.data,
+                                         count)
+        if c2 < 0:
+            raise NotImplementedError("corrupt DELTA_BYTE_ARRAY suffixes")
+        offs = np.zeros(count + 1, dtype=np.int64)
+        np.cumsum(pre + suf, out=offs[1:])
+        total = int(offs[-1])
+        outbuf = np.empty(max(total, 1), dtype=np.uint8)[:total]
+        rc = self.ext.delta_ba_concat_host(
+            pre.ctypes.data, suf.ctypes.data, arr.ctypes.data + c1 + c2,
+            len(arr) - c1 - c2, count, outbuf.ctypes.data, offs.ctypes.data)
+        if rc < 0:
+            raise NotImplementedError("corrupt DELTA_BYTE_ARRAY payload")
+        return offs, outbuf
+
     def _byte_array_dense(self, data: bytes, count: int):
         """Parse parquet length-prefixed BYTE_ARRAY records -> (offsets
         int32 [count+1] cuda, compact bytes cuda). The serial record walk
@@ -509,25 +579,20 @@ class _ChunkDecoder:
                                        n_valid, self.s)
             return Column(self.dtype, n, out, mask, null_count=None)
         if encoding == PLAIN and self.phys == "BYTE_ARRAY":
-            from ..ops import gpu_backend as gb
-
             offs_dense, out_bytes = self._byte_array_dense(values, n_valid)
-            dense_col = Column(DType.string(), n_valid, out_bytes, None,
-                               offs_dense, 0)
-            if not nulls:
-                return dense_col
-            # scatter dense -> rows with nulls: build ridx then string-gather
-            ridx = torch.full((n,), -1, dtype=torch.int32, device="cuda")
-            dense_iota = torch.empty(max(n_valid, 1), dtype=torch.int32,
-                                     device="cuda")[:n_valid]
-            if n_valid:
-                self.ext.iota_i32(dense_iota.data_ptr(), n_valid, self.s)
-                self.ext.scatter_fixed(4, dense_iota.data_ptr(),
-                                       valid_idx.data_ptr(), ridx.data_ptr(),
-                                       n_valid, self.s)
-            out = gb._gather_col(dense_col, ridx, n, maybe_negative=True)
-            return Column(self.dtype, n, out.data, mask, out.offsets,
-                          null_count=None)
+            return self._string_dense_to_rows(offs_dense, out_bytes, n,
+                                              mask, valid_idx, n_valid)
+        if encoding in (DELTA_LENGTH_BYTE_ARRAY, DELTA_BYTE_ARRAY) \
+                and self.phys == "BYTE_ARRAY":
+            offs_h, payload = self._delta_byte_array_dense(
+                values, n_valid, encoding)
+            offs_dense = torch.from_numpy(
+                offs_h.astype(np.int32)).cuda()
+            out_bytes = _upload_parts([payload], self.ext, self.s) \
+                if len(payload) else torch.zeros(0, dtype=torch.uint8,
+                                                 device="cuda")
+            return self._string_dense_to_rows(offs_dense, out_bytes, n,
+                                              mask, valid_idx, n_valid)
         if encoding == DELTA_BINARY_PACKED and self.phys in ("INT32",
                                                              "INT64"):
             page = torch.from_numpy(np.frombuffer(
