@@ -20,7 +20,9 @@ from .expr.expressions import (CaseWhen, ascii_, coalesce, col, concat_ws,
                                datediff, dayofweek, greatest, hour, isin,
                                least, lit, quarter, to_date,
                                unix_timestamp,
-                               minute, round_, second, when)
+                               minute, round_, second, when,
+                               create_map, map_keys, map_values,
+                               map_entries)
 from .expr.windows import (dense_rank, lag, lead, nth_value, ntile,
                            rank, row_number, win_avg,
                            win_count, win_max, win_min, win_sum)

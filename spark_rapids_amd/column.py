@@ -122,7 +122,7 @@ class Column:
         self._minmax = None  # cached (min, max) for dense-key group-by
         if dtype.id is TypeId.STRING:
             assert offsets is not None and offsets.numel() == size + 1
-        if dtype.id is TypeId.LIST:
+        if dtype.id in (TypeId.LIST, TypeId.MAP):
             assert offsets is not None and child is not None
         if dtype.id is TypeId.STRUCT:
             assert isinstance(child, tuple) and \
@@ -227,15 +227,17 @@ class Column:
                          make_validity(valid) if not valid.all() else None,
                          None, None, tuple(kids))
             return col.to(device) if device != "cpu" else col
-        if dtype.id is TypeId.LIST:
+        if dtype.id in (TypeId.LIST, TypeId.MAP):
             valid = np.array([v is not None for v in values], dtype=bool)
             flat: list = []
             offsets = np.zeros(n + 1, dtype=np.int32)
             for i, v in enumerate(values):
                 if v is not None:
-                    flat.extend(v)
+                    flat.extend(v.items() if isinstance(v, dict) else v)
                 offsets[i + 1] = len(flat)
-            child = Column.from_pylist(flat, dtype.children[0])
+            elem_dt = dtype.entry_dtype if dtype.id is TypeId.MAP \
+                else dtype.children[0]
+            child = Column.from_pylist(flat, elem_dt)
             col = Column(dtype, n, torch.zeros(0, dtype=torch.uint8),
                          make_validity(valid) if not valid.all() else None,
                          torch.from_numpy(offsets), None, child)
@@ -342,6 +344,12 @@ class Column:
             elems = self.child.to_pylist()
             return [list(elems[offs[i]:offs[i + 1]]) if valid[i] else None
                     for i in range(self.size)]
+        if self.dtype.id is TypeId.MAP:
+            offs = self.offsets.cpu().numpy()
+            elems = self.child.to_pylist()  # entry dicts {key, value}
+            return [{e["key"]: e["value"]
+                     for e in elems[offs[i]:offs[i + 1]]}
+                    if valid[i] else None for i in range(self.size)]
         if self.dtype.id is TypeId.STRING:
             offs = self.offsets.cpu().numpy()
             raw = self.data.cpu().numpy().tobytes()

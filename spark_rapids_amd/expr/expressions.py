@@ -150,13 +150,13 @@ class Expression:
         empty parts dropped)."""
         return StrSplit(self, delimiter)
 
-    def element_at(self, index: int) -> "ElementAt":
-        """1-based element of a LIST value; NULL beyond the list length
-        (Spark element_at)."""
+    def element_at(self, index) -> "ElementAt":
+        """1-based element of a LIST value (NULL beyond the length), or
+        the value for a key of a MAP value (Spark element_at)."""
         return ElementAt(self, index)
 
     def size(self) -> "ArraySize":
-        """Element count of a LIST value (null list -> null)."""
+        """Entry count of a LIST or MAP value (null -> null)."""
         return ArraySize(self)
 
     def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
@@ -633,10 +633,10 @@ class StrSplit(Expression):
 
 class ElementAt(Expression):
     """element_at(array, k) with 1-based k (negative k counts from the
-    end); NULL when |k| exceeds the length (GpuElementAt analogue)."""
+    end); NULL when |k| exceeds the length. element_at(map, key) returns
+    the value for key or NULL (GpuElementAt analogue)."""
 
-    def __init__(self, child: Expression, index: int):
-        assert index != 0, "element_at is 1-based (Spark semantics)"
+    def __init__(self, child: Expression, index):
         self.child = child
         self.index = index
 
@@ -645,10 +645,16 @@ class ElementAt(Expression):
         return (self.child,)
 
     def dtype(self, schema: Schema) -> DType:
-        return self.child.dtype(schema).children[0]
+        cdt = self.child.dtype(schema)
+        return cdt.children[1] if cdt.id is TypeId.MAP else cdt.children[0]
 
     def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
-        return ops.element_at(self.child.eval(batch, schema), self.index)
+        c = self.child.eval(batch, schema)
+        if c.dtype.id is TypeId.MAP:
+            return ops.map_get(c, self.index)
+        assert isinstance(self.index, int) and self.index != 0, \
+            "element_at(array, k) is 1-based (Spark semantics)"
+        return ops.element_at(c, self.index)
 
     def __str__(self):
         return f"element_at({self.child}, {self.index})"
@@ -1224,3 +1230,88 @@ def named_struct(**fields) -> CreateNamedStruct:
 
 def get_field(struct_expr, name: str) -> GetStructField:
     return GetStructField(struct_expr, name)
+
+
+class CreateMap(Expression):
+    """create_map(k1, v1, k2, v2, ...) -> MAP column (reference:
+    GpuCreateMap). Entries keep source order; lookups are last-win."""
+
+    def __init__(self, exprs):
+        assert exprs and len(exprs) % 2 == 0, \
+            "create_map needs key1, value1, key2, value2, ..."
+        self.keys = [_as_expr(e) for e in exprs[0::2]]
+        self.vals = [_as_expr(e) for e in exprs[1::2]]
+
+    @property
+    def children(self):
+        return tuple(self.keys) + tuple(self.vals)
+
+    def dtype(self, schema: Schema) -> DType:
+        return DType.map_(self.keys[0].dtype(schema),
+                          self.vals[0].dtype(schema))
+
+    def nullable(self, schema: Schema) -> bool:
+        return False
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        kc = [e.eval(batch, schema) for e in self.keys]
+        vc = [e.eval(batch, schema) for e in self.vals]
+        return ops.make_map(kc, vc)
+
+    def output_name(self) -> str:
+        return "map"
+
+    def __str__(self):
+        inner = ", ".join(f"{k}, {v}" for k, v in zip(self.keys, self.vals))
+        return f"map({inner})"
+
+
+class MapView(Expression):
+    """map_keys / map_values / map_entries (zero-copy layout views;
+    reference: GpuMapKeys/GpuMapValues/GpuMapEntries)."""
+
+    def __init__(self, child, mode: str):
+        self.child = _as_expr(child)
+        self.mode = mode  # keys | values | entries
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        mt = self.child.dtype(schema)
+        if mt.id is not TypeId.MAP:
+            raise TypeError(f"map_{self.mode} on non-map {mt}")
+        if self.mode == "keys":
+            return DType.list_(mt.children[0])
+        if self.mode == "values":
+            return DType.list_(mt.children[1])
+        return DType.list_(mt.entry_dtype)
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        c = self.child.eval(batch, schema)
+        fn = {"keys": ops.map_keys, "values": ops.map_values,
+              "entries": ops.map_entries}[self.mode]
+        return fn(c)
+
+    def output_name(self) -> str:
+        return f"map_{self.mode}({self.child.output_name()})"
+
+    def __str__(self):
+        return f"map_{self.mode}({self.child})"
+
+
+def create_map(*exprs) -> CreateMap:
+    return CreateMap(list(exprs))
+
+
+def map_keys(m) -> MapView:
+    return MapView(m, "keys")
+
+
+def map_values(m) -> MapView:
+    return MapView(m, "values")
+
+
+def map_entries(m) -> MapView:
+    return MapView(m, "entries")

@@ -35,6 +35,10 @@ def _spark_type(dt: DType) -> object:
     if dt.id is TypeId.LIST:
         return {"type": "array", "elementType": _spark_type(dt.children[0]),
                 "containsNull": True}
+    if dt.id is TypeId.MAP:
+        return {"type": "map", "keyType": _spark_type(dt.children[0]),
+                "valueType": _spark_type(dt.children[1]),
+                "valueContainsNull": True}
     if dt.id is TypeId.STRUCT:
         return {"type": "struct", "fields": [
             {"name": n, "type": _spark_type(c), "nullable": True,

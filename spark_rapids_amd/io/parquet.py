@@ -62,6 +62,9 @@ def arrow_to_dtype(t) -> DType:
         return STRING
     if pa.types.is_decimal(t):
         return DType.decimal(t.precision, t.scale)
+    if pa.types.is_map(t):
+        return DType.map_(arrow_to_dtype(t.key_type),
+                          arrow_to_dtype(t.item_type))
     if pa.types.is_list(t) or pa.types.is_large_list(t):
         return DType.list_(arrow_to_dtype(t.value_type))
     if pa.types.is_struct(t):
@@ -436,6 +439,9 @@ def _dtype_to_arrow(dtype: DType):
 
     if dtype.id is TypeId.LIST:
         return pa.list_(_dtype_to_arrow(dtype.children[0]))
+    if dtype.id is TypeId.MAP:
+        return pa.map_(_dtype_to_arrow(dtype.children[0]),
+                       _dtype_to_arrow(dtype.children[1]))
     if dtype.id is TypeId.STRUCT:
         return pa.struct([(n, _dtype_to_arrow(t)) for n, t in
                           zip(dtype.field_names, dtype.children)])

@@ -58,6 +58,39 @@ def element_at(col: Column, index: int) -> Column:
     return backend_for(col).element_at(col, index)
 
 
+def map_keys(col: Column) -> Column:
+    """MAP -> LIST<key>: shares the offsets and the entry key child
+    buffer (zero-copy, reference GpuMapKeys)."""
+    return Column(DType.list_(col.dtype.children[0]), col.size, col.data,
+                  col.validity, col.offsets, col._null_count,
+                  col.child.child[0])
+
+
+def map_values(col: Column) -> Column:
+    return Column(DType.list_(col.dtype.children[1]), col.size, col.data,
+                  col.validity, col.offsets, col._null_count,
+                  col.child.child[1])
+
+
+def map_entries(col: Column) -> Column:
+    """MAP -> LIST<STRUCT<key,value>> (zero-copy view of the entries)."""
+    return Column(DType.list_(col.dtype.entry_dtype), col.size, col.data,
+                  col.validity, col.offsets, col._null_count, col.child)
+
+
+def map_get(col: Column, key) -> Column:
+    """element_at(map, key): the value for `key` per row, NULL when the
+    key is absent or the map is null (GpuElementAt over maps)."""
+    return backend_for(col).map_get(col, key)
+
+
+def make_map(kcols: Sequence[Column], vcols: Sequence[Column]) -> Column:
+    """create_map(k1,v1,k2,v2,...): one map per row with a fixed entry
+    list; entries keep source order (no dedup; lookups are LAST_WIN —
+    map_get returns the last match, matching dict() view semantics)."""
+    return backend_for(*kcols, *vcols).make_map(list(kcols), list(vcols))
+
+
 def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     return backend_for(col).regexp_extract(col, pattern, group)
 

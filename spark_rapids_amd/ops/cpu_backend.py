@@ -656,6 +656,24 @@ def element_at(col: Column, index: int) -> Column:
     return Column.from_pylist(out, col.dtype.children[0])
 
 
+def map_get(col: Column, key) -> Column:
+    out = []
+    for v in col.to_pylist():  # dicts (last-win on duplicate keys)
+        out.append(None if v is None else v.get(key))
+    return Column.from_pylist(out, col.dtype.children[1])
+
+
+def make_map(kcols, vcols) -> Column:
+    from ..types import DType as _DT
+
+    dtype = _DT.map_(kcols[0].dtype, vcols[0].dtype)
+    kl = [c.to_pylist() for c in kcols]
+    vl = [c.to_pylist() for c in vcols]
+    rows = [list(zip(krow, vrow))
+            for krow, vrow in zip(zip(*kl), zip(*vl))]
+    return Column.from_pylist(rows, dtype)
+
+
 def regexp_extract(col: Column, pattern: str, group: int) -> Column:
     import re as _re
 
@@ -813,7 +831,7 @@ def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> C
 def _gather_idx(batch: ColumnBatch, idx: np.ndarray, row_ok: np.ndarray) -> ColumnBatch:
     cols = []
     for c in batch.columns:
-        if c.dtype.id in (TypeId.LIST, TypeId.STRUCT):
+        if c.dtype.is_nested:
             vals_py = c.to_pylist()
             out = [vals_py[i] if ok and 0 <= i < len(vals_py) else None
                    for i, ok in zip(idx, row_ok)]
@@ -836,7 +854,7 @@ def concat_batches(batches: List[ColumnBatch]) -> ColumnBatch:
     cols = []
     for i in range(ncols):
         dtype = batches[0].columns[i].dtype
-        if dtype.id in (TypeId.LIST, TypeId.STRUCT):
+        if dtype.is_nested:
             vals = []
             for b in batches:
                 vals.extend(b.columns[i].to_pylist())

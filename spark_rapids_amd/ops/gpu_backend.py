@@ -657,6 +657,74 @@ def element_at(col: Column, index: int) -> Column:
     return _gather_col(col.child, srcm.data, n, maybe_negative=True)
 
 
+def map_get(col: Column, key) -> Column:
+    """element_at(map, key): entry-level key compare + per-segment
+    last-match reduction (torch scatter_reduce amax over the entry iota),
+    then a nullable gather of the value child."""
+    n = col.size
+    vdt = col.dtype.children[1]
+    if n == 0:
+        return _empty_col(vdt)
+    entry = col.child
+    kcol, vcol = entry.child
+    ne = kcol.size
+    if ne == 0:
+        out = _gather_col(vcol, torch.full((n,), -1, dtype=torch.int32,
+                                           device="cuda"), n,
+                          maybe_negative=True)
+        return out
+    if kcol.dtype.id is TypeId.STRING:
+        sw = str_predicate("starts_with", kcol, key)
+        ln = unary_op("length", kcol, DType.int32())
+        lm = binary_op_scalar("eq", ln, len(key), DType.bool_())
+        match = binary_op("and", sw, lm, DType.bool_())
+    else:
+        match = binary_op_scalar("eq", kcol, key, DType.bool_())
+    mt = match.data[:ne].bool()
+    offs = col.offsets.long()
+    counts = offs[1:n + 1] - offs[:n]
+    seg = torch.repeat_interleave(
+        torch.arange(n, device="cuda", dtype=torch.int64), counts)
+    eidx = torch.arange(ne, device="cuda", dtype=torch.int64)
+    res = torch.full((n,), -1, dtype=torch.int64, device="cuda")
+    if mt.any():
+        res.scatter_reduce_(0, seg[mt], eidx[mt], reduce="amax",
+                            include_self=True)
+    if col.validity is not None:
+        rv = torch.empty(n, dtype=torch.uint8, device="cuda")
+        ext.mask_expand(col.validity.data_ptr(), rv.data_ptr(), False, n,
+                        _stream())
+        res = torch.where(rv.bool(), res,
+                          torch.full_like(res, -1))
+    return _gather_col(vcol, res.to(torch.int32), n, maybe_negative=True)
+
+
+def make_map(kcols, vcols) -> Column:
+    """create_map: interleave the per-pair key/value columns into entry
+    children (block concat + permutation gather) with row-major offsets."""
+    npairs = len(kcols)
+    n = kcols[0].size
+    dtype = DType.map_(kcols[0].dtype, vcols[0].dtype)
+    perm_np = (np.arange(n * npairs, dtype=np.int32) % npairs) * n \
+        + np.arange(n * npairs, dtype=np.int32) // npairs
+    perm = torch.from_numpy(perm_np).cuda()
+    def inter(cols):
+        if len(cols) == 1:
+            return cols[0]
+        blk = concat_batches([ColumnBatch([c], n) for c in cols]).columns[0]
+        return _gather_col(blk, perm, n * npairs, maybe_negative=False)
+    kc = inter(kcols)
+    vc = inter(vcols)
+    entry = Column(dtype.entry_dtype, n * npairs,
+                   torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                   None, None, 0, (kc, vc))
+    offsets = torch.arange(0, (n + 1) * npairs, npairs,
+                           dtype=torch.int32, device="cuda")
+    return Column(dtype, n, torch.zeros(0, dtype=torch.uint8,
+                                        device="cuda"),
+                  None, offsets, 0, entry)
+
+
 def array_size(col: Column) -> Column:
     n = col.size
     s = _stream()

@@ -151,7 +151,16 @@ class Tagger:
             if any(c in e.delimiter for c in ".\\+*?()[]{}|^$"):
                 out.append("regex split delimiters run on CPU")
         elif type(e).__name__ in ("ArraySize", "ElementAt"):
-            pass
+            cdt = e.child.dtype(schema)
+            if cdt.id is TypeId.MAP and type(e).__name__ == "ElementAt" \
+                    and cdt.children[1].is_nested:
+                out.append("element_at over map with nested values on CPU")
+        elif type(e).__name__ in ("CreateMap", "MapView"):
+            mt = e.dtype(schema)
+            if any(c.is_nested for c in
+                   (mt.children if mt.id is TypeId.MAP
+                    else mt.children[0].children)):
+                out.append("maps of nested keys/values on CPU")
         elif type(e).__name__ in ("PadExpr", "LocateExpr"):
             pass  # lpad/rpad/locate/instr device kernels (k_str_pad/locate)
         elif type(e).__name__ in ("DateFormat", "ToTimestamp", "TzConvert"):

@@ -44,7 +44,7 @@ def _pad_to8(sections: List[torch.Tensor], t: torch.Tensor, dev):
 def _ser_col(c: Column, header: List[int], sections: List[torch.Tensor],
              dev):
     has_valid = c.validity is not None
-    if c.dtype.id is TypeId.LIST:
+    if c.dtype.id in (TypeId.LIST, TypeId.MAP):
         child = c.child
         header.extend([1 if has_valid else 0, child.size])
         _pad_to8(sections, _as_bytes(c.offsets), dev)
@@ -103,12 +103,14 @@ class _Reader:
 
 
 def _deser_col(dtype: DType, n: int, r: _Reader) -> Column:
-    if dtype.id is TypeId.LIST:
+    if dtype.id in (TypeId.LIST, TypeId.MAP):
         has_valid = r.take_header() != 0
         child_rows = r.take_header()
         offsets = r.take((n + 1) * 4, pad=True).view(torch.int32)
         validity = r.take(mask_nbytes(n), pad=False) if has_valid else None
-        child = _deser_col(dtype.children[0], child_rows, r)
+        elem_dt = dtype.entry_dtype if dtype.id is TypeId.MAP \
+            else dtype.children[0]
+        child = _deser_col(elem_dt, child_rows, r)
         return Column(dtype, n, torch.zeros(0, dtype=torch.uint8,
                                             device=r.buf.device),
                       validity, offsets, None if has_valid else 0, child)

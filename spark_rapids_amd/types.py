@@ -30,6 +30,7 @@ class TypeId(enum.Enum):
     NULL = "void"
     LIST = "array"
     STRUCT = "struct"
+    MAP = "map"
 
 
 _FIXED_WIDTH_BYTES = {
@@ -126,6 +127,18 @@ class DType:
         return DType(TypeId.LIST, children=(elem,))
 
     @staticmethod
+    def map_(key: "DType", value: "DType") -> "DType":
+        """Spark MapType. Physical layout = LIST of STRUCT<key,value>
+        entries (arrow map layout): offsets + one entry struct child."""
+        return DType(TypeId.MAP, children=(key, value))
+
+    @property
+    def entry_dtype(self) -> "DType":
+        """MAP only: the entry struct dtype (key, value)."""
+        return DType.struct_([("key", self.children[0]),
+                              ("value", self.children[1])])
+
+    @staticmethod
     def struct(*fields: "DType") -> "DType":
         names = tuple(f"c{i}" for i in range(len(fields)))
         return DType(TypeId.STRUCT, children=tuple(fields),
@@ -174,7 +187,7 @@ class DType:
 
     @property
     def is_nested(self) -> bool:
-        return self.id in (TypeId.LIST, TypeId.STRUCT)
+        return self.id in (TypeId.LIST, TypeId.STRUCT, TypeId.MAP)
 
     def numpy_dtype(self):
         return np.dtype(_NUMPY_DTYPES[self.id])
@@ -184,6 +197,8 @@ class DType:
             return f"decimal({self.precision},{self.scale})"
         if self.id is TypeId.LIST:
             return f"array<{self.children[0]}>"
+        if self.id is TypeId.MAP:
+            return f"map<{self.children[0]},{self.children[1]}>"
         if self.id is TypeId.STRUCT:
             inner = ", ".join(f"{n}:{c}" for n, c in
                               zip(self.field_names, self.children))
