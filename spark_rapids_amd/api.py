@@ -452,6 +452,11 @@ class Session:
 
             device_pool.activate(self.conf.get(MEM_POOL_FRACTION))
             _spill.configure_watermark(self.conf.get(MEM_SPILL_WATERMARK))
+            from .config import PINNED_POOL_SIZE
+            from .memory import host_pool as _hp
+
+            if _hp.pool() is None:
+                _hp.configure(self.conf.get(PINNED_POOL_SIZE))
 
     # ---- conf ----------------------------------------------------------
     def set(self, key: str, value) -> "Session":

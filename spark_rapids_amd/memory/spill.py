@@ -56,9 +56,44 @@ class SpillableBatch:
             if self._state != DEVICE or self._batch is None:
                 return 0
             n = self._batch.nbytes
-            self._batch = self._batch.to("cpu")
+            self._batch = self._to_host_pinned(self._batch)
             self._state = HOST
             return n
+
+    def _to_host_pinned(self, batch):
+        """D2H through the pinned pool (HostAlloc analogue) when
+        configured; pageable fallback otherwise."""
+        from ..column import ColumnBatch
+        from . import host_pool
+
+        if host_pool.pool() is None or not batch.is_cuda:
+            return batch.to("cpu")
+        self._pinned = getattr(self, "_pinned", [])
+        cols = [self._copy_col_host(c) for c in batch.columns]
+        torch.cuda.synchronize()  # copies must land before device frees
+        return ColumnBatch(cols, batch.num_rows)
+
+    def _copy_col_host(self, c):
+        from ..column import Column
+        from . import host_pool
+
+        child = c.child
+        if isinstance(child, tuple):
+            child = tuple(self._copy_col_host(k) for k in child)
+        elif child is not None:
+            child = self._copy_col_host(child)
+        cp = lambda t: host_pool.copy_tensor_to_host(t, self._pinned)
+        return Column(c.dtype, c.size, cp(c.data),
+                      cp(c.validity) if c.validity is not None else None,
+                      cp(c.offsets) if c.offsets is not None else None,
+                      c._null_count, child)
+
+    def _release_pinned(self):
+        from . import host_pool
+
+        held = getattr(self, "_pinned", None)
+        if held:
+            host_pool.release(held)
 
     def spill_to_disk(self):
         with self._lock:
@@ -75,6 +110,7 @@ class SpillableBatch:
             self._meta = meta
             self._disk_path = path
             self._batch = None
+            self._release_pinned()
             self._state = DISK
             return n
 
@@ -89,12 +125,18 @@ class SpillableBatch:
                 self._state = HOST
             if self._state == HOST and self._orig_device != "cpu":
                 self._batch = self._batch.to(self._orig_device)
+                # pinned H2D uploads are async on the current stream; the
+                # pinned staging can only be reused after they land
+                if getattr(self, "_pinned", None):
+                    torch.cuda.synchronize()
+                self._release_pinned()
                 self._state = DEVICE
             return self._batch
 
     def close(self):
         with self._lock:
             self._batch = None
+            self._release_pinned()
             if self._disk_path and os.path.exists(self._disk_path):
                 os.unlink(self._disk_path)
             self._disk_path = None

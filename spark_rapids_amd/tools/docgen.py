@@ -171,6 +171,12 @@ _EXPR_ROWS = [
     # complex types
     ("CreateNamedStruct", "named_struct", "GPU", "struct columns, child-wise"),
     ("GetStructField", "get_field", "GPU", "parent null mask merged"),
+    ("CreateMap", "create_map", "GPU", "entry interleave, last-win lookups"),
+    ("MapKeys", "map_keys", "GPU", "zero-copy entries view"),
+    ("MapValues", "map_values", "GPU", "zero-copy entries view"),
+    ("MapEntries", "map_entries", "GPU", "LIST<STRUCT<key,value>> view"),
+    ("ElementAt (map)", "col.element_at(key)", "GPU",
+     "segment last-match reduction; nested values on CPU"),
     ("Explode", "explode", "GPU", ""),
     ("PosExplode", "posexplode", "GPU", ""),
     ("Explode_outer", "explode(outer=True)", "CPU", "padding path"),

@@ -401,3 +401,59 @@ def test_c_api_consumer():
                        timeout=120, cwd=REPO)
     assert r.returncode == 0, r.stdout + r.stderr
     assert "C_API_OK" in r.stdout, r.stdout
+
+
+class TestPinnedHostPool:
+    def test_alloc_free_coalesce(self):
+        from spark_rapids_amd.memory.host_pool import PinnedPool
+
+        p = PinnedPool(1 << 20, slab_bytes=4096)
+        a = p.alloc(1000)
+        b = p.alloc(1000)
+        c = p.alloc(1000)
+        assert a is not None and b is not None and c is not None
+        assert p.used == 3 * 1008  # 16-aligned
+        p.free(b)
+        b2 = p.alloc(900)  # reuses the hole
+        assert p.stats()["slabs"] == 1
+        p.free(a)
+        p.free(b2)
+        p.free(c)
+        assert p.used == 0
+        # coalesced back to one range covering the slab
+        assert p._free[0] == [(0, 4096)]
+
+    def test_capacity_limit(self):
+        from spark_rapids_amd.memory.host_pool import PinnedPool
+
+        p = PinnedPool(8192, slab_bytes=4096)
+        a = p.alloc(4000)
+        b = p.alloc(4000)
+        assert a is not None and b is not None
+        assert p.alloc(4000) is None  # over capacity -> pageable fallback
+
+    @pytest.mark.gpu
+    def test_spill_uses_pool_and_releases(self):
+        import numpy as np
+
+        from spark_rapids_amd.column import Column, ColumnBatch
+        from spark_rapids_amd.memory import host_pool
+        from spark_rapids_amd.memory.spill import SpillableBatch
+        from spark_rapids_amd.types import INT64, STRING
+
+        host_pool.configure(1 << 22)
+        try:
+            c = Column.from_numpy(np.arange(1000, dtype=np.int64), INT64)
+            sc = Column.from_pylist([f"s{i}" for i in range(1000)], STRING)
+            b = ColumnBatch([c, sc], 1000).cuda()
+            h = SpillableBatch(b)
+            assert h.spill_to_host() > 0
+            assert host_pool.pool().used > 0  # staged in pinned views
+            back = h.get()
+            assert host_pool.pool().used == 0  # released after upload
+            assert back.is_cuda
+            assert back.columns[0].cpu().to_pylist() == c.to_pylist()
+            assert back.columns[1].cpu().to_pylist() == sc.to_pylist()
+            h.close()
+        finally:
+            host_pool.configure(None)

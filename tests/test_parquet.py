@@ -456,7 +456,7 @@ def test_gpu_delta_byte_array_scan(tmp_path):
                          "t": "DELTA_LENGTH_BYTE_ARRAY"})
     from spark_rapids_amd.io.parquet_gpu import read_parquet_gpu
 
-    batch = read_parquet_gpu(f)
+    batch = read_parquet_gpu(f, ["s", "t"])
     got_s = batch.columns[0].cpu().to_pylist()
     got_t = batch.columns[1].cpu().to_pylist()
     assert got_s == vals
