@@ -8,6 +8,7 @@ spark.rapids.* config registry.
 from .api import DataFrame, Session
 from .column import Column, ColumnBatch, Field, Schema
 from .config import RapidsConf, help_doc
+from .metrics import reset_task_metrics, task_metrics
 from .expr.aggregates import (approx_count_distinct, approx_percentile,
                               avg, bit_and, bit_or, bit_xor, collect_list,
                               collect_set, percentile,

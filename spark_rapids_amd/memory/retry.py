@@ -78,12 +78,15 @@ def _split_batch(batch) -> List:
 def with_retry_split(fn: Callable, batch, max_splits: int = 8) -> List:
     """Run fn(batch); on OOM retry after releasing memory, then
     split-and-retry recursively. Returns a list of result batches."""
+    from ..metrics import task_metric_add
+
     try:
         oom_injector.maybe_throw()
         return [fn(batch)]
     except GpuSplitAndRetryOOM:
         pass  # go straight to split
     except _OOM_TYPES:
+        task_metric_add("retryCount", 1)
         _release_device_memory()
         try:
             return [fn(batch)]
@@ -91,6 +94,7 @@ def with_retry_split(fn: Callable, batch, max_splits: int = 8) -> List:
             pass
     if max_splits <= 0:
         raise MemoryError("GPU OOM: retry budget exhausted")
+    task_metric_add("splitAndRetryCount", 1)
     parts = _split_batch(batch)
     out: List = []
     for p in parts:

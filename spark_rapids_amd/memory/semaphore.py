@@ -23,6 +23,9 @@ class PrioritySemaphore:
         self.contended = 0  # acquisitions that had to wait (metrics)
 
     def acquire(self, priority: int = 0):
+        import time as _time
+
+        t0 = _time.perf_counter()
         with self._cond:
             self._seq += 1
             me = (priority, self._seq)
@@ -36,6 +39,11 @@ class PrioritySemaphore:
             heapq.heappop(self._waiters)
             self._permits -= 1
             self._cond.notify_all()
+        if waited:
+            from ..metrics import task_metric_add
+
+            task_metric_add("semaphoreWaitMs",
+                            (_time.perf_counter() - t0) * 1e3)
 
     def release(self):
         with self._cond:

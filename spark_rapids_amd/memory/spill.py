@@ -52,12 +52,20 @@ class SpillableBatch:
             return self._batch.nbytes if self._batch is not None else 0
 
     def spill_to_host(self):
+        import time as _time
+
         with self._lock:
             if self._state != DEVICE or self._batch is None:
                 return 0
+            t0 = _time.perf_counter()
             n = self._batch.nbytes
             self._batch = self._to_host_pinned(self._batch)
             self._state = HOST
+            from ..metrics import task_metric_add
+
+            task_metric_add("spillToHostBytes", n)
+            task_metric_add("spillTimeMs",
+                            (_time.perf_counter() - t0) * 1e3)
             return n
 
     def _to_host_pinned(self, batch):
@@ -112,6 +120,9 @@ class SpillableBatch:
             self._batch = None
             self._release_pinned()
             self._state = DISK
+            from ..metrics import task_metric_add
+
+            task_metric_add("spillToDiskBytes", n)
             return n
 
     def get(self):

@@ -72,6 +72,43 @@ def instrument(exec_) -> None:
     _wrap(exec_)
 
 
+_task_lock = __import__("threading").Lock()
+_TASK_ZERO = {
+    "semaphoreWaitMs": 0.0, "spillToHostBytes": 0, "spillToDiskBytes": 0,
+    "spillTimeMs": 0.0, "retryCount": 0, "splitAndRetryCount": 0,
+}
+_task = dict(_TASK_ZERO)
+
+
+def task_metric_add(key: str, value) -> None:
+    """Bump a task-level accumulator (reference: GpuTaskMetrics.scala —
+    semaphore wait, retry counts, spill time/bytes, max device memory)."""
+    with _task_lock:
+        _task[key] = _task.get(key, 0) + value
+
+
+def task_metrics() -> Dict:
+    """Current task-level accumulators; maxDeviceMemoryBytes reads the
+    device pool's live high watermark when the pool is active."""
+    with _task_lock:
+        out = dict(_task)
+    try:
+        from .memory import device_pool
+
+        if device_pool.is_active():
+            out["maxDeviceMemoryBytes"] = \
+                device_pool.stats()["high_watermark"]
+    except Exception:  # noqa: BLE001 - pool not active / no GPU
+        pass
+    return out
+
+
+def reset_task_metrics() -> None:
+    with _task_lock:
+        _task.clear()
+        _task.update(_TASK_ZERO)
+
+
 def collect_metrics(exec_) -> List[Dict]:
     out = []
 

@@ -457,3 +457,30 @@ class TestPinnedHostPool:
             h.close()
         finally:
             host_pool.configure(None)
+
+
+def test_task_metrics_accumulators():
+    """GpuTaskMetrics analogue: spill bytes/time, retry counts and
+    semaphore wait surface through sr.task_metrics()."""
+    import numpy as np
+
+    import spark_rapids_amd as sr
+    from spark_rapids_amd.column import Column, ColumnBatch
+    from spark_rapids_amd.memory.retry import oom_injector, with_retry_split
+    from spark_rapids_amd.memory.spill import SpillableBatch
+    from spark_rapids_amd.metrics import reset_task_metrics, task_metrics
+    from spark_rapids_amd.types import INT64
+
+    reset_task_metrics()
+    c = Column.from_numpy(np.arange(100, dtype=np.int64), INT64)
+    h = SpillableBatch(ColumnBatch([c], 100))
+    h.spill_to_disk()  # host -> disk (host batch: to_host is no-op)
+    assert task_metrics()["spillToDiskBytes"] > 0
+    h.close()
+
+    oom_injector.arm(1)  # throw on first allocation check
+    out = with_retry_split(lambda b: b, ColumnBatch([c], 100))
+    assert len(out) >= 1
+    tm = task_metrics()
+    assert tm["retryCount"] + tm["splitAndRetryCount"] >= 1
+    reset_task_metrics()
