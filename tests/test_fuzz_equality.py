@@ -8,7 +8,8 @@ import numpy as np
 import pytest
 
 import spark_rapids_amd as sr
-from spark_rapids_amd import DType, col, count_star, sum_, min_, max_, avg
+from spark_rapids_amd import (DType, col, count, count_star, sum_, min_,
+                              max_, avg)
 
 decimal.getcontext().prec = 60
 
@@ -91,3 +92,64 @@ def test_fuzz_plan_equality(seed):
                 assert a == pytest.approx(b, rel=1e-6, abs=1e-9), (rg, rc)
             else:
                 assert a == b, (rg, rc)
+
+
+# ---------------------------------------------------------------------------
+# Special-value fuzz (reference analogue: data_gen.py special values fed
+# through assert_gpu_and_cpu_are_equal)
+# ---------------------------------------------------------------------------
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(4))
+def test_fuzz_special_values_exprs(seed):
+    from spark_rapids_amd.testing import (assert_gpu_and_cpu_are_equal,
+                                          gen_column)
+    from spark_rapids_amd.types import INT32, INT64, FLOAT64, STRING
+
+    n = 4000
+    data = {
+        "i": gen_column(INT32, n, seed * 11 + 1),
+        "l": gen_column(INT64, n, seed * 11 + 2),
+        "f": gen_column(FLOAT64, n, seed * 11 + 3),
+        "s": gen_column(STRING, n, seed * 11 + 4),
+        "d": gen_column(DType.decimal(9, 2), n, seed * 11 + 5),
+    }
+
+    def q(s):
+        df = s.create_dataframe(
+            {k: list(v) for k, v in data.items()},
+            dtypes={"d": DType.decimal(9, 2)})
+        return (df
+                .with_column("a", col("i").cast(sr.INT64) + col("l"))
+                .with_column("b", col("f") * 2.0 - col("f"))
+                .with_column("c", col("s").length())
+                .with_column("e", col("d") + col("d"))
+                .with_column("g", col("f") > 0.0)
+                .with_column("h", col("s").contains("a"))
+                .filter(col("i").is_not_null() | col("f").is_null()))
+
+    assert_gpu_and_cpu_are_equal(
+        q, conf={"spark.rapids.sql.incompatibleOps.enabled": True})
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(3))
+def test_fuzz_special_values_agg_sort(seed):
+    from spark_rapids_amd.testing import (assert_gpu_and_cpu_are_equal,
+                                          gen_column)
+    from spark_rapids_amd.types import INT32, FLOAT64, STRING
+
+    n = 6000
+    data = {
+        "k": gen_column(INT32, n, seed * 7 + 1, null_frac=0.15),
+        "f": gen_column(FLOAT64, n, seed * 7 + 2),
+        "s": gen_column(STRING, n, seed * 7 + 3),
+    }
+
+    def q(s):
+        df = s.create_dataframe({k: list(v) for k, v in data.items()})
+        return (df.group_by("k")
+                .agg(sum_(col("f")), count_star(), count(col("s")))
+                .sort("k"))
+
+    assert_gpu_and_cpu_are_equal(q, rel=1e-6)
