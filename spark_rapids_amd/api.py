@@ -235,8 +235,19 @@ class DataFrame:
         instrument(exec_)
         self._last_exec = exec_
         sem = GpuSemaphore.get()
-        with sem.held():
-            batches = [b.cpu() for b in exec_.execute()]
+        try:
+            with sem.held():
+                batches = [b.cpu() for b in exec_.execute()]
+        except (RuntimeError, MemoryError) as e:
+            from .tools import crashdump
+
+            p = crashdump.dump(e, exec_.tree_string())
+            if p:
+                import sys
+
+                print(f"[rapids] GPU error; crash bundle: {p}",
+                      file=sys.stderr)
+            raise
         if not batches:
             return ColumnBatch(
                 [Column.from_pylist([], f.dtype) for f in self.schema.fields], 0)

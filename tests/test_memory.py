@@ -484,3 +484,20 @@ def test_task_metrics_accumulators():
     tm = task_metrics()
     assert tm["retryCount"] + tm["splitAndRetryCount"] >= 1
     reset_task_metrics()
+
+
+def test_crashdump_bundle(tmp_path, monkeypatch):
+    """Fatal HIP-style errors write a diagnostic bundle (core-dump
+    handler analogue); non-GPU errors do not."""
+    import json
+
+    from spark_rapids_amd.tools import crashdump
+
+    monkeypatch.setenv("RAPIDS_CRASH_DIR", str(tmp_path))
+    p = crashdump.dump(RuntimeError("HIP error: out of memory"), "plan")
+    assert p is not None
+    info = json.loads(open(p).read())
+    assert "out of memory" in info["error"]
+    assert info["plan"] == "plan"
+    assert crashdump.dump(ValueError("not gpu")) is None
+    assert crashdump.dump(RuntimeError("plain python error")) is None
