@@ -97,6 +97,9 @@ def main():
     ap.add_argument("--partitions", type=int, default=8,
                     help="fact parquet files per rank")
     ap.add_argument("--data-dir", default=None)
+    ap.add_argument("--io-threads", type=int, default=None,
+                    help="override spark.rapids.sql.format.parquet."
+                         "multiThreadedRead.numThreads")
     ap.add_argument("--queries", default=None,
                     help="comma list to restrict the suite (debug)")
     ap.add_argument("--per-query", action="store_true",
@@ -127,7 +130,11 @@ def main():
         torch.distributed.barrier()
     stage_s = time.perf_counter() - stage_t0
 
-    session = Session({"spark.rapids.sql.enabled": use_gpu})
+    conf = {"spark.rapids.sql.enabled": use_gpu}
+    if args.io_threads:
+        conf["spark.rapids.sql.format.parquet.multiThreadedRead"
+             ".numThreads"] = args.io_threads
+    session = Session(conf)
     tables = _open_tables(session, paths)
     queries = args.queries.split(",") if args.queries else None
 
