@@ -36,7 +36,10 @@ class PinnedPool:
     def _new_slab(self, need: int) -> Optional[int]:
         size = max(self.slab_bytes, need)
         if self.reserved + size > self.capacity:
-            return None
+            # cap the last slab to the remaining capacity
+            size = self.capacity - self.reserved
+            if size < need:
+                return None
         try:
             t = torch.empty(size, dtype=torch.uint8, pin_memory=True)
         except RuntimeError:

@@ -154,6 +154,16 @@ class SpillableBatch:
         spill_store.unregister(self)
 
 
+def _compact(t: torch.Tensor) -> torch.Tensor:
+    """Host copy with exactly-sized storage: torch.save serializes the
+    WHOLE backing storage, so a small view of a pinned-pool slab must be
+    cloned or the save writes the entire slab to disk."""
+    h = t.cpu()
+    if h.untyped_storage().nbytes() != h.numel() * h.element_size():
+        h = h.clone()
+    return h
+
+
 def _flatten(batch):
     tensors: List[torch.Tensor] = []
     meta = {"num_rows": batch.num_rows, "cols": []}
@@ -162,11 +172,11 @@ def _flatten(batch):
                  "has_validity": c.validity is not None,
                  "has_offsets": c.offsets is not None,
                  "null_count": c._null_count}
-        tensors.append(c.data.cpu())
+        tensors.append(_compact(c.data))
         if c.validity is not None:
-            tensors.append(c.validity.cpu())
+            tensors.append(_compact(c.validity))
         if c.offsets is not None:
-            tensors.append(c.offsets.cpu())
+            tensors.append(_compact(c.offsets))
         meta["cols"].append(entry)
     return tensors, meta
 
