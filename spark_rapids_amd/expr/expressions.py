@@ -159,6 +159,11 @@ class Expression:
         """Entry count of a LIST or MAP value (null -> null)."""
         return ArraySize(self)
 
+    def array_contains(self, value) -> "ArrayContains":
+        """True when the LIST value contains `value` (Spark
+        array_contains; null list -> null)."""
+        return ArrayContains(self, value)
+
     def regexp_extract(self, pattern: str, group: int = 1) -> "RegexpExtract":
         return RegexpExtract(self, pattern, group)
 
@@ -658,6 +663,28 @@ class ElementAt(Expression):
 
     def __str__(self):
         return f"element_at({self.child}, {self.index})"
+
+
+class ArrayContains(Expression):
+    """array_contains(array, value) -> bool (GpuArrayContains)."""
+
+    def __init__(self, child: Expression, value):
+        self.child = child
+        self.value = value
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def dtype(self, schema: Schema) -> DType:
+        return BOOL
+
+    def eval(self, batch: ColumnBatch, schema: Schema) -> Column:
+        return ops.array_contains(self.child.eval(batch, schema),
+                                  self.value)
+
+    def __str__(self):
+        return f"array_contains({self.child}, {self.value!r})"
 
 
 class ArraySize(Expression):

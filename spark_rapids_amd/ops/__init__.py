@@ -54,6 +54,12 @@ def array_size(col: Column) -> Column:
     return backend_for(col).array_size(col)
 
 
+def array_contains(col: Column, value) -> Column:
+    """array_contains(array, value) -> bool; null array -> null
+    (GpuArrayContains analogue)."""
+    return backend_for(col).array_contains(col, value)
+
+
 def element_at(col: Column, index: int) -> Column:
     return backend_for(col).element_at(col, index)
 

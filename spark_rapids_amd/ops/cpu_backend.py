@@ -656,6 +656,11 @@ def element_at(col: Column, index: int) -> Column:
     return Column.from_pylist(out, col.dtype.children[0])
 
 
+def array_contains(col: Column, value) -> Column:
+    out = [None if v is None else (value in v) for v in col.to_pylist()]
+    return Column.from_pylist(out, DType.bool_())
+
+
 def map_get(col: Column, key) -> Column:
     out = []
     for v in col.to_pylist():  # dicts (last-win on duplicate keys)

@@ -657,6 +657,42 @@ def element_at(col: Column, index: int) -> Column:
     return _gather_col(col.child, srcm.data, n, maybe_negative=True)
 
 
+def array_contains(col: Column, value) -> Column:
+    """Entry-level compare + per-segment any() via scatter amax."""
+    n = col.size
+    if n == 0:
+        return _empty_col(DType.bool_())
+    elem = col.child
+    ne = elem.size
+    hit = torch.zeros(n, dtype=torch.uint8, device="cuda")
+    if ne:
+        if elem.dtype.id is TypeId.STRING:
+            sw = str_predicate("starts_with", elem, value)
+            ln = unary_op("length", elem, DType.int32())
+            lm = binary_op_scalar("eq", ln, len(value), DType.bool_())
+            match = binary_op("and", sw, lm, DType.bool_())
+        else:
+            match = binary_op_scalar("eq", elem, value, DType.bool_())
+        mt = match.data[:ne].bool()
+        if elem.validity is not None:
+            ev = torch.empty(ne, dtype=torch.uint8, device="cuda")
+            ext.mask_expand(elem.validity.data_ptr(), ev.data_ptr(),
+                            False, ne, _stream())
+            mt &= ev.bool()
+        offs = col.offsets.long()
+        counts = offs[1:n + 1] - offs[:n]
+        seg = torch.repeat_interleave(
+            torch.arange(n, device="cuda", dtype=torch.int64), counts)
+        if mt.any():
+            hit.scatter_reduce_(0, seg[mt],
+                                torch.ones(int(mt.sum()),
+                                           dtype=torch.uint8,
+                                           device="cuda"),
+                                reduce="amax", include_self=True)
+    v = col.validity.clone() if col.validity is not None else None
+    return Column(DType.bool_(), n, hit, v, null_count=col._null_count)
+
+
 def map_get(col: Column, key) -> Column:
     """element_at(map, key): entry-level key compare + per-segment
     last-match reduction (torch scatter_reduce amax over the entry iota),

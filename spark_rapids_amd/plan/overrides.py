@@ -150,6 +150,10 @@ class Tagger:
         elif type(e).__name__ == "StrSplit":
             if any(c in e.delimiter for c in ".\\+*?()[]{}|^$"):
                 out.append("regex split delimiters run on CPU")
+        elif type(e).__name__ == "ArrayContains":
+            et = e.child.dtype(schema).children[0]
+            if et.is_nested:
+                out.append("array_contains over nested elements on CPU")
         elif type(e).__name__ in ("ArraySize", "ElementAt"):
             cdt = e.child.dtype(schema)
             if cdt.id is TypeId.MAP and type(e).__name__ == "ElementAt" \

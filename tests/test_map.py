@@ -127,3 +127,43 @@ def test_map_int_key_gpu_matches_cpu():
     g = q(sr.Session())
     c = q(sr.Session({"spark.rapids.sql.enabled": False}))
     assert g == c
+
+
+def test_array_contains_cpu(cpu):
+    from spark_rapids_amd.types import DType, INT64
+
+    lt = DType.list_(INT64)
+    df = cpu.create_dataframe(
+        {"a": [[1, 2, 3], [], None, [5, None]]}, dtypes={"a": lt})
+    out = df.select(col("a").array_contains(2).alias("c2"),
+                    col("a").array_contains(5).alias("c5")).to_pydict()
+    assert out["c2"] == [True, False, None, False]
+    assert out["c5"] == [False, False, None, True]
+
+
+@pytest.mark.gpu
+def test_array_contains_gpu_matches_cpu():
+    rng = np.random.default_rng(19)
+    n = 4000
+    vals = [None if i % 17 == 0 else
+            [int(v) for v in rng.integers(0, 10, int(rng.integers(0, 6)))]
+            for i in range(n)]
+    words = ["aa", "bb", "cc", "dd"]
+    svals = [None if i % 13 == 0 else
+             [words[int(v)] for v in rng.integers(0, 4,
+                                                  int(rng.integers(0, 5)))]
+             for i in range(n)]
+    from spark_rapids_amd.types import DType, INT64, STRING
+
+    def q(s):
+        df = s.create_dataframe(
+            {"a": [list(v) if v is not None else None for v in vals],
+             "s": [list(v) if v is not None else None for v in svals]},
+            dtypes={"a": DType.list_(INT64), "s": DType.list_(STRING)})
+        return df.select(col("a").array_contains(7).alias("c"),
+                         col("s").array_contains("bb").alias("d")
+                         ).to_pydict()
+
+    g = q(sr.Session())
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert g == c
