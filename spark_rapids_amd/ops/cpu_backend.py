@@ -193,6 +193,21 @@ def _binary_impl(op, a, av, b, bv, in_dtype: DType, out_dtype: DType) -> Column:
         if in_dtype.id is TypeId.STRING:
             # elementwise python compare on object arrays
             res = np.array([_str_cmp(op, x, y) for x, y in zip(a, b)], dtype=bool)
+        elif a.dtype.kind == "f" or (hasattr(b, "dtype")
+                                     and getattr(b, "dtype", None) is not None
+                                     and getattr(b.dtype, "kind", "") == "f"):
+            # Spark float ordering: NaN == NaN and NaN is GREATER than
+            # every other value (matches the GPU kernels and Spark SQL)
+            na = np.isnan(a.astype(np.float64))
+            bb = np.asarray(b, dtype=np.float64)
+            nb = np.isnan(bb) if bb.shape else np.full(n, np.isnan(bb))
+            with np.errstate(invalid="ignore"):
+                eq = (a == b) | (na & nb)
+                lt = np.where(na, False, np.where(nb, True, a < b))
+                gt = np.where(nb, False, np.where(na, True, a > b))
+            res = {"eq": lambda: eq, "ne": lambda: ~eq,
+                   "lt": lambda: lt, "le": lambda: lt | eq,
+                   "gt": lambda: gt, "ge": lambda: gt | eq}[op]()
         else:
             with np.errstate(invalid="ignore"):
                 res = {

@@ -106,3 +106,15 @@ def test_int_promotion(session):
 def test_filter_drops_null_predicate(df):
     rows = df.filter(col("i") > 1).select("i").to_pydict()["i"]
     assert rows == [2, 5]
+
+
+def test_float_nan_ordering_spark_semantics(session):
+    """Spark float ordering: NaN == NaN, NaN greater than everything
+    (found by the special-value fuzz; CPU backend previously used raw
+    numpy semantics where all NaN compares are false)."""
+    nan, inf = float("nan"), float("inf")
+    df = session.create_dataframe({"f": [nan, 1.0, -0.0, inf]})
+    out = df.select((col("f") > 0.0).alias("g"),
+                    (col("f") == col("f")).alias("e")).to_pydict()
+    assert out["g"] == [True, True, False, True]
+    assert out["e"] == [True, True, True, True]
