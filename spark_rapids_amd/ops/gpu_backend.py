@@ -1381,6 +1381,11 @@ def reduce(op: str, col: Column):
                                 "max": -2 ** 63, "count": 0}[op],
                          dtype=torch.int64, device="cuda")
     cnt = torch.zeros(1, dtype=torch.int64, device="cuda")
+    if op == "count" and not col.dtype.is_fixed_width:
+        # count over STRING/nested: only the validity matters
+        nn8 = torch.ones(max(n, 1), dtype=torch.uint8, device="cuda")[:n]
+        col = Column(DType.bool_(), n, nn8, col.validity,
+                     null_count=col._null_count)
     ext.reduce(_RED.get(op, 0), _ht(col.dtype), col.data.data_ptr(),
                _ptr(col.validity), acc.data_ptr(), cnt.data_ptr(), n, s)
     c = int(cnt.item())
@@ -1530,6 +1535,13 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                           int(op.split(":", 1)[1]), s)
             allocs.append(("collect", out_dtype, False, col, None))
             continue
+        if op == "count" and vc is not None and not vc.dtype.is_fixed_width:
+            # count over STRING/nested only needs the validity: swap in a
+            # u8 non-null indicator column so no typed accumulator is hit
+            nn8 = torch.ones(max(vc.size, 1), dtype=torch.uint8,
+                             device="cuda")[:vc.size]
+            vc = Column(DType.bool_(), vc.size, nn8, vc.validity,
+                        null_count=vc._null_count)
         acc_is_double = out_dtype.is_floating or (
             vc is not None and vc.dtype.is_floating)
         acc = torch.empty(max(nrep * ngroups, 1),
