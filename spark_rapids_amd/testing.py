@@ -20,6 +20,8 @@ def _rows_equal(g, c, approx_float: bool, rel: float) -> bool:
     if isinstance(c, float) and isinstance(g, float):
         if math.isnan(c) or math.isnan(g):
             return math.isnan(c) and math.isnan(g)
+        if math.isinf(c) or math.isinf(g):
+            return g == c  # same-signed infinity only
         if approx_float:
             tol = rel * max(abs(c), abs(g), 1e-300)
             return abs(g - c) <= max(tol, 1e-12)
