@@ -153,3 +153,41 @@ def test_fuzz_special_values_agg_sort(seed):
                 .sort("k"))
 
     assert_gpu_and_cpu_are_equal(q, rel=1e-6)
+
+
+@pytest.mark.gpu
+@pytest.mark.parametrize("seed", range(3))
+def test_fuzz_special_values_window_join(seed):
+    from spark_rapids_amd import win_sum, row_number
+    from spark_rapids_amd.testing import (assert_gpu_and_cpu_are_equal,
+                                          gen_column)
+    from spark_rapids_amd.types import INT32, FLOAT64
+
+    n = 5000
+    data = {
+        "p": [abs(v) % 40 if v is not None else None
+              for v in gen_column(INT32, n, seed * 13 + 1, null_frac=0.05)],
+        "o": gen_column(FLOAT64, n, seed * 13 + 2, null_frac=0.05),
+        "v": gen_column(FLOAT64, n, seed * 13 + 3),
+        "k": gen_column(INT32, n, seed * 13 + 4, null_frac=0.2),
+        "t": list(range(n)),
+    }
+    rdata = {
+        "k": gen_column(INT32, 800, seed * 13 + 5, null_frac=0.1),
+        "w": gen_column(FLOAT64, 800, seed * 13 + 6),
+    }
+
+    def q(s):
+        df = s.create_dataframe({k: list(v) for k, v in data.items()})
+        r = s.create_dataframe({k: list(v) for k, v in rdata.items()})
+        # unique tiebreaker: special values repeat, and row_number over
+        # ties is backend-nondeterministic
+        df = df.with_column(
+            "rn", row_number().over(["p"], ["o", "t"]))
+        df = df.with_column(
+            "ws", win_sum(col("v")).over(["p"], ["o", "t"],
+                                         rows_between=(-2, 2)))
+        return df.join(r, on="k", how="left").select(
+            "p", "rn", "ws", "w")
+
+    assert_gpu_and_cpu_are_equal(q, rel=1e-6)
