@@ -18,9 +18,13 @@ import sys
 from collections import defaultdict
 
 
-def _find_csv(d, needle):
-    for p in sorted(glob.glob(os.path.join(d, "**", "*.csv"),
-                              recursive=True)):
+def _find_csv(d, needle, prefer=None):
+    cands = sorted(glob.glob(os.path.join(d, "**", "*.csv"),
+                             recursive=True))
+    if prefer:
+        cands = [p for p in cands if prefer in os.path.basename(p)] + \
+            [p for p in cands if prefer not in os.path.basename(p)]
+    for p in cands:
         with open(p) as f:
             head = f.readline()
         if needle in head:
@@ -32,7 +36,8 @@ def load_stats(d):
     """kernel -> (total_ns, calls) from *_kernel_stats.csv or kernel
     trace."""
     out = {}
-    p = _find_csv(d, "TotalDurationNs") or _find_csv(d, "DurationNs")
+    p = _find_csv(d, "TotalDurationNs", prefer="kernel_stats") \
+        or _find_csv(d, "DurationNs", prefer="kernel_trace")
     if p is None:
         return out
     with open(p) as f:
