@@ -45,7 +45,8 @@ void hipdf_gather_table(const void*, int, const void*, int64_t, hipStream_t);
 void hipdf_gather_str_lens(const void*, const void*, void*, int64_t,
                            hipStream_t);
 void hipdf_gather_str_bytes(const void*, const void*, const void*,
-                            const void*, void*, int64_t, hipStream_t);
+                            const void*, void*, int64_t, int64_t,
+                            hipStream_t);
 void hipdf_narrow_i64_i32(const void*, void*, int64_t, hipStream_t);
 void hipdf_copy_valid_range(const void*, int, int64_t, void*, int64_t,
                             hipStream_t);
@@ -675,9 +676,10 @@ PYBIND11_MODULE(hipdf, m) {
   });
   m.def("gather_str_bytes", [](int64_t in_bytes, int64_t in_offs, int64_t idx,
                                int64_t out_offs, int64_t out_bytes,
-                               int64_t n_out, int64_t stream) {
+                               int64_t n_out, int64_t total_bytes,
+                               int64_t stream) {
     hipdf_gather_str_bytes(P(in_bytes), P(in_offs), P(idx), P(out_offs),
-                           PM(out_bytes), n_out, S(stream));
+                           PM(out_bytes), n_out, total_bytes, S(stream));
     check_async();
   });
   m.def("narrow_i64_i32", [](int64_t in, int64_t out, int64_t n,

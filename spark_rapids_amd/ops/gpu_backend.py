@@ -1158,7 +1158,7 @@ def _gather_col(c: Column, idx: torch.Tensor, n_out: int,
         if total:
             ext.gather_str_bytes(c.data.data_ptr(), c.offsets.data_ptr(),
                                  idx.data_ptr(), scanned.data_ptr(),
-                                 out_bytes.data_ptr(), n_out, s)
+                                 out_bytes.data_ptr(), n_out, total, s)
         offs = torch.empty(n_out + 1, dtype=torch.int32, device="cuda")
         ext.narrow_i64_i32(scanned.data_ptr(), offs.data_ptr(), n_out, s)
         offs[n_out] = total
