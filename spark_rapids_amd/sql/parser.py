@@ -698,13 +698,31 @@ class Parser:
 
             return StrSplit(args[0], args[1].value)
         if name == "element_at":
-            from ..expr.expressions import ElementAt
+            from ..expr.expressions import ElementAt, Literal
 
+            # int literal -> array index; any other literal -> map key
+            if isinstance(args[1], Literal) and \
+                    not isinstance(args[1].value, int):
+                return ElementAt(args[0], args[1].value)
             return ElementAt(args[0], self._int_arg(args[1]))
         if name == "size":
             from ..expr.expressions import ArraySize
 
             return ArraySize(args[0])
+        if name == "array_contains":
+            from ..expr.expressions import ArrayContains, Literal
+
+            return ArrayContains(args[0], args[1].value
+                                 if isinstance(args[1], Literal)
+                                 else args[1])
+        if name in ("map_keys", "map_values", "map_entries"):
+            from ..expr.expressions import MapView
+
+            return MapView(args[0], name.split("_", 1)[1])
+        if name == "map":
+            from ..expr.expressions import CreateMap
+
+            return CreateMap(args)
         if name == "repeat":
             from ..expr.expressions import repeat_str
 

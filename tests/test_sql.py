@@ -256,3 +256,23 @@ def test_sql_conditional_join(s):
                  "AND a < b").to_pydict()
     assert sorted(zip(out2["a"], out2["b"]), key=repr) == \
         [(10, 15), (20, None), (30, None)]
+
+
+def test_sql_map_array_functions(s):
+    from spark_rapids_amd.types import DType, INT64
+
+    df = s.create_dataframe(
+        {"a": [[1, 2], [3], None], "x": [10, 20, 30]},
+        dtypes={"a": DType.list_(INT64)})
+    s.register("maf_t", df)
+    out = s.sql("SELECT size(a) AS n, element_at(a, 1) AS e, "
+                "array_contains(a, 3) AS c FROM maf_t").to_pydict()
+    assert out["n"] == [2, 1, None]
+    assert out["e"] == [1, 3, None]
+    assert out["c"] == [False, True, None]
+    out2 = s.sql("SELECT element_at(map('k', x), 'k') AS v "
+                 "FROM maf_t").to_pydict()
+    assert out2["v"] == [10, 20, 30]
+    out3 = s.sql("SELECT size(map_keys(map('k', x, 'j', x))) AS nk "
+                 "FROM maf_t").to_pydict()
+    assert out3["nk"] == [2, 2, 2]
