@@ -129,6 +129,22 @@ class DataFrame:
                          L.NestedLoopJoin(self.plan, other.plan, [], [],
                                           how, condition=condition))
 
+    def cache(self) -> "DataFrame":
+        """Materialize this DataFrame's result on first action as
+        compressed in-memory parquet batches and serve later actions
+        from them (ParquetCachedBatchSerializer analogue)."""
+        if not isinstance(self.plan, L.CacheData):
+            self.plan = L.CacheData(self.plan)
+        return self
+
+    persist = cache
+
+    def unpersist(self) -> "DataFrame":
+        if isinstance(self.plan, L.CacheData):
+            self.plan.store = None
+            self.plan = self.plan.child
+        return self
+
     def sort(self, *keys: str, descending: Union[bool, List[bool]] = False) -> "DataFrame":
         ks = list(keys)
         if isinstance(descending, bool):

@@ -263,6 +263,24 @@ class Column:
             col = Column(dtype, n, data, validity,
                          null_count=int(n - valid.sum()) if validity is not None else 0)
             return col.to(device) if device != "cpu" else col
+        if dtype.id is TypeId.DECIMAL64:
+            # same input convention as dec128_pack: Decimal/float inputs
+            # are VALUES (scaled here), plain ints are raw unscaled units
+            import decimal as _dec
+
+            conv = []
+            for v in values:
+                if v is None:
+                    conv.append(0)
+                elif isinstance(v, float):
+                    conv.append(int(round(v * 10 ** dtype.scale)))
+                elif isinstance(v, _dec.Decimal):
+                    conv.append(int(v.scaleb(dtype.scale)))
+                else:
+                    conv.append(int(v))
+            dense = np.array(conv, dtype=np.int64)
+            return Column.from_numpy(dense, dtype,
+                                     None if valid.all() else valid, device)
         np_dt = dtype.numpy_dtype()
         fill = 0
         dense = np.array([v if v is not None else fill for v in values], dtype=np_dt)

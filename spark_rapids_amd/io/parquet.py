@@ -487,3 +487,29 @@ def _column_to_arrow(c: Column, dtype: DType):
     if dtype.id is TypeId.TIMESTAMP:
         return pa.array(c.to_numpy(), type=pa.timestamp("us"), mask=mask)
     return pa.array(c.to_numpy(), mask=mask)
+
+
+def batch_to_parquet_bytes(batch: ColumnBatch, schema: Schema,
+                           compression: str = "zstd") -> bytes:
+    """Serialize a host batch to an in-memory compressed parquet blob
+    (reference analogue: ParquetCachedBatchSerializer — df.cache()
+    stores compressed parquet batches instead of raw columns)."""
+    import io as _io
+
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    arrays = [_column_to_arrow(c.cpu(), f.dtype)
+              for f, c in zip(schema.fields, batch.columns)]
+    tbl = pa.table(dict(zip(schema.names, arrays)))
+    buf = _io.BytesIO()
+    pq.write_table(tbl, buf, compression=compression)
+    return buf.getvalue()
+
+
+def parquet_bytes_to_batch(blob: bytes) -> ColumnBatch:
+    import io as _io
+
+    import pyarrow.parquet as pq
+
+    return arrow_table_to_batch(pq.read_table(_io.BytesIO(blob)))

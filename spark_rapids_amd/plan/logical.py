@@ -158,6 +158,28 @@ class Join(LogicalPlan):
         return f"Join({self.how})"
 
 
+class CacheData(LogicalPlan):
+    """df.cache(): materialize the child once as compressed in-memory
+    parquet blobs and serve every later action from them (reference
+    analogue: ParquetCachedBatchSerializer — Spark df.cache() with
+    parquet-compressed cached batches)."""
+
+    def __init__(self, child: LogicalPlan):
+        self.child = child
+        self.store = None  # List[bytes] after first materialization
+
+    @property
+    def children(self):
+        return (self.child,)
+
+    def schema(self) -> Schema:
+        return self.child.schema()
+
+    def name(self) -> str:
+        state = "materialized" if self.store is not None else "lazy"
+        return f"CacheData({state})"
+
+
 class NestedLoopJoin(Join):
     """Join on an arbitrary (non-equi) condition with no equality keys:
     every (left, right) row pair is tested, chunked to bound memory

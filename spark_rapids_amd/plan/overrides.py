@@ -403,6 +403,10 @@ def _convert(node: L.LogicalPlan, conf: RapidsConf, tagger: Tagger,
             return P.CoalesceBatchesExec(scan, conf.get(BATCH_SIZE_BYTES))
         return scan
 
+    if isinstance(node, L.CacheData):
+        inner = _convert(node.child, conf, tagger, gpu_wanted)
+        return P.CacheDataExec(node, _ensure_device(inner, "cpu"))
+
     if isinstance(node, L.Cached):
         # one physical exec per logical node, however many parents
         # reference it (hierarchical rollup levels share their base)

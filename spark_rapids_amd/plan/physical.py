@@ -815,6 +815,32 @@ class HashJoinExec(PhysicalExec):
         return f"{self.name()}({self.how}, {pairs})"
 
 
+class CacheDataExec(PhysicalExec):
+    """Serve a CacheData node: first execution materializes the child
+    into compressed parquet blobs held on the logical node (so the store
+    survives across actions on the same DataFrame); later executions
+    decode the blobs (PCBS analogue). Always a host-side exec — the
+    planner inserts a transfer when the parent runs on GPU."""
+
+    def __init__(self, node, child: PhysicalExec):
+        super().__init__("cpu", child.schema, [child])
+        self.node = node
+
+    def execute(self) -> Iterator[ColumnBatch]:
+        from ..io.parquet import (batch_to_parquet_bytes,
+                                  parquet_bytes_to_batch)
+
+        if self.node.store is None:
+            blobs = []
+            for b in self.children[0].execute():
+                if b.num_rows:
+                    blobs.append(batch_to_parquet_bytes(b.cpu(),
+                                                        self.schema))
+            self.node.store = blobs
+        for blob in self.node.store:
+            yield parquet_bytes_to_batch(blob)
+
+
 class NestedLoopJoinExec(HashJoinExec):
     """Broadcast nested-loop join: all (left, right) row pairs streamed
     through the join condition in bounded chunks (reference analogue:
