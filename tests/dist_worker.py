@@ -202,6 +202,26 @@ def main():
     td.all_reduce(t2)
     assert int(t2.item()) == n_total
 
+    # conditional hash join across ranks: per-rank sharded probe against
+    # a replicated build, non-equi condition compacts pairs; the keyless
+    # count aggregates globally so every rank reports the global count
+    cright = s.create_dataframe({
+        "k": np.arange(97, dtype=np.int64),
+        "thr": np.arange(97, dtype=np.float64) * 50.0,
+    }, replicated=True)
+    cj = df.join(cright, on="k", condition=col("v") < col("thr")) \
+        .agg(count_star()).collect()
+    exp_cj = sum(1 for v in all_rows.tolist()
+                 if float(v) < (v % 97) * 50.0)
+    assert cj[0][0] == exp_cj, (cj, exp_cj)
+
+    # nested-loop join across ranks (replicated right, semi pairs)
+    nlr = s.create_dataframe({"b": [100.0, 5000.0]}, replicated=True)
+    nl = df.join_nl(nlr, col("v") < col("b"), "semi") \
+        .agg(count_star()).collect()
+    exp_nl = sum(1 for v in all_rows.tolist() if float(v) < 5000.0)
+    assert nl[0][0] == exp_nl, (nl, exp_nl)
+
     td.barrier()
     if rank == 0:
         print("DIST_OK")
