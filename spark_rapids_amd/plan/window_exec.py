@@ -349,7 +349,29 @@ class WindowExec(PhysicalExec):
             work_t = FLOAT64 if use_f64 else INT64
             v64 = gb.cast(Column(vc.dtype, n, vc.data, None, null_count=0),
                           work_t)
-            vz = gb.binary_op("mul", v64, gb.cast(nn_col, work_t), work_t)
+            nonfin = None
+            if use_f64:
+                # prefix-sum framing is wrong for non-finite values (one
+                # inf/NaN poisons every later frame) and a null slot's
+                # garbage bits could be non-finite: zero both out of the
+                # scan and patch affected frames from exact non-finite
+                # frame COUNTS afterwards (found by special-value fuzz)
+                nnb = valid_u8[:n].to(torch.bool)
+                vv = v64.data[:n]
+                pos_m = torch.isposinf(vv) & nnb
+                neg_m = torch.isneginf(vv) & nnb
+                nan_m = torch.isnan(vv) & nnb
+                bad = pos_m | neg_m | nan_m
+                if bool(bad.any()):
+                    nonfin = (pos_m, neg_m, nan_m)
+                vz = Column(work_t, n,
+                            torch.where(nnb & ~bad, vv,
+                                        torch.zeros((), dtype=vv.dtype,
+                                                    device="cuda")),
+                            None, null_count=0)
+            else:
+                vz = gb.binary_op("mul", v64, gb.cast(nn_col, work_t),
+                                  work_t)
             nn64 = gb.cast(nn_col, INT64)
             # inclusive global prefix sums; gather_fixed zero-fills index -1
             if use_f64:
@@ -384,6 +406,26 @@ class WindowExec(PhysicalExec):
             if op == "count":
                 return cnt_r
             sum_r = gb.binary_op("sub", incl_b, incl_a, work_t)
+            if nonfin is not None:
+                def _frame_count(mask):
+                    m64 = mask.to(torch.int64).contiguous()
+                    excl_m, _ = gb._exclusive_scan_i64(m64)
+                    incl_m = excl_m + m64
+                    cb = self._gather_i32(incl_m, b_idx.data, n)
+                    ca = self._gather_i32(incl_m, am1.data, n)
+                    return cb - ca
+
+                pc = _frame_count(nonfin[0])
+                nc = _frame_count(nonfin[1])
+                qc = _frame_count(nonfin[2])
+                st = sum_r.data[:n]
+                st = torch.where((qc > 0) | ((pc > 0) & (nc > 0)),
+                                 torch.full_like(st, float("nan")), st)
+                st = torch.where((pc > 0) & (nc == 0) & (qc == 0),
+                                 torch.full_like(st, float("inf")), st)
+                st = torch.where((nc > 0) & (pc == 0) & (qc == 0),
+                                 torch.full_like(st, float("-inf")), st)
+                sum_r = Column(work_t, n, st, None, null_count=0)
             ov = torch.empty(mask_nbytes(n), dtype=torch.uint8,
                              device="cuda")
             ext.mask_from_nonzero(cnt_r.data.data_ptr(), ov.data_ptr(), n, s)
@@ -397,7 +439,20 @@ class WindowExec(PhysicalExec):
             work_t = FLOAT64 if use_f64 else INT64
             v64 = gb.cast(Column(vc.dtype, n, vc.data, None, null_count=0),
                           work_t)
-            vz = gb.binary_op("mul", v64, gb.cast(nn_col, work_t), work_t)
+            if use_f64:
+                # select (not multiply) so non-finite garbage in null
+                # slots cannot leak a NaN into the running prefix; valid
+                # non-finite VALUES keep exact prefix semantics here
+                # (frame is always [start, i])
+                nnb = valid_u8[:n].to(torch.bool)
+                vz = Column(work_t, n,
+                            torch.where(nnb, v64.data[:n],
+                                        torch.zeros((), dtype=torch.float64,
+                                                    device="cuda")),
+                            None, null_count=0)
+            else:
+                vz = gb.binary_op("mul", v64, gb.cast(nn_col, work_t),
+                                  work_t)
             run_cnt = self._running_sum_i64(
                 gb.cast(nn_col, INT64).data, heads, seg_start_col, n)
             if op == "count":

@@ -75,7 +75,10 @@ def test_cache_gpu_plan(tmp_path):
          "v": [float(v) for v in rng.uniform(0, 1, 20000)]}).cache()
     g1 = df.group_by("k").agg(sum_(col("v"))).sort("k").collect()
     g2 = df.group_by("k").agg(sum_(col("v"))).sort("k").collect()
-    assert g1 == g2
+    assert [r[0] for r in g1] == [r[0] for r in g2]
+    for a, b in zip(g1, g2):
+        # float atomics reassociate between runs; values match approx
+        assert abs(a[1] - b[1]) <= 1e-9 * max(abs(a[1]), 1.0)
     c = sr.Session({"spark.rapids.sql.enabled": False})
     dfc = c.create_dataframe(
         {"k": [r[0] for r in g1], "s": [r[1] for r in g1]})
