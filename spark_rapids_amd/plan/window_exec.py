@@ -745,12 +745,33 @@ def _compute(w: WindowExpr, table: ColumnBatch, cs, n, heads, ochange, idx,
                     ok[i] = True
             res = res.astype(out_dt.numpy_dtype())
             return _make(res, ok if not ok.all() else None, out_dt)
-        csv = np.cumsum(vv)
+        # exact non-finite frame semantics: zero NaN/inf out of the
+        # prefix (one would poison every later frame) and patch frames
+        # from their non-finite COUNTS (mirrors the GPU path)
+        pos_m = np.isposinf(vv)
+        neg_m = np.isneginf(vv)
+        nan_m = np.isnan(vv)
+        vclean = np.where(pos_m | neg_m | nan_m, 0.0, vv)
+        csv = np.cumsum(vclean)
         csn = np.cumsum(cnt_f)
         am1 = np.maximum(a - 1, 0)
+
+        def _fcnt(mask):
+            cs = np.cumsum(mask.astype(np.int64))
+            basec = np.where(a > 0, cs[am1], 0)
+            return np.where(empty_frame, 0, cs[b] - basec)
+
         base_v = np.where(a > 0, csv[am1], 0.0)
         base_n = np.where(a > 0, csn[am1], 0)
         sum_r = np.where(empty_frame, 0.0, csv[b] - base_v)
+        if pos_m.any() or neg_m.any() or nan_m.any():
+            pc, nc, qc = _fcnt(pos_m), _fcnt(neg_m), _fcnt(nan_m)
+            sum_r = np.where((qc > 0) | ((pc > 0) & (nc > 0)), np.nan,
+                             sum_r)
+            sum_r = np.where((pc > 0) & (nc == 0) & (qc == 0), np.inf,
+                             sum_r)
+            sum_r = np.where((nc > 0) & (pc == 0) & (qc == 0), -np.inf,
+                             sum_r)
         cnt_r = np.where(empty_frame, 0, csn[b] - base_n)
         if op == "count":
             return _make(cnt_r.astype(np.int64), None, out_dt)

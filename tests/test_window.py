@@ -398,3 +398,28 @@ def test_gpu_minmax_windows_match_cpu():
             assert (a is None) == (b is None), k
             if a is not None:
                 assert a == pytest.approx(b), k
+
+
+def test_framed_sum_nonfinite_frames(session):
+    """A NaN/inf value must only affect frames CONTAINING it — prefix-sum
+    framing used to poison every later frame (found by the special-value
+    fuzz; both backends now patch from non-finite frame counts)."""
+    inf, nan = float("inf"), float("nan")
+    df = session.create_dataframe({
+        "p": [1] * 6, "t": list(range(6)),
+        "v": [1.0, inf, 2.0, 4.0, nan, 8.0]})
+    out = df.with_column("w", win_sum(col("v")).over(
+        ["p"], ["t"], rows_between=(0, 1))).to_pydict()["w"]
+    # frames: [1,inf]=inf, [inf,2]=inf, [2,4]=6, [4,nan]=nan,
+    # [nan,8]=nan, [8]=8
+    import math
+    assert out[0] == inf and out[1] == inf
+    assert out[2] == 6.0
+    assert math.isnan(out[3]) and math.isnan(out[4])
+    assert out[5] == 8.0
+    df2 = session.create_dataframe({
+        "p": [1] * 3, "t": [0, 1, 2], "v": [inf, -inf, 5.0]})
+    out2 = df2.with_column("w", win_sum(col("v")).over(
+        ["p"], ["t"], rows_between=(-1, 0))).to_pydict()["w"]
+    assert out2[0] == inf and math.isnan(out2[1])
+    assert out2[2] == -inf  # frame [-inf, 5]
