@@ -199,7 +199,8 @@ class _ChunkDecoder:
     def __init__(self, raw: bytes, phys: str, dtype: DType, max_def: int,
                  codec, num_values: int, ext, stream: int,
                  max_rep: int = 0, list_dtype: Optional[DType] = None,
-                 opt_list: bool = True, opt_elem: bool = True):
+                 opt_list: bool = True, opt_elem: bool = True,
+                 capture_def: bool = False):
         self.raw = raw
         self.phys = phys
         self.dtype = dtype  # ELEMENT dtype when max_rep > 0
@@ -214,6 +215,11 @@ class _ChunkDecoder:
         self.list_dtype = list_dtype
         self.opt_list = opt_list
         self.opt_elem = opt_elem
+        # when set, the chunk's decoded def levels are stashed on
+        # self.captured_def (None = trivially all max_def) so a STRUCT
+        # parent can derive its own validity from any leaf's levels
+        self.capture_def = capture_def
+        self.captured_def: Optional[torch.Tensor] = None
         self.dict_fixed: Optional[torch.Tensor] = None  # device dict values
         self.dict_str: Optional[tuple] = None  # (offsets cuda, bytes cuda)
 
@@ -241,12 +247,20 @@ class _ChunkDecoder:
         if fast is not None:
             return fast
         page_cols = []
+        cap = []
         for pg in pages:
             levels = None
             if not pg.all_valid and self.max_def > 0:
                 levels = self._decode_levels(pg.levels, pg.n)
+            cap.append(levels)
             page_cols.append(self._materialize(pg.values, pg.n, levels,
                                                pg.encoding))
+        if self.capture_def and any(c is not None for c in cap):
+            full = [c if c is not None else
+                    torch.full((pg.n,), self.max_def, dtype=torch.int32,
+                               device="cuda")
+                    for c, pg in zip(cap, pages)]
+            self.captured_def = torch.cat(full) if len(full) > 1 else full[0]
         if len(page_cols) == 1:
             return page_cols[0]
         from ..ops import gpu_backend
@@ -472,8 +486,11 @@ class _ChunkDecoder:
         """(mask, valid_idx, n_valid) for the whole chunk: decode all
         pages' def-level RLE streams at once, then run the validity
         chain once over the contiguous levels array."""
+        bw = max(1, self.max_def.bit_length())
         levels = self._rle_streams_to_i32(
-            [(p.levels, p.n, 1) for p in pages], total)
+            [(p.levels, p.n, bw) for p in pages], total)
+        if self.capture_def:
+            self.captured_def = levels
         return self._valid_parts(levels, total)
 
     def _chunk_plain_fixed(self, pages, total, all_valid) -> Column:
@@ -623,12 +640,8 @@ class _ChunkDecoder:
 
     # -- data pages ------------------------------------------------------
     def _decode_levels(self, rle: bytes, n: int) -> torch.Tensor:
-        dev = torch.from_numpy(
-            np.frombuffer(rle, dtype=np.uint8).copy()).cuda()
-        out = torch.empty(n, dtype=torch.int32, device="cuda")
-        self.ext.rle_hybrid_decode(dev.data_ptr(), dev.numel(), 1,
-                                   out.data_ptr(), n, self.s)
-        return out
+        return self._decode_level_stream(rle, n,
+                                         max(1, self.max_def.bit_length()))
 
     def _valid_parts(self, levels: Optional[torch.Tensor], n: int):
         """(validity_mask or None, valid_idx tensor, n_valid)."""
@@ -856,42 +869,82 @@ def read_parquet_gpu(path: str, columns: List[str],
                 continue  # pruned by row-group min/max statistics
             rgmd = md.row_group(rg)
             cols = []
-            for name in columns:
-                field = arrow_schema.field(name)
-                dtype = arrow_to_dtype(field.type)
-                j = name_to_idx.get(name)
-                list_kw = {}
-                if j is None:
-                    # nested field: leaves carry dotted paths
-                    cands = [jj for p, jj in name_to_idx.items()
-                             if p.startswith(name + ".")]
-                    if dtype.id is TypeId.LIST and len(cands) == 1 \
-                            and not dtype.children[0].is_nested:
-                        j = cands[0]
-                        list_kw = dict(
-                            max_rep=1, list_dtype=dtype,
-                            opt_list=field.nullable,
-                            opt_elem=field.type.value_field.nullable)
-                        dtype = dtype.children[0]
-                    else:
-                        raise NotImplementedError(
-                            f"nested parquet column {name}")
+
+            def _mk_dec(j, leaf_dtype, **kw):
                 cmd = rgmd.column(j)
-                max_def = pq_schema.column(j).max_definition_level
                 if pq_schema.column(j).max_repetition_level > 0 \
-                        and not list_kw:
+                        and not kw.get("max_rep"):
                     raise NotImplementedError("nested parquet column")
                 start = cmd.dictionary_page_offset \
                     if cmd.dictionary_page_offset is not None \
                     else cmd.data_page_offset
                 t1 = _time.perf_counter()
                 raw = mv[start:start + cmd.total_compressed_size]
+                PHASE_STATS["io_s"] += _time.perf_counter() - t1
+                return _ChunkDecoder(
+                    raw, cmd.physical_type, leaf_dtype,
+                    pq_schema.column(j).max_definition_level,
+                    _codec(cmd.compression), cmd.num_values, ext, s, **kw)
+
+            for name in columns:
+                field = arrow_schema.field(name)
+                dtype = arrow_to_dtype(field.type)
+                j = name_to_idx.get(name)
                 t2 = _time.perf_counter()
-                PHASE_STATS["io_s"] += t2 - t1
-                dec = _ChunkDecoder(raw, cmd.physical_type, dtype, max_def,
-                                    _codec(cmd.compression), cmd.num_values,
-                                    ext, s, **list_kw)
-                cols.append(dec.decode())
+                if j is not None:
+                    cols.append(_mk_dec(j, dtype).decode())
+                elif dtype.id is TypeId.LIST \
+                        and not dtype.children[0].is_nested:
+                    cands = [jj for p, jj in name_to_idx.items()
+                             if p.startswith(name + ".")]
+                    if len(cands) != 1:
+                        raise NotImplementedError(
+                            f"nested parquet column {name}")
+                    dec = _mk_dec(
+                        cands[0], dtype.children[0], max_rep=1,
+                        list_dtype=dtype, opt_list=field.nullable,
+                        opt_elem=field.type.value_field.nullable)
+                    cols.append(dec.decode())
+                elif dtype.id is TypeId.STRUCT \
+                        and not any(c.is_nested for c in dtype.children):
+                    # device STRUCT decode: each leaf is a flat chunk
+                    # (dotted path); the struct's own validity comes from
+                    # the first leaf's def levels (struct-null rows have
+                    # def 0 on EVERY leaf)
+                    kids = []
+                    sv = None
+                    for ci, (cname, cdt) in enumerate(
+                            zip(dtype.field_names, dtype.children)):
+                        jj = name_to_idx.get(f"{name}.{cname}")
+                        if jj is None:
+                            raise NotImplementedError(
+                                f"struct leaf {name}.{cname}")
+                        dec = _mk_dec(jj, cdt,
+                                      capture_def=(ci == 0
+                                                   and field.nullable))
+                        kids.append(dec.decode())
+                        if ci == 0:
+                            sv = dec.captured_def
+                    n_rows = rgmd.num_rows
+                    validity = None
+                    null_count = 0
+                    if sv is not None:
+                        rv = (sv >= 1).to(torch.int32).contiguous()
+                        if not bool((rv == 1).all()):
+                            mask = torch.empty(mask_nbytes(n_rows),
+                                               dtype=torch.uint8,
+                                               device="cuda")
+                            ext.levels_to_mask(rv.data_ptr(), 1,
+                                               mask.data_ptr(), n_rows, s)
+                            validity = mask
+                            null_count = None
+                    cols.append(Column(
+                        dtype, n_rows,
+                        torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                        validity, None, null_count, tuple(kids)))
+                else:
+                    raise NotImplementedError(
+                        f"nested parquet column {name}")
                 PHASE_STATS["decode_s"] += _time.perf_counter() - t2
             rg_batches.append(ColumnBatch(cols, rgmd.num_rows))
     if len(rg_batches) == 1:
