@@ -905,6 +905,38 @@ def read_parquet_gpu(path: str, columns: List[str],
                         list_dtype=dtype, opt_list=field.nullable,
                         opt_elem=field.type.value_field.nullable)
                     cols.append(dec.decode())
+                elif dtype.id is TypeId.MAP \
+                        and not any(c.is_nested for c in dtype.children):
+                    # MAP = two rep-1 leaves (key, value) sharing the
+                    # entry offsets; decode each through the LIST path
+                    # and zip the entry children into the entry struct
+                    leaves = {p: jj for p, jj in name_to_idx.items()
+                              if p.startswith(name + ".")}
+                    jk = next((jj for p, jj in leaves.items()
+                               if p.endswith(".key")), None)
+                    jv = next((jj for p, jj in leaves.items()
+                               if p.endswith(".value")), None)
+                    if jk is None or jv is None:
+                        raise NotImplementedError(
+                            f"map leaf layout for {name}")
+                    kt, vt = dtype.children
+                    kl = _mk_dec(jk, kt, max_rep=1,
+                                 list_dtype=DType.list_(kt),
+                                 opt_list=field.nullable,
+                                 opt_elem=False).decode()
+                    vl = _mk_dec(jv, vt, max_rep=1,
+                                 list_dtype=DType.list_(vt),
+                                 opt_list=field.nullable,
+                                 opt_elem=field.type.item_field.nullable
+                                 ).decode()
+                    entry = Column(
+                        dtype.entry_dtype, kl.child.size,
+                        torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                        None, None, 0, (kl.child, vl.child))
+                    cols.append(Column(
+                        dtype, kl.size,
+                        torch.zeros(0, dtype=torch.uint8, device="cuda"),
+                        kl.validity, kl.offsets, kl._null_count, entry))
                 elif dtype.id is TypeId.STRUCT \
                         and not any(c.is_nested for c in dtype.children):
                     # device STRUCT decode: each leaf is a flat chunk

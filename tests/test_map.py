@@ -167,3 +167,34 @@ def test_array_contains_gpu_matches_cpu():
     g = q(sr.Session())
     c = q(sr.Session({"spark.rapids.sql.enabled": False}))
     assert g == c
+
+
+@pytest.mark.gpu
+def test_gpu_map_parquet_scan(tmp_path):
+    """Device MAP decode: key/value leaves through the LIST machinery,
+    shared entry offsets, null maps and null values."""
+    import pyarrow as pa
+    import pyarrow.parquet as pq
+
+    from spark_rapids_amd.types import DType as _DT
+
+    vals = [{"a": 1, "b": 2}, None, {}, {"c": None, "d": 9}] * 700
+    f = str(tmp_path / "m.parquet")
+    pq.write_table(pa.table({
+        "m": pa.array(vals, type=pa.map_(pa.string(), pa.int64())),
+        "k": pa.array(np.arange(2800, dtype=np.int64))}), f)
+    from spark_rapids_amd.io.parquet_gpu import read_parquet_gpu
+
+    batch = read_parquet_gpu(f, ["m", "k"])
+    got = batch.columns[0].cpu().to_pylist()
+    assert got == vals
+    # element_at on the GPU-scanned column, no fallback
+    from spark_rapids_amd.io.parquet import SCAN_STATS
+
+    s = sr.Session()
+    before = SCAN_STATS["fallback_files"]
+    out = s.read_parquet(f).select(
+        col("m").element_at("d").alias("d")).to_pydict()
+    assert SCAN_STATS["fallback_files"] == before, \
+        SCAN_STATS["last_fallback"]
+    assert out["d"] == [None, None, None, 9] * 700
