@@ -1535,6 +1535,12 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                           int(op.split(":", 1)[1]), s)
             allocs.append(("collect", out_dtype, False, col, None))
             continue
+        if op in ("min", "max") and vc is not None \
+                and vc.dtype.id is TypeId.STRING:
+            col = _gb_minmax_str(vc, row_gid, sel, n, ngroups,
+                                 op == "max", s)
+            allocs.append(("collect", out_dtype, False, col, None))
+            continue
         if op == "count" and vc is not None and not vc.dtype.is_fixed_width:
             # count over STRING/nested only needs the validity: swap in a
             # u8 non-null indicator column so no typed accumulator is hit
@@ -1754,6 +1760,33 @@ def _gb_percentile(vc: Column, row_gid: torch.Tensor, selp, n: int,
     ov = _alloc_mask(ngroups)
     ext.mask_from_nonzero(vcnt.data_ptr(), ov.data_ptr(), ngroups, s)
     return Column(DType.float64(), ngroups, out, ov, null_count=None)
+
+
+def _gb_minmax_str(vc: Column, row_gid: torch.Tensor, sel, n: int,
+                   ngroups: int, is_max: bool, s) -> Column:
+    """min/max over STRING values per group, sort-based: stable radix
+    sort of (gid, value) (string keys via the chunked big-endian LSD
+    machinery), then the head row of each gid segment is the extreme.
+    Nulls sort last so any non-null wins; an all-null group's head is a
+    null row and gathers as NULL (Spark ignore-null semantics)."""
+    gid_col = Column(DType.int32(), n, row_gid[:n], None, null_count=0)
+    if sel is not None:
+        vals = _gather_col(vc, sel, n, maybe_negative=False)
+    else:
+        vals = vc
+    pair = ColumnBatch([gid_col, vals], n)
+    perm = sort_order(pair, [0, 1], [False, is_max], [False, True])
+    pg = perm.data[:n].long()
+    sg = row_gid[:n].long()[pg] if sel is None else \
+        row_gid[:n].long()[pg]
+    head = torch.ones(n, dtype=torch.bool, device="cuda")
+    if n > 1:
+        head[1:] = sg[1:] != sg[:-1]
+    hp = head.nonzero(as_tuple=True)[0]
+    orig = pg[hp] if sel is None else sel[:n].long()[pg[hp]]
+    ridx = torch.full((ngroups,), -1, dtype=torch.int32, device="cuda")
+    ridx[sg[hp]] = orig.to(torch.int32)
+    return _gather_col(vc, ridx, ngroups, maybe_negative=True)
 
 
 def _gb_collect(vc: Column, row_gid: torch.Tensor, selp, n: int,

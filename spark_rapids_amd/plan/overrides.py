@@ -240,7 +240,7 @@ class Tagger:
                         continue
                     r = _NUMERIC.supports(t)
                     if t.id is TypeId.STRING and a.op not in (
-                            "count", "count_all"):
+                            "count", "count_all", "min", "max"):
                         reasons.append(
                             f"agg {a.op} over strings not on GPU yet")
                     elif r and a.op not in ("count", "count_all", "min",

@@ -640,3 +640,34 @@ def test_nested_loop_join_gpu_matches_cpu(how):
                 assert x == pytest.approx(y, rel=1e-9), (how, gr, cr)
             else:
                 assert x == y, (how, gr, cr)
+
+
+def test_group_min_max_strings_gpu():
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, min_, max_
+
+    rng = np.random.default_rng(41)
+    n = 30_000
+    words = ["", "a", "zz", "alpha", "beta", "Zeta", "ä", "m"]
+    data = {"k": [int(v) for v in rng.integers(0, 200, n)],
+            "s": [None if i % 11 == 0 else words[int(v)]
+                  for i, v in enumerate(rng.integers(0, 8, n))]}
+
+    def q(s):
+        df = s.create_dataframe(data)
+        return sorted(df.group_by("k")
+                      .agg(min_(col("s")), max_(col("s"))).collect())
+
+    sg = sr.Session()
+    df = sg.create_dataframe(data).group_by("k").agg(min_(col("s")))
+    assert "GpuHashAggregate" in df.physical_plan().tree_string()
+    g = q(sg)
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert g == c
+    # keyless reduction over strings
+    gk = sg.create_dataframe(data).agg(min_(col("s")),
+                                       max_(col("s"))).collect()
+    ck = sr.Session({"spark.rapids.sql.enabled": False}) \
+        .create_dataframe(data).agg(min_(col("s")),
+                                    max_(col("s"))).collect()
+    assert gk == ck
