@@ -1141,6 +1141,15 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                 lists = [list(dict.fromkeys(l)) for l in lists]
             out_cols.append(Column.from_pylist(lists, out_dtype))
             continue
+        if op in ("min", "max") and \
+                batch.columns[vidx].dtype.id is TypeId.STRING:
+            best: list = [None] * ngroups
+            fn = min if op == "min" else max
+            for g, v, ok in zip(codes, vc.to_pylist(), av):
+                if ok:
+                    best[g] = v if best[g] is None else fn(best[g], v)
+            out_cols.append(Column.from_pylist(best, out_dtype))
+            continue
         if batch.columns[vidx].dtype.id is TypeId.DECIMAL128 and \
                 op not in ("sum", "count", "count_all"):
             raise NotImplementedError(
