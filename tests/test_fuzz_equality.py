@@ -149,7 +149,8 @@ def test_fuzz_special_values_agg_sort(seed):
     def q(s):
         df = s.create_dataframe({k: list(v) for k, v in data.items()})
         return (df.group_by("k")
-                .agg(sum_(col("f")), count_star(), count(col("s")))
+                .agg(sum_(col("f")), count_star(), count(col("s")),
+                     min_(col("s")), max_(col("s")))
                 .sort("k"))
 
     assert_gpu_and_cpu_are_equal(q, rel=1e-6)
