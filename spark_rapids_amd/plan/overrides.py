@@ -232,8 +232,7 @@ class Tagger:
                         reasons += self.expr_reasons(a.child, cs)
                         continue
                     if a.op in ("collect_list", "collect_set"):
-                        if t.is_nested or (t.id is TypeId.STRING
-                                           and a.op == "collect_set"):
+                        if t.is_nested:
                             reasons.append(
                                 f"{a.op} over {t} not on GPU yet")
                         reasons += self.expr_reasons(a.child, cs)

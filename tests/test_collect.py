@@ -280,3 +280,27 @@ def test_nested_group_key_clear_error(cpu):
     df = cpu.create_dataframe({"p": [["a"], ["b"]]})
     with pytest.raises(NotImplementedError, match="nested"):
         df.group_by("p").agg(count_star())
+
+
+@pytest.mark.gpu
+def test_collect_set_strings_gpu_matches_cpu():
+    import numpy as np
+
+    import spark_rapids_amd as sr
+    from spark_rapids_amd import col, collect_set
+
+    rng = np.random.default_rng(9)
+    n = 20_000
+    words = ["a", "bb", "ccc", "", "zz", "Aa"]
+    data = {"k": [int(v) for v in rng.integers(0, 100, n)],
+            "s": [None if i % 13 == 0 else words[int(v)]
+                  for i, v in enumerate(rng.integers(0, 6, n))]}
+
+    def q(s):
+        df = s.create_dataframe(data)
+        out = df.group_by("k").agg(collect_set(col("s"))).collect()
+        return sorted((k, sorted(v)) for k, v in out)
+
+    g = q(sr.Session())
+    c = q(sr.Session({"spark.rapids.sql.enabled": False}))
+    assert g == c
