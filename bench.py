@@ -93,7 +93,7 @@ def main():
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=3)
     ap.add_argument("--warmup", type=int, default=1)
-    ap.add_argument("--rows-per-gpu", type=int, default=40_000_000)
+    ap.add_argument("--rows-per-gpu", type=int, default=60_000_000)
     ap.add_argument("--partitions", type=int, default=8,
                     help="fact parquet files per rank")
     ap.add_argument("--data-dir", default=None)
