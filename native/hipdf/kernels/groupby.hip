@@ -230,6 +230,11 @@ struct AggDesc {
   int op;             // GbOp
   int type;           // HType of the value column
   int acc_is_double;  // accumulator: double (1) or int64 (0)
+  int skip_cnt;       // sum/min/max over a non-null column: the count
+                      // atomic is pure overhead (group validity is
+                      // implied by group existence) — skip it in the
+                      // global-atomic path; the LDS path still counts
+                      // (its flush is gated on lcnt)
   const void* vals;
   const uint64_t* valid;
   void* acc;          // [ngroups] double or int64
@@ -310,7 +315,8 @@ __global__ void k_gb_agg_multi(const AggDesc* __restrict__ aggs, int naggs,
         continue;
       }
       if (!valid_bit(d.valid, i)) continue;
-      atomicAdd((unsigned long long*)&cnt_p[g], 1ull);
+      if (USE_LDS || !d.skip_cnt)
+        atomicAdd((unsigned long long*)&cnt_p[g], 1ull);
       if (d.op == GB_COUNT) continue;
       if (d.acc_is_double) {
         double* acc_p = USE_LDS ? (double*)base : (double*)d.acc + rep_off;

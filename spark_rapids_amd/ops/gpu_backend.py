@@ -1518,28 +1518,28 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                     ext.gb_sum_i128(vc.data.data_ptr(), _ptr(vc.validity),
                                     row_gid.data_ptr(), selp, acc.data_ptr(),
                                     cnt.data_ptr(), n, s)
-            allocs.append(("sum_d128", out_dtype, False, acc, cnt))
+            allocs.append(("sum_d128", out_dtype, False, acc, cnt, 0))
             continue
         if op in ("collect_list", "collect_set"):
             col = _gb_collect(vc, row_gid, selp, n, ngroups, out_dtype,
                               op == "collect_set", s)
-            allocs.append(("collect", out_dtype, False, col, None))
+            allocs.append(("collect", out_dtype, False, col, None, 0))
             continue
         if op.startswith("percentile:"):
             col = _gb_percentile(vc, row_gid, selp, n, ngroups,
                                  float(op.split(":", 1)[1]), s)
-            allocs.append(("collect", out_dtype, False, col, None))
+            allocs.append(("collect", out_dtype, False, col, None, 0))
             continue
         if op.startswith("hll:"):
             col = _gb_hll(vc, row_gid, selp, n, ngroups,
                           int(op.split(":", 1)[1]), s)
-            allocs.append(("collect", out_dtype, False, col, None))
+            allocs.append(("collect", out_dtype, False, col, None, 0))
             continue
         if op in ("min", "max") and vc is not None \
                 and vc.dtype.id is TypeId.STRING:
             col = _gb_minmax_str(vc, row_gid, sel, n, ngroups,
                                  op == "max", s)
-            allocs.append(("collect", out_dtype, False, col, None))
+            allocs.append(("collect", out_dtype, False, col, None, 0))
             continue
         if op == "count" and vc is not None and not vc.dtype.is_fixed_width:
             # count over STRING/nested only needs the validity: swap in a
@@ -1559,9 +1559,15 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         if op not in ("count", "count_all"):
             ext.gb_acc_init(_GB[op], acc.data_ptr(), acc_is_double,
                             nrep * ngroups, s)
-        allocs.append((op, out_dtype, acc_is_double, acc, cnt))
+        # sum/min/max over a non-null column: group validity is implied
+        # by group existence, so the per-row count atomic is pure traffic
+        # (roofline: k_gb_agg_multi writes were ~2x the accumulator data)
+        skip_cnt = 1 if (op in ("sum", "min", "max") and vc is not None
+                         and vc.validity is None and n > 0) else 0
+        allocs.append((op, out_dtype, acc_is_double, acc, cnt,
+                       skip_cnt))
         blobs.append(struct.pack(
-            "<iiiiqqqq", _GB[op], t, acc_is_double, 0,
+            "<iiiiqqqq", _GB[op], t, acc_is_double, skip_cnt,
             _ptr(vc.data if vc is not None else None),
             _ptr(vc.validity if vc is not None else None),
             acc.data_ptr(), cnt.data_ptr()))
@@ -1571,13 +1577,13 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         ext.gb_agg_multi(desc.data_ptr(), len(blobs), row_gid.data_ptr(),
                          selp, ngroups, nrep, n, s)
         if nrep > 1:
-            for op, _, acc_is_double, acc, cnt in allocs:
+            for op, _, acc_is_double, acc, cnt, _skip in allocs:
                 if op in ("sum_d128", "collect"):
                     continue
                 ext.gb_reduce_reps(_GB.get(op, 0), acc.data_ptr(),
                                    1 if acc_is_double else 0,
                                    cnt.data_ptr(), ngroups, nrep, s)
-    for op, out_dtype, acc_is_double, acc, cnt in allocs:
+    for op, out_dtype, acc_is_double, acc, cnt, skip_cnt in allocs:
         if op == "collect":
             out_cols.append(acc)
             continue
@@ -1599,10 +1605,15 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
                      out_data.data_ptr(), ngroups, s)
         else:
             out_data = data
-        ov = _alloc_mask(ngroups)
-        ext.mask_from_nonzero(cnt.data_ptr(), ov.data_ptr(), ngroups, s)
-        out_cols.append(Column(out_dtype, ngroups, out_data, ov,
-                               null_count=None))
+        if skip_cnt:
+            # every observed group has >= 1 contributing row
+            out_cols.append(Column(out_dtype, ngroups, out_data, None,
+                                   null_count=0))
+        else:
+            ov = _alloc_mask(ngroups)
+            ext.mask_from_nonzero(cnt.data_ptr(), ov.data_ptr(), ngroups, s)
+            out_cols.append(Column(out_dtype, ngroups, out_data, ov,
+                                   null_count=None))
     if dense is not None:
         return _compact_dense(out_cols, allocs, keys, dense_info, ngroups,
                               row_gid, n, s)
@@ -1665,7 +1676,7 @@ def _compact_dense(out_cols, allocs, keys, dense_info, ngroups, row_gid,
     surviving ids (kmin + (g // stride) % range per key)."""
     mins, ranges, strides = dense_info
     counts = None
-    for (op, _, _, acc, cnt) in allocs:
+    for (op, _, _, acc, cnt, _sk) in allocs:
         if op == "count_all":
             counts = cnt
             break
