@@ -144,10 +144,16 @@ def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
     return backend_for(*batch.columns, mask).apply_boolean_mask(batch, mask)
 
 
-def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> ColumnBatch:
+def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False,
+           negatives: bool = True) -> ColumnBatch:
     """Take rows by int32 index column; negative index -> null row
-    (OutOfBoundsPolicy.NULLIFY analogue for join gather maps)."""
-    return backend_for(*batch.columns, indices).gather(batch, indices, check_bounds)
+    (OutOfBoundsPolicy.NULLIFY analogue for join gather maps).
+    negatives=False: the caller guarantees no -1 indices (inner/cross
+    maps), so non-null source columns stay mask-free — otherwise a
+    validity mask re-materializes on every gathered column and defeats
+    the downstream non-null fast paths."""
+    return backend_for(*batch.columns, indices).gather(
+        batch, indices, check_bounds, negatives)
 
 
 def murmur3_hash(cols: Sequence[Column], seed: int = 42) -> Column:

@@ -840,8 +840,11 @@ def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
     return _gather_idx(batch, idx, np.ones(len(idx), dtype=bool))
 
 
-def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> ColumnBatch:
+def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False,
+           negatives: bool = True) -> ColumnBatch:
     idx = _vals(indices).astype(np.int64)
+    if not negatives and not check_bounds:
+        return _gather_idx(batch, idx, np.ones(len(idx), dtype=bool))
     ok = idx >= 0
     if check_bounds:
         ok &= idx < batch.num_rows

@@ -1127,9 +1127,10 @@ def apply_boolean_mask(batch: ColumnBatch, mask: Column) -> ColumnBatch:
     return _gather_by_idx(batch, idx, idx.numel(), maybe_negative=False)
 
 
-def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False) -> ColumnBatch:
+def gather(batch: ColumnBatch, indices: Column, check_bounds: bool = False,
+           negatives: bool = True) -> ColumnBatch:
     return _gather_by_idx(batch, indices.data, indices.size,
-                          maybe_negative=True)
+                          maybe_negative=negatives)
 
 
 def _gather_col(c: Column, idx: torch.Tensor, n_out: int,

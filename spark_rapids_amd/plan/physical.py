@@ -600,12 +600,13 @@ class HashJoinExec(PhysicalExec):
             lmap, rmap = ops.join_gather_maps(lbatch, rtable, lkidx, rkidx,
                                               self.how, right_matched)
             if self.how in ("semi", "anti"):
-                out = ops.gather(lbatch, lmap)
+                out = ops.gather(lbatch, lmap, negatives=False)
                 if out.num_rows:
                     yield out
                 continue
-            lout = ops.gather(lbatch, lmap)
-            rout = ops.gather(rtable, rmap)
+            lout = ops.gather(lbatch, lmap, negatives=False)
+            rout = ops.gather(rtable, rmap,
+                              negatives=self.how in ("left", "full"))
             if lout.num_rows:
                 yield ColumnBatch(lout.columns +
                                   self._right_out(rout), lout.num_rows)
@@ -666,12 +667,13 @@ class HashJoinExec(PhysicalExec):
                 lmap, rmap = ops.join_gather_maps(lb, rb, lkidx, rkidx,
                                                   self.how, right_matched)
                 if self.how in ("semi", "anti"):
-                    out = ops.gather(lb, lmap)
+                    out = ops.gather(lb, lmap, negatives=False)
                     if out.num_rows:
                         yield out
                 else:
-                    lout = ops.gather(lb, lmap)
-                    rout = ops.gather(rb, rmap)
+                    lout = ops.gather(lb, lmap, negatives=False)
+                    rout = ops.gather(
+                        rb, rmap, negatives=self.how in ("left", "full"))
                     if lout.num_rows:
                         yield ColumnBatch(
                             lout.columns + self._right_out(rout),
@@ -695,8 +697,8 @@ class HashJoinExec(PhysicalExec):
                                           None)
         if lmap.size == 0:
             return lmap, rmap
-        lout = ops.gather(lb, lmap)
-        rout = ops.gather(rb, rmap)
+        lout = ops.gather(lb, lmap, negatives=False)
+        rout = ops.gather(rb, rmap, negatives=False)
         pair = ColumnBatch(list(lout.columns) + list(rout.columns),
                            lout.num_rows)
         mask = self.condition.eval(pair, self._pair_schema())
@@ -739,8 +741,8 @@ class HashJoinExec(PhysicalExec):
             else:
                 right_matched[idx.cpu().numpy()] = True
         if lmap_f.size:
-            lout = ops.gather(lb, lmap_f)
-            rout = ops.gather(rb, rmap_f)
+            lout = ops.gather(lb, lmap_f, negatives=False)
+            rout = ops.gather(rb, rmap_f, negatives=False)
             yield ColumnBatch(list(lout.columns) + self._right_out(rout),
                               lout.num_rows)
         if how in ("left", "full"):
@@ -907,8 +909,8 @@ class NestedLoopJoinExec(HashJoinExec):
                 e0 = min(nl, s0 + rows_per)
                 chunk = _slice_rows(lbatch, s0, e0)
                 lmap, rmap = self._cross_maps(e0 - s0, nr)
-                lout = ops.gather(chunk, lmap)
-                rout = ops.gather(rtable, rmap)
+                lout = ops.gather(chunk, lmap, negatives=False)
+                rout = ops.gather(rtable, rmap, negatives=False)
                 pair = ColumnBatch(list(lout.columns) + list(rout.columns),
                                    lout.num_rows)
                 mask = self.condition.eval(pair, self._pair_schema())
@@ -1005,8 +1007,8 @@ class CrossJoinExec(PhysicalExec):
                 ri = np.tile(np.arange(nr, dtype=np.int32), nl)
                 lmap = Column.from_numpy(li)
                 rmap = Column.from_numpy(ri)
-            lout = ops.gather(lbatch, lmap)
-            rout = ops.gather(rtable, rmap)
+            lout = ops.gather(lbatch, lmap, negatives=False)
+            rout = ops.gather(rtable, rmap, negatives=False)
             yield ColumnBatch(lout.columns + rout.columns, total)
 
 
