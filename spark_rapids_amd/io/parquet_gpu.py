@@ -7,11 +7,15 @@ Reference analogue: GpuParquetScan + libcudf's parquet decode kernels
 (CpuCompressionConfig, GpuParquetScan.scala:1549) is the model for the
 decompress-on-host/decode-on-device split used here.
 
-Scope: flat schemas; INT32/INT64/FLOAT/DOUBLE/BOOLEAN (+date32/timestamp/
-decimal on those physical types, including width-widening INT32-physical
-decimals) PLAIN, dictionary or DELTA_BINARY_PACKED; BYTE_ARRAY strings
-PLAIN or dictionary-encoded. Anything else raises NotImplementedError and
-the caller falls back to the CPU (hybrid) reader per file.
+Scope: flat schemas plus one-level LIST (repetition levels decoded on
+device), STRUCT (dotted leaves, struct validity from def levels) and MAP
+(key/value leaves through the LIST path); INT32/INT64/FLOAT/DOUBLE/
+BOOLEAN (+date32/timestamp/decimal on those physical types, including
+width-widening INT32-physical decimals) PLAIN, dictionary or
+DELTA_BINARY_PACKED; BYTE_ARRAY strings PLAIN, dictionary,
+DELTA_BYTE_ARRAY or DELTA_LENGTH_BYTE_ARRAY. Anything else raises
+NotImplementedError and the caller falls back to the CPU (hybrid) reader
+per file.
 """
 from __future__ import annotations
 
