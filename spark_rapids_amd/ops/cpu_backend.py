@@ -1194,8 +1194,21 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         elif op in ("min", "max"):
             init = np.inf if op == "min" else -np.inf
             res = np.full(ngroups, init)
-            ufunc = np.minimum if op == "min" else np.maximum
+            # Spark float ordering: NaN is the GREATEST value — min must
+            # prefer any non-NaN (fmin drops NaN), max must return NaN
+            # when present (maximum propagates it); all-NaN groups stay
+            # NaN via fmin(nan,nan)=nan. Matches the GPU CAS ordering.
+            ufunc = np.fmin if op == "min" else np.maximum
             ufunc.at(res, gc, af)
+            if op == "min":
+                nanm = np.isnan(af)
+                if nanm.any():
+                    # groups whose only values are NaN: fmin left init
+                    only_nan = np.full(ngroups, True)
+                    only_nan[gc[~nanm]] = False
+                    seen = np.full(ngroups, False)
+                    seen[gc] = True
+                    res[only_nan & seen] = np.nan
         else:
             raise NotImplementedError(f"cpu groupby agg {op}")
         if out_dtype.is_floating:
