@@ -361,8 +361,29 @@ __global__ void k_gb_reduce_reps(int op, ACC* __restrict__ acc,
       int64_t cr = cnt[(size_t)r * ngroups + g];
       ACC vr = acc[(size_t)r * ngroups + g];
       if (op == GB_SUM) v = v + vr;
-      else if (op == GB_MIN) v = (cr && (!c || vr < v)) ? vr : v;
-      else if (op == GB_MAX) v = (cr && (!c || vr > v)) ? vr : v;
+      else if (op == GB_MIN) {
+        // integers: the init identity (INT64_MAX) can never win a min,
+        // so the fold needs no cnt — which also makes it correct for
+        // skip_cnt aggs whose cnt replicas are all zero. Floats keep
+        // cnt (empty-replica +inf is ambiguous with a real +inf when
+        // NaN values exist) and compare in Spark order (NaN greatest).
+        bool take;
+        if constexpr (std::is_floating_point<ACC>::value)
+          take = cr && (!c || vr < v ||
+                        (isnan((double)v) && !isnan((double)vr)));
+        else
+          take = vr < v;
+        if (take) v = vr;
+      }
+      else if (op == GB_MAX) {
+        bool take;
+        if constexpr (std::is_floating_point<ACC>::value)
+          take = cr && (!c || (vr > v && !isnan((double)v)) ||
+                        isnan((double)vr));
+        else
+          take = vr > v;
+        if (take) v = vr;
+      }
       else if (op == GB_BITAND)
         v = (ACC)((int64_t)v & (int64_t)vr);
       else if (op == GB_BITOR)

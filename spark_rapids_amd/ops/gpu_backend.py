@@ -1560,11 +1560,16 @@ def group_by_aggregate(batch: ColumnBatch, key_idx: List[int],
         if op not in ("count", "count_all"):
             ext.gb_acc_init(_GB[op], acc.data_ptr(), acc_is_double,
                             nrep * ngroups, s)
-        # sum/min/max over a non-null column: group validity is implied
-        # by group existence, so the per-row count atomic is pure traffic
-        # (roofline: k_gb_agg_multi writes were ~2x the accumulator data)
-        skip_cnt = 1 if (op in ("sum", "min", "max") and vc is not None
-                         and vc.validity is None and n > 0) else 0
+        # sum / integer min-max over a non-null column: group validity
+        # is implied by group existence, so the per-row count atomic is
+        # pure traffic (roofline: k_gb_agg_multi writes were ~2x the
+        # accumulator data). FLOAT min/max keep the count: the replica
+        # fold needs it to tell an empty replica's ±inf init from a real
+        # ±inf when NaN values are present (Spark: NaN is greatest).
+        skip_cnt = 1 if (vc is not None and vc.validity is None and n > 0
+                         and (op == "sum"
+                              or (op in ("min", "max")
+                                  and not vc.dtype.is_floating))) else 0
         allocs.append((op, out_dtype, acc_is_double, acc, cnt,
                        skip_cnt))
         blobs.append(struct.pack(
