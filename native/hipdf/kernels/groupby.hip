@@ -171,8 +171,16 @@ __device__ __forceinline__ void acc_atomic<double>(int op, double* addr,
 
 template <typename ACC>
 __device__ __forceinline__ ACC acc_init(int op) {
-  if (op == GB_MIN) return std::numeric_limits<ACC>::max();
-  if (op == GB_MAX) return std::numeric_limits<ACC>::lowest();
+  // float identities must be ±infinity (not ±DBL_MAX): a group whose
+  // real extreme IS ±inf must beat the init under plain compare
+  if (op == GB_MIN)
+    return std::numeric_limits<ACC>::has_infinity
+               ? std::numeric_limits<ACC>::infinity()
+               : std::numeric_limits<ACC>::max();
+  if (op == GB_MAX)
+    return std::numeric_limits<ACC>::has_infinity
+               ? -std::numeric_limits<ACC>::infinity()
+               : std::numeric_limits<ACC>::lowest();
   if (op == GB_BITAND) return (ACC)-1;  // all ones identity
   return (ACC)0;
 }
