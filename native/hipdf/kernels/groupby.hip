@@ -161,8 +161,9 @@ __device__ __forceinline__ void acc_atomic<double>(int op, double* addr,
   do {
     assumed = old;
     double cur = __longlong_as_double((long long)assumed);
-    bool better = op == GB_MIN ? (v < cur || (isnan(cur) && !isnan(v)))
-                               : (v > cur && !isnan(cur)) || (isnan(v));
+    bool better = op == GB_MIN
+                      ? (!isnan(v) && (isnan(cur) || v < cur))
+                      : ((v > cur && !isnan(cur)) || (isnan(v) && !isnan(cur)));
     // Spark ordering: NaN is the greatest value
     if (!better) break;
     old = atomicCAS(up, assumed, (unsigned long long)__double_as_longlong(v));
@@ -174,8 +175,11 @@ __device__ __forceinline__ ACC acc_init(int op) {
   // float identities must be ±infinity (not ±DBL_MAX): a group whose
   // real extreme IS ±inf must beat the init under plain compare
   if (op == GB_MIN)
+    // float min identity is NaN: under Spark's total order NaN is the
+    // GREATEST value (above +inf), so min{all NaN} must stay NaN while
+    // any real value (incl. +inf) replaces it
     return std::numeric_limits<ACC>::has_infinity
-               ? std::numeric_limits<ACC>::infinity()
+               ? std::numeric_limits<ACC>::quiet_NaN()
                : std::numeric_limits<ACC>::max();
   if (op == GB_MAX)
     return std::numeric_limits<ACC>::has_infinity
@@ -377,8 +381,8 @@ __global__ void k_gb_reduce_reps(int op, ACC* __restrict__ acc,
         // NaN values exist) and compare in Spark order (NaN greatest).
         bool take;
         if constexpr (std::is_floating_point<ACC>::value)
-          take = cr && (!c || vr < v ||
-                        (isnan((double)v) && !isnan((double)vr)));
+          take = cr && (!c || (!isnan((double)vr) &&
+                               (isnan((double)v) || vr < v)));
         else
           take = vr < v;
         if (take) v = vr;
@@ -387,7 +391,7 @@ __global__ void k_gb_reduce_reps(int op, ACC* __restrict__ acc,
         bool take;
         if constexpr (std::is_floating_point<ACC>::value)
           take = cr && (!c || (vr > v && !isnan((double)v)) ||
-                        isnan((double)vr));
+                        (isnan((double)vr) && !isnan((double)v)));
         else
           take = vr > v;
         if (take) v = vr;
